@@ -1,1 +1,1 @@
-from . import hashing  # noqa: F401
+from . import aggregate, copying, hashing, join  # noqa: F401

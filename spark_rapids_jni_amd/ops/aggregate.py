@@ -1,0 +1,159 @@
+"""Hash group-by aggregation (COUNT/SUM/MIN/MAX) on GPU.
+
+Spark semantics: nulls form their own group (null-safe key equality);
+SUM/MIN/MAX skip nulls (all-null group -> null result); COUNT(col) counts
+non-null, COUNT(*) counts rows. Integer SUM accumulates in int64, float in
+float64 (Spark's widened accumulators). The reference delegates this to cudf
+groupby + Aggregation64Utils overflow helpers; here it
+is a first-class MI355X kernel (src/gpu/hashtable.hip groupby_kernel).
+"""
+import struct
+from enum import IntEnum
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+from .. import _native
+from ..columnar import Column, DType, Table, pack_descriptors
+from .copying import gather_column
+
+
+class Agg(IntEnum):
+    COUNT_ALL = 0
+    COUNT_VALID = 1
+    SUM = 2
+    MIN = 4
+    MAX = 5
+
+
+_AGGDESC_FMT = "<iiQQQ"  # op, in_dtype, data, valid, state
+_AGGDESC_SZ = struct.calcsize(_AGGDESC_FMT)
+
+
+def _next_pow2(n):
+    p = 1
+    while p < n:
+        p <<= 1
+    return p
+
+
+def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
+            num_groups_hint: Optional[int] = None):
+    """Group rows of `keys` and aggregate.
+
+    aggs: list of (Agg, value_column_or_None). Returns
+    (key_table, [result Column, ...]) with one row per group.
+    """
+    kcols = keys.columns if isinstance(keys, Table) else (
+        [keys] if isinstance(keys, Column) else list(keys))
+    n = kcols[0].size
+    assert n < 2**31, "group-by input capped at 2^31-1 rows per batch"
+    g = _native.gpu()
+    stream = _native.current_stream()
+    dev = kcols[0].device
+
+    cap_groups = num_groups_hint if num_groups_hint else n
+    capacity = max(_next_pow2(min(cap_groups, n) * 2), 64)
+    slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
+
+    # native agg list = user aggs + a hidden COUNT_VALID per nullable SUM col
+    # (to derive the all-null-group -> null result in the SAME compaction pass,
+    # since compaction order is nondeterministic).
+    native: List[Tuple[int, Optional[Column], torch.Tensor]] = []
+    metas = []  # (user op, is_float, native_index, hidden_count_index|None)
+
+    def add_native(native_op, col, st):
+        native.append((native_op, col, st))
+        return len(native) - 1
+
+    for op, col in aggs:
+        if op in (Agg.COUNT_ALL, Agg.COUNT_VALID):
+            st = torch.zeros(capacity, dtype=torch.int64, device=dev)
+            idx = add_native(int(op), col, st)
+            metas.append((op, False, idx, None))
+            continue
+        assert col is not None
+        is_float = col.dtype in (DType.FLOAT32, DType.FLOAT64)
+        if op == Agg.SUM:
+            st = torch.zeros(capacity, dtype=torch.float64 if is_float
+                             else torch.int64, device=dev)
+            idx = add_native(3 if is_float else 2, col, st)
+            hidden = None
+            stc = torch.zeros(capacity, dtype=torch.int64, device=dev)
+            hidden = add_native(1, col, stc)
+            metas.append((op, is_float, idx, hidden))
+        elif op == Agg.MIN:
+            st = torch.full((capacity,), float("inf") if is_float else 2**63 - 1,
+                            dtype=torch.float64 if is_float else torch.int64,
+                            device=dev)
+            idx = add_native(6 if is_float else 4, col, st)
+            stc = torch.zeros(capacity, dtype=torch.int64, device=dev)
+            metas.append((op, is_float, idx, add_native(1, col, stc)))
+        else:  # MAX
+            st = torch.full((capacity,), float("-inf") if is_float else -2**63,
+                            dtype=torch.float64 if is_float else torch.int64,
+                            device=dev)
+            idx = add_native(7 if is_float else 5, col, st)
+            stc = torch.zeros(capacity, dtype=torch.int64, device=dev)
+            metas.append((op, is_float, idx, add_native(1, col, stc)))
+
+    naggs = len(native)
+    raw = bytearray(max(naggs, 1) * _AGGDESC_SZ)
+    for i, (native_op, col, st) in enumerate(native):
+        struct.pack_into(
+            _AGGDESC_FMT, raw, i * _AGGDESC_SZ, native_op,
+            int(col.dtype) if col is not None else 0,
+            col.data.data_ptr() if col is not None else 0,
+            (col.validity.data_ptr() if col is not None and col.validity is not None
+             else 0),
+            st.data_ptr())
+    agg_desc = torch.frombuffer(raw, dtype=torch.uint8).to(dev)
+
+    kdesc, ktop, keep = pack_descriptors(kcols)
+    g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n, slots.data_ptr(),
+              capacity, agg_desc.data_ptr(), naggs, stream)
+
+    counter = torch.zeros(1, dtype=torch.int64, device=dev)
+    out_repr = torch.empty(capacity, dtype=torch.int64, device=dev)
+    out_agg = torch.empty(max(naggs, 1) * capacity, dtype=torch.int64, device=dev)
+    g.groupby_compact(slots.data_ptr(), capacity, agg_desc.data_ptr(), naggs,
+                      counter.data_ptr(), out_repr.data_ptr(), out_agg.data_ptr(),
+                      capacity, stream)
+    ngroups = int(counter.item())
+    out_repr = out_repr[:ngroups]
+
+    def agg_vals(idx):
+        return out_agg[idx * capacity:idx * capacity + ngroups]
+
+    key_out = Table([gather_column(c, out_repr) for c in kcols])
+    results: List[Column] = []
+    for (op, is_float, idx, hidden) in metas:
+        vals = agg_vals(idx).clone()
+        if op in (Agg.COUNT_ALL, Agg.COUNT_VALID):
+            results.append(Column(DType.INT64, ngroups, vals))
+        else:  # SUM / MIN / MAX: null result for all-null groups
+            cnt = agg_vals(hidden)
+            validity = _validity_from_bool(cnt > 0)
+            if is_float:
+                out_vals = vals.view(torch.float64).clone()
+                out_vals[cnt == 0] = 0.0
+                results.append(Column(DType.FLOAT64, ngroups, out_vals, validity,
+                                      null_count=None))
+            else:
+                out_vals = vals.clone()
+                out_vals[cnt == 0] = 0
+                results.append(Column(DType.INT64, ngroups, out_vals, validity,
+                                      null_count=None))
+    return key_out, results
+
+
+def _validity_from_bool(flags: torch.Tensor):
+    n = flags.numel()
+    dev = flags.device
+    nbytes = ((n + 63) // 64) * 8
+    bits = torch.zeros(nbytes * 8, dtype=torch.bool, device=dev)
+    bits[:n] = flags
+    weights = torch.tensor([1, 2, 4, 8, 16, 32, 64, 128], dtype=torch.uint8,
+                           device=dev)
+    packed = (bits.view(-1, 8).to(torch.uint8) * weights).sum(1, dtype=torch.uint8)
+    return packed

@@ -1,0 +1,154 @@
+"""Composable join building blocks.
+
+Java API parity: com.nvidia.spark.rapids.jni.JoinPrimitives
+(reference join_primitives.hpp:64-237 / JoinPrimitives.java:80-302):
+hash_inner_join producing gather maps, plus gather-map algebra
+(make_left_outer / make_full_outer / semi / anti).
+
+MI355X design: a multimap of packed (fingerprint32|row+1) u64 slots at 50%
+load, linear probing (src/gpu/hashtable.hip). Build side is capped at 2^31-1
+rows (the reference has the same per-column row cap); probe side is int64.
+"""
+from typing import List, Optional, Sequence, Tuple, Union
+
+import torch
+
+from .. import _native
+from ..columnar import Column, Table, pack_descriptors
+
+
+def _keys(x) -> List[Column]:
+    if isinstance(x, Table):
+        return x.columns
+    if isinstance(x, Column):
+        return [x]
+    return list(x)
+
+
+def _next_pow2(n: int) -> int:
+    p = 1
+    while p < n:
+        p <<= 1
+    return p
+
+
+class HashJoinTable:
+    """Build-side hash table reusable across probes (reference: the build/probe
+    split of hash_inner_join)."""
+
+    def __init__(self, build_keys: List[Column], slots: torch.Tensor,
+                 capacity: int, keep):
+        self.build_keys = build_keys
+        self.slots = slots
+        self.capacity = capacity
+        self._keep = keep
+        self.num_build_rows = build_keys[0].size
+
+    @staticmethod
+    def build(keys: Union[Column, Table, Sequence[Column]]) -> "HashJoinTable":
+        cols = _keys(keys)
+        n = cols[0].size
+        assert n < 2**31, "build side capped at 2^31-1 rows (chunk the build)"
+        g = _native.gpu()
+        dev = cols[0].device
+        capacity = max(_next_pow2(n * 2), 64)
+        slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
+        desc, top, keep = pack_descriptors(cols)
+        g.join_build(desc.data_ptr(), top.data_ptr(), len(cols), n,
+                     slots.data_ptr(), capacity, _native.current_stream())
+        return HashJoinTable(cols, slots, capacity, (desc, top, keep))
+
+    def _descs(self, probe_cols):
+        bdesc, btop, bkeep = pack_descriptors(self.build_keys)
+        pdesc, ptop, pkeep = pack_descriptors(probe_cols)
+        return bdesc, btop, pdesc, ptop, (bkeep, pkeep, bdesc, btop, pdesc, ptop)
+
+    def inner_join(self, probe: Union[Column, Table, Sequence[Column]],
+                   out_hint: Optional[int] = None,
+                   track_build_matches: bool = False):
+        """Returns (build_idx int32, probe_idx int64[, build_matched uint8])."""
+        pcols = _keys(probe)
+        nprobe = pcols[0].size
+        g = _native.gpu()
+        stream = _native.current_stream()
+        dev = pcols[0].device
+        bdesc, btop, pdesc, ptop, keep = self._descs(pcols)
+        counter = torch.zeros(1, dtype=torch.int64, device=dev)
+        if out_hint is None:
+            g.join_probe_count(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
+                               ptop.data_ptr(), len(pcols), nprobe,
+                               self.slots.data_ptr(), self.capacity,
+                               counter.data_ptr(), stream)
+            total = int(counter.item())
+            counter.zero_()
+        else:
+            total = out_hint
+        out_build = torch.empty(max(total, 1), dtype=torch.int32, device=dev)
+        out_probe = torch.empty(max(total, 1), dtype=torch.int64, device=dev)
+        matched = (torch.zeros(self.num_build_rows, dtype=torch.uint8, device=dev)
+                   if track_build_matches else None)
+        g.join_probe_fill(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
+                          ptop.data_ptr(), len(pcols), nprobe,
+                          self.slots.data_ptr(), self.capacity, counter.data_ptr(),
+                          out_build.data_ptr(), out_probe.data_ptr(), total,
+                          matched.data_ptr() if matched is not None else 0, stream)
+        actual = int(counter.item())
+        if actual > total:
+            # hint was too small: rerun with the exact size
+            return self.inner_join(probe, out_hint=actual,
+                                   track_build_matches=track_build_matches)
+        out_build = out_build[:actual]
+        out_probe = out_probe[:actual]
+        if track_build_matches:
+            return out_build, out_probe, matched
+        return out_build, out_probe
+
+    def semi_join(self, probe, anti: bool = False) -> torch.Tensor:
+        """Left semi/anti join: probe-side row indices with (no) match."""
+        pcols = _keys(probe)
+        nprobe = pcols[0].size
+        g = _native.gpu()
+        stream = _native.current_stream()
+        dev = pcols[0].device
+        bdesc, btop, pdesc, ptop, keep = self._descs(pcols)
+        counter = torch.zeros(1, dtype=torch.int64, device=dev)
+        out = torch.empty(nprobe, dtype=torch.int64, device=dev)
+        g.join_semi(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
+                    ptop.data_ptr(), len(pcols), nprobe, self.slots.data_ptr(),
+                    self.capacity, counter.data_ptr(), out.data_ptr(), nprobe,
+                    1 if anti else 0, stream)
+        return out[:int(counter.item())]
+
+
+def hash_inner_join(build, probe) -> Tuple[torch.Tensor, torch.Tensor]:
+    """reference join_primitives.hpp:87 hash_inner_join"""
+    return HashJoinTable.build(build).inner_join(probe)
+
+
+def make_left_outer(probe_size: int, build_idx: torch.Tensor,
+                    probe_idx: torch.Tensor):
+    """Extend an inner-join gather map pair to LEFT OUTER (probe side = left):
+    unmatched probe rows appear with build index -1 (null gather).
+    reference join_primitives.hpp:130 make_left_outer."""
+    dev = probe_idx.device
+    matched = torch.zeros(probe_size, dtype=torch.bool, device=dev)
+    matched[probe_idx] = True
+    unmatched = torch.nonzero(~matched, as_tuple=False).view(-1)
+    lo_build = torch.cat([build_idx.long(),
+                          torch.full((unmatched.numel(),), -1, dtype=torch.int64,
+                                     device=dev)])
+    lo_probe = torch.cat([probe_idx, unmatched])
+    return lo_build, lo_probe
+
+
+def make_full_outer(build_matched: torch.Tensor, build_idx: torch.Tensor,
+                    probe_idx: torch.Tensor, probe_size: int):
+    """FULL OUTER: left-outer plus unmatched build rows with probe index -1.
+    reference join_primitives.hpp:150 make_full_outer."""
+    lo_build, lo_probe = make_left_outer(probe_size, build_idx, probe_idx)
+    unmatched_b = torch.nonzero(build_matched == 0, as_tuple=False).view(-1)
+    fo_build = torch.cat([lo_build, unmatched_b])
+    fo_probe = torch.cat([lo_probe,
+                          torch.full((unmatched_b.numel(),), -1,
+                                     dtype=torch.int64, device=lo_probe.device)])
+    return fo_build, fo_probe

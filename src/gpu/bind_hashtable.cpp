@@ -1,0 +1,84 @@
+// Bindings for join/aggregate hash tables (Java API parity: JoinPrimitives.java).
+#include "srj_bind.hpp"
+
+extern "C" {
+void srj_join_build(const void*, const int32_t*, int32_t, int64_t, uint64_t*,
+                    int64_t, hipStream_t);
+void srj_join_probe_count(const void*, const int32_t*, const void*, const int32_t*,
+                          int32_t, int64_t, const uint64_t*, int64_t, uint64_t*,
+                          hipStream_t);
+void srj_join_probe_fill(const void*, const int32_t*, const void*, const int32_t*,
+                         int32_t, int64_t, const uint64_t*, int64_t, uint64_t*,
+                         int32_t*, int64_t*, int64_t, uint8_t*, hipStream_t);
+void srj_join_semi(const void*, const int32_t*, const void*, const int32_t*,
+                   int32_t, int64_t, const uint64_t*, int64_t, uint64_t*,
+                   int64_t*, int64_t, int32_t, hipStream_t);
+void srj_groupby(const void*, const int32_t*, int32_t, int64_t, uint64_t*, int64_t,
+                 const void*, int32_t, hipStream_t);
+void srj_groupby_compact(const uint64_t*, int64_t, const void*, int32_t, uint64_t*,
+                         int64_t*, int64_t*, int64_t, hipStream_t);
+}
+
+void register_hashtable(py::module_& m) {
+  m.def("join_build",
+        [](uintptr_t cols, uintptr_t top, int32_t ntop, int64_t nrows,
+           uintptr_t slots, int64_t capacity, uintptr_t stream) {
+          srj_join_build(as_ptr<void>(cols), as_ptr<int32_t>(top), ntop, nrows,
+                         as_ptr<uint64_t>(slots), capacity, as_stream(stream));
+          check_hip("join_build");
+        });
+  m.def("join_probe_count",
+        [](uintptr_t bcols, uintptr_t btop, uintptr_t pcols, uintptr_t ptop,
+           int32_t ntop, int64_t nprobe, uintptr_t slots, int64_t capacity,
+           uintptr_t counter, uintptr_t stream) {
+          srj_join_probe_count(as_ptr<void>(bcols), as_ptr<int32_t>(btop),
+                               as_ptr<void>(pcols), as_ptr<int32_t>(ptop), ntop,
+                               nprobe, as_ptr<uint64_t>(slots), capacity,
+                               as_ptr<uint64_t>(counter), as_stream(stream));
+          check_hip("join_probe_count");
+        });
+  m.def("join_probe_fill",
+        [](uintptr_t bcols, uintptr_t btop, uintptr_t pcols, uintptr_t ptop,
+           int32_t ntop, int64_t nprobe, uintptr_t slots, int64_t capacity,
+           uintptr_t counter, uintptr_t out_build, uintptr_t out_probe,
+           int64_t out_capacity, uintptr_t build_matched, uintptr_t stream) {
+          srj_join_probe_fill(as_ptr<void>(bcols), as_ptr<int32_t>(btop),
+                              as_ptr<void>(pcols), as_ptr<int32_t>(ptop), ntop,
+                              nprobe, as_ptr<uint64_t>(slots), capacity,
+                              as_ptr<uint64_t>(counter), as_ptr<int32_t>(out_build),
+                              as_ptr<int64_t>(out_probe), out_capacity,
+                              as_ptr<uint8_t>(build_matched), as_stream(stream));
+          check_hip("join_probe_fill");
+        });
+  m.def("join_semi",
+        [](uintptr_t bcols, uintptr_t btop, uintptr_t pcols, uintptr_t ptop,
+           int32_t ntop, int64_t nprobe, uintptr_t slots, int64_t capacity,
+           uintptr_t counter, uintptr_t out_probe, int64_t out_capacity,
+           int32_t anti, uintptr_t stream) {
+          srj_join_semi(as_ptr<void>(bcols), as_ptr<int32_t>(btop),
+                        as_ptr<void>(pcols), as_ptr<int32_t>(ptop), ntop, nprobe,
+                        as_ptr<uint64_t>(slots), capacity, as_ptr<uint64_t>(counter),
+                        as_ptr<int64_t>(out_probe), out_capacity, anti,
+                        as_stream(stream));
+          check_hip("join_semi");
+        });
+  m.def("groupby",
+        [](uintptr_t cols, uintptr_t top, int32_t ntop, int64_t nrows,
+           uintptr_t slots, int64_t capacity, uintptr_t aggs, int32_t naggs,
+           uintptr_t stream) {
+          srj_groupby(as_ptr<void>(cols), as_ptr<int32_t>(top), ntop, nrows,
+                      as_ptr<uint64_t>(slots), capacity, as_ptr<void>(aggs), naggs,
+                      as_stream(stream));
+          check_hip("groupby");
+        });
+  m.def("groupby_compact",
+        [](uintptr_t slots, int64_t capacity, uintptr_t aggs, int32_t naggs,
+           uintptr_t counter, uintptr_t out_repr, uintptr_t out_agg,
+           int64_t out_capacity, uintptr_t stream) {
+          srj_groupby_compact(as_ptr<uint64_t>(slots), capacity, as_ptr<void>(aggs),
+                              naggs, as_ptr<uint64_t>(counter),
+                              as_ptr<int64_t>(out_repr), as_ptr<int64_t>(out_agg),
+                              out_capacity, as_stream(stream));
+          check_hip("groupby_compact");
+        });
+}

@@ -2,11 +2,9 @@
 // replaced by its own bind_*.cpp as the corresponding kernels land.
 #include "srj_bind.hpp"
 
-void register_hashtable(py::module_&) {}
 void register_rowconv(py::module_&) {}
 void register_shuffle(py::module_&) {}
 void register_cast(py::module_&) {}
-void register_copying(py::module_&) {}
 void register_datetime(py::module_&) {}
 void register_json(py::module_&) {}
 void register_misc(py::module_&) {}

@@ -1,0 +1,196 @@
+"""Join / aggregate / gather / partition tests against Python oracles."""
+import random
+from collections import defaultdict
+
+import pytest
+import torch
+
+from spark_rapids_jni_amd.columnar import Column, DType, Table
+
+random.seed(7)
+
+
+def _pairs(build_idx, probe_idx):
+    return set(zip(build_idx.cpu().tolist(), probe_idx.cpu().tolist()))
+
+
+@pytest.mark.gpu
+def test_inner_join_int64():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    nb, np_ = 2000, 5000
+    bvals = [random.randint(0, 700) for _ in range(nb)]
+    pvals = [random.randint(0, 1000) for _ in range(np_)]
+    b = Column.from_pylist(bvals, DType.INT64, "cuda")
+    p = Column.from_pylist(pvals, DType.INT64, "cuda")
+    tbl = HashJoinTable.build(b)
+    bi, pi = tbl.inner_join(p)
+    expect = set()
+    index = defaultdict(list)
+    for i, v in enumerate(bvals):
+        index[v].append(i)
+    for j, v in enumerate(pvals):
+        for i in index.get(v, ()):
+            expect.add((i, j))
+    assert _pairs(bi, pi) == expect
+
+
+@pytest.mark.gpu
+def test_inner_join_nulls_never_match():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    bvals = [1, None, 2, 3, None]
+    pvals = [None, 1, 3, None, 9]
+    b = Column.from_pylist(bvals, DType.INT64, "cuda")
+    p = Column.from_pylist(pvals, DType.INT64, "cuda")
+    bi, pi = HashJoinTable.build(b).inner_join(p)
+    assert _pairs(bi, pi) == {(0, 1), (3, 2)}
+
+
+@pytest.mark.gpu
+def test_inner_join_multicol_string():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    nb, np_ = 300, 800
+    bk1 = [random.randint(0, 20) for _ in range(nb)]
+    bk2 = [random.choice(["a", "bb", "ccc", None, "dd"]) for _ in range(nb)]
+    pk1 = [random.randint(0, 25) for _ in range(np_)]
+    pk2 = [random.choice(["a", "bb", "ccc", None, "x"]) for _ in range(np_)]
+    b = [Column.from_pylist(bk1, DType.INT32, "cuda"),
+         Column.from_pylist(bk2, DType.STRING, "cuda")]
+    p = [Column.from_pylist(pk1, DType.INT32, "cuda"),
+         Column.from_pylist(pk2, DType.STRING, "cuda")]
+    bi, pi = HashJoinTable.build(b).inner_join(p)
+    expect = set()
+    for i in range(nb):
+        if bk2[i] is None:
+            continue
+        for j in range(np_):
+            if pk2[j] is None:
+                continue
+            if bk1[i] == pk1[j] and bk2[i] == pk2[j]:
+                expect.add((i, j))
+    assert _pairs(bi, pi) == expect
+
+
+@pytest.mark.gpu
+def test_semi_anti_join():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    bvals = [1, 2, 3]
+    pvals = [0, 1, 2, 5, 2, None]
+    tbl = HashJoinTable.build(Column.from_pylist(bvals, DType.INT64, "cuda"))
+    p = Column.from_pylist(pvals, DType.INT64, "cuda")
+    semi = sorted(tbl.semi_join(p).cpu().tolist())
+    anti = sorted(tbl.semi_join(p, anti=True).cpu().tolist())
+    assert semi == [1, 2, 4]
+    assert anti == [0, 3, 5]  # null key: no match -> anti emits it
+
+
+@pytest.mark.gpu
+def test_left_outer_full_outer():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable, make_full_outer, make_left_outer
+    bvals = [10, 20, 30]
+    pvals = [20, 99, 10]
+    tbl = HashJoinTable.build(Column.from_pylist(bvals, DType.INT64, "cuda"))
+    p = Column.from_pylist(pvals, DType.INT64, "cuda")
+    bi, pi, matched = tbl.inner_join(p, track_build_matches=True)
+    lb, lp = make_left_outer(3, bi, pi)
+    pairs = set(zip(lb.cpu().tolist(), lp.cpu().tolist()))
+    assert pairs == {(1, 0), (0, 2), (-1, 1)}
+    fb, fp = make_full_outer(matched, bi, pi, 3)
+    pairs = set(zip(fb.cpu().tolist(), fp.cpu().tolist()))
+    assert pairs == {(1, 0), (0, 2), (-1, 1), (2, -1)}
+
+
+@pytest.mark.gpu
+def test_gather_with_nulls():
+    from spark_rapids_jni_amd.ops.copying import gather_column
+    vals = [1.5, None, 3.5, 4.5]
+    col = Column.from_pylist(vals, DType.FLOAT64, "cuda")
+    gmap = torch.tensor([3, -1, 1, 0, 2], dtype=torch.int64, device="cuda")
+    out = gather_column(col, gmap)
+    assert out.to_pylist() == [4.5, None, None, 1.5, 3.5]
+
+
+@pytest.mark.gpu
+def test_gather_strings():
+    from spark_rapids_jni_amd.ops.copying import gather_column
+    vals = ["hello", None, "", "world", "xyzzy"]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    gmap = torch.tensor([4, 0, -1, 2, 1, 3], dtype=torch.int64, device="cuda")
+    out = gather_column(col, gmap)
+    assert out.to_pylist() == ["xyzzy", "hello", None, "", None, "world"]
+
+
+@pytest.mark.gpu
+def test_groupby_sums_counts():
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 5000
+    keys = [random.randint(0, 50) if random.random() > 0.05 else None
+            for _ in range(n)]
+    vals = [random.randint(-100, 100) if random.random() > 0.1 else None
+            for _ in range(n)]
+    fvals = [random.random() * 10 if v is not None else None for v in vals]
+    kc = Column.from_pylist(keys, DType.INT64, "cuda")
+    vc = Column.from_pylist(vals, DType.INT64, "cuda")
+    fc = Column.from_pylist(fvals, DType.FLOAT64, "cuda")
+    kt, res = groupby(kc, [(Agg.COUNT_ALL, None), (Agg.COUNT_VALID, vc),
+                           (Agg.SUM, vc), (Agg.MIN, vc), (Agg.MAX, vc),
+                           (Agg.SUM, fc)])
+    out_keys = kt.columns[0].to_pylist()
+    got = {k: (c, cv, s, mn, mx, fs) for k, c, cv, s, mn, mx, fs in zip(
+        out_keys, res[0].to_pylist(), res[1].to_pylist(), res[2].to_pylist(),
+        res[3].to_pylist(), res[4].to_pylist(), res[5].to_pylist())}
+    exp = defaultdict(lambda: [0, 0, None, None, None, None])
+    for k, v, f in zip(keys, vals, fvals):
+        e = exp[k]
+        e[0] += 1
+        if v is not None:
+            e[1] += 1
+            e[2] = v if e[2] is None else e[2] + v
+            e[3] = v if e[3] is None else min(e[3], v)
+            e[4] = v if e[4] is None else max(e[4], v)
+            e[5] = f if e[5] is None else e[5] + f
+    assert set(got.keys()) == set(exp.keys())
+    for k, e in exp.items():
+        c, cv, s, mn, mx, fs = got[k]
+        assert (c, cv, s, mn, mx) == tuple(e[:5]), f"key {k}"
+        if e[5] is None:
+            assert fs is None
+        else:
+            assert abs(fs - e[5]) < 1e-6
+
+
+@pytest.mark.gpu
+def test_groupby_multicol_keys():
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 2000
+    k1 = [random.randint(0, 5) for _ in range(n)]
+    k2 = [random.choice(["x", "y", None]) for _ in range(n)]
+    c1 = Column.from_pylist(k1, DType.INT32, "cuda")
+    c2 = Column.from_pylist(k2, DType.STRING, "cuda")
+    kt, res = groupby([c1, c2], [(Agg.COUNT_ALL, None)])
+    gk1 = kt.columns[0].to_pylist()
+    gk2 = kt.columns[1].to_pylist()
+    got = {(a, b): c for a, b, c in zip(gk1, gk2, res[0].to_pylist())}
+    exp = defaultdict(int)
+    for a, b in zip(k1, k2):
+        exp[(a, b)] += 1
+    assert got == dict(exp)
+
+
+@pytest.mark.gpu
+def test_partition_roundtrip():
+    from spark_rapids_jni_amd.ops import hashing
+    from spark_rapids_jni_amd.ops.copying import partition_map, spark_partition_ids
+    n, nparts = 10000, 8
+    vals = [random.randint(-10**9, 10**9) for _ in range(n)]
+    col = Column.from_pylist(vals, DType.INT64, "cuda")
+    h = hashing.murmur3([col])
+    pids = spark_partition_ids(h, nparts)
+    offsets, perm = partition_map(pids, nparts)
+    offs = offsets.cpu().tolist()
+    permh = perm.cpu().tolist()
+    pidsh = pids.cpu().tolist()
+    assert offs[0] == 0 and offs[-1] == n
+    assert sorted(permh) == list(range(n))
+    for p in range(nparts):
+        for d in range(offs[p], offs[p + 1]):
+            assert pidsh[permh[d]] == p
