@@ -1,0 +1,202 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: hash-join build+probe on MI355X (BASELINE config[2]:
+"hash-join build+probe 10B x 1B int64 keys, 1xMI355X"), scaled weakly over N
+GPUs with the Spark-style shuffle (murmur3 hash partition -> RCCL all-to-all
+over xGMI -> local probe), one rank per GPU.
+
+One step (per rank):
+  * build a hash table from the rank's ~1B-key build shard
+  * for each of `chunks` probe chunks (1B rows each):
+      - murmur3-hash the keys, pmod-partition into WORLD_SIZE buckets,
+        gather + all_to_all_single over RCCL (skipped at N=1)
+      - probe the local table, materializing the inner-join gather maps
+        (int32 build idx + int64 probe idx)
+Metric: probe rows/s over the whole job (all ranks), max step time over ranks.
+
+Run: python bench.py [--gpus N] [--steps K] [--warmup W]
+Multi-GPU: torchrun --nproc-per-node N bench.py --gpus N (driver contract).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+BUILD_ROWS_DEFAULT = 1_000_000_000
+PROBE_ROWS_DEFAULT = 10_000_000_000
+CHUNK_ROWS = 1_000_000_000
+
+
+def log(rank, *a):
+    if rank == 0:
+        print(*a, file=sys.stderr, flush=True)
+
+
+def setup_dist(args):
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")  # = RCCL on ROCm
+    else:
+        torch.cuda.set_device(local_rank)
+    return world, rank, local_rank
+
+
+def barrier_sync(world):
+    if world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+    torch.cuda.synchronize()
+
+
+def shuffle_exchange(keys: torch.Tensor, world: int):
+    """Spark shuffle write+exchange: murmur3 -> pmod partition -> RCCL a2a."""
+    from spark_rapids_jni_amd.columnar import Column
+    from spark_rapids_jni_amd.ops import hashing
+    from spark_rapids_jni_amd.ops.copying import partition_map, spark_partition_ids
+    import torch.distributed as dist
+
+    from spark_rapids_jni_amd.ops.copying import gather_column
+    col = Column.from_torch(keys)
+    h = hashing.murmur3([col])
+    pids = spark_partition_ids(h, world)
+    offsets, perm = partition_map(pids, world)
+    send = gather_column(col, perm, has_nulls=False).data  # partition-contiguous
+    counts = offsets.diff()
+    recv_counts = torch.empty_like(counts)
+    dist.all_to_all_single(recv_counts, counts)
+    in_splits = counts.cpu().tolist()
+    out_splits = recv_counts.cpu().tolist()
+    out = torch.empty(sum(out_splits), dtype=keys.dtype, device=keys.device)
+    dist.all_to_all_single(out, send, out_splits, in_splits)
+    return out
+
+
+def make_build_shard(n_build: int, world: int, rank: int, device):
+    """Rank r's build shard: global key space [0, world*n_build) routed to the
+    owning rank by pmod(murmur3), like Spark's build-side shuffle (one-time)."""
+    local = torch.arange(rank * n_build, (rank + 1) * n_build, dtype=torch.int64,
+                         device=device)
+    if world == 1:
+        return local
+    return shuffle_exchange(local, world)
+
+
+def one_step(tbl, build_col, n_probe, n_chunk, world, key_space, device, out_hint):
+    from spark_rapids_jni_amd.columnar import Column
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+
+    # rebuild the table each step (config is build+probe)
+    tbl = HashJoinTable.build(build_col)
+    done = 0
+    while done < n_probe:
+        m = min(n_chunk, n_probe - done)
+        probe = torch.randint(0, key_space, (m,), dtype=torch.int64, device=device)
+        if world > 1:
+            probe = shuffle_exchange(probe, world)
+        pcol = Column.from_torch(probe)
+        bi, pi = tbl.inner_join(pcol, out_hint=probe.numel() + out_hint)
+        done += m
+        del probe, pcol, bi, pi
+    return tbl
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--build-rows", type=int, default=BUILD_ROWS_DEFAULT,
+                    help="build rows per GPU (default = named config)")
+    ap.add_argument("--probe-rows", type=int, default=PROBE_ROWS_DEFAULT,
+                    help="probe rows per GPU per step (default = named config)")
+    ap.add_argument("--chunk-rows", type=int, default=CHUNK_ROWS)
+    args = ap.parse_args()
+
+    world, rank, local_rank = setup_dist(args)
+    if args.gpus > 1 and world != args.gpus:
+        raise RuntimeError(
+            f"--gpus {args.gpus} needs torchrun with --nproc-per-node {args.gpus} "
+            f"(WORLD_SIZE is {world})")
+    device = torch.device("cuda", local_rank)
+
+    free, total = torch.cuda.mem_get_info(device)
+    need = (args.build_rows * 28              # build shard + table slots (2.2x pow2)
+            + args.chunk_rows * (40 if world > 1 else 8)  # chunk + partition temps
+            + args.chunk_rows * 13            # join output maps
+            ) * 1.15
+    if need > free:
+        raise MemoryError(
+            f"config needs ~{need/2**30:.0f} GiB, only {free/2**30:.0f} GiB free "
+            "on this GPU — refusing to run a reduced config silently")
+
+    from spark_rapids_jni_amd.columnar import Column
+
+    log(rank, f"[bench] generating build shard ({args.build_rows} rows/GPU)")
+    build = make_build_shard(args.build_rows, world, rank, device)
+    build_col = Column.from_torch(build)
+    key_space = world * args.build_rows
+    # probe rows landing on this rank can exceed the chunk a bit at world>1
+    out_hint = 0 if world == 1 else args.chunk_rows // 16
+
+    log(rank, f"[bench] warmup {args.warmup} steps")
+    for _ in range(args.warmup):
+        one_step(None, build_col, args.probe_rows, args.chunk_rows, world,
+                 key_space, device, out_hint)
+    barrier_sync(world)
+
+    log(rank, f"[bench] timing {args.steps} steps")
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step(None, build_col, args.probe_rows, args.chunk_rows, world,
+                 key_space, device, out_hint)
+    barrier_sync(world)
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_rows = world * args.probe_rows * args.steps
+    value = total_rows / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "hash_join_probe_rows_per_sec",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "model": "hash-join build+probe 10B x 1B int64 keys",
+                "build_rows_per_gpu": args.build_rows,
+                "probe_rows_per_gpu": args.probe_rows,
+                "global_batch": world * args.probe_rows,
+                "seq_len": 0,
+                "parallelism": f"1 executor/GPU x{world}, RCCL all-to-all shuffle",
+            },
+        }))
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
