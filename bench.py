@@ -89,23 +89,18 @@ def make_build_shard(n_build: int, world: int, rank: int, device):
     return shuffle_exchange(local, world)
 
 
-def one_step(tbl, build_col, n_probe, n_chunk, world, key_space, device, out_hint):
+def one_step(build_col, chunks, world, out_hint):
     from spark_rapids_jni_amd.columnar import Column
     from spark_rapids_jni_amd.ops.join import HashJoinTable
 
     # rebuild the table each step (config is build+probe)
     tbl = HashJoinTable.build(build_col)
-    done = 0
-    while done < n_probe:
-        m = min(n_chunk, n_probe - done)
-        probe = torch.randint(0, key_space, (m,), dtype=torch.int64, device=device)
-        if world > 1:
-            probe = shuffle_exchange(probe, world)
+    for chunk in chunks:
+        probe = shuffle_exchange(chunk, world) if world > 1 else chunk
         pcol = Column.from_torch(probe)
         bi, pi = tbl.inner_join(pcol, out_hint=probe.numel() + out_hint)
-        done += m
         del probe, pcol, bi, pi
-    return tbl
+    del tbl
 
 
 def main():
@@ -129,7 +124,8 @@ def main():
 
     free, total = torch.cuda.mem_get_info(device)
     need = (args.build_rows * 28              # build shard + table slots (2.2x pow2)
-            + args.chunk_rows * (40 if world > 1 else 8)  # chunk + partition temps
+            + args.probe_rows * 8             # resident probe chunks
+            + args.chunk_rows * (32 if args.gpus > 1 else 0)  # partition temps
             + args.chunk_rows * 13            # join output maps
             ) * 1.15
     if need > free:
@@ -146,17 +142,25 @@ def main():
     # probe rows landing on this rank can exceed the chunk a bit at world>1
     out_hint = 0 if world == 1 else args.chunk_rows // 16
 
+    # pre-generate probe chunks (synthetic data stays fixed across steps;
+    # RNG cost is outside the timed region)
+    chunks = []
+    done = 0
+    while done < args.probe_rows:
+        m = min(args.chunk_rows, args.probe_rows - done)
+        chunks.append(torch.randint(0, key_space, (m,), dtype=torch.int64,
+                                    device=device))
+        done += m
+
     log(rank, f"[bench] warmup {args.warmup} steps")
     for _ in range(args.warmup):
-        one_step(None, build_col, args.probe_rows, args.chunk_rows, world,
-                 key_space, device, out_hint)
+        one_step(build_col, chunks, world, out_hint)
     barrier_sync(world)
 
     log(rank, f"[bench] timing {args.steps} steps")
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        one_step(None, build_col, args.probe_rows, args.chunk_rows, world,
-                 key_space, device, out_hint)
+        one_step(build_col, chunks, world, out_hint)
     barrier_sync(world)
     elapsed = time.perf_counter() - t0
 

@@ -40,8 +40,10 @@ def _pyb_includes():
     return [pybind11.get_include(), sysconfig.get_paths()["include"]]
 
 
+# NOTE: no -ffast-math — Spark-exact semantics require IEEE NaN/-0.0 behavior
+# (float normalization, NaN-greatest ordering). Individual kernels may opt in.
 HIP_FLAGS = [f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
-             "-ffast-math", "-fno-gpu-rdc", "-Wall"]
+             "-fno-gpu-rdc", "-Wall"]
 CXX_FLAGS = ["-O2", "-std=c++17", "-fPIC", "-Wall",
              "-D__HIP_PLATFORM_AMD__", f"-I{ROCM}/include"]
 LINK_FLAGS = ["-shared", f"-L{ROCM}/lib", "-lamdhip64", "-pthread"]

@@ -202,19 +202,22 @@ __device__ __host__ inline uint32_t mm3_hash_bytes(const char* p, int32_t len,
   return mm3_fmix(h1, (uint32_t)len);
 }
 
-// normalization Spark applies before hashing floats
+// normalization Spark applies before hashing floats (NaN -> canonical NaN,
+// -0.0 -> +0.0). Bit-based so it is immune to fast-math flags.
 __device__ __host__ inline int32_t norm_float_bits(float f) {
-  if (f != f) return 0x7fc00000;  // canonical NaN
-  if (f == 0.0f) f = 0.0f;        // -0.0 -> +0.0
   int32_t b;
   __builtin_memcpy(&b, &f, 4);
+  if ((b & 0x7f800000) == 0x7f800000 && (b & 0x007fffff)) return 0x7fc00000;
+  if ((b & 0x7fffffff) == 0) return 0;
   return b;
 }
 __device__ __host__ inline int64_t norm_double_bits(double d) {
-  if (d != d) return 0x7ff8000000000000LL;
-  if (d == 0.0) d = 0.0;
   int64_t b;
   __builtin_memcpy(&b, &d, 8);
+  if ((b & 0x7ff0000000000000LL) == 0x7ff0000000000000LL &&
+      (b & 0x000fffffffffffffLL))
+    return 0x7ff8000000000000LL;
+  if ((b & 0x7fffffffffffffffLL) == 0) return 0;
   return b;
 }
 
