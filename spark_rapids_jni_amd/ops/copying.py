@@ -70,7 +70,9 @@ def partition_map(parts: torch.Tensor, nparts: int):
     g.partition_hist(parts.data_ptr(), n, nparts, hist.data_ptr(), stream)
     offsets = torch.zeros(nparts + 1, dtype=torch.int64, device=dev)
     torch.cumsum(hist, 0, out=offsets[1:].view(nparts))
-    cursors = offsets[:nparts].contiguous()
+    # clone: the scatter kernel advances the cursors in place and a slice's
+    # .contiguous() would alias the offsets storage
+    cursors = offsets[:nparts].clone()
     perm = torch.empty(n, dtype=torch.int64, device=dev)
     g.partition_scatter(parts.data_ptr(), n, nparts, cursors.data_ptr(),
                         perm.data_ptr(), stream)

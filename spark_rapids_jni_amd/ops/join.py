@@ -32,31 +32,51 @@ def _next_pow2(n: int) -> int:
     return p
 
 
+def _is_i64_fast(cols: List[Column]) -> bool:
+    from ..columnar import DType
+    return (len(cols) == 1 and cols[0].dtype == DType.INT64)
+
+
 class HashJoinTable:
     """Build-side hash table reusable across probes (reference: the build/probe
-    split of hash_inner_join)."""
+    split of hash_inner_join).
+
+    Two layouts: a generic (fingerprint32|row+1) single-word table for any key
+    schema, and a specialized 16-byte {key,row} table for single-int64 keys
+    (the NDS surrogate-key hot path) with software-pipelined probing — see
+    src/gpu/hashtable_i64.hip.
+    """
 
     def __init__(self, build_keys: List[Column], slots: torch.Tensor,
-                 capacity: int, keep):
+                 capacity: int, keep, i64_fast: bool):
         self.build_keys = build_keys
         self.slots = slots
         self.capacity = capacity
         self._keep = keep
+        self.i64_fast = i64_fast
         self.num_build_rows = build_keys[0].size
 
     @staticmethod
-    def build(keys: Union[Column, Table, Sequence[Column]]) -> "HashJoinTable":
+    def build(keys: Union[Column, Table, Sequence[Column]],
+              force_generic: bool = False) -> "HashJoinTable":
         cols = _keys(keys)
         n = cols[0].size
         assert n < 2**31, "build side capped at 2^31-1 rows (chunk the build)"
         g = _native.gpu()
         dev = cols[0].device
         capacity = max(_next_pow2(n * 2), 64)
+        if _is_i64_fast(cols) and not force_generic:
+            c = cols[0]
+            slots = torch.zeros(capacity * 2, dtype=torch.int64, device=dev)
+            g.join_build_i64(c.data.data_ptr(),
+                             c.validity.data_ptr() if c.validity is not None else 0,
+                             n, slots.data_ptr(), capacity, _native.current_stream())
+            return HashJoinTable(cols, slots, capacity, None, True)
         slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
         desc, top, keep = pack_descriptors(cols)
         g.join_build(desc.data_ptr(), top.data_ptr(), len(cols), n,
                      slots.data_ptr(), capacity, _native.current_stream())
-        return HashJoinTable(cols, slots, capacity, (desc, top, keep))
+        return HashJoinTable(cols, slots, capacity, (desc, top, keep), False)
 
     def _descs(self, probe_cols):
         bdesc, btop, bkeep = pack_descriptors(self.build_keys)
@@ -72,13 +92,22 @@ class HashJoinTable:
         g = _native.gpu()
         stream = _native.current_stream()
         dev = pcols[0].device
-        bdesc, btop, pdesc, ptop, keep = self._descs(pcols)
+        fast = self.i64_fast and _is_i64_fast(pcols)
         counter = torch.zeros(1, dtype=torch.int64, device=dev)
+        if not fast:
+            bdesc, btop, pdesc, ptop, keep = self._descs(pcols)
         if out_hint is None:
-            g.join_probe_count(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
-                               ptop.data_ptr(), len(pcols), nprobe,
-                               self.slots.data_ptr(), self.capacity,
-                               counter.data_ptr(), stream)
+            if fast:
+                p = pcols[0]
+                g.join_probe_i64(p.data.data_ptr(),
+                                 p.validity.data_ptr() if p.validity is not None else 0,
+                                 nprobe, self.slots.data_ptr(), self.capacity,
+                                 counter.data_ptr(), 0, 0, 0, 0, 0, stream)
+            else:
+                g.join_probe_count(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
+                                   ptop.data_ptr(), len(pcols), nprobe,
+                                   self.slots.data_ptr(), self.capacity,
+                                   counter.data_ptr(), stream)
             total = int(counter.item())
             counter.zero_()
         else:
@@ -87,11 +116,21 @@ class HashJoinTable:
         out_probe = torch.empty(max(total, 1), dtype=torch.int64, device=dev)
         matched = (torch.zeros(self.num_build_rows, dtype=torch.uint8, device=dev)
                    if track_build_matches else None)
-        g.join_probe_fill(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
-                          ptop.data_ptr(), len(pcols), nprobe,
-                          self.slots.data_ptr(), self.capacity, counter.data_ptr(),
-                          out_build.data_ptr(), out_probe.data_ptr(), total,
-                          matched.data_ptr() if matched is not None else 0, stream)
+        if fast:
+            p = pcols[0]
+            g.join_probe_i64(p.data.data_ptr(),
+                             p.validity.data_ptr() if p.validity is not None else 0,
+                             nprobe, self.slots.data_ptr(), self.capacity,
+                             counter.data_ptr(), out_build.data_ptr(),
+                             out_probe.data_ptr(), total,
+                             matched.data_ptr() if matched is not None else 0,
+                             1, stream)
+        else:
+            g.join_probe_fill(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
+                              ptop.data_ptr(), len(pcols), nprobe,
+                              self.slots.data_ptr(), self.capacity, counter.data_ptr(),
+                              out_build.data_ptr(), out_probe.data_ptr(), total,
+                              matched.data_ptr() if matched is not None else 0, stream)
         actual = int(counter.item())
         if actual > total:
             # hint was too small: rerun with the exact size
@@ -105,6 +144,21 @@ class HashJoinTable:
 
     def semi_join(self, probe, anti: bool = False) -> torch.Tensor:
         """Left semi/anti join: probe-side row indices with (no) match."""
+        if self.i64_fast:
+            # semi/anti runs on the generic slot layout
+            if not hasattr(self, "_generic"):
+                g = _native.gpu()
+                n = self.build_keys[0].size
+                slots = torch.zeros(self.capacity, dtype=torch.int64,
+                                    device=self.build_keys[0].device)
+                desc, top, keep = pack_descriptors(self.build_keys)
+                g.join_build(desc.data_ptr(), top.data_ptr(), 1, n,
+                             slots.data_ptr(), self.capacity,
+                             _native.current_stream())
+                self._generic = HashJoinTable(self.build_keys, slots,
+                                              self.capacity, (desc, top, keep),
+                                              False)
+            return self._generic.semi_join(probe, anti)
         pcols = _keys(probe)
         nprobe = pcols[0].size
         g = _native.gpu()

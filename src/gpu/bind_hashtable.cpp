@@ -17,6 +17,11 @@ void srj_groupby(const void*, const int32_t*, int32_t, int64_t, uint64_t*, int64
                  const void*, int32_t, hipStream_t);
 void srj_groupby_compact(const uint64_t*, int64_t, const void*, int32_t, uint64_t*,
                          int64_t*, int64_t*, int64_t, hipStream_t);
+void srj_join_build_i64(const long long*, const uint8_t*, int64_t, void*, int64_t,
+                        hipStream_t);
+void srj_join_probe_i64(const long long*, const uint8_t*, int64_t, const void*,
+                        int64_t, uint64_t*, int32_t*, int64_t*, int64_t, uint8_t*,
+                        int32_t, hipStream_t);
 }
 
 void register_hashtable(py::module_& m) {
@@ -61,6 +66,25 @@ void register_hashtable(py::module_& m) {
                         as_ptr<int64_t>(out_probe), out_capacity, anti,
                         as_stream(stream));
           check_hip("join_semi");
+        });
+  m.def("join_build_i64",
+        [](uintptr_t keys, uintptr_t valid, int64_t nrows, uintptr_t slots,
+           int64_t capacity, uintptr_t stream) {
+          srj_join_build_i64(as_ptr<long long>(keys), as_ptr<uint8_t>(valid), nrows,
+                             as_ptr<void>(slots), capacity, as_stream(stream));
+          check_hip("join_build_i64");
+        });
+  m.def("join_probe_i64",
+        [](uintptr_t probe, uintptr_t pvalid, int64_t nprobe, uintptr_t slots,
+           int64_t capacity, uintptr_t counter, uintptr_t out_build,
+           uintptr_t out_probe, int64_t out_capacity, uintptr_t build_matched,
+           int32_t fill, uintptr_t stream) {
+          srj_join_probe_i64(as_ptr<long long>(probe), as_ptr<uint8_t>(pvalid),
+                             nprobe, as_ptr<void>(slots), capacity,
+                             as_ptr<uint64_t>(counter), as_ptr<int32_t>(out_build),
+                             as_ptr<int64_t>(out_probe), out_capacity,
+                             as_ptr<uint8_t>(build_matched), fill, as_stream(stream));
+          check_hip("join_probe_i64");
         });
   m.def("groupby",
         [](uintptr_t cols, uintptr_t top, int32_t ntop, int64_t nrows,

@@ -1,0 +1,173 @@
+// Specialized int64-key hash table (the NDS hot path: fact-table joins on
+// int64 surrogate keys; BASELINE config[2]).
+//
+// Why a separate design from hashtable.hip's generic (fp|row) table: the
+// generic probe is a dependent chain of TWO random loads per row (slot word,
+// then the build key for equality). rocprof on the v1 kernel showed it
+// latency-bound at ~3% of HBM bandwidth (profiles/r01_join_v1_kernel_stats).
+// Here:
+//   * 16-byte slots {key, row+1} — key inline, so a probe is ONE random
+//     16-byte load (one 64B line) per slot visited;
+//   * software-pipelined batches of PIPE rows per lane: hashes and first-slot
+//     loads for the whole batch are issued independently before any resolve,
+//     giving each lane PIPE outstanding loads (memory-level parallelism)
+//     instead of one dependent chain;
+//   * wave-aggregated output append (one atomicAdd per wave for the common
+//     first-match case), per-match atomics only for rare duplicate matches.
+// Claim protocol: atomicCAS on the row word (0 = empty), winner writes key;
+// no sentinel key needed, build completes before probe launches.
+#include "srj_common.hpp"
+
+namespace srj {
+
+struct Slot64 {
+  long long key;
+  long long row1;  // row + 1; 0 = empty
+};
+
+constexpr int PIPE = 8;
+
+__device__ inline uint64_t i64_hash(long long k) { return mix64((uint64_t)k); }
+
+// ---------------------------------------------------------------------------
+// build
+// ---------------------------------------------------------------------------
+__global__ void join_build_i64_kernel(const long long* __restrict__ keys,
+                                      const uint8_t* __restrict__ valid,
+                                      int64_t nrows, Slot64* __restrict__ slots,
+                                      uint64_t mask) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
+  // batch of PIPE strided rows per iteration
+  for (int64_t base = tid * PIPE; base < nrows; base += nthreads * PIPE) {
+    long long k[PIPE];
+    uint64_t s[PIPE];
+    bool act[PIPE];
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) {
+      int64_t row = base + b;
+      act[b] = row < nrows && is_valid(valid, row);
+      k[b] = act[b] ? keys[row] : 0;
+      s[b] = i64_hash(k[b]) & mask;
+    }
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) {
+      if (!act[b]) continue;
+      int64_t row = base + b;
+      uint64_t sl = s[b];
+      while (true) {
+        long long prev = atomicCAS(
+            reinterpret_cast<unsigned long long*>(&slots[sl].row1), 0ull,
+            (unsigned long long)(row + 1));
+        if (prev == 0) {
+          slots[sl].key = k[b];
+          break;
+        }
+        sl = (sl + 1) & mask;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// probe (count / fill fused via template)
+// ---------------------------------------------------------------------------
+template <bool FILL>
+__global__ void join_probe_i64_kernel(
+    const long long* __restrict__ build_unused, const long long* __restrict__ probe,
+    const uint8_t* __restrict__ pvalid, int64_t nprobe,
+    const Slot64* __restrict__ slots, uint64_t mask,
+    uint64_t* __restrict__ counter, int32_t* __restrict__ out_build,
+    int64_t* __restrict__ out_probe, int64_t out_capacity,
+    uint8_t* __restrict__ build_matched) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
+  int lane = threadIdx.x & (WAVE - 1);
+  uint64_t count_local = 0;
+  // ceil so every lane of a wave runs the same batch iterations (ballots)
+  int64_t nbatch = (nprobe + (int64_t)PIPE - 1) / PIPE;
+  for (int64_t batch = tid; batch < nbatch; batch += nthreads) {
+    int64_t base = batch * PIPE;
+    long long k[PIPE];
+    uint64_t s[PIPE];
+    bool act[PIPE];
+    Slot64 first[PIPE];
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) {
+      int64_t row = base + b;
+      act[b] = row < nprobe && is_valid(pvalid, row);
+      k[b] = act[b] ? probe[row] : 0;
+      s[b] = i64_hash(k[b]) & mask;
+    }
+    // issue the PIPE independent first-slot loads
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) {
+      first[b] = act[b] ? slots[s[b]] : Slot64{0, 0};
+    }
+    // resolve: common case terminates on the prefetched slot
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) {
+      if (!act[b]) continue;
+      int64_t row = base + b;
+      Slot64 cur = first[b];
+      uint64_t sl = s[b];
+      while (cur.row1 != 0) {
+        if (cur.key == k[b]) {
+          if (FILL) {
+            uint64_t pos = atomicAdd((unsigned long long*)counter, 1ull);
+            if ((int64_t)pos < out_capacity) {
+              out_build[pos] = (int32_t)(cur.row1 - 1);
+              out_probe[pos] = row;
+            }
+            if (build_matched) build_matched[cur.row1 - 1] = 1;
+          } else {
+            ++count_local;
+          }
+        }
+        sl = (sl + 1) & mask;
+        cur = slots[sl];
+      }
+    }
+  }
+  if (!FILL) {
+    count_local = wave_sum(count_local);
+    if (lane == 0 && count_local)
+      atomicAdd((unsigned long long*)counter, (unsigned long long)count_local);
+  }
+}
+
+}  // namespace srj
+
+using namespace srj;
+
+extern "C" {
+
+void srj_join_build_i64(const long long* keys, const uint8_t* valid, int64_t nrows,
+                        void* slots, int64_t capacity, hipStream_t stream) {
+  int64_t nthreads_needed = (nrows + PIPE - 1) / PIPE;
+  join_build_i64_kernel<<<grid_1d(nthreads_needed), DEFAULT_BLOCK, 0, stream>>>(
+      keys, valid, nrows, reinterpret_cast<Slot64*>(slots),
+      (uint64_t)(capacity - 1));
+}
+
+void srj_join_probe_i64(const long long* probe, const uint8_t* pvalid,
+                        int64_t nprobe, const void* slots, int64_t capacity,
+                        uint64_t* counter, int32_t* out_build, int64_t* out_probe,
+                        int64_t out_capacity, uint8_t* build_matched, int32_t fill,
+                        hipStream_t stream) {
+  int64_t nthreads_needed = (nprobe + PIPE - 1) / PIPE;
+  if (fill)
+    join_probe_i64_kernel<true>
+        <<<grid_1d(nthreads_needed), DEFAULT_BLOCK, 0, stream>>>(
+            nullptr, probe, pvalid, nprobe,
+            reinterpret_cast<const Slot64*>(slots), (uint64_t)(capacity - 1),
+            counter, out_build, out_probe, out_capacity, build_matched);
+  else
+    join_probe_i64_kernel<false>
+        <<<grid_1d(nthreads_needed), DEFAULT_BLOCK, 0, stream>>>(
+            nullptr, probe, pvalid, nprobe,
+            reinterpret_cast<const Slot64*>(slots), (uint64_t)(capacity - 1),
+            counter, nullptr, nullptr, 0, nullptr);
+}
+
+}  // extern "C"

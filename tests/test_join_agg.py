@@ -194,3 +194,21 @@ def test_partition_roundtrip():
     for p in range(nparts):
         for d in range(offs[p], offs[p + 1]):
             assert pidsh[permh[d]] == p
+
+
+@pytest.mark.gpu
+def test_i64_fast_path_matches_generic():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    nb, np_ = 3000, 9000
+    bvals = [random.randint(0, 1500) if random.random() > 0.02 else None
+             for _ in range(nb)]
+    pvals = [random.randint(0, 2000) if random.random() > 0.02 else None
+             for _ in range(np_)]
+    b = Column.from_pylist(bvals, DType.INT64, "cuda")
+    p = Column.from_pylist(pvals, DType.INT64, "cuda")
+    fast = HashJoinTable.build(b)
+    slow = HashJoinTable.build(b, force_generic=True)
+    assert fast.i64_fast and not slow.i64_fast
+    fb, fp = fast.inner_join(p)
+    sb, sp = slow.inner_join(p)
+    assert _pairs(fb, fp) == _pairs(sb, sp)
