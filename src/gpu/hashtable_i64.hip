@@ -72,60 +72,94 @@ __global__ void join_build_i64_kernel(const long long* __restrict__ keys,
 // ---------------------------------------------------------------------------
 // probe (count / fill fused via template)
 // ---------------------------------------------------------------------------
-template <bool FILL>
+// Probe with count-then-emit per batch:
+//  * stage 1: one validity byte covers the whole 8-row batch (PIPE == 8);
+//    probe keys and first slots load unconditionally (clamped), back-to-back
+//    — the compiler then emits 8 independent dwordx4 loads with counted
+//    vmcnt, which is the MLP this kernel lives on.
+//  * stage 2 (resolve): count this lane's matches; slot lines end L1-warm.
+//  * stage 3 (emit, FILL only): ONE wave-wide atomicAdd reserves output space
+//    (wave prefix sum gives per-lane bases), then a cache-warm rescan writes
+//    the pairs with plain stores — no per-match atomics.
+template <bool FILL, bool HAS_VALID>
 __global__ void join_probe_i64_kernel(
-    const long long* __restrict__ build_unused, const long long* __restrict__ probe,
-    const uint8_t* __restrict__ pvalid, int64_t nprobe,
-    const Slot64* __restrict__ slots, uint64_t mask,
+    const long long* __restrict__ probe, const uint8_t* __restrict__ pvalid,
+    int64_t nprobe, const Slot64* __restrict__ slots, uint64_t mask,
     uint64_t* __restrict__ counter, int32_t* __restrict__ out_build,
     int64_t* __restrict__ out_probe, int64_t out_capacity,
     uint8_t* __restrict__ build_matched) {
+  static_assert(PIPE == 8, "validity byte per batch assumes PIPE == 8");
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   int lane = threadIdx.x & (WAVE - 1);
   uint64_t count_local = 0;
-  // ceil so every lane of a wave runs the same batch iterations (ballots)
   int64_t nbatch = (nprobe + (int64_t)PIPE - 1) / PIPE;
-  for (int64_t batch = tid; batch < nbatch; batch += nthreads) {
-    int64_t base = batch * PIPE;
+  // pad so every lane of a wave runs the same iterations (wave shuffles)
+  int64_t nbatch_pad = (nbatch + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t batch = tid; batch < nbatch_pad; batch += nthreads) {
+    bool batch_ok = batch < nbatch;
+    int64_t base = batch_ok ? batch * PIPE : 0;
+    uint32_t vbits = 0xff;
+    if (HAS_VALID) vbits = pvalid[base >> 3];
+    if (base + PIPE > nprobe) {
+      int64_t tail = nprobe - base;
+      vbits &= (uint32_t)((1u << (tail < 0 ? 0 : tail)) - 1u);
+    }
+    if (!batch_ok) vbits = 0;
     long long k[PIPE];
     uint64_t s[PIPE];
-    bool act[PIPE];
     Slot64 first[PIPE];
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) {
       int64_t row = base + b;
-      act[b] = row < nprobe && is_valid(pvalid, row);
-      k[b] = act[b] ? probe[row] : 0;
+      k[b] = probe[row < nprobe ? row : 0];  // clamped, unconditional
       s[b] = i64_hash(k[b]) & mask;
     }
-    // issue the PIPE independent first-slot loads
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) first[b] = slots[s[b]];
+    // resolve: count matches (and for !FILL that's all)
+    uint32_t nm = 0;
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) {
-      first[b] = act[b] ? slots[s[b]] : Slot64{0, 0};
-    }
-    // resolve: common case terminates on the prefetched slot
-#pragma unroll
-    for (int b = 0; b < PIPE; ++b) {
-      if (!act[b]) continue;
-      int64_t row = base + b;
+      if (!((vbits >> b) & 1)) continue;
       Slot64 cur = first[b];
       uint64_t sl = s[b];
       while (cur.row1 != 0) {
-        if (cur.key == k[b]) {
-          if (FILL) {
-            uint64_t pos = atomicAdd((unsigned long long*)counter, 1ull);
-            if ((int64_t)pos < out_capacity) {
-              out_build[pos] = (int32_t)(cur.row1 - 1);
-              out_probe[pos] = row;
-            }
-            if (build_matched) build_matched[cur.row1 - 1] = 1;
-          } else {
-            ++count_local;
-          }
-        }
+        nm += (cur.key == k[b]);
         sl = (sl + 1) & mask;
         cur = slots[sl];
+      }
+    }
+    if (!FILL) {
+      count_local += nm;
+      continue;
+    }
+    // emit: one atomic per wave, then cache-warm rescan writes the pairs
+    uint32_t incl = wave_prefix_incl(nm);
+    uint32_t total = __shfl(incl, WAVE - 1, WAVE);
+    uint64_t wave_base = 0;
+    if (lane == WAVE - 1 && total)
+      wave_base = atomicAdd((unsigned long long*)counter, (unsigned long long)total);
+    wave_base = __shfl(wave_base, WAVE - 1, WAVE);
+    int64_t pos = (int64_t)(wave_base + incl - nm);
+    if (nm) {
+#pragma unroll
+      for (int b = 0; b < PIPE; ++b) {
+        if (!((vbits >> b) & 1)) continue;
+        Slot64 cur = first[b];
+        uint64_t sl = s[b];
+        while (cur.row1 != 0) {
+          if (cur.key == k[b]) {
+            if (pos < out_capacity) {
+              out_build[pos] = (int32_t)(cur.row1 - 1);
+              out_probe[pos] = base + b;
+              if (build_matched) build_matched[cur.row1 - 1] = 1;
+            }
+            ++pos;
+          }
+          sl = (sl + 1) & mask;
+          cur = slots[sl];
+        }
       }
     }
   }
@@ -156,18 +190,26 @@ void srj_join_probe_i64(const long long* probe, const uint8_t* pvalid,
                         int64_t out_capacity, uint8_t* build_matched, int32_t fill,
                         hipStream_t stream) {
   int64_t nthreads_needed = (nprobe + PIPE - 1) / PIPE;
-  if (fill)
-    join_probe_i64_kernel<true>
-        <<<grid_1d(nthreads_needed), DEFAULT_BLOCK, 0, stream>>>(
-            nullptr, probe, pvalid, nprobe,
-            reinterpret_cast<const Slot64*>(slots), (uint64_t)(capacity - 1),
-            counter, out_build, out_probe, out_capacity, build_matched);
-  else
-    join_probe_i64_kernel<false>
-        <<<grid_1d(nthreads_needed), DEFAULT_BLOCK, 0, stream>>>(
-            nullptr, probe, pvalid, nprobe,
-            reinterpret_cast<const Slot64*>(slots), (uint64_t)(capacity - 1),
-            counter, nullptr, nullptr, 0, nullptr);
+  int64_t g = grid_1d(nthreads_needed);
+  const Slot64* sl = reinterpret_cast<const Slot64*>(slots);
+  uint64_t mask = (uint64_t)(capacity - 1);
+  if (fill) {
+    if (pvalid)
+      join_probe_i64_kernel<true, true><<<g, DEFAULT_BLOCK, 0, stream>>>(
+          probe, pvalid, nprobe, sl, mask, counter, out_build, out_probe,
+          out_capacity, build_matched);
+    else
+      join_probe_i64_kernel<true, false><<<g, DEFAULT_BLOCK, 0, stream>>>(
+          probe, pvalid, nprobe, sl, mask, counter, out_build, out_probe,
+          out_capacity, build_matched);
+  } else {
+    if (pvalid)
+      join_probe_i64_kernel<false, true><<<g, DEFAULT_BLOCK, 0, stream>>>(
+          probe, pvalid, nprobe, sl, mask, counter, nullptr, nullptr, 0, nullptr);
+    else
+      join_probe_i64_kernel<false, false><<<g, DEFAULT_BLOCK, 0, stream>>>(
+          probe, pvalid, nprobe, sl, mask, counter, nullptr, nullptr, 0, nullptr);
+  }
 }
 
 }  // extern "C"

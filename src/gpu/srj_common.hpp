@@ -127,6 +127,18 @@ __device__ inline T wave_sum(T v) {
   return v;
 }
 
+// inclusive prefix sum across the wave; all 64 lanes must participate
+template <typename T>
+__device__ inline T wave_prefix_incl(T v) {
+  int lane = threadIdx.x & (WAVE - 1);
+#pragma unroll
+  for (int off = 1; off < WAVE; off <<= 1) {
+    T n = __shfl_up(v, off, WAVE);
+    if (lane >= off) v += n;
+  }
+  return v;
+}
+
 template <typename T>
 __device__ inline T wave_max(T v) {
 #pragma unroll
