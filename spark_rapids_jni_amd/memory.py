@@ -1,0 +1,197 @@
+"""Memory/resource layer: RmmSpark facade + typed OOM exceptions.
+
+Java API parity (reference): RmmSpark.java (static facade), GpuRetryOOM /
+GpuSplitAndRetryOOM / CpuRetryOOM / CpuSplitAndRetryOOM / GpuOOM exception
+hierarchy, SparkResourceAdaptor.java + SparkResourceAdaptorJni.cpp state
+machine (see src/host/resource_adaptor.cpp and docs/memory_management.md).
+
+The native core exposes two integration modes:
+  * simulated pool (`alloc`/`dealloc`): full state machine against a byte
+    budget — used by tests, the Monte Carlo stress harness, and host-memory
+    tracking;
+  * external hooks (`pre_alloc_external`/...): bracket real device
+    allocations made through torch's caching allocator.
+"""
+import threading
+from typing import Optional
+
+from . import _native
+
+
+class GpuOOM(RuntimeError):
+    """Base of GPU OOM errors (reference GpuOOM.java)."""
+
+
+class GpuRetryOOM(GpuOOM):
+    """Roll back to a spillable state and retry (reference GpuRetryOOM.java:17)."""
+
+
+class GpuSplitAndRetryOOM(GpuOOM):
+    """Split the input and retry (reference GpuSplitAndRetryOOM.java)."""
+
+
+class CpuRetryOOM(RuntimeError):
+    pass
+
+
+class CpuSplitAndRetryOOM(RuntimeError):
+    pass
+
+
+class OffHeapOOM(RuntimeError):
+    pass
+
+
+class ThreadRemovedError(RuntimeError):
+    """Thread was unregistered while blocked (REMOVE_THROW)."""
+
+
+_CODE_TO_EXC = {
+    1: GpuRetryOOM,
+    2: GpuSplitAndRetryOOM,
+    3: CpuRetryOOM,
+    4: CpuSplitAndRetryOOM,
+    5: ThreadRemovedError,
+    6: OffHeapOOM,
+}
+
+
+def _raise_for(code: int, what: str = "allocation"):
+    if code == 0:
+        return
+    exc = _CODE_TO_EXC.get(code, RuntimeError)
+    raise exc(f"{what} -> {exc.__name__}")
+
+
+class TaskPriority:
+    """Monotonic per-task-attempt priority (reference task_priority.hpp:26,
+    TaskPriority.java:28). Lower number = higher priority."""
+    _lock = threading.Lock()
+    _next = 0
+    _assigned = {}
+
+    @classmethod
+    def get_task_priority(cls, task_attempt_id: int) -> int:
+        with cls._lock:
+            if task_attempt_id not in cls._assigned:
+                cls._assigned[task_attempt_id] = cls._next
+                cls._next += 1
+            return cls._assigned[task_attempt_id]
+
+    @classmethod
+    def task_done(cls, task_attempt_id: int) -> None:
+        with cls._lock:
+            cls._assigned.pop(task_attempt_id, None)
+
+
+class RmmSpark:
+    """Static facade over the resource adaptor (reference RmmSpark.java)."""
+
+    _adaptor = None
+    _lock = threading.Lock()
+
+    # -- lifecycle (reference RmmSpark.setEventHandler / clearEventHandler) --
+    @classmethod
+    def set_event_handler(cls, pool_limit: int = -1, host_limit: int = -1):
+        with cls._lock:
+            if cls._adaptor is not None:
+                raise RuntimeError("event handler already set")
+            cls._adaptor = _native.host().SparkResourceAdaptor(pool_limit,
+                                                               host_limit)
+            return cls._adaptor
+
+    @classmethod
+    def clear_event_handler(cls):
+        with cls._lock:
+            cls._adaptor = None
+
+    @classmethod
+    def adaptor(cls):
+        if cls._adaptor is None:
+            raise RuntimeError("RmmSpark.set_event_handler was not called")
+        return cls._adaptor
+
+    @staticmethod
+    def current_thread_id() -> int:
+        return threading.get_ident()
+
+    # -- thread/task association --------------------------------------------
+    @classmethod
+    def start_dedicated_task_thread(cls, thread_id: int, task_id: int):
+        cls.adaptor().start_dedicated_task_thread(thread_id, task_id)
+
+    @classmethod
+    def current_thread_is_dedicated_to_task(cls, task_id: int):
+        cls.start_dedicated_task_thread(cls.current_thread_id(), task_id)
+
+    @classmethod
+    def shuffle_thread_working_on_tasks(cls, task_ids):
+        cls.adaptor().pool_thread_working_on_tasks(cls.current_thread_id(),
+                                                   list(task_ids), True)
+
+    @classmethod
+    def pool_thread_working_on_tasks(cls, is_for_shuffle, task_ids,
+                                     thread_id: Optional[int] = None):
+        cls.adaptor().pool_thread_working_on_tasks(
+            thread_id if thread_id is not None else cls.current_thread_id(),
+            list(task_ids), bool(is_for_shuffle))
+
+    @classmethod
+    def pool_thread_finished_for_tasks(cls, task_ids,
+                                       thread_id: Optional[int] = None):
+        cls.adaptor().pool_thread_finished_for_tasks(
+            thread_id if thread_id is not None else cls.current_thread_id(),
+            list(task_ids))
+
+    @classmethod
+    def remove_current_thread_association(cls):
+        cls.adaptor().remove_thread_association(cls.current_thread_id())
+
+    @classmethod
+    def task_done(cls, task_id: int):
+        cls.adaptor().task_done(task_id)
+        TaskPriority.task_done(task_id)
+
+    # -- allocation (simulated pool; raises the typed OOMs) ------------------
+    @classmethod
+    def alloc(cls, nbytes: int, thread_id: Optional[int] = None):
+        tid = thread_id if thread_id is not None else cls.current_thread_id()
+        _raise_for(cls.adaptor().alloc_sim(tid, nbytes), f"alloc({nbytes})")
+
+    @classmethod
+    def dealloc(cls, nbytes: int, thread_id: Optional[int] = None):
+        tid = thread_id if thread_id is not None else cls.current_thread_id()
+        cls.adaptor().dealloc_sim(tid, nbytes)
+
+    @classmethod
+    def block_thread_until_ready(cls, thread_id: Optional[int] = None):
+        tid = thread_id if thread_id is not None else cls.current_thread_id()
+        _raise_for(cls.adaptor().block_thread_until_ready(tid),
+                   "block_thread_until_ready")
+
+    # -- test/fault injection (reference RmmSpark.forceRetryOOM:534) ---------
+    @classmethod
+    def force_retry_oom(cls, thread_id: int, num_ooms: int = 1):
+        cls.adaptor().force_retry_oom(thread_id, num_ooms)
+
+    @classmethod
+    def force_split_and_retry_oom(cls, thread_id: int, num_ooms: int = 1):
+        cls.adaptor().force_split_and_retry_oom(thread_id, num_ooms)
+
+    # -- spill ranges (reference RmmSpark.spillRangeStart/Done:867-880) ------
+    @classmethod
+    def spill_range_start(cls):
+        cls.adaptor().spill_range_start(cls.current_thread_id())
+
+    @classmethod
+    def spill_range_done(cls):
+        cls.adaptor().spill_range_done(cls.current_thread_id())
+
+    # -- metrics (reference RmmSpark.getAndReset*:663-767) -------------------
+    @classmethod
+    def get_and_reset_metrics(cls, task_id: int) -> dict:
+        return cls.adaptor().get_and_reset_metrics(task_id)
+
+    @classmethod
+    def get_state_of(cls, thread_id: int) -> str:
+        return cls.adaptor().get_state_of(thread_id)
