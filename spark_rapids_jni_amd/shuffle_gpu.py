@@ -1,0 +1,341 @@
+"""Device-side Kudo shuffle: split a GPU table into per-partition Kudo
+records and assemble received records back into one GPU table.
+
+Java API parity: kudo/KudoGpuSerializer.java:50,71
+(splitAndSerializeToDevice / assembleFromDeviceRaw) and the native
+shuffle_split / shuffle_assemble pair.
+
+The per-partition output is a byte-exact Kudo record (same format as
+spark_rapids_jni_amd/kudo.py writes on the host — cross-validated in
+tests/test_shuffle_gpu.py), so records can be merged by either path. The
+buffers are laid out contiguously per destination rank and feed RCCL
+all_to_all_single directly — no host bounce.
+
+Current device-path scope: flat schemas (fixed-width + string columns);
+nested types take the host kudo path.
+"""
+import struct
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from . import _native
+from .columnar import Column, DType, FIXED_WIDTH, Table, validity_nbytes
+from .kudo import KudoTableHeader, MAGIC
+
+_COPYSEG = "<QQq"          # src, dst, nbytes
+_VALIDSEG = "<QQqqq"       # src, dst, src_start_bit, dst_start_bit, nbits
+_OFFSEG = "<QQqii"         # src, dst, n, base, write_last
+COPY_CHUNK = 16384
+
+
+class _SegBatch:
+    """Accumulates copy segments + chunk prefix, then launches one kernel."""
+
+    def __init__(self):
+        self.segs = []
+        self.chunks = []
+
+    def add(self, src_ptr: int, dst_ptr: int, nbytes: int):
+        if nbytes <= 0:
+            return
+        self.segs.append(struct.pack(_COPYSEG, src_ptr, dst_ptr, nbytes))
+        self.chunks.append((nbytes + COPY_CHUNK - 1) // COPY_CHUNK)
+
+    def run(self, device, g, stream):
+        if not self.segs:
+            return
+        raw = b"".join(self.segs)
+        prefix = np.zeros(len(self.chunks), dtype=np.int64)
+        np.cumsum(self.chunks[:-1], out=prefix[1:])
+        total = int(prefix[-1] + self.chunks[-1])
+        segs_t = torch.frombuffer(bytearray(raw), dtype=torch.uint8).to(device)
+        pref_t = torch.from_numpy(prefix).to(device)
+        g.segmented_copy(segs_t.data_ptr(), pref_t.data_ptr(), len(self.chunks),
+                         total, stream)
+        return segs_t, pref_t  # keepalive
+
+
+def _check_flat(cols: Sequence[Column]):
+    for c in cols:
+        assert not c.children, "device shuffle path supports flat schemas"
+
+
+def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
+                                  perm: torch.Tensor
+                                  ) -> Tuple[torch.Tensor, List[int]]:
+    """Split rows (already partition-grouped by `perm`) into P contiguous
+    device Kudo records. Returns (one uint8 buffer, per-partition byte sizes).
+    """
+    from .ops.copying import gather
+    _check_flat(table.columns)
+    g = _native.gpu()
+    stream = _native.current_stream()
+    dev = table.device
+    G = gather(table, perm)
+    offs = offsets.cpu().tolist()
+    nparts = len(offs) - 1
+
+    ncols = len(G.columns)
+    hlen = 28 + (ncols + 7) // 8
+
+    # string char boundaries per partition (one small D2H)
+    str_cols = [c for c in G.columns if c.dtype == DType.STRING]
+    char_bounds = {}
+    if str_cols:
+        for ci, c in enumerate(G.columns):
+            if c.dtype == DType.STRING:
+                idx = torch.tensor(offs, dtype=torch.int64, device=dev)
+                char_bounds[ci] = c.offsets[idx].cpu().tolist()
+
+    # plan per-partition record layout
+    headers = bytearray()
+    sizes = []
+    seg_plan = []  # (src_ptr, rec_off, nbytes) rec-relative; headers separate
+    positions = []
+    pos = 0
+    for p in range(nparts):
+        start, end = offs[p], offs[p + 1]
+        n = end - start
+        bitset = bytearray((ncols + 7) // 8)
+        validity_parts = []  # (src_ptr, nbytes)
+        offset_parts = []
+        data_parts = []
+        for ci, c in enumerate(G.columns):
+            if c.validity is not None and n > 0:
+                # n == 0: prefer "no validity" (kudo spec) over a filler byte
+                bitset[ci // 8] |= 1 << (ci % 8)
+                b0 = start // 8
+                b1 = max((start + n + 7) // 8, b0 + 1)
+                validity_parts.append((c.validity.data_ptr() + b0, b1 - b0))
+            if c.dtype == DType.STRING:
+                if n > 0:
+                    offset_parts.append((c.offsets.data_ptr() + start * 4,
+                                         (n + 1) * 4))
+                    cb = char_bounds[ci]
+                    nchars = cb[p + 1] - cb[p]
+                    if nchars > 0:
+                        data_parts.append((c.data.data_ptr() + cb[p], nchars))
+            else:
+                w = FIXED_WIDTH[c.dtype]
+                if n > 0:
+                    data_parts.append((c.data.data_ptr() + start * w, n * w))
+        vlen = sum(x[1] for x in validity_parts)
+        vpad = (4 - (hlen + vlen) % 4) % 4
+        olen = sum(x[1] for x in offset_parts)
+        dlen = sum(x[1] for x in data_parts)
+        dpad = (4 - dlen % 4) % 4
+        h = KudoTableHeader(start, n, vlen + vpad, olen, 0, ncols, bytes(bitset))
+        h.total_len = h.validity_len + olen + dlen + dpad
+        import io
+        b = io.BytesIO()
+        h.write(b)
+        headers.extend(b.getvalue())
+        rec_len = hlen + h.total_len
+        # record-relative positions of body parts
+        rp = hlen
+        for src, nb in validity_parts:
+            seg_plan.append((src, pos + rp, nb))
+            rp += nb
+        rp += vpad
+        for src, nb in offset_parts:
+            seg_plan.append((src, pos + rp, nb))
+            rp += nb
+        for src, nb in data_parts:
+            seg_plan.append((src, pos + rp, nb))
+            rp += nb
+        positions.append(pos)
+        sizes.append(rec_len)
+        pos += rec_len
+
+    out = torch.zeros(max(pos, 1), dtype=torch.uint8, device=dev)
+    base = out.data_ptr()
+    batch = _SegBatch()
+    # headers: staged once, copied into place
+    hdr_stage = torch.frombuffer(bytes(headers) or b"\x00",
+                                 dtype=torch.uint8).to(dev)
+    for p in range(nparts):
+        batch.add(hdr_stage.data_ptr() + p * hlen, base + positions[p], hlen)
+    for src, dst_off, nb in seg_plan:
+        batch.add(src, base + dst_off, nb)
+    keep = batch.run(dev, g, stream)
+    return out, sizes
+
+
+def assemble_from_device(buffers: Sequence[torch.Tensor],
+                         schema: Sequence[Column]) -> Table:
+    """Merge device Kudo records into one device table (inverse of split)."""
+    _check_flat(schema)
+    g = _native.gpu()
+    stream = _native.current_stream()
+    dev = buffers[0].device
+    ncols = len(schema)
+    hlen = 28 + (ncols + 7) // 8
+
+    # parse headers (small D2H)
+    headers: List[KudoTableHeader] = []
+    for b in buffers:
+        import io
+        raw = b[:hlen].cpu().numpy().tobytes()
+        h = KudoTableHeader.read(io.BytesIO(raw))
+        headers.append(h)
+
+    # per piece/column section positions (record-relative, host arithmetic)
+    # order: validities | pad | offsets | data
+    piece_cols = []  # [piece][col] = dict(valid_ptr, off_ptr, data_ptr?, ...)
+    gather_addrs = []  # addresses of first/last offset values per string piece
+    for b, h in zip(buffers, headers):
+        base = b.data_ptr()
+        start_bit = h.offset % 8
+        vpos = hlen
+        opos = hlen + h.validity_len
+        dpos = opos + h.offset_len
+        cols = []
+        for ci, c in enumerate(schema):
+            rec = {"n": h.num_rows, "start_bit": start_bit}
+            if h.has_validity_buffer(ci):
+                nb = (start_bit + h.num_rows + 7) // 8 if h.num_rows else 1
+                rec["valid_ptr"] = base + vpos
+                vpos += nb
+            if c.dtype == DType.STRING:
+                if h.num_rows > 0:
+                    rec["off_ptr"] = base + opos
+                    opos += (h.num_rows + 1) * 4
+                    gather_addrs.append(rec["off_ptr"])
+                    gather_addrs.append(rec["off_ptr"] + h.num_rows * 4)
+            cols.append(rec)
+        # data positions need char counts -> fill after gather
+        rec_meta = {"base": base, "dpos": dpos, "cols": cols}
+        piece_cols.append(rec_meta)
+
+    # fetch first/last offsets of every string piece in one kernel + D2H
+    char_counts = {}
+    if gather_addrs:
+        addrs = torch.tensor(gather_addrs, dtype=torch.int64, device=dev)
+        vals = torch.empty(len(gather_addrs), dtype=torch.int32, device=dev)
+        g.gather_i32_at(addrs.data_ptr(), len(gather_addrs), vals.data_ptr(),
+                        stream)
+        flat = vals.cpu().tolist()
+        k = 0
+        for pi, meta in enumerate(piece_cols):
+            for ci, c in enumerate(schema):
+                if c.dtype == DType.STRING and "off_ptr" in meta["cols"][ci]:
+                    first, last = flat[k], flat[k + 1]
+                    k += 2
+                    char_counts[(pi, ci)] = (first, last - first)
+
+    # walk data section per piece
+    for pi, meta in enumerate(piece_cols):
+        dpos = meta["dpos"]
+        for ci, c in enumerate(schema):
+            rec = meta["cols"][ci]
+            if rec["n"] == 0:
+                continue
+            if c.dtype == DType.STRING:
+                first, nchars = char_counts.get((pi, ci), (0, 0))
+                rec["first_off"] = first
+                if nchars > 0:
+                    rec["data_ptr"] = meta["base"] + dpos
+                    rec["nchars"] = nchars
+                    dpos += nchars
+            else:
+                w = FIXED_WIDTH[c.dtype]
+                rec["data_ptr"] = meta["base"] + dpos
+                dpos += rec["n"] * w
+
+    total_rows = sum(h.num_rows for h in headers)
+    out_cols = []
+    copy_batch = _SegBatch()
+    vsegs, vwords = [], []
+    osegs, ocounts = [], []
+    keepalive = []
+    for ci, c in enumerate(schema):
+        any_valid = any("valid_ptr" in meta["cols"][ci] for meta in piece_cols)
+        validity = None
+        if any_valid:
+            validity = torch.zeros(validity_nbytes(total_rows), dtype=torch.uint8,
+                                   device=dev)
+        if c.dtype == DType.STRING:
+            out_offs = torch.empty(total_rows + 1, dtype=torch.int32, device=dev)
+            nchars_total = sum(char_counts.get((pi, ci), (0, 0))[1]
+                               for pi in range(len(piece_cols)))
+            out_chars = torch.empty(max(nchars_total, 1), dtype=torch.uint8,
+                                    device=dev)
+        else:
+            from .columnar import TORCH_DTYPE
+            numel = total_rows * (2 if c.dtype == DType.DECIMAL128 else 1)
+            out_data = torch.empty(numel, dtype=TORCH_DTYPE[c.dtype],
+                                   device=dev)
+        row_pos = 0
+        char_pos = 0
+        for pi, meta in enumerate(piece_cols):
+            rec = meta["cols"][ci]
+            n = rec["n"]
+            if validity is not None and n > 0:
+                src = rec.get("valid_ptr", 0)
+                sbit = rec["start_bit"] if src else 0
+                vsegs.append(struct.pack(_VALIDSEG, src, validity.data_ptr(),
+                                         sbit, row_pos, n))
+                w0 = row_pos // 64
+                w1 = (row_pos + n - 1) // 64
+                vwords.append(w1 - w0 + 1)
+            if c.dtype == DType.STRING and n > 0:
+                nch = char_counts.get((pi, ci), (0, 0))[1]
+                is_last = row_pos + n == total_rows
+                osegs.append(struct.pack(
+                    _OFFSEG, rec["off_ptr"], out_offs.data_ptr() + row_pos * 4,
+                    n, char_pos, 1 if is_last else 0))
+                ocounts.append(n + (1 if is_last else 0))
+                if nch > 0:
+                    copy_batch.add(rec["data_ptr"],
+                                   out_chars.data_ptr() + char_pos, nch)
+                char_pos += nch
+            elif n > 0 and c.dtype != DType.STRING:
+                w = FIXED_WIDTH[c.dtype]
+                copy_batch.add(rec["data_ptr"], out_data.data_ptr() + row_pos * w,
+                               n * w)
+            row_pos += n
+        if c.dtype == DType.STRING:
+            if total_rows == 0:
+                out_offs = torch.zeros(1, dtype=torch.int32, device=dev)
+            out_cols.append(Column(DType.STRING, total_rows, out_chars, validity,
+                                   out_offs, null_count=None))
+        else:
+            out_cols.append(Column(c.dtype, total_rows, out_data, validity,
+                                   scale=c.scale, null_count=None))
+
+    keepalive.append(copy_batch.run(dev, g, stream))
+    if vsegs:
+        raw = b"".join(vsegs)
+        prefix = np.zeros(len(vwords), dtype=np.int64)
+        np.cumsum(vwords[:-1], out=prefix[1:])
+        total_words = int(prefix[-1] + vwords[-1])
+        segs_t = torch.frombuffer(bytearray(raw), dtype=torch.uint8).to(dev)
+        pref_t = torch.from_numpy(prefix).to(dev)
+        g.validity_merge(segs_t.data_ptr(), pref_t.data_ptr(), len(vwords),
+                         total_words, stream)
+        keepalive.append((segs_t, pref_t))
+    if osegs:
+        raw = b"".join(osegs)
+        prefix = np.zeros(len(ocounts), dtype=np.int64)
+        np.cumsum(ocounts[:-1], out=prefix[1:])
+        total = int(prefix[-1] + ocounts[-1])
+        segs_t = torch.frombuffer(bytearray(raw), dtype=torch.uint8).to(dev)
+        pref_t = torch.from_numpy(prefix).to(dev)
+        g.offsets_rebase(segs_t.data_ptr(), pref_t.data_ptr(), len(ocounts),
+                         total, stream)
+        keepalive.append((segs_t, pref_t))
+    torch.cuda.synchronize()  # keepalive tensors may go out of scope
+    return Table(out_cols)
+
+
+def assemble_from_device_raw(buffer: torch.Tensor, sizes: Sequence[int],
+                             schema: Sequence[Column]) -> Table:
+    """KudoGpuSerializer.assembleFromDeviceRaw: one concatenated buffer."""
+    views = []
+    pos = 0
+    for s in sizes:
+        views.append(buffer[pos:pos + s])
+        pos += s
+    return assemble_from_device(views, schema)

@@ -3,7 +3,6 @@
 #include "srj_bind.hpp"
 
 void register_rowconv(py::module_&) {}
-void register_shuffle(py::module_&) {}
 void register_cast(py::module_&) {}
 void register_datetime(py::module_&) {}
 void register_json(py::module_&) {}
