@@ -70,7 +70,8 @@ def convert_to_rows(table: Table) -> List[Tuple[torch.Tensor, int]]:
                 c.dtype, m, c.data[start * elems_per_row:],
                 c.validity[start // 8:] if c.validity is not None else None,
                 scale=c.scale))
-        buf = torch.empty(max(m * row_size, 1), dtype=torch.uint8, device=dev)
+        # zeros: padding bytes are part of the format contract
+        buf = torch.zeros(max(m * row_size, 1), dtype=torch.uint8, device=dev)
         desc = _pack_desc(views, offs, dev)
         g.to_rows(desc.data_ptr(), len(cols), m, row_size, validity_off,
                   buf.data_ptr(), stream)

@@ -153,7 +153,7 @@ def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
     base = out.data_ptr()
     batch = _SegBatch()
     # headers: staged once, copied into place
-    hdr_stage = torch.frombuffer(bytes(headers) or b"\x00",
+    hdr_stage = torch.frombuffer(headers or bytearray(1),
                                  dtype=torch.uint8).to(dev)
     for p in range(nparts):
         batch.add(hdr_stage.data_ptr() + p * hlen, base + positions[p], hlen)
