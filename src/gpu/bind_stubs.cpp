@@ -2,7 +2,6 @@
 // replaced by its own bind_*.cpp as the corresponding kernels land.
 #include "srj_bind.hpp"
 
-void register_cast(py::module_&) {}
 void register_datetime(py::module_&) {}
 void register_json(py::module_&) {}
 void register_misc(py::module_&) {}

@@ -1,0 +1,300 @@
+"""Cast kernel tests vs Spark-semantics Python oracles."""
+import datetime
+import math
+import random
+import struct
+from decimal import ROUND_HALF_UP, Decimal, InvalidOperation, localcontext
+
+import pytest
+import torch
+
+from spark_rapids_jni_amd.columnar import Column, DType
+
+random.seed(47)
+
+
+# --- oracles ----------------------------------------------------------------
+
+def oracle_int(s, bits):
+    if s is None:
+        return None
+    t = s.strip(''.join(chr(c) for c in range(0x21)))
+    if not t:
+        return None
+    i, neg = 0, False
+    if t[0] in "+-":
+        neg = t[0] == "-"
+        i = 1
+    digits = ""
+    seen_dot = False
+    for j in range(i, len(t)):
+        c = t[j]
+        if c == ".":
+            if not t[j + 1:].isdigit() and t[j + 1:] != "":
+                return None
+            if not all(ch.isdigit() for ch in t[j + 1:]):
+                return None
+            seen_dot = True
+            break
+        if not c.isdigit():
+            return None
+        digits += c
+    if not digits:
+        return None
+    v = int(digits) * (-1 if neg else 1)
+    lo, hi = -(1 << (bits - 1)), (1 << (bits - 1)) - 1
+    if v < lo or v > hi:
+        return None
+    return v
+
+
+BOOL_TRUE = {"t", "true", "y", "yes", "1"}
+BOOL_FALSE = {"f", "false", "n", "no", "0"}
+
+
+def oracle_bool(s):
+    if s is None:
+        return None
+    t = s.strip(''.join(chr(c) for c in range(0x21))).lower()
+    if t in BOOL_TRUE:
+        return True
+    if t in BOOL_FALSE:
+        return False
+    return None
+
+
+def oracle_float(s):
+    if s is None:
+        return None
+    t = s.strip(''.join(chr(c) for c in range(0x21)))
+    if not t:
+        return None
+    sign = 1.0
+    body = t
+    if body[:1] in "+-":
+        sign = -1.0 if body[0] == "-" else 1.0
+        body = body[1:]
+    low = body.lower()
+    if low in ("inf", "infinity"):
+        return sign * math.inf
+    if low == "nan":
+        return math.nan
+    if low and low[-1] in "dDfF" and not low.endswith("inf"):
+        body = body[:-1]
+        low = low[:-1]
+    try:
+        v = float(body)
+    except ValueError:
+        return None
+    if "x" in low or "_" in low or low.startswith("n") or low.startswith("i"):
+        return None
+    return sign * v
+
+
+def oracle_decimal(s, precision, scale):
+    if s is None:
+        return None
+    t = s.strip(''.join(chr(c) for c in range(0x21)))
+    if not t:
+        return None
+    try:
+        with localcontext() as ctx:
+            ctx.prec = 60
+            d = Decimal(t)
+            q = d.quantize(Decimal(1).scaleb(-scale), rounding=ROUND_HALF_UP)
+    except (InvalidOperation, ValueError):
+        return None
+    unscaled = int(q.scaleb(scale))
+    if abs(unscaled) >= 10 ** precision:
+        return None
+    return unscaled
+
+
+# --- tests ------------------------------------------------------------------
+
+INT_CASES = ["0", "1", "-1", "+42", "  17  ", "2147483647", "2147483648",
+             "-2147483648", "-2147483649", "9223372036854775807",
+             "9223372036854775808", "-9223372036854775808", "1.5", "-3.99",
+             "1.5.2", "1.", ".5", "", " ", "abc", "12a", "+", "-",
+             "127", "128", "-128", "-129", "32767", "32768", None, "00012",
+             "1e3", "  -00  "]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dtype,bits", [(DType.INT8, 8), (DType.INT16, 16),
+                                        (DType.INT32, 32), (DType.INT64, 64)])
+def test_string_to_int(dtype, bits):
+    from spark_rapids_jni_amd.ops import cast
+    col = Column.from_pylist(INT_CASES, DType.STRING, "cuda")
+    got = cast.to_integer(col, dtype=dtype).to_pylist()
+    for s, gv in zip(INT_CASES, got):
+        assert gv == oracle_int(s, bits), f"{s!r} ({bits}b): {gv}"
+
+
+@pytest.mark.gpu
+def test_string_to_int_ansi_reports_first_bad_row():
+    from spark_rapids_jni_amd.ops import cast
+    vals = ["1", "2", "bad", "4", "worse"]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    with pytest.raises(cast.CastException) as ei:
+        cast.to_integer(col, ansi=True)
+    assert ei.value.row_with_error == 2
+
+
+@pytest.mark.gpu
+def test_string_to_bool():
+    from spark_rapids_jni_amd.ops import cast
+    vals = ["true", "TRUE", "t", "y", "yes", "1", "false", "F", "n", "no", "0",
+            " true ", "tr", "2", "", None, "yess"]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    got = cast.to_bool(col).to_pylist()
+    for s, gv in zip(vals, got):
+        assert gv == oracle_bool(s), f"{s!r}"
+
+
+FLOAT_CASES = ["0", "-0.0", "1.5", "  3.25  ", "1e10", "1E-10", "-2.5e3",
+               "3.4028235e38", "1.7976931348623157e308", "1e309", "-1e309",
+               "1e-400", "inf", "-inf", "Infinity", "-INFINITY", "NaN", "nan",
+               ".5", "5.", "1.5d", "2.5f", "1.5D", "", "abc", "1..2", "1e",
+               "e5", None, "0.1", "123456789.123456789", "-7",
+               "4.9e-324", "2.2250738585072014e-308"]
+
+
+@pytest.mark.gpu
+def test_string_to_double():
+    from spark_rapids_jni_amd.ops import cast
+    col = Column.from_pylist(FLOAT_CASES, DType.STRING, "cuda")
+    got = cast.to_float(col, dtype=DType.FLOAT64).to_pylist()
+    for s, gv in zip(FLOAT_CASES, got):
+        exp = oracle_float(s)
+        if exp is None:
+            assert gv is None, f"{s!r} -> {gv}"
+        elif math.isnan(exp):
+            assert gv is not None and math.isnan(gv), f"{s!r}"
+        elif exp == 0 or math.isinf(exp):
+            assert gv == exp and math.copysign(1, gv) == math.copysign(1, exp), f"{s!r}"
+        else:
+            # within 1 ulp (documented gap vs bit-exact reference)
+            assert gv is not None
+            assert abs(gv - exp) <= abs(exp) * 2.3e-16, f"{s!r}: {gv} != {exp}"
+
+
+@pytest.mark.gpu
+def test_string_to_decimal():
+    from spark_rapids_jni_amd.ops import cast
+    cases = ["0", "1.23", "-1.23", "999.99", "1000.00", "0.005", "-0.005",
+             "12.345", "1.2e2", "  7.5 ", "", "abc", None, "99999.99",
+             "-99999.99", "123456.78"]
+    precision, scale = 7, 2
+    col = Column.from_pylist(cases, DType.STRING, "cuda")
+    out = cast.to_decimal(col, precision, scale)
+    got = out.to_pylist()
+    assert out.dtype == DType.DECIMAL32
+    for s, gv in zip(cases, got):
+        assert gv == oracle_decimal(s, precision, scale), f"{s!r}: {gv}"
+
+
+@pytest.mark.gpu
+def test_string_to_decimal128():
+    from spark_rapids_jni_amd.ops import cast
+    cases = ["123456789012345678901234567.123", "-99999999999999999999.5", "1"]
+    col = Column.from_pylist(cases, DType.STRING, "cuda")
+    out = cast.to_decimal(col, 38, 3)
+    assert out.dtype == DType.DECIMAL128
+    # DECIMAL128 data = 2 int64 words/row little-endian
+    words = out.data.cpu().tolist()
+    for i, s in enumerate(cases):
+        exp = oracle_decimal(s, 38, 3)
+        lo, hi = words[2 * i] & (2**64 - 1), words[2 * i + 1]
+        got = (hi << 64) | lo
+        if got >= 2**127:
+            got -= 2**128
+        assert got == exp, f"{s}: {got} != {exp}"
+
+
+def _date_oracle(s):
+    if s is None:
+        return None
+    t = s.strip()
+    try:
+        parts = t.split("T")[0].split(" ")[0].split("-")
+        if t.startswith("-"):
+            return None  # negative years: not in these tests
+        if len(parts) == 1:
+            d = datetime.date(int(parts[0]), 1, 1)
+        elif len(parts) == 2:
+            d = datetime.date(int(parts[0]), int(parts[1]), 1)
+        else:
+            d = datetime.date(int(parts[0]), int(parts[1]), int(parts[2]))
+        return (d - datetime.date(1970, 1, 1)).days
+    except ValueError:
+        return None
+
+
+@pytest.mark.gpu
+def test_string_to_date():
+    from spark_rapids_jni_amd.ops import cast
+    vals = ["2020-01-01", "2020-1-1", "1970-01-01", "2020-02-29", "2019-02-29",
+            "2020-13-01", "2020-00-10", "2020", "2020-06",
+            "2020-06-15T23:59:59", "2020-06-15 garbage-is-ok-after-sep",
+            "", "not a date", None, "1969-12-31", "0001-01-01", "9999-12-31"]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    got = cast.to_date(col).to_pylist()
+    for s, gv in zip(vals, got):
+        exp = _date_oracle(s)
+        assert gv == exp, f"{s!r}: {gv} != {exp}"
+
+
+@pytest.mark.gpu
+def test_string_to_timestamp():
+    from spark_rapids_jni_amd.ops import cast
+    epoch = datetime.datetime(1970, 1, 1, tzinfo=datetime.timezone.utc)
+
+    def us(y, mo, d, h=0, mi=0, s=0, micro=0, tzs=0):
+        t = datetime.datetime(y, mo, d, h, mi, s, micro,
+                              tzinfo=datetime.timezone.utc)
+        return int((t - epoch).total_seconds()) * 1000000 + micro - tzs * 1000000
+
+    cases = {
+        "2020-01-02 03:04:05": us(2020, 1, 2, 3, 4, 5),
+        "2020-01-02T03:04:05.123456": us(2020, 1, 2, 3, 4, 5, 123456),
+        "2020-01-02 03:04:05.1": us(2020, 1, 2, 3, 4, 5, 100000),
+        "2020-01-02": us(2020, 1, 2),
+        "2020-01-02 03:04:05Z": us(2020, 1, 2, 3, 4, 5),
+        "2020-01-02 03:04:05+05:30": us(2020, 1, 2, 3, 4, 5, 0, 5 * 3600 + 1800),
+        "2020-01-02 03:04:05-08:00": us(2020, 1, 2, 3, 4, 5, 0, -8 * 3600),
+        "epoch": 0,
+        "2020-01-02 25:00:00": None,
+        "2020-01-02 03:61:00": None,
+        "junk": None,
+        "": None,
+    }
+    vals = list(cases.keys())
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    got = cast.to_timestamp(col).to_pylist()
+    for s, gv in zip(vals, got):
+        exp = cases[s]
+        micro = exp
+        assert gv == exp, f"{s!r}: {gv} != {exp}"
+
+
+@pytest.mark.gpu
+def test_integer_to_string():
+    from spark_rapids_jni_amd.ops import cast
+    vals = [0, 1, -1, 42, -12345, 2**31 - 1, -2**31, None, 7]
+    col = Column.from_pylist(vals, DType.INT32, "cuda")
+    got = cast.from_integer(col).to_pylist()
+    assert got == [str(v) if v is not None else None for v in vals]
+    bvals = [True, False, None]
+    bcol = Column.from_pylist(bvals, DType.BOOL8, "cuda")
+    assert cast.from_integer(bcol).to_pylist() == ["true", "false", None]
+
+
+@pytest.mark.gpu
+def test_decimal_to_string():
+    from spark_rapids_jni_amd.ops import cast
+    # unscaled values at scale 2: 12345 -> "123.45"
+    col = Column.from_pylist([12345, -12345, 5, -5, 0, None], DType.DECIMAL64,
+                             "cuda", scale=2)
+    got = cast.from_integer(col).to_pylist()
+    assert got == ["123.45", "-123.45", "0.05", "-0.05", "0.00", None]
