@@ -152,6 +152,78 @@ __device__ inline double pow10_pos(int e) {
   return r * tbl[e];
 }
 
+// --- Eisel-Lemire exactly-rounded decimal->double (Lemire, "Number Parsing
+// at a Gigabyte per Second"; same algorithm family as the reference's exact
+// string_to_float path). w = significand (<= 19 digits), q = decimal exp.
+#include "pow5_table.inc"
+
+__device__ inline bool eisel_lemire(uint64_t w, int64_t q, bool neg,
+                                    double* out) {
+  if (w == 0 || q < -342) {
+    *out = neg ? -0.0 : 0.0;
+    return true;
+  }
+  if (q > 308) {
+    *out = neg ? -INFINITY : INFINITY;
+    return true;
+  }
+  int lz = __clzll((long long)w);
+  w <<= lz;
+  const unsigned long long* p5 = SRJ_POW5[q + 342];
+  // 128-bit product approximation with 55-bit precision requirement
+  uint64_t lo = w * p5[0];
+  uint64_t hi = __umul64hi(w, p5[0]);
+  constexpr uint64_t precision_mask = 0x1FFull;  // 64 - (53 + 3) + 1... (2^9-1)
+  if ((hi & precision_mask) == precision_mask) {
+    uint64_t lo2 = w * p5[1];
+    uint64_t hi2 = __umul64hi(w, p5[1]);
+    uint64_t sum = lo + hi2;
+    if (sum < lo) ++hi;
+    lo = sum;
+    if ((hi & precision_mask) == precision_mask && lo + w < lo) {
+      return false;  // ambiguous: caller falls back
+    }
+    (void)lo2;
+  }
+  uint64_t upperbit = hi >> 63;
+  uint64_t mantissa = hi >> (upperbit + 9);  // 55 bits (53 + 2 round bits)
+  // power(q) + upperbit - lz - minimum_exponent(-1023)
+  int32_t power2 = (int32_t)(((152170LL + 65536LL) * q) >> 16) + 63 +
+                   (int32_t)upperbit - lz + 1023;
+  if (power2 <= 0) {  // subnormal or zero
+    if (-power2 + 1 >= 64) {
+      *out = neg ? -0.0 : 0.0;
+      return true;
+    }
+    mantissa >>= -power2 + 1;
+    mantissa += mantissa & 1;
+    mantissa >>= 1;
+    power2 = mantissa >= (1ull << 52) ? 1 : 0;
+    uint64_t bits = mantissa | ((uint64_t)power2 << 52) |
+                    ((uint64_t)neg << 63);
+    __builtin_memcpy(out, &bits, 8);
+    return true;
+  }
+  // round-ties-to-even edge: product exactly halfway
+  if (lo <= 1 && q >= -4 && q <= 23 && (mantissa & 3) == 1) {
+    if ((mantissa << (upperbit + 9)) == hi) mantissa &= ~1ull;
+  }
+  mantissa += mantissa & 1;
+  mantissa >>= 1;
+  if (mantissa >= (1ull << 53)) {
+    mantissa = 1ull << 52;
+    ++power2;
+  }
+  mantissa &= ~(1ull << 52);
+  if (power2 >= 0x7FF) {
+    *out = neg ? -INFINITY : INFINITY;
+    return true;
+  }
+  uint64_t bits = mantissa | ((uint64_t)power2 << 52) | ((uint64_t)neg << 63);
+  __builtin_memcpy(out, &bits, 8);
+  return true;
+}
+
 __device__ bool parse_double(StrView s, double* out) {
   s = trim_all(s);
   if (s.len == 0) return false;
@@ -186,7 +258,7 @@ __device__ bool parse_double(StrView s, double* out) {
   }
   uint64_t mant = 0;
   int ndig = 0, exp_adj = 0;
-  bool any = false, dot = false;
+  bool any = false, dot = false, dropped_nonzero = false, dropped = false;
   for (; i < s.len; ++i) {
     char c = s.ptr[i];
     if (c >= '0' && c <= '9') {
@@ -195,8 +267,10 @@ __device__ bool parse_double(StrView s, double* out) {
         mant = mant * 10 + (c - '0');
         if (mant) ++ndig;
         if (dot) --exp_adj;
-      } else if (!dot) {
-        ++exp_adj;
+      } else {
+        dropped = true;
+        if (c != '0') dropped_nonzero = true;
+        if (!dot) ++exp_adj;
       }
     } else if (c == '.') {
       if (dot) return false;
@@ -231,14 +305,34 @@ __device__ bool parse_double(StrView s, double* out) {
     }
   }
   if (!any) return false;
+  // exact path: Eisel-Lemire; bracket w/w+1 when digits were truncated
+  double exact;
+  if (!dropped || !dropped_nonzero) {
+    if (eisel_lemire(mant, exp_adj, neg, &exact)) {
+      *out = exact;
+      return true;
+    }
+  } else {
+    double lo_v, hi_v;
+    if (eisel_lemire(mant, exp_adj, neg, &lo_v) &&
+        eisel_lemire(mant + 1, exp_adj, neg, &hi_v)) {
+      int64_t lb, hb;
+      __builtin_memcpy(&lb, &lo_v, 8);
+      __builtin_memcpy(&hb, &hi_v, 8);
+      if (lb == hb) {
+        *out = lo_v;
+        return true;
+      }
+    }
+  }
+  // fallback (rare ambiguous cases): within-1-ulp approximation
   double d = (double)mant;
   if (exp_adj > 0) {
     if (exp_adj > 308 + 19) { *out = neg ? -INFINITY : INFINITY; return true; }
     d *= pow10_pos(exp_adj);
   } else if (exp_adj < 0) {
     int e = -exp_adj;
-    if (e > 342) { *out = neg ? -0.0 : 0.0; return true; }
-    // divide in <=22 exact chunks to limit rounding error
+    if (e > 342 + 19) { *out = neg ? -0.0 : 0.0; return true; }
     while (e > 22) { d /= 1e22; e -= 22; }
     d /= pow10_pos(e);
   }

@@ -102,9 +102,9 @@ def oracle_decimal(s, precision, scale):
             ctx.prec = 60
             d = Decimal(t)
             q = d.quantize(Decimal(1).scaleb(-scale), rounding=ROUND_HALF_UP)
+            unscaled = int(q.scaleb(scale))
     except (InvalidOperation, ValueError):
         return None
-    unscaled = int(q.scaleb(scale))
     if abs(unscaled) >= 10 ** precision:
         return None
     return unscaled
@@ -174,9 +174,39 @@ def test_string_to_double():
         elif exp == 0 or math.isinf(exp):
             assert gv == exp and math.copysign(1, gv) == math.copysign(1, exp), f"{s!r}"
         else:
-            # within 1 ulp (documented gap vs bit-exact reference)
+            # bit-exact (Eisel-Lemire path)
             assert gv is not None
-            assert abs(gv - exp) <= abs(exp) * 2.3e-16, f"{s!r}: {gv} != {exp}"
+            assert struct.pack("<d", gv) == struct.pack("<d", exp), \
+                f"{s!r}: {gv!r} != {exp!r}"
+
+
+@pytest.mark.gpu
+def test_string_to_double_random_bit_exact():
+    from spark_rapids_jni_amd.ops import cast
+    rng = random.Random(1234)
+    cases = []
+    for _ in range(3000):
+        nd = rng.randint(1, 19)
+        digits = "".join(rng.choice("0123456789") for _ in range(nd))
+        dot = rng.randint(0, nd)
+        body = digits[:dot] + "." + digits[dot:] if rng.random() < 0.7 else digits
+        e = rng.randint(-320, 310)
+        s = ("-" if rng.random() < 0.5 else "") + body + \
+            (f"e{e}" if rng.random() < 0.6 else "")
+        cases.append(s)
+    col = Column.from_pylist(cases, DType.STRING, "cuda")
+    got = cast.to_float(col, dtype=DType.FLOAT64).to_pylist()
+    for s, gv in zip(cases, got):
+        try:
+            exp = float(s)
+        except ValueError:
+            exp = None
+        if exp is None:
+            assert gv is None, s
+            continue
+        assert gv is not None, s
+        assert struct.pack("<d", gv) == struct.pack("<d", exp), \
+            f"{s!r}: {gv!r} != {exp!r}"
 
 
 @pytest.mark.gpu
