@@ -4,6 +4,5 @@
 
 void register_datetime(py::module_&) {}
 void register_json(py::module_&) {}
-void register_misc(py::module_&) {}
 void register_parquet(py::module_&) {}
 void register_sort(py::module_&) {}

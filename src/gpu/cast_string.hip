@@ -475,15 +475,6 @@ __global__ void string_to_decimal_kernel(ColDesc in, int64_t nrows,
 // ---------------------------------------------------------------------------
 // string -> date / timestamp
 // ---------------------------------------------------------------------------
-__device__ inline int64_t days_from_civil(int y, int m, int d) {
-  y -= m <= 2;
-  int era = (y >= 0 ? y : y - 399) / 400;
-  unsigned yoe = (unsigned)(y - era * 400);
-  unsigned doy = (153u * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1;
-  unsigned doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
-  return (int64_t)era * 146097 + (int64_t)doe - 719468;
-}
-
 __device__ inline bool valid_ymd(int y, int m, int d) {
   if (m < 1 || m > 12 || d < 1) return false;
   const int dim[12] = {31, 28, 31, 30, 31, 30, 31, 31, 30, 31, 30, 31};

@@ -325,6 +325,16 @@ __device__ __host__ inline uint64_t xxhash64_fixed(uint64_t bits, int width,
   return xxhash64_bytes(buf, width, seed);
 }
 
+// --- proleptic Gregorian civil-date helpers (Howard Hinnant's algorithms) ---
+__device__ __host__ inline int64_t days_from_civil(int y, int m, int d) {
+  y -= m <= 2;
+  int64_t era = (y >= 0 ? y : y - 399) / 400;
+  unsigned yoe = (unsigned)(y - era * 400);
+  unsigned doy = (153u * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1;
+  unsigned doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
+  return (int64_t)era * 146097 + (int64_t)doe - 719468;
+}
+
 // --- 64-bit mix for internal hash tables (not Spark-visible) ---
 __device__ __host__ inline uint64_t mix64(uint64_t x) {
   x ^= x >> 33;
