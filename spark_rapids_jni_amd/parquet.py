@@ -1,0 +1,485 @@
+"""Parquet: host footer/page-header parsing (Thrift compact via the native
+_host parser) + GPU page decode.
+
+Reference parity: NativeParquetJni.cpp (footer parse/prune with
+TCompactProtocol, column pruning, row-group filtering by split offset with
+preserved row indexes) + the page decode the reference delegates to libcudf
+(built fresh here for CDNA4 — SURVEY.md §7 item 5).
+
+Scope v1: flat schemas; physical types BOOLEAN/INT32/INT64/FLOAT/DOUBLE/
+BYTE_ARRAY; PLAIN + RLE_DICTIONARY/PLAIN_DICTIONARY encodings; UNCOMPRESSED
+and SNAPPY (host-decompressed) pages; data page V1 and V2.
+"""
+import struct
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from . import _native
+from .columnar import Column, DType, Table, make_validity
+
+MAGIC = b"PAR1"
+
+# parquet physical types
+T_BOOLEAN, T_INT32, T_INT64, T_INT96, T_FLOAT, T_DOUBLE, T_BYTE_ARRAY, \
+    T_FIXED_LEN_BYTE_ARRAY = range(8)
+
+ENC_PLAIN = 0
+ENC_PLAIN_DICTIONARY = 2
+ENC_RLE = 3
+ENC_RLE_DICTIONARY = 8
+
+CODEC_UNCOMPRESSED = 0
+CODEC_SNAPPY = 1
+
+_SCATTER_FMT = "<QQQQqqqii"     # ScatterDesc (64B)
+_RLE_FMT = "<QqQqii"            # RleDesc (40B)
+_STRIDX_FMT = "<QqqQQ"          # StrIndexDesc (40B)
+_STRCPY_FMT = "<QQQQQQqqq"      # StrCopyDesc (72B)
+
+
+@dataclass
+class SchemaField:
+    name: str
+    physical_type: int
+    repetition: int          # 0 required, 1 optional, 2 repeated
+    converted_type: Optional[int]
+    scale: int = 0
+    precision: int = 0
+    logical: Optional[dict] = None
+
+
+@dataclass
+class ColumnChunkMeta:
+    path: Tuple[str, ...]
+    physical_type: int
+    encodings: List[int]
+    codec: int
+    num_values: int
+    total_compressed_size: int
+    total_uncompressed_size: int
+    data_page_offset: int
+    dictionary_page_offset: Optional[int]
+
+    @property
+    def start_offset(self):
+        o = self.data_page_offset
+        if self.dictionary_page_offset is not None and \
+                0 < self.dictionary_page_offset < o:
+            o = self.dictionary_page_offset
+        return o
+
+
+@dataclass
+class RowGroupMeta:
+    columns: List[ColumnChunkMeta]
+    num_rows: int
+    total_byte_size: int
+
+
+@dataclass
+class ParquetFooter:
+    """reference ParquetFooter.java — parsed footer with prune support."""
+    version: int
+    num_rows: int
+    schema: List[SchemaField]
+    row_groups: List[RowGroupMeta]
+    created_by: Optional[str] = None
+
+    def prune(self, keep_columns: Sequence[str]) -> "ParquetFooter":
+        keep = {c.lower() for c in keep_columns}
+        fields = [f for f in self.schema if f.name.lower() in keep]
+        rgs = []
+        for rg in self.row_groups:
+            cols = [c for c in rg.columns if c.path[0].lower() in keep]
+            rgs.append(RowGroupMeta(cols, rg.num_rows, rg.total_byte_size))
+        return ParquetFooter(self.version, self.num_rows, fields, rgs,
+                             self.created_by)
+
+    def filter_row_groups(self, part_offset: int, part_length: int
+                          ) -> "ParquetFooter":
+        """Keep row groups whose midpoint falls in [offset, offset+length)
+        (Spark split semantics, reference NativeParquetJni.cpp:795-810)."""
+        rgs = []
+        for rg in self.row_groups:
+            if not rg.columns:
+                continue
+            start = min(c.start_offset for c in rg.columns)
+            mid = start + rg.total_byte_size // 2
+            if part_offset <= mid < part_offset + part_length:
+                rgs.append(rg)
+        return ParquetFooter(self.version, sum(r.num_rows for r in rgs),
+                             self.schema, rgs, self.created_by)
+
+
+def read_footer(path_or_bytes) -> ParquetFooter:
+    raw = (open(path_or_bytes, "rb").read()
+           if isinstance(path_or_bytes, str) else path_or_bytes)
+    assert raw[:4] == MAGIC and raw[-4:] == MAGIC, "not a parquet file"
+    flen = struct.unpack("<I", raw[-8:-4])[0]
+    fmd, _ = _native.host().thrift_parse(raw[-8 - flen:-8], 0)
+
+    schema_elems = fmd[2]
+    root = schema_elems[0]
+    assert root.get(5, 0) == len(schema_elems) - 1, \
+        "nested parquet schemas not supported yet"
+    fields = []
+    for se in schema_elems[1:]:
+        fields.append(SchemaField(
+            name=se[4].decode(), physical_type=se.get(1, -1),
+            repetition=se.get(3, 0), converted_type=se.get(6),
+            scale=se.get(7, 0), precision=se.get(8, 0), logical=se.get(10)))
+
+    row_groups = []
+    for rg in fmd[4]:
+        cols = []
+        for cc in rg[1]:
+            md = cc[3]
+            cols.append(ColumnChunkMeta(
+                path=tuple(p.decode() for p in md[3]),
+                physical_type=md[1], encodings=list(md[2]), codec=md[4],
+                num_values=md[5], total_compressed_size=md[7],
+                total_uncompressed_size=md[6], data_page_offset=md[9],
+                dictionary_page_offset=md.get(11)))
+        row_groups.append(RowGroupMeta(cols, rg[3], rg[2]))
+    return ParquetFooter(fmd.get(1, 1), fmd[3], fields, row_groups,
+                         fmd.get(6, b"").decode() if 6 in fmd else None)
+
+
+_PHYS_TO_DTYPE = {
+    T_BOOLEAN: DType.BOOL8,
+    T_INT32: DType.INT32,
+    T_INT64: DType.INT64,
+    T_FLOAT: DType.FLOAT32,
+    T_DOUBLE: DType.FLOAT64,
+    T_BYTE_ARRAY: DType.STRING,
+}
+
+_PHYS_WIDTH = {T_INT32: 4, T_INT64: 8, T_FLOAT: 4, T_DOUBLE: 8}
+
+
+def _field_dtype(f: SchemaField) -> DType:
+    if f.physical_type == T_INT32 and f.converted_type == 6:  # DATE
+        return DType.DATE32
+    if f.physical_type == T_INT64 and f.converted_type in (9, 10):
+        return DType.TIMESTAMP_US
+    if f.logical and 2 in (f.logical or {}):
+        pass
+    return _PHYS_TO_DTYPE[f.physical_type]
+
+
+@dataclass
+class _Page:
+    kind: int                # 0 data v1, 1 dict, 2 data v2
+    num_values: int
+    encoding: int
+    data: bytes              # uncompressed page payload
+    def_bytes: int = 0       # v2: definition level byte length
+    num_nulls: int = -1      # v2 only
+
+
+def _decompress(codec, data, uncompressed_size):
+    if codec == CODEC_UNCOMPRESSED:
+        return data
+    if codec == CODEC_SNAPPY:
+        import pyarrow as pa
+        return pa.Codec("snappy").decompress(
+            data, decompressed_size=uncompressed_size).to_pybytes()
+    raise NotImplementedError(f"parquet codec {codec}")
+
+
+def _walk_pages(raw: bytes, chunk: ColumnChunkMeta) -> List[_Page]:
+    h = _native.host()
+    pos = chunk.start_offset
+    pages = []
+    values_seen = 0
+    while values_seen < chunk.num_values:
+        ph, end = h.thrift_parse(raw, pos)
+        ptype = ph[1]
+        uncomp = ph[2]
+        comp = ph[3]
+        payload = raw[end:end + comp]
+        pos = end + comp
+        if ptype == 0:  # DATA_PAGE v1
+            dph = ph[5]
+            data = _decompress(chunk.codec, payload, uncomp)
+            pages.append(_Page(0, dph[1], dph[2], data))
+            values_seen += dph[1]
+        elif ptype == 2:  # DICTIONARY_PAGE
+            dph = ph[7]
+            data = _decompress(chunk.codec, payload, uncomp)
+            pages.append(_Page(1, dph[1], dph.get(2, ENC_PLAIN), data))
+        elif ptype == 3:  # DATA_PAGE_V2
+            dph = ph[8]
+            nv = dph[1]
+            dlen = dph.get(5, 0)
+            rlen = dph.get(6, 0)
+            assert rlen == 0, "nested (repeated) columns not supported yet"
+            levels = payload[:dlen]
+            body = payload[dlen:]
+            if dph.get(7, True) and chunk.codec != CODEC_UNCOMPRESSED:
+                body = _decompress(chunk.codec, body, uncomp - dlen)
+            pages.append(_Page(2, nv, dph[4], levels + body, def_bytes=dlen,
+                               num_nulls=dph.get(2, -1)))
+            values_seen += nv
+        else:
+            pass  # index page: skip
+    return pages
+
+
+def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
+                 total_rows: int, device) -> Column:
+    g = _native.gpu()
+    stream = _native.current_stream()
+    dev = torch.device(device)
+    nullable = f.repetition == 1
+    dtype = _field_dtype(f)
+
+    # flatten pages across row groups
+    pages: List[_Page] = []
+    dict_per_page: List[int] = []   # index into dict list, -1 none
+    dicts: List[_Page] = []
+    for ch in chunks:
+        cur_dict = -1
+        for p in _walk_pages(raw, ch):
+            if p.kind == 1:
+                dicts.append(p)
+                cur_dict = len(dicts) - 1
+            else:
+                pages.append(p)
+                dict_per_page.append(cur_dict)
+
+    # upload all page payloads in one buffer
+    blobs = [p.data for p in pages] + [p.data for p in dicts]
+    offs = np.zeros(len(blobs) + 1, dtype=np.int64)
+    for i, b in enumerate(blobs):
+        offs[i + 1] = offs[i] + ((len(b) + 7) & ~7)
+    big = torch.zeros(max(int(offs[-1]), 1), dtype=torch.uint8, device=dev)
+    hbuf = np.zeros(int(offs[-1]), dtype=np.uint8)
+    for i, b in enumerate(blobs):
+        hbuf[offs[i]:offs[i] + len(b)] = np.frombuffer(b, dtype=np.uint8)
+    if len(hbuf):
+        big[:len(hbuf)] = torch.from_numpy(hbuf)
+    pbase = [big.data_ptr() + int(offs[i]) for i in range(len(blobs))]
+    dict_base_idx = len(pages)
+
+    # page row starts
+    row_starts = np.zeros(len(pages) + 1, dtype=np.int64)
+    for i, p in enumerate(pages):
+        row_starts[i + 1] = row_starts[i] + p.num_values
+    assert row_starts[-1] == total_rows, (row_starts[-1], total_rows)
+
+    # ---- phase 1: def levels -> def bytes per row -------------------------
+    def_t = None
+    vprefix = None
+    page_vbase = [0] * len(pages)
+    if nullable:
+        def_t = torch.ones(max(total_rows, 1), dtype=torch.uint8, device=dev)
+        rle_descs = bytearray()
+        nrle = 0
+        for i, p in enumerate(pages):
+            if p.kind == 0:
+                # v1: def levels = 4-byte len + RLE (bit width 1)
+                (dl,) = struct.unpack_from("<I", p.data, 0)
+                src = pbase[i] + 4
+                src_len = dl
+            else:
+                src = pbase[i]
+                src_len = p.def_bytes
+            rle_descs += struct.pack(_RLE_FMT, src, src_len,
+                                     def_t.data_ptr() + int(row_starts[i]),
+                                     p.num_values, 1, 0)
+            nrle += 1
+        dt = torch.frombuffer(rle_descs or bytearray(1), dtype=torch.uint8).to(dev)
+        g.pq_rle_decode(dt.data_ptr(), nrle, stream)
+        # column-wide exclusive valid count
+        incl = torch.cumsum(def_t.to(torch.int64), 0)
+        vprefix = incl - def_t.to(torch.int64)
+        bases = vprefix[torch.from_numpy(row_starts[:-1]).to(dev)].cpu().tolist() \
+            if len(pages) else []
+        page_vbase = [int(b) for b in bases]
+
+    # ---- phase 2: decode dictionaries & dict indices ----------------------
+    # dict values: PLAIN-encoded; fixed width -> direct pointer; strings ->
+    # index via string_plain_index
+    dict_fixed_ptr = {}
+    dict_str = {}
+    if dicts:
+        sidx_descs = bytearray()
+        scount = 0
+        dict_meta = []
+        for di, dp in enumerate(dicts):
+            base = pbase[dict_base_idx + di]
+            if f.physical_type == T_BYTE_ARRAY:
+                voff = torch.empty(max(dp.num_values, 1), dtype=torch.int64,
+                                   device=dev)
+                vlen = torch.empty(max(dp.num_values, 1), dtype=torch.int32,
+                                   device=dev)
+                sidx_descs += struct.pack(_STRIDX_FMT, base, len(dp.data),
+                                          dp.num_values, voff.data_ptr(),
+                                          vlen.data_ptr())
+                scount += 1
+                dict_str[di] = (base, voff, vlen)
+            else:
+                dict_fixed_ptr[di] = base
+        if scount:
+            st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
+            g.pq_string_plain_index(st.data_ptr(), scount, stream)
+
+    def _body_off(i, p):
+        if p.kind == 2:
+            return p.def_bytes
+        if nullable:
+            (dl,) = struct.unpack_from("<I", p.data, 0)
+            return 4 + dl
+        return 0
+
+    # decode dict indices for dict-encoded pages
+    idx_tensors = {}
+    rle_descs = bytearray()
+    rle_meta = []
+    for i, p in enumerate(pages):
+        if p.encoding in (ENC_PLAIN_DICTIONARY, ENC_RLE_DICTIONARY):
+            nvals = p.num_values
+            body_off = _body_off(i, p)
+            bw = p.data[body_off]
+            idx = torch.empty(max(nvals, 1), dtype=torch.int32, device=dev)
+            idx_tensors[i] = idx
+            rle_descs += struct.pack(_RLE_FMT, pbase[i] + body_off + 1,
+                                     len(p.data) - body_off - 1,
+                                     idx.data_ptr(), nvals, bw, 1)
+            rle_meta.append(i)
+    if rle_meta:
+        rt = torch.frombuffer(rle_descs, dtype=torch.uint8).to(dev)
+        g.pq_rle_decode(rt.data_ptr(), len(rle_meta), stream)
+
+    vprefix_ptr = vprefix.data_ptr() if vprefix is not None else 0
+    def_ptr = def_t.data_ptr() if def_t is not None else 0
+
+    # ---- phase 3: values ---------------------------------------------------
+    if f.physical_type == T_BYTE_ARRAY:
+        # PLAIN pages need value indexing first
+        sidx_descs = bytearray()
+        plain_meta = []
+        for i, p in enumerate(pages):
+            if p.encoding == ENC_PLAIN:
+                bo = _body_off(i, p)
+                nvalid = p.num_values  # upper bound; walker stops at src end
+                voff = torch.empty(max(nvalid, 1), dtype=torch.int64, device=dev)
+                vlen = torch.empty(max(nvalid, 1), dtype=torch.int32, device=dev)
+                sidx_descs += struct.pack(_STRIDX_FMT, pbase[i] + bo,
+                                          len(p.data) - bo, nvalid,
+                                          voff.data_ptr(), vlen.data_ptr())
+                plain_meta.append((i, voff, vlen))
+        if plain_meta:
+            st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
+            g.pq_string_plain_index(st.data_ptr(), len(plain_meta), stream)
+        pm = {i: (voff, vlen) for i, voff, vlen in plain_meta}
+
+        cp_descs = bytearray()
+        for i, p in enumerate(pages):
+            bo = _body_off(i, p)
+            if p.encoding == ENC_PLAIN:
+                voff, vlen = pm[i]
+                cp_descs += struct.pack(
+                    _STRCPY_FMT, pbase[i] + bo, voff.data_ptr(), vlen.data_ptr(),
+                    0, def_ptr, vprefix_ptr, int(row_starts[i]), p.num_values,
+                    page_vbase[i])
+            else:
+                di = dict_per_page[i]
+                base, voff, vlen = dict_str[di]
+                cp_descs += struct.pack(
+                    _STRCPY_FMT, base, voff.data_ptr(), vlen.data_ptr(),
+                    idx_tensors[i].data_ptr(), def_ptr, vprefix_ptr,
+                    int(row_starts[i]), p.num_values, page_vbase[i])
+        ct = torch.frombuffer(cp_descs or bytearray(1), dtype=torch.uint8).to(dev)
+        lens = torch.empty(max(total_rows, 1), dtype=torch.int32, device=dev)
+        g.pq_string_copy(ct.data_ptr(), len(pages), 0, lens.data_ptr(), 0, 0,
+                         stream)
+        offsets = torch.zeros(total_rows + 1, dtype=torch.int32, device=dev)
+        if total_rows:
+            torch.cumsum(lens[:total_rows], 0, out=offsets[1:].view(total_rows))
+        nchars = int(offsets[-1].item())
+        chars = torch.empty(max(nchars, 1), dtype=torch.uint8, device=dev)
+        g.pq_string_copy(ct.data_ptr(), len(pages), 1, 0, offsets.data_ptr(),
+                         chars.data_ptr(), stream)
+        validity = None
+        if nullable:
+            validity = make_validity(total_rows, dev)
+            g.pq_def_to_validity(def_ptr, total_rows, validity.data_ptr(), stream)
+        return Column(DType.STRING, total_rows, chars[:nchars], validity,
+                      offsets, null_count=None)
+
+    if f.physical_type == T_BOOLEAN:
+        # PLAIN booleans: bit-packed LE; decode via rle path is not right --
+        # use a simple host fallback for this rare type
+        out = torch.zeros(max(total_rows, 1), dtype=torch.int8, device=dev)
+        host_vals = []
+        for i, p in enumerate(pages):
+            bo = _body_off(i, p)
+            bits = np.unpackbits(np.frombuffer(p.data[bo:], dtype=np.uint8),
+                                 bitorder="little")
+            host_vals.append(bits)
+        if host_vals:
+            allv = np.concatenate(host_vals)
+        # place values accounting for nulls on host (small columns expected)
+        if nullable:
+            defs = def_t.cpu().numpy()
+            vals = np.zeros(total_rows, dtype=np.int8)
+            vi = 0
+            for r in range(total_rows):
+                if defs[r]:
+                    vals[r] = allv[vi]
+                    vi += 1
+            out[:total_rows] = torch.from_numpy(vals).to(dev)
+        else:
+            out[:total_rows] = torch.from_numpy(
+                allv[:total_rows].astype(np.int8)).to(dev)
+        validity = None
+        if nullable:
+            validity = make_validity(total_rows, dev)
+            g.pq_def_to_validity(def_ptr, total_rows, validity.data_ptr(), stream)
+        return Column(DType.BOOL8, total_rows, out[:total_rows], validity,
+                      null_count=None)
+
+    width = _PHYS_WIDTH[f.physical_type]
+    from .columnar import TORCH_DTYPE
+    out = torch.zeros(max(total_rows, 1), dtype=TORCH_DTYPE[dtype], device=dev)
+    sc_descs = bytearray()
+    for i, p in enumerate(pages):
+        bo = _body_off(i, p)
+        if p.encoding == ENC_PLAIN:
+            sc_descs += struct.pack(_SCATTER_FMT, pbase[i] + bo, 0, def_ptr,
+                                    vprefix_ptr, int(row_starts[i]),
+                                    p.num_values, page_vbase[i], width, 0)
+        else:
+            di = dict_per_page[i]
+            sc_descs += struct.pack(_SCATTER_FMT, idx_tensors[i].data_ptr(),
+                                    dict_fixed_ptr[di], def_ptr, vprefix_ptr,
+                                    int(row_starts[i]), p.num_values,
+                                    page_vbase[i], width, 1)
+    st = torch.frombuffer(sc_descs or bytearray(1), dtype=torch.uint8).to(dev)
+    g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(), stream)
+    validity = None
+    if nullable:
+        validity = make_validity(total_rows, dev)
+        g.pq_def_to_validity(def_ptr, total_rows, validity.data_ptr(), stream)
+    return Column(dtype, total_rows, out[:total_rows], validity,
+                  scale=0, null_count=None)
+
+
+def read_table(path: str, columns: Optional[Sequence[str]] = None,
+               device="cuda") -> Table:
+    """Scan a parquet file into a GPU Table (footer + page decode)."""
+    raw = open(path, "rb").read()
+    footer = read_footer(raw)
+    if columns is not None:
+        footer = footer.prune(columns)
+    total_rows = sum(rg.num_rows for rg in footer.row_groups)
+    cols = []
+    for fi, f in enumerate(footer.schema):
+        chunks = [rg.columns[fi] for rg in footer.row_groups]
+        cols.append(_read_column(raw, f, chunks, total_rows, device))
+    return Table(cols)

@@ -4,5 +4,4 @@
 
 void register_datetime(py::module_&) {}
 void register_json(py::module_&) {}
-void register_parquet(py::module_&) {}
 void register_sort(py::module_&) {}
