@@ -175,3 +175,27 @@ def from_integer(col: Column) -> Column:
                         chars.data_ptr(), validity.data_ptr(), stream)
     return Column(DType.STRING, n, chars[:nchars], validity, offsets,
                   null_count=None)
+
+
+def from_floats(col: Column) -> Column:
+    """float/double -> string, Java Double.toString/Float.toString format
+    (Ryu shortest round-trip; reference ftos_converter.cuh)."""
+    g = _native.gpu()
+    stream = _native.current_stream()
+    n = col.size
+    dev = col.device
+    width = FIXED_WIDTH[col.dtype]
+    lens = torch.empty(n, dtype=torch.int32, device=dev)
+    vptr = col.validity.data_ptr() if col.validity is not None else 0
+    g.float_to_string(col.data.data_ptr(), vptr, n, width, 0, lens.data_ptr(),
+                      0, 0, 0, stream)
+    offsets = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+    torch.cumsum(lens, 0, out=offsets[1:].view(n))
+    nchars = int(offsets[-1].item())
+    chars = torch.empty(max(nchars, 1), dtype=torch.uint8, device=dev)
+    validity = make_validity(n, dev)
+    g.float_to_string(col.data.data_ptr(), vptr, n, width, 1, 0,
+                      offsets.data_ptr(), chars.data_ptr(), validity.data_ptr(),
+                      stream)
+    return Column(DType.STRING, n, chars[:nchars], validity, offsets,
+                  null_count=None)

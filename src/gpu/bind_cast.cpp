@@ -16,6 +16,8 @@ void srj_string_to_timestamp(const void*, int64_t, int64_t, int64_t, int64_t,
                              int64_t*, uint8_t*, int64_t*, hipStream_t);
 void srj_integer_to_string(const void*, int64_t, int32_t, int32_t*,
                            const int32_t*, char*, uint8_t*, hipStream_t);
+void srj_float_to_string(const void*, const uint8_t*, int64_t, int32_t, int32_t,
+                         int32_t*, const int32_t*, char*, uint8_t*, hipStream_t);
 }
 
 void register_cast(py::module_& m) {
@@ -66,6 +68,16 @@ void register_cast(py::module_& m) {
                             as_ptr<int64_t>(out), as_ptr<uint8_t>(valid),
                             as_ptr<int64_t>(err), as_stream(stream));
     check_hip("string_to_timestamp");
+  });
+  m.def("float_to_string", [](uintptr_t in, uintptr_t valid, int64_t n,
+                              int32_t width, int32_t phase, uintptr_t lens,
+                              uintptr_t offsets, uintptr_t chars,
+                              uintptr_t out_valid, uintptr_t stream) {
+    srj_float_to_string(as_ptr<void>(in), as_ptr<uint8_t>(valid), n, width, phase,
+                        as_ptr<int32_t>(lens), as_ptr<int32_t>(offsets),
+                        as_ptr<char>(chars), as_ptr<uint8_t>(out_valid),
+                        as_stream(stream));
+    check_hip("float_to_string");
   });
   m.def("integer_to_string", [](uintptr_t in, int64_t n, int32_t phase,
                                 uintptr_t lens, uintptr_t offsets, uintptr_t chars,

@@ -328,3 +328,85 @@ def test_decimal_to_string():
                              "cuda", scale=2)
     got = cast.from_integer(col).to_pylist()
     assert got == ["123.45", "-123.45", "0.05", "-0.05", "0.00", None]
+
+
+def _java_double_str(v):
+    import math
+    from decimal import Decimal
+    if math.isnan(v):
+        return "NaN"
+    if math.isinf(v):
+        return "Infinity" if v > 0 else "-Infinity"
+    if v == 0:
+        return "-0.0" if math.copysign(1, v) < 0 else "0.0"
+    neg = v < 0
+    d = Decimal(repr(abs(v)))
+    sign, digits, exp = d.as_tuple()
+    ds = "".join(map(str, digits)).rstrip("0") or "0"
+    exp += len(digits) - len(ds)
+    point = len(ds) + exp
+    if 0 < point <= 7:
+        intpart = ds[:point].ljust(point, "0") if len(ds) >= point else \
+            ds.ljust(point, "0")
+        frac = ds[point:] or "0"
+        s = f"{intpart}.{frac}"
+    elif -3 < point <= 0:
+        s = "0." + "0" * (-point) + ds
+    else:
+        frac = ds[1:] or "0"
+        s = f"{ds[0]}.{frac}E{point - 1}"
+    return ("-" if neg else "") + s
+
+
+@pytest.mark.gpu
+def test_double_to_string():
+    from spark_rapids_jni_amd.ops import cast
+    vals = [0.0, -0.0, 1.0, -1.5, 3.14159, 1e7, 9999999.0, 1e-3, 1e-4,
+            123000.0, float("nan"), float("inf"), float("-inf"),
+            1.7976931348623157e308, 4.9e-324, 2.2250738585072014e-308,
+            0.1, 1/3, None, 12345.678]
+    col = Column.from_pylist(vals, DType.FLOAT64, "cuda")
+    got = cast.from_floats(col).to_pylist()
+    for v, gs in zip(vals, got):
+        if v is None:
+            assert gs is None
+            continue
+        assert gs == _java_double_str(v), f"{v!r}: {gs!r}"
+
+
+@pytest.mark.gpu
+def test_double_to_string_random_roundtrip():
+    from spark_rapids_jni_amd.ops import cast
+    rng = random.Random(555)
+    vals = []
+    for _ in range(5000):
+        b = rng.getrandbits(64)
+        (v,) = struct.unpack("<d", struct.pack("<Q", b))
+        if v != v or v in (float("inf"), float("-inf")):
+            continue
+        vals.append(v)
+    col = Column.from_pylist(vals, DType.FLOAT64, "cuda")
+    got = cast.from_floats(col).to_pylist()
+    for v, gs in zip(vals, got):
+        assert gs == _java_double_str(v), f"{v!r} ({struct.pack('<d', v).hex()}): {gs!r}"
+
+
+@pytest.mark.gpu
+def test_float_to_string_roundtrip():
+    from spark_rapids_jni_amd.ops import cast
+    rng = random.Random(777)
+    vals = [1.5, 0.1, 100.0, 3.4028235e38, 1.17549435e-38, 1.4e-45]
+    for _ in range(3000):
+        b = rng.getrandbits(32)
+        (v,) = struct.unpack("<f", struct.pack("<I", b))
+        if v != v or v in (float("inf"), float("-inf")):
+            continue
+        vals.append(v)
+    col = Column.from_pylist(vals, DType.FLOAT32, "cuda")
+    got = cast.from_floats(col).to_pylist()
+    for v, gs in zip(vals, got):
+        # shortest round-trip property: parses back to the same float32
+        f = struct.unpack("<f", struct.pack("<f", float(gs)))[0]
+        assert struct.pack("<f", f) == struct.pack("<f", v), f"{v!r}: {gs!r}"
+        # and is it in Java format shape
+        assert "." in gs
