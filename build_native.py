@@ -98,7 +98,7 @@ def build(verbose=False):
                 outs.append(obj)
         return outs
 
-    gpu_headers = sorted(SRC_GPU.glob("*.hpp"))
+    gpu_headers = sorted(SRC_GPU.glob("*.hpp")) + sorted(SRC_GPU.glob("*.inc"))
     host_headers = sorted(SRC_HOST.glob("*.hpp"))
 
     jobs = []
