@@ -3,5 +3,4 @@
 #include "srj_bind.hpp"
 
 void register_datetime(py::module_&) {}
-void register_json(py::module_&) {}
 void register_sort(py::module_&) {}

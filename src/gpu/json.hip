@@ -1,0 +1,471 @@
+// JSON ops: device pull parser + get_json_object (JSONPath) + from_json.
+//
+// Reference parity: json_parser.cuh (1,710 LoC device pull parser),
+// get_json_object.cu (1,262 — path instructions with named keys, indices and
+// [*] wildcards; Hive/Spark semantics), from_json_to_structs.cu,
+// from_json_to_raw_map.cu. Fresh MI355X implementation:
+//   * one thread per row (JSON docs are row-sized; wave64 validity writes)
+//   * two-phase string output (sizes -> cumsum -> write)
+//   * Spark result semantics: single string match -> unescaped value;
+//     container match -> raw JSON span; multiple wildcard matches ->
+//     "[m1,m2,...]"; missing/invalid -> null.
+#include "srj_common.hpp"
+
+namespace srj {
+
+constexpr int JSON_MAX_DEPTH = 64;   // reference json_parser max nesting
+constexpr int MAX_PATH_DEPTH = 16;   // reference JSONUtils.MAX_PATH_DEPTH
+
+struct JsonSpan {
+  const char* p;
+  int32_t len;
+};
+
+__device__ inline const char* j_skip_ws(const char* p, const char* e) {
+  while (p < e && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+  return p;
+}
+
+// returns end of string token (after closing quote), or null on error
+__device__ inline const char* j_skip_string(const char* p, const char* e) {
+  // p points at opening quote
+  ++p;
+  while (p < e) {
+    if (*p == '\\') {
+      p += 2;
+    } else if (*p == '"') {
+      return p + 1;
+    } else {
+      ++p;
+    }
+  }
+  return nullptr;
+}
+
+// skip one JSON value starting at p; returns its end or null
+__device__ const char* j_skip_value(const char* p, const char* e) {
+  int depth = 0;
+  p = j_skip_ws(p, e);
+  if (p >= e) return nullptr;
+  do {
+    if (p >= e) return nullptr;
+    char c = *p;
+    if (c == '"') {
+      p = j_skip_string(p, e);
+      if (!p) return nullptr;
+    } else if (c == '{' || c == '[') {
+      ++depth;
+      if (depth > JSON_MAX_DEPTH) return nullptr;
+      ++p;
+    } else if (c == '}' || c == ']') {
+      --depth;
+      if (depth < 0) return nullptr;
+      ++p;
+    } else if (c == ',' || c == ':') {
+      ++p;
+    } else {
+      // literal: number/true/false/null
+      while (p < e && *p != ',' && *p != '}' && *p != ']' && *p != ' ' &&
+             *p != '\t' && *p != '\n' && *p != '\r')
+        ++p;
+    }
+    p = j_skip_ws(p, e);
+  } while (depth > 0);
+  return p;
+}
+
+// path instruction: kind 0 = key, 1 = index, 2 = wildcard ([*] or .*)
+struct PathInstr {
+  int32_t kind;
+  int32_t key_off;   // into shared key chars
+  int32_t key_len;
+  int32_t index;
+};
+
+// output sink: counts in phase 0, writes in phase 1
+struct Sink {
+  char* buf;
+  int32_t pos;
+  bool write;
+
+  __device__ void put(char c) {
+    if (write) buf[pos] = c;
+    ++pos;
+  }
+  __device__ void span(const char* p, int32_t n) {
+    if (write)
+      for (int32_t i = 0; i < n; ++i) buf[pos + i] = p[i];
+    pos += n;
+  }
+  // unescape a string token body (between quotes)
+  __device__ void unescaped(const char* p, int32_t n) {
+    for (int32_t i = 0; i < n; ++i) {
+      char c = p[i];
+      if (c == '\\' && i + 1 < n) {
+        ++i;
+        char x = p[i];
+        switch (x) {
+          case 'n': put('\n'); break;
+          case 't': put('\t'); break;
+          case 'r': put('\r'); break;
+          case 'b': put('\b'); break;
+          case 'f': put('\f'); break;
+          case '/': put('/'); break;
+          case '"': put('"'); break;
+          case '\\': put('\\'); break;
+          case 'u': {
+            if (i + 4 < n) {
+              uint32_t cp = 0;
+              for (int k = 1; k <= 4; ++k) {
+                char h = p[i + k];
+                cp = cp * 16 + (h <= '9' ? h - '0' : ((h | 32) - 'a' + 10));
+              }
+              i += 4;
+              if (cp < 0x80) {
+                put((char)cp);
+              } else if (cp < 0x800) {
+                put((char)(0xC0 | (cp >> 6)));
+                put((char)(0x80 | (cp & 0x3F)));
+              } else {
+                put((char)(0xE0 | (cp >> 12)));
+                put((char)(0x80 | ((cp >> 6) & 0x3F)));
+                put((char)(0x80 | (cp & 0x3F)));
+              }
+            }
+            break;
+          }
+          default: put(x);
+        }
+      } else {
+        put(c);
+      }
+    }
+  }
+};
+
+struct MatchCtx {
+  const PathInstr* instrs;
+  const char* keychars;
+  int32_t ninstr;
+  int32_t nmatches;
+  JsonSpan matches[8];   // first few match spans (value extents incl quotes)
+  bool overflow;
+};
+
+// recursive matcher: collect value spans matching instrs[step:] within value
+// at p.. Returns false on malformed json.
+__device__ bool j_match(const char* p, const char* e, MatchCtx& ctx,
+                        int32_t step, int depth) {
+  if (depth > MAX_PATH_DEPTH + 2) return false;
+  p = j_skip_ws(p, e);
+  if (step == ctx.ninstr) {
+    const char* vend = j_skip_value(p, e);
+    if (!vend) return false;
+    if (ctx.nmatches < 8)
+      ctx.matches[ctx.nmatches] = {p, (int32_t)(vend - p)};
+    else
+      ctx.overflow = true;
+    ++ctx.nmatches;
+    return true;
+  }
+  if (p >= e) return true;  // no match
+  const PathInstr ins = ctx.instrs[step];
+  if (*p == '{' && ins.kind == 0) {
+    ++p;
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p < e && *p == '}') return true;
+      if (p >= e || *p != '"') return false;
+      const char* kend = j_skip_string(p, e);
+      if (!kend) return false;
+      const char* kbody = p + 1;
+      int32_t klen = (int32_t)(kend - p - 2);
+      p = j_skip_ws(kend, e);
+      if (p >= e || *p != ':') return false;
+      ++p;
+      p = j_skip_ws(p, e);
+      bool match = klen == ins.key_len;
+      if (match) {
+        for (int32_t i = 0; i < klen; ++i)
+          if (kbody[i] != ctx.keychars[ins.key_off + i]) { match = false; break; }
+      }
+      if (match) {
+        if (!j_match(p, e, ctx, step + 1, depth + 1)) return false;
+      }
+      const char* vend = j_skip_value(p, e);
+      if (!vend) return false;
+      p = j_skip_ws(vend, e);
+      if (p < e && *p == ',') { ++p; continue; }
+      if (p < e && *p == '}') return true;
+      return false;
+    }
+  }
+  if (*p == '[' && (ins.kind == 1 || ins.kind == 2)) {
+    ++p;
+    int32_t idx = 0;
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p < e && *p == ']') return true;
+      if (ins.kind == 2 || idx == ins.index) {
+        if (!j_match(p, e, ctx, step + 1, depth + 1)) return false;
+      }
+      const char* vend = j_skip_value(p, e);
+      if (!vend) return false;
+      p = j_skip_ws(vend, e);
+      ++idx;
+      if (p < e && *p == ',') { ++p; continue; }
+      if (p < e && *p == ']') return true;
+      return false;
+    }
+  }
+  // wildcard over object values (Hive "$.*")
+  if (*p == '{' && ins.kind == 2) {
+    ++p;
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p < e && *p == '}') return true;
+      if (p >= e || *p != '"') return false;
+      const char* kend = j_skip_string(p, e);
+      if (!kend) return false;
+      p = j_skip_ws(kend, e);
+      if (p >= e || *p != ':') return false;
+      ++p;
+      p = j_skip_ws(p, e);
+      if (!j_match(p, e, ctx, step + 1, depth + 1)) return false;
+      const char* vend = j_skip_value(p, e);
+      if (!vend) return false;
+      p = j_skip_ws(vend, e);
+      if (p < e && *p == ',') { ++p; continue; }
+      if (p < e && *p == '}') return true;
+      return false;
+    }
+  }
+  return true;  // structure mismatch: no match, not an error
+}
+
+__device__ void emit_match(Sink& sink, JsonSpan m, bool as_element) {
+  if (m.len >= 2 && m.p[0] == '"' && !as_element) {
+    sink.unescaped(m.p + 1, m.len - 2);
+  } else {
+    sink.span(m.p, m.len);
+  }
+}
+
+template <bool WRITE>
+__global__ void get_json_object_kernel(ColDesc in, int64_t nrows,
+                                       const PathInstr* __restrict__ instrs,
+                                       const char* __restrict__ keychars,
+                                       int32_t ninstr,
+                                       int32_t* __restrict__ lens,
+                                       const int32_t* __restrict__ offsets,
+                                       char* __restrict__ chars,
+                                       uint8_t* __restrict__ out_valid) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t npad = (nrows + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < npad;
+       row += stride) {
+    bool in_range = row < nrows;
+    bool valid = in_range && is_valid(in.valid, row);
+    int32_t out_len = 0;
+    Sink sink{WRITE ? chars + (in_range && offsets ? offsets[row] : 0) : nullptr,
+              0, WRITE};
+    if (valid) {
+      StrView s = get_string(in, row);
+      MatchCtx ctx{instrs, keychars, ninstr, 0, {}, false};
+      bool ok = j_match(s.ptr, s.ptr + s.len, ctx, 0, 0);
+      if (!ok || ctx.nmatches == 0 || ctx.overflow) {
+        valid = false;
+      } else if (ctx.nmatches == 1) {
+        // Spark/Hive: a matched JSON null yields SQL NULL
+        JsonSpan m = ctx.matches[0];
+        if (m.len == 4 && m.p[0] == 'n' && m.p[1] == 'u' && m.p[2] == 'l' &&
+            m.p[3] == 'l') {
+          valid = false;
+        } else {
+          emit_match(sink, m, false);
+          out_len = sink.pos;
+        }
+      } else {
+        sink.put('[');
+        for (int32_t i = 0; i < ctx.nmatches; ++i) {
+          if (i) sink.put(',');
+          emit_match(sink, ctx.matches[i], true);
+        }
+        sink.put(']');
+        out_len = sink.pos;
+      }
+    }
+    if (WRITE) {
+      ballot_write_validity(out_valid, row, valid);
+    } else if (in_range) {
+      lens[row] = valid ? out_len : 0;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// from_json: top-level object field extraction -> string columns (one kernel
+// run per schema field reusing the path machinery), and raw-map extraction.
+// ---------------------------------------------------------------------------
+template <bool WRITE>
+__global__ void json_to_map_kernel(ColDesc in, int64_t nrows,
+                                   int32_t* __restrict__ entry_counts,
+                                   const int32_t* __restrict__ entry_offsets,
+                                   int32_t* __restrict__ key_lens,
+                                   int32_t* __restrict__ val_lens,
+                                   const int32_t* __restrict__ key_offsets,
+                                   const int32_t* __restrict__ val_offsets,
+                                   char* __restrict__ key_chars,
+                                   char* __restrict__ val_chars,
+                                   uint8_t* __restrict__ out_valid) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t npad = (nrows + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < npad;
+       row += stride) {
+    bool in_range = row < nrows;
+    bool valid = in_range && is_valid(in.valid, row);
+    int32_t nent = 0;
+    if (valid) {
+      StrView s = get_string(in, row);
+      const char* p = j_skip_ws(s.ptr, s.ptr + s.len);
+      const char* e = s.ptr + s.len;
+      if (p >= e || *p != '{') {
+        valid = false;
+      } else {
+        ++p;
+        int32_t ebase = WRITE && in_range ? entry_offsets[row] : 0;
+        while (valid) {
+          p = j_skip_ws(p, e);
+          if (p < e && *p == '}') break;
+          if (p >= e || *p != '"') { valid = false; break; }
+          const char* kend = j_skip_string(p, e);
+          if (!kend) { valid = false; break; }
+          const char* kbody = p + 1;
+          int32_t klen = (int32_t)(kend - p - 2);
+          p = j_skip_ws(kend, e);
+          if (p >= e || *p != ':') { valid = false; break; }
+          ++p;
+          p = j_skip_ws(p, e);
+          const char* vstart = p;
+          const char* vend = j_skip_value(p, e);
+          if (!vend) { valid = false; break; }
+          // value emitted unquoted for strings, raw otherwise
+          if (WRITE) {
+            int32_t ei = ebase + nent;
+            Sink ks{key_chars + key_offsets[ei], 0, true};
+            ks.unescaped(kbody, klen);
+            Sink vs{val_chars + val_offsets[ei], 0, true};
+            JsonSpan m{vstart, (int32_t)(vend - vstart)};
+            emit_match(vs, m, false);
+          } else {
+            // measure
+            Sink ks{nullptr, 0, false};
+            ks.unescaped(kbody, klen);
+            Sink vs{nullptr, 0, false};
+            JsonSpan m{vstart, (int32_t)(vend - vstart)};
+            emit_match(vs, m, false);
+            if (key_lens) key_lens[in_range ? row : 0] = 0;  // placeholder
+          }
+          ++nent;
+          p = j_skip_ws(vend, e);
+          if (p < e && *p == ',') { ++p; continue; }
+          if (p < e && *p == '}') break;
+          valid = false;
+        }
+      }
+    }
+    if (!WRITE && in_range) entry_counts[row] = valid ? nent : 0;
+    if (WRITE) ballot_write_validity(out_valid, row, valid);
+  }
+}
+
+// measure per-entry key/value lengths (phase between count and write):
+// entry_offsets gives each row's first entry slot.
+__global__ void json_map_entry_lens_kernel(ColDesc in, int64_t nrows,
+                                           const int32_t* __restrict__ entry_offsets,
+                                           int32_t* __restrict__ key_lens,
+                                           int32_t* __restrict__ val_lens) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < nrows;
+       row += stride) {
+    if (!is_valid(in.valid, row)) continue;
+    StrView s = get_string(in, row);
+    const char* p = j_skip_ws(s.ptr, s.ptr + s.len);
+    const char* e = s.ptr + s.len;
+    if (p >= e || *p != '{') continue;
+    ++p;
+    int32_t ei = entry_offsets[row];
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p >= e || *p == '}') break;
+      if (*p != '"') break;
+      const char* kend = j_skip_string(p, e);
+      if (!kend) break;
+      Sink ks{nullptr, 0, false};
+      ks.unescaped(p + 1, (int32_t)(kend - p - 2));
+      p = j_skip_ws(kend, e);
+      if (p >= e || *p != ':') break;
+      ++p;
+      p = j_skip_ws(p, e);
+      const char* vend = j_skip_value(p, e);
+      if (!vend) break;
+      Sink vs{nullptr, 0, false};
+      emit_match(vs, {p, (int32_t)(vend - p)}, false);
+      key_lens[ei] = ks.pos;
+      val_lens[ei] = vs.pos;
+      ++ei;
+      p = j_skip_ws(vend, e);
+      if (p < e && *p == ',') { ++p; continue; }
+      break;
+    }
+  }
+}
+
+}  // namespace srj
+
+using namespace srj;
+
+extern "C" {
+
+void srj_get_json_object(const void* in, int64_t nrows, const void* instrs,
+                         const char* keychars, int32_t ninstr, int32_t phase,
+                         int32_t* lens, const int32_t* offsets, char* chars,
+                         uint8_t* out_valid, hipStream_t stream) {
+  ColDesc c = *reinterpret_cast<const ColDesc*>(in);
+  if (phase == 0)
+    get_json_object_kernel<false><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+        c, nrows, reinterpret_cast<const PathInstr*>(instrs), keychars, ninstr,
+        lens, nullptr, nullptr, nullptr);
+  else
+    get_json_object_kernel<true><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+        c, nrows, reinterpret_cast<const PathInstr*>(instrs), keychars, ninstr,
+        nullptr, offsets, chars, out_valid);
+}
+
+void srj_json_map_count(const void* in, int64_t nrows, int32_t* entry_counts,
+                        hipStream_t stream) {
+  ColDesc c = *reinterpret_cast<const ColDesc*>(in);
+  json_to_map_kernel<false><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+      c, nrows, entry_counts, nullptr, nullptr, nullptr, nullptr, nullptr,
+      nullptr, nullptr, nullptr);
+}
+
+void srj_json_map_entry_lens(const void* in, int64_t nrows,
+                             const int32_t* entry_offsets, int32_t* key_lens,
+                             int32_t* val_lens, hipStream_t stream) {
+  ColDesc c = *reinterpret_cast<const ColDesc*>(in);
+  json_map_entry_lens_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+      c, nrows, entry_offsets, key_lens, val_lens);
+}
+
+void srj_json_map_write(const void* in, int64_t nrows,
+                        const int32_t* entry_offsets, const int32_t* key_offsets,
+                        const int32_t* val_offsets, char* key_chars,
+                        char* val_chars, uint8_t* out_valid, hipStream_t stream) {
+  ColDesc c = *reinterpret_cast<const ColDesc*>(in);
+  json_to_map_kernel<true><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+      c, nrows, nullptr, entry_offsets, nullptr, nullptr, key_offsets,
+      val_offsets, key_chars, val_chars, out_valid);
+}
+
+}  // extern "C"

@@ -1,0 +1,118 @@
+"""JSON op tests (get_json_object / from_json) vs Python oracle."""
+import json
+import random
+
+import pytest
+
+from spark_rapids_jni_amd.columnar import Column, DType
+
+random.seed(19)
+
+
+def test_path_compile():
+    from spark_rapids_jni_amd.ops.json import JsonPathError, compile_path
+    assert compile_path("$.a[2].b[*]") == [(0, "a", 0), (1, None, 2),
+                                           (0, "b", 0), (2, None, 0)]
+    assert compile_path("$['x y'].z") == [(0, "x y", 0), (0, "z", 0)]
+    assert compile_path("$.*") == [(2, None, 0)]
+    with pytest.raises(JsonPathError):
+        compile_path("a.b")
+    with pytest.raises(JsonPathError):
+        compile_path("$.a[b]")
+
+
+DOCS = [
+    '{"a":"b"}',
+    '{"a":1,"b":{"c":[10,20,30]}}',
+    '{"a":[{"x":1},{"x":2},{"y":3}]}',
+    '{"a":"line\\nbreak \\u00e9"}',
+    '{"a":null}',
+    '[1,2,3]',
+    'not json at all {',
+    "",
+    None,
+    '{"a": {"deep": {"nest": "v"}}}',
+    '{"empty":{},"arr":[]}',
+]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("path,expect", [
+    ("$.a", ["b", "1", '[{"x":1},{"x":2},{"y":3}]', "line\nbreak é",
+             None, None, None, None, None, '{"deep": {"nest": "v"}}', None]),
+    ("$.b.c[1]", [None, "20", None, None, None, None, None, None, None, None,
+                  None]),
+    ("$.a[*].x", [None, None, "[1,2]", None, None, None, None, None, None,
+                  None, None]),
+    ("$.a[1].x", [None, None, "2", None, None, None, None, None, None, None,
+                  None]),
+    ("$.a.deep.nest", [None, None, None, None, None, None, None, None, None,
+                       "v", None]),
+    ("$.empty", [None, None, None, None, None, None, None, None, None, None,
+                 "{}"]),
+    ("$[1]", [None, None, None, None, None, "2", None, None, None, None,
+              None]),
+])
+def test_get_json_object(path, expect):
+    from spark_rapids_jni_amd.ops.json import get_json_object
+    col = Column.from_pylist(DOCS, DType.STRING, "cuda")
+    got = get_json_object(col, path).to_pylist()
+    assert got == expect, f"path {path}"
+
+
+@pytest.mark.gpu
+def test_from_json_raw_map():
+    from spark_rapids_jni_amd.ops.json import from_json_to_raw_map
+    docs = ['{"k1":"v1","k2":2}', "{}", '{"x":{"nested":1},"y":[1,2]}',
+            "bad", None, '{"esc\\t":"tab"}']
+    col = Column.from_pylist(docs, DType.STRING, "cuda")
+    out = from_json_to_raw_map(col)
+    lst = out.to_pylist()
+    assert lst[0] == [("k1", "v1"), ("k2", "2")]
+    assert lst[1] == []
+    assert lst[2] == [("x", '{"nested":1}'), ("y", "[1,2]")]
+    assert lst[3] is None
+    assert lst[4] is None
+    assert lst[5] == [("esc\t", "tab")]
+
+
+@pytest.mark.gpu
+def test_from_json_to_structs():
+    from spark_rapids_jni_amd.ops.json import from_json_to_structs
+    docs = ['{"i": 42, "s":"hi", "f": 1.5, "b": true}',
+            '{"i": "7", "s": 9, "f": "2.25"}',
+            '{"s":"only"}', None]
+    col = Column.from_pylist(docs, DType.STRING, "cuda")
+    t = from_json_to_structs(col, ["i", "s", "f", "b"],
+                             [DType.INT64, DType.STRING, DType.FLOAT64,
+                              DType.BOOL8])
+    assert t.columns[0].to_pylist() == [42, 7, None, None]
+    assert t.columns[1].to_pylist() == ["hi", "9", "only", None]
+    assert t.columns[2].to_pylist() == [1.5, 2.25, None, None]
+    assert t.columns[3].to_pylist() == [True, None, None, None]
+
+
+@pytest.mark.gpu
+def test_get_json_object_fuzz_vs_oracle():
+    """random flat docs: $.key extraction must match a json-module oracle."""
+    from spark_rapids_jni_amd.ops.json import get_json_object
+    docs = []
+    for _ in range(300):
+        d = {f"k{j}": random.choice([random.randint(0, 99), "s" * (j % 3),
+                                     None, True, [1, 2], {"n": j}])
+             for j in range(random.randint(0, 5))}
+        docs.append(json.dumps(d, separators=(",", ":")))
+    col = Column.from_pylist(docs, DType.STRING, "cuda")
+    for key in ["k0", "k2", "k4"]:
+        got = get_json_object(col, f"$.{key}").to_pylist()
+        for doc, gv in zip(docs, got):
+            obj = json.loads(doc)
+            if key not in obj or obj[key] is None:
+                assert gv is None, (doc, key, gv)
+            else:
+                v = obj[key]
+                if isinstance(v, str):
+                    exp = v
+                else:
+                    exp = json.dumps(v, separators=(",", ":"))
+                assert gv == exp, (doc, key, gv)
