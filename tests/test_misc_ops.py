@@ -255,3 +255,47 @@ def test_truncate_timestamp():
                   - epoch) // datetime.timedelta(microseconds=1))
     assert wk == exp_wk
     assert truncate_timestamp(col, "DAY").to_pylist()[1] is None
+
+
+@pytest.mark.gpu
+def test_radix_sort_i64():
+    from spark_rapids_jni_amd.ops.sort import sort_pairs_i64
+    for n in [0, 1, 63, 2048, 5000, 100000]:
+        t = torch.randint(-2**62, 2**62, (n,), dtype=torch.int64, device="cuda")
+        if n > 10:
+            t[:5] = torch.tensor([0, -1, 1, -2**62, 2**62 - 1], device="cuda")
+        sk, order = sort_pairs_i64(t)
+        ref, ref_idx = torch.sort(t, stable=True)
+        assert torch.equal(sk, ref), f"n={n}"
+        assert torch.equal(t[order], sk), f"n={n} payload"
+
+
+@pytest.mark.gpu
+def test_radix_sort_stability():
+    from spark_rapids_jni_amd.ops.sort import sort_pairs_i64
+    t = torch.tensor([5, 3, 5, 3, 5, 3] * 700, dtype=torch.int64,
+                     device="cuda")
+    sk, order = sort_pairs_i64(t)
+    oh = order.cpu().tolist()
+    threes = [i for i in oh[:2100]]
+    assert threes == sorted(threes)  # stable: original order preserved
+    fives = [i for i in oh[2100:]]
+    assert fives == sorted(fives)
+
+
+@pytest.mark.gpu
+def test_sort_merge_join_matches_hash_join():
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    from spark_rapids_jni_amd.ops.sort import sort_merge_inner_join
+    import random as rnd
+    rnd.seed(3)
+    bvals = [rnd.randint(0, 500) if rnd.random() > 0.02 else None
+             for _ in range(1500)]
+    pvals = [rnd.randint(0, 600) if rnd.random() > 0.02 else None
+             for _ in range(4000)]
+    b = Column.from_pylist(bvals, DType.INT64, "cuda")
+    p = Column.from_pylist(pvals, DType.INT64, "cuda")
+    sb, sp = sort_merge_inner_join(b, p)
+    hb, hp = HashJoinTable.build(b).inner_join(p)
+    assert set(zip(sb.cpu().tolist(), sp.cpu().tolist())) == \
+        set(zip(hb.cpu().tolist(), hp.cpu().tolist()))
