@@ -3,4 +3,3 @@
 #include "srj_bind.hpp"
 
 void register_datetime(py::module_&) {}
-void register_sort(py::module_&) {}
