@@ -20,7 +20,7 @@ PYBIND11_MODULE(_gpu, m) {
   register_rowconv(m);
   register_shuffle(m);
   register_cast(m);
-  register_datetime(m);
+  register_misc2(m);
   register_json(m);
   register_misc(m);
   register_parquet(m);

@@ -1,5 +1,2 @@
-// Temporary empty registrations for subsystems not yet implemented; each is
-// replaced by its own bind_*.cpp as the corresponding kernels land.
+// (no pending stubs)
 #include "srj_bind.hpp"
-
-void register_datetime(py::module_&) {}

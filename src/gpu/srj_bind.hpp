@@ -34,7 +34,7 @@ void register_rowconv(py::module_& m);
 void register_shuffle(py::module_& m);
 void register_cast(py::module_& m);
 void register_copying(py::module_& m);
-void register_datetime(py::module_& m);
+void register_misc2(py::module_& m);
 void register_json(py::module_& m);
 void register_misc(py::module_& m);
 void register_parquet(py::module_& m);

@@ -1,0 +1,146 @@
+"""SHA/HLL/percentile/conv/parse_url/GBK tests vs Python oracles."""
+import hashlib
+import random
+import zlib
+
+import pytest
+import torch
+
+from spark_rapids_jni_amd.columnar import Column, DType
+
+random.seed(29)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("bits,fn", [(224, hashlib.sha224),
+                                     (256, hashlib.sha256),
+                                     (384, hashlib.sha384),
+                                     (512, hashlib.sha512)])
+def test_sha2(bits, fn):
+    from spark_rapids_jni_amd.ops.sketch import sha2
+    vals = ["", "abc", "hello world", None,
+            "x" * 55, "y" * 56, "z" * 63, "w" * 64, "v" * 65,
+            "long " * 100,
+            "".join(chr(random.randint(32, 126)) for _ in range(200))]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    got = sha2(col, bits).to_pylist()
+    for v, gs in zip(vals, got):
+        if v is None:
+            assert gs is None
+        else:
+            assert gs == fn(v.encode()).hexdigest(), f"{bits}: {v[:20]!r}"
+
+
+def test_crc32_host():
+    from spark_rapids_jni_amd.ops.sketch import crc32_host
+    assert crc32_host(b"hello") == zlib.crc32(b"hello")
+
+
+@pytest.mark.gpu
+def test_hllpp_estimate_and_pack():
+    from spark_rapids_jni_amd.ops.sketch import HyperLogLogPlusPlus
+    n = 200000
+    distinct = 50000
+    vals = [random.randrange(distinct) for _ in range(n)]
+    col = Column.from_pylist(vals, DType.INT64, "cuda")
+    h = HyperLogLogPlusPlus(precision=9)
+    h.update(col)
+    est = h.estimate()
+    assert abs(est - len(set(vals))) / len(set(vals)) < 0.15
+    # pack/unpack roundtrip is exact
+    longs = h.to_longs()
+    h2 = HyperLogLogPlusPlus.from_longs(longs, 9)
+    assert torch.equal(h.registers, h2.registers)
+    # merge
+    vals2 = [distinct + random.randrange(distinct) for _ in range(n)]
+    h3 = HyperLogLogPlusPlus(precision=9)
+    h3.update(Column.from_pylist(vals2, DType.INT64, "cuda"))
+    h.merge(h3)
+    est2 = h.estimate()
+    total_distinct = len(set(vals) | set(vals2))
+    assert abs(est2 - total_distinct) / total_distinct < 0.15
+
+
+@pytest.mark.gpu
+def test_percentile_from_histogram():
+    from spark_rapids_jni_amd.ops.sketch import percentile_from_histogram
+    # one histogram: values 1..4 with freqs 1,1,1,1 -> median = 2.5
+    offsets = torch.tensor([0, 4, 6], dtype=torch.int32, device="cuda")
+    values = Column.from_pylist([1.0, 2.0, 3.0, 4.0, 10.0, 20.0],
+                                DType.FLOAT64, "cuda")
+    freqs = Column.from_pylist([1, 1, 1, 1, 3, 1], DType.INT64, "cuda")
+    out = percentile_from_histogram(offsets, values, freqs, [0.5, 0.0, 1.0])
+    got = out.to_pylist()
+    assert got[0] == pytest.approx(2.5)
+    assert got[1] == pytest.approx(1.0)
+    assert got[2] == pytest.approx(4.0)
+    # second histogram: [10 x3, 20 x1]: p50 over ranks 0..3 -> 10
+    assert got[3] == pytest.approx(10.0)
+    assert got[5] == pytest.approx(20.0)
+
+
+@pytest.mark.gpu
+def test_conv():
+    from spark_rapids_jni_amd.ops.sketch import convert_base
+    vals = ["100", "ff", "-10", "", "zz", None, "12ab", " 1f "]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    got = convert_base(col, 16, 10).to_pylist()
+    assert got[0] == "256"
+    assert got[1] == "255"
+    # -10 (hex) = -16 -> unsigned two's complement
+    assert got[2] == str(2**64 - 16)
+    assert got[3] is None
+    assert got[4] is None  # 'z' invalid in base 16, no digits at all
+    assert got[5] is None
+    assert got[6] == str(0x12AB)
+    assert got[7] == "31"
+    got2 = convert_base(col, 16, -10).to_pylist()
+    assert got2[2] == "-16"
+    got3 = convert_base(Column.from_pylist(["255"], DType.STRING, "cuda"),
+                        10, 16).to_pylist()
+    assert got3[0] == "FF"
+
+
+@pytest.mark.gpu
+def test_parse_uri():
+    from spark_rapids_jni_amd.ops.sketch import UriPart, parse_uri
+    vals = ["https://www.example.com:8080/path/to/x?a=1&bb=2#frag",
+            "http://user@host.org/p",
+            "ftp://h/",
+            "not a uri",
+            "mailto:someone@example.com",
+            None,
+            "https://bad host/x"]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    assert parse_uri(col, UriPart.PROTOCOL).to_pylist() == \
+        ["https", "http", "ftp", None, "mailto", None, None]
+    assert parse_uri(col, UriPart.HOST).to_pylist() == \
+        ["www.example.com", "host.org", "h", None, None, None, None]
+    assert parse_uri(col, UriPart.PATH).to_pylist() == \
+        ["/path/to/x", "/p", "/", None, "someone@example.com", None, None]
+    assert parse_uri(col, UriPart.QUERY).to_pylist() == \
+        ["a=1&bb=2", None, None, None, None, None, None]
+    assert parse_uri(col, UriPart.QUERY_KEY, "bb").to_pylist() == \
+        ["2", None, None, None, None, None, None]
+    assert parse_uri(col, UriPart.QUERY_KEY, "a").to_pylist() == \
+        ["1", None, None, None, None, None, None]
+
+
+@pytest.mark.gpu
+def test_gbk_decode():
+    from spark_rapids_jni_amd.ops.sketch import CharsetDecodeError, gbk_decode
+    samples = ["hello", "中文", "混合mixed文本", "", None]
+    raw = [s.encode("gbk") if s is not None else None for s in samples]
+    # build a binary column: store gbk bytes as a "string" column
+    col = Column.from_pylist(raw, DType.STRING, "cuda")
+    got = gbk_decode(col).to_pylist()
+    for s, gs in zip(samples, got):
+        assert gs == s
+    # invalid sequence: REPLACE mode inserts U+FFFD, REPORT raises
+    bad = Column.from_pylist([b"ok", b"\x81\x20bad"], DType.STRING, "cuda")
+    rep = gbk_decode(bad).to_pylist()
+    assert rep[0] == "ok"
+    assert "�" in rep[1]
+    with pytest.raises(CharsetDecodeError) as ei:
+        gbk_decode(bad, report=True)
+    assert ei.value.row_with_error == 1
