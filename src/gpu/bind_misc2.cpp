@@ -18,6 +18,9 @@ void srj_parse_uri(const void*, int64_t, int32_t, const char*, int32_t, int32_t,
                    int32_t*, const int32_t*, char*, uint8_t*, hipStream_t);
 void srj_gbk_decode(const void*, int64_t, int32_t, int32_t, int32_t*,
                     const int32_t*, char*, uint8_t*, int64_t*, hipStream_t);
+void srj_tz_convert(const int64_t*, const uint8_t*, int64_t, const int64_t*,
+                    const int64_t*, const int64_t*, const int32_t*, int32_t,
+                    int32_t, int64_t*, hipStream_t);
 }
 
 void register_misc2(py::module_& m) {
@@ -77,6 +80,16 @@ void register_misc2(py::module_& m) {
                   as_ptr<int32_t>(lens), as_ptr<int32_t>(offsets),
                   as_ptr<char>(chars), as_ptr<uint8_t>(valid), as_stream(stream));
     check_hip("parse_uri");
+  });
+  m.def("tz_convert", [](uintptr_t in, uintptr_t valid, int64_t n,
+                         uintptr_t utc_us, uintptr_t local_us, uintptr_t off,
+                         uintptr_t zoffs, int32_t zidx, int32_t to_utc,
+                         uintptr_t out, uintptr_t stream) {
+    srj_tz_convert(as_ptr<int64_t>(in), as_ptr<uint8_t>(valid), n,
+                   as_ptr<int64_t>(utc_us), as_ptr<int64_t>(local_us),
+                   as_ptr<int64_t>(off), as_ptr<int32_t>(zoffs), zidx, to_utc,
+                   as_ptr<int64_t>(out), as_stream(stream));
+    check_hip("tz_convert");
   });
   m.def("gbk_decode", [](uintptr_t in, int64_t n, int32_t report, int32_t phase,
                          uintptr_t lens, uintptr_t offsets, uintptr_t chars,

@@ -446,6 +446,37 @@ __global__ void gbk_decode_kernel(ColDesc in, int64_t nrows, int32_t report,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// timezone conversion: binary search pre-expanded transitions
+// (reference timezones.cu convert_timestamp_to_utc / _to_timezone)
+// ---------------------------------------------------------------------------
+__global__ void tz_convert_kernel(const int64_t* __restrict__ in,
+                                  const uint8_t* __restrict__ valid, int64_t n,
+                                  const int64_t* __restrict__ utc_us,
+                                  const int64_t* __restrict__ local_us,
+                                  const int64_t* __restrict__ off_sec,
+                                  const int32_t* __restrict__ zone_offsets,
+                                  int32_t zone_idx, int32_t to_utc,
+                                  int64_t* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int32_t s = zone_offsets[zone_idx], e = zone_offsets[zone_idx + 1];
+  const int64_t* arr = to_utc ? local_us : utc_us;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t t = in[i];
+    if (!is_valid(valid, i)) { out[i] = t; continue; }
+    int32_t lo = s, hi = e - 1;
+    while (lo < hi) {  // last index with arr[idx] <= t
+      int32_t mid = (lo + hi + 1) >> 1;
+      if (arr[mid] <= t) lo = mid;
+      else hi = mid - 1;
+    }
+    int64_t off = off_sec[lo] * 1000000LL;
+    out[i] = to_utc ? t - off : t + off;
+  }
+}
+
 }  // namespace srj
 
 using namespace srj;
@@ -520,6 +551,14 @@ void srj_gbk_decode(const void* in, int64_t nrows, int32_t report, int32_t phase
         c, nrows, report, nullptr, offsets, chars, out_valid, err_row);
 }
 
-void srj_sha2_dispatch() {}
+void srj_tz_convert(const int64_t* in, const uint8_t* valid, int64_t n,
+                    const int64_t* utc_us, const int64_t* local_us,
+                    const int64_t* off_sec, const int32_t* zone_offsets,
+                    int32_t zone_idx, int32_t to_utc, int64_t* out,
+                    hipStream_t stream) {
+  tz_convert_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(
+      in, valid, n, utc_us, local_us, off_sec, zone_offsets, zone_idx, to_utc,
+      out);
+}
 
 }  // extern "C"

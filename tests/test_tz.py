@@ -1,0 +1,59 @@
+"""Timezone subsystem tests (transition extraction CPU + device conversion)."""
+import datetime
+import zoneinfo
+
+import pytest
+import torch
+
+from spark_rapids_jni_amd.columnar import Column, DType
+
+LA = "America/Los_Angeles"
+EPOCH = datetime.datetime(1970, 1, 1, tzinfo=datetime.timezone.utc)
+
+
+def _us(y, mo, d, h=0, mi=0, s=0):
+    t = datetime.datetime(y, mo, d, h, mi, s, tzinfo=datetime.timezone.utc)
+    return int((t - EPOCH).total_seconds()) * 1_000_000
+
+
+def test_extract_transitions_la():
+    from spark_rapids_jni_amd.tz import extract_transitions
+    trans = extract_transitions(LA, until_year=2030)
+    # 2021 DST: Mar 14 10:00 UTC (-> PDT), Nov 7 09:00 UTC (-> PST)
+    secs = {t for t, _ in trans}
+    assert _us(2021, 3, 14, 10) // 10**6 in secs
+    assert _us(2021, 11, 7, 9) // 10**6 in secs
+    offs = dict(trans)
+    assert offs[_us(2021, 3, 14, 10) // 10**6] == -7 * 3600
+    assert offs[_us(2021, 11, 7, 9) // 10**6] == -8 * 3600
+
+
+def test_extract_dst_rules():
+    from spark_rapids_jni_amd.tz import extract_dst_rules, validate_rules_stable
+    rules = extract_dst_rules(LA)
+    assert len(rules) == 2
+    months = sorted(r.month for r in rules)
+    assert months == [3, 11]
+    assert validate_rules_stable(LA)
+
+
+@pytest.mark.gpu
+def test_tz_convert_device():
+    from spark_rapids_jni_amd.tz import GpuTimeZoneDB
+    db = GpuTimeZoneDB.instance()
+    tz = zoneinfo.ZoneInfo(LA)
+    cases_utc = [
+        _us(2021, 1, 15, 12), _us(2021, 7, 15, 12),
+        _us(2021, 3, 14, 9, 59), _us(2021, 3, 14, 10, 1),
+        _us(1999, 12, 31, 23, 59), _us(2085, 6, 1),
+    ]
+    col = Column.from_pylist(cases_utc, DType.TIMESTAMP_US, "cuda")
+    local = db.convert_utc_timestamp_to_timezone(col, LA).to_pylist()
+    for u, lv in zip(cases_utc, local):
+        dt = EPOCH + datetime.timedelta(microseconds=u)
+        exp = u + int(dt.astimezone(tz).utcoffset().total_seconds()) * 10**6
+        assert lv == exp, datetime.datetime.utcfromtimestamp(u / 1e6)
+    # roundtrip local -> utc (unambiguous times)
+    lcol = Column.from_pylist(local, DType.TIMESTAMP_US, "cuda")
+    back = db.convert_timestamp_to_utc(lcol, LA).to_pylist()
+    assert back == cases_utc
