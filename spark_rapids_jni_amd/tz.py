@@ -45,7 +45,8 @@ def extract_transitions(name: str,
     end = int((datetime.datetime(until_year, 1, 1,
                                  tzinfo=datetime.timezone.utc)
                - _EPOCH).total_seconds())
-    out = [(-(2**62), _utc_offset_at(tz, start))]
+    # sentinel must survive *1e6 in int64: ~year -139000
+    out = [(-(2**42), _utc_offset_at(tz, start))]
     step = 6 * 3600  # DST shifts are at least hours apart; 6h probe misses none
     prev_off = out[0][1]
     t = start
