@@ -152,77 +152,7 @@ __device__ inline double pow10_pos(int e) {
   return r * tbl[e];
 }
 
-// --- Eisel-Lemire exactly-rounded decimal->double (Lemire, "Number Parsing
-// at a Gigabyte per Second"; same algorithm family as the reference's exact
-// string_to_float path). w = significand (<= 19 digits), q = decimal exp.
-#include "pow5_table.inc"
-
-__device__ inline bool eisel_lemire(uint64_t w, int64_t q, bool neg,
-                                    double* out) {
-  if (w == 0 || q < -342) {
-    *out = neg ? -0.0 : 0.0;
-    return true;
-  }
-  if (q > 308) {
-    *out = neg ? -INFINITY : INFINITY;
-    return true;
-  }
-  int lz = __clzll((long long)w);
-  w <<= lz;
-  const unsigned long long* p5 = SRJ_POW5[q + 342];
-  // 128-bit product approximation with 55-bit precision requirement
-  uint64_t lo = w * p5[0];
-  uint64_t hi = __umul64hi(w, p5[0]);
-  constexpr uint64_t precision_mask = 0x1FFull;  // 64 - (53 + 3) + 1... (2^9-1)
-  if ((hi & precision_mask) == precision_mask) {
-    uint64_t lo2 = w * p5[1];
-    uint64_t hi2 = __umul64hi(w, p5[1]);
-    uint64_t sum = lo + hi2;
-    if (sum < lo) ++hi;
-    lo = sum;
-    if ((hi & precision_mask) == precision_mask && lo + w < lo) {
-      return false;  // ambiguous: caller falls back
-    }
-    (void)lo2;
-  }
-  uint64_t upperbit = hi >> 63;
-  uint64_t mantissa = hi >> (upperbit + 9);  // 55 bits (53 + 2 round bits)
-  // power(q) + upperbit - lz - minimum_exponent(-1023)
-  int32_t power2 = (int32_t)(((152170LL + 65536LL) * q) >> 16) + 63 +
-                   (int32_t)upperbit - lz + 1023;
-  if (power2 <= 0) {  // subnormal or zero
-    if (-power2 + 1 >= 64) {
-      *out = neg ? -0.0 : 0.0;
-      return true;
-    }
-    mantissa >>= -power2 + 1;
-    mantissa += mantissa & 1;
-    mantissa >>= 1;
-    power2 = mantissa >= (1ull << 52) ? 1 : 0;
-    uint64_t bits = mantissa | ((uint64_t)power2 << 52) |
-                    ((uint64_t)neg << 63);
-    __builtin_memcpy(out, &bits, 8);
-    return true;
-  }
-  // round-ties-to-even edge: product exactly halfway
-  if (lo <= 1 && q >= -4 && q <= 23 && (mantissa & 3) == 1) {
-    if ((mantissa << (upperbit + 9)) == hi) mantissa &= ~1ull;
-  }
-  mantissa += mantissa & 1;
-  mantissa >>= 1;
-  if (mantissa >= (1ull << 53)) {
-    mantissa = 1ull << 52;
-    ++power2;
-  }
-  mantissa &= ~(1ull << 52);
-  if (power2 >= 0x7FF) {
-    *out = neg ? -INFINITY : INFINITY;
-    return true;
-  }
-  uint64_t bits = mantissa | ((uint64_t)power2 << 52) | ((uint64_t)neg << 63);
-  __builtin_memcpy(out, &bits, 8);
-  return true;
-}
+#include "exact_fp.inc"
 
 __device__ bool parse_double(StrView s, double* out) {
   s = trim_all(s);
@@ -339,6 +269,7 @@ __device__ bool parse_double(StrView s, double* out) {
   *out = neg ? -d : d;
   return true;
 }
+
 
 template <typename T>
 __global__ void string_to_float_kernel(ColDesc in, int64_t nrows,

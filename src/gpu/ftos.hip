@@ -11,6 +11,7 @@
 // scientific d.dddE±x (no '+' sign, e.g. "1.0E10", "-4.9E-324").
 #include "srj_common.hpp"
 #include "ryu_tables.inc"
+#include "exact_fp.inc"
 
 namespace srj {
 
@@ -387,6 +388,59 @@ __global__ void float_to_string_kernel(const T* __restrict__ in,
   }
 }
 
+// ---------------------------------------------------------------------------
+// round_float: Spark round/bround on doubles at decimal scale d, exact via
+// shortest decimal digits (Ryu) -> decimal-space rounding -> exact re-parse
+// (Eisel-Lemire). HALF_UP (round) or HALF_EVEN (bround).
+// ---------------------------------------------------------------------------
+__global__ void round_double_kernel(const double* __restrict__ in,
+                                    const uint8_t* __restrict__ valid, int64_t n,
+                                    int32_t scale, int32_t half_even,
+                                    double* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    double x = in[i];
+    if (!is_valid(valid, i)) { out[i] = 0; continue; }
+    uint64_t bits;
+    __builtin_memcpy(&bits, &x, 8);
+    uint32_t e = (uint32_t)((bits >> 52) & 0x7FF);
+    uint64_t m = bits & ((1ull << 52) - 1);
+    bool neg = bits >> 63;
+    if (e == 0x7FF || x == 0.0) { out[i] = x; continue; }
+    Dec64 d = ryu_d2d(m, e);
+    // round digits*10^exponent at decimal position -scale
+    int32_t drop = -scale - d.exponent;  // digits to drop from the right
+    if (drop <= 0) { out[i] = x; continue; }
+    uint64_t v = d.digits;
+    int nd = 0;
+    for (uint64_t t = v; t; t /= 10) ++nd;
+    if (drop >= nd + 1) { out[i] = neg ? -0.0 : 0.0; continue; }
+    uint64_t p10 = 1;
+    for (int k = 0; k < drop; ++k) p10 *= 10;
+    uint64_t q = v / p10;
+    uint64_t r = v - q * p10;
+    uint64_t half = p10 / 2;
+    bool up;
+    if (half_even) {
+      up = r > half || (r == half && (q & 1));
+    } else {
+      up = r >= half;
+    }
+    if (up) ++q;
+    double res;
+    if (!eisel_lemire(q, d.exponent + drop, neg, &res)) {
+      // fallback: scale by double arithmetic (ambiguous cases are rare)
+      res = (neg ? -1.0 : 1.0) * (double)q;
+      int32_t ee = d.exponent + drop;
+      while (ee > 0) { res *= 10.0; --ee; }
+      while (ee < 0) { res /= 10.0; ++ee; }
+    }
+    out[i] = res;
+  }
+}
+
+
 }  // namespace srj
 
 using namespace srj;
@@ -413,6 +467,13 @@ void srj_float_to_string(const void* in, const uint8_t* valid, int64_t nrows,
       float_to_string_kernel<float, true><<<g, DEFAULT_BLOCK, 0, stream>>>(
           (const float*)in, valid, nrows, nullptr, offsets, chars, out_valid);
   }
+}
+
+void srj_round_double(const double* in, const uint8_t* valid, int64_t n,
+                      int32_t scale, int32_t half_even, double* out,
+                      hipStream_t stream) {
+  round_double_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(
+      in, valid, n, scale, half_even, out);
 }
 
 }  // extern "C"

@@ -39,3 +39,4 @@ void register_json(py::module_& m);
 void register_misc(py::module_& m);
 void register_parquet(py::module_& m);
 void register_sort(py::module_& m);
+void register_lists(py::module_& m);
