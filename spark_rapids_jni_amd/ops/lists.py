@@ -137,8 +137,8 @@ def map_zip(m1: Column, m2: Column):
     else:
         data = torch.where(from_m2, keys_b.data, keys_a.data)
         keys = Column(keys_a.dtype, nent, data)
-    va = gather_column(v1, v1map[:nent])
-    vb = gather_column(v2, v2map[:nent])
+    va = gather_column(v1, v1map[:nent], has_nulls=True)
+    vb = gather_column(v2, v2map[:nent], has_nulls=True)
     entry = Column(DType.STRUCT, nent, None, None, None,
                    [keys, Column(DType.STRUCT, nent, None, None, None, [va, vb])])
     return Column(DType.LIST, n, None, m1.validity, out_offsets, [entry],

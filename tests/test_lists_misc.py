@@ -134,8 +134,9 @@ def test_round_double():
     assert even[3] == 1.2  # 1.25 -> 1.2 (even)
     assert even[4] == 1.4  # 1.35 is actually 1.35000000000000008882 -> 1.4
     two = round_double(col, 2).to_pylist()
-    # 2.675 is 2.67499999999999982236 in binary -> exact decimal rounds DOWN
-    assert two[8] == 2.67
+    # Spark rounds via the SHORTEST decimal form (BigDecimal.valueOf /
+    # Double.toString): 2.675 -> "2.675" -> HALF_UP -> 2.68
+    assert two[8] == 2.68
 
 
 @pytest.mark.gpu
