@@ -46,7 +46,8 @@ HIP_FLAGS = [f"--offload-arch={ARCH}", "-O3", "-std=c++17", "-fPIC",
              "-fno-gpu-rdc", "-Wall"]
 CXX_FLAGS = ["-O2", "-std=c++17", "-fPIC", "-Wall",
              "-D__HIP_PLATFORM_AMD__", f"-I{ROCM}/include"]
-LINK_FLAGS = ["-shared", f"-L{ROCM}/lib", "-lamdhip64", "-pthread"]
+LINK_FLAGS = ["-shared", f"-L{ROCM}/lib", "-lamdhip64",
+              "-lrocprofiler-sdk-roctx", "-pthread"]
 
 
 def _load_cache():
@@ -119,6 +120,19 @@ def build(verbose=False):
 
     if gpu_objs:
         link(gpu_objs, PKG / "_gpu.so")
+    # fault-injection preload shim (reference faultinj/ — LD_PRELOAD analog)
+    fi = ROOT / "src" / "tools" / "faultinj.cpp"
+    if fi.exists():
+        out = PKG / "libsrjfaultinj.so"
+        key = _key(fi, CXX_FLAGS, [])
+        ck = str(out)
+        if not out.exists() or cache.get(ck) != key:
+            r = subprocess.run([CXX] + CXX_FLAGS + ["-shared", str(fi), "-o",
+                                str(out), "-ldl"], capture_output=True,
+                               text=True)
+            if r.returncode != 0:
+                raise RuntimeError(f"faultinj build failed:\n{r.stderr}")
+            cache[ck] = key
     if host_objs:
         link(host_objs, PKG / "_host.so")
     CACHE.write_text(json.dumps(cache))

@@ -1,0 +1,164 @@
+"""Profiler: op-level activity capture + offline converter.
+
+Reference parity: Profiler.java:36-106 (init/start/stop with writeBufferSize
++ flushPeriodMillis), ProfilerJni.cpp CUPTI capture -> size-prefixed
+flatbuffers stream, spark_rapids_profile_converter (JSON / nsys-rep output).
+
+MI355X design: two mechanisms, like the reference:
+  * rocTX ranges (native, in _gpu.so): `srj_func_range()` pushes/pops a range
+    in rocprofv3 traces — the SRJ_FUNC_RANGE() analog.
+  * an in-process activity recorder: every traced section writes a
+    size-prefixed binary record (schema below, the profiler.fbs analog) to a
+    double-buffered writer with periodic flush; `convert(path)` renders JSON
+    (Chrome trace-event format loadable in Perfetto — the nsys-rep analog).
+
+Record: <u32 size> <u8 kind> <u64 start_ns> <u64 dur_ns> <u32 tid>
+        <u16 name_len> <name utf-8>    kind: 0 range, 1 marker, 2 kernel
+"""
+import contextlib
+import io
+import json
+import os
+import struct
+import threading
+import time
+from typing import Optional
+
+from .. import _native
+
+_MAGIC = b"SRJPROF1"
+
+
+class Profiler:
+    _active: Optional["Profiler"] = None
+
+    def __init__(self, path: str, write_buffer_size: int = 1 << 20,
+                 flush_period_millis: int = 0):
+        self.path = path
+        self.write_buffer_size = write_buffer_size
+        self.flush_period_millis = flush_period_millis
+        self._buf = io.BytesIO()
+        self._lock = threading.Lock()
+        self._f = None
+        self._running = False
+        self._last_flush = time.monotonic()
+
+    # -- lifecycle (Profiler.init/start/stop/shutdown) ----------------------
+    @classmethod
+    def init(cls, path: str, **kw) -> "Profiler":
+        p = Profiler(path, **kw)
+        p._f = open(path, "wb")
+        p._f.write(_MAGIC)
+        cls._active = p
+        return p
+
+    def start(self):
+        self._running = True
+
+    def stop(self):
+        self._running = False
+        self.flush()
+
+    def shutdown(self):
+        self.stop()
+        if self._f:
+            self._f.close()
+            self._f = None
+        if Profiler._active is self:
+            Profiler._active = None
+
+    # -- recording ----------------------------------------------------------
+    def record(self, kind: int, name: str, start_ns: int, dur_ns: int):
+        nb = name.encode()
+        rec = struct.pack("<BQQIH", kind, start_ns, dur_ns,
+                          threading.get_ident() & 0xFFFFFFFF, len(nb)) + nb
+        with self._lock:
+            self._buf.write(struct.pack("<I", len(rec)))
+            self._buf.write(rec)
+            now = time.monotonic()
+            if (self._buf.tell() >= self.write_buffer_size or
+                    (self.flush_period_millis and
+                     (now - self._last_flush) * 1000 >= self.flush_period_millis)):
+                self._flush_locked()
+
+    def _flush_locked(self):
+        if self._f and self._buf.tell():
+            self._f.write(self._buf.getvalue())
+            self._f.flush()
+            self._buf = io.BytesIO()
+            self._last_flush = time.monotonic()
+
+    def flush(self):
+        with self._lock:
+            self._flush_locked()
+
+
+@contextlib.contextmanager
+def srj_func_range(name: str, use_roctx: bool = True):
+    """SRJ_FUNC_RANGE() analog: rocTX push/pop (visible in rocprofv3) plus a
+    record in the active Profiler stream."""
+    g = None
+    if use_roctx:
+        try:
+            g = _native.gpu()
+            g.roctx_range_push(name)
+        except Exception:
+            g = None
+    t0 = time.monotonic_ns()
+    try:
+        yield
+    finally:
+        dur = time.monotonic_ns() - t0
+        if g is not None:
+            g.roctx_range_pop()
+        p = Profiler._active
+        if p is not None and p._running:
+            p.record(0, name, t0, dur)
+
+
+def mark(name: str):
+    try:
+        _native.gpu().roctx_mark(name)
+    except Exception:
+        pass
+    p = Profiler._active
+    if p is not None and p._running:
+        p.record(1, name, time.monotonic_ns(), 0)
+
+
+# -- offline converter (spark_rapids_profile_converter analog) --------------
+
+def read_records(path: str):
+    out = []
+    with open(path, "rb") as f:
+        assert f.read(8) == _MAGIC, "not an SRJ profile"
+        while True:
+            szb = f.read(4)
+            if len(szb) < 4:
+                break
+            (sz,) = struct.unpack("<I", szb)
+            rec = f.read(sz)
+            kind, start, dur, tid, nlen = struct.unpack_from("<BQQIH", rec)
+            name = rec[23:23 + nlen].decode()
+            out.append(dict(kind=kind, name=name, start_ns=start, dur_ns=dur,
+                            tid=tid))
+    return out
+
+
+def convert_to_json(path: str, out_path: Optional[str] = None) -> str:
+    """Chrome trace-event JSON (Perfetto-loadable)."""
+    events = []
+    for r in read_records(path):
+        if r["kind"] == 1:
+            events.append({"name": r["name"], "ph": "i", "pid": 0,
+                           "tid": r["tid"], "ts": r["start_ns"] / 1000.0,
+                           "s": "t"})
+        else:
+            events.append({"name": r["name"], "ph": "X", "pid": 0,
+                           "tid": r["tid"], "ts": r["start_ns"] / 1000.0,
+                           "dur": r["dur_ns"] / 1000.0})
+    doc = json.dumps({"traceEvents": events})
+    if out_path:
+        with open(out_path, "w") as f:
+            f.write(doc)
+    return doc

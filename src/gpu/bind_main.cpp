@@ -40,4 +40,5 @@ PYBIND11_MODULE(_gpu, m) {
   register_misc(m);
   register_parquet(m);
   register_lists(m);
+  register_tools(m);
 }

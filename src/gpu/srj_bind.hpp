@@ -40,3 +40,4 @@ void register_misc(py::module_& m);
 void register_parquet(py::module_& m);
 void register_sort(py::module_& m);
 void register_lists(py::module_& m);
+void register_tools(py::module_& m);
