@@ -141,6 +141,21 @@ template <typename T>
 T real(const char* name) {
   static_assert(sizeof(T) == sizeof(void*), "fn ptr");
   void* p = dlsym(RTLD_NEXT, name);
+  if (!p) {
+    // torch dlopens libamdhip64 after the preload (RTLD_LOCAL), so
+    // RTLD_NEXT may not see it — resolve through an explicit handle.
+    static void* h = [] {
+      void* x = dlopen("libamdhip64.so", RTLD_LAZY | RTLD_NOLOAD);
+      if (!x) x = dlopen("libamdhip64.so.7", RTLD_LAZY | RTLD_NOLOAD);
+      if (!x) x = dlopen("libamdhip64.so", RTLD_LAZY);
+      return x;
+    }();
+    if (h) p = dlsym(h, name);
+  }
+  if (!p) {
+    fprintf(stderr, "[srj-faultinj] FATAL: cannot resolve %s\n", name);
+    abort();
+  }
   return reinterpret_cast<T>(p);
 }
 
