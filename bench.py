@@ -113,9 +113,16 @@ def main():
     ap.add_argument("--probe-rows", type=int, default=PROBE_ROWS_DEFAULT,
                     help="probe rows per GPU per step (default = named config)")
     ap.add_argument("--chunk-rows", type=int, default=CHUNK_ROWS)
+    ap.add_argument("--op", choices=["join", "groupby", "q3"], default="join",
+                    help="flagship join (default), BASELINE config[1] "
+                         "group-by, or the NDS q3-shaped pipeline")
+    ap.add_argument("--groups", type=int, default=1_000_000,
+                    help="distinct groups for --op groupby")
     args = ap.parse_args()
 
     world, rank, local_rank = setup_dist(args)
+    if args.op != "join":
+        return run_secondary(args, world, rank, local_rank)
     if args.gpus > 1 and world != args.gpus:
         raise RuntimeError(
             f"--gpus {args.gpus} needs torchrun with --nproc-per-node {args.gpus} "
@@ -196,6 +203,73 @@ def main():
                 "seq_len": 0,
                 "parallelism": f"1 executor/GPU x{world}, RCCL all-to-all shuffle",
             },
+        }))
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+def run_secondary(args, world, rank, local_rank):
+    """Secondary benches: BASELINE config[1] hash-aggregate and the q3-shaped
+    pipeline. Single-GPU metrics (weak-scaled aggregate when launched under
+    torchrun)."""
+    device = torch.device("cuda", local_rank)
+    from spark_rapids_jni_amd.columnar import Column
+    if args.op == "groupby":
+        n = args.build_rows if args.build_rows != BUILD_ROWS_DEFAULT \
+            else 1_000_000_000
+        keys = torch.randint(0, args.groups, (n,), dtype=torch.int64,
+                             device=device)
+        vals = torch.randint(0, 1000, (n,), dtype=torch.int64, device=device)
+        kc, vc = Column.from_torch(keys), Column.from_torch(vals)
+        from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+
+        def step():
+            kt, res = groupby(kc, [(Agg.SUM, vc), (Agg.COUNT_ALL, None)],
+                              num_groups_hint=args.groups * 2)
+            del kt, res
+        metric, model = "hash_aggregate_rows_per_sec", \
+            "hash-aggregate sum/count group-by int64, 1B rows"
+        rows_per_step = n
+    else:  # q3
+        from spark_rapids_jni_amd import exec as ex
+        n = args.probe_rows if args.probe_rows != PROBE_ROWS_DEFAULT \
+            else 1_000_000_000
+        ss = ex.gen_store_sales(n, device=device)
+        dd = ex.gen_date_dim(device=device)
+        it = ex.gen_item(device=device)
+
+        def step():
+            keys, res = ex.q3_like(ss, dd, it, year=2000, manufact_id=50)
+            del keys, res
+        metric, model = "nds_q3_rows_per_sec", \
+            "q3-shape scan->filter->join x2->groupby over store_sales"
+        rows_per_step = n
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync(world)
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync(world)
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    if rank == 0:
+        print(json.dumps({
+            "metric": metric,
+            "value": world * rows_per_step * args.steps / elapsed,
+            "unit": "rows/s", "n_gpus": world, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+            "dtype": "int64", "data": "synthetic",
+            "config": {"model": model, "global_batch": world * rows_per_step,
+                       "seq_len": 0, "parallelism": f"x{world}"},
         }))
     if world > 1:
         import torch.distributed as dist

@@ -1,0 +1,94 @@
+"""Minimal columnar execution layer + NDS-shaped synthetic data.
+
+The reference is a kernel library driven by the Spark plugin; the analogous
+driver here is this module: enough operator glue (scan, filter, project,
+join, aggregate, exchange) to run NDS-class query shapes end-to-end on
+MI355X — used by bench.py --op q3 and the integration tests.
+"""
+from typing import List, Optional, Sequence
+
+import torch
+
+from . import parquet as pq
+from .columnar import Column, DType, Table
+from .ops import hashing
+from .ops.aggregate import Agg, groupby
+from .ops.copying import gather, gather_column
+from .ops.join import HashJoinTable
+
+
+def apply_boolean_mask(table: Table, mask: torch.Tensor) -> Table:
+    """Keep rows where mask != 0 (Spark Filter)."""
+    sel = torch.nonzero(mask, as_tuple=False).view(-1)
+    return gather(table, sel)
+
+
+def scan_parquet(path: str, columns=None, device="cuda") -> Table:
+    return pq.read_table(path, columns=columns, device=device)
+
+
+# --- synthetic TPC-DS-shaped generators (no network: random in the right
+#     shapes/cardinalities; SURVEY.md §7 stage 8) ---------------------------
+
+def gen_date_dim(device="cuda", start_year=1998, n_years=7) -> Table:
+    n = n_years * 365
+    d_date_sk = torch.arange(2450000, 2450000 + n, dtype=torch.int64,
+                             device=device)
+    d_year = (torch.arange(n, device=device) // 365 + start_year).to(torch.int64)
+    d_moy = ((torch.arange(n, device=device) % 365) // 31 + 1).clamp(1, 12) \
+        .to(torch.int64)
+    return Table([Column.from_torch(d_date_sk), Column.from_torch(d_year),
+                  Column.from_torch(d_moy)])
+
+
+def gen_store_sales(n: int, n_items: int = 18000, n_dates: int = 7 * 365,
+                    device="cuda", seed: int = 42) -> Table:
+    g = torch.Generator(device=device)
+    g.manual_seed(seed)
+    sold_date_sk = torch.randint(2450000, 2450000 + n_dates, (n,),
+                                 dtype=torch.int64, device=device, generator=g)
+    item_sk = torch.randint(1, n_items + 1, (n,), dtype=torch.int64,
+                            device=device, generator=g)
+    qty = torch.randint(1, 100, (n,), dtype=torch.int64, device=device,
+                        generator=g)
+    price = torch.rand(n, dtype=torch.float64, device=device, generator=g) * 100
+    return Table([Column.from_torch(sold_date_sk), Column.from_torch(item_sk),
+                  Column.from_torch(qty),
+                  Column(DType.FLOAT64, n, price)])
+
+
+def gen_item(n_items: int = 18000, device="cuda", seed: int = 7) -> Table:
+    g = torch.Generator(device=device)
+    g.manual_seed(seed)
+    i_item_sk = torch.arange(1, n_items + 1, dtype=torch.int64, device=device)
+    i_brand_id = torch.randint(1, 1000, (n_items,), dtype=torch.int64,
+                               device=device, generator=g)
+    i_manufact_id = torch.randint(1, 100, (n_items,), dtype=torch.int64,
+                                  device=device, generator=g)
+    return Table([Column.from_torch(i_item_sk), Column.from_torch(i_brand_id),
+                  Column.from_torch(i_manufact_id)])
+
+
+def q3_like(store_sales: Table, date_dim: Table, item: Table, year: int,
+            manufact_id: int):
+    """NDS q3 shape: date filter -> join store_sales⋈date_dim ⋈ item with an
+    item filter -> group by brand -> sum(price).
+
+    store_sales: [sold_date_sk, item_sk, qty, price]
+    date_dim:    [d_date_sk, d_year, d_moy]
+    item:        [i_item_sk, i_brand_id, i_manufact_id]
+    """
+    # filter dims
+    dd = apply_boolean_mask(date_dim, date_dim.columns[1].data == year)
+    it = apply_boolean_mask(item, item.columns[2].data == manufact_id)
+    # join: build on the (small) dims, probe the fact table
+    dd_tbl = HashJoinTable.build(dd.columns[0])
+    bi, pi = dd_tbl.inner_join(store_sales.columns[0])
+    ss1 = gather(store_sales, pi)
+    it_tbl = HashJoinTable.build(it.columns[0])
+    bi2, pi2 = it_tbl.inner_join(ss1.columns[1])
+    ss2 = gather(ss1, pi2)
+    brand = gather_column(it.columns[1], bi2.long())
+    keys, results = groupby(brand, [(Agg.SUM, ss2.columns[3]),
+                                    (Agg.COUNT_ALL, None)])
+    return keys, results

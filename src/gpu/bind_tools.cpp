@@ -4,6 +4,9 @@
 
 #include <rocprofiler-sdk-roctx/roctx.h>
 
+extern "C" void srj_pb_decode(const void*, int64_t, void*, int32_t, uint8_t*,
+                              int32_t, hipStream_t);
+
 void register_tools(py::module_& m) {
   m.def("roctx_range_push", [](const std::string& name) {
     return roctxRangePush(name.c_str());
@@ -14,4 +17,11 @@ void register_tools(py::module_& m) {
     return (int64_t)roctxRangeStartA(name.c_str());
   });
   m.def("roctx_range_stop", [](int64_t id) { roctxRangeStop((roctx_range_id_t)id); });
+  m.def("pb_decode", [](uintptr_t in, int64_t n, uintptr_t fields,
+                        int32_t nfields, uintptr_t row_ok, int32_t write_bytes,
+                        uintptr_t stream) {
+    srj_pb_decode(as_ptr<void>(in), n, as_ptr<void>(fields), nfields,
+                  as_ptr<uint8_t>(row_ok), write_bytes, as_stream(stream));
+    check_hip("pb_decode");
+  });
 }

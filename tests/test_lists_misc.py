@@ -214,3 +214,33 @@ def test_arms():
         pass
     assert c1.closed and c2.closed
     close_quietly(None, 5)  # no-op, no raise
+
+
+@pytest.mark.gpu
+def test_q3_pipeline_vs_torch_oracle():
+    from spark_rapids_jni_amd import exec as ex
+    n = 200000
+    ss = ex.gen_store_sales(n, n_items=500, device="cuda")
+    dd = ex.gen_date_dim(device="cuda")
+    it = ex.gen_item(500, device="cuda")
+    keys, res = ex.q3_like(ss, dd, it, year=2000, manufact_id=50)
+    got = dict(zip(keys.columns[0].to_pylist(), res[0].to_pylist()))
+    # torch oracle
+    sold, item_sk = ss.columns[0].data, ss.columns[1].data
+    qty, price = ss.columns[2].data, ss.columns[3].data
+    d_sk, d_year = dd.columns[0].data, dd.columns[1].data
+    i_sk, i_brand, i_man = (it.columns[0].data, it.columns[1].data,
+                            it.columns[2].data)
+    good_dates = set(d_sk[d_year == 2000].cpu().tolist())
+    man_items = {int(s): int(b) for s, b, m in
+                 zip(i_sk.cpu().tolist(), i_brand.cpu().tolist(),
+                     i_man.cpu().tolist()) if m == 50}
+    exp = {}
+    for s, isk, p in zip(sold.cpu().tolist(), item_sk.cpu().tolist(),
+                         price.cpu().tolist()):
+        if s in good_dates and isk in man_items:
+            b = man_items[isk]
+            exp[b] = exp.get(b, 0.0) + p
+    assert set(got) == set(exp)
+    for b in exp:
+        assert abs(got[b] - exp[b]) < 1e-6 * max(1.0, abs(exp[b]))

@@ -1,0 +1,79 @@
+"""Protobuf decode tests with a hand-written Python encoder as oracle."""
+import random
+
+import pytest
+
+from spark_rapids_jni_amd.columnar import Column, DType
+
+random.seed(37)
+
+
+def _varint(v):
+    out = b""
+    v &= (1 << 64) - 1
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out += bytes([b | 0x80])
+        else:
+            out += bytes([b])
+            return out
+
+
+def _tag(fnum, wire):
+    return _varint((fnum << 3) | wire)
+
+
+def encode(fields):
+    """fields: list of (fnum, kind, value)"""
+    import struct as st
+    out = b""
+    for fnum, kind, v in fields:
+        if kind in ("int64", "int32", "bool"):
+            out += _tag(fnum, 0) + _varint(int(v))
+        elif kind == "sint64":
+            zz = (v << 1) ^ (v >> 63)
+            out += _tag(fnum, 0) + _varint(zz)
+        elif kind == "double":
+            out += _tag(fnum, 1) + st.pack("<d", v)
+        elif kind == "float":
+            out += _tag(fnum, 5) + st.pack("<f", v)
+        elif kind in ("string", "bytes"):
+            b = v.encode() if isinstance(v, str) else v
+            out += _tag(fnum, 2) + _varint(len(b)) + b
+    return out
+
+
+@pytest.mark.gpu
+def test_protobuf_decode():
+    from spark_rapids_jni_amd.ops import protobuf
+    msgs = [
+        encode([(1, "int64", 42), (2, "string", "hello"), (3, "double", 1.5)]),
+        encode([(2, "string", ""), (1, "int64", -1)]),
+        encode([(3, "double", -2.25), (9, "int64", 99)]),  # 9: unknown field
+        b"",
+        None,
+        encode([(1, "int64", 7), (1, "int64", 8)]),  # last-one-wins
+        encode([(4, "sint64", -5), (5, "bool", 1), (6, "float", 0.5)]),
+    ]
+    col = Column.from_pylist(msgs, DType.STRING, "cuda")
+    t = protobuf.decode(col, [(1, "int64"), (2, "string"), (3, "double"),
+                              (4, "sint64"), (5, "bool"), (6, "float")])
+    assert t.columns[0].to_pylist() == [42, -1, None, None, None, 8, None]
+    assert t.columns[1].to_pylist() == ["hello", "", None, None, None, None,
+                                        None]
+    assert t.columns[2].to_pylist() == [1.5, None, -2.25, None, None, None,
+                                        None]
+    assert t.columns[3].to_pylist() == [None, None, None, None, None, None, -5]
+    assert t.columns[4].to_pylist() == [None] * 6 + [True]
+    assert t.columns[5].to_pylist() == [None] * 6 + [0.5]
+
+
+@pytest.mark.gpu
+def test_protobuf_malformed():
+    from spark_rapids_jni_amd.ops import protobuf
+    msgs = [encode([(1, "int64", 5)]), b"\xff\xff\xff", b"\x0a\xff"]
+    col = Column.from_pylist(msgs, DType.STRING, "cuda")
+    t = protobuf.decode(col, [(1, "int64")])
+    assert t.columns[0].to_pylist() == [5, None, None]
