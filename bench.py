@@ -130,7 +130,7 @@ def main():
     device = torch.device("cuda", local_rank)
 
     free, total = torch.cuda.mem_get_info(device)
-    need = (args.build_rows * 28              # build shard + table slots (2.2x pow2)
+    need = (args.build_rows * 84              # build shard + 25%-load slot table
             + args.probe_rows * 8             # resident probe chunks
             + args.chunk_rows * (32 if args.gpus > 1 else 0)  # partition temps
             + args.chunk_rows * 13            # join output maps

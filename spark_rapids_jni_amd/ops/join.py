@@ -64,8 +64,12 @@ class HashJoinTable:
         assert n < 2**31, "build side capped at 2^31-1 rows (chunk the build)"
         g = _native.gpu()
         dev = cols[0].device
+        # int64 fast path runs at 25% max load: random-probe scans touch
+        # ~1.3 slots instead of ~2.5 at 50% load (measured faster than
+        # double-slot prefetching); generic path stays at 50%.
         capacity = max(_next_pow2(n * 2), 64)
         if _is_i64_fast(cols) and not force_generic:
+            capacity = max(_next_pow2(n * 4), 64)
             c = cols[0]
             slots = torch.zeros(capacity * 2, dtype=torch.int64, device=dev)
             g.join_build_i64(c.data.data_ptr(),
