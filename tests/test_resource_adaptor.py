@@ -243,3 +243,11 @@ def test_monte_carlo_no_livelock(rmm):
     for th in threads:
         th.join(timeout=30)
     assert len(done) == nthreads, f"only {len(done)}/{nthreads} finished"
+
+
+def test_monte_carlo_cli_harness():
+    from spark_rapids_jni_amd.memory import RmmSpark
+    from spark_rapids_jni_amd.tools.rmm_monte_carlo import run
+    RmmSpark.clear_event_handler()
+    stats = run(6, 30, 2 * 2**20, 2**20 // 2, skewed=True)
+    assert stats["done"] == 6 and stats["failed"] == 0
