@@ -109,19 +109,9 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
             st.data_ptr())
     agg_desc = torch.frombuffer(raw, dtype=torch.uint8).to(dev)
 
-    # fast path: single non-null int64 key -> pipelined slot assignment
-    # (src/gpu/hashtable_i64.hip) + generic accumulate over assigned slots
-    if (len(kcols) == 1 and kcols[0].dtype == DType.INT64
-            and kcols[0].validity is None):
-        slot_idx = torch.empty(n, dtype=torch.int32, device=dev)
-        g.groupby_assign_i64(kcols[0].data.data_ptr(), n, slots.data_ptr(),
-                             capacity, slot_idx.data_ptr(), stream)
-        g.accumulate_by_slot(slot_idx.data_ptr(), n, agg_desc.data_ptr(),
-                             naggs, stream)
-    else:
-        kdesc, ktop, keep = pack_descriptors(kcols)
-        g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n,
-                  slots.data_ptr(), capacity, agg_desc.data_ptr(), naggs, stream)
+    kdesc, ktop, keep = pack_descriptors(kcols)
+    g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n, slots.data_ptr(),
+              capacity, agg_desc.data_ptr(), naggs, stream)
 
     counter = torch.zeros(1, dtype=torch.int64, device=dev)
     out_repr = torch.empty(capacity, dtype=torch.int64, device=dev)

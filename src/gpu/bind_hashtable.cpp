@@ -22,10 +22,6 @@ void srj_join_build_i64(const long long*, const uint8_t*, int64_t, void*, int64_
 void srj_join_probe_i64(const long long*, const uint8_t*, int64_t, const void*,
                         int64_t, uint64_t*, int32_t*, int64_t*, int64_t, uint8_t*,
                         int32_t, hipStream_t);
-void srj_groupby_assign_i64(const long long*, int64_t, uint64_t*, int64_t,
-                            int32_t*, hipStream_t);
-void srj_accumulate_by_slot(const int32_t*, int64_t, const void*, int32_t,
-                            hipStream_t);
 }
 
 void register_hashtable(py::module_& m) {
@@ -89,21 +85,6 @@ void register_hashtable(py::module_& m) {
                              as_ptr<int64_t>(out_probe), out_capacity,
                              as_ptr<uint8_t>(build_matched), fill, as_stream(stream));
           check_hip("join_probe_i64");
-        });
-  m.def("groupby_assign_i64",
-        [](uintptr_t keys, int64_t n, uintptr_t slots, int64_t capacity,
-           uintptr_t out_slot, uintptr_t stream) {
-          srj_groupby_assign_i64(as_ptr<long long>(keys), n,
-                                 as_ptr<uint64_t>(slots), capacity,
-                                 as_ptr<int32_t>(out_slot), as_stream(stream));
-          check_hip("groupby_assign_i64");
-        });
-  m.def("accumulate_by_slot",
-        [](uintptr_t slot_idx, int64_t n, uintptr_t aggs, int32_t naggs,
-           uintptr_t stream) {
-          srj_accumulate_by_slot(as_ptr<int32_t>(slot_idx), n, as_ptr<void>(aggs),
-                                 naggs, as_stream(stream));
-          check_hip("accumulate_by_slot");
         });
   m.def("groupby",
         [](uintptr_t cols, uintptr_t top, int32_t ntop, int64_t nrows,

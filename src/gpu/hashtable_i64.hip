@@ -17,7 +17,6 @@
 // Claim protocol: atomicCAS on the row word (0 = empty), winner writes key;
 // no sentinel key needed, build completes before probe launches.
 #include "srj_common.hpp"
-#include "agg_common.hpp"
 
 namespace srj {
 
@@ -171,136 +170,11 @@ __global__ void join_probe_i64_kernel(
   }
 }
 
-// ---------------------------------------------------------------------------
-// group-by fast path: pipelined insert-or-find assigning a slot index per
-// row (single non-null int64 key). Accumulation runs in a separate generic
-// kernel over the assigned slots, so every agg op composes with this path.
-// Slots are (fp32|row+1) packed words; key equality reads keys[repr] from the
-// read-only input (hot reprs stay in L1/L2).
-// ---------------------------------------------------------------------------
-__global__ void groupby_assign_i64_kernel(const long long* __restrict__ keys,
-                                          int64_t nrows,
-                                          uint64_t* __restrict__ slots,
-                                          uint64_t mask,
-                                          int32_t* __restrict__ out_slot) {
-  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
-  int64_t nbatch = (nrows + (int64_t)PIPE - 1) / PIPE;
-  for (int64_t batch = tid; batch < nbatch; batch += nthreads) {
-    int64_t base = batch * PIPE;
-    long long k[PIPE];
-    uint64_t s[PIPE];
-    uint64_t first[PIPE];
-#pragma unroll
-    for (int b = 0; b < PIPE; ++b) {
-      int64_t row = base + b;
-      k[b] = keys[row < nrows ? row : 0];
-      s[b] = mix64((uint64_t)k[b]) & mask;
-    }
-#pragma unroll
-    for (int b = 0; b < PIPE; ++b) first[b] = slots[s[b]];
-#pragma unroll
-    for (int b = 0; b < PIPE; ++b) {
-      int64_t row = base + b;
-      if (row >= nrows) continue;
-      uint64_t h = mix64((uint64_t)k[b]);
-      uint32_t fp = (uint32_t)(h >> 32);
-      if (fp == 0) fp = 1;
-      uint64_t packed = ((uint64_t)fp << 32) | (uint64_t)(row + 1);
-      uint64_t sl = s[b];
-      uint64_t word = first[b];
-      while (true) {
-        if (word == 0) {
-          word = atomicCAS(reinterpret_cast<unsigned long long*>(slots + sl),
-                           0ull, (unsigned long long)packed);
-          if (word == 0) break;  // claimed: this row is the representative
-        }
-        if ((uint32_t)(word >> 32) == fp) {
-          int64_t repr = (int64_t)(word & 0xffffffffu) - 1;
-          if (repr == row || keys[repr] == k[b]) break;
-        }
-        sl = (sl + 1) & mask;
-        word = slots[sl];
-      }
-      out_slot[row] = (int32_t)sl;
-    }
-  }
-}
-
-// accumulate into pre-assigned slots (any agg list)
-__global__ void accumulate_by_slot_kernel(const int32_t* __restrict__ slot_idx,
-                                          int64_t nrows,
-                                          const AggDesc* __restrict__ aggs,
-                                          int32_t naggs) {
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < nrows;
-       row += stride) {
-    int64_t s = slot_idx[row];
-    for (int32_t a = 0; a < naggs; ++a) {
-      const AggDesc& g = aggs[a];
-      switch (g.op) {
-        case AGG_COUNT_ALL:
-          atomicAdd((unsigned long long*)g.state + s, 1ull);
-          break;
-        case AGG_COUNT_VALID:
-          if (is_valid(g.valid, row))
-            atomicAdd((unsigned long long*)g.state + s, 1ull);
-          break;
-        case AGG_SUM_INT64:
-          if (is_valid(g.valid, row))
-            atomicAdd((unsigned long long*)g.state + s,
-                      (unsigned long long)fetch_int64(g.data, g.in_dtype, row));
-          break;
-        case AGG_SUM_FLOAT64:
-          if (is_valid(g.valid, row))
-            atomicAdd(reinterpret_cast<double*>(g.state) + s,
-                      fetch_double(g.data, g.in_dtype, row));
-          break;
-        case AGG_MIN_INT64:
-          if (is_valid(g.valid, row))
-            atomic_min_i64(reinterpret_cast<int64_t*>(g.state) + s,
-                           fetch_int64(g.data, g.in_dtype, row));
-          break;
-        case AGG_MAX_INT64:
-          if (is_valid(g.valid, row))
-            atomic_max_i64(reinterpret_cast<int64_t*>(g.state) + s,
-                           fetch_int64(g.data, g.in_dtype, row));
-          break;
-        case AGG_MIN_FLOAT64:
-          if (is_valid(g.valid, row))
-            atomic_min_f64(reinterpret_cast<double*>(g.state) + s,
-                           fetch_double(g.data, g.in_dtype, row));
-          break;
-        case AGG_MAX_FLOAT64:
-          if (is_valid(g.valid, row))
-            atomic_max_f64(reinterpret_cast<double*>(g.state) + s,
-                           fetch_double(g.data, g.in_dtype, row));
-          break;
-      }
-    }
-  }
-}
-
 }  // namespace srj
 
 using namespace srj;
 
 extern "C" {
-
-void srj_groupby_assign_i64(const long long* keys, int64_t nrows, uint64_t* slots,
-                            int64_t capacity, int32_t* out_slot,
-                            hipStream_t stream) {
-  int64_t nthreads_needed = (nrows + PIPE - 1) / PIPE;
-  groupby_assign_i64_kernel<<<grid_1d(nthreads_needed), DEFAULT_BLOCK, 0,
-                              stream>>>(keys, nrows, slots,
-                                        (uint64_t)(capacity - 1), out_slot);
-}
-
-void srj_accumulate_by_slot(const int32_t* slot_idx, int64_t nrows,
-                            const void* aggs, int32_t naggs, hipStream_t stream) {
-  accumulate_by_slot_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
-      slot_idx, nrows, reinterpret_cast<const AggDesc*>(aggs), naggs);
-}
 
 void srj_join_build_i64(const long long* keys, const uint8_t* valid, int64_t nrows,
                         void* slots, int64_t capacity, hipStream_t stream) {
