@@ -41,4 +41,5 @@ PYBIND11_MODULE(_gpu, m) {
   register_parquet(m);
   register_lists(m);
   register_tools(m);
+  register_dec128(m);
 }

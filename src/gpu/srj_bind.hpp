@@ -41,3 +41,4 @@ void register_parquet(py::module_& m);
 void register_sort(py::module_& m);
 void register_lists(py::module_& m);
 void register_tools(py::module_& m);
+void register_dec128(py::module_& m);
