@@ -199,3 +199,52 @@ def from_floats(col: Column) -> Column:
                       stream)
     return Column(DType.STRING, n, chars[:nchars], validity, offsets,
                   null_count=None)
+
+
+_FMT_KINDS = {"y": 1, "M": 2, "d": 3, "H": 4, "m": 5, "s": 6, "S": 7}
+
+
+def compile_timestamp_format(fmt: str):
+    """Spark to_timestamp pattern subset: y/M/d/H/m/s/S runs + literals
+    (reference parse_timestamp_with_format.cu:386)."""
+    toks = []
+    i = 0
+    while i < len(fmt):
+        c = fmt[i]
+        if c in _FMT_KINDS:
+            j = i
+            while j < len(fmt) and fmt[j] == c:
+                j += 1
+            toks.append((_FMT_KINDS[c], j - i))
+            i = j
+        elif c == "'":
+            j = fmt.index("'", i + 1)
+            for ch in fmt[i + 1:j]:
+                toks.append((0, ord(ch)))
+            i = j + 1
+        else:
+            toks.append((0, ord(c)))
+            i += 1
+    return toks
+
+
+def to_timestamp_with_format(col: Column, fmt: str, ansi: bool = False,
+                             default_tz_offset_sec: int = 0) -> Column:
+    """Spark to_timestamp(col, fmt) (reference parse_timestamp_with_format)."""
+    g = _native.gpu()
+    n = col.size
+    dev = col.device
+    toks = compile_timestamp_format(fmt)
+    raw = bytearray()
+    for kind, count in toks:
+        raw += struct.pack("<ii", kind, count)
+    tt = torch.frombuffer(raw or bytearray(1), dtype=torch.uint8).to(dev)
+    out = torch.empty(n, dtype=torch.int64, device=dev)
+    validity = make_validity(n, dev)
+    err, err_ptr = _err_buf(ansi, dev)
+    desc, keep = _desc_for(col)
+    g.parse_timestamp_fmt(desc.data_ptr(), n, tt.data_ptr(), len(toks),
+                          default_tz_offset_sec, out.data_ptr(),
+                          validity.data_ptr(), err_ptr, _native.current_stream())
+    _check_err(err, col, ansi)
+    return Column(DType.TIMESTAMP_US, n, out, validity, null_count=None)

@@ -599,6 +599,125 @@ __global__ void string_to_timestamp_kernel(ColDesc in, int64_t nrows,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// to_timestamp with format pattern (reference parse_timestamp_with_format.cu
+// :386 — Spark to_timestamp/unix_timestamp pattern subset).
+// Token kinds compiled host-side; CORRECTED-mode digit-count rules
+// (count==1: 1-2 digits greedy; count>=2: exactly count digits).
+// ---------------------------------------------------------------------------
+enum FmtKind : int32_t {
+  FMT_LITERAL = 0, FMT_YEAR = 1, FMT_MONTH = 2, FMT_DAY = 3,
+  FMT_HOUR = 4, FMT_MINUTE = 5, FMT_SECOND = 6, FMT_FRACTION = 7,
+};
+
+struct FmtToken {
+  int32_t kind;
+  int32_t count;   // repeat count (or the literal char for FMT_LITERAL)
+};
+
+__device__ inline bool parse_digits_n(StrView s, int* pos, int count,
+                                      bool greedy2, long* out) {
+  long v = 0;
+  int k = 0;
+  int maxd = greedy2 ? 2 : count;
+  while (*pos < s.len && k < maxd && s.ptr[*pos] >= '0' &&
+         s.ptr[*pos] <= '9') {
+    v = v * 10 + (s.ptr[*pos] - '0');
+    ++(*pos);
+    ++k;
+  }
+  if (greedy2 ? k < 1 : k < count) return false;
+  *out = v;
+  return true;
+}
+
+__global__ void parse_timestamp_fmt_kernel(ColDesc in, int64_t nrows,
+                                           const FmtToken* __restrict__ toks,
+                                           int32_t ntoks,
+                                           int64_t default_tz_offset_sec,
+                                           int64_t* __restrict__ out,
+                                           uint8_t* __restrict__ out_valid,
+                                           int64_t* __restrict__ err_row) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t npad = (nrows + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < npad;
+       row += stride) {
+    bool in_range = row < nrows;
+    bool valid = in_range && is_valid(in.valid, row);
+    int64_t v = 0;
+    if (valid) {
+      StrView s = trim_all(get_string(in, row));
+      int pos = 0;
+      long y = 1970, mo = 1, d = 1, hh = 0, mi = 0, ss = 0, us = 0;
+      for (int32_t t = 0; t < ntoks && valid; ++t) {
+        FmtToken tk = toks[t];
+        switch (tk.kind) {
+          case FMT_LITERAL:
+            if (pos >= s.len || s.ptr[pos] != (char)tk.count) valid = false;
+            else ++pos;
+            break;
+          case FMT_YEAR: {
+            // years allow up to max(4, count) digits when count == 4 ("yyyy")
+            long vv = 0;
+            int k = 0, maxd = tk.count <= 2 ? 4 : tk.count;
+            while (pos < s.len && k < maxd && s.ptr[pos] >= '0' &&
+                   s.ptr[pos] <= '9') {
+              vv = vv * 10 + (s.ptr[pos] - '0');
+              ++pos; ++k;
+            }
+            if (k < (tk.count <= 2 ? 1 : tk.count)) valid = false;
+            else y = vv;
+            break;
+          }
+          case FMT_MONTH:
+            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &mo);
+            break;
+          case FMT_DAY:
+            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &d);
+            break;
+          case FMT_HOUR:
+            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &hh);
+            break;
+          case FMT_MINUTE:
+            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &mi);
+            break;
+          case FMT_SECOND:
+            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &ss);
+            break;
+          case FMT_FRACTION: {
+            long f = 0;
+            int k = 0;
+            while (pos < s.len && k < tk.count && s.ptr[pos] >= '0' &&
+                   s.ptr[pos] <= '9') {
+              f = f * 10 + (s.ptr[pos] - '0');
+              ++pos; ++k;
+            }
+            if (k == 0) { valid = false; break; }
+            for (int z = k; z < 6; ++z) f *= 10;
+            for (int z = 6; z < k; ++z) f /= 10;
+            us = f;
+            break;
+          }
+        }
+      }
+      if (pos != s.len) valid = false;  // whole string must be consumed
+      if (valid && (!valid_ymd((int)y, (int)mo, (int)d) || hh > 23 ||
+                    mi > 59 || ss > 59))
+        valid = false;
+      if (valid) {
+        v = (days_from_civil((int)y, (int)mo, (int)d) * 86400LL +
+             hh * 3600 + mi * 60 + ss) * 1000000LL + us -
+            default_tz_offset_sec * 1000000LL;
+      } else if (err_row) {
+        atomicMin(reinterpret_cast<long long*>(err_row), (long long)row);
+      }
+    }
+    if (in_range) out[row] = v;
+    ballot_write_validity(out_valid, row, valid);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // integer/decimal -> string (two-phase: sizes then write)
 // ---------------------------------------------------------------------------
@@ -759,6 +878,16 @@ void srj_string_to_timestamp(const void* in, int64_t nrows, int64_t now_us,
   string_to_timestamp_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       *reinterpret_cast<const ColDesc*>(in), nrows, now_us, today_days,
       default_tz_offset_sec, out, out_valid, err_row);
+}
+
+void srj_parse_timestamp_fmt(const void* in, int64_t nrows, const void* toks,
+                             int32_t ntoks, int64_t default_tz_offset_sec,
+                             int64_t* out, uint8_t* out_valid, int64_t* err_row,
+                             hipStream_t stream) {
+  parse_timestamp_fmt_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+      *reinterpret_cast<const ColDesc*>(in), nrows,
+      reinterpret_cast<const FmtToken*>(toks), ntoks, default_tz_offset_sec,
+      out, out_valid, err_row);
 }
 
 void srj_integer_to_string(const void* in, int64_t nrows, int32_t phase,

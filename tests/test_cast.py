@@ -410,3 +410,35 @@ def test_float_to_string_roundtrip():
         assert struct.pack("<f", f) == struct.pack("<f", v), f"{v!r}: {gs!r}"
         # and is it in Java format shape
         assert "." in gs
+
+
+@pytest.mark.gpu
+def test_to_timestamp_with_format():
+    from spark_rapids_jni_amd.ops import cast
+    epoch = datetime.datetime(1970, 1, 1, tzinfo=datetime.timezone.utc)
+
+    def us(*a, micro=0):
+        t = datetime.datetime(*a, tzinfo=datetime.timezone.utc)
+        return int((t - epoch) // datetime.timedelta(microseconds=1)) + micro
+
+    vals = ["2021/07/15 13:45:59", "1999/01/02 00:00:00", "2021/7/15 1:2:3",
+            "bad", "2021/07/15", None, "2021/07/15 13:45:59 extra"]
+    col = Column.from_pylist(vals, DType.STRING, "cuda")
+    got = cast.to_timestamp_with_format(col, "yyyy/MM/dd HH:mm:ss").to_pylist()
+    assert got[0] == us(2021, 7, 15, 13, 45, 59)
+    assert got[1] == us(1999, 1, 2)
+    assert got[2] is None  # MM requires 2 digits
+    assert got[3] is None
+    assert got[4] is None  # incomplete
+    assert got[5] is None
+    assert got[6] is None  # trailing junk
+
+    col2 = Column.from_pylist(["15-7-2021 9.5", "1-12-1999 23.59"],
+                              DType.STRING, "cuda")
+    got2 = cast.to_timestamp_with_format(col2, "d-M-yyyy H.m").to_pylist()
+    assert got2[0] == us(2021, 7, 15, 9, 5)
+    assert got2[1] == us(1999, 12, 1, 23, 59)
+
+    col3 = Column.from_pylist(["2020-01-02 03:04:05.123"], DType.STRING, "cuda")
+    got3 = cast.to_timestamp_with_format(col3, "yyyy-MM-dd HH:mm:ss.SSS")
+    assert got3.to_pylist()[0] == us(2020, 1, 2, 3, 4, 5, micro=123000)
