@@ -94,6 +94,123 @@ __global__ void from_rows_kernel(const RowColDesc* __restrict__ cols,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// variable-width (strings) JCUDF path (reference copy_strings_to_rows
+// row_conversion.cu:839 / copy_strings_from_rows:1145): string columns hold
+// an (offset-in-row, length) int32 pair in the fixed section; string bytes
+// are appended after the validity bytes; rows are 8-byte aligned with a
+// row-offsets array (build_string_row_offsets analog runs as a size kernel +
+// host cumsum).
+// ---------------------------------------------------------------------------
+__global__ void var_row_sizes_kernel(const RowColDesc* __restrict__ cols,
+                                     int32_t ncols, int64_t nrows,
+                                     int32_t fixed_size,
+                                     int32_t* __restrict__ sizes) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < nrows;
+       row += stride) {
+    int32_t sz = fixed_size;
+    for (int32_t c = 0; c < ncols; ++c) {
+      const RowColDesc& d = cols[c];
+      if (d.width != 0) continue;  // fixed col
+      const int32_t* offs = reinterpret_cast<const int32_t*>(d.data);
+      if (is_valid(d.valid, row)) sz += offs[row + 1] - offs[row];
+    }
+    sizes[row] = (sz + 7) & ~7;
+  }
+}
+
+// var col RowColDesc: width == 0, data = offsets ptr, row_off = fixed-section
+// position of the (offset,len) pair; chars pointer passed separately.
+__global__ void to_rows_var_kernel(const RowColDesc* __restrict__ cols,
+                                   const uint64_t* __restrict__ char_ptrs,
+                                   int32_t ncols, int64_t nrows,
+                                   int32_t fixed_size, int32_t validity_off,
+                                   const int32_t* __restrict__ row_offsets,
+                                   uint8_t* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < nrows;
+       row += stride) {
+    uint8_t* r = out + row_offsets[row];
+    int32_t var_pos = fixed_size;
+    uint8_t vbyte = 0;
+    for (int32_t c = 0; c < ncols; ++c) {
+      const RowColDesc& d = cols[c];
+      bool valid = is_valid(d.valid, row);
+      if (d.width == 0) {
+        const int32_t* offs = reinterpret_cast<const int32_t*>(d.data);
+        int32_t len = valid ? offs[row + 1] - offs[row] : 0;
+        reinterpret_cast<int32_t*>(r + d.row_off)[0] = var_pos;
+        reinterpret_cast<int32_t*>(r + d.row_off)[1] = len;
+        if (valid) {
+          const char* src = reinterpret_cast<const char*>(char_ptrs[c]) +
+                            offs[row];
+          for (int32_t k = 0; k < len; ++k) r[var_pos + k] = src[k];
+        }
+        var_pos += len;
+      } else if (valid) {
+        copy_elem(r + d.row_off,
+                  reinterpret_cast<const uint8_t*>(d.data) + row * d.width,
+                  d.width);
+      } else {
+        for (int b = 0; b < d.width; ++b) r[d.row_off + b] = 0;
+      }
+      vbyte |= (uint8_t)valid << (c & 7);
+      if ((c & 7) == 7) {
+        r[validity_off + (c >> 3)] = vbyte;
+        vbyte = 0;
+      }
+    }
+    if (ncols & 7) r[validity_off + (ncols >> 3)] = vbyte;
+  }
+}
+
+// from rows: phase 0 emits per-row string lengths per var column; phase 1
+// copies fixed cols + chars using the per-column char offsets (cumsum'd).
+template <int PHASE>
+__global__ void from_rows_var_kernel(const RowColDesc* __restrict__ cols,
+                                     const uint64_t* __restrict__ char_ptrs,
+                                     const uint64_t* __restrict__ len_ptrs,
+                                     int32_t ncols, int64_t nrows,
+                                     int32_t validity_off,
+                                     const int32_t* __restrict__ row_offsets,
+                                     const uint8_t* __restrict__ in) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t npad = (nrows + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       row < npad; row += stride) {
+    bool in_range = row < nrows;
+    const uint8_t* r = in + (in_range ? row_offsets[row] : 0);
+    for (int32_t c = 0; c < ncols; ++c) {
+      const RowColDesc& d = cols[c];
+      bool valid = in_range && ((r[validity_off + (c >> 3)] >> (c & 7)) & 1);
+      if (d.width == 0) {
+        int32_t pos = in_range ? reinterpret_cast<const int32_t*>(r + d.row_off)[0] : 0;
+        int32_t len = (in_range && valid)
+                          ? reinterpret_cast<const int32_t*>(r + d.row_off)[1]
+                          : 0;
+        if (PHASE == 0) {
+          if (in_range)
+            reinterpret_cast<int32_t*>(len_ptrs[c])[row] = len;
+        } else if (in_range && valid) {
+          // out offsets (cumsum of lengths) live where len_ptrs points now
+          const int32_t* ooffs = reinterpret_cast<const int32_t*>(len_ptrs[c]);
+          char* dst = reinterpret_cast<char*>(char_ptrs[c]) + ooffs[row];
+          for (int32_t k = 0; k < len; ++k) dst[k] = (char)r[pos + k];
+        }
+      } else if (PHASE == 1 && in_range) {
+        copy_elem(const_cast<uint8_t*>(
+                      reinterpret_cast<const uint8_t*>(d.data)) + row * d.width,
+                  r + d.row_off, d.width);
+      }
+      if (PHASE == 1 && d.valid != nullptr) {
+        ballot_write_validity(const_cast<uint8_t*>(d.valid), row, valid);
+      }
+    }
+  }
+}
+
 }  // namespace srj
 
 using namespace srj;
@@ -113,6 +230,36 @@ void srj_from_rows(const void* cols, int32_t ncols, int64_t nrows,
   from_rows_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
       validity_off, in);
+}
+
+void srj_var_row_sizes(const void* cols, int32_t ncols, int64_t nrows,
+                       int32_t fixed_size, int32_t* sizes, hipStream_t stream) {
+  var_row_sizes_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+      reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, fixed_size,
+      sizes);
+}
+
+void srj_to_rows_var(const void* cols, const uint64_t* char_ptrs, int32_t ncols,
+                     int64_t nrows, int32_t fixed_size, int32_t validity_off,
+                     const int32_t* row_offsets, uint8_t* out,
+                     hipStream_t stream) {
+  to_rows_var_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+      reinterpret_cast<const RowColDesc*>(cols), char_ptrs, ncols, nrows,
+      fixed_size, validity_off, row_offsets, out);
+}
+
+void srj_from_rows_var(const void* cols, const uint64_t* char_ptrs,
+                       const uint64_t* len_ptrs, int32_t ncols, int64_t nrows,
+                       int32_t validity_off, const int32_t* row_offsets,
+                       const uint8_t* in, int32_t phase, hipStream_t stream) {
+  if (phase == 0)
+    from_rows_var_kernel<0><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+        reinterpret_cast<const RowColDesc*>(cols), char_ptrs, len_ptrs, ncols,
+        nrows, validity_off, row_offsets, in);
+  else
+    from_rows_var_kernel<1><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+        reinterpret_cast<const RowColDesc*>(cols), char_ptrs, len_ptrs, ncols,
+        nrows, validity_off, row_offsets, in);
 }
 
 }  // extern "C"

@@ -89,3 +89,57 @@ def test_row_roundtrip():
     back = convert_from_rows(batches, dtypes)
     for orig, got, vals in zip(cols, back.columns, colvals):
         assert got.to_pylist() == vals
+
+
+def test_var_row_layout():
+    from spark_rapids_jni_amd.ops.row_conversion import var_row_layout
+    # INT64, STRING, INT8: string slot is 8B aligned to 4
+    offs, voff, fixed = var_row_layout([DType.INT64, DType.STRING, DType.INT8])
+    assert offs == [0, 8, 16]
+    assert voff == 17
+    assert fixed == 24
+
+
+@pytest.mark.gpu
+def test_var_row_roundtrip():
+    from spark_rapids_jni_amd.ops.row_conversion import (
+        convert_from_rows_varwidth, convert_to_rows_varwidth)
+    n = 777
+    svals = [None if i % 13 == 4 else ("s%d" % i) * (i % 9) for i in range(n)]
+    ivals = [None if i % 7 == 2 else i * 3 for i in range(n)]
+    s2vals = ["x" * (i % 4) for i in range(n)]
+    dtypes = [DType.STRING, DType.INT64, DType.STRING]
+    cols = [Column.from_pylist(svals, DType.STRING, "cuda"),
+            Column.from_pylist(ivals, DType.INT64, "cuda"),
+            Column.from_pylist(s2vals, DType.STRING, "cuda")]
+    buf, row_offs = convert_to_rows_varwidth(Table(cols))
+    assert row_offs.numel() == n + 1
+    # every row 8-byte aligned
+    ro = row_offs.cpu().tolist()
+    assert all(o % 8 == 0 for o in ro)
+    back = convert_from_rows_varwidth(buf, row_offs, dtypes)
+    assert back.columns[0].to_pylist() == svals
+    assert back.columns[1].to_pylist() == ivals
+    assert back.columns[2].to_pylist() == s2vals
+
+
+@pytest.mark.gpu
+def test_var_row_layout_bytes():
+    """Check the on-wire fixed-section layout: (offset,len) pair + chars
+    after validity."""
+    from spark_rapids_jni_amd.ops.row_conversion import (
+        convert_to_rows_varwidth, var_row_layout)
+    dtypes = [DType.INT32, DType.STRING]
+    cols = [Column.from_pylist([7, 8], DType.INT32, "cuda"),
+            Column.from_pylist(["abc", ""], DType.STRING, "cuda")]
+    buf, row_offs = convert_to_rows_varwidth(Table(cols))
+    offs, voff, fixed = var_row_layout(dtypes)
+    raw = buf.cpu().numpy().tobytes()
+    ro = row_offs.cpu().tolist()
+    r0 = raw[ro[0]:ro[1]]
+    i32, spos, slen = struct.unpack_from("<i", r0, offs[0])[0], \
+        struct.unpack_from("<ii", r0, offs[1])[0], \
+        struct.unpack_from("<ii", r0, offs[1])[1]
+    assert i32 == 7 and spos == fixed and slen == 3
+    assert r0[spos:spos + 3] == b"abc"
+    assert r0[voff] == 0b11
