@@ -483,3 +483,85 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
         chunks = [rg.columns[fi] for rg in footer.row_groups]
         cols.append(_read_column(raw, f, chunks, total_rows, device))
     return Table(cols)
+
+
+# ---------------------------------------------------------------------------
+# Footer re-serialization (reference NativeParquetJni.cpp:692-719: prune +
+# rewrite the FileMetaData thrift so a pruned footer can be handed to any
+# parquet reader). Operates on the typed thrift tree to preserve wire types.
+# ---------------------------------------------------------------------------
+
+def _tv(struct_dict, fid, default=None):
+    tv = struct_dict.get(fid)
+    return default if tv is None else tv[1]
+
+
+def rewrite_footer(raw: bytes, keep_columns: Optional[Sequence[str]] = None,
+                   part_offset: Optional[int] = None,
+                   part_length: Optional[int] = None) -> bytes:
+    """Parse a footer (FileMetaData thrift bytes, no magic/length), prune
+    columns and/or filter row groups, and re-serialize to thrift bytes."""
+    host = _native.host()
+    fmd, _ = host.thrift_parse_typed(raw, 0)
+
+    if keep_columns is not None:
+        keep = {c.lower() for c in keep_columns}
+        etype, elems = fmd[2][1]
+        root = elems[0]
+        kept_elems = [root] + [
+            se for se in elems[1:] if _tv(se, 4).decode().lower() in keep]
+        root[5] = (5, len(kept_elems) - 1)   # num_children (i32)
+        fmd[2] = (fmd[2][0], (etype, kept_elems))
+        rgs_t, rgs = fmd[4][1]
+        for rg in rgs:
+            ct, cols = rg[1][1]
+            kept = [cc for cc in cols
+                    if _tv(_tv(cc, 3), 3)[1][0].decode().lower() in keep]
+            rg[1] = (rg[1][0], (ct, kept))
+        # column_orders (field 7) is one entry per leaf column
+        if 7 in fmd:
+            n_leaves = sum(1 for se in kept_elems[1:] if 5 not in se)
+            ot, orders = fmd[7][1]
+            fmd[7] = (fmd[7][0], (ot, orders[:n_leaves]))
+
+    if part_offset is not None:
+        rgs_t, rgs = fmd[4][1]
+        kept_rgs = []
+        for rg in rgs:
+            cols = rg[1][1][1]
+            if not cols:
+                continue
+            starts = []
+            for cc in cols:
+                md = _tv(cc, 3)
+                o = _tv(md, 9)
+                dp = _tv(md, 11)
+                if dp is not None and 0 < dp < o:
+                    o = dp
+                starts.append(o)
+            mid = min(starts) + _tv(rg, 2) // 2
+            if part_offset <= mid < part_offset + part_length:
+                kept_rgs.append(rg)
+        fmd[4] = (fmd[4][0], (rgs_t, kept_rgs))
+        fmd[3] = (fmd[3][0], sum(_tv(rg, 3) for rg in kept_rgs))
+
+    return host.thrift_write(fmd)
+
+
+def rewrite_parquet_file(src_path: str, dst_path: str,
+                         keep_columns: Optional[Sequence[str]] = None,
+                         part_offset: Optional[int] = None,
+                         part_length: Optional[int] = None) -> None:
+    """Copy a parquet file replacing its footer with a pruned rewrite (data
+    pages are byte-identical; only FileMetaData changes)."""
+    raw = open(src_path, "rb").read()
+    assert raw[:4] == MAGIC and raw[-4:] == MAGIC
+    flen = struct.unpack("<I", raw[-8:-4])[0]
+    body_end = len(raw) - 8 - flen
+    new_footer = rewrite_footer(raw[body_end:-8], keep_columns, part_offset,
+                                part_length)
+    with open(dst_path, "wb") as f:
+        f.write(raw[:body_end])
+        f.write(new_footer)
+        f.write(struct.pack("<I", len(new_footer)))
+        f.write(MAGIC)

@@ -72,7 +72,13 @@ struct Reader {
         uint64_t n = h >> 4;
         if (n == 15) n = varint();
         py::list out;
-        for (uint64_t i = 0; i < n; ++i) out.append(value(etype));
+        for (uint64_t i = 0; i < n; ++i) {
+          // list bool elements are full bytes (1=true), unlike struct fields
+          if (etype == 1 || etype == 2)
+            out.append(py::bool_(byte() == 1));
+          else
+            out.append(value(etype));
+        }
         return out;
       }
       case 11: {  // map
@@ -109,6 +115,184 @@ struct Reader {
   }
 };
 
+// --------------------------------------------------------------------------
+// Typed parse + writer (footer re-serialization, reference
+// NativeParquetJni.cpp:692-719 rewrites pruned footers with TCompactProtocol).
+// Typed tree: struct -> {id: (wire_type, value)}; list/set -> (etype, [v..]);
+// map -> (ktype, vtype, [(k, v)..]); bool normalized to wire_type 1.
+// --------------------------------------------------------------------------
+
+struct TypedReader : Reader {
+  py::object tvalue(int type) {
+    switch (type) {
+      case 1: return py::bool_(true);
+      case 2: return py::bool_(false);
+      case 3: return py::int_((int64_t)(int8_t)byte());
+      case 4:
+      case 5:
+      case 6: return py::int_(zigzag());
+      case 7: {
+        if (pos + 8 > len) throw std::runtime_error("thrift: bad double");
+        double d;
+        memcpy(&d, p + pos, 8);
+        pos += 8;
+        return py::float_(d);
+      }
+      case 8: return binary();
+      case 9:
+      case 10: {
+        uint8_t h = byte();
+        int etype = h & 0x0F;
+        uint64_t n = h >> 4;
+        if (n == 15) n = varint();
+        py::list out;
+        for (uint64_t i = 0; i < n; ++i) {
+          if (etype == 1 || etype == 2)
+            out.append(py::bool_(byte() == 1));
+          else
+            out.append(tvalue(etype));
+        }
+        return py::make_tuple(etype, out);
+      }
+      case 11: {
+        uint64_t n = varint();
+        int ktype = 0, vtype = 0;
+        py::list pairs;
+        if (n > 0) {
+          uint8_t kv = byte();
+          ktype = kv >> 4;
+          vtype = kv & 0x0F;
+          for (uint64_t i = 0; i < n; ++i) {
+            py::object k = tvalue(ktype);
+            pairs.append(py::make_tuple(k, tvalue(vtype)));
+          }
+        }
+        return py::make_tuple(ktype, vtype, pairs);
+      }
+      case 12: return tstrct();
+      default:
+        throw std::runtime_error("thrift: unknown type " + std::to_string(type));
+    }
+  }
+
+  py::dict tstrct() {
+    py::dict out;
+    int16_t last_id = 0;
+    while (true) {
+      uint8_t h = byte();
+      if (h == 0) return out;
+      int type = h & 0x0F;
+      int delta = h >> 4;
+      int16_t id = delta ? (int16_t)(last_id + delta) : (int16_t)zigzag();
+      last_id = id;
+      int norm = type == 2 ? 1 : type;  // bool value lives in the nibble
+      out[py::int_(id)] = py::make_tuple(norm, tvalue(type));
+    }
+  }
+};
+
+struct Writer {
+  std::string out;
+
+  void byte(uint8_t b) { out.push_back((char)b); }
+
+  void varint(uint64_t v) {
+    while (v >= 0x80) {
+      byte((uint8_t)(v | 0x80));
+      v >>= 7;
+    }
+    byte((uint8_t)v);
+  }
+
+  void zigzag(int64_t v) { varint(((uint64_t)v << 1) ^ (uint64_t)(v >> 63)); }
+
+  void value(int type, py::handle v) {
+    switch (type) {
+      case 1:
+      case 2: return;  // bool encoded in the type nibble / element header
+      case 3: byte((uint8_t)(int8_t)py::cast<int64_t>(v)); return;
+      case 4:
+      case 5:
+      case 6: zigzag(py::cast<int64_t>(v)); return;
+      case 7: {
+        double d = py::cast<double>(v);
+        char buf[8];
+        memcpy(buf, &d, 8);
+        out.append(buf, 8);
+        return;
+      }
+      case 8: {
+        std::string s = py::cast<std::string>(v);
+        varint(s.size());
+        out += s;
+        return;
+      }
+      case 9:
+      case 10: {
+        auto t = py::cast<py::tuple>(v);
+        int etype = py::cast<int>(t[0]);
+        auto elems = py::cast<py::list>(t[1]);
+        size_t n = elems.size();
+        if (n < 15) {
+          byte((uint8_t)((n << 4) | etype));
+        } else {
+          byte((uint8_t)(0xF0 | etype));
+          varint(n);
+        }
+        for (auto e : elems) {
+          // bool list elements carry the value as a 1/2 byte
+          if (etype == 1 || etype == 2)
+            byte(py::cast<bool>(e) ? 1 : 2);
+          else
+            value(etype, e);
+        }
+        return;
+      }
+      case 11: {
+        auto t = py::cast<py::tuple>(v);
+        int ktype = py::cast<int>(t[0]), vtype = py::cast<int>(t[1]);
+        auto pairs = py::cast<py::list>(t[2]);
+        varint(pairs.size());
+        if (pairs.size() > 0) {
+          byte((uint8_t)((ktype << 4) | vtype));
+          for (auto pr : pairs) {
+            auto kv = py::cast<py::tuple>(pr);
+            value(ktype, kv[0]);
+            value(vtype, kv[1]);
+          }
+        }
+        return;
+      }
+      case 12: strct(py::cast<py::dict>(v)); return;
+      default:
+        throw std::runtime_error("thrift write: unknown type " +
+                                 std::to_string(type));
+    }
+  }
+
+  void strct(py::dict d) {
+    int16_t last_id = 0;
+    for (auto item : d) {
+      int16_t id = (int16_t)py::cast<int64_t>(item.first);
+      auto tv = py::cast<py::tuple>(item.second);
+      int type = py::cast<int>(tv[0]);
+      py::handle v = tv[1];
+      int wire = type;
+      if (type == 1 || type == 2) wire = py::cast<bool>(v) ? 1 : 2;
+      int delta = id - last_id;
+      if (delta >= 1 && delta <= 15) {
+        byte((uint8_t)((delta << 4) | wire));
+      } else {
+        byte((uint8_t)wire);
+        zigzag(id);
+      }
+      last_id = id;
+      value(wire, v);
+    }
+    byte(0);  // STOP
+  }
+};
+
 }  // namespace
 
 void register_thrift(py::module_& m) {
@@ -124,4 +308,26 @@ void register_thrift(py::module_& m) {
         },
         py::arg("data"), py::arg("offset") = 0,
         "Parse one compact-protocol struct; returns ({field_id: value}, end)");
+  m.def("thrift_parse_typed",
+        [](py::bytes data, size_t offset) {
+          char* buf;
+          py::ssize_t n;
+          if (PyBytes_AsStringAndSize(data.ptr(), &buf, &n) != 0)
+            throw std::runtime_error("bad bytes");
+          TypedReader r;
+          r.p = reinterpret_cast<const uint8_t*>(buf);
+          r.len = (size_t)n;
+          r.pos = offset;
+          py::dict d = r.tstrct();
+          return py::make_tuple(d, r.pos);
+        },
+        py::arg("data"), py::arg("offset") = 0,
+        "Parse one struct keeping wire types: {id: (type, value)}");
+  m.def("thrift_write",
+        [](py::dict d) {
+          Writer w;
+          w.strct(d);
+          return py::bytes(w.out);
+        },
+        "Serialize a typed struct tree back to compact protocol");
 }
