@@ -82,8 +82,8 @@ def to_float(col: Column, ansi: bool = False,
              dtype: DType = DType.FLOAT64) -> Column:
     """CastStrings.toFloat (reference cast_string_to_float.cu:828).
 
-    Known gap vs the reference: rounding is within 1 ulp of correctly-rounded
-    for long-mantissa inputs (no Eisel-Lemire fallback yet)."""
+    Exactly rounded via Eisel-Lemire for both FLOAT64 and FLOAT32 (binary32
+    rounds directly from the decimal digits — no double intermediate)."""
     g = _native.gpu()
     n = col.size
     out = torch.empty(n, dtype=TORCH_DTYPE[dtype], device=col.device)

@@ -10,8 +10,8 @@
 //     ExceptionWithRowIndex-style errors (exception_with_row_index.hpp:25).
 //   * bool: t/true/y/yes/1 and f/false/n/no/0, case-insensitive.
 //   * float: trim, special literals (inf/infinity/nan, signed), decimal +
-//     exponent parse. NOTE: rounding is within 1 ulp of correctly-rounded
-//     (double-double steps), not yet bit-exact Eisel-Lemire — tracked gap.
+//     exponent parse; exactly-rounded Eisel-Lemire for BOTH binary64 and
+//     binary32 (direct, no double-rounding through an intermediate double).
 //   * decimal: digits -> __int128 unscaled with HALF_UP rescale to the
 //     target scale, precision overflow -> null/ANSI error.
 //   * date/timestamp: Spark patterns yyyy[-M[-d]][ |T[h:m:s[.us][zone]]],
@@ -154,9 +154,18 @@ __device__ inline double pow10_pos(int e) {
 
 #include "exact_fp.inc"
 
-__device__ bool parse_double(StrView s, double* out) {
+struct DecParts {
+  uint64_t mant;
+  int exp_adj;
+  bool neg;
+  bool dropped_nonzero;
+  bool dropped;
+};
+
+// Shared decimal-literal scan: returns 0 invalid, 1 number, 2 inf, 3 nan.
+__device__ int parse_dec_parts(StrView s, DecParts* o) {
   s = trim_all(s);
-  if (s.len == 0) return false;
+  if (s.len == 0) return 0;
   int i = 0;
   bool neg = false;
   if (s.ptr[0] == '+' || s.ptr[0] == '-') { neg = s.ptr[0] == '-'; i = 1; }
@@ -175,15 +184,14 @@ __device__ bool parse_double(StrView s, double* out) {
           return true;
         };
         if (eq("inf", 3) || eq("infinity", 8)) {
-          *out = neg ? -INFINITY : INFINITY;
-          return true;
+          o->neg = neg;
+          return 2;
         }
         if (eq("nan", 3)) {
-          *out = NAN;
-          return true;
+          return 3;
         }
       }
-      if (c0 == 'n' || c0 == 'i') return false;
+      if (c0 == 'n' || c0 == 'i') return 0;
     }
   }
   uint64_t mant = 0;
@@ -203,17 +211,17 @@ __device__ bool parse_double(StrView s, double* out) {
         if (!dot) ++exp_adj;
       }
     } else if (c == '.') {
-      if (dot) return false;
+      if (dot) return 0;
       dot = true;
     } else if (c == 'e' || c == 'E') {
-      if (!any) return false;
+      if (!any) return 0;
       ++i;
       bool eneg = false;
       if (i < s.len && (s.ptr[i] == '+' || s.ptr[i] == '-')) {
         eneg = s.ptr[i] == '-';
         ++i;
       }
-      if (i >= s.len) return false;
+      if (i >= s.len) return 0;
       long e = 0;
       for (; i < s.len; ++i) {
         char ec = s.ptr[i];
@@ -221,7 +229,7 @@ __device__ bool parse_double(StrView s, double* out) {
           if ((ec == 'd' || ec == 'D' || ec == 'f' || ec == 'F') &&
               i == s.len - 1)
             break;  // Java-style suffix accepted by Spark's parse
-          return false;
+          return 0;
         }
         if (e < 100000) e = e * 10 + (ec - '0');
       }
@@ -231,21 +239,51 @@ __device__ bool parse_double(StrView s, double* out) {
                i == s.len - 1) {
       break;  // trailing type suffix
     } else {
-      return false;
+      return 0;
     }
   }
-  if (!any) return false;
+  if (!any) return 0;
+  o->mant = mant;
+  o->exp_adj = exp_adj;
+  o->neg = neg;
+  o->dropped_nonzero = dropped_nonzero;
+  o->dropped = dropped;
+  return 1;
+}
+
+__device__ inline double approx_pow10_scale(uint64_t mant, int exp_adj,
+                                            bool neg) {
+  // fallback (rare ambiguous cases): within-1-ulp approximation
+  double d = (double)mant;
+  if (exp_adj > 0) {
+    if (exp_adj > 308 + 19) return neg ? -INFINITY : INFINITY;
+    d *= pow10_pos(exp_adj);
+  } else if (exp_adj < 0) {
+    int e = -exp_adj;
+    if (e > 342 + 19) return neg ? -0.0 : 0.0;
+    while (e > 22) { d /= 1e22; e -= 22; }
+    d /= pow10_pos(e);
+  }
+  return neg ? -d : d;
+}
+
+__device__ bool parse_double(StrView s, double* out) {
+  DecParts p;
+  int r = parse_dec_parts(s, &p);
+  if (r == 0) return false;
+  if (r == 2) { *out = p.neg ? -INFINITY : INFINITY; return true; }
+  if (r == 3) { *out = NAN; return true; }
   // exact path: Eisel-Lemire; bracket w/w+1 when digits were truncated
   double exact;
-  if (!dropped || !dropped_nonzero) {
-    if (eisel_lemire(mant, exp_adj, neg, &exact)) {
+  if (!p.dropped || !p.dropped_nonzero) {
+    if (eisel_lemire(p.mant, p.exp_adj, p.neg, &exact)) {
       *out = exact;
       return true;
     }
   } else {
     double lo_v, hi_v;
-    if (eisel_lemire(mant, exp_adj, neg, &lo_v) &&
-        eisel_lemire(mant + 1, exp_adj, neg, &hi_v)) {
+    if (eisel_lemire(p.mant, p.exp_adj, p.neg, &lo_v) &&
+        eisel_lemire(p.mant + 1, p.exp_adj, p.neg, &hi_v)) {
       int64_t lb, hb;
       __builtin_memcpy(&lb, &lo_v, 8);
       __builtin_memcpy(&hb, &hi_v, 8);
@@ -255,19 +293,46 @@ __device__ bool parse_double(StrView s, double* out) {
       }
     }
   }
-  // fallback (rare ambiguous cases): within-1-ulp approximation
-  double d = (double)mant;
-  if (exp_adj > 0) {
-    if (exp_adj > 308 + 19) { *out = neg ? -INFINITY : INFINITY; return true; }
-    d *= pow10_pos(exp_adj);
-  } else if (exp_adj < 0) {
-    int e = -exp_adj;
-    if (e > 342 + 19) { *out = neg ? -0.0 : 0.0; return true; }
-    while (e > 22) { d /= 1e22; e -= 22; }
-    d /= pow10_pos(e);
-  }
-  *out = neg ? -d : d;
+  *out = approx_pow10_scale(p.mant, p.exp_adj, p.neg);
   return true;
+}
+
+// Direct exactly-rounded string->float (no double intermediate, so no
+// double-rounding; reference cast_string_to_float.cu parses per-type too).
+__device__ bool parse_float(StrView s, float* out) {
+  DecParts p;
+  int r = parse_dec_parts(s, &p);
+  if (r == 0) return false;
+  if (r == 2) { *out = p.neg ? -INFINITY : INFINITY; return true; }
+  if (r == 3) { *out = NAN; return true; }
+  float exact;
+  if (!p.dropped || !p.dropped_nonzero) {
+    if (eisel_lemire_f32(p.mant, p.exp_adj, p.neg, &exact)) {
+      *out = exact;
+      return true;
+    }
+  } else {
+    float lo_v, hi_v;
+    if (eisel_lemire_f32(p.mant, p.exp_adj, p.neg, &lo_v) &&
+        eisel_lemire_f32(p.mant + 1, p.exp_adj, p.neg, &hi_v)) {
+      int32_t lb, hb;
+      __builtin_memcpy(&lb, &lo_v, 4);
+      __builtin_memcpy(&hb, &hi_v, 4);
+      if (lb == hb) {
+        *out = lo_v;
+        return true;
+      }
+    }
+  }
+  *out = (float)approx_pow10_scale(p.mant, p.exp_adj, p.neg);
+  return true;
+}
+
+__device__ inline bool parse_fp(StrView s, double* out) {
+  return parse_double(s, out);
+}
+__device__ inline bool parse_fp(StrView s, float* out) {
+  return parse_float(s, out);
 }
 
 
@@ -282,12 +347,12 @@ __global__ void string_to_float_kernel(ColDesc in, int64_t nrows,
        row += stride) {
     bool in_range = row < nrows;
     bool valid = false;
-    double v = 0;
+    T v = 0;
     if (in_range && is_valid(in.valid, row)) {
-      valid = parse_double(get_string(in, row), &v);
+      valid = parse_fp(get_string(in, row), &v);
       if (!valid) record_error(err_row, row);
     }
-    if (in_range) out[row] = (T)v;
+    if (in_range) out[row] = v;
     ballot_write_validity(out_valid, row, valid);
   }
 }

@@ -442,3 +442,52 @@ def test_to_timestamp_with_format():
     col3 = Column.from_pylist(["2020-01-02 03:04:05.123"], DType.STRING, "cuda")
     got3 = cast.to_timestamp_with_format(col3, "yyyy-MM-dd HH:mm:ss.SSS")
     assert got3.to_pylist()[0] == us(2020, 1, 2, 3, 4, 5, micro=123000)
+
+
+@pytest.mark.gpu
+def test_string_to_float32_exact_rounding():
+    """Direct binary32 rounding (no double intermediate): oracle picks the
+    nearest float32 by exact Fraction comparison among the 1-ulp neighbors
+    of the double-rounded candidate."""
+    import math
+    import struct as st
+    from fractions import Fraction
+
+    import numpy as np
+    from spark_rapids_jni_amd.ops import cast as srj_cast
+
+    rnd = random.Random(97)
+    cases = ["7.038531e-26", "1.1754944e-38", "16777217", "33554431",
+             "0.1", "2.3509887e-38", "3.4028236e38", "1e-45", "7e-46",
+             "1.17549435e-38", "0.000001", "8388609.499999999"]
+    for _ in range(3000):
+        mant = rnd.randint(0, 10**rnd.randint(1, 17))
+        frac = rnd.randint(0, 10**rnd.randint(1, 10))
+        exp = rnd.randint(-45, 39)
+        cases.append(f"{mant}.{frac}e{exp}")
+    col = Column.from_pylist(cases, DType.STRING, "cuda")
+    got = srj_cast.to_float(col, dtype=DType.FLOAT32).data.cpu().numpy()
+
+    def oracle(s):
+        v = Fraction(s.split("e")[0]) \
+            * Fraction(10) ** int(s.split("e")[1] if "e" in s else 0)
+        # round-to-nearest overflow threshold: (2 - 2^-24) * 2^127
+        if v >= Fraction(2**128) - Fraction(2**103):
+            return np.float32("inf")
+        cand = np.float32(min(float(s), 3.4028234e38))  # within 1 ulp
+        bits = st.unpack("<I", st.pack("<f", cand))[0]
+        best, bestkey = None, None
+        for b in {max(bits - 1, 0), bits, bits + 1}:
+            f = st.unpack("<f", st.pack("<I", b))[0]
+            if math.isinf(f) or math.isnan(f):
+                continue
+            key = (abs(Fraction(f) - v), b & 1)  # ties-to-even
+            if bestkey is None or key < bestkey:
+                best, bestkey = np.float32(f), key
+        return best
+
+    for i, s in enumerate(cases):
+        exp = oracle(s)
+        g = got[i]
+        assert st.pack("<f", g) == st.pack("<f", exp), \
+            f"{s}: got {g!r} want {exp!r}"
