@@ -40,6 +40,45 @@ def gather_column(col: Column, gmap: torch.Tensor,
                            stream)
         return Column(DType.STRING, n, chars[:nchars], validity, offsets,
                       null_count=None)
+    if col.dtype == DType.STRUCT:
+        validity = None
+        if out_nullable:
+            validity = make_validity(n, dev)
+            g.gather_validity(
+                col.validity.data_ptr() if col.validity is not None else 0,
+                gmap.data_ptr(), n, validity.data_ptr(), stream)
+        children = [gather_column(ch, gmap, has_nulls) for ch in col.children]
+        return Column(DType.STRUCT, n, None, validity, children=children,
+                      null_count=None)
+    if col.dtype == DType.LIST:
+        lens = torch.empty(max(n, 1), dtype=torch.int32, device=dev)
+        g.gather_str_lengths(col.offsets.data_ptr(), gmap.data_ptr(), n,
+                             lens.data_ptr(), stream)
+        offsets = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+        if n:
+            torch.cumsum(lens[:n], 0, out=offsets[1:].view(n))
+        validity = None
+        if out_nullable:
+            validity = make_validity(n, dev)
+            g.gather_validity(
+                col.validity.data_ptr() if col.validity is not None else 0,
+                gmap.data_ptr(), n, validity.data_ptr(), stream)
+        # child gather map: src child rows of each gathered list, in order
+        lens64 = lens[:n].to(torch.int64)
+        total_child = int(offsets[-1].item())
+        if total_child > 0:
+            starts = col.offsets.to(torch.int64)[gmap.clamp(min=0)]
+            base = torch.repeat_interleave(starts, lens64)
+            first = torch.repeat_interleave(offsets[:-1].to(torch.int64),
+                                            lens64)
+            child_map = base + (
+                torch.arange(total_child, dtype=torch.int64, device=dev) -
+                first)
+        else:
+            child_map = torch.empty(0, dtype=torch.int64, device=dev)
+        child = gather_column(col.children[0], child_map, has_nulls)
+        return Column(DType.LIST, n, None, validity, offsets,
+                      children=[child], null_count=None)
     width = FIXED_WIDTH[col.dtype]
     numel = n * 2 if col.dtype == DType.DECIMAL128 else n
     out = torch.empty(numel, dtype=col.data.dtype, device=dev)

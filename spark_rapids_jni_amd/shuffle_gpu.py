@@ -11,8 +11,10 @@ tests/test_shuffle_gpu.py), so records can be merged by either path. The
 buffers are laid out contiguously per destination rank and feed RCCL
 all_to_all_single directly — no host bounce.
 
-Current device-path scope: flat schemas (fixed-width + string columns);
-nested types take the host kudo path.
+Nested schemas (STRUCT/LIST, arbitrarily deep) are handled natively:
+child row ranges derive from parent offset values (fetched at partition
+boundaries on split; staged offset sections on assemble), matching the
+host-path slicing semantics exactly.
 """
 import struct
 from typing import List, Optional, Sequence, Tuple
@@ -57,9 +59,32 @@ class _SegBatch:
         return segs_t, pref_t  # keepalive
 
 
-def _check_flat(cols: Sequence[Column]):
+def _flatten_plan(cols, bounds, dev):
+    """Depth-first flatten with per-partition absolute row bounds.
+
+    bounds: list of nparts+1 absolute row indices for this nesting level.
+    Returns [(col, bounds, char_bounds)] — char_bounds only for STRING
+    (absolute char positions at partition boundaries, one small D2H per
+    offset-bearing column, mirroring the host writer kudo.py:152).
+    """
+    import torch as _t
+    out = []
     for c in cols:
-        assert not c.children, "device shuffle path supports flat schemas"
+        cb = None
+        child_bounds = None
+        if c.dtype in (DType.STRING, DType.LIST):
+            idx = _t.tensor(bounds, dtype=_t.int64, device=dev)
+            vals = c.offsets[idx].cpu().tolist()
+            if c.dtype == DType.STRING:
+                cb = vals
+            else:
+                child_bounds = vals
+        out.append((c, bounds, cb))
+        if c.dtype == DType.STRUCT:
+            out.extend(_flatten_plan(c.children, bounds, dev))
+        elif c.dtype == DType.LIST:
+            out.extend(_flatten_plan([c.children[0]], child_bounds, dev))
+    return out
 
 
 def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
@@ -67,9 +92,10 @@ def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
                                   ) -> Tuple[torch.Tensor, List[int]]:
     """Split rows (already partition-grouped by `perm`) into P contiguous
     device Kudo records. Returns (one uint8 buffer, per-partition byte sizes).
+    Supports nested schemas (STRUCT/LIST): child row ranges derive from the
+    parent's offset values at partition boundaries, as in the host writer.
     """
     from .ops.copying import gather
-    _check_flat(table.columns)
     g = _native.gpu()
     stream = _native.current_stream()
     dev = table.device
@@ -77,17 +103,9 @@ def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
     offs = offsets.cpu().tolist()
     nparts = len(offs) - 1
 
-    ncols = len(G.columns)
+    flat = _flatten_plan(G.columns, offs, dev)
+    ncols = len(flat)
     hlen = 28 + (ncols + 7) // 8
-
-    # string char boundaries per partition (one small D2H)
-    str_cols = [c for c in G.columns if c.dtype == DType.STRING]
-    char_bounds = {}
-    if str_cols:
-        for ci, c in enumerate(G.columns):
-            if c.dtype == DType.STRING:
-                idx = torch.tensor(offs, dtype=torch.int64, device=dev)
-                char_bounds[ci] = c.offsets[idx].cpu().tolist()
 
     # plan per-partition record layout
     headers = bytearray()
@@ -96,28 +114,30 @@ def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
     positions = []
     pos = 0
     for p in range(nparts):
-        start, end = offs[p], offs[p + 1]
-        n = end - start
+        top_start, top_n = offs[p], offs[p + 1] - offs[p]
         bitset = bytearray((ncols + 7) // 8)
         validity_parts = []  # (src_ptr, nbytes)
         offset_parts = []
         data_parts = []
-        for ci, c in enumerate(G.columns):
+        for ci, (c, bnds, cb) in enumerate(flat):
+            start, end = bnds[p], bnds[p + 1]
+            n = end - start
             if c.validity is not None and n > 0:
                 # n == 0: prefer "no validity" (kudo spec) over a filler byte
                 bitset[ci // 8] |= 1 << (ci % 8)
                 b0 = start // 8
                 b1 = max((start + n + 7) // 8, b0 + 1)
                 validity_parts.append((c.validity.data_ptr() + b0, b1 - b0))
-            if c.dtype == DType.STRING:
+            if c.dtype in (DType.STRING, DType.LIST):
                 if n > 0:
                     offset_parts.append((c.offsets.data_ptr() + start * 4,
                                          (n + 1) * 4))
-                    cb = char_bounds[ci]
+            if c.dtype == DType.STRING:
+                if n > 0:
                     nchars = cb[p + 1] - cb[p]
                     if nchars > 0:
                         data_parts.append((c.data.data_ptr() + cb[p], nchars))
-            else:
+            elif c.dtype not in (DType.STRUCT, DType.LIST):
                 w = FIXED_WIDTH[c.dtype]
                 if n > 0:
                     data_parts.append((c.data.data_ptr() + start * w, n * w))
@@ -126,7 +146,8 @@ def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
         olen = sum(x[1] for x in offset_parts)
         dlen = sum(x[1] for x in data_parts)
         dpad = (4 - dlen % 4) % 4
-        h = KudoTableHeader(start, n, vlen + vpad, olen, 0, ncols, bytes(bitset))
+        h = KudoTableHeader(top_start, top_n, vlen + vpad, olen, 0, ncols,
+                            bytes(bitset))
         h.total_len = h.validity_len + olen + dlen + dpad
         import io
         b = io.BytesIO()
@@ -163,147 +184,207 @@ def split_and_serialize_to_device(table: Table, offsets: torch.Tensor,
     return out, sizes
 
 
+def _flat_count(cols) -> int:
+    n = 0
+    for c in cols:
+        n += 1
+        if c.dtype == DType.STRUCT:
+            n += _flat_count(c.children)
+        elif c.dtype == DType.LIST:
+            n += _flat_count([c.children[0]])
+    return n
+
+
 def assemble_from_device(buffers: Sequence[torch.Tensor],
                          schema: Sequence[Column]) -> Table:
-    """Merge device Kudo records into one device table (inverse of split)."""
-    _check_flat(schema)
+    """Merge device Kudo records into one device table (inverse of split).
+
+    Nested schemas supported: each piece's offsets section is staged to host
+    once for layout planning (counts/positions); validity merge, offset
+    rebase and data copies all run in batched device kernels.
+    """
     g = _native.gpu()
     stream = _native.current_stream()
     dev = buffers[0].device
-    ncols = len(schema)
+    ncols = _flat_count(schema)
     hlen = 28 + (ncols + 7) // 8
 
-    # parse headers (small D2H)
+    import io
+
+    # parse headers + stage offset sections (small D2H per piece)
     headers: List[KudoTableHeader] = []
+    off_sections: List[np.ndarray] = []
     for b in buffers:
-        import io
         raw = b[:hlen].cpu().numpy().tobytes()
         h = KudoTableHeader.read(io.BytesIO(raw))
+        assert h.num_columns == ncols, \
+            f"schema mismatch: record has {h.num_columns} flat columns"
         headers.append(h)
+        o0 = hlen + h.validity_len
+        sec = (b[o0:o0 + h.offset_len].cpu().numpy().view(np.int32)
+               if h.offset_len else np.empty(0, dtype=np.int32))
+        off_sections.append(sec)
 
-    # per piece/column section positions (record-relative, host arithmetic)
-    # order: validities | pad | offsets | data
-    piece_cols = []  # [piece][col] = dict(valid_ptr, off_ptr, data_ptr?, ...)
-    gather_addrs = []  # addresses of first/last offset values per string piece
-    for b, h in zip(buffers, headers):
+    # walk each piece's flattened schema computing per-column layout
+    # rec: dict(n, start_bit, valid_ptr?, off_ptr?, first_off?, data_ptr?, ...)
+    piece_cols: List[List[dict]] = []
+    for b, h, osec in zip(buffers, headers, off_sections):
         base = b.data_ptr()
-        start_bit = h.offset % 8
         vpos = hlen
         opos = hlen + h.validity_len
         dpos = opos + h.offset_len
-        cols = []
-        for ci, c in enumerate(schema):
-            rec = {"n": h.num_rows, "start_bit": start_bit}
-            if h.has_validity_buffer(ci):
-                nb = (start_bit + h.num_rows + 7) // 8 if h.num_rows else 1
-                rec["valid_ptr"] = base + vpos
-                vpos += nb
-            if c.dtype == DType.STRING:
-                if h.num_rows > 0:
+        oidx = 0  # index into osec (int32 words)
+        cols: List[dict] = []
+
+        def walk(cs, start_bit, n):
+            nonlocal vpos, opos, dpos, oidx
+            for c in cs:
+                ci = len(cols)
+                rec = {"n": n, "start_bit": start_bit}
+                cols.append(rec)
+                if h.has_validity_buffer(ci):
+                    if n > 0:
+                        rec["valid_ptr"] = base + vpos
+                        vpos += (start_bit + n + 7) // 8
+                    else:
+                        vpos += 1  # host-writer filler byte
+                first = last = 0
+                if c.dtype in (DType.STRING, DType.LIST) and n > 0:
                     rec["off_ptr"] = base + opos
-                    opos += (h.num_rows + 1) * 4
-                    gather_addrs.append(rec["off_ptr"])
-                    gather_addrs.append(rec["off_ptr"] + h.num_rows * 4)
-            cols.append(rec)
-        # data positions need char counts -> fill after gather
-        rec_meta = {"base": base, "dpos": dpos, "cols": cols}
-        piece_cols.append(rec_meta)
+                    first = int(osec[oidx])
+                    last = int(osec[oidx + n])
+                    rec["first_off"] = first
+                    opos += (n + 1) * 4
+                    oidx += n + 1
+                if c.dtype == DType.STRING:
+                    nch = last - first
+                    rec["nchars"] = nch
+                    if nch > 0:
+                        rec["data_ptr"] = base + dpos
+                        dpos += nch
+                elif c.dtype == DType.STRUCT:
+                    walk(c.children, start_bit, n)
+                elif c.dtype == DType.LIST:
+                    walk([c.children[0]], first % 8, last - first)
+                elif n > 0:
+                    w = FIXED_WIDTH[c.dtype]
+                    rec["data_ptr"] = base + dpos
+                    dpos += n * w
+        walk(list(schema), h.offset % 8, h.num_rows)
+        piece_cols.append(cols)
 
-    # fetch first/last offsets of every string piece in one kernel + D2H
-    char_counts = {}
-    if gather_addrs:
-        addrs = torch.tensor(gather_addrs, dtype=torch.int64, device=dev)
-        vals = torch.empty(len(gather_addrs), dtype=torch.int32, device=dev)
-        g.gather_i32_at(addrs.data_ptr(), len(gather_addrs), vals.data_ptr(),
-                        stream)
-        flat = vals.cpu().tolist()
-        k = 0
-        for pi, meta in enumerate(piece_cols):
-            for ci, c in enumerate(schema):
-                if c.dtype == DType.STRING and "off_ptr" in meta["cols"][ci]:
-                    first, last = flat[k], flat[k + 1]
-                    k += 2
-                    char_counts[(pi, ci)] = (first, last - first)
-
-    # walk data section per piece
-    for pi, meta in enumerate(piece_cols):
-        dpos = meta["dpos"]
-        for ci, c in enumerate(schema):
-            rec = meta["cols"][ci]
-            if rec["n"] == 0:
-                continue
-            if c.dtype == DType.STRING:
-                first, nchars = char_counts.get((pi, ci), (0, 0))
-                rec["first_off"] = first
-                if nchars > 0:
-                    rec["data_ptr"] = meta["base"] + dpos
-                    rec["nchars"] = nchars
-                    dpos += nchars
-            else:
-                w = FIXED_WIDTH[c.dtype]
-                rec["data_ptr"] = meta["base"] + dpos
-                dpos += rec["n"] * w
-
-    total_rows = sum(h.num_rows for h in headers)
-    out_cols = []
     copy_batch = _SegBatch()
     vsegs, vwords = [], []
     osegs, ocounts = [], []
     keepalive = []
-    for ci, c in enumerate(schema):
-        any_valid = any("valid_ptr" in meta["cols"][ci] for meta in piece_cols)
-        validity = None
-        if any_valid:
-            validity = torch.zeros(validity_nbytes(total_rows), dtype=torch.uint8,
-                                   device=dev)
-        if c.dtype == DType.STRING:
-            out_offs = torch.empty(total_rows + 1, dtype=torch.int32, device=dev)
-            nchars_total = sum(char_counts.get((pi, ci), (0, 0))[1]
-                               for pi in range(len(piece_cols)))
-            out_chars = torch.empty(max(nchars_total, 1), dtype=torch.uint8,
-                                    device=dev)
-        else:
+    flat_idx = [0]
+
+    def build(cs) -> List[Column]:
+        out_cols = []
+        for c in cs:
+            ci = flat_idx[0]
+            flat_idx[0] += 1
+            recs = [pc[ci] for pc in piece_cols]
+            total_rows = sum(r["n"] for r in recs)
+            any_valid = any("valid_ptr" in r for r in recs)
+            validity = None
+            if any_valid:
+                validity = torch.zeros(validity_nbytes(total_rows),
+                                       dtype=torch.uint8, device=dev)
+                row_pos = 0
+                for r in recs:
+                    n = r["n"]
+                    if n > 0:
+                        src = r.get("valid_ptr", 0)
+                        sbit = r["start_bit"] if src else 0
+                        vsegs.append(struct.pack(
+                            _VALIDSEG, src, validity.data_ptr(), sbit,
+                            row_pos, n))
+                        w0 = row_pos // 64
+                        w1 = (row_pos + n - 1) // 64
+                        vwords.append(w1 - w0 + 1)
+                    row_pos += n
+            offsets_t = None
+            if c.dtype in (DType.STRING, DType.LIST):
+                offsets_t = torch.zeros(total_rows + 1, dtype=torch.int32,
+                                        device=dev)
+                row_pos = 0
+                base_val = 0
+                for r in recs:
+                    n = r["n"]
+                    if n > 0:
+                        child_n = (r["nchars"] if c.dtype == DType.STRING
+                                   else pcount(r, c))
+                        is_last = row_pos + n == total_rows
+                        osegs.append(struct.pack(
+                            _OFFSEG, r["off_ptr"],
+                            offsets_t.data_ptr() + row_pos * 4, n, base_val,
+                            1 if is_last else 0))
+                        ocounts.append(n + (1 if is_last else 0))
+                        base_val += child_n
+                        row_pos += n
+            if c.dtype == DType.STRUCT:
+                children = build(c.children)
+                out_cols.append(Column(DType.STRUCT, total_rows, None,
+                                       validity, children=children,
+                                       null_count=None))
+                continue
+            if c.dtype == DType.LIST:
+                children = build([c.children[0]])
+                out_cols.append(Column(DType.LIST, total_rows, None, validity,
+                                       offsets_t, children, c.scale,
+                                       null_count=None))
+                continue
+            if c.dtype == DType.STRING:
+                nchars_total = sum(r.get("nchars", 0) for r in recs)
+                out_chars = torch.empty(max(nchars_total, 1),
+                                        dtype=torch.uint8, device=dev)
+                char_pos = 0
+                for r in recs:
+                    nch = r.get("nchars", 0)
+                    if nch > 0:
+                        copy_batch.add(r["data_ptr"],
+                                       out_chars.data_ptr() + char_pos, nch)
+                    char_pos += nch
+                out_cols.append(Column(DType.STRING, total_rows, out_chars,
+                                       validity, offsets_t, null_count=None))
+                continue
             from .columnar import TORCH_DTYPE
+            w = FIXED_WIDTH[c.dtype]
             numel = total_rows * (2 if c.dtype == DType.DECIMAL128 else 1)
-            out_data = torch.empty(numel, dtype=TORCH_DTYPE[c.dtype],
+            out_data = torch.empty(max(numel, 1), dtype=TORCH_DTYPE[c.dtype],
                                    device=dev)
-        row_pos = 0
-        char_pos = 0
-        for pi, meta in enumerate(piece_cols):
-            rec = meta["cols"][ci]
-            n = rec["n"]
-            if validity is not None and n > 0:
-                src = rec.get("valid_ptr", 0)
-                sbit = rec["start_bit"] if src else 0
-                vsegs.append(struct.pack(_VALIDSEG, src, validity.data_ptr(),
-                                         sbit, row_pos, n))
-                w0 = row_pos // 64
-                w1 = (row_pos + n - 1) // 64
-                vwords.append(w1 - w0 + 1)
-            if c.dtype == DType.STRING and n > 0:
-                nch = char_counts.get((pi, ci), (0, 0))[1]
-                is_last = row_pos + n == total_rows
-                osegs.append(struct.pack(
-                    _OFFSEG, rec["off_ptr"], out_offs.data_ptr() + row_pos * 4,
-                    n, char_pos, 1 if is_last else 0))
-                ocounts.append(n + (1 if is_last else 0))
-                if nch > 0:
-                    copy_batch.add(rec["data_ptr"],
-                                   out_chars.data_ptr() + char_pos, nch)
-                char_pos += nch
-            elif n > 0 and c.dtype != DType.STRING:
-                w = FIXED_WIDTH[c.dtype]
-                copy_batch.add(rec["data_ptr"], out_data.data_ptr() + row_pos * w,
-                               n * w)
-            row_pos += n
-        if c.dtype == DType.STRING:
-            if total_rows == 0:
-                out_offs = torch.zeros(1, dtype=torch.int32, device=dev)
-            out_cols.append(Column(DType.STRING, total_rows, out_chars, validity,
-                                   out_offs, null_count=None))
-        else:
+            row_pos = 0
+            for r in recs:
+                n = r["n"]
+                if n > 0:
+                    copy_batch.add(r["data_ptr"],
+                                   out_data.data_ptr() + row_pos * w, n * w)
+                row_pos += n
             out_cols.append(Column(c.dtype, total_rows, out_data, validity,
                                    scale=c.scale, null_count=None))
+        return out_cols
+
+    def pcount(rec, c):
+        # child rows contributed by this piece for a LIST column
+        return rec["_child_n"] if rec["n"] > 0 else 0
+
+    # fill _child_n from the already-walked per-piece layouts
+    def fill_child_counts(cs, idx):
+        for c in cs:
+            ci = idx[0]
+            idx[0] += 1
+            if c.dtype == DType.LIST:
+                child_idx = idx[0]
+                for pc in piece_cols:
+                    r = pc[ci]
+                    r["_child_n"] = pc[child_idx]["n"] if r["n"] > 0 else 0
+                fill_child_counts([c.children[0]], idx)
+            elif c.dtype == DType.STRUCT:
+                fill_child_counts(c.children, idx)
+    fill_child_counts(list(schema), [0])
+
+    out_cols = build(list(schema))
 
     keepalive.append(copy_batch.run(dev, g, stream))
     if vsegs:

@@ -4,6 +4,8 @@
 extern "C" {
 void srj_gather_fixed(const void*, const uint8_t*, const int64_t*, int64_t, void*,
                       uint8_t*, int32_t, hipStream_t);
+void srj_gather_validity(const uint8_t*, const int64_t*, int64_t, uint8_t*,
+                         hipStream_t);
 void srj_gather_str_lengths(const int32_t*, const int64_t*, int64_t, int32_t*,
                             hipStream_t);
 void srj_gather_str_chars(const char*, const int32_t*, const uint8_t*,
@@ -23,6 +25,13 @@ void register_copying(py::module_& m) {
                            as_ptr<int64_t>(map), n, as_ptr<void>(out),
                            as_ptr<uint8_t>(out_valid), elem_size, as_stream(stream));
           check_hip("gather_fixed");
+        });
+  m.def("gather_validity",
+        [](uintptr_t in_valid, uintptr_t map, int64_t n, uintptr_t out_valid,
+           uintptr_t stream) {
+          srj_gather_validity(as_ptr<uint8_t>(in_valid), as_ptr<int64_t>(map),
+                              n, as_ptr<uint8_t>(out_valid), as_stream(stream));
+          check_hip("gather_validity");
         });
   m.def("gather_str_lengths",
         [](uintptr_t offsets, uintptr_t map, int64_t n, uintptr_t lens,

@@ -71,6 +71,21 @@ __global__ void gather_str_chars_kernel(const char* __restrict__ in_chars,
   }
 }
 
+// validity-only gather (STRUCT/LIST columns have no data buffer)
+__global__ void gather_validity_kernel(const uint8_t* __restrict__ in_valid,
+                                       const int64_t* __restrict__ map,
+                                       int64_t n,
+                                       uint8_t* __restrict__ out_valid) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < ((n + WAVE - 1) & ~(int64_t)(WAVE - 1)); i += stride) {
+    bool in_range = i < n;
+    int64_t idx = in_range ? map[i] : -1;
+    bool valid = in_range && idx >= 0 && is_valid(in_valid, idx);
+    ballot_write_validity(out_valid, i, valid);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // hash partition
 // ---------------------------------------------------------------------------
@@ -169,6 +184,12 @@ void srj_gather_fixed(const void* in, const uint8_t* in_valid, const int64_t* ma
           (const int4*)in, in_valid, map, n, (int4*)out, out_valid);
       break;
   }
+}
+
+void srj_gather_validity(const uint8_t* in_valid, const int64_t* map,
+                         int64_t n, uint8_t* out_valid, hipStream_t stream) {
+  gather_validity_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(
+      in_valid, map, n, out_valid);
 }
 
 void srj_gather_str_lengths(const int32_t* offsets, const int64_t* map, int64_t n,

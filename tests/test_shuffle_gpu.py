@@ -105,3 +105,131 @@ def test_device_roundtrip_with_empty_partitions():
     permh = perm.cpu().tolist()
     assert got.columns[0].to_pylist() == [ints[i] for i in permh]
     assert got.columns[2].to_pylist() == [strs[i] for i in permh]
+
+
+def _mk_nested(n, device):
+    """table: [LIST<INT64> (nullable), STRUCT<INT32, STRING>, LIST<STRING>]"""
+    child_vals, offs = [], [0]
+    for i in range(n):
+        ln = random.randint(0, 4)
+        child_vals.extend(random.randint(0, 99) for _ in range(ln))
+        offs.append(offs[-1] + ln)
+    from spark_rapids_jni_amd.columnar import validity_from_bools
+    lvalid = [i % 11 != 7 for i in range(n)]
+    lst = Column(DType.LIST, n, None,
+                 validity_from_bools(lvalid, device),
+                 torch.tensor(offs, dtype=torch.int32, device=device),
+                 [Column.from_pylist(child_vals, DType.INT64, device)],
+                 null_count=None)
+    a = Column.from_pylist([None if i % 6 == 0 else i for i in range(n)],
+                           DType.INT32, device)
+    b = Column.from_pylist([f"v{i}" if i % 3 else None for i in range(n)],
+                           DType.STRING, device)
+    st = Column(DType.STRUCT, n, None, None, None, [a, b])
+    soffs, svals = [0], []
+    for i in range(n):
+        ln = random.randint(0, 3)
+        svals.extend(f"s{i}-{k}" for k in range(ln))
+        soffs.append(soffs[-1] + ln)
+    lstr = Column(DType.LIST, n, None, None,
+                  torch.tensor(soffs, dtype=torch.int32, device=device),
+                  [Column.from_pylist(svals, DType.STRING, device)])
+    return Table([lst, st, lstr])
+
+
+@pytest.mark.gpu
+def test_nested_gather():
+    from spark_rapids_jni_amd.ops.copying import gather
+    n = 200
+    tbl = _mk_nested(n, "cuda")
+    exp = [c.to_pylist() for c in tbl.columns]
+    gmap = torch.tensor([random.randrange(n) for _ in range(n * 2)],
+                        dtype=torch.int64, device="cuda")
+    G = gather(tbl, gmap)
+    gm = gmap.cpu().tolist()
+    for c, e in zip(G.columns, exp):
+        assert c.to_pylist() == [e[i] for i in gm]
+
+
+@pytest.mark.gpu
+def test_nested_device_split_host_merge():
+    from spark_rapids_jni_amd import shuffle_gpu
+    n, nparts = 500, 4
+    tbl = _mk_nested(n, "cuda")
+    exp = [c.to_pylist() for c in tbl.columns]
+    pids_h, offsets, perm = _partition(n, nparts, "cuda")
+    buf, sizes = shuffle_gpu.split_and_serialize_to_device(tbl, offsets, perm)
+    raw = buf.cpu().numpy().tobytes()
+    pieces, pos = [], 0
+    for s in sizes:
+        pieces.append(raw[pos:pos + s])
+        pos += s
+    schema = [c.to("cpu") for c in tbl.columns]
+    merged = kudo.merge_on_host(pieces, schema)
+    permh = perm.cpu().tolist()
+    for mc, e in zip(merged, exp):
+        assert mc.to_pylist() == [e[i] for i in permh]
+
+
+@pytest.mark.gpu
+def test_nested_host_write_device_assemble():
+    from spark_rapids_jni_amd import shuffle_gpu
+    n = 300
+    tbl_cpu = _mk_nested(n, "cpu")
+    exp = [c.to_pylist() for c in tbl_cpu.columns]
+    slices = [(0, 100), (100, 50), (150, 0), (150, 150)]
+    bufs = []
+    for off, cnt in slices:
+        out = io.BytesIO()
+        kudo.write_partition(tbl_cpu.columns, off, cnt, out)
+        bufs.append(torch.frombuffer(bytearray(out.getvalue()),
+                                     dtype=torch.uint8).to("cuda"))
+    got = shuffle_gpu.assemble_from_device(bufs, tbl_cpu.columns)
+    exp_rows = [i for off, cnt in slices for i in range(off, off + cnt)]
+    for gc, e in zip(got.columns, exp):
+        assert gc.to_pylist() == [e[i] for i in exp_rows]
+
+
+@pytest.mark.gpu
+def test_nested_device_roundtrip():
+    from spark_rapids_jni_amd import shuffle_gpu
+    n, nparts = 400, 3
+    tbl = _mk_nested(n, "cuda")
+    exp = [c.to_pylist() for c in tbl.columns]
+    pids_h, offsets, perm = _partition(n, nparts, "cuda")
+    buf, sizes = shuffle_gpu.split_and_serialize_to_device(tbl, offsets, perm)
+    got = shuffle_gpu.assemble_from_device_raw(buf, sizes,
+                                               [c.to("cpu") for c in
+                                                tbl.columns])
+    permh = perm.cpu().tolist()
+    for gc, e in zip(got.columns, exp):
+        assert gc.to_pylist() == [e[i] for i in permh]
+
+
+@pytest.mark.gpu
+def test_deep_nested_device_roundtrip():
+    """LIST<LIST<INT64>> exercises the level-chained offset resolution."""
+    from spark_rapids_jni_amd import shuffle_gpu
+    n, nparts = 150, 3
+    inner_vals, ioffs = [], [0]
+    for _ in range(400):
+        ln = random.randint(0, 3)
+        inner_vals.extend(random.randint(0, 999) for _ in range(ln))
+        ioffs.append(ioffs[-1] + ln)
+    inner = Column(DType.LIST, 400, None, None,
+                   torch.tensor(ioffs, dtype=torch.int32, device="cuda"),
+                   [Column.from_pylist(inner_vals, DType.INT64, "cuda")])
+    ooffs = [0]
+    for i in range(n):
+        ooffs.append(min(ooffs[-1] + random.randint(0, 5), 400))
+    outer = Column(DType.LIST, n, None, None,
+                   torch.tensor(ooffs, dtype=torch.int32, device="cuda"),
+                   [inner])
+    tbl = Table([outer])
+    exp = outer.to_pylist()
+    pids_h, offsets, perm = _partition(n, nparts, "cuda")
+    buf, sizes = shuffle_gpu.split_and_serialize_to_device(tbl, offsets, perm)
+    got = shuffle_gpu.assemble_from_device_raw(
+        buf, sizes, [outer.to("cpu")])
+    permh = perm.cpu().tolist()
+    assert got.columns[0].to_pylist() == [exp[i] for i in permh]
