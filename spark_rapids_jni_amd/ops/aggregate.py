@@ -52,9 +52,24 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     stream = _native.current_stream()
     dev = kcols[0].device
 
+    # specialized path: single non-null int64-valued key — 16B {key,row}
+    # slots claimed by CAS-on-key, software-pipelined probing
+    # (src/gpu/hashtable_i64.hip groupby_i64_kernel)
+    i64_fast = (len(kcols) == 1 and kcols[0].validity is None
+                and kcols[0].data is not None
+                and kcols[0].data.dtype == torch.int64
+                and kcols[0].dtype not in (DType.FLOAT64,))
     cap_groups = num_groups_hint if num_groups_hint else n
     capacity = max(_next_pow2(min(cap_groups, n) * 2), 64)
-    slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
+    if i64_fast:
+        # interleaved {key, row1} pairs; keys init to the EMPTY sentinel
+        # (INT64_MIN), +1 reserved slot for rows whose key equals it
+        slots = torch.zeros(2 * (capacity + 1), dtype=torch.int64, device=dev)
+        slots.view(-1, 2)[:, 0] = -2**63
+        nstates = capacity + 1
+    else:
+        slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
+        nstates = capacity
 
     # native agg list = user aggs + a hidden COUNT_VALID per nullable SUM col
     # (to derive the all-null-group -> null result in the SAME compaction pass,
@@ -66,36 +81,41 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
         native.append((native_op, col, st))
         return len(native) - 1
 
+    def _hidden_count(col, add_native):
+        # all-null-group detection needs a per-group valid count — but a
+        # column with no validity buffer can never produce an all-null
+        # group, so skip the extra atomic stream entirely (the bench shape:
+        # one atomicAdd per row saved)
+        if col.validity is None:
+            return None
+        stc = torch.zeros(nstates, dtype=torch.int64, device=dev)
+        return add_native(1, col, stc)
+
     for op, col in aggs:
         if op in (Agg.COUNT_ALL, Agg.COUNT_VALID):
-            st = torch.zeros(capacity, dtype=torch.int64, device=dev)
+            st = torch.zeros(nstates, dtype=torch.int64, device=dev)
             idx = add_native(int(op), col, st)
             metas.append((op, False, idx, None))
             continue
         assert col is not None
         is_float = col.dtype in (DType.FLOAT32, DType.FLOAT64)
         if op == Agg.SUM:
-            st = torch.zeros(capacity, dtype=torch.float64 if is_float
+            st = torch.zeros(nstates, dtype=torch.float64 if is_float
                              else torch.int64, device=dev)
             idx = add_native(3 if is_float else 2, col, st)
-            hidden = None
-            stc = torch.zeros(capacity, dtype=torch.int64, device=dev)
-            hidden = add_native(1, col, stc)
-            metas.append((op, is_float, idx, hidden))
+            metas.append((op, is_float, idx, _hidden_count(col, add_native)))
         elif op == Agg.MIN:
-            st = torch.full((capacity,), float("inf") if is_float else 2**63 - 1,
+            st = torch.full((nstates,), float("inf") if is_float else 2**63 - 1,
                             dtype=torch.float64 if is_float else torch.int64,
                             device=dev)
             idx = add_native(6 if is_float else 4, col, st)
-            stc = torch.zeros(capacity, dtype=torch.int64, device=dev)
-            metas.append((op, is_float, idx, add_native(1, col, stc)))
+            metas.append((op, is_float, idx, _hidden_count(col, add_native)))
         else:  # MAX
-            st = torch.full((capacity,), float("-inf") if is_float else -2**63,
+            st = torch.full((nstates,), float("-inf") if is_float else -2**63,
                             dtype=torch.float64 if is_float else torch.int64,
                             device=dev)
             idx = add_native(7 if is_float else 5, col, st)
-            stc = torch.zeros(capacity, dtype=torch.int64, device=dev)
-            metas.append((op, is_float, idx, add_native(1, col, stc)))
+            metas.append((op, is_float, idx, _hidden_count(col, add_native)))
 
     naggs = len(native)
     raw = bytearray(max(naggs, 1) * _AGGDESC_SZ)
@@ -109,21 +129,29 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
             st.data_ptr())
     agg_desc = torch.frombuffer(raw, dtype=torch.uint8).to(dev)
 
-    kdesc, ktop, keep = pack_descriptors(kcols)
-    g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n, slots.data_ptr(),
-              capacity, agg_desc.data_ptr(), naggs, stream)
-
     counter = torch.zeros(1, dtype=torch.int64, device=dev)
-    out_repr = torch.empty(capacity, dtype=torch.int64, device=dev)
-    out_agg = torch.empty(max(naggs, 1) * capacity, dtype=torch.int64, device=dev)
-    g.groupby_compact(slots.data_ptr(), capacity, agg_desc.data_ptr(), naggs,
-                      counter.data_ptr(), out_repr.data_ptr(), out_agg.data_ptr(),
-                      capacity, stream)
+    out_repr = torch.empty(nstates, dtype=torch.int64, device=dev)
+    out_agg = torch.empty(max(naggs, 1) * nstates, dtype=torch.int64, device=dev)
+    if i64_fast:
+        g.groupby_i64(kcols[0].data.data_ptr(), n, slots.data_ptr(), capacity,
+                      agg_desc.data_ptr(), naggs, stream)
+        g.groupby_compact_i64(slots.data_ptr(), capacity + 1,
+                              agg_desc.data_ptr(), naggs, counter.data_ptr(),
+                              out_repr.data_ptr(), out_agg.data_ptr(), nstates,
+                              stream)
+    else:
+        kdesc, ktop, keep = pack_descriptors(kcols)
+        g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n,
+                  slots.data_ptr(), capacity, agg_desc.data_ptr(), naggs,
+                  stream)
+        g.groupby_compact(slots.data_ptr(), capacity, agg_desc.data_ptr(),
+                          naggs, counter.data_ptr(), out_repr.data_ptr(),
+                          out_agg.data_ptr(), nstates, stream)
     ngroups = int(counter.item())
     out_repr = out_repr[:ngroups]
 
     def agg_vals(idx):
-        return out_agg[idx * capacity:idx * capacity + ngroups]
+        return out_agg[idx * nstates:idx * nstates + ngroups]
 
     key_out = Table([gather_column(c, out_repr) for c in kcols])
     results: List[Column] = []
@@ -132,6 +160,13 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
         if op in (Agg.COUNT_ALL, Agg.COUNT_VALID):
             results.append(Column(DType.INT64, ngroups, vals))
         else:  # SUM / MIN / MAX: null result for all-null groups
+            if hidden is None:  # value column had no nulls
+                if is_float:
+                    results.append(Column(DType.FLOAT64, ngroups,
+                                          vals.view(torch.float64).clone()))
+                else:
+                    results.append(Column(DType.INT64, ngroups, vals))
+                continue
             cnt = agg_vals(hidden)
             validity = _validity_from_bool(cnt > 0)
             if is_float:

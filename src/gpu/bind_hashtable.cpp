@@ -19,6 +19,11 @@ void srj_groupby_compact(const uint64_t*, int64_t, const void*, int32_t, uint64_
                          int64_t*, int64_t*, int64_t, hipStream_t);
 void srj_join_build_i64(const long long*, const uint8_t*, int64_t, void*, int64_t,
                         hipStream_t);
+void srj_groupby_i64(const long long*, int64_t, void*, int64_t, const void*,
+                     int32_t, hipStream_t);
+void srj_groupby_compact_i64(const void*, int64_t, const void*, int32_t,
+                             uint64_t*, int64_t*, int64_t*, int64_t,
+                             hipStream_t);
 void srj_join_probe_i64(const long long*, const uint8_t*, int64_t, const void*,
                         int64_t, uint64_t*, int32_t*, int64_t*, int64_t, uint8_t*,
                         int32_t, hipStream_t);
@@ -73,6 +78,26 @@ void register_hashtable(py::module_& m) {
           srj_join_build_i64(as_ptr<long long>(keys), as_ptr<uint8_t>(valid), nrows,
                              as_ptr<void>(slots), capacity, as_stream(stream));
           check_hip("join_build_i64");
+        });
+  m.def("groupby_i64",
+        [](uintptr_t keys, int64_t nrows, uintptr_t slots, int64_t capacity,
+           uintptr_t aggs, int32_t naggs, uintptr_t stream) {
+          srj_groupby_i64(as_ptr<long long>(keys), nrows, as_ptr<void>(slots),
+                          capacity, as_ptr<void>(aggs), naggs,
+                          as_stream(stream));
+          check_hip("groupby_i64");
+        });
+  m.def("groupby_compact_i64",
+        [](uintptr_t slots, int64_t capacity1, uintptr_t aggs, int32_t naggs,
+           uintptr_t counter, uintptr_t out_repr, uintptr_t out_agg,
+           int64_t out_capacity, uintptr_t stream) {
+          srj_groupby_compact_i64(as_ptr<void>(slots), capacity1,
+                                  as_ptr<void>(aggs), naggs,
+                                  as_ptr<uint64_t>(counter),
+                                  as_ptr<int64_t>(out_repr),
+                                  as_ptr<int64_t>(out_agg), out_capacity,
+                                  as_stream(stream));
+          check_hip("groupby_compact_i64");
         });
   m.def("join_probe_i64",
         [](uintptr_t probe, uintptr_t pvalid, int64_t nprobe, uintptr_t slots,

@@ -212,3 +212,52 @@ def test_i64_fast_path_matches_generic():
     fb, fp = fast.inner_join(p)
     sb, sp = slow.inner_join(p)
     assert _pairs(fb, fp) == _pairs(sb, sp)
+
+
+@pytest.mark.gpu
+def test_groupby_i64_fast_path():
+    """Single non-null int64 key takes the specialized Slot64 kernel; must
+    agree with a host oracle, including the INT64_MIN sentinel key."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 20000
+    keys = [random.choice([-2**63, -2**63 + 1, 0, 7, 2**62, -5])
+            if random.random() < 0.3 else random.randint(-10**9, 10**9)
+            for _ in range(n)]
+    vals = [random.randint(-1000, 1000) for _ in range(n)]
+    kc = Column.from_pylist(keys, DType.INT64, "cuda")
+    vc = Column.from_pylist(vals, DType.INT64, "cuda")
+    assert kc.validity is None  # must select the fast path
+    kt, res = groupby(kc, [(Agg.COUNT_ALL, None), (Agg.SUM, vc),
+                           (Agg.MIN, vc), (Agg.MAX, vc)])
+    got = {k: (c, s, mn, mx) for k, c, s, mn, mx in zip(
+        kt.columns[0].to_pylist(), res[0].to_pylist(), res[1].to_pylist(),
+        res[2].to_pylist(), res[3].to_pylist())}
+    exp = {}
+    for k, v in zip(keys, vals):
+        if k not in exp:
+            exp[k] = [0, 0, v, v]
+        e = exp[k]
+        e[0] += 1
+        e[1] += v
+        e[2] = min(e[2], v)
+        e[3] = max(e[3], v)
+    assert len(got) == len(exp)
+    for k, e in exp.items():
+        assert got[k] == tuple(e), f"key {k}"
+
+
+@pytest.mark.gpu
+def test_groupby_i64_fast_path_large_sparse():
+    """Mostly-unique keys stress the insert/CAS path of the Slot64 table."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 100_000
+    keys = torch.randint(-2**62, 2**62, (n,), dtype=torch.int64,
+                         device="cuda")
+    kc = Column(DType.INT64, n, keys)
+    kt, res = groupby(kc, [(Agg.COUNT_ALL, None)])
+    total = int(res[0].data.sum().item())
+    assert total == n
+    import collections
+    cnt = collections.Counter(keys.cpu().tolist())
+    got = dict(zip(kt.columns[0].to_pylist(), res[0].to_pylist()))
+    assert got == dict(cnt)
