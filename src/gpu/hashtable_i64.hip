@@ -118,24 +118,35 @@ __global__ void join_probe_i64_kernel(
     }
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) first[b] = slots[s[b]];
-    // resolve: count matches (and for !FILL that's all)
+    // resolve: count matches; carry the FIRST match's build row in a
+    // register so the (overwhelmingly common) exactly-one-match case emits
+    // without re-walking the slot chain
     uint32_t nm = 0;
+    uint32_t nmb[PIPE];
+    long long hit1[PIPE];
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) {
+      nmb[b] = 0;
+      hit1[b] = 0;
       if (!((vbits >> b) & 1)) continue;
       Slot64 cur = first[b];
       uint64_t sl = s[b];
       while (cur.row1 != 0) {
-        nm += (cur.key == k[b]);
+        if (cur.key == k[b]) {
+          if (nmb[b] == 0) hit1[b] = cur.row1;
+          ++nmb[b];
+        }
         sl = (sl + 1) & mask;
         cur = slots[sl];
       }
+      nm += nmb[b];
     }
     if (!FILL) {
       count_local += nm;
       continue;
     }
-    // emit: one atomic per wave, then cache-warm rescan writes the pairs
+    // emit: one atomic per wave; single-match rows write straight from the
+    // carried register, only multi-match rows rescan (cache-warm)
     uint32_t incl = wave_prefix_incl(nm);
     uint32_t total = __shfl(incl, WAVE - 1, WAVE);
     uint64_t wave_base = 0;
@@ -146,7 +157,16 @@ __global__ void join_probe_i64_kernel(
     if (nm) {
 #pragma unroll
       for (int b = 0; b < PIPE; ++b) {
-        if (!((vbits >> b) & 1)) continue;
+        if (nmb[b] == 0) continue;
+        if (nmb[b] == 1) {
+          if (pos < out_capacity) {
+            out_build[pos] = (int32_t)(hit1[b] - 1);
+            out_probe[pos] = base + b;
+            if (build_matched) build_matched[hit1[b] - 1] = 1;
+          }
+          ++pos;
+          continue;
+        }
         Slot64 cur = first[b];
         uint64_t sl = s[b];
         while (cur.row1 != 0) {
