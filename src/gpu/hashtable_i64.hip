@@ -44,23 +44,36 @@ __global__ void join_build_i64_kernel(const long long* __restrict__ keys,
     long long k[PIPE];
     uint64_t s[PIPE];
     bool act[PIPE];
+    long long prev[PIPE];
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) {
       int64_t row = base + b;
       act[b] = row < nrows && is_valid(valid, row);
-      k[b] = act[b] ? keys[row] : 0;
-      s[b] = i64_hash(k[b]) & mask;
+      k[b] = act[b] ? keys[row < nrows ? row : 0] : 0;
+      s[b] = act[b] ? (i64_hash(k[b]) & mask) : 0;
+    }
+    // first-slot claims issued back-to-back (independent atomics pipeline
+    // with counted vmcnt); inactive lanes CAS 0->0 on slot 0, a no-op
+#pragma unroll
+    for (int b = 0; b < PIPE; ++b) {
+      prev[b] = atomicCAS(
+          reinterpret_cast<unsigned long long*>(&slots[s[b]].row1), 0ull,
+          (unsigned long long)(act[b] ? base + b + 1 : 0));
     }
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) {
       if (!act[b]) continue;
+      if (prev[b] == 0) {
+        slots[s[b]].key = k[b];
+        continue;
+      }
       int64_t row = base + b;
-      uint64_t sl = s[b];
+      uint64_t sl = (s[b] + 1) & mask;
       while (true) {
-        long long prev = atomicCAS(
+        long long p = atomicCAS(
             reinterpret_cast<unsigned long long*>(&slots[sl].row1), 0ull,
             (unsigned long long)(row + 1));
-        if (prev == 0) {
+        if (p == 0) {
           slots[sl].key = k[b];
           break;
         }
