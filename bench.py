@@ -89,14 +89,38 @@ def make_build_shard(n_build: int, world: int, rank: int, device):
     return shuffle_exchange(local, world)
 
 
-def one_step(build_col, chunks, world, out_hint):
+def one_step(build_col, chunks, world, out_hint, comm_stream=None):
     from spark_rapids_jni_amd.columnar import Column
     from spark_rapids_jni_amd.ops.join import HashJoinTable
 
     # rebuild the table each step (config is build+probe)
     tbl = HashJoinTable.build(build_col)
-    for chunk in chunks:
-        probe = shuffle_exchange(chunk, world) if world > 1 else chunk
+    if world == 1:
+        for chunk in chunks:
+            pcol = Column.from_torch(chunk)
+            bi, pi = tbl.inner_join(pcol, out_hint=chunk.numel() + out_hint)
+            del pcol, bi, pi
+        del tbl
+        return
+
+    # world > 1: run the shuffle of chunk i+1 on a side stream while chunk i
+    # probes on the default stream (xGMI all-to-all overlaps with compute)
+    main = torch.cuda.current_stream()
+
+    def start_exchange(chunk):
+        with torch.cuda.stream(comm_stream):
+            probe = shuffle_exchange(chunk, world)
+            ev = torch.cuda.Event()
+            ev.record(comm_stream)
+        return probe, ev
+
+    nxt = start_exchange(chunks[0])
+    for i in range(len(chunks)):
+        probe, ev = nxt
+        if i + 1 < len(chunks):
+            nxt = start_exchange(chunks[i + 1])
+        main.wait_event(ev)
+        probe.record_stream(main)
         pcol = Column.from_torch(probe)
         bi, pi = tbl.inner_join(pcol, out_hint=probe.numel() + out_hint)
         del probe, pcol, bi, pi
@@ -132,7 +156,7 @@ def main():
     free, total = torch.cuda.mem_get_info(device)
     need = (args.build_rows * 84              # build shard + 25%-load slot table
             + args.probe_rows * 8             # resident probe chunks
-            + args.chunk_rows * (32 if args.gpus > 1 else 0)  # partition temps
+            + args.chunk_rows * (48 if args.gpus > 1 else 0)  # 2 in-flight shuffles
             + args.chunk_rows * 13            # join output maps
             ) * 1.15
     if need > free:
@@ -159,15 +183,16 @@ def main():
                                     device=device))
         done += m
 
+    comm_stream = torch.cuda.Stream(device=device) if world > 1 else None
     log(rank, f"[bench] warmup {args.warmup} steps")
     for _ in range(args.warmup):
-        one_step(build_col, chunks, world, out_hint)
+        one_step(build_col, chunks, world, out_hint, comm_stream)
     barrier_sync(world)
 
     log(rank, f"[bench] timing {args.steps} steps")
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        one_step(build_col, chunks, world, out_hint)
+        one_step(build_col, chunks, world, out_hint, comm_stream)
     barrier_sync(world)
     elapsed = time.perf_counter() - t0
 
