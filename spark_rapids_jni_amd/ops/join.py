@@ -100,13 +100,22 @@ class HashJoinTable:
         counter = torch.zeros(1, dtype=torch.int64, device=dev)
         if not fast:
             bdesc, btop, pdesc, ptop, keep = self._descs(pcols)
+        # NOTE measured negative: radix-partitioning the probe keys by hash
+        # prefix (part_hist/part_scatter + idxmap probe) did NOT speed up the
+        # 10B x 1B config — each 64B table line is touched only ~1.4x per 1B
+        # chunk, so LLC locality saves almost no DRAM traffic while the
+        # scatter adds 29 ms/chunk (profiles/r01_bench_join_22.1B.json).
+        # The kernels stay exposed (g.part_hist/part_scatter, idxmap probe)
+        # for schema-partitioned shuffle use; the probe here stays direct.
+        probe_keys_ptr = pcols[0].data.data_ptr() if fast else 0
+        idxmap_ptr = 0
         if out_hint is None:
             if fast:
                 p = pcols[0]
-                g.join_probe_i64(p.data.data_ptr(),
+                g.join_probe_i64(probe_keys_ptr,
                                  p.validity.data_ptr() if p.validity is not None else 0,
                                  nprobe, self.slots.data_ptr(), self.capacity,
-                                 counter.data_ptr(), 0, 0, 0, 0, 0, stream)
+                                 counter.data_ptr(), 0, 0, 0, 0, 0, 0, stream)
             else:
                 g.join_probe_count(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
                                    ptop.data_ptr(), len(pcols), nprobe,
@@ -122,13 +131,13 @@ class HashJoinTable:
                    if track_build_matches else None)
         if fast:
             p = pcols[0]
-            g.join_probe_i64(p.data.data_ptr(),
+            g.join_probe_i64(probe_keys_ptr,
                              p.validity.data_ptr() if p.validity is not None else 0,
                              nprobe, self.slots.data_ptr(), self.capacity,
                              counter.data_ptr(), out_build.data_ptr(),
                              out_probe.data_ptr(), total,
                              matched.data_ptr() if matched is not None else 0,
-                             1, stream)
+                             1, idxmap_ptr, stream)
         else:
             g.join_probe_fill(bdesc.data_ptr(), btop.data_ptr(), pdesc.data_ptr(),
                               ptop.data_ptr(), len(pcols), nprobe,

@@ -24,9 +24,12 @@ void srj_groupby_i64(const long long*, int64_t, void*, int64_t, const void*,
 void srj_groupby_compact_i64(const void*, int64_t, const void*, int32_t,
                              uint64_t*, int64_t*, int64_t*, int64_t,
                              hipStream_t);
+void srj_part_hist(const long long*, int64_t, int32_t, int64_t*, hipStream_t);
+void srj_part_scatter(const long long*, int64_t, int32_t, int64_t*, long long*,
+                      int32_t*, hipStream_t);
 void srj_join_probe_i64(const long long*, const uint8_t*, int64_t, const void*,
                         int64_t, uint64_t*, int32_t*, int64_t*, int64_t, uint8_t*,
-                        int32_t, hipStream_t);
+                        int32_t, const int32_t*, hipStream_t);
 }
 
 void register_hashtable(py::module_& m) {
@@ -103,13 +106,29 @@ void register_hashtable(py::module_& m) {
         [](uintptr_t probe, uintptr_t pvalid, int64_t nprobe, uintptr_t slots,
            int64_t capacity, uintptr_t counter, uintptr_t out_build,
            uintptr_t out_probe, int64_t out_capacity, uintptr_t build_matched,
-           int32_t fill, uintptr_t stream) {
+           int32_t fill, uintptr_t idxmap, uintptr_t stream) {
           srj_join_probe_i64(as_ptr<long long>(probe), as_ptr<uint8_t>(pvalid),
                              nprobe, as_ptr<void>(slots), capacity,
                              as_ptr<uint64_t>(counter), as_ptr<int32_t>(out_build),
                              as_ptr<int64_t>(out_probe), out_capacity,
-                             as_ptr<uint8_t>(build_matched), fill, as_stream(stream));
+                             as_ptr<uint8_t>(build_matched), fill,
+                             as_ptr<int32_t>(idxmap), as_stream(stream));
           check_hip("join_probe_i64");
+        });
+  m.def("part_hist",
+        [](uintptr_t keys, int64_t n, int32_t pbits, uintptr_t hist,
+           uintptr_t stream) {
+          srj_part_hist(as_ptr<long long>(keys), n, pbits,
+                        as_ptr<int64_t>(hist), as_stream(stream));
+          check_hip("part_hist");
+        });
+  m.def("part_scatter",
+        [](uintptr_t keys, int64_t n, int32_t pbits, uintptr_t cursors,
+           uintptr_t out_keys, uintptr_t out_idx, uintptr_t stream) {
+          srj_part_scatter(as_ptr<long long>(keys), n, pbits,
+                           as_ptr<int64_t>(cursors), as_ptr<long long>(out_keys),
+                           as_ptr<int32_t>(out_idx), as_stream(stream));
+          check_hip("part_scatter");
         });
   m.def("groupby",
         [](uintptr_t cols, uintptr_t top, int32_t ntop, int64_t nrows,

@@ -37,6 +37,13 @@ constexpr int PIPE = 8;
 
 __device__ inline uint64_t i64_hash(long long k) { return mix64((uint64_t)k); }
 
+// Slot from the TOP hash bits: the table is then ordered by hash prefix, so
+// keys radix-partitioned by the same prefix probe a contiguous table window
+// (LLC-sized at 512 partitions) instead of the whole table. mask = 2^k - 1.
+__device__ inline uint64_t slot_of(uint64_t h, uint64_t mask) {
+  return h >> __builtin_clzll(mask);
+}
+
 // ---------------------------------------------------------------------------
 // build
 // ---------------------------------------------------------------------------
@@ -57,7 +64,7 @@ __global__ void join_build_i64_kernel(const long long* __restrict__ keys,
       int64_t row = base + b;
       act[b] = row < nrows && is_valid(valid, row);
       k[b] = act[b] ? keys[row < nrows ? row : 0] : 0;
-      s[b] = act[b] ? (i64_hash(k[b]) & mask) : 0;
+      s[b] = act[b] ? slot_of(i64_hash(k[b]), mask) : 0;
     }
     // first-slot claims issued back-to-back (independent atomics pipeline
     // with counted vmcnt); inactive lanes CAS 0->0 on slot 0, a no-op
@@ -113,7 +120,8 @@ __global__ void join_probe_i64_kernel(
     int64_t nprobe, const Slot64* __restrict__ slots, uint64_t mask,
     uint64_t* __restrict__ counter, int32_t* __restrict__ out_build,
     int64_t* __restrict__ out_probe, int64_t out_capacity,
-    uint8_t* __restrict__ build_matched) {
+    uint8_t* __restrict__ build_matched,
+    const int32_t* __restrict__ idxmap /* source rows of partitioned keys */) {
   static_assert(PPIPE == 8 || PPIPE == 16, "validity bytes per batch");
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
@@ -142,7 +150,7 @@ __global__ void join_probe_i64_kernel(
     for (int b = 0; b < PPIPE; ++b) {
       int64_t row = base + b;
       k[b] = probe[row < nprobe ? row : 0];  // clamped, unconditional
-      s[b] = i64_hash(k[b]) & mask;
+      s[b] = slot_of(i64_hash(k[b]), mask);
     }
 #pragma unroll
     for (int b = 0; b < PPIPE; ++b) first[b] = slots[s[b]];
@@ -190,7 +198,7 @@ __global__ void join_probe_i64_kernel(
         if (nmb[b] == 1) {
           if (pos < out_capacity) {
             out_build[pos] = (int32_t)(hit1[b] - 1);
-            out_probe[pos] = base + b;
+            out_probe[pos] = idxmap ? (int64_t)idxmap[base + b] : base + b;
             if (build_matched) build_matched[hit1[b] - 1] = 1;
           }
           ++pos;
@@ -203,7 +211,7 @@ __global__ void join_probe_i64_kernel(
             long long r1 = cur.row1 & SLOT_ROW;
             if (pos < out_capacity) {
               out_build[pos] = (int32_t)(r1 - 1);
-              out_probe[pos] = base + b;
+              out_probe[pos] = idxmap ? (int64_t)idxmap[base + b] : base + b;
               if (build_matched) build_matched[r1 - 1] = 1;
             }
             ++pos;
@@ -254,7 +262,7 @@ __global__ void groupby_i64_kernel(const long long* __restrict__ keys,
     for (int b = 0; b < PIPE; ++b) {
       int64_t row = base + b;
       k[b] = keys[row < nrows ? row : 0];  // clamped, unconditional
-      s[b] = i64_hash(k[b]) & mask;
+      s[b] = slot_of(i64_hash(k[b]), mask);
     }
 #pragma unroll
     for (int b = 0; b < PIPE; ++b) first[b] = slots[s[b]];
@@ -352,7 +360,7 @@ void srj_join_probe_i64(const long long* probe, const uint8_t* pvalid,
                         int64_t nprobe, const void* slots, int64_t capacity,
                         uint64_t* counter, int32_t* out_build, int64_t* out_probe,
                         int64_t out_capacity, uint8_t* build_matched, int32_t fill,
-                        hipStream_t stream) {
+                        const int32_t* idxmap, hipStream_t stream) {
   int64_t nthreads_needed = (nprobe + PROBE_PIPE - 1) / PROBE_PIPE;
   int64_t g = grid_1d(nthreads_needed);
   const Slot64* sl = reinterpret_cast<const Slot64*>(slots);
@@ -361,18 +369,20 @@ void srj_join_probe_i64(const long long* probe, const uint8_t* pvalid,
     if (pvalid)
       join_probe_i64_kernel<true, true, PROBE_PIPE><<<g, DEFAULT_BLOCK, 0, stream>>>(
           probe, pvalid, nprobe, sl, mask, counter, out_build, out_probe,
-          out_capacity, build_matched);
+          out_capacity, build_matched, idxmap);
     else
       join_probe_i64_kernel<true, false, PROBE_PIPE><<<g, DEFAULT_BLOCK, 0, stream>>>(
           probe, pvalid, nprobe, sl, mask, counter, out_build, out_probe,
-          out_capacity, build_matched);
+          out_capacity, build_matched, idxmap);
   } else {
     if (pvalid)
       join_probe_i64_kernel<false, true, PROBE_PIPE><<<g, DEFAULT_BLOCK, 0, stream>>>(
-          probe, pvalid, nprobe, sl, mask, counter, nullptr, nullptr, 0, nullptr);
+          probe, pvalid, nprobe, sl, mask, counter, nullptr, nullptr, 0,
+          nullptr, nullptr);
     else
       join_probe_i64_kernel<false, false, PROBE_PIPE><<<g, DEFAULT_BLOCK, 0, stream>>>(
-          probe, pvalid, nprobe, sl, mask, counter, nullptr, nullptr, 0, nullptr);
+          probe, pvalid, nprobe, sl, mask, counter, nullptr, nullptr, 0,
+          nullptr, nullptr);
   }
 }
 

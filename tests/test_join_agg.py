@@ -261,3 +261,46 @@ def test_groupby_i64_fast_path_large_sparse():
     cnt = collections.Counter(keys.cpu().tolist())
     got = dict(zip(kt.columns[0].to_pylist(), res[0].to_pylist()))
     assert got == dict(cnt)
+
+
+@pytest.mark.gpu
+def test_partitioned_probe_matches_direct():
+    """Force the radix-partitioned probe path (large capacity via a big build)
+    on a small validation: results must equal the direct probe as multisets."""
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    from spark_rapids_jni_amd import _native
+    g = _native.gpu()
+    n = 200_000
+    keys = torch.randint(0, n // 2, (n,), dtype=torch.int64, device="cuda")
+    bcol = Column(DType.INT64, n, keys)
+    tbl = HashJoinTable.build(bcol)
+    probe = torch.randint(0, n, (n,), dtype=torch.int64, device="cuda")
+    pcol = Column(DType.INT64, n, probe)
+    bi, pi = tbl.inner_join(pcol)
+    # partition manually and probe with idxmap — must match the direct result
+    pbits = 4
+    np_ = 1 << pbits
+    stream = _native.current_stream()
+    hist = torch.zeros(np_, dtype=torch.int64, device="cuda")
+    g.part_hist(probe.data_ptr(), n, pbits, hist.data_ptr(), stream)
+    assert int(hist.sum().item()) == n
+    cursors = torch.zeros(np_, dtype=torch.int64, device="cuda")
+    torch.cumsum(hist[:np_ - 1], 0, out=cursors[1:])
+    part_keys = torch.empty(n, dtype=torch.int64, device="cuda")
+    part_idx = torch.empty(n, dtype=torch.int32, device="cuda")
+    g.part_scatter(probe.data_ptr(), n, pbits, cursors.data_ptr(),
+                   part_keys.data_ptr(), part_idx.data_ptr(), stream)
+    # partitioned keys are a permutation of the input
+    assert torch.equal(part_keys.sort().values, probe.sort().values)
+    assert torch.equal(probe[part_idx.long()], part_keys)
+    counter = torch.zeros(1, dtype=torch.int64, device="cuda")
+    out_b = torch.empty(bi.numel(), dtype=torch.int32, device="cuda")
+    out_p = torch.empty(bi.numel(), dtype=torch.int64, device="cuda")
+    g.join_probe_i64(part_keys.data_ptr(), 0, n, tbl.slots.data_ptr(),
+                     tbl.capacity, counter.data_ptr(), out_b.data_ptr(),
+                     out_p.data_ptr(), bi.numel(), 0, 1, part_idx.data_ptr(),
+                     stream)
+    assert int(counter.item()) == bi.numel()
+    direct = sorted(zip(pi.cpu().tolist(), bi.cpu().tolist()))
+    parted = sorted(zip(out_p.cpu().tolist(), out_b.cpu().tolist()))
+    assert direct == parted
