@@ -23,6 +23,32 @@ struct RowColDesc {
   int32_t row_off;        // byte offset within row
 };
 
+// load unconditionally, store value-or-zero: keeps the column loads
+// branch-free so consecutive elements pipeline (counted vmcnt) instead of
+// stalling on a per-element validity branch
+__device__ inline void copy_elem_z(uint8_t* dst, const uint8_t* src, int w,
+                                   bool valid) {
+  switch (w) {
+    case 1: *dst = valid ? *src : 0; break;
+    case 2: *reinterpret_cast<uint16_t*>(dst) =
+                valid ? *reinterpret_cast<const uint16_t*>(src) : 0;
+            break;
+    case 4: *reinterpret_cast<uint32_t*>(dst) =
+                valid ? *reinterpret_cast<const uint32_t*>(src) : 0;
+            break;
+    case 8: *reinterpret_cast<uint64_t*>(dst) =
+                valid ? *reinterpret_cast<const uint64_t*>(src) : 0;
+            break;
+    case 16: {
+      uint64_t lo = *reinterpret_cast<const uint64_t*>(src);
+      uint64_t hi = *reinterpret_cast<const uint64_t*>(src + 8);
+      *reinterpret_cast<uint64_t*>(dst) = valid ? lo : 0;
+      *reinterpret_cast<uint64_t*>(dst + 8) = valid ? hi : 0;
+      break;
+    }
+  }
+}
+
 __device__ inline void copy_elem(uint8_t* dst, const uint8_t* src, int w) {
   switch (w) {
     case 1: *dst = *src; break;
@@ -94,6 +120,127 @@ __global__ void from_rows_kernel(const RowColDesc* __restrict__ cols,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// LDS-tiled fixed-width transpose (reference copy_to_rows/copy_from_rows
+// tile design, row_conversion.cu:591/912): a workgroup stages TILE_ROWS rows
+// in LDS, so the global side of both directions is fully coalesced — column
+// reads are 64-lane contiguous AND row writes leave as wide dword4 copies
+// instead of per-element scattered stores. Used when 64 rows of the layout
+// fit in LDS; the one-thread-per-row kernels above remain the wide-row
+// fallback.
+// ---------------------------------------------------------------------------
+constexpr int TILE_ROWS = 64;
+
+__global__ void to_rows_tiled_kernel(const RowColDesc* __restrict__ cols,
+                                     int32_t ncols, int64_t nrows,
+                                     int32_t row_size, int32_t validity_off,
+                                     uint8_t* __restrict__ out) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint8_t* tile = reinterpret_cast<uint8_t*>(smem);
+  int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
+  for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    int64_t row0 = t * TILE_ROWS;
+    int32_t m = (int32_t)(nrows - row0 < TILE_ROWS ? nrows - row0 : TILE_ROWS);
+    // zero the tile (padding + null slots are zero by contract)
+    for (int32_t i = threadIdx.x * 16; i < TILE_ROWS * row_size;
+         i += blockDim.x * 16)
+      *reinterpret_cast<uint4*>(tile + i) = uint4{0, 0, 0, 0};
+    __syncthreads();
+    // data: flat (column, tile-row) work — each wave covers one column's 64
+    // consecutive rows, so the global column loads are fully coalesced and
+    // the whole block stays busy
+    for (int32_t w = threadIdx.x; w < ncols * TILE_ROWS; w += blockDim.x) {
+      int32_t c = w >> 6;
+      int32_t r = w & (TILE_ROWS - 1);
+      if (r >= m) continue;
+      int64_t row = row0 + r;
+      const RowColDesc& d = cols[c];
+      copy_elem_z(tile + (int64_t)r * row_size + d.row_off,
+                  reinterpret_cast<const uint8_t*>(d.data) + row * d.width,
+                  d.width, is_valid(d.valid, row));
+    }
+    // validity bytes: one thread per tile row (no cross-thread byte RMW)
+    for (int32_t r = threadIdx.x; r < TILE_ROWS; r += blockDim.x) {
+      if (r >= m) continue;
+      int64_t row = row0 + r;
+      uint8_t* dst = tile + (int64_t)r * row_size;
+      uint8_t vbyte = 0;
+      for (int32_t c = 0; c < ncols; ++c) {
+        vbyte |= (uint8_t)is_valid(cols[c].valid, row) << (c & 7);
+        if ((c & 7) == 7) {
+          dst[validity_off + (c >> 3)] = vbyte;
+          vbyte = 0;
+        }
+      }
+      if (ncols & 7) dst[validity_off + (ncols >> 3)] = vbyte;
+    }
+    __syncthreads();
+    // coalesced wide copy LDS -> global
+    uint8_t* gdst = out + row0 * row_size;
+    int32_t nbytes = m * row_size;
+    int32_t i = threadIdx.x * 16;
+    for (; i + 16 <= nbytes; i += blockDim.x * 16)
+      *reinterpret_cast<uint4*>(gdst + i) = *reinterpret_cast<uint4*>(tile + i);
+    // tail (row_size is a multiple of 8, so 8B granularity suffices)
+    if (threadIdx.x == 0)
+      for (int32_t j = nbytes & ~15; j < nbytes; j += 8)
+        *reinterpret_cast<uint64_t*>(gdst + j) =
+            *reinterpret_cast<uint64_t*>(tile + j);
+    __syncthreads();
+  }
+}
+
+__global__ void from_rows_tiled_kernel(const RowColDesc* __restrict__ cols,
+                                       int32_t ncols, int64_t nrows,
+                                       int32_t row_size, int32_t validity_off,
+                                       const uint8_t* __restrict__ in) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  uint8_t* tile = reinterpret_cast<uint8_t*>(smem);
+  int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
+  for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    int64_t row0 = t * TILE_ROWS;
+    int32_t m = (int32_t)(nrows - row0 < TILE_ROWS ? nrows - row0 : TILE_ROWS);
+    const uint8_t* gsrc = in + row0 * row_size;
+    int32_t nbytes = m * row_size;
+    for (int32_t i = threadIdx.x * 16; i + 16 <= nbytes; i += blockDim.x * 16)
+      *reinterpret_cast<uint4*>(tile + i) =
+          *reinterpret_cast<const uint4*>(gsrc + i);
+    if (threadIdx.x == 0)
+      for (int32_t j = nbytes & ~15; j < nbytes; j += 8)
+        *reinterpret_cast<uint64_t*>(tile + j) =
+            *reinterpret_cast<const uint64_t*>(gsrc + j);
+    __syncthreads();
+    // data: flat (column, tile-row) work — coalesced global column writes
+    for (int32_t w = threadIdx.x; w < ncols * TILE_ROWS; w += blockDim.x) {
+      int32_t c = w >> 6;
+      int32_t r = w & (TILE_ROWS - 1);
+      if (r >= m) continue;
+      const RowColDesc& d = cols[c];
+      copy_elem(const_cast<uint8_t*>(
+                    reinterpret_cast<const uint8_t*>(d.data)) +
+                    (row0 + r) * d.width,
+                tile + (int64_t)r * row_size + d.row_off, d.width);
+    }
+    // validity: wave 0 only (TILE_ROWS == WAVE, rows map 1:1 onto lanes —
+    // the full wave participates in each ballot)
+    if (threadIdx.x < WAVE) {
+      int32_t r = threadIdx.x;
+      bool in_range = r < m;
+      const uint8_t* src = tile + (int64_t)(in_range ? r : 0) * row_size;
+      int64_t row = row0 + r;
+      for (int32_t c = 0; c < ncols; ++c) {
+        const RowColDesc& d = cols[c];
+        bool valid =
+            in_range && ((src[validity_off + (c >> 3)] >> (c & 7)) & 1);
+        if (d.valid != nullptr) {
+          ballot_write_validity(const_cast<uint8_t*>(d.valid), row, valid);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
 
 // ---------------------------------------------------------------------------
 // variable-width (strings) JCUDF path (reference copy_strings_to_rows
@@ -219,6 +366,15 @@ extern "C" {
 
 void srj_to_rows(const void* cols, int32_t ncols, int64_t nrows, int32_t row_size,
                  int32_t validity_off, uint8_t* out, hipStream_t stream) {
+  size_t lds = (size_t)TILE_ROWS * row_size;
+  if (lds <= 64 * 1024) {
+    int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
+    int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
+    to_rows_tiled_kernel<<<nblk, DEFAULT_BLOCK, lds, stream>>>(
+        reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
+        validity_off, out);
+    return;
+  }
   to_rows_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
       validity_off, out);
@@ -227,6 +383,15 @@ void srj_to_rows(const void* cols, int32_t ncols, int64_t nrows, int32_t row_siz
 void srj_from_rows(const void* cols, int32_t ncols, int64_t nrows,
                    int32_t row_size, int32_t validity_off, const uint8_t* in,
                    hipStream_t stream) {
+  size_t lds = (size_t)TILE_ROWS * row_size;
+  if (lds <= 64 * 1024) {
+    int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
+    int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
+    from_rows_tiled_kernel<<<nblk, DEFAULT_BLOCK, lds, stream>>>(
+        reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
+        validity_off, in);
+    return;
+  }
   from_rows_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
       validity_off, in);
