@@ -49,7 +49,8 @@ def main():
     from spark_rapids_jni_amd.ops import cast, hashing, misc
     from spark_rapids_jni_amd.ops.row_conversion import (convert_from_rows,
                                                          convert_to_rows)
-    from spark_rapids_jni_amd.ops.json import get_json_object
+    from spark_rapids_jni_amd.ops.json import (get_json_object,
+                                               get_json_object_multiple_paths)
     from spark_rapids_jni_amd.ops.sketch import parse_uri, UriPart
 
     g = torch.Generator(device=dev)
@@ -112,8 +113,10 @@ def main():
     paths = ["$.a", "$.b.c", "$.d[1]", "$.e"]
     for k in (1, 2, 4):
         def run_paths(k=k):
-            for p in paths[:k]:
-                get_json_object(jc, p)
+            if k == 1:
+                get_json_object(jc, paths[0])
+            else:
+                get_json_object_multiple_paths(jc, paths[:k])
         secs = timeit(run_paths, args.iters)
         emit(f"get_json_object_{k}path", jc.size * k, secs,
              int(jc.offsets[-1].item()) * k)

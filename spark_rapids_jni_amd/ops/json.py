@@ -88,9 +88,89 @@ def get_json_object(col: Column, path: str) -> Column:
                   null_count=None)
 
 
+_MULTIOUT_FMT = "<QQQQQ"  # lens, offsets, chars, out_valid, overflow
+
+
 def get_json_object_multiple_paths(col: Column, paths: List[str]) -> List[Column]:
-    """reference JSONUtils multi-path variant (shared input scan)."""
-    return [get_json_object(col, p) for p in paths]
+    """reference JSONUtils.getJsonObjectMultiplePaths (get_json_object.cu
+    multi-path kernel): the document is tokenized ONCE per row and a bitmask
+    of still-viable paths rides the single walk — N paths cost roughly one
+    scan instead of N. Falls back per path when a row exceeds the multi-path
+    match cap (>4 wildcard matches)."""
+    if len(paths) == 0:
+        return []
+    if len(paths) == 1:
+        return [get_json_object(col, paths[0])]
+    out: List[Optional[Column]] = [None] * len(paths)
+    for c0 in range(0, len(paths), 8):
+        chunk = paths[c0:c0 + 8]
+        cols = _multi_path_chunk(col, chunk)
+        for j, oc in enumerate(cols):
+            out[c0 + j] = oc
+    return out  # type: ignore[return-value]
+
+
+def _multi_path_chunk(col: Column, paths: List[str]) -> List[Optional[Column]]:
+    g = _native.gpu()
+    stream = _native.current_stream()
+    n = col.size
+    dev = col.device
+    npaths = len(paths)
+    all_instrs = []
+    path_off = []
+    path_len = []
+    for p in paths:
+        ins = compile_path(p)
+        path_off.append(len(all_instrs))
+        path_len.append(len(ins))
+        all_instrs.extend(ins)
+    it, kt = _pack_instrs(all_instrs, dev)
+    poff = torch.tensor(path_off, dtype=torch.int32).to(dev)
+    plen = torch.tensor(path_len, dtype=torch.int32).to(dev)
+    desc, top, keep = pack_descriptors([col])
+
+    lens = [torch.empty(n, dtype=torch.int32, device=dev) for _ in paths]
+    overflow = torch.zeros(npaths, dtype=torch.int32, device=dev)
+
+    def pack_outs(offsets=None, chars=None, valid=None):
+        raw = bytearray(npaths * struct.calcsize(_MULTIOUT_FMT))
+        for i in range(npaths):
+            struct.pack_into(
+                _MULTIOUT_FMT, raw, i * struct.calcsize(_MULTIOUT_FMT),
+                lens[i].data_ptr(),
+                offsets[i].data_ptr() if offsets else 0,
+                chars[i].data_ptr() if chars else 0,
+                valid[i].data_ptr() if valid else 0,
+                overflow.data_ptr() + i * 4)
+        return torch.frombuffer(raw, dtype=torch.uint8).to(dev)
+
+    outs0 = pack_outs()
+    g.get_json_multi(desc.data_ptr(), n, it.data_ptr(), kt.data_ptr(),
+                     poff.data_ptr(), plen.data_ptr(), npaths, 0,
+                     outs0.data_ptr(), stream)
+    offsets = []
+    for i in range(npaths):
+        o = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+        torch.cumsum(lens[i], 0, out=o[1:].view(n))
+        offsets.append(o)
+    totals = torch.stack([o[-1] for o in offsets]).cpu().tolist()  # one D2H
+    chars = [torch.empty(max(t, 1), dtype=torch.uint8, device=dev)
+             for t in totals]
+    valid = [make_validity(n, dev) for _ in range(npaths)]
+    outs1 = pack_outs(offsets, chars, valid)
+    g.get_json_multi(desc.data_ptr(), n, it.data_ptr(), kt.data_ptr(),
+                     poff.data_ptr(), plen.data_ptr(), npaths, 1,
+                     outs1.data_ptr(), stream)
+    ovf = overflow.cpu().tolist()
+    res: List[Optional[Column]] = []
+    for i in range(npaths):
+        if ovf[i]:
+            # some row had >4 matches: redo this path on the exact kernel
+            res.append(get_json_object(col, paths[i]))
+        else:
+            res.append(Column(DType.STRING, n, chars[i][:totals[i]], valid[i],
+                              offsets[i], null_count=None))
+    return res
 
 
 def from_json_to_raw_map(col: Column) -> Column:

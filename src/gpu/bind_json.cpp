@@ -5,6 +5,9 @@ extern "C" {
 void srj_get_json_object(const void*, int64_t, const void*, const char*, int32_t,
                          int32_t, int32_t*, const int32_t*, char*, uint8_t*,
                          hipStream_t);
+void srj_get_json_multi(const void*, int64_t, const void*, const char*,
+                        const int32_t*, const int32_t*, int32_t, int32_t,
+                        const void*, hipStream_t);
 void srj_json_map_count(const void*, int64_t, int32_t*, hipStream_t);
 void srj_json_map_entry_lens(const void*, int64_t, const int32_t*, int32_t*,
                              int32_t*, hipStream_t);
@@ -23,6 +26,16 @@ void register_json(py::module_& m) {
                         as_ptr<char>(chars), as_ptr<uint8_t>(valid),
                         as_stream(stream));
     check_hip("get_json_object");
+  });
+  m.def("get_json_multi", [](uintptr_t in, int64_t n, uintptr_t instrs,
+                             uintptr_t keychars, uintptr_t path_off,
+                             uintptr_t path_len, int32_t npaths, int32_t phase,
+                             uintptr_t outs, uintptr_t stream) {
+    srj_get_json_multi(as_ptr<void>(in), n, as_ptr<void>(instrs),
+                       as_ptr<char>(keychars), as_ptr<int32_t>(path_off),
+                       as_ptr<int32_t>(path_len), npaths, phase,
+                       as_ptr<void>(outs), as_stream(stream));
+    check_hip("get_json_multi");
   });
   m.def("json_map_count", [](uintptr_t in, int64_t n, uintptr_t counts,
                              uintptr_t stream) {

@@ -303,6 +303,195 @@ __global__ void get_json_object_kernel(ColDesc in, int64_t nrows,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// multi-path shared scan (reference get_json_object.cu multi-path kernel with
+// num_threads_per_row / JSONUtils.getJsonObjectMultiplePaths): every path
+// instruction consumes exactly one nesting level, so ONE document walk can
+// carry a bitmask of still-viable paths and match them all simultaneously —
+// the doc is tokenized once instead of once per path.
+// ---------------------------------------------------------------------------
+constexpr int MAX_MULTI_PATHS = 8;
+constexpr int MULTI_MATCH_CAP = 4;
+
+struct MultiCtx {
+  const PathInstr* instrs;   // concatenated
+  const char* keychars;
+  const int32_t* path_off;   // [npaths] start into instrs
+  const int32_t* path_len;   // [npaths] instruction count
+  int32_t npaths;
+  uint8_t nmatches[MAX_MULTI_PATHS];
+  bool overflow[MAX_MULTI_PATHS];
+  JsonSpan matches[MAX_MULTI_PATHS][MULTI_MATCH_CAP];
+};
+
+__device__ inline void multi_record(MultiCtx& ctx, int i, const char* p,
+                                    const char* vend) {
+  if (ctx.nmatches[i] < MULTI_MATCH_CAP)
+    ctx.matches[i][ctx.nmatches[i]] = {p, (int32_t)(vend - p)};
+  else
+    ctx.overflow[i] = true;
+  ++ctx.nmatches[i];
+}
+
+// single walk carrying `mask` of paths viable at this nesting level
+__device__ bool j_match_multi(const char* p, const char* e, MultiCtx& ctx,
+                              int32_t depth, uint32_t mask) {
+  if (depth > MAX_PATH_DEPTH + 2) return false;
+  p = j_skip_ws(p, e);
+  // terminal paths match this whole value
+  uint32_t term = 0;
+  for (int i = 0; i < ctx.npaths; ++i)
+    if ((mask >> i) & 1 && ctx.path_len[i] == depth) term |= 1u << i;
+  if (term) {
+    const char* vend = j_skip_value(p, e);
+    if (!vend) return false;
+    for (int i = 0; i < ctx.npaths; ++i)
+      if ((term >> i) & 1) multi_record(ctx, i, p, vend);
+  }
+  uint32_t desc = mask & ~term;
+  if (!desc || p >= e) return true;
+  if (*p == '{') {
+    ++p;
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p < e && *p == '}') return true;
+      if (p >= e || *p != '"') return false;
+      const char* kend = j_skip_string(p, e);
+      if (!kend) return false;
+      const char* kbody = p + 1;
+      int32_t klen = (int32_t)(kend - p - 2);
+      p = j_skip_ws(kend, e);
+      if (p >= e || *p != ':') return false;
+      ++p;
+      p = j_skip_ws(p, e);
+      uint32_t m2 = 0;
+      for (int i = 0; i < ctx.npaths; ++i) {
+        if (!((desc >> i) & 1)) continue;
+        const PathInstr ins = ctx.instrs[ctx.path_off[i] + depth];
+        if (ins.kind == 2) {
+          m2 |= 1u << i;
+        } else if (ins.kind == 0 && ins.key_len == klen) {
+          bool eq = true;
+          for (int32_t k = 0; k < klen; ++k)
+            if (kbody[k] != ctx.keychars[ins.key_off + k]) { eq = false; break; }
+          if (eq) m2 |= 1u << i;
+        }
+      }
+      if (m2) {
+        if (!j_match_multi(p, e, ctx, depth + 1, m2)) return false;
+      }
+      const char* vend = j_skip_value(p, e);
+      if (!vend) return false;
+      p = j_skip_ws(vend, e);
+      if (p < e && *p == ',') { ++p; continue; }
+      if (p < e && *p == '}') return true;
+      return false;
+    }
+  }
+  if (*p == '[') {
+    ++p;
+    int32_t idx = 0;
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p < e && *p == ']') return true;
+      uint32_t m2 = 0;
+      for (int i = 0; i < ctx.npaths; ++i) {
+        if (!((desc >> i) & 1)) continue;
+        const PathInstr ins = ctx.instrs[ctx.path_off[i] + depth];
+        if (ins.kind == 2 || (ins.kind == 1 && ins.index == idx)) m2 |= 1u << i;
+      }
+      if (m2) {
+        if (!j_match_multi(p, e, ctx, depth + 1, m2)) return false;
+      }
+      const char* vend = j_skip_value(p, e);
+      if (!vend) return false;
+      p = j_skip_ws(vend, e);
+      ++idx;
+      if (p < e && *p == ',') { ++p; continue; }
+      if (p < e && *p == ']') return true;
+      return false;
+    }
+  }
+  return true;  // structure mismatch: no match, not an error
+}
+
+// per-path output block pointers (lens / offsets / chars / validity)
+struct MultiOut {
+  int32_t* lens;
+  const int32_t* offsets;
+  char* chars;
+  uint8_t* out_valid;
+  int32_t* overflow;  // one flag per path (host triggers per-path fallback)
+};
+
+template <bool WRITE>
+__global__ void get_json_multi_kernel(ColDesc in, int64_t nrows,
+                                      const PathInstr* __restrict__ instrs,
+                                      const char* __restrict__ keychars,
+                                      const int32_t* __restrict__ path_off,
+                                      const int32_t* __restrict__ path_len,
+                                      int32_t npaths,
+                                      const MultiOut* __restrict__ outs) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t npad = (nrows + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < npad;
+       row += stride) {
+    bool in_range = row < nrows;
+    bool doc_valid = in_range && is_valid(in.valid, row);
+    MultiCtx ctx;
+    ctx.instrs = instrs;
+    ctx.keychars = keychars;
+    ctx.path_off = path_off;
+    ctx.path_len = path_len;
+    ctx.npaths = npaths;
+    for (int i = 0; i < npaths; ++i) {
+      ctx.nmatches[i] = 0;
+      ctx.overflow[i] = false;
+    }
+    bool ok = false;
+    if (doc_valid) {
+      StrView s = get_string(in, row);
+      ok = j_match_multi(s.ptr, s.ptr + s.len, ctx, 0,
+                         (uint32_t)((1u << npaths) - 1));
+    }
+    for (int i = 0; i < npaths; ++i) {
+      const MultiOut o = outs[i];
+      bool valid = doc_valid && ok && ctx.nmatches[i] > 0 && !ctx.overflow[i];
+      int32_t out_len = 0;
+      Sink sink{WRITE && in_range && o.offsets
+                    ? o.chars + o.offsets[row] : nullptr,
+                0, WRITE};
+      if (valid) {
+        if (ctx.nmatches[i] == 1) {
+          JsonSpan m = ctx.matches[i][0];
+          if (m.len == 4 && m.p[0] == 'n' && m.p[1] == 'u' && m.p[2] == 'l' &&
+              m.p[3] == 'l') {
+            valid = false;
+          } else {
+            emit_match(sink, m, false);
+            out_len = sink.pos;
+          }
+        } else {
+          sink.put('[');
+          for (int32_t k = 0; k < (int32_t)ctx.nmatches[i]; ++k) {
+            if (k) sink.put(',');
+            emit_match(sink, ctx.matches[i][k], true);
+          }
+          sink.put(']');
+          out_len = sink.pos;
+        }
+      }
+      if (ctx.overflow[i] && in_range) atomicOr(o.overflow + 0, 1);
+      if (WRITE) {
+        ballot_write_validity(o.out_valid, row, valid);
+      } else if (in_range) {
+        o.lens[row] = valid ? out_len : 0;
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // from_json: top-level object field extraction -> string columns (one kernel
 // run per schema field reusing the path machinery), and raw-map extraction.
@@ -440,6 +629,21 @@ void srj_get_json_object(const void* in, int64_t nrows, const void* instrs,
     get_json_object_kernel<true><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
         c, nrows, reinterpret_cast<const PathInstr*>(instrs), keychars, ninstr,
         nullptr, offsets, chars, out_valid);
+}
+
+void srj_get_json_multi(const void* in, int64_t nrows, const void* instrs,
+                        const char* keychars, const int32_t* path_off,
+                        const int32_t* path_len, int32_t npaths, int32_t phase,
+                        const void* outs, hipStream_t stream) {
+  ColDesc c = *reinterpret_cast<const ColDesc*>(in);
+  if (phase == 0)
+    get_json_multi_kernel<false><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+        c, nrows, reinterpret_cast<const PathInstr*>(instrs), keychars,
+        path_off, path_len, npaths, reinterpret_cast<const MultiOut*>(outs));
+  else
+    get_json_multi_kernel<true><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
+        c, nrows, reinterpret_cast<const PathInstr*>(instrs), keychars,
+        path_off, path_len, npaths, reinterpret_cast<const MultiOut*>(outs));
 }
 
 void srj_json_map_count(const void* in, int64_t nrows, int32_t* entry_counts,

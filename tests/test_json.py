@@ -116,3 +116,33 @@ def test_get_json_object_fuzz_vs_oracle():
                 else:
                     exp = json.dumps(v, separators=(",", ":"))
                 assert gv == exp, (doc, key, gv)
+
+@pytest.mark.gpu
+def test_multi_path_shared_scan():
+    """Multi-path results must equal per-path get_json_object exactly."""
+    from spark_rapids_jni_amd.ops.json import (get_json_object,
+                                               get_json_object_multiple_paths)
+    docs = ['{"a":%d,"b":{"c":"x%d"},"d":[%d,%d,%d],"e":"v","f":null}' %
+            (i, i % 9, i, i + 1, i + 2) for i in range(3000)]
+    docs += ['{"a":1', None, '[]', '{"d":[5]}']
+    col = Column.from_pylist(docs, DType.STRING, "cuda")
+    paths = ["$.a", "$.b.c", "$.d[1]", "$.d[*]", "$.e", "$.f", "$.missing",
+             "$.b", "$.d[9]"]
+    multi = get_json_object_multiple_paths(col, paths)
+    for p, mc in zip(paths, multi):
+        sc = get_json_object(col, p)
+        assert mc.to_pylist() == sc.to_pylist(), f"path {p}"
+
+
+@pytest.mark.gpu
+def test_multi_path_overflow_fallback():
+    """>4 wildcard matches per row triggers the per-path fallback."""
+    from spark_rapids_jni_amd.ops.json import (get_json_object,
+                                               get_json_object_multiple_paths)
+    docs = ['[1,2,3,4,5,6,7]', '[1,2]']
+    col = Column.from_pylist(docs, DType.STRING, "cuda")
+    paths = ["$[*]", "$[0]"]
+    multi = get_json_object_multiple_paths(col, paths)
+    for p, mc in zip(paths, multi):
+        sc = get_json_object(col, p)
+        assert mc.to_pylist() == sc.to_pylist(), f"path {p}"
