@@ -92,3 +92,71 @@ def q3_like(store_sales: Table, date_dim: Table, item: Table, year: int,
     keys, results = groupby(brand, [(Agg.SUM, ss2.columns[3]),
                                     (Agg.COUNT_ALL, None)])
     return keys, results
+
+
+def q9_like(store_sales: Table, buckets: Sequence[int]):
+    """NDS q9 shape: bucketed conditional aggregates over one fact scan —
+    CASE WHEN qty BETWEEN lo AND hi THEN ... per bucket, avg(price) within
+    each quantity bucket (select_first_true_index + masked aggregates).
+
+    Returns [(count, avg_price) per bucket].
+    """
+    from .ops.misc import select_first_true_index
+    qty = store_sales.columns[2].data
+    price = store_sales.columns[3]
+    conds = []
+    lo = 0
+    for hi in buckets:
+        conds.append(Column.from_torch(((qty > lo) & (qty <= hi)).to(torch.int8),
+                                       dtype=DType.BOOL8))
+        lo = hi
+    bucket_idx = select_first_true_index(conds)
+    out = []
+    for b in range(len(buckets)):
+        mask = bucket_idx.data == b
+        cnt = int(mask.sum().item())
+        s = float(price.data[mask].sum().item()) if cnt else 0.0
+        out.append((cnt, s / cnt if cnt else None))
+    return out
+
+
+def q95_like(web_sales: Table, web_returns_order_sk: Column,
+             ship_date_sk: Column):
+    """NDS q95 shape: orders shipped in a date window whose order number
+    also appears in returns (left SEMI join) and orders that never appear
+    (ANTI join) — exercises the semi/anti gather-map algebra.
+
+    web_sales: [order_number, ship_date_sk, net_profit]
+    Returns (semi_count, anti_count, semi_profit_sum).
+    """
+    dd_tbl = HashJoinTable.build(ship_date_sk)
+    sel = dd_tbl.semi_join(web_sales.columns[1])
+    ws = gather(web_sales, sel)
+    ret_tbl = HashJoinTable.build(web_returns_order_sk)
+    semi = ret_tbl.semi_join(ws.columns[0])
+    anti = ret_tbl.semi_join(ws.columns[0], anti=True)
+    profit = gather_column(ws.columns[2], semi)
+    return semi.numel(), anti.numel(), float(profit.data.sum().item())
+
+
+def q1_like(store_returns: Table, avg_factor: float = 1.2):
+    """NDS q1 shape: per (customer, store) return totals, HAVING
+    total > avg_factor * store average (group-by -> second group-by ->
+    broadcast join back -> filter).
+
+    store_returns: [customer_sk, store_sk, return_amt]
+    Returns the customer_sk column of qualifying rows.
+    """
+    cust, store, amt = store_returns.columns
+    pair_keys, pair_res = groupby(Table([cust, store]), [(Agg.SUM, amt)])
+    totals = pair_res[0]
+    # per-store average of the per-customer totals
+    store_keys, store_res = groupby(pair_keys.columns[1],
+                                    [(Agg.SUM, totals), (Agg.COUNT_ALL, None)])
+    avg = store_res[0].data.to(torch.float64) / store_res[1].data.clamp(min=1)
+    st_tbl = HashJoinTable.build(store_keys.columns[0])
+    bi, pi = st_tbl.inner_join(pair_keys.columns[1])
+    thresh = avg[bi.long()] * avg_factor
+    tot = totals.data.to(torch.float64)[pi]
+    qual = pi[tot > thresh]
+    return gather_column(pair_keys.columns[0], qual)
