@@ -10,6 +10,7 @@ Scope v1: flat schemas; physical types BOOLEAN/INT32/INT64/FLOAT/DOUBLE/
 BYTE_ARRAY; PLAIN + RLE_DICTIONARY/PLAIN_DICTIONARY encodings; UNCOMPRESSED
 and SNAPPY (host-decompressed) pages; data page V1 and V2.
 """
+import mmap
 import struct
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Tuple
@@ -118,8 +119,8 @@ def read_footer(path_or_bytes) -> ParquetFooter:
     raw = (open(path_or_bytes, "rb").read()
            if isinstance(path_or_bytes, str) else path_or_bytes)
     assert raw[:4] == MAGIC and raw[-4:] == MAGIC, "not a parquet file"
-    flen = struct.unpack("<I", raw[-8:-4])[0]
-    fmd, _ = _native.host().thrift_parse(raw[-8 - flen:-8], 0)
+    flen = struct.unpack("<I", bytes(raw[-8:-4]))[0]
+    fmd, _ = _native.host().thrift_parse(bytes(raw[-8 - flen:-8]), 0)
 
     schema_elems = fmd[2]
     root = schema_elems[0]
@@ -190,13 +191,29 @@ def _decompress(codec, data, uncompressed_size):
     raise NotImplementedError(f"parquet codec {codec}")
 
 
-def _walk_pages(raw: bytes, chunk: ColumnChunkMeta) -> List[_Page]:
+def _parse_page_header(h, raw, pos):
+    """Parse one thrift page header from a buffer without materializing the
+    whole file as bytes (headers are small; grow the window on truncation)."""
+    if isinstance(raw, bytes):
+        ph, end = h.thrift_parse(raw, pos)
+        return ph, end
+    for sz in (4096, 1 << 16, 1 << 22):
+        try:
+            ph, end = h.thrift_parse(bytes(raw[pos:pos + sz]), 0)
+            return ph, pos + end
+        except RuntimeError:
+            continue
+    ph, end = h.thrift_parse(bytes(raw[pos:]), 0)
+    return ph, pos + end
+
+
+def _walk_pages(raw, chunk: ColumnChunkMeta) -> List[_Page]:
     h = _native.host()
     pos = chunk.start_offset
     pages = []
     values_seen = 0
     while values_seen < chunk.num_values:
-        ph, end = h.thrift_parse(raw, pos)
+        ph, end = _parse_page_header(h, raw, pos)
         ptype = ph[1]
         uncomp = ph[2]
         comp = ph[3]
@@ -217,11 +234,12 @@ def _walk_pages(raw: bytes, chunk: ColumnChunkMeta) -> List[_Page]:
             dlen = dph.get(5, 0)
             rlen = dph.get(6, 0)
             assert rlen == 0, "nested (repeated) columns not supported yet"
-            levels = payload[:dlen]
-            body = payload[dlen:]
             if dph.get(7, True) and chunk.codec != CODEC_UNCOMPRESSED:
-                body = _decompress(chunk.codec, body, uncomp - dlen)
-            pages.append(_Page(2, nv, dph[4], levels + body, def_bytes=dlen,
+                body = _decompress(chunk.codec, payload[dlen:], uncomp - dlen)
+                data = bytes(payload[:dlen]) + bytes(body)
+            else:
+                data = payload  # levels + body are already contiguous
+            pages.append(_Page(2, nv, dph[4], data, def_bytes=dlen,
                                num_nulls=dph.get(2, -1)))
             values_seen += nv
         else:
@@ -251,17 +269,32 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                 pages.append(p)
                 dict_per_page.append(cur_dict)
 
-    # upload all page payloads in one buffer
+    # upload all page payloads in one buffer: stage through PINNED host
+    # memory (CachingHostAllocator keeps the source alive across the async
+    # H2D) — with an mmap'd file the page bytes are copied exactly once on
+    # the host before the DMA
     blobs = [p.data for p in pages] + [p.data for p in dicts]
     offs = np.zeros(len(blobs) + 1, dtype=np.int64)
     for i, b in enumerate(blobs):
         offs[i + 1] = offs[i] + ((len(b) + 7) & ~7)
-    big = torch.zeros(max(int(offs[-1]), 1), dtype=torch.uint8, device=dev)
-    hbuf = np.zeros(int(offs[-1]), dtype=np.uint8)
-    for i, b in enumerate(blobs):
-        hbuf[offs[i]:offs[i] + len(b)] = np.frombuffer(b, dtype=np.uint8)
-    if len(hbuf):
-        big[:len(hbuf)] = torch.from_numpy(hbuf)
+    if int(offs[-1]) and dev.type == "cuda":
+        hbuf_t = torch.empty(int(offs[-1]), dtype=torch.uint8,
+                             pin_memory=True)
+        views = [np.frombuffer(b, dtype=np.uint8) for b in blobs]
+        _native.host().copy_blobs(
+            hbuf_t.data_ptr(),
+            [(v.ctypes.data, int(offs[i]), v.nbytes)
+             for i, v in enumerate(views) if v.nbytes],
+            8)
+        del views
+        big = hbuf_t.to(dev, non_blocking=True)
+    else:
+        big = torch.zeros(max(int(offs[-1]), 1), dtype=torch.uint8, device=dev)
+        hbuf = np.zeros(int(offs[-1]), dtype=np.uint8)
+        for i, b in enumerate(blobs):
+            hbuf[offs[i]:offs[i] + len(b)] = np.frombuffer(b, dtype=np.uint8)
+        if len(hbuf):
+            big[:len(hbuf)] = torch.from_numpy(hbuf)
     pbase = [big.data_ptr() + int(offs[i]) for i in range(len(blobs))]
     dict_base_idx = len(pages)
 
@@ -470,10 +503,37 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                   scale=0, null_count=None)
 
 
+_MMAP_CACHE: Dict[tuple, memoryview] = {}
+
+
+def _mmap_file(path: str) -> memoryview:
+    import os
+    st = os.stat(path)
+    key = (path, st.st_mtime_ns, st.st_size)
+    mv = _MMAP_CACHE.get(key)
+    if mv is None:
+        with open(path, "rb") as f:
+            mm = mmap.mmap(f.fileno(), 0, access=mmap.ACCESS_READ)
+        try:
+            mm.madvise(mmap.MADV_WILLNEED)
+        except (AttributeError, OSError):
+            pass
+        mv = memoryview(mm)
+        if len(_MMAP_CACHE) >= 8:
+            _MMAP_CACHE.clear()
+        _MMAP_CACHE[key] = mv
+    return mv
+
+
 def read_table(path: str, columns: Optional[Sequence[str]] = None,
                device="cuda") -> Table:
-    """Scan a parquet file into a GPU Table (footer + page decode)."""
-    raw = open(path, "rb").read()
+    """Scan a parquet file into a GPU Table (footer + page decode).
+
+    The file is mmap'd: only the footer and the selected columns' page
+    ranges are ever touched, and page bytes flow mmap -> pinned staging ->
+    device with a single host copy. Mappings are cached per (path, mtime,
+    size) so repeated scans pay the soft page faults once."""
+    raw = _mmap_file(path)
     footer = read_footer(raw)
     if columns is not None:
         footer = footer.prune(columns)

@@ -5,10 +5,14 @@
 // walker that materializes structs as {field_id: value} Python dicts; the
 // Parquet schema semantics live in spark_rapids_jni_amd/parquet.py.
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
 
 #include <cstdint>
+#include <cstring>
 #include <stdexcept>
 #include <string>
+#include <thread>
+#include <vector>
 
 namespace py = pybind11;
 
@@ -330,4 +334,41 @@ void register_thrift(py::module_& m) {
           return py::bytes(w.out);
         },
         "Serialize a typed struct tree back to compact protocol");
+  // multithreaded scatter-memcpy for the parquet page staging path: copies
+  // many (src, dst_off, len) blobs into one pinned buffer in parallel — a
+  // single-threaded Python loop runs at ~20 GB/s, this saturates host DRAM.
+  m.def("copy_blobs",
+        [](uintptr_t dst,
+           std::vector<std::tuple<uintptr_t, uint64_t, uint64_t>> parts,
+           int nthreads) {
+          py::gil_scoped_release rel;
+          if (nthreads < 1) nthreads = 1;
+          uint64_t total = 0;
+          for (auto& p : parts) total += std::get<2>(p);
+          uint64_t per = (total + nthreads - 1) / nthreads;
+          // assign whole blobs to threads by running-byte ranges; large
+          // blobs are split at stripe boundaries
+          std::vector<std::thread> ts;
+          for (int t = 0; t < nthreads; ++t) {
+            uint64_t lo = per * t, hi = per * (t + 1);
+            ts.emplace_back([&, lo, hi]() {
+              uint64_t pos = 0;
+              for (auto& p : parts) {
+                uint64_t len = std::get<2>(p);
+                uint64_t a = pos, b = pos + len;
+                pos = b;
+                uint64_t s0 = a < lo ? lo : a;
+                uint64_t s1 = b > hi ? hi : b;
+                if (s0 >= s1) continue;
+                std::memcpy(reinterpret_cast<char*>(dst) + std::get<1>(p) +
+                                (s0 - a),
+                            reinterpret_cast<const char*>(std::get<0>(p)) +
+                                (s0 - a),
+                            s1 - s0);
+              }
+            });
+          }
+          for (auto& t : ts) t.join();
+        },
+        py::arg("dst"), py::arg("parts"), py::arg("nthreads") = 8);
 }
