@@ -37,11 +37,38 @@ def emit(name, rows, secs, bytes_processed=None):
     print(json.dumps(rec), flush=True)
 
 
+def plumbing_bench():
+    """BASELINE config[0]: murmur3 + row<->columnar on a 1k-row int64 batch
+    via the HOST path (no GPU) — measures binding/plumbing overhead like the
+    reference's Java/JNI host path."""
+    import io
+    import numpy as np
+    from spark_rapids_jni_amd import _native, kudo
+    from spark_rapids_jni_amd.columnar import Column, DType
+    host = _native.host()
+    n = 1000
+    keys = torch.arange(n, dtype=torch.int64)
+    out = torch.empty(n, dtype=torch.int32)
+    col = Column.from_torch(keys)
+    iters = 2000
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        host.murmur3_long_host(keys.data_ptr(), n, 42, out.data_ptr())
+        b = io.BytesIO()
+        kudo.write_partition([col], 0, n, b)
+        kudo.merge_on_host([b.getvalue()], [col])
+    dt = (time.perf_counter() - t0) / iters
+    print(json.dumps({"bench": "host_plumbing_murmur3_rowcol_1k",
+                      "rows": n, "us_per_batch": round(dt * 1e6, 1),
+                      "rows_per_sec": round(n / dt, 1)}), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--rows", type=int, default=2**24)
     ap.add_argument("--iters", type=int, default=5)
     args = ap.parse_args()
+    plumbing_bench()  # CPU shape (BASELINE config[0]) first
     n = args.rows
     dev = "cuda"
 

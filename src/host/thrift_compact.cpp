@@ -18,6 +18,31 @@ namespace py = pybind11;
 
 namespace {
 
+// Spark murmur3 for int64 (host mirror of src/gpu/srj_common.hpp mm3_*:
+// two 4-byte little-endian blocks per long) — the BASELINE config[0]
+// "plumbing" path and a host/GPU cross-check.
+inline uint32_t h_rotl32(uint32_t x, int r) { return (x << r) | (x >> (32 - r)); }
+inline uint32_t h_mm3_mix_k1(uint32_t k1) {
+  k1 *= 0xcc9e2d51u;
+  k1 = h_rotl32(k1, 15);
+  k1 *= 0x1b873593u;
+  return k1;
+}
+inline uint32_t h_mm3_mix_h1(uint32_t h1, uint32_t k1) {
+  h1 ^= k1;
+  h1 = h_rotl32(h1, 13);
+  return h1 * 5u + 0xe6546b64u;
+}
+inline uint32_t h_mm3_fmix(uint32_t h1, uint32_t len) {
+  h1 ^= len;
+  h1 ^= h1 >> 16;
+  h1 *= 0x85ebca6bu;
+  h1 ^= h1 >> 13;
+  h1 *= 0xc2b2ae35u;
+  h1 ^= h1 >> 16;
+  return h1;
+}
+
 struct Reader {
   const uint8_t* p;
   size_t len;
@@ -371,4 +396,17 @@ void register_thrift(py::module_& m) {
           for (auto& t : ts) t.join();
         },
         py::arg("dst"), py::arg("parts"), py::arg("nthreads") = 8);
+  m.def("murmur3_long_host",
+        [](uintptr_t data, int64_t n, uint32_t seed, uintptr_t out) {
+          const int64_t* v = reinterpret_cast<const int64_t*>(data);
+          int32_t* o = reinterpret_cast<int32_t*>(out);
+          for (int64_t i = 0; i < n; ++i) {
+            uint32_t lo = (uint32_t)v[i];
+            uint32_t hi = (uint32_t)(((uint64_t)v[i]) >> 32);
+            uint32_t h1 = h_mm3_mix_h1(seed, h_mm3_mix_k1(lo));
+            h1 = h_mm3_mix_h1(h1, h_mm3_mix_k1(hi));
+            o[i] = (int32_t)h_mm3_fmix(h1, 8);
+          }
+        },
+        "Spark murmur3 of an int64 array on the host (config[0] plumbing)");
 }
