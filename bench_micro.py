@@ -148,6 +148,22 @@ def main():
         emit(f"get_json_object_{k}path", jc.size * k, secs,
              int(jc.offsets[-1].item()) * k)
 
+    # 5b. generic (multi-column) hash join probe — the non-int64 join path
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    nb = n // 2
+    npr = n
+    k1 = torch.randint(0, nb, (nb,), dtype=torch.int32, device=dev, generator=g)
+    k2 = torch.randint(0, 8, (nb,), dtype=torch.int32, device=dev, generator=g)
+    btbl = Table([Column.from_torch(k1), Column.from_torch(k2)])
+    p1 = torch.randint(0, nb, (npr,), dtype=torch.int32, device=dev,
+                       generator=g)
+    p2 = torch.randint(0, 8, (npr,), dtype=torch.int32, device=dev,
+                       generator=g)
+    ptbl = Table([Column.from_torch(p1), Column.from_torch(p2)])
+    tbl = HashJoinTable.build(btbl)
+    secs = timeit(lambda: tbl.inner_join(ptbl, out_hint=npr // 4), args.iters)
+    emit("generic_join_probe_2col", npr, secs)
+
     # 6. parse_uri
     uris = ["https://host%d.example.com:80/p/%d?k=%d&z=9" % (i % 50, i, i)
             for i in range(200_000)]

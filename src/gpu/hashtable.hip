@@ -55,6 +55,8 @@ __global__ void join_build_kernel(const ColDesc* __restrict__ cols,
 // probe: two-phase (count then fill). COUNT uses one wave-level reduction +
 // a single atomicAdd per wave. FILL appends match pairs at a global cursor.
 // ---------------------------------------------------------------------------
+constexpr int GPIPE = 8;  // probe rows in flight per lane (as hashtable_i64)
+
 template <bool FILL>
 __global__ void join_probe_kernel(
     const ColDesc* __restrict__ bcols, const int32_t* __restrict__ btop,
@@ -63,33 +65,53 @@ __global__ void join_probe_kernel(
     uint64_t mask, uint64_t* __restrict__ counter,
     int32_t* __restrict__ out_build, int64_t* __restrict__ out_probe,
     int64_t out_capacity, uint8_t* __restrict__ build_matched) {
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  // software-pipelined batches: hashes and first slot words for GPIPE rows
+  // are issued before any chain resolves (the slot-word loads are THE
+  // random accesses this kernel is bound by; same structure as the int64
+  // fast path in hashtable_i64.hip)
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   uint64_t local = 0;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < nprobe;
-       row += stride) {
-    if (row_has_null_key(pcols, ptop, ntop, row)) continue;
-    uint64_t h = row_hash64(pcols, ptop, ntop, row);
-    uint32_t fp = fingerprint(h);
-    uint64_t s = h & mask;
-    while (true) {
-      uint64_t word = slots[s];
-      if (word == 0) break;
-      if ((uint32_t)(word >> 32) == fp) {
-        int64_t brow = (int64_t)(word & 0xffffffffu) - 1;
-        if (rows_equal(bcols, btop, ntop, brow, pcols, ptop, row)) {
-          if (FILL) {
-            uint64_t pos = atomicAdd((unsigned long long*)counter, 1ull);
-            if ((int64_t)pos < out_capacity) {
-              out_build[pos] = (int32_t)brow;
-              out_probe[pos] = row;
+  for (int64_t base = tid * GPIPE; base < nprobe; base += nthreads * GPIPE) {
+    bool act[GPIPE];
+    uint64_t h[GPIPE];
+    uint64_t s0[GPIPE];
+    uint64_t w[GPIPE];
+#pragma unroll
+    for (int b = 0; b < GPIPE; ++b) {
+      int64_t row = base + b;
+      act[b] = row < nprobe && !row_has_null_key(pcols, ptop, ntop, row);
+      h[b] = act[b] ? row_hash64(pcols, ptop, ntop, row) : 0;
+      s0[b] = h[b] & mask;
+    }
+#pragma unroll
+    for (int b = 0; b < GPIPE; ++b) w[b] = slots[s0[b]];
+#pragma unroll
+    for (int b = 0; b < GPIPE; ++b) {
+      if (!act[b]) continue;
+      int64_t row = base + b;
+      uint32_t fp = fingerprint(h[b]);
+      uint64_t s = s0[b];
+      uint64_t word = w[b];
+      while (word != 0) {
+        if ((uint32_t)(word >> 32) == fp) {
+          int64_t brow = (int64_t)(word & 0xffffffffu) - 1;
+          if (rows_equal(bcols, btop, ntop, brow, pcols, ptop, row)) {
+            if (FILL) {
+              uint64_t pos = atomicAdd((unsigned long long*)counter, 1ull);
+              if ((int64_t)pos < out_capacity) {
+                out_build[pos] = (int32_t)brow;
+                out_probe[pos] = row;
+              }
+              if (build_matched) build_matched[brow] = 1;
+            } else {
+              ++local;
             }
-            if (build_matched) build_matched[brow] = 1;
-          } else {
-            ++local;
           }
         }
+        s = (s + 1) & mask;
+        word = slots[s];
       }
-      s = (s + 1) & mask;
     }
   }
   if (!FILL) {
@@ -232,7 +254,8 @@ void srj_join_probe_count(const void* bcols, const int32_t* btop,
                           const void* pcols, const int32_t* ptop, int32_t ntop,
                           int64_t nprobe, const uint64_t* slots, int64_t capacity,
                           uint64_t* counter, hipStream_t stream) {
-  join_probe_kernel<false><<<grid_1d(nprobe), DEFAULT_BLOCK, 0, stream>>>(
+  join_probe_kernel<false><<<grid_1d((nprobe + GPIPE - 1) / GPIPE),
+                             DEFAULT_BLOCK, 0, stream>>>(
       reinterpret_cast<const ColDesc*>(bcols), btop,
       reinterpret_cast<const ColDesc*>(pcols), ptop, ntop, nprobe, slots,
       (uint64_t)(capacity - 1), counter, nullptr, nullptr, 0, nullptr);
