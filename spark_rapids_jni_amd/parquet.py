@@ -39,6 +39,7 @@ _SCATTER_FMT = "<QQQQqqqii"     # ScatterDesc (64B)
 _RLE_FMT = "<QqQqii"            # RleDesc (40B)
 _STRIDX_FMT = "<QqqQQ"          # StrIndexDesc (40B)
 _STRCPY_FMT = "<QQQQQQqqq"      # StrCopyDesc (72B)
+_SNAP_FMT = "<QqQq"             # SnapDesc (32B)
 
 
 @dataclass
@@ -176,9 +177,11 @@ class _Page:
     kind: int                # 0 data v1, 1 dict, 2 data v2
     num_values: int
     encoding: int
-    data: bytes              # uncompressed page payload
+    data: bytes              # page payload (still compressed when comp=True)
     def_bytes: int = 0       # v2: definition level byte length
     num_nulls: int = -1      # v2 only
+    comp: bool = False       # True: snappy payload decompressed ON DEVICE
+    uncomp: int = 0          # uncompressed size when comp=True
 
 
 def _decompress(codec, data, uncompressed_size):
@@ -207,7 +210,8 @@ def _parse_page_header(h, raw, pos):
     return ph, pos + end
 
 
-def _walk_pages(raw, chunk: ColumnChunkMeta) -> List[_Page]:
+def _walk_pages(raw, chunk: ColumnChunkMeta,
+                keep_compressed: bool = False) -> List[_Page]:
     h = _native.host()
     pos = chunk.start_offset
     pages = []
@@ -221,13 +225,21 @@ def _walk_pages(raw, chunk: ColumnChunkMeta) -> List[_Page]:
         pos = end + comp
         if ptype == 0:  # DATA_PAGE v1
             dph = ph[5]
-            data = _decompress(chunk.codec, payload, uncomp)
-            pages.append(_Page(0, dph[1], dph[2], data))
+            if keep_compressed and chunk.codec == CODEC_SNAPPY:
+                pages.append(_Page(0, dph[1], dph[2], payload, comp=True,
+                                   uncomp=uncomp))
+            else:
+                data = _decompress(chunk.codec, payload, uncomp)
+                pages.append(_Page(0, dph[1], dph[2], data))
             values_seen += dph[1]
         elif ptype == 2:  # DICTIONARY_PAGE
             dph = ph[7]
-            data = _decompress(chunk.codec, payload, uncomp)
-            pages.append(_Page(1, dph[1], dph.get(2, ENC_PLAIN), data))
+            if keep_compressed and chunk.codec == CODEC_SNAPPY:
+                pages.append(_Page(1, dph[1], dph.get(2, ENC_PLAIN), payload,
+                                   comp=True, uncomp=uncomp))
+            else:
+                data = _decompress(chunk.codec, payload, uncomp)
+                pages.append(_Page(1, dph[1], dph.get(2, ENC_PLAIN), data))
         elif ptype == 3:  # DATA_PAGE_V2
             dph = ph[8]
             nv = dph[1]
@@ -255,13 +267,15 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     nullable = f.repetition == 1
     dtype = _field_dtype(f)
 
-    # flatten pages across row groups
+    # flatten pages across row groups; snappy payloads stay compressed and
+    # are decompressed ON DEVICE (one wave per page) for non-BOOLEAN columns
+    keep_comp = dev.type == "cuda" and f.physical_type != T_BOOLEAN
     pages: List[_Page] = []
     dict_per_page: List[int] = []   # index into dict list, -1 none
     dicts: List[_Page] = []
     for ch in chunks:
         cur_dict = -1
-        for p in _walk_pages(raw, ch):
+        for p in _walk_pages(raw, ch, keep_compressed=keep_comp):
             if p.kind == 1:
                 dicts.append(p)
                 cur_dict = len(dicts) - 1
@@ -298,6 +312,43 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     pbase = [big.data_ptr() + int(offs[i]) for i in range(len(blobs))]
     dict_base_idx = len(pages)
 
+    # device snappy decompression of compressed blobs; pbase is repointed at
+    # the decompressed copies so everything downstream is codec-agnostic
+    allp = pages + dicts
+    comp_ids = [j for j, p in enumerate(allp) if p.comp]
+    if comp_ids:
+        offs2 = np.zeros(len(comp_ids) + 1, dtype=np.int64)
+        for k, j in enumerate(comp_ids):
+            offs2[k + 1] = offs2[k] + ((allp[j].uncomp + 7) & ~7)
+        big2 = torch.empty(max(int(offs2[-1]), 1), dtype=torch.uint8,
+                           device=dev)
+        sd = bytearray()
+        for k, j in enumerate(comp_ids):
+            sd += struct.pack(_SNAP_FMT, pbase[j], len(allp[j].data),
+                              big2.data_ptr() + int(offs2[k]), allp[j].uncomp)
+        sdt = torch.frombuffer(sd, dtype=torch.uint8).to(dev)
+        g.pq_snappy_decomp(sdt.data_ptr(), len(comp_ids), stream)
+        for k, j in enumerate(comp_ids):
+            pbase[j] = big2.data_ptr() + int(offs2[k])
+
+    def _plen(j):
+        # payload length as visible to the decode kernels (decompressed)
+        return allp[j].uncomp if allp[j].comp else len(allp[j].data)
+
+    # def-level lengths of compressed nullable v1 pages live inside the
+    # device-decompressed bytes: fetch them in one small gather + D2H
+    dl_map = {}
+    if nullable:
+        need = [i for i, p in enumerate(pages) if p.comp and p.kind == 0]
+        if need:
+            addrs = torch.tensor([pbase[i] for i in need], dtype=torch.int64,
+                                 device=dev)
+            vals = torch.empty(len(need), dtype=torch.int32, device=dev)
+            g.gather_i32_at(addrs.data_ptr(), len(need), vals.data_ptr(),
+                            stream)
+            for i, v in zip(need, vals.cpu().tolist()):
+                dl_map[i] = int(v)
+
     # page row starts
     row_starts = np.zeros(len(pages) + 1, dtype=np.int64)
     for i, p in enumerate(pages):
@@ -315,7 +366,8 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
         for i, p in enumerate(pages):
             if p.kind == 0:
                 # v1: def levels = 4-byte len + RLE (bit width 1)
-                (dl,) = struct.unpack_from("<I", p.data, 0)
+                dl = dl_map[i] if p.comp else \
+                    struct.unpack_from("<I", p.data, 0)[0]
                 src = pbase[i] + 4
                 src_len = dl
             else:
@@ -350,7 +402,8 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                                    device=dev)
                 vlen = torch.empty(max(dp.num_values, 1), dtype=torch.int32,
                                    device=dev)
-                sidx_descs += struct.pack(_STRIDX_FMT, base, len(dp.data),
+                sidx_descs += struct.pack(_STRIDX_FMT, base,
+                                          _plen(dict_base_idx + di),
                                           dp.num_values, voff.data_ptr(),
                                           vlen.data_ptr())
                 scount += 1
@@ -365,6 +418,8 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
         if p.kind == 2:
             return p.def_bytes
         if nullable:
+            if p.comp:
+                return 4 + dl_map[i]
             (dl,) = struct.unpack_from("<I", p.data, 0)
             return 4 + dl
         return 0
@@ -373,17 +428,29 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     idx_tensors = {}
     rle_descs = bytearray()
     rle_meta = []
-    for i, p in enumerate(pages):
-        if p.encoding in (ENC_PLAIN_DICTIONARY, ENC_RLE_DICTIONARY):
-            nvals = p.num_values
-            body_off = _body_off(i, p)
-            bw = p.data[body_off]
-            idx = torch.empty(max(nvals, 1), dtype=torch.int32, device=dev)
-            idx_tensors[i] = idx
-            rle_descs += struct.pack(_RLE_FMT, pbase[i] + body_off + 1,
-                                     len(p.data) - body_off - 1,
-                                     idx.data_ptr(), nvals, bw, 1)
-            rle_meta.append(i)
+    dict_pages = [i for i, p in enumerate(pages)
+                  if p.encoding in (ENC_PLAIN_DICTIONARY, ENC_RLE_DICTIONARY)]
+    bw_map = {}
+    comp_dp = [i for i in dict_pages if pages[i].comp]
+    if comp_dp:
+        addrs = torch.tensor(
+            [pbase[i] + _body_off(i, pages[i]) for i in comp_dp],
+            dtype=torch.int64, device=dev)
+        bws = torch.empty(len(comp_dp), dtype=torch.uint8, device=dev)
+        g.gather_u8_at(addrs.data_ptr(), len(comp_dp), bws.data_ptr(), stream)
+        for i, b in zip(comp_dp, bws.cpu().tolist()):
+            bw_map[i] = int(b)
+    for i in dict_pages:
+        p = pages[i]
+        nvals = p.num_values
+        body_off = _body_off(i, p)
+        bw = bw_map[i] if p.comp else p.data[body_off]
+        idx = torch.empty(max(nvals, 1), dtype=torch.int32, device=dev)
+        idx_tensors[i] = idx
+        rle_descs += struct.pack(_RLE_FMT, pbase[i] + body_off + 1,
+                                 _plen(i) - body_off - 1,
+                                 idx.data_ptr(), nvals, bw, 1)
+        rle_meta.append(i)
     if rle_meta:
         rt = torch.frombuffer(rle_descs, dtype=torch.uint8).to(dev)
         g.pq_rle_decode(rt.data_ptr(), len(rle_meta), stream)
@@ -403,7 +470,7 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                 voff = torch.empty(max(nvalid, 1), dtype=torch.int64, device=dev)
                 vlen = torch.empty(max(nvalid, 1), dtype=torch.int32, device=dev)
                 sidx_descs += struct.pack(_STRIDX_FMT, pbase[i] + bo,
-                                          len(p.data) - bo, nvalid,
+                                          _plen(i) - bo, nvalid,
                                           voff.data_ptr(), vlen.data_ptr())
                 plain_meta.append((i, voff, vlen))
         if plain_meta:

@@ -254,6 +254,93 @@ __global__ void def_to_validity_kernel(const uint8_t* __restrict__ def,
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// snappy page decompression (device): one wave per page, all 64 lanes parse
+// the tag stream redundantly in lockstep (identical control flow, no
+// divergence) and cooperate on the literal/match copies. The reference
+// delegates page decompression to nvcomp/libcudf; here it is ~80 lines of
+// CDNA4 HIP. Overlapped matches (offset < wave width) fall back to lane 0.
+// ---------------------------------------------------------------------------
+struct SnapDesc {
+  const uint8_t* src;
+  int64_t src_len;
+  uint8_t* dst;
+  int64_t dst_len;
+};
+
+__global__ void snappy_decomp_kernel(const SnapDesc* __restrict__ descs,
+                                     int32_t n) {
+  int waves_per_block = blockDim.x / WAVE;
+  int wid = blockIdx.x * waves_per_block + (int)(threadIdx.x / WAVE);
+  int lane = threadIdx.x & (WAVE - 1);
+  if (wid >= n) return;
+  SnapDesc d = descs[wid];
+  const uint8_t* p = d.src;
+  const uint8_t* e = d.src + d.src_len;
+  while (p < e && (*p & 0x80)) ++p;  // preamble: uncompressed length varint
+  ++p;
+  uint8_t* o = d.dst;
+  uint8_t* oend = d.dst + d.dst_len;
+  while (p < e && o < oend) {
+    uint8_t tag = *p++;
+    int op = tag & 3;
+    if (op == 0) {  // literal run
+      int64_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        int nb = (int)len - 60;
+        len = 0;
+        for (int i = 0; i < nb; ++i) len |= (int64_t)p[i] << (8 * i);
+        ++len;
+        p += nb;
+      }
+      if (o + len > oend || p + len > e) return;  // corrupt
+      for (int64_t k = lane; k < len; k += WAVE) o[k] = p[k];
+      o += len;
+      p += len;
+    } else {  // back-reference
+      int64_t len, off;
+      if (op == 1) {
+        if (p >= e) return;
+        len = ((tag >> 2) & 7) + 4;
+        off = ((int64_t)(tag >> 5) << 8) | p[0];
+        p += 1;
+      } else if (op == 2) {
+        if (p + 2 > e) return;
+        len = (tag >> 2) + 1;
+        off = (int64_t)p[0] | ((int64_t)p[1] << 8);
+        p += 2;
+      } else {
+        if (p + 4 > e) return;
+        len = (tag >> 2) + 1;
+        off = (int64_t)p[0] | ((int64_t)p[1] << 8) | ((int64_t)p[2] << 16) |
+              ((int64_t)p[3] << 24);
+        p += 4;
+      }
+      if (off <= 0 || o - off < d.dst || o + len > oend) return;  // corrupt
+      const uint8_t* s2 = o - off;
+      if (off >= len) {
+        for (int64_t k = lane; k < len; k += WAVE) o[k] = s2[k];
+      } else if (off == 1) {
+        uint8_t b = s2[0];  // RLE fill
+        for (int64_t k = lane; k < len; k += WAVE) o[k] = b;
+      } else if (lane == 0) {
+        // overlapped pattern: serial on one lane (rare)
+        for (int64_t k = 0; k < len; ++k) o[k] = s2[k];
+      }
+      o += len;
+    }
+  }
+}
+
+// fetch single bytes at arbitrary device addresses (host planning needs the
+// RLE bit-width byte / def-level length inside device-decompressed pages)
+__global__ void gather_u8_at_kernel(const uint64_t* __restrict__ addrs,
+                                    int64_t n, uint8_t* __restrict__ out) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) out[i] = *reinterpret_cast<const uint8_t*>(addrs[i]);
+}
+
 }  // namespace srj
 
 using namespace srj;
@@ -302,6 +389,18 @@ void srj_def_to_validity(const uint8_t* def, int64_t nrows, uint8_t* validity,
                          hipStream_t stream) {
   def_to_validity_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       def, nrows, validity);
+}
+
+void srj_pq_snappy_decomp(const void* descs, int32_t n, hipStream_t stream) {
+  int waves_per_block = DEFAULT_BLOCK / 64;
+  int blocks = (n + waves_per_block - 1) / waves_per_block;
+  snappy_decomp_kernel<<<blocks, DEFAULT_BLOCK, 0, stream>>>(
+      reinterpret_cast<const SnapDesc*>(descs), n);
+}
+
+void srj_gather_u8_at(const uint64_t* addrs, int64_t n, uint8_t* out,
+                      hipStream_t stream) {
+  gather_u8_at_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(addrs, n, out);
 }
 
 }  // extern "C"

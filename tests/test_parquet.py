@@ -163,3 +163,30 @@ def test_footer_rewrite_split_filter(tmp_path):
     assert f.num_rows == 200
     t2 = pq.read_table(dst)
     assert t2.num_rows == 200
+
+
+@pytest.mark.gpu
+def test_decode_snappy_large_plain(tmp_path):
+    """Large snappy pages without dictionary force long literal/match streams
+    through the device decompressor (one wave per page)."""
+    t, oracle = _make_table(200_000)
+    p = str(tmp_path / "snl.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=False,
+                   row_group_size=60_000, data_page_size=256 * 1024)
+    _check(p, oracle)
+
+
+@pytest.mark.gpu
+def test_decode_snappy_repetitive(tmp_path):
+    """Highly repetitive data creates overlapped back-references (offset <
+    length) — the serial-lane fallback path of the snappy kernel."""
+    n = 50_000
+    ints = [7 for _ in range(n)]
+    strs = [("ab" * 30) for _ in range(n)]
+    t = pa.table({"i": pa.array(ints, type=pa.int64()),
+                  "s": pa.array(strs, type=pa.string())})
+    p = str(tmp_path / "snr.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=False)
+    got = srj_pq.read_table(p, device="cuda")
+    assert got.columns[0].to_pylist() == ints
+    assert got.columns[1].to_pylist() == strs
