@@ -144,6 +144,8 @@ def main():
                          "config[3] parquet scan->filter->project")
     ap.add_argument("--groups", type=int, default=1_000_000,
                     help="distinct groups for --op groupby")
+    ap.add_argument("--codec", choices=["none", "snappy"], default="none",
+                    help="parquet codec for --op parquet")
     args = ap.parse_args()
 
     world, rank, local_rank = setup_dist(args)
@@ -271,7 +273,8 @@ def run_secondary(args, world, rank, local_rank):
         from spark_rapids_jni_amd import parquet as srj_pq
         n = args.probe_rows if args.probe_rows != PROBE_ROWS_DEFAULT \
             else 100_000_000
-        path = f"/tmp/bench_store_sales_{rank}.parquet"
+        codec = "SNAPPY" if args.codec == "snappy" else "NONE"
+        path = f"/tmp/bench_store_sales_{rank}_{args.codec}.parquet"
         rng = np.random.default_rng(3)
         log(rank, f"[bench] writing {n}-row store_sales parquet (one-time)")
         papq.write_table(pa.table({
@@ -281,7 +284,7 @@ def run_secondary(args, world, rank, local_rank):
             "ss_quantity": rng.integers(1, 100, n, dtype=np.int64),
             "ss_sales_price": rng.random(n) * 100.0,
             "ss_ext_sales_price": rng.random(n) * 1000.0,
-        }), path, compression="NONE", row_group_size=8_000_000)
+        }), path, compression=codec, row_group_size=8_000_000)
 
         def step():
             t = srj_pq.read_table(path, columns=["ss_sold_date_sk",
@@ -295,7 +298,7 @@ def run_secondary(args, world, rank, local_rank):
             del t, mask, price, s
         metric, model = "parquet_scan_rows_per_sec", \
             "parquet scan->filter->project, store_sales-shaped " \
-            f"({n} rows, 5 cols, uncompressed)"
+            f"({n} rows, 5 cols, {args.codec})"
         rows_per_step = n
     else:  # q3
         from spark_rapids_jni_amd import exec as ex

@@ -164,6 +164,13 @@ def main():
     secs = timeit(lambda: tbl.inner_join(ptbl, out_hint=npr // 4), args.iters)
     emit("generic_join_probe_2col", npr, secs)
 
+    # 5c. radix sort (north-star hot op; reference delegates to cub/cudf)
+    from spark_rapids_jni_amd.ops.sort import sort_pairs_i64
+    skeys = torch.randint(-2**62, 2**62, (n,), dtype=torch.int64, device=dev,
+                          generator=g)
+    secs = timeit(lambda: sort_pairs_i64(skeys), args.iters)
+    emit("radix_sort_int64", n, secs, n * 8)
+
     # 6. parse_uri
     uris = ["https://host%d.example.com:80/p/%d?k=%d&z=9" % (i % 50, i, i)
             for i in range(200_000)]
