@@ -274,6 +274,32 @@ def _flatten(cols: Sequence[Column]):
     return flat, top
 
 
+# pinned staging for descriptor uploads. During hipGraph capture no pinned
+# allocation (hipHostRegister) is permitted, so capture-time descriptors are
+# bump-allocated from a pre-created pinned arena; regions referenced by a
+# captured graph are never recycled (the graph re-reads them at replay).
+_PIN_ARENA = None
+_PIN_POS = 0
+
+
+def _stage_pinned(raw) -> torch.Tensor:
+    global _PIN_ARENA, _PIN_POS
+    host = torch.frombuffer(bytearray(raw) if not isinstance(raw, bytearray)
+                            else raw, dtype=torch.uint8)
+    if not torch.cuda.is_current_stream_capturing():
+        if _PIN_ARENA is None:
+            _PIN_ARENA = torch.empty(1 << 22, dtype=torch.uint8,
+                                     pin_memory=True)
+        return host.pin_memory()
+    n = (len(host) + 255) & ~255
+    assert _PIN_ARENA is not None and _PIN_POS + n <= _PIN_ARENA.numel(), \
+        "pinned arena exhausted during graph capture"
+    view = _PIN_ARENA[_PIN_POS:_PIN_POS + len(host)]
+    view.copy_(host)  # host->pinned memcpy, no stream work
+    _PIN_POS += n
+    return view
+
+
 def pack_descriptors(cols: Sequence[Column], device=None):
     """Pack columns into a device ColDesc array + top-level index array.
 
@@ -290,6 +316,16 @@ def pack_descriptors(cols: Sequence[Column], device=None):
         struct.pack_into(_COLDESC_FMT, raw, i * COLDESC_BYTES,
                          int(c.dtype), c.scale, data_ptr, valid_ptr, offs_ptr,
                          len(c.children), getattr(c, "_child0", 0), c.size)
+    if torch.cuda.is_available() and str(device).startswith("cuda"):
+        # pinned + async: hipGraph-capturable (pageable H2D is not) and
+        # faster on the hot path
+        pinned = _stage_pinned(raw)
+        desc = pinned.to(device, non_blocking=True)
+        traw = bytearray(len(top) * 4)
+        struct.pack_into(f"<{len(top)}i", traw, 0, *top)
+        tpin = _stage_pinned(traw)
+        top_t = tpin.view(torch.int32).to(device, non_blocking=True)
+        return desc, top_t, (flat, pinned, tpin)
     desc_host = torch.frombuffer(raw, dtype=torch.uint8)
     desc = desc_host.to(device)
     top_t = torch.tensor(top, dtype=torch.int32, device=device)
