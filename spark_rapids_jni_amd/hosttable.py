@@ -47,7 +47,8 @@ class HostTable:
                 ent["data"] = (pos, nb, str(c.data.dtype))
                 pos = _align8(pos + nb)
             layout.append(ent)
-        buf = torch.empty(max(pos, 1), dtype=torch.uint8, pin_memory=True)
+        buf = torch.empty(max(pos, 1), dtype=torch.uint8,
+                          pin_memory=torch.cuda.is_available())
         ctx = torch.cuda.stream(stream) if stream else _null_ctx()
         with ctx:
             for c, ent in zip(flat, layout):
