@@ -195,3 +195,37 @@ class RmmSpark:
     @classmethod
     def get_state_of(cls, thread_id: int) -> str:
         return cls.adaptor().get_state_of(thread_id)
+
+
+def with_retry(fn, spill=None, split=None, max_retries: int = 16,
+               thread_id: Optional[int] = None):
+    """Run `fn` under the OOM-retry protocol (the plugin's
+    RmmRapidsRetryIterator.withRetry over this library's state machine):
+
+      * GpuRetryOOM: call `spill()` (e.g. SpillManager.spill_until), wait
+        via blockThreadUntilReady, re-run `fn`.
+      * GpuSplitAndRetryOOM: call `split()` to halve the working set first
+        (raises if no split function was provided).
+    """
+    attempts = 0
+    while True:
+        try:
+            return fn()
+        except GpuSplitAndRetryOOM:
+            if split is None:
+                raise
+            attempts += 1
+            if attempts > max_retries:
+                raise
+            split()
+            continue  # split made room; retry immediately
+        except GpuRetryOOM:
+            attempts += 1
+            if attempts > max_retries:
+                raise
+            if spill is not None:
+                spill()
+            try:
+                RmmSpark.block_thread_until_ready(thread_id)
+            except Exception:
+                pass

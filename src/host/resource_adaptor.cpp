@@ -260,7 +260,9 @@ class SparkResourceAdaptor {
       while (t.state == ThreadState::BUFN) t.cv.wait(lk);
       add_block_time(lk, t, now_ns() - start);
       if (t.state == ThreadState::SPLIT_THROW) {
-        set_state(lk, t, ThreadState::BUFN_WAIT, "split thrown");
+        // the split victim is the designated progress-maker: it halves its
+        // input and retries immediately (doc step 3) — do not re-block it
+        set_state(lk, t, ThreadState::RUNNING, "split thrown");
         bump_task_metric(lk, t, &TaskMetrics::num_split_retry);
         return THROW_GPU_SPLIT;
       }
@@ -386,7 +388,7 @@ class SparkResourceAdaptor {
         while (t.state == ThreadState::BUFN) t.cv.wait(lk);
         add_block_time(lk, t, now_ns() - start);
         if (t.state == ThreadState::SPLIT_THROW) {
-          set_state(lk, t, ThreadState::BUFN_WAIT, "split thrown");
+          set_state(lk, t, ThreadState::RUNNING, "split thrown");
           bump_task_metric(lk, t, &TaskMetrics::num_split_retry);
           return THROW_GPU_SPLIT;
         }
@@ -437,7 +439,7 @@ class SparkResourceAdaptor {
     }
     if (t.split_oom_injected > 0) {
       t.split_oom_injected--;
-      set_state(lk, t, ThreadState::BUFN_WAIT, "injected split OOM");
+      set_state(lk, t, ThreadState::RUNNING, "injected split OOM");
       bump_task_metric(lk, t, &TaskMetrics::num_split_retry);
       return is_gpu ? THROW_GPU_SPLIT : THROW_CPU_SPLIT;
     }
@@ -460,7 +462,7 @@ class SparkResourceAdaptor {
         bump_task_metric(lk, t, &TaskMetrics::num_retry);
         return is_gpu ? THROW_GPU_RETRY : THROW_CPU_RETRY;
       case ThreadState::SPLIT_THROW:
-        set_state(lk, t, ThreadState::BUFN_WAIT, "split OOM thrown");
+        set_state(lk, t, ThreadState::RUNNING, "split OOM thrown");
         bump_task_metric(lk, t, &TaskMetrics::num_split_retry);
         return is_gpu ? THROW_GPU_SPLIT : THROW_CPU_SPLIT;
       case ThreadState::REMOVE_THROW: {

@@ -251,3 +251,42 @@ def test_monte_carlo_cli_harness():
     RmmSpark.clear_event_handler()
     stats = run(6, 30, 2 * 2**20, 2**20 // 2, skewed=True)
     assert stats["done"] == 6 and stats["failed"] == 0
+
+
+def test_with_retry_spills_then_succeeds(rmm):
+    from spark_rapids_jni_amd.memory import with_retry
+    _adaptor(rmm)
+    tid = 41
+    rmm.start_dedicated_task_thread(tid, 9)
+    rmm.force_retry_oom(tid, 1)
+    spills = []
+    halves = []
+
+    def alloc():
+        rmm.alloc(10, thread_id=tid)
+        return "ok"
+
+    # single thread: the retry blocks, the deadlock-breaker escalates the
+    # next attempt to split-and-retry — with_retry rides both transitions
+    got = with_retry(alloc, spill=lambda: spills.append(1),
+                     split=lambda: halves.append(1), thread_id=tid)
+    assert got == "ok"
+    assert len(spills) >= 1
+    rmm.task_done(9)
+
+
+def test_with_retry_split_path(rmm):
+    from spark_rapids_jni_amd.memory import with_retry
+    _adaptor(rmm)
+    tid = 43
+    rmm.start_dedicated_task_thread(tid, 11)
+    rmm.force_split_and_retry_oom(tid, 1)
+    halves = []
+
+    def alloc():
+        rmm.alloc(10, thread_id=tid)
+        return len(halves)
+
+    got = with_retry(alloc, split=lambda: halves.append(1), thread_id=tid)
+    assert got == 1 and len(halves) == 1
+    rmm.task_done(11)
