@@ -57,8 +57,13 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     # (src/gpu/hashtable_i64.hip groupby_i64_kernel)
     i64_fast = (len(kcols) == 1 and kcols[0].validity is None
                 and kcols[0].data is not None
-                and kcols[0].data.dtype == torch.int64
-                and kcols[0].dtype not in (DType.FLOAT64,))
+                and kcols[0].data.dtype in (torch.int64, torch.int32,
+                                            torch.int16, torch.int8)
+                and kcols[0].dtype not in (DType.FLOAT64, DType.FLOAT32))
+    i64_keys = None
+    if i64_fast:
+        i64_keys = (kcols[0].data if kcols[0].data.dtype == torch.int64
+                    else kcols[0].data.to(torch.int64))
     cap_groups = num_groups_hint if num_groups_hint else n
     capacity = max(_next_pow2(min(cap_groups, n) * 2), 64)
     if i64_fast:
@@ -133,7 +138,7 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     out_repr = torch.empty(nstates, dtype=torch.int64, device=dev)
     out_agg = torch.empty(max(naggs, 1) * nstates, dtype=torch.int64, device=dev)
     if i64_fast:
-        g.groupby_i64(kcols[0].data.data_ptr(), n, slots.data_ptr(), capacity,
+        g.groupby_i64(i64_keys.data_ptr(), n, slots.data_ptr(), capacity,
                       agg_desc.data_ptr(), naggs, stream)
         g.groupby_compact_i64(slots.data_ptr(), capacity + 1,
                               agg_desc.data_ptr(), naggs, counter.data_ptr(),

@@ -32,9 +32,27 @@ def _next_pow2(n: int) -> int:
     return p
 
 
+_FAST_KEY_DTYPES = None
+
+
 def _is_i64_fast(cols: List[Column]) -> bool:
+    """Single integer-typed key -> the specialized 16B-slot table (narrower
+    ints are upcast to int64 once; sign extension preserves equality, and
+    real NDS surrogate keys are INT32)."""
+    global _FAST_KEY_DTYPES
     from ..columnar import DType
-    return (len(cols) == 1 and cols[0].dtype == DType.INT64)
+    if _FAST_KEY_DTYPES is None:
+        _FAST_KEY_DTYPES = {DType.INT64, DType.INT32, DType.INT16, DType.INT8,
+                            DType.DATE32, DType.TIMESTAMP_US}
+    return len(cols) == 1 and cols[0].dtype in _FAST_KEY_DTYPES
+
+
+def _as_i64_keys(c: Column) -> Column:
+    """int64 view/upcast of an integer key column (validity shared)."""
+    if c.data.dtype == torch.int64:
+        return c
+    return Column(c.dtype, c.size, c.data.to(torch.int64), c.validity,
+                  null_count=None)
 
 
 class HashJoinTable:
@@ -70,7 +88,7 @@ class HashJoinTable:
         capacity = max(_next_pow2(n * 2), 64)
         if _is_i64_fast(cols) and not force_generic:
             capacity = max(_next_pow2(n * 4), 64)
-            c = cols[0]
+            c = _as_i64_keys(cols[0])
             slots = torch.zeros(capacity * 2, dtype=torch.int64, device=dev)
             g.join_build_i64(c.data.data_ptr(),
                              c.validity.data_ptr() if c.validity is not None else 0,
@@ -97,6 +115,8 @@ class HashJoinTable:
         stream = _native.current_stream()
         dev = pcols[0].device
         fast = self.i64_fast and _is_i64_fast(pcols)
+        if fast:
+            pcols = [_as_i64_keys(pcols[0])]
         counter = torch.zeros(1, dtype=torch.int64, device=dev)
         if not fast:
             bdesc, btop, pdesc, ptop, keep = self._descs(pcols)

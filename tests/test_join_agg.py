@@ -304,3 +304,45 @@ def test_partitioned_probe_matches_direct():
     direct = sorted(zip(pi.cpu().tolist(), bi.cpu().tolist()))
     parted = sorted(zip(out_p.cpu().tolist(), out_b.cpu().tolist()))
     assert direct == parted
+
+
+@pytest.mark.gpu
+def test_int32_keys_take_fast_path():
+    """int32 (NDS surrogate-key dtype) joins route through the specialized
+    table via an int64 upcast and agree with the generic path."""
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    n = 40_000
+    bk = torch.randint(-2**31, 2**31 - 1, (n,), dtype=torch.int32,
+                       device="cuda")
+    pk = torch.cat([bk[:n // 2], torch.randint(-2**31, 2**31 - 1, (n,),
+                                               dtype=torch.int32,
+                                               device="cuda")])
+    bcol = Column.from_torch(bk)
+    pcol = Column.from_torch(pk)
+    tbl = HashJoinTable.build(bcol)
+    assert tbl.i64_fast
+    bi, pi = tbl.inner_join(pcol)
+    gtbl = HashJoinTable.build(bcol, force_generic=True)
+    gbi, gpi = gtbl.inner_join(pcol)
+    assert sorted(zip(pi.cpu().tolist(), bi.cpu().tolist())) == \
+        sorted(zip(gpi.cpu().tolist(), gbi.cpu().tolist()))
+
+
+@pytest.mark.gpu
+def test_groupby_int32_key_fast_path():
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 30_000
+    keys = torch.randint(-1000, 1000, (n,), dtype=torch.int32, device="cuda")
+    vals = torch.randint(0, 100, (n,), dtype=torch.int64, device="cuda")
+    kt, res = groupby(Column.from_torch(keys),
+                      [(Agg.COUNT_ALL, None),
+                       (Agg.SUM, Column.from_torch(vals))])
+    import collections
+    exp_c = collections.Counter(keys.cpu().tolist())
+    exp_s = collections.defaultdict(int)
+    for k, v in zip(keys.cpu().tolist(), vals.cpu().tolist()):
+        exp_s[k] += v
+    got = {k: (c, s) for k, c, s in zip(kt.columns[0].to_pylist(),
+                                        res[0].to_pylist(),
+                                        res[1].to_pylist())}
+    assert got == {k: (exp_c[k], exp_s[k]) for k in exp_c}
