@@ -46,8 +46,10 @@ class HyperLogLogPlusPlus:
     packed 10 x 6 bits per long for Spark interchange
     (reference HyperLogLogPlusPlusHostUDF.java:20-35).
 
-    Known gap: estimate() uses linear counting + raw HLL bias-free formula,
-    without Spark's empirical bias-correction interpolation tables — estimates
+    estimate() uses linear counting + bias-corrected raw HLL; the bias
+    interpolation tables are OUR OWN simulation-generated curves (_hll_bias
+    .py) rather than Spark's paper-appendix tables — sketches/merges stay
+    Spark byte-compatible, estimates
     in the mid-range deviate by up to ~1-2% from Spark's.
     """
 
@@ -103,7 +105,27 @@ class HyperLogLogPlusPlus:
         zeros = sum(1 for r in regs if r == 0)
         if e <= 2.5 * m and zeros > 0:
             return m * math.log(m / zeros)  # linear counting
+        if e <= 5.0 * m:
+            e -= self._bias(e)
         return e
+
+    def _bias(self, raw: float) -> float:
+        """Linear interpolation over the generated bias curve (HLL++-style
+        bias correction; see _hll_bias.py header for provenance)."""
+        from ._hll_bias import BIAS, RAW_ESTIMATE
+        xs = RAW_ESTIMATE.get(self.precision)
+        if xs is None:
+            return 0.0
+        ys = BIAS[self.precision]
+        if raw <= xs[0]:
+            return ys[0]
+        if raw >= xs[-1]:
+            return ys[-1]
+        import bisect
+        i = bisect.bisect_left(xs, raw)
+        x0, x1 = xs[i - 1], xs[i]
+        t = (raw - x0) / (x1 - x0)
+        return ys[i - 1] * (1 - t) + ys[i] * t
 
 
 def create_histogram_if_valid(values: Column, freqs: Column,
