@@ -25,8 +25,30 @@ _KINDS = {"int64": (0, DType.INT64, torch.int64),
 _FIELD_FMT = "<iiQQQQQ"  # field_number, kind, data, valid, lens, offsets, chars
 
 
-def decode(col: Column, schema: List[Tuple[int, str]]) -> Table:
-    """Decode one serialized protobuf message per row into columns."""
+def decode(col: Column, schema) -> Table:
+    """Decode one serialized protobuf message per row into columns.
+
+    schema: list of (field_number, kind); kind is a scalar kind name from
+    _KINDS or ("message", child_schema) — nested messages decode
+    recursively into STRUCT columns (reference Protobuf.java nested-message
+    passes; here the length-delimited blob pass feeds a recursive decode).
+    """
+    def _is_msg(k):
+        return isinstance(k, tuple) and k[0] == "message"
+
+    if any(_is_msg(k) for _, k in schema):
+        flat = [(f, "bytes" if _is_msg(k) else k) for f, k in schema]
+        tbl = decode(col, flat)
+        out = []
+        for (f, k), c in zip(schema, tbl.columns):
+            if _is_msg(k):
+                sub = decode(c, k[1])
+                out.append(Column(DType.STRUCT, c.size, None, c.validity,
+                                  children=list(sub.columns),
+                                  null_count=None))
+            else:
+                out.append(c)
+        return Table(out)
     assert len(schema) <= 64
     g = _native.gpu()
     stream = _native.current_stream()

@@ -77,3 +77,38 @@ def test_protobuf_malformed():
     col = Column.from_pylist(msgs, DType.STRING, "cuda")
     t = protobuf.decode(col, [(1, "int64")])
     assert t.columns[0].to_pylist() == [5, None, None]
+
+
+@pytest.mark.gpu
+def test_protobuf_nested_messages():
+    """Nested message fields decode recursively into STRUCT columns."""
+    def inner_msg(a, s):
+        return encode([(1, "int64", a), (2, "string", s)])
+
+    rows = []
+    exp = []
+    for i in range(500):
+        if i % 11 == 3:
+            rows.append(encode([(3, "int64", i)]))  # nested field absent
+            exp.append((i, None))
+        else:
+            sub = inner_msg(i * 7, f"n{i}")
+            rows.append(encode([(3, "int64", i)]) +
+                        _tag(4, 2) + _varint(len(sub)) + sub)
+            exp.append((i, (i * 7, f"n{i}")))
+    col = Column.from_pylist([bytes(r) for r in rows], DType.STRING, "cuda")
+    from spark_rapids_jni_amd.ops.protobuf import decode
+    tbl = decode(col, [(3, "int64"),
+                       (4, ("message", [(1, "int64"), (2, "string")]))])
+    top = tbl.columns[0].to_pylist()
+    nested = tbl.columns[1]
+    assert nested.dtype == DType.STRUCT
+    got_a = nested.children[0].to_pylist()
+    got_s = nested.children[1].to_pylist()
+    for i, (t, sub) in enumerate(exp):
+        assert top[i] == t
+        if sub is None:
+            assert nested.is_valid_host(i) is False or \
+                (got_a[i] is None and got_s[i] is None)
+        else:
+            assert (got_a[i], got_s[i]) == sub
