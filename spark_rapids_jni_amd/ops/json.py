@@ -212,15 +212,24 @@ def from_json_to_raw_map(col: Column) -> Column:
 
 
 def from_json_to_structs(col: Column, field_names: List[str],
-                         field_types: Optional[List[DType]] = None) -> Table:
+                         field_types: Optional[List] = None) -> Table:
     """Spark from_json to STRUCT (reference from_json_to_structs.cu).
-    Extracts each field via the path engine, then coerces to the requested
-    type with the Spark-exact cast kernels."""
+    All fields are extracted in ONE shared-scan pass (multi-path kernel),
+    then coerced to the requested types with the Spark-exact cast kernels.
+    A field type may be ("struct", child_names, child_types) — nested
+    struct schemas recurse on the extracted sub-object text."""
     from . import cast as cast_ops
+    extracted = get_json_object_multiple_paths(
+        col, [f"$.{name}" for name in field_names])
     cols = []
     for i, name in enumerate(field_names):
-        s = get_json_object(col, f"$.{name}")
+        s = extracted[i]
         dt = field_types[i] if field_types else DType.STRING
+        if isinstance(dt, tuple) and dt[0] == "struct":
+            sub = from_json_to_structs(s, dt[1], dt[2])
+            cols.append(Column(DType.STRUCT, s.size, None, s.validity,
+                               children=list(sub.columns), null_count=None))
+            continue
         if dt == DType.STRING:
             cols.append(s)
         elif dt in (DType.INT8, DType.INT16, DType.INT32, DType.INT64):

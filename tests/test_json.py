@@ -146,3 +146,24 @@ def test_multi_path_overflow_fallback():
     for p, mc in zip(paths, multi):
         sc = get_json_object(col, p)
         assert mc.to_pylist() == sc.to_pylist(), f"path {p}"
+
+
+@pytest.mark.gpu
+def test_from_json_nested_struct():
+    from spark_rapids_jni_amd.ops.json import from_json_to_structs
+    docs = ['{"a":%d,"b":{"x":%d,"y":"s%d"}}' % (i, i * 2, i)
+            if i % 7 != 3 else '{"a":%d}' % i for i in range(300)]
+    col = Column.from_pylist(docs, DType.STRING, "cuda")
+    tbl = from_json_to_structs(
+        col, ["a", "b"],
+        [DType.INT64, ("struct", ["x", "y"], [DType.INT64, DType.STRING])])
+    assert tbl.columns[0].to_pylist() == list(range(300))
+    b = tbl.columns[1]
+    assert b.dtype == DType.STRUCT
+    xs = b.children[0].to_pylist()
+    ys = b.children[1].to_pylist()
+    for i in range(300):
+        if i % 7 == 3:
+            assert not b.is_valid_host(i)
+        else:
+            assert xs[i] == i * 2 and ys[i] == f"s{i}"
