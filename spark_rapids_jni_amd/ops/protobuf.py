@@ -2,8 +2,10 @@
 ProtobufSchemaDescriptor.java flattened field tables).
 
 Schema: list of (field_number, kind) where kind in {"int64","int32","bool",
-"sint64","double","float","string","bytes"}. v1 decodes top-level scalar
-fields; nested messages come out as raw BYTES blobs.
+"sint64","double","float","string","bytes","repeated_int64"} or
+("message", child_schema). Nested messages decode recursively to STRUCT
+columns; repeated_int64 (packed or unpacked varints) decodes to
+LIST<INT64>.
 """
 import struct
 from typing import List, Tuple
@@ -22,7 +24,7 @@ _KINDS = {"int64": (0, DType.INT64, torch.int64),
           "string": (6, DType.STRING, None),
           "bytes": (6, DType.STRING, None)}
 
-_FIELD_FMT = "<iiQQQQQ"  # field_number, kind, data, valid, lens, offsets, chars
+_FIELD_FMT = "<iiiiQQQQQ"  # fnum, kind, rep_slot, pad, data, valid, lens, offsets, chars
 
 
 def decode(col: Column, schema) -> Table:
@@ -57,7 +59,16 @@ def decode(col: Column, schema) -> Table:
     desc, top, keep = pack_descriptors([col])
 
     outs = []
+    nrep = 0
     for fnum, kindname in schema:
+        if kindname == "repeated_int64":
+            assert nrep < 8, "at most 8 repeated fields per message"
+            outs.append({"kind": 7, "dtype": DType.LIST, "fnum": fnum,
+                         "rep_slot": nrep,
+                         "lens": torch.zeros(n, dtype=torch.int32, device=dev),
+                         "valid": make_validity(n, dev, fill_valid=False)})
+            nrep += 1
+            continue
         kind, dt, tdt = _KINDS[kindname]
         if dt == DType.STRING:
             outs.append({"kind": kind, "dtype": dt, "fnum": fnum,
@@ -72,7 +83,7 @@ def decode(col: Column, schema) -> Table:
         raw = bytearray()
         for o in outs:
             raw += struct.pack(
-                _FIELD_FMT, o["fnum"], o["kind"],
+                _FIELD_FMT, o["fnum"], o["kind"], o.get("rep_slot", 0), 0,
                 o["data"].data_ptr() if "data" in o else 0,
                 o["valid"].data_ptr(),
                 o["lens"].data_ptr() if "lens" in o and phase == 0 else 0,
@@ -84,7 +95,7 @@ def decode(col: Column, schema) -> Table:
     ft = pack(0)
     g.pb_decode(desc.data_ptr(), n, ft.data_ptr(), len(outs),
                 row_ok.data_ptr(), 0, stream)
-    # phase 2: allocate string outputs, re-run writing bytes
+    # phase 2: allocate string/list outputs, re-run writing bytes/values
     any_bytes = False
     for o in outs:
         if "lens" in o:
@@ -93,14 +104,24 @@ def decode(col: Column, schema) -> Table:
             torch.cumsum(o["lens"], 0, out=offsets[1:].view(n))
             o["offsets"] = offsets
             nch = int(offsets[-1].item())
-            o["chars"] = torch.empty(max(nch, 1), dtype=torch.uint8, device=dev)
+            if o["kind"] == 7:
+                o["data"] = torch.empty(max(nch, 1), dtype=torch.int64,
+                                        device=dev)
+            else:
+                o["chars"] = torch.empty(max(nch, 1), dtype=torch.uint8,
+                                         device=dev)
     if any_bytes:
         ft2 = pack(1)
         g.pb_decode(desc.data_ptr(), n, ft2.data_ptr(), len(outs),
                     0, 1, stream)
     cols = []
     for o in outs:
-        if "offsets" in o:
+        if o["kind"] == 7:
+            nch = int(o["offsets"][-1].item())
+            child = Column(DType.INT64, nch, o["data"][:max(nch, 1)])
+            cols.append(Column(DType.LIST, n, None, o["valid"],
+                               o["offsets"], [child], null_count=None))
+        elif "offsets" in o:
             nch = int(o["offsets"][-1].item())
             cols.append(Column(DType.STRING, n, o["chars"][:nch], o["valid"],
                                o["offsets"], null_count=None))

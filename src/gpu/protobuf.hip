@@ -21,15 +21,20 @@ enum PbKind : int32_t {
   PB_FIXED64_DOUBLE = 4,
   PB_FIXED32_FLOAT = 5,
   PB_BYTES = 6,          // string/bytes/submessage blob
+  PB_REP_I64 = 7,        // repeated int64 varint (packed or unpacked) -> LIST
 };
+
+constexpr int PB_MAX_REPEATED = 8;
 
 struct PbField {
   int32_t field_number;
   int32_t kind;
+  int32_t rep_slot;  // REP_*: index into the per-row repeated cursors
+  int32_t _pad;
   void* data;        // output fixed-width data (null for BYTES in phase 0)
   uint8_t* valid;    // output validity
-  int32_t* lens;     // BYTES phase 0: per-row length
-  const int32_t* offsets;  // BYTES phase 1: column offsets
+  int32_t* lens;     // BYTES/REP phase 0: per-row length/count
+  const int32_t* offsets;  // BYTES/REP phase 1: column offsets
   char* chars;       // BYTES phase 1
 };
 
@@ -61,6 +66,9 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
     bool valid = in_range && is_valid(in.valid, row);
     // per-field presence this row (up to 64 fields via bitmask)
     uint64_t present = 0;
+    int32_t rcnt[PB_MAX_REPEATED];
+#pragma unroll
+    for (int k = 0; k < PB_MAX_REPEATED; ++k) rcnt[k] = 0;
     bool ok = valid;
     if (valid) {
       StrView s = get_string(in, row);
@@ -83,6 +91,13 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
               PbField& f = fields[fi];
               present |= 1ull << fi;
               switch (f.kind) {
+                case PB_REP_I64: {
+                  int32_t c0 = rcnt[f.rep_slot]++;
+                  if (WRITE_BYTES)
+                    reinterpret_cast<int64_t*>(f.data)[f.offsets[row] + c0] =
+                        (int64_t)v;
+                  break;
+                }
                 case PB_VARINT_I64:
                   reinterpret_cast<int64_t*>(f.data)[row] = (int64_t)v;
                   break;
@@ -140,6 +155,19 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
               } else {
                 f.lens[row] = (int32_t)blen;
               }
+            } else if (fi >= 0 && fields[fi].kind == PB_REP_I64) {
+              // packed repeated varints
+              PbField& f = fields[fi];
+              present |= 1ull << fi;
+              int64_t bpos = pos, bend = pos + (int64_t)blen;
+              while (bpos < bend) {
+                uint64_t v;
+                if (!pb_varint(p, bend, &bpos, &v)) { ok = false; break; }
+                int32_t c0 = rcnt[f.rep_slot]++;
+                if (WRITE_BYTES)
+                  reinterpret_cast<int64_t*>(f.data)[f.offsets[row] + c0] =
+                      (int64_t)v;
+              }
             }
             pos += blen;
             break;
@@ -156,6 +184,8 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
       ballot_write_validity(fields[k].valid, row, fv);
       if (in_range && !fv && fields[k].kind == PB_BYTES && !WRITE_BYTES)
         fields[k].lens[row] = 0;
+      if (in_range && fields[k].kind == PB_REP_I64 && !WRITE_BYTES)
+        fields[k].lens[row] = fv ? rcnt[fields[k].rep_slot] : 0;
     }
   }
 }

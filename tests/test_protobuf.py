@@ -112,3 +112,40 @@ def test_protobuf_nested_messages():
                 (got_a[i] is None and got_s[i] is None)
         else:
             assert (got_a[i], got_s[i]) == sub
+
+
+@pytest.mark.gpu
+def test_protobuf_repeated_int64():
+    """repeated int64 fields (both unpacked tags and packed blobs) decode
+    to LIST<INT64>."""
+    rows = []
+    exp = []
+    for i in range(400):
+        vals = list(range(i % 6))
+        body = b""
+        if i % 3 == 0:
+            # packed: one length-delimited blob of varints
+            blob = b"".join(_varint(v * 3) for v in vals)
+            body += _tag(5, 2) + _varint(len(blob)) + blob
+        else:
+            for v in vals:
+                body += _tag(5, 0) + _varint(v * 3)
+        body += _tag(1, 0) + _varint(i)
+        rows.append(body)
+        exp.append([v * 3 for v in vals] if (vals or i % 3 == 0 and i % 6 == 0)
+                   else ([v * 3 for v in vals] if vals else None))
+    col = Column.from_pylist([bytes(r) for r in rows], DType.STRING, "cuda")
+    from spark_rapids_jni_amd.ops.protobuf import decode
+    tbl = decode(col, [(1, "int64"), (5, "repeated_int64")])
+    assert tbl.columns[0].to_pylist() == list(range(400))
+    lst = tbl.columns[1]
+    got = lst.to_pylist()
+    for i in range(400):
+        vals = [v * 3 for v in range(i % 6)]
+        if i % 3 == 0:
+            # packed blob always present (possibly empty)
+            assert got[i] == vals, i
+        elif vals:
+            assert got[i] == vals, i
+        else:
+            assert got[i] is None, i
