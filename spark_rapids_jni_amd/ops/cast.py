@@ -248,3 +248,13 @@ def to_timestamp_with_format(col: Column, fmt: str, ansi: bool = False,
                           validity.data_ptr(), err_ptr, _native.current_stream())
     _check_err(err, col, ansi)
     return Column(DType.TIMESTAMP_US, n, out, validity, null_count=None)
+
+
+def float_to_decimal(col: Column, precision: int, scale: int,
+                     ansi: bool = False) -> Column:
+    """float/double -> DECIMAL (reference decimal_utils floatingPointToDecimal).
+
+    Spark semantics: Decimal(BigDecimal.valueOf(d)) goes through
+    Double.toString — composing our Ryu shortest-round-trip formatter with
+    the exact string->decimal parser reproduces that bit-for-bit."""
+    return to_decimal(from_floats(col), precision, scale, ansi=ansi)

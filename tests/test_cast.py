@@ -491,3 +491,25 @@ def test_string_to_float32_exact_rounding():
         g = got[i]
         assert st.pack("<f", g) == st.pack("<f", exp), \
             f"{s}: got {g!r} want {exp!r}"
+
+
+@pytest.mark.gpu
+def test_float_to_decimal():
+    """double -> decimal matches Python Decimal(repr(d)) quantization
+    (Spark: Decimal(BigDecimal.valueOf(d)) via Double.toString)."""
+    from decimal import ROUND_HALF_UP, Decimal
+    from spark_rapids_jni_amd.ops.cast import float_to_decimal
+    vals = [0.0, 1.5, -2.25, 123.456, 1e-3, 99999.999, -0.1,
+            3.141592653589793, 2.5, -2.5] + \
+        [random.uniform(-1e6, 1e6) for _ in range(500)]
+    col = Column.from_pylist(vals, DType.FLOAT64, "cuda")
+    got = float_to_decimal(col, 18, 3)
+    out = got.data.cpu().tolist()
+    for i, v in enumerate(vals):
+        exp = Decimal(repr(v)).quantize(Decimal("0.001"),
+                                        rounding=ROUND_HALF_UP)
+        exp_unscaled = int(exp.scaleb(3))
+        if abs(exp_unscaled) >= 10**18:
+            assert not got.is_valid_host(i), i
+        else:
+            assert out[i] == exp_unscaled, (i, v, out[i], exp_unscaled)
