@@ -51,6 +51,11 @@ class SchemaField:
     scale: int = 0
     precision: int = 0
     logical: Optional[dict] = None
+    # LIST<primitive> (parquet 3-level encoding): element leaf + level depths
+    is_list: bool = False
+    element: Optional["SchemaField"] = None
+    max_def: int = 0         # leaf max definition level
+    max_rep: int = 0         # leaf max repetition level
 
 
 @dataclass
@@ -125,14 +130,42 @@ def read_footer(path_or_bytes) -> ParquetFooter:
 
     schema_elems = fmd[2]
     root = schema_elems[0]
-    assert root.get(5, 0) == len(schema_elems) - 1, \
-        "nested parquet schemas not supported yet"
     fields = []
-    for se in schema_elems[1:]:
+    i = 1
+    nchild_root = root.get(5, 0)
+    for _ in range(nchild_root):
+        se = schema_elems[i]
+        nch = se.get(5, 0)
+        if nch == 0:
+            fields.append(SchemaField(
+                name=se[4].decode(), physical_type=se.get(1, -1),
+                repetition=se.get(3, 0), converted_type=se.get(6),
+                scale=se.get(7, 0), precision=se.get(8, 0),
+                logical=se.get(10)))
+            i += 1
+            continue
+        # 3-level LIST group: optional group (LIST) { repeated group list {
+        #   <element leaf> } }
+        assert se.get(6) == 3 or (se.get(10) and 3 in se.get(10, {})), \
+            f"nested non-LIST schema at {se[4].decode()!r} not supported"
+        rep_grp = schema_elems[i + 1]
+        assert rep_grp.get(3, 0) == 2 and rep_grp.get(5, 0) == 1, \
+            "unrecognized LIST encoding"
+        leaf = schema_elems[i + 2]
+        assert leaf.get(5, 0) == 0, "LIST of nested types not supported"
+        elem = SchemaField(
+            name=leaf[4].decode(), physical_type=leaf.get(1, -1),
+            repetition=leaf.get(3, 0), converted_type=leaf.get(6),
+            scale=leaf.get(7, 0), precision=leaf.get(8, 0),
+            logical=leaf.get(10))
+        outer_opt = 1 if se.get(3, 0) == 1 else 0
+        elem_opt = 1 if leaf.get(3, 0) == 1 else 0
         fields.append(SchemaField(
-            name=se[4].decode(), physical_type=se.get(1, -1),
-            repetition=se.get(3, 0), converted_type=se.get(6),
-            scale=se.get(7, 0), precision=se.get(8, 0), logical=se.get(10)))
+            name=se[4].decode(), physical_type=-1, repetition=se.get(3, 0),
+            converted_type=3, is_list=True, element=elem,
+            max_def=outer_opt + 1 + elem_opt, max_rep=1))
+        i += 3
+    assert i == len(schema_elems), "unsupported schema shape"
 
     row_groups = []
     for rg in fmd[4]:
@@ -182,6 +215,7 @@ class _Page:
     num_nulls: int = -1      # v2 only
     comp: bool = False       # True: snappy payload decompressed ON DEVICE
     uncomp: int = 0          # uncompressed size when comp=True
+    rep_bytes: int = 0       # v2: repetition level byte length
 
 
 def _decompress(codec, data, uncompressed_size):
@@ -245,14 +279,15 @@ def _walk_pages(raw, chunk: ColumnChunkMeta,
             nv = dph[1]
             dlen = dph.get(5, 0)
             rlen = dph.get(6, 0)
-            assert rlen == 0, "nested (repeated) columns not supported yet"
+            # v2 layout: [rep levels][def levels][body]
+            lv = rlen + dlen
             if dph.get(7, True) and chunk.codec != CODEC_UNCOMPRESSED:
-                body = _decompress(chunk.codec, payload[dlen:], uncomp - dlen)
-                data = bytes(payload[:dlen]) + bytes(body)
+                body = _decompress(chunk.codec, payload[lv:], uncomp - lv)
+                data = bytes(payload[:lv]) + bytes(body)
             else:
                 data = payload  # levels + body are already contiguous
             pages.append(_Page(2, nv, dph[4], data, def_bytes=dlen,
-                               num_nulls=dph.get(2, -1)))
+                               num_nulls=dph.get(2, -1), rep_bytes=rlen))
             values_seen += nv
         else:
             pass  # index page: skip
@@ -592,6 +627,248 @@ def _mmap_file(path: str) -> memoryview:
     return mv
 
 
+def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
+                      total_rows: int, device) -> Column:
+    """LIST<primitive|string> decode (parquet 3-level encoding).
+
+    Levels (repetition + definition) decode with the same RLE kernel as the
+    flat path; the row structure (list offsets, null/empty lists) is derived
+    from them with torch scans, and the VALUE decode reuses the flat-path
+    kernels verbatim by treating ELEMENT space as row space."""
+    g = _native.gpu()
+    stream = _native.current_stream()
+    dev = torch.device(device)
+    elem = f.element
+    elem_nullable = elem.repetition == 1
+    max_def = f.max_def
+    def_bw = max(max_def.bit_length(), 1)
+
+    pages: List[_Page] = []
+    dict_per_page: List[int] = []
+    dicts: List[_Page] = []
+    for ch in chunks:
+        cur_dict = -1
+        for p in _walk_pages(raw, ch):  # host decompression on this path
+            if p.kind == 1:
+                dicts.append(p)
+                cur_dict = len(dicts) - 1
+            else:
+                pages.append(p)
+                dict_per_page.append(cur_dict)
+
+    # upload payloads
+    blobs = [p.data for p in pages] + [p.data for p in dicts]
+    offs = np.zeros(len(blobs) + 1, dtype=np.int64)
+    for i, b in enumerate(blobs):
+        offs[i + 1] = offs[i] + ((len(b) + 7) & ~7)
+    big = torch.zeros(max(int(offs[-1]), 1), dtype=torch.uint8, device=dev)
+    hbuf = np.zeros(int(offs[-1]), dtype=np.uint8)
+    for i, b in enumerate(blobs):
+        hbuf[offs[i]:offs[i] + len(b)] = np.frombuffer(b, dtype=np.uint8)
+    if len(hbuf):
+        big[:len(hbuf)] = torch.from_numpy(hbuf)
+    pbase = [big.data_ptr() + int(offs[i]) for i in range(len(blobs))]
+    dict_base_idx = len(pages)
+
+    # ---- decode rep + def levels into position space ----------------------
+    pos_starts = np.zeros(len(pages) + 1, dtype=np.int64)
+    for i, p in enumerate(pages):
+        pos_starts[i + 1] = pos_starts[i] + p.num_values
+    npos = int(pos_starts[-1])
+    rep_t = torch.zeros(max(npos, 1), dtype=torch.uint8, device=dev)
+    def_t = torch.zeros(max(npos, 1), dtype=torch.uint8, device=dev)
+    rle_descs = bytearray()
+    nrle = 0
+    body_offs = []
+    for i, p in enumerate(pages):
+        if p.kind == 0:  # v1: [u32 rl][rep][u32 dl][def][body]
+            (rl,) = struct.unpack_from("<I", p.data, 0)
+            rep_src, rep_len = pbase[i] + 4, rl
+            (dl,) = struct.unpack_from("<I", p.data, 4 + rl)
+            def_src, def_len = pbase[i] + 4 + rl + 4, dl
+            body_offs.append(4 + rl + 4 + dl)
+        else:            # v2: [rep][def][body], lengths in the header
+            rep_src, rep_len = pbase[i], p.rep_bytes
+            def_src, def_len = pbase[i] + p.rep_bytes, p.def_bytes
+            body_offs.append(p.rep_bytes + p.def_bytes)
+        rle_descs += struct.pack(_RLE_FMT, rep_src, rep_len,
+                                 rep_t.data_ptr() + int(pos_starts[i]),
+                                 p.num_values, 1, 0)
+        rle_descs += struct.pack(_RLE_FMT, def_src, def_len,
+                                 def_t.data_ptr() + int(pos_starts[i]),
+                                 p.num_values, def_bw, 0)
+        nrle += 2
+    if nrle:
+        rt = torch.frombuffer(rle_descs, dtype=torch.uint8).to(dev)
+        g.pq_rle_decode(rt.data_ptr(), nrle, stream)
+
+    rep64 = rep_t[:npos].to(torch.int64)
+    def64 = def_t[:npos].to(torch.int64)
+    row_mask = rep64 == 0
+    nrows_here = int(row_mask.sum().item())
+    assert nrows_here == total_rows, (nrows_here, total_rows)
+    row_id = torch.cumsum(row_mask.to(torch.int64), 0) - 1
+    elem_slot_def = max_def - (1 if elem_nullable else 0)
+    elem_mask = def64 >= elem_slot_def
+    lengths = torch.bincount(row_id[elem_mask], minlength=total_rows)
+    offsets = torch.zeros(total_rows + 1, dtype=torch.int32, device=dev)
+    if total_rows:
+        offsets[1:] = torch.cumsum(lengths, 0).to(torch.int32)
+    total_elems = int(offsets[-1].item())
+
+    # element-space def (1 = element non-null) + exclusive valid prefix
+    elem_def = (def64[elem_mask] == max_def).to(torch.uint8)
+    if total_elems == 0:
+        elem_def = torch.zeros(1, dtype=torch.uint8, device=dev)
+    incl = torch.cumsum(elem_def.to(torch.int64), 0)
+    elem_vprefix = incl - elem_def.to(torch.int64)
+
+    # per-page element ranges (element start + count)
+    emask_cum = torch.zeros(npos + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(elem_mask.to(torch.int64), 0, out=emask_cum[1:])
+    bounds = emask_cum[torch.from_numpy(pos_starts).to(dev)].cpu().tolist()
+    elem_starts = [int(b) for b in bounds[:-1]]
+    elems_in_page = [int(bounds[k + 1] - bounds[k]) for k in range(len(pages))]
+    vbase = (elem_vprefix[torch.tensor(elem_starts, dtype=torch.int64,
+                                       device=dev)].cpu().tolist()
+             if pages and total_elems else [0] * len(pages))
+
+    def_ptr = elem_def.data_ptr()
+    vprefix_ptr = elem_vprefix.data_ptr()
+
+    # ---- dictionaries -----------------------------------------------------
+    dict_fixed_ptr = {}
+    dict_str = {}
+    if dicts:
+        sidx_descs = bytearray()
+        scount = 0
+        for di, dp in enumerate(dicts):
+            base = pbase[dict_base_idx + di]
+            if elem.physical_type == T_BYTE_ARRAY:
+                voff = torch.empty(max(dp.num_values, 1), dtype=torch.int64,
+                                   device=dev)
+                vlen = torch.empty(max(dp.num_values, 1), dtype=torch.int32,
+                                   device=dev)
+                sidx_descs += struct.pack(_STRIDX_FMT, base, len(dp.data),
+                                          dp.num_values, voff.data_ptr(),
+                                          vlen.data_ptr())
+                scount += 1
+                dict_str[di] = (base, voff, vlen)
+            else:
+                dict_fixed_ptr[di] = base
+        if scount:
+            st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
+            g.pq_string_plain_index(st.data_ptr(), scount, stream)
+
+    # dict-encoded pages: decode indices (one per non-null element)
+    idx_tensors = {}
+    rle_descs = bytearray()
+    rle_meta = []
+    for i, p in enumerate(pages):
+        if p.encoding in (ENC_PLAIN_DICTIONARY, ENC_RLE_DICTIONARY):
+            bo = body_offs[i]
+            bw = p.data[bo]
+            nvals = max(elems_in_page[i], 1)
+            idx = torch.empty(nvals, dtype=torch.int32, device=dev)
+            idx_tensors[i] = idx
+            rle_descs += struct.pack(_RLE_FMT, pbase[i] + bo + 1,
+                                     len(p.data) - bo - 1,
+                                     idx.data_ptr(), elems_in_page[i], bw, 1)
+            rle_meta.append(i)
+    if rle_meta:
+        rt = torch.frombuffer(rle_descs, dtype=torch.uint8).to(dev)
+        g.pq_rle_decode(rt.data_ptr(), len(rle_meta), stream)
+
+    # ---- element values (element space == the flat path's row space) ------
+    elem_validity = None
+    if elem_nullable and total_elems:
+        elem_validity = make_validity(total_elems, dev)
+        g.pq_def_to_validity(def_ptr, total_elems, elem_validity.data_ptr(),
+                             stream)
+
+    if elem.physical_type == T_BYTE_ARRAY:
+        sidx_descs = bytearray()
+        plain_meta = []
+        for i, p in enumerate(pages):
+            if p.encoding == ENC_PLAIN:
+                bo = body_offs[i]
+                nvalid = max(elems_in_page[i], 1)
+                voff = torch.empty(nvalid, dtype=torch.int64, device=dev)
+                vlen = torch.empty(nvalid, dtype=torch.int32, device=dev)
+                sidx_descs += struct.pack(_STRIDX_FMT, pbase[i] + bo,
+                                          len(p.data) - bo, elems_in_page[i],
+                                          voff.data_ptr(), vlen.data_ptr())
+                plain_meta.append((i, voff, vlen))
+        if plain_meta:
+            st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
+            g.pq_string_plain_index(st.data_ptr(), len(plain_meta), stream)
+        pm = {i: (voff, vlen) for i, voff, vlen in plain_meta}
+        cp_descs = bytearray()
+        for i, p in enumerate(pages):
+            if p.encoding == ENC_PLAIN:
+                voff, vlen = pm[i]
+                cp_descs += struct.pack(
+                    _STRCPY_FMT, pbase[i] + body_offs[i], voff.data_ptr(),
+                    vlen.data_ptr(), 0, def_ptr, vprefix_ptr, elem_starts[i],
+                    elems_in_page[i], vbase[i])
+            else:
+                di = dict_per_page[i]
+                base, voff, vlen = dict_str[di]
+                cp_descs += struct.pack(
+                    _STRCPY_FMT, base, voff.data_ptr(), vlen.data_ptr(),
+                    idx_tensors[i].data_ptr(), def_ptr, vprefix_ptr,
+                    elem_starts[i], elems_in_page[i], vbase[i])
+        ct = torch.frombuffer(cp_descs or bytearray(1),
+                              dtype=torch.uint8).to(dev)
+        lens = torch.empty(max(total_elems, 1), dtype=torch.int32, device=dev)
+        g.pq_string_copy(ct.data_ptr(), len(pages), 0, lens.data_ptr(), 0, 0,
+                         stream)
+        soffs = torch.zeros(total_elems + 1, dtype=torch.int32, device=dev)
+        if total_elems:
+            torch.cumsum(lens[:total_elems], 0,
+                         out=soffs[1:].view(total_elems))
+        nchars = int(soffs[-1].item())
+        chars = torch.empty(max(nchars, 1), dtype=torch.uint8, device=dev)
+        g.pq_string_copy(ct.data_ptr(), len(pages), 1, 0, soffs.data_ptr(),
+                         chars.data_ptr(), stream)
+        child = Column(DType.STRING, total_elems, chars[:nchars],
+                       elem_validity, soffs, null_count=None)
+    else:
+        width = _PHYS_WIDTH[elem.physical_type]
+        from .columnar import TORCH_DTYPE
+        edt = _field_dtype(elem)
+        out = torch.zeros(max(total_elems, 1), dtype=TORCH_DTYPE[edt],
+                          device=dev)
+        sc_descs = bytearray()
+        for i, p in enumerate(pages):
+            if p.encoding == ENC_PLAIN:
+                sc_descs += struct.pack(_SCATTER_FMT, pbase[i] + body_offs[i],
+                                        0, def_ptr, vprefix_ptr,
+                                        elem_starts[i], elems_in_page[i],
+                                        vbase[i], width, 0)
+            else:
+                di = dict_per_page[i]
+                sc_descs += struct.pack(_SCATTER_FMT,
+                                        idx_tensors[i].data_ptr(),
+                                        dict_fixed_ptr[di], def_ptr,
+                                        vprefix_ptr, elem_starts[i],
+                                        elems_in_page[i], vbase[i], width, 1)
+        st = torch.frombuffer(sc_descs or bytearray(1),
+                              dtype=torch.uint8).to(dev)
+        g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(), stream)
+        child = Column(edt, total_elems, out[:max(total_elems, 1)],
+                       elem_validity, scale=elem.scale, null_count=None)
+
+    # ---- list validity (null vs empty/populated lists) --------------------
+    validity = None
+    if f.repetition == 1 and total_rows:
+        from .ops.aggregate import _validity_from_bool
+        def_at_row = def64[row_mask]
+        validity = _validity_from_bool(def_at_row >= 1)
+    return Column(DType.LIST, total_rows, None, validity, offsets, [child],
+                  null_count=None)
+
+
 def read_table(path: str, columns: Optional[Sequence[str]] = None,
                device="cuda") -> Table:
     """Scan a parquet file into a GPU Table (footer + page decode).
@@ -608,7 +885,10 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
     cols = []
     for fi, f in enumerate(footer.schema):
         chunks = [rg.columns[fi] for rg in footer.row_groups]
-        cols.append(_read_column(raw, f, chunks, total_rows, device))
+        if f.is_list:
+            cols.append(_read_list_column(raw, f, chunks, total_rows, device))
+        else:
+            cols.append(_read_column(raw, f, chunks, total_rows, device))
     return Table(cols)
 
 

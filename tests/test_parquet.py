@@ -190,3 +190,58 @@ def test_decode_snappy_repetitive(tmp_path):
     got = srj_pq.read_table(p, device="cuda")
     assert got.columns[0].to_pylist() == ints
     assert got.columns[1].to_pylist() == strs
+
+
+def _list_oracle(n, with_nulls=True, strings=False):
+    rows = []
+    for i in range(n):
+        if with_nulls and i % 13 == 4:
+            rows.append(None)
+        elif i % 7 == 2:
+            rows.append([])
+        else:
+            ln = i % 5
+            if strings:
+                rows.append([None if with_nulls and (i + k) % 11 == 5
+                             else f"v{i}-{k}" for k in range(ln)])
+            else:
+                rows.append([None if with_nulls and (i + k) % 11 == 5
+                             else i * 10 + k for k in range(ln)])
+    return rows
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("pagever", ["1.0", "2.0"])
+def test_decode_list_int64(tmp_path, pagever):
+    rows = _list_oracle(4000)
+    t = pa.table({"l": pa.array(rows, type=pa.list_(pa.int64()))})
+    p = str(tmp_path / f"l{pagever}.parquet")
+    pq.write_table(t, p, compression="NONE", use_dictionary=False,
+                   data_page_version=pagever, row_group_size=1500)
+    got = srj_pq.read_table(p, device="cuda")
+    assert got.columns[0].dtype.name == "LIST"
+    assert got.columns[0].to_pylist() == rows
+
+
+@pytest.mark.gpu
+def test_decode_list_strings_dict(tmp_path):
+    rows = _list_oracle(3000, strings=True)
+    t = pa.table({"s": pa.array(rows, type=pa.list_(pa.string()))})
+    p = str(tmp_path / "ls.parquet")
+    pq.write_table(t, p, compression="NONE", use_dictionary=True,
+                   row_group_size=1000)
+    got = srj_pq.read_table(p, device="cuda")
+    assert got.columns[0].to_pylist() == rows
+
+
+@pytest.mark.gpu
+def test_decode_list_mixed_with_flat(tmp_path):
+    rows = _list_oracle(2000)
+    ints = list(range(2000))
+    t = pa.table({"i": pa.array(ints, type=pa.int64()),
+                  "l": pa.array(rows, type=pa.list_(pa.int64()))})
+    p = str(tmp_path / "lm.parquet")
+    pq.write_table(t, p, compression="SNAPPY", row_group_size=700)
+    got = srj_pq.read_table(p, device="cuda")
+    assert got.columns[0].to_pylist() == ints
+    assert got.columns[1].to_pylist() == rows
