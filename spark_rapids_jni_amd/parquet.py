@@ -200,8 +200,14 @@ def _field_dtype(f: SchemaField) -> DType:
         return DType.DATE32
     if f.physical_type == T_INT64 and f.converted_type in (9, 10):
         return DType.TIMESTAMP_US
-    if f.logical and 2 in (f.logical or {}):
-        pass
+    if f.converted_type == 5:  # DECIMAL backed by int32/int64 physical
+        if f.physical_type == T_INT32:
+            return DType.DECIMAL32
+        if f.physical_type == T_INT64:
+            return DType.DECIMAL64
+        raise NotImplementedError(
+            "FIXED_LEN_BYTE_ARRAY decimals not supported (write with "
+            "store_decimal_as_integer / Spark int-backed decimals)")
     return _PHYS_TO_DTYPE[f.physical_type]
 
 
@@ -602,7 +608,7 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
         validity = make_validity(total_rows, dev)
         g.pq_def_to_validity(def_ptr, total_rows, validity.data_ptr(), stream)
     return Column(dtype, total_rows, out[:total_rows], validity,
-                  scale=0, null_count=None)
+                  scale=f.scale, null_count=None)
 
 
 _MMAP_CACHE: Dict[tuple, memoryview] = {}

@@ -245,3 +245,27 @@ def test_decode_list_mixed_with_flat(tmp_path):
     got = srj_pq.read_table(p, device="cuda")
     assert got.columns[0].to_pylist() == ints
     assert got.columns[1].to_pylist() == rows
+
+
+@pytest.mark.gpu
+def test_decode_decimal_int_backed(tmp_path):
+    from decimal import Decimal
+    rows32 = [None if i % 9 == 1 else Decimal(i * 7 - 500).scaleb(-2)
+              for i in range(2000)]
+    rows64 = [None if i % 5 == 2 else Decimal(i * 977 - 10**6).scaleb(-4)
+              for i in range(2000)]
+    t = pa.table({"d32": pa.array(rows32, type=pa.decimal128(7, 2)),
+                  "d64": pa.array(rows64, type=pa.decimal128(15, 4))})
+    p = str(tmp_path / "dec.parquet")
+    pq.write_table(t, p, compression="NONE", store_decimal_as_integer=True)
+    got = srj_pq.read_table(p, device="cuda")
+    d32, d64 = got.columns
+    assert d32.dtype.name == "DECIMAL32" and d32.scale == 2
+    assert d64.dtype.name == "DECIMAL64" and d64.scale == 4
+    g32 = d32.to_pylist()
+    g64 = d64.to_pylist()
+    for i in range(2000):
+        exp = rows32[i]
+        assert g32[i] == (None if exp is None else int(exp.scaleb(2))), i
+        exp = rows64[i]
+        assert g64[i] == (None if exp is None else int(exp.scaleb(4))), i
