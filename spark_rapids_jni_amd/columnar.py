@@ -330,3 +330,89 @@ def pack_descriptors(cols: Sequence[Column], device=None):
     desc = desc_host.to(device)
     top_t = torch.tensor(top, dtype=torch.int32, device=device)
     return desc, top_t, (flat, desc_host)
+
+
+# ---------------------------------------------------------------------------
+# Arrow interop: the Column layout IS Arrow's (validity bitmap LSB-first,
+# int32 string offsets + data buffer), so conversion is buffer re-wrapping,
+# not element-wise work.
+# ---------------------------------------------------------------------------
+
+_ARROW_TYPES = None
+
+
+def _arrow_map():
+    global _ARROW_TYPES
+    if _ARROW_TYPES is None:
+        import pyarrow as pa
+        _ARROW_TYPES = {
+            DType.INT8: pa.int8(), DType.INT16: pa.int16(),
+            DType.INT32: pa.int32(), DType.INT64: pa.int64(),
+            DType.FLOAT32: pa.float32(), DType.FLOAT64: pa.float64(),
+            DType.BOOL8: pa.int8(), DType.DATE32: pa.date32(),
+            DType.TIMESTAMP_US: pa.timestamp("us"),
+            DType.STRING: pa.string(),
+        }
+    return _ARROW_TYPES
+
+
+def from_arrow(arr) -> Column:
+    """pyarrow Array/ChunkedArray -> host Column (zero-copy where the Arrow
+    buffers are already in our layout; bools are unpacked to int8)."""
+    import pyarrow as pa
+    if isinstance(arr, pa.ChunkedArray):
+        arr = arr.combine_chunks()
+    if arr.offset != 0 or isinstance(arr, pa.BooleanArray):
+        # normalize slices / bit-packed bools through pylist (cold path)
+        dt = ({pa.bool_(): DType.BOOL8}.get(arr.type)
+              or next(k for k, v in _arrow_map().items() if v == arr.type))
+        return Column.from_pylist(arr.to_pylist(), dt)
+    import numpy as np
+    bufs = arr.buffers()
+    n = len(arr)
+    validity = None
+    if bufs[0] is not None and arr.null_count:
+        vb = np.frombuffer(bufs[0], dtype=np.uint8)
+        padded = np.zeros(validity_nbytes(n), dtype=np.uint8)
+        padded[:len(vb)] = vb
+        validity = torch.from_numpy(padded)
+    if pa.types.is_string(arr.type) or pa.types.is_binary(arr.type):
+        offs = np.frombuffer(bufs[1], dtype=np.int32)[:n + 1].copy()
+        chars = (np.frombuffer(bufs[2], dtype=np.uint8).copy()
+                 if bufs[2] is not None else np.zeros(0, dtype=np.uint8))
+        return Column(DType.STRING, n, torch.from_numpy(chars), validity,
+                      torch.from_numpy(offs), null_count=arr.null_count)
+    dt = next(k for k, v in _arrow_map().items() if v == arr.type)
+    np_dt = {DType.INT8: np.int8, DType.INT16: np.int16, DType.INT32: np.int32,
+             DType.INT64: np.int64, DType.FLOAT32: np.float32,
+             DType.FLOAT64: np.float64, DType.DATE32: np.int32,
+             DType.TIMESTAMP_US: np.int64}[dt]
+    data = np.frombuffer(bufs[1], dtype=np_dt)[:n].copy()
+    return Column(dt, n, torch.from_numpy(data), validity,
+                  null_count=arr.null_count)
+
+
+def to_arrow(col: Column):
+    """host/device Column -> pyarrow Array (host copy if on GPU)."""
+    import numpy as np
+    import pyarrow as pa
+    c = col.to("cpu")
+    n = c.size
+    at = _arrow_map()[c.dtype]
+    vbuf = None
+    if c.validity is not None:
+        vbuf = pa.py_buffer(c.validity.numpy().tobytes())
+    if c.dtype == DType.STRING:
+        return pa.StringArray.from_buffers(
+            n, pa.py_buffer(c.offsets.numpy().tobytes()),
+            pa.py_buffer((c.data.numpy().tobytes()
+                          if c.data is not None else b"")),
+            vbuf, c.null_count if c.validity is not None else 0)
+    if c.dtype == DType.BOOL8:
+        return pa.array(
+            [None if not c.is_valid_host(i) else bool(int(c.data[i]))
+             for i in range(n)], type=pa.bool_())
+    buf = pa.py_buffer(c.data.numpy().tobytes())
+    return pa.Array.from_buffers(
+        at, n, [vbuf, buf],
+        null_count=c.null_count if c.validity is not None else 0)

@@ -146,3 +146,21 @@ def test_body_alignment():
     assert (t.header.header_len() + t.header.validity_len) % 4 == 0
     # whole record (header + body) ends 4-byte aligned
     assert (t.header.header_len() + t.header.total_len) % 4 == 0
+
+
+def test_arrow_interop_roundtrip():
+    pa = pytest.importorskip("pyarrow")
+    from spark_rapids_jni_amd.columnar import from_arrow, to_arrow
+    cases = [
+        pa.array([1, None, 3, -7], type=pa.int64()),
+        pa.array([1.5, None, -2.25], type=pa.float64()),
+        pa.array(["ab", None, "", "xyz"], type=pa.string()),
+        pa.array(list(range(300)), type=pa.int32()),
+        pa.array([None] * 5, type=pa.int64()),
+    ]
+    for arr in cases:
+        col = from_arrow(arr)
+        assert col.to_pylist() == arr.to_pylist()
+        back = to_arrow(col)
+        assert back.to_pylist() == arr.to_pylist()
+        assert back.type == arr.type
