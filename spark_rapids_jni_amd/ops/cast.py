@@ -258,3 +258,31 @@ def float_to_decimal(col: Column, precision: int, scale: int,
     Double.toString — composing our Ryu shortest-round-trip formatter with
     the exact string->decimal parser reproduces that bit-for-bit."""
     return to_decimal(from_floats(col), precision, scale, ansi=ansi)
+
+
+def format_number(col: Column, d: int) -> Column:
+    """Spark format_number(expr, d) (reference format_float.cu): HALF_EVEN
+    rounding of the shortest decimal representation at d places with
+    thousands separators (java.text.DecimalFormat semantics)."""
+    g = _native.gpu()
+    stream = _native.current_stream()
+    n = col.size
+    dev = col.device
+    data = col.data
+    if col.dtype == DType.FLOAT32:
+        data = data.to(torch.float64)
+    elif col.dtype != DType.FLOAT64:
+        data = data.to(torch.float64)
+    vptr = col.validity.data_ptr() if col.validity is not None else 0
+    lens = torch.empty(n, dtype=torch.int32, device=dev)
+    g.format_number(data.data_ptr(), vptr, n, d, 0, lens.data_ptr(), 0, 0, 0,
+                    stream)
+    offsets = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+    torch.cumsum(lens, 0, out=offsets[1:].view(n))
+    nchars = int(offsets[-1].item())
+    chars = torch.empty(max(nchars, 1), dtype=torch.uint8, device=dev)
+    validity = make_validity(n, dev)
+    g.format_number(data.data_ptr(), vptr, n, d, 1, 0, offsets.data_ptr(),
+                    chars.data_ptr(), validity.data_ptr(), stream)
+    return Column(DType.STRING, n, chars[:nchars], validity, offsets,
+                  null_count=None)

@@ -16,6 +16,9 @@ void srj_string_to_timestamp(const void*, int64_t, int64_t, int64_t, int64_t,
                              int64_t*, uint8_t*, int64_t*, hipStream_t);
 void srj_integer_to_string(const void*, int64_t, int32_t, int32_t*,
                            const int32_t*, char*, uint8_t*, hipStream_t);
+void srj_format_number(const double*, const uint8_t*, int64_t, int32_t,
+                       int32_t, int32_t*, const int32_t*, char*, uint8_t*,
+                       hipStream_t);
 void srj_float_to_string(const void*, const uint8_t*, int64_t, int32_t, int32_t,
                          int32_t*, const int32_t*, char*, uint8_t*, hipStream_t);
 void srj_parse_timestamp_fmt(const void*, int64_t, const void*, int32_t, int64_t,
@@ -79,6 +82,16 @@ void register_cast(py::module_& m) {
                             tz_off, as_ptr<int64_t>(out), as_ptr<uint8_t>(valid),
                             as_ptr<int64_t>(err), as_stream(stream));
     check_hip("parse_timestamp_fmt");
+  });
+  m.def("format_number", [](uintptr_t in, uintptr_t valid, int64_t n,
+                            int32_t d, int32_t phase, uintptr_t lens,
+                            uintptr_t offsets, uintptr_t chars,
+                            uintptr_t out_valid, uintptr_t stream) {
+    srj_format_number(as_ptr<double>(in), as_ptr<uint8_t>(valid), n, d, phase,
+                      as_ptr<int32_t>(lens), as_ptr<int32_t>(offsets),
+                      as_ptr<char>(chars), as_ptr<uint8_t>(out_valid),
+                      as_stream(stream));
+    check_hip("format_number");
   });
   m.def("float_to_string", [](uintptr_t in, uintptr_t valid, int64_t n,
                               int32_t width, int32_t phase, uintptr_t lens,

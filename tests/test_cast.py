@@ -513,3 +513,33 @@ def test_float_to_decimal():
             assert not got.is_valid_host(i), i
         else:
             assert out[i] == exp_unscaled, (i, v, out[i], exp_unscaled)
+
+
+@pytest.mark.gpu
+def test_format_number():
+    """DecimalFormat semantics: HALF_EVEN on the shortest repr + commas."""
+    from decimal import ROUND_HALF_EVEN, Decimal
+    from spark_rapids_jni_amd.ops.cast import format_number
+
+    def oracle(v, d):
+        if v is None:
+            return None
+        if math.isnan(v):
+            return "NaN"
+        if math.isinf(v):
+            return "-Infinity" if v < 0 else "Infinity"
+        q = Decimal(repr(v)).quantize(Decimal(1).scaleb(-d),
+                                      rounding=ROUND_HALF_EVEN)
+        if q == 0:
+            q = abs(q)
+        return f"{q:,.{d}f}"
+
+    vals = [0.0, -0.0, 1.5, 2.675, -2.675, 1234567.891, 0.005, -0.004,
+            1e12 + 0.5, 999.995, -999999.999, 1e-5, None, float("nan"),
+            float("inf"), float("-inf"), 0.125, -0.125] + \
+        [random.uniform(-1e9, 1e9) for _ in range(400)]
+    col = Column.from_pylist(vals, DType.FLOAT64, "cuda")
+    for d in (0, 2, 5):
+        got = format_number(col, d).to_pylist()
+        for v, gv in zip(vals, got):
+            assert gv == oracle(v, d), (v, d, gv, oracle(v, d))
