@@ -269,3 +269,67 @@ def test_decode_decimal_int_backed(tmp_path):
         assert g32[i] == (None if exp is None else int(exp.scaleb(2))), i
         exp = rows64[i]
         assert g64[i] == (None if exp is None else int(exp.scaleb(4))), i
+
+
+def test_thrift_typed_fuzz_roundtrip():
+    """Random typed trees -> compact-protocol write -> typed parse -> equal."""
+    import random as rnd
+    from spark_rapids_jni_amd import _native
+    host = _native.host()
+    r = rnd.Random(271)
+
+    def rand_value(ty, depth):
+        if ty == 1:
+            return r.random() < 0.5
+        if ty == 3:
+            return r.randint(-128, 127)
+        if ty in (4, 5, 6):
+            bits = {4: 15, 5: 31, 6: 63}[ty]
+            return r.randint(-(2**bits), 2**bits - 1)
+        if ty == 7:
+            return r.uniform(-1e18, 1e18)
+        if ty == 8:
+            return bytes(r.randrange(256) for _ in range(r.randrange(20)))
+        if ty == 9:
+            ety = r.choice([3, 5, 6, 8] + ([12] if depth < 2 else []))
+            return (ety, [rand_value(ety, depth + 1)
+                          for _ in range(r.randrange(18))])
+        if ty == 11:
+            kt, vt = r.choice([(5, 8), (8, 6)])
+            return (kt, vt, [(rand_value(kt, depth + 1),
+                              rand_value(vt, depth + 1))
+                             for _ in range(r.randrange(5))])
+        if ty == 12:
+            return rand_struct(depth + 1)
+        raise AssertionError(ty)
+
+    def rand_struct(depth=0):
+        d = {}
+        fid = 0
+        for _ in range(r.randrange(1, 8)):
+            fid += r.randint(1, 40)
+            ty = r.choice([1, 3, 4, 5, 6, 7, 8, 9, 11] +
+                          ([12] if depth < 3 else []))
+            d[fid] = (ty, rand_value(ty, depth))
+        return d
+
+    def norm(d):
+        # dict -> sorted comparable structure (floats exact: same bits)
+        out = {}
+        for k, (ty, v) in d.items():
+            if ty == 12:
+                v = norm(v)
+            elif ty == 9:
+                v = (v[0], [norm(x) if v[0] == 12 else x for x in v[1]])
+            elif ty == 11:
+                # empty maps lose their kv types on the wire (no kv byte)
+                v = (0, 0, []) if not v[2] else (v[0], v[1], list(v[2]))
+            out[k] = (ty, v)
+        return out
+
+    for _ in range(60):
+        tree = rand_struct()
+        raw = host.thrift_write(tree)
+        back, end = host.thrift_parse_typed(raw, 0)
+        assert end == len(raw)
+        assert norm(back) == norm(tree)
