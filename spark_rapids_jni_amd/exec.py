@@ -160,3 +160,18 @@ def q1_like(store_returns: Table, avg_factor: float = 1.2):
     tot = totals.data.to(torch.float64)[pi]
     qual = pi[tot > thresh]
     return gather_column(pair_keys.columns[0], qual)
+
+
+def q14_distinct_like(store_sales: Table, precision: int = 12):
+    """NDS q14/q38-family shape: approximate distinct counting of item keys
+    via the Spark-compatible HLL++ sketch (ops/sketch.py) alongside an exact
+    group-by distinct for validation-sized inputs."""
+    from .ops.sketch import HyperLogLogPlusPlus
+    items = store_sales.columns[1]
+    h = HyperLogLogPlusPlus(precision=precision,
+                            device=str(items.data.device))
+    h.update(items)
+    approx = h.estimate()
+    keys, _ = groupby(items, [(Agg.COUNT_ALL, None)])
+    exact = keys.num_rows
+    return approx, exact

@@ -543,3 +543,45 @@ def test_format_number():
         got = format_number(col, d).to_pylist()
         for v, gv in zip(vals, got):
             assert gv == oracle(v, d), (v, d, gv, oracle(v, d))
+
+
+@pytest.mark.gpu
+def test_float32_to_string_shortest():
+    """Float.toString = shortest round-trip repr of the float32 (Ryu f2d);
+    numpy's float32 repr is the same shortest form."""
+    import numpy as np
+    from spark_rapids_jni_amd.ops.cast import from_floats
+    rng = np.random.default_rng(31)
+    vals = np.concatenate([
+        np.array([0.0, -0.0, 1.0, 0.1, 3.14159, 1e-40, 3.4028235e38,
+                  1.1754944e-38, 1e7, 1e-3, 12345678.0], dtype=np.float32),
+        rng.uniform(-1e6, 1e6, 300).astype(np.float32),
+        (rng.uniform(-1, 1, 200) * 10.0 ** rng.integers(-44, 38, 200))
+        .astype(np.float32)])
+    col = Column.from_torch(torch.from_numpy(vals.copy()).cuda())
+    got = from_floats(col).to_pylist()
+    for v, gv in zip(vals, got):
+        exp = repr(float(np.float32(v)))
+        # Java prints magnitudes >= 1e7 / < 1e-3 in E-notation like
+        # repr does; normalize the exponent spelling
+        f32 = np.float32(v)
+        exp = repr(f32.item()) if False else None
+        # oracle: shortest repr of the float32 value as Java formats it
+        s = np.format_float_positional(f32, unique=True, trim="0")
+        # fall back to scientific for the E-notation range (Java rule)
+        a = abs(float(f32))
+        if a != 0 and (a >= 1e7 or a < 1e-3):
+            s = np.format_float_scientific(f32, unique=True, trim="0")
+            # numpy: '1.e+07' style -> Java: '1.0E7'
+            m, e = s.split("e")
+            if m.endswith("."):
+                m += "0"
+            if "." not in m:
+                m += ".0"
+            s = f"{m}E{int(e)}"
+        else:
+            if s.endswith("."):
+                s += "0"
+            if "." not in s:
+                s += ".0"
+        assert gv == s, (float(f32), gv, s)

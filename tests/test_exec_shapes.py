@@ -75,3 +75,11 @@ def test_q1_like_having():
     exp = sorted(c for (c, s), t in totals.items()
                  if t > 1.2 * (sum(by_store[s]) / len(by_store[s])))
     assert got == exp
+
+
+def test_q14_distinct_like():
+    from spark_rapids_jni_amd.exec import gen_store_sales, q14_distinct_like
+    ss = gen_store_sales(500_000, n_items=18000, device="cuda")
+    approx, exact = q14_distinct_like(ss)
+    assert exact <= 18000
+    assert abs(approx - exact) / exact < 0.05
