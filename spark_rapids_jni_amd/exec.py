@@ -81,15 +81,20 @@ def q3_like(store_sales: Table, date_dim: Table, item: Table, year: int,
     # filter dims
     dd = apply_boolean_mask(date_dim, date_dim.columns[1].data == year)
     it = apply_boolean_mask(item, item.columns[2].data == manufact_id)
-    # join: build on the (small) dims, probe the fact table
+    # join: build on the (small) dims, probe the fact table. Dimension keys
+    # are unique, so matches <= probe rows: pass out_hint to skip the COUNT
+    # probe entirely (the exact-size retry path still guards correctness).
+    nfact = store_sales.num_rows
     dd_tbl = HashJoinTable.build(dd.columns[0])
-    bi, pi = dd_tbl.inner_join(store_sales.columns[0])
-    ss1 = gather(store_sales, pi)
+    bi, pi = dd_tbl.inner_join(store_sales.columns[0], out_hint=nfact)
+    # only the columns the rest of the plan reads survive the gather
+    ss1 = Table([gather_column(store_sales.columns[1], pi),
+                 gather_column(store_sales.columns[3], pi)])
     it_tbl = HashJoinTable.build(it.columns[0])
-    bi2, pi2 = it_tbl.inner_join(ss1.columns[1])
-    ss2 = gather(ss1, pi2)
+    bi2, pi2 = it_tbl.inner_join(ss1.columns[0], out_hint=ss1.num_rows)
+    price = gather_column(ss1.columns[1], pi2)
     brand = gather_column(it.columns[1], bi2.long())
-    keys, results = groupby(brand, [(Agg.SUM, ss2.columns[3]),
+    keys, results = groupby(brand, [(Agg.SUM, price),
                                     (Agg.COUNT_ALL, None)])
     return keys, results
 
