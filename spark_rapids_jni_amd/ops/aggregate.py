@@ -138,8 +138,28 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     out_repr = torch.empty(nstates, dtype=torch.int64, device=dev)
     out_agg = torch.empty(max(naggs, 1) * nstates, dtype=torch.int64, device=dev)
     if i64_fast:
-        g.groupby_i64(i64_keys.data_ptr(), n, slots.data_ptr(), capacity,
-                      agg_desc.data_ptr(), naggs, stream)
+        overflow = torch.zeros(1, dtype=torch.int32, device=dev)
+        lds_ok = (num_groups_hint is not None and num_groups_hint <= 1024
+                  and naggs <= 3)
+        if lds_ok:
+            # low-cardinality: per-workgroup LDS pre-aggregation collapses
+            # per-row global atomics into per-(block x group) merges
+            _IDENT = {0: 0, 1: 0, 2: 0, 3: 0,
+                      4: 2**63 - 1, 5: -2**63,
+                      6: 0x7FF0000000000000,           # +inf bits
+                      7: 0xFFF0000000000000 - 2**64}   # -inf bits
+            idents = torch.tensor([_IDENT[op] for op, _c, _s in native],
+                                  dtype=torch.int64).to(dev)
+            g.groupby_i64_lds(i64_keys.data_ptr(), n, slots.data_ptr(),
+                              capacity, agg_desc.data_ptr(), naggs,
+                              idents.data_ptr(), overflow.data_ptr(), stream)
+        else:
+            g.groupby_i64(i64_keys.data_ptr(), n, slots.data_ptr(), capacity,
+                          agg_desc.data_ptr(), naggs, overflow.data_ptr(),
+                          stream)
+        if int(overflow.item()):
+            # cardinality hint undersized the table: re-run unhinted
+            return groupby(keys, aggs, num_groups_hint=None)
         g.groupby_compact_i64(slots.data_ptr(), capacity + 1,
                               agg_desc.data_ptr(), naggs, counter.data_ptr(),
                               out_repr.data_ptr(), out_agg.data_ptr(), nstates,

@@ -20,7 +20,10 @@ void srj_groupby_compact(const uint64_t*, int64_t, const void*, int32_t, uint64_
 void srj_join_build_i64(const long long*, const uint8_t*, int64_t, void*, int64_t,
                         hipStream_t);
 void srj_groupby_i64(const long long*, int64_t, void*, int64_t, const void*,
-                     int32_t, hipStream_t);
+                     int32_t, int32_t*, hipStream_t);
+void srj_groupby_i64_lds(const long long*, int64_t, void*, int64_t,
+                         const void*, int32_t, const int64_t*, int32_t*,
+                         hipStream_t);
 void srj_groupby_compact_i64(const void*, int64_t, const void*, int32_t,
                              uint64_t*, int64_t*, int64_t*, int64_t,
                              hipStream_t);
@@ -84,11 +87,23 @@ void register_hashtable(py::module_& m) {
         });
   m.def("groupby_i64",
         [](uintptr_t keys, int64_t nrows, uintptr_t slots, int64_t capacity,
-           uintptr_t aggs, int32_t naggs, uintptr_t stream) {
+           uintptr_t aggs, int32_t naggs, uintptr_t overflow,
+           uintptr_t stream) {
           srj_groupby_i64(as_ptr<long long>(keys), nrows, as_ptr<void>(slots),
                           capacity, as_ptr<void>(aggs), naggs,
-                          as_stream(stream));
+                          as_ptr<int32_t>(overflow), as_stream(stream));
           check_hip("groupby_i64");
+        });
+  m.def("groupby_i64_lds",
+        [](uintptr_t keys, int64_t nrows, uintptr_t slots, int64_t capacity,
+           uintptr_t aggs, int32_t naggs, uintptr_t identities,
+           uintptr_t overflow, uintptr_t stream) {
+          srj_groupby_i64_lds(as_ptr<long long>(keys), nrows,
+                              as_ptr<void>(slots), capacity,
+                              as_ptr<void>(aggs), naggs,
+                              as_ptr<int64_t>(identities),
+                              as_ptr<int32_t>(overflow), as_stream(stream));
+          check_hip("groupby_i64_lds");
         });
   m.def("groupby_compact_i64",
         [](uintptr_t slots, int64_t capacity1, uintptr_t aggs, int32_t naggs,

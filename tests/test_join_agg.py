@@ -346,3 +346,34 @@ def test_groupby_int32_key_fast_path():
                                         res[0].to_pylist(),
                                         res[1].to_pylist())}
     assert got == {k: (exp_c[k], exp_s[k]) for k in exp_c}
+
+
+@pytest.mark.gpu
+def test_groupby_lds_low_cardinality():
+    """hinted low-cardinality group-by takes the LDS pre-aggregation kernel;
+    results must match the host oracle exactly (incl. min/max/nullable sum
+    and hint-exceeding cardinality falling through to the global path)."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 200_000
+    for true_groups, hint in ((50, 64), (900, 1000), (5000, 64)):
+        keys = torch.randint(0, true_groups, (n,), dtype=torch.int64,
+                             device="cuda")
+        vals = [None if i % 13 == 5 else (i % 2001) - 1000 for i in range(n)]
+        vc = Column.from_pylist(vals, DType.INT64, "cuda")
+        kt, res = groupby(Column.from_torch(keys),
+                          [(Agg.COUNT_ALL, None), (Agg.SUM, vc),
+                           (Agg.MIN, vc)], num_groups_hint=hint)
+        import collections
+        exp = collections.defaultdict(lambda: [0, None, None])
+        for k, v in zip(keys.cpu().tolist(), vals):
+            e = exp[k]
+            e[0] += 1
+            if v is not None:
+                e[1] = v if e[1] is None else e[1] + v
+                e[2] = v if e[2] is None else min(e[2], v)
+        got = {k: (c, s, mn) for k, c, s, mn in zip(
+            kt.columns[0].to_pylist(), res[0].to_pylist(),
+            res[1].to_pylist(), res[2].to_pylist())}
+        assert len(got) == len(exp), (true_groups, hint)
+        for k, e in exp.items():
+            assert got[k] == tuple(e), (true_groups, hint, k)
