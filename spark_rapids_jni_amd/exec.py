@@ -94,8 +94,11 @@ def q3_like(store_sales: Table, date_dim: Table, item: Table, year: int,
     bi2, pi2 = it_tbl.inner_join(ss1.columns[0], out_hint=ss1.num_rows)
     price = gather_column(ss1.columns[1], pi2)
     brand = gather_column(it.columns[1], bi2.long())
+    # brand cardinality is bounded by the filtered item rows (planner
+    # statistic): a low hint engages the LDS pre-aggregation kernel
     keys, results = groupby(brand, [(Agg.SUM, price),
-                                    (Agg.COUNT_ALL, None)])
+                                    (Agg.COUNT_ALL, None)],
+                            num_groups_hint=min(it.num_rows, 1_000_000))
     return keys, results
 
 
