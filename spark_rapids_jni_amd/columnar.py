@@ -332,6 +332,13 @@ def pack_descriptors(cols: Sequence[Column], device=None):
     return desc, top_t, (flat, desc_host)
 
 
+def validity_to_bool(mask: torch.Tensor, n: int) -> torch.Tensor:
+    """LSB-first validity bitmask -> bool tensor of length n (device op)."""
+    shifts = torch.arange(8, dtype=torch.uint8, device=mask.device)
+    bits = (mask.unsqueeze(1) >> shifts) & 1
+    return bits.reshape(-1)[:n].to(torch.bool)
+
+
 # ---------------------------------------------------------------------------
 # Arrow interop: the Column layout IS Arrow's (validity bitmap LSB-first,
 # int32 string offsets + data buffer), so conversion is buffer re-wrapping,

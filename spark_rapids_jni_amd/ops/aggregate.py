@@ -55,15 +55,55 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     # specialized path: single non-null int64-valued key — 16B {key,row}
     # slots claimed by CAS-on-key, software-pipelined probing
     # (src/gpu/hashtable_i64.hip groupby_i64_kernel)
+    def _int_keyable(c):
+        return (c.data is not None
+                and c.data.dtype in (torch.int64, torch.int32, torch.int16,
+                                     torch.int8)
+                and c.dtype not in (DType.FLOAT64, DType.FLOAT32))
+
     i64_fast = (len(kcols) == 1 and kcols[0].validity is None
-                and kcols[0].data is not None
-                and kcols[0].data.dtype in (torch.int64, torch.int32,
-                                            torch.int16, torch.int8)
-                and kcols[0].dtype not in (DType.FLOAT64, DType.FLOAT32))
+                and _int_keyable(kcols[0]))
     i64_keys = None
     if i64_fast:
         i64_keys = (kcols[0].data if kcols[0].data.dtype == torch.int64
                     else kcols[0].data.to(torch.int64))
+    elif len(kcols) > 1 and all(_int_keyable(c) for c in kcols):
+        # multi-key packing (spark-rapids-style): if the per-column value
+        # ranges multiply into < 2^63, fold the key tuple into ONE int64 and
+        # take the specialized path. Nulls are null-safe-equal group keys:
+        # each column packs as 0 = null, value - min + 1 otherwise. The
+        # output keys are gathered from the ORIGINAL columns by the
+        # representative rows, so packing never leaks into results.
+        widths = []
+        mins = []
+        ok = True
+        for c in kcols:
+            lo, hi = torch.aminmax(c.data)
+            lo, hi = int(lo.item()), int(hi.item())
+            span = hi - lo + 2  # +1 for the null code
+            if span <= 0:
+                ok = False
+                break
+            widths.append(span)
+            mins.append(lo)
+        if ok:
+            total = 1
+            for w in widths:
+                total *= w
+                if total >= 2**62:
+                    ok = False
+                    break
+        if ok:
+            packed = torch.zeros(n, dtype=torch.int64, device=dev)
+            for c, lo, w in zip(kcols, mins, widths):
+                code = (c.data.to(torch.int64) - lo) + 1
+                if c.validity is not None:
+                    from ..columnar import validity_to_bool
+                    code = torch.where(validity_to_bool(c.validity, n), code,
+                                       torch.zeros_like(code))
+                packed = packed * w + code
+            i64_fast = True
+            i64_keys = packed
     cap_groups = num_groups_hint if num_groups_hint else n
     capacity = max(_next_pow2(min(cap_groups, n) * 2), 64)
     if i64_fast:

@@ -258,7 +258,10 @@ __device__ inline int64_t gb64_find_or_claim(Slot64* __restrict__ slots,
   }
   uint64_t sl = slot_of(i64_hash(k), mask);
   Slot64 cur = slots[sl];
-  for (int64_t probes = 0; probes <= (int64_t)mask; ++probes) {
+  // longest linear-probe cluster at <=50% load is O(log n) (~100 at 4B
+  // slots); a chain past 1024 means the table is overloaded (bad hint)
+  int64_t bound = (int64_t)mask < 1024 ? (int64_t)mask : 1024;
+  for (int64_t probes = 0; probes <= bound; ++probes) {
     if (cur.key == k) return (int64_t)sl;
     if (cur.key == GB_EMPTY_KEY) {
       long long prev = atomicCAS(
@@ -307,6 +310,7 @@ __global__ void groupby_i64_lds_kernel(
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        row < nrows; row += stride) {
+    if (*overflow) return;  // undersized hint: host re-runs unhinted anyway
     long long k = keys[row];
     int lidx = -1;
     if (k != GB_EMPTY_KEY) {
@@ -439,6 +443,7 @@ __global__ void groupby_i64_kernel(const long long* __restrict__ keys,
   int64_t nthreads = (int64_t)gridDim.x * blockDim.x;
   int64_t capacity = (int64_t)mask + 1;
   for (int64_t base = tid * PIPE; base < nrows; base += nthreads * PIPE) {
+    if (*overflow) return;  // undersized hint: host re-runs unhinted anyway
     long long k[PIPE];
     uint64_t s[PIPE];
     Slot64 first[PIPE];
@@ -464,7 +469,8 @@ __global__ void groupby_i64_kernel(const long long* __restrict__ keys,
         idx = -1;
         Slot64 cur = first[b];
         uint64_t sl = s[b];
-        for (int64_t probes = 0; probes <= (int64_t)mask; ++probes) {
+        int64_t bound = (int64_t)mask < 1024 ? (int64_t)mask : 1024;
+        for (int64_t probes = 0; probes <= bound; ++probes) {
           if (cur.key == k[b]) { idx = (int64_t)sl; break; }
           if (cur.key == GB_EMPTY_KEY) {
             long long prev = atomicCAS(

@@ -377,3 +377,29 @@ def test_groupby_lds_low_cardinality():
         assert len(got) == len(exp), (true_groups, hint)
         for k, e in exp.items():
             assert got[k] == tuple(e), (true_groups, hint, k)
+
+
+@pytest.mark.gpu
+def test_groupby_multicol_packed_keys():
+    """Narrow multi-int keys pack into one int64 (null-safe) and take the
+    specialized path; results must equal the generic-path oracle exactly."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 50_000
+    k1 = [None if i % 19 == 7 else i % 7 for i in range(n)]
+    k2 = [(i * 31) % 1000 - 500 for i in range(n)]
+    vals = [i % 100 for i in range(n)]
+    c1 = Column.from_pylist(k1, DType.INT32, "cuda")
+    c2 = Column.from_pylist(k2, DType.INT64, "cuda")
+    vc = Column.from_pylist(vals, DType.INT64, "cuda")
+    kt, res = groupby(Table([c1, c2]), [(Agg.COUNT_ALL, None), (Agg.SUM, vc)])
+    import collections
+    exp = collections.defaultdict(lambda: [0, 0])
+    for a, b, v in zip(k1, k2, vals):
+        exp[(a, b)][0] += 1
+        exp[(a, b)][1] += v
+    got = {}
+    g1 = kt.columns[0].to_pylist()
+    g2 = kt.columns[1].to_pylist()
+    for i, (c, sm) in enumerate(zip(res[0].to_pylist(), res[1].to_pylist())):
+        got[(g1[i], g2[i])] = [c, sm]
+    assert got == dict(exp)
