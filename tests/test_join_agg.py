@@ -403,3 +403,31 @@ def test_groupby_multicol_packed_keys():
     for i, (c, sm) in enumerate(zip(res[0].to_pylist(), res[1].to_pylist())):
         got[(g1[i], g2[i])] = [c, sm]
     assert got == dict(exp)
+
+
+@pytest.mark.gpu
+def test_multikey_packed_join():
+    """Multi-int join keys pack into one int64; results must equal the
+    generic path, including null keys (never match) and probe values
+    outside the build-side range."""
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    n = 30_000
+    b1 = [i % 500 for i in range(n)]
+    b2 = [None if i % 23 == 9 else (i * 7) % 40 - 20 for i in range(n)]
+    # first half mirrors build rows (guaranteed matches), second half random
+    # with values outside the build range
+    p1 = b1[:n // 2] + [(i * 3) % 700 for i in range(n // 2)]
+    p2 = b2[:n // 2] + [None if i % 17 == 3 else (i * 5) % 44 - 22
+                        for i in range(n // 2)]
+    bt = Table([Column.from_pylist(b1, DType.INT32, "cuda"),
+                Column.from_pylist(b2, DType.INT64, "cuda")])
+    pt = Table([Column.from_pylist(p1, DType.INT32, "cuda"),
+                Column.from_pylist(p2, DType.INT64, "cuda")])
+    tbl = HashJoinTable.build(bt)
+    assert tbl.i64_fast and getattr(tbl, "_pack", None) is not None
+    bi, pi = tbl.inner_join(pt)
+    gt = HashJoinTable.build(bt, force_generic=True)
+    gbi, gpi = gt.inner_join(pt)
+    assert sorted(zip(pi.cpu().tolist(), bi.cpu().tolist())) == \
+        sorted(zip(gpi.cpu().tolist(), gbi.cpu().tolist()))
+    assert bi.numel() > 0
