@@ -205,10 +205,13 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
                               out_repr.data_ptr(), out_agg.data_ptr(), nstates,
                               stream)
     else:
+        overflow = torch.zeros(1, dtype=torch.int32, device=dev)
         kdesc, ktop, keep = pack_descriptors(kcols)
         g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n,
                   slots.data_ptr(), capacity, agg_desc.data_ptr(), naggs,
-                  stream)
+                  overflow.data_ptr(), stream)
+        if int(overflow.item()):
+            return groupby(keys, aggs, num_groups_hint=None)
         g.groupby_compact(slots.data_ptr(), capacity, agg_desc.data_ptr(),
                           naggs, counter.data_ptr(), out_repr.data_ptr(),
                           out_agg.data_ptr(), nstates, stream)

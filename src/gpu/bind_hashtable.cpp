@@ -14,7 +14,7 @@ void srj_join_semi(const void*, const int32_t*, const void*, const int32_t*,
                    int32_t, int64_t, const uint64_t*, int64_t, uint64_t*,
                    int64_t*, int64_t, int32_t, hipStream_t);
 void srj_groupby(const void*, const int32_t*, int32_t, int64_t, uint64_t*, int64_t,
-                 const void*, int32_t, hipStream_t);
+                 const void*, int32_t, int32_t*, hipStream_t);
 void srj_groupby_compact(const uint64_t*, int64_t, const void*, int32_t, uint64_t*,
                          int64_t*, int64_t*, int64_t, hipStream_t);
 void srj_join_build_i64(const long long*, const uint8_t*, int64_t, void*, int64_t,
@@ -148,10 +148,10 @@ void register_hashtable(py::module_& m) {
   m.def("groupby",
         [](uintptr_t cols, uintptr_t top, int32_t ntop, int64_t nrows,
            uintptr_t slots, int64_t capacity, uintptr_t aggs, int32_t naggs,
-           uintptr_t stream) {
+           uintptr_t overflow, uintptr_t stream) {
           srj_groupby(as_ptr<void>(cols), as_ptr<int32_t>(top), ntop, nrows,
                       as_ptr<uint64_t>(slots), capacity, as_ptr<void>(aggs), naggs,
-                      as_stream(stream));
+                      as_ptr<int32_t>(overflow), as_stream(stream));
           check_hip("groupby");
         });
   m.def("groupby_compact",

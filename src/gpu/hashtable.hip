@@ -174,27 +174,38 @@ __global__ void groupby_kernel(const ColDesc* __restrict__ cols,
                                const int32_t* __restrict__ top, int32_t ntop,
                                int64_t nrows, uint64_t* __restrict__ slots,
                                uint64_t mask, const AggDesc* __restrict__ aggs,
-                               int32_t naggs) {
+                               int32_t naggs, int32_t* __restrict__ overflow) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < nrows;
        row += stride) {
+    if (*overflow) return;  // undersized hint: host re-runs unhinted
     uint64_t h = row_hash64(cols, top, ntop, row);
     uint32_t fp = fingerprint(h);
     uint64_t packed = ((uint64_t)fp << 32) | (uint64_t)(row + 1);
     uint64_t s = h & mask;
-    // insert-or-find
-    while (true) {
+    // insert-or-find, probe-bounded (a chain past 1024 means the table is
+    // overloaded from a bad cardinality hint -- flag and bail)
+    int64_t bound = (int64_t)mask < 1024 ? (int64_t)mask : 1024;
+    bool found = false;
+    for (int64_t probes = 0; probes <= bound; ++probes) {
       uint64_t word = slots[s];
       if (word == 0) {
         word = atomicCAS(reinterpret_cast<unsigned long long*>(slots + s), 0ull,
                          (unsigned long long)packed);
-        if (word == 0) break;  // claimed: this row is the representative
+        if (word == 0) { found = true; break; }
       }
       if ((uint32_t)(word >> 32) == fp) {
         int64_t repr = (int64_t)(word & 0xffffffffu) - 1;
-        if (repr == row || rows_equal(cols, top, ntop, repr, cols, top, row)) break;
+        if (repr == row || rows_equal(cols, top, ntop, repr, cols, top, row)) {
+          found = true;
+          break;
+        }
       }
       s = (s + 1) & mask;
+    }
+    if (!found) {
+      atomicOr(overflow, 1);
+      continue;
     }
     agg_accumulate(aggs, naggs, row, (int64_t)s);
   }
@@ -293,10 +304,12 @@ void srj_join_semi(const void* bcols, const int32_t* btop, const void* pcols,
 
 void srj_groupby(const void* cols, const int32_t* top, int32_t ntop,
                  int64_t nrows, uint64_t* slots, int64_t capacity,
-                 const void* aggs, int32_t naggs, hipStream_t stream) {
+                 const void* aggs, int32_t naggs, int32_t* overflow,
+                 hipStream_t stream) {
   groupby_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       reinterpret_cast<const ColDesc*>(cols), top, ntop, nrows, slots,
-      (uint64_t)(capacity - 1), reinterpret_cast<const AggDesc*>(aggs), naggs);
+      (uint64_t)(capacity - 1), reinterpret_cast<const AggDesc*>(aggs), naggs,
+      overflow);
 }
 
 void srj_groupby_compact(const uint64_t* slots, int64_t capacity, const void* aggs,

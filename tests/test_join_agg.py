@@ -431,3 +431,22 @@ def test_multikey_packed_join():
     assert sorted(zip(pi.cpu().tolist(), bi.cpu().tolist())) == \
         sorted(zip(gpi.cpu().tolist(), gbi.cpu().tolist()))
     assert bi.numel() > 0
+
+
+@pytest.mark.gpu
+def test_groupby_bad_hint_recovers():
+    """A cardinality hint far below the true group count must not hang or
+    miscount — both specialized and generic paths flag overflow and re-run."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 300_000
+    # int64 path
+    keys = torch.randint(0, 50_000, (n,), dtype=torch.int64, device="cuda")
+    kt, res = groupby(Column.from_torch(keys), [(Agg.COUNT_ALL, None)],
+                      num_groups_hint=16)
+    assert int(res[0].data.sum().item()) == n
+    # generic (string-keyed) path
+    svals = [f"k{i % 20000}" for i in range(n)]
+    sc = Column.from_pylist(svals, DType.STRING, "cuda")
+    kt2, res2 = groupby(sc, [(Agg.COUNT_ALL, None)], num_groups_hint=16)
+    assert int(res2[0].data.sum().item()) == n
+    assert kt2.num_rows == 20000
