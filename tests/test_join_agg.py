@@ -450,3 +450,26 @@ def test_groupby_bad_hint_recovers():
     kt2, res2 = groupby(sc, [(Agg.COUNT_ALL, None)], num_groups_hint=16)
     assert int(res2[0].data.sum().item()) == n
     assert kt2.num_rows == 20000
+
+
+@pytest.mark.gpu
+def test_groupby_multicol_wide_range_generic():
+    """Key ranges whose product overflows the int64 packing domain must stay
+    on the generic representative-row path and still be exact."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    n = 20_000
+    k1 = torch.randint(-2**60, 2**60, (n,), dtype=torch.int64,
+                       device="cuda") % 97
+    k2 = torch.randint(-2**60, 2**60, (n,), dtype=torch.int64, device="cuda")
+    # k2 spans ~2^61 values -> range product overflows -> generic path
+    kt, res = groupby(Table([Column.from_torch(k1), Column.from_torch(k2)]),
+                      [(Agg.COUNT_ALL, None)])
+    assert int(res[0].data.sum().item()) == n
+    import collections
+    exp = collections.Counter(zip(k1.cpu().tolist(), k2.cpu().tolist()))
+    got = collections.Counter()
+    g1 = kt.columns[0].to_pylist()
+    g2 = kt.columns[1].to_pylist()
+    for i, c in enumerate(res[0].to_pylist()):
+        got[(g1[i], g2[i])] = c
+    assert got == exp
