@@ -282,14 +282,21 @@ _PIN_ARENA = None
 _PIN_POS = 0
 
 
+def ensure_pinned_arena() -> None:
+    """Create the capture-time pinned staging arena. MUST be called before
+    hipGraph capture (pinned allocation is illegal mid-capture);
+    graphs.CapturedPipeline does this automatically."""
+    global _PIN_ARENA
+    if _PIN_ARENA is None:
+        _PIN_ARENA = torch.empty(1 << 22, dtype=torch.uint8, pin_memory=True)
+
+
 def _stage_pinned(raw) -> torch.Tensor:
     global _PIN_ARENA, _PIN_POS
     host = torch.frombuffer(bytearray(raw) if not isinstance(raw, bytearray)
                             else raw, dtype=torch.uint8)
     if not torch.cuda.is_current_stream_capturing():
-        if _PIN_ARENA is None:
-            _PIN_ARENA = torch.empty(1 << 22, dtype=torch.uint8,
-                                     pin_memory=True)
+        ensure_pinned_arena()
         return host.pin_memory()
     n = (len(host) + 255) & ~255
     assert _PIN_ARENA is not None and _PIN_POS + n <= _PIN_ARENA.numel(), \
@@ -316,15 +323,17 @@ def pack_descriptors(cols: Sequence[Column], device=None):
         struct.pack_into(_COLDESC_FMT, raw, i * COLDESC_BYTES,
                          int(c.dtype), c.scale, data_ptr, valid_ptr, offs_ptr,
                          len(c.children), getattr(c, "_child0", 0), c.size)
-    if torch.cuda.is_available() and str(device).startswith("cuda"):
-        # pinned + async: hipGraph-capturable (pageable H2D is not) and
-        # faster on the hot path
+    if (torch.cuda.is_available() and str(device).startswith("cuda")
+            and torch.cuda.is_current_stream_capturing()):
+        # hipGraph capture: pageable H2D is not capturable — stage through
+        # the pre-created pinned arena (regions live as long as the graph)
         pinned = _stage_pinned(raw)
         desc = pinned.to(device, non_blocking=True)
-        traw = bytearray(len(top) * 4)
+        traw = bytearray(max(len(top), 1) * 4)
         struct.pack_into(f"<{len(top)}i", traw, 0, *top)
         tpin = _stage_pinned(traw)
-        top_t = tpin.view(torch.int32).to(device, non_blocking=True)
+        top_t = tpin.view(torch.int32)[:len(top)].to(device,
+                                                     non_blocking=True)
         return desc, top_t, (flat, pinned, tpin)
     desc_host = torch.frombuffer(raw, dtype=torch.uint8)
     desc = desc_host.to(device)

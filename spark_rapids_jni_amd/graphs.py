@@ -24,6 +24,8 @@ class CapturedPipeline:
 
     def __init__(self, fn: Callable, static_inputs: Sequence[torch.Tensor],
                  warmup: int = 3):
+        from .columnar import ensure_pinned_arena
+        ensure_pinned_arena()  # pinned allocation is illegal mid-capture
         self.inputs = list(static_inputs)
         self.fn = fn
         # warm up on a side stream so allocations settle before capture
