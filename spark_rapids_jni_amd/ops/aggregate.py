@@ -67,7 +67,7 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     if i64_fast:
         i64_keys = (kcols[0].data if kcols[0].data.dtype == torch.int64
                     else kcols[0].data.to(torch.int64))
-    elif len(kcols) > 1 and all(_int_keyable(c) for c in kcols):
+    elif n > 0 and len(kcols) > 1 and all(_int_keyable(c) for c in kcols):
         # multi-key packing (spark-rapids-style): if the per-column value
         # ranges multiply into < 2^63, fold the key tuple into ONE int64 and
         # take the specialized path. Nulls are null-safe-equal group keys:

@@ -67,6 +67,8 @@ def _int_packable(cols: List[Column]) -> bool:
 def _pack_ranges(cols: List[Column]):
     """(mins, widths) for folding an int-key tuple into one int64, or None
     if the range product overflows."""
+    if cols[0].size == 0:
+        return None  # nothing to scan; generic path handles empties
     mins, widths = [], []
     total = 1
     for c in cols:

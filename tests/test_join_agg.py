@@ -473,3 +473,23 @@ def test_groupby_multicol_wide_range_generic():
     for i, c in enumerate(res[0].to_pylist()):
         got[(g1[i], g2[i])] = c
     assert got == exp
+
+
+@pytest.mark.gpu
+def test_empty_inputs_all_paths():
+    """0-row build/probe/group-by inputs on every dispatch path."""
+    from spark_rapids_jni_amd.ops.aggregate import Agg, groupby
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    e64 = Column.from_torch(torch.empty(0, dtype=torch.int64, device="cuda"))
+    e32 = Column.from_torch(torch.empty(0, dtype=torch.int32, device="cuda"))
+    tbl = HashJoinTable.build(e64)
+    bi, pi = tbl.inner_join(Column.from_torch(
+        torch.arange(10, dtype=torch.int64, device="cuda")))
+    assert bi.numel() == 0
+    tbl2 = HashJoinTable.build(Table([e32, e64]))
+    bi2, pi2 = tbl2.inner_join(Table([e32, e64]))
+    assert bi2.numel() == 0
+    kt, res = groupby(Table([e32, e64]), [(Agg.COUNT_ALL, None)])
+    assert kt.num_rows == 0
+    kt2, res2 = groupby(e64, [(Agg.COUNT_ALL, None)], num_groups_hint=8)
+    assert kt2.num_rows == 0
