@@ -310,7 +310,9 @@ __global__ void groupby_i64_lds_kernel(
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        row < nrows; row += stride) {
-    if (*overflow) return;  // undersized hint: host re-runs unhinted anyway
+    // on overflow: BREAK (not return) so every wave still reaches the
+    // __syncthreads() before the flush — the host re-runs unhinted anyway
+    if (*overflow) break;
     long long k = keys[row];
     int lidx = -1;
     if (k != GB_EMPTY_KEY) {
