@@ -233,3 +233,59 @@ def test_deep_nested_device_roundtrip():
         buf, sizes, [outer.to("cpu")])
     permh = perm.cpu().tolist()
     assert got.columns[0].to_pylist() == [exp[i] for i in permh]
+
+
+@pytest.mark.gpu
+def test_device_kudo_fuzz_roundtrip():
+    """Random nested schemas + random partition maps through the DEVICE
+    split/assemble pair."""
+    import random as rnd
+    r = rnd.Random(311)
+    from spark_rapids_jni_amd import shuffle_gpu
+    from spark_rapids_jni_amd.columnar import validity_from_bools
+
+    def rand_col(n, depth=0):
+        t = r.choice(["i64", "str", "f64"] +
+                     (["struct", "list"] if depth < 2 else []))
+        if t == "i64":
+            vals = [None if r.random() < 0.15 else r.randint(-10**9, 10**9)
+                    for _ in range(n)]
+            return Column.from_pylist(vals, DType.INT64, "cuda")
+        if t == "f64":
+            vals = [None if r.random() < 0.1 else r.random() * 100
+                    for _ in range(n)]
+            return Column.from_pylist(vals, DType.FLOAT64, "cuda")
+        if t == "str":
+            vals = [None if r.random() < 0.2 else "x" * r.randrange(5)
+                    for _ in range(n)]
+            return Column.from_pylist(vals, DType.STRING, "cuda")
+        if t == "struct":
+            a = rand_col(n, depth + 1)
+            b = rand_col(n, depth + 1)
+            vmask = validity_from_bools([r.random() > 0.1 for _ in range(n)],
+                                        "cuda")
+            return Column(DType.STRUCT, n, None, vmask, None, [a, b],
+                          null_count=None)
+        offs = [0]
+        total = 0
+        for _ in range(n):
+            total += r.randrange(3)
+            offs.append(total)
+        child = rand_col(total, depth + 1)
+        return Column(DType.LIST, n, None, None,
+                      torch.tensor(offs, dtype=torch.int32, device="cuda"),
+                      [child])
+
+    for trial in range(8):
+        n = r.randrange(1, 400)
+        tbl = Table([rand_col(n) for _ in range(r.randrange(1, 4))])
+        exp = [c.to_pylist() for c in tbl.columns]
+        nparts = r.randrange(1, 5)
+        pids_h, offsets, perm = _partition(n, nparts, "cuda")
+        buf, sizes = shuffle_gpu.split_and_serialize_to_device(tbl, offsets,
+                                                               perm)
+        got = shuffle_gpu.assemble_from_device_raw(
+            buf, sizes, [c.to("cpu") for c in tbl.columns])
+        permh = perm.cpu().tolist()
+        for gc, e in zip(got.columns, exp):
+            assert gc.to_pylist() == [e[i] for i in permh], f"trial {trial}"
