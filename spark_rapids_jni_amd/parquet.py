@@ -56,6 +56,9 @@ class SchemaField:
     element: Optional["SchemaField"] = None
     max_def: int = 0         # leaf max definition level
     max_rep: int = 0         # leaf max repetition level
+    # flat STRUCT group: leaf fields (each mapping to one column chunk)
+    is_struct: bool = False
+    children: Optional[list] = None
 
 
 @dataclass
@@ -144,10 +147,28 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                 logical=se.get(10)))
             i += 1
             continue
+        if not (se.get(6) == 3 or (se.get(10) and 3 in se.get(10, {}))):
+            # plain group = STRUCT of leaves (non-repeated)
+            outer_opt = 1 if se.get(3, 0) == 1 else 0
+            leaves = []
+            for k in range(nch):
+                lf = schema_elems[i + 1 + k]
+                assert lf.get(5, 0) == 0, \
+                    "nested struct-of-struct not supported"
+                leaves.append(SchemaField(
+                    name=lf[4].decode(), physical_type=lf.get(1, -1),
+                    repetition=lf.get(3, 0), converted_type=lf.get(6),
+                    scale=lf.get(7, 0), precision=lf.get(8, 0),
+                    logical=lf.get(10),
+                    max_def=outer_opt + (1 if lf.get(3, 0) == 1 else 0)))
+            fields.append(SchemaField(
+                name=se[4].decode(), physical_type=-1,
+                repetition=se.get(3, 0), converted_type=None, is_struct=True,
+                children=leaves))
+            i += 1 + nch
+            continue
         # 3-level LIST group: optional group (LIST) { repeated group list {
         #   <element leaf> } }
-        assert se.get(6) == 3 or (se.get(10) and 3 in se.get(10, {})), \
-            f"nested non-LIST schema at {se[4].decode()!r} not supported"
         rep_grp = schema_elems[i + 1]
         assert rep_grp.get(3, 0) == 2 and rep_grp.get(5, 0) == 1, \
             "unrecognized LIST encoding"
@@ -875,6 +896,226 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                   null_count=None)
 
 
+def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
+                        total_rows: int, device) -> Column:
+    """Flat STRUCT<primitive|string> decode: every leaf is a plain column
+    whose definition levels carry BOTH the struct's and the leaf's
+    nullability (position space == row space, no repetition). Values reuse
+    the flat-path kernels; the struct validity falls out of the first
+    leaf's levels (def >= 1 means the struct is present)."""
+    g = _native.gpu()
+    stream = _native.current_stream()
+    dev = torch.device(device)
+    outer_opt = f.repetition == 1
+    children = []
+    struct_validity = None
+
+    for li, leaf in enumerate(f.children):
+        chunks = [rg.columns[leaf0 + li] for rg in row_groups]
+        max_def = leaf.max_def
+        def_bw = max(max_def.bit_length(), 1)
+        pages: List[_Page] = []
+        dict_per_page: List[int] = []
+        dicts: List[_Page] = []
+        for ch in chunks:
+            cur_dict = -1
+            for p in _walk_pages(raw, ch):
+                if p.kind == 1:
+                    dicts.append(p)
+                    cur_dict = len(dicts) - 1
+                else:
+                    pages.append(p)
+                    dict_per_page.append(cur_dict)
+        blobs = [p.data for p in pages] + [p.data for p in dicts]
+        offs = np.zeros(len(blobs) + 1, dtype=np.int64)
+        for i, b in enumerate(blobs):
+            offs[i + 1] = offs[i] + ((len(b) + 7) & ~7)
+        big = torch.zeros(max(int(offs[-1]), 1), dtype=torch.uint8, device=dev)
+        hbuf = np.zeros(int(offs[-1]), dtype=np.uint8)
+        for i, b in enumerate(blobs):
+            hbuf[offs[i]:offs[i] + len(b)] = np.frombuffer(b, dtype=np.uint8)
+        if len(hbuf):
+            big[:len(hbuf)] = torch.from_numpy(hbuf)
+        pbase = [big.data_ptr() + int(offs[i]) for i in range(len(blobs))]
+        dict_base_idx = len(pages)
+
+        row_starts = np.zeros(len(pages) + 1, dtype=np.int64)
+        for i, p in enumerate(pages):
+            row_starts[i + 1] = row_starts[i] + p.num_values
+        assert row_starts[-1] == total_rows
+
+        # decode definition levels (bit width up to 2) into row space
+        lev_t = torch.zeros(max(total_rows, 1), dtype=torch.uint8, device=dev)
+        body_offs = []
+        if max_def > 0:
+            rle_descs = bytearray()
+            for i, p in enumerate(pages):
+                if p.kind == 0:
+                    (dl,) = struct.unpack_from("<I", p.data, 0)
+                    src, src_len = pbase[i] + 4, dl
+                    body_offs.append(4 + dl)
+                else:
+                    src, src_len = pbase[i], p.def_bytes
+                    body_offs.append(p.rep_bytes + p.def_bytes)
+                rle_descs += struct.pack(_RLE_FMT, src, src_len,
+                                         lev_t.data_ptr() + int(row_starts[i]),
+                                         p.num_values, def_bw, 0)
+            rt = torch.frombuffer(rle_descs or bytearray(1),
+                                  dtype=torch.uint8).to(dev)
+            g.pq_rle_decode(rt.data_ptr(), len(pages), stream)
+        else:
+            body_offs = [0] * len(pages)
+            lev_t.fill_(0)
+
+        if struct_validity is None and outer_opt:
+            from .ops.aggregate import _validity_from_bool
+            struct_validity = _validity_from_bool(
+                lev_t[:total_rows].to(torch.int64) >= 1)
+
+        leaf_def = (lev_t[:max(total_rows, 1)].to(torch.int64) ==
+                    max_def).to(torch.uint8) if max_def > 0 else \
+            torch.ones(max(total_rows, 1), dtype=torch.uint8, device=dev)
+        incl = torch.cumsum(leaf_def.to(torch.int64), 0)
+        vprefix = incl - leaf_def.to(torch.int64)
+        def_ptr = leaf_def.data_ptr()
+        vprefix_ptr = vprefix.data_ptr()
+        vbase = (vprefix[torch.from_numpy(row_starts[:-1]).to(dev)]
+                 .cpu().tolist() if len(pages) else [])
+
+        # dictionaries
+        dict_fixed_ptr = {}
+        dict_str = {}
+        if dicts:
+            sidx_descs = bytearray()
+            scount = 0
+            for di, dp in enumerate(dicts):
+                base = pbase[dict_base_idx + di]
+                if leaf.physical_type == T_BYTE_ARRAY:
+                    voff = torch.empty(max(dp.num_values, 1),
+                                       dtype=torch.int64, device=dev)
+                    vlen = torch.empty(max(dp.num_values, 1),
+                                       dtype=torch.int32, device=dev)
+                    sidx_descs += struct.pack(_STRIDX_FMT, base, len(dp.data),
+                                              dp.num_values, voff.data_ptr(),
+                                              vlen.data_ptr())
+                    scount += 1
+                    dict_str[di] = (base, voff, vlen)
+                else:
+                    dict_fixed_ptr[di] = base
+            if scount:
+                st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
+                g.pq_string_plain_index(st.data_ptr(), scount, stream)
+
+        # dict-index pages
+        idx_tensors = {}
+        rle_descs = bytearray()
+        rle_meta = []
+        for i, p in enumerate(pages):
+            if p.encoding in (ENC_PLAIN_DICTIONARY, ENC_RLE_DICTIONARY):
+                bo = body_offs[i]
+                bw = p.data[bo]
+                idx = torch.empty(max(p.num_values, 1), dtype=torch.int32,
+                                  device=dev)
+                idx_tensors[i] = idx
+                rle_descs += struct.pack(_RLE_FMT, pbase[i] + bo + 1,
+                                         len(p.data) - bo - 1, idx.data_ptr(),
+                                         p.num_values, bw, 1)
+                rle_meta.append(i)
+        if rle_meta:
+            rt = torch.frombuffer(rle_descs, dtype=torch.uint8).to(dev)
+            g.pq_rle_decode(rt.data_ptr(), len(rle_meta), stream)
+
+        leaf_validity = None
+        if max_def > 0 and total_rows:
+            leaf_validity = make_validity(total_rows, dev)
+            g.pq_def_to_validity(def_ptr, total_rows,
+                                 leaf_validity.data_ptr(), stream)
+
+        if leaf.physical_type == T_BYTE_ARRAY:
+            sidx_descs = bytearray()
+            plain_meta = []
+            for i, p in enumerate(pages):
+                if p.encoding == ENC_PLAIN:
+                    bo = body_offs[i]
+                    voff = torch.empty(max(p.num_values, 1),
+                                       dtype=torch.int64, device=dev)
+                    vlen = torch.empty(max(p.num_values, 1),
+                                       dtype=torch.int32, device=dev)
+                    sidx_descs += struct.pack(_STRIDX_FMT, pbase[i] + bo,
+                                              len(p.data) - bo, p.num_values,
+                                              voff.data_ptr(),
+                                              vlen.data_ptr())
+                    plain_meta.append((i, voff, vlen))
+            if plain_meta:
+                st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
+                g.pq_string_plain_index(st.data_ptr(), len(plain_meta),
+                                        stream)
+            pm = {i: (voff, vlen) for i, voff, vlen in plain_meta}
+            cp_descs = bytearray()
+            for i, p in enumerate(pages):
+                if p.encoding == ENC_PLAIN:
+                    voff, vlen = pm[i]
+                    cp_descs += struct.pack(
+                        _STRCPY_FMT, pbase[i] + body_offs[i], voff.data_ptr(),
+                        vlen.data_ptr(), 0, def_ptr, vprefix_ptr,
+                        int(row_starts[i]), p.num_values, int(vbase[i]))
+                else:
+                    di = dict_per_page[i]
+                    base, voff, vlen = dict_str[di]
+                    cp_descs += struct.pack(
+                        _STRCPY_FMT, base, voff.data_ptr(), vlen.data_ptr(),
+                        idx_tensors[i].data_ptr(), def_ptr, vprefix_ptr,
+                        int(row_starts[i]), p.num_values, int(vbase[i]))
+            ct = torch.frombuffer(cp_descs or bytearray(1),
+                                  dtype=torch.uint8).to(dev)
+            lens = torch.empty(max(total_rows, 1), dtype=torch.int32,
+                               device=dev)
+            g.pq_string_copy(ct.data_ptr(), len(pages), 0, lens.data_ptr(),
+                             0, 0, stream)
+            soffs = torch.zeros(total_rows + 1, dtype=torch.int32, device=dev)
+            if total_rows:
+                torch.cumsum(lens[:total_rows], 0,
+                             out=soffs[1:].view(total_rows))
+            nchars = int(soffs[-1].item())
+            chars = torch.empty(max(nchars, 1), dtype=torch.uint8, device=dev)
+            g.pq_string_copy(ct.data_ptr(), len(pages), 1, 0,
+                             soffs.data_ptr(), chars.data_ptr(), stream)
+            children.append(Column(DType.STRING, total_rows, chars[:nchars],
+                                   leaf_validity, soffs, null_count=None))
+        else:
+            width = _PHYS_WIDTH[leaf.physical_type]
+            from .columnar import TORCH_DTYPE
+            ldt = _field_dtype(leaf)
+            out = torch.zeros(max(total_rows, 1), dtype=TORCH_DTYPE[ldt],
+                              device=dev)
+            sc_descs = bytearray()
+            for i, p in enumerate(pages):
+                if p.encoding == ENC_PLAIN:
+                    sc_descs += struct.pack(_SCATTER_FMT,
+                                            pbase[i] + body_offs[i], 0,
+                                            def_ptr, vprefix_ptr,
+                                            int(row_starts[i]), p.num_values,
+                                            int(vbase[i]), width, 0)
+                else:
+                    di = dict_per_page[i]
+                    sc_descs += struct.pack(_SCATTER_FMT,
+                                            idx_tensors[i].data_ptr(),
+                                            dict_fixed_ptr[di], def_ptr,
+                                            vprefix_ptr, int(row_starts[i]),
+                                            p.num_values, int(vbase[i]),
+                                            width, 1)
+            st = torch.frombuffer(sc_descs or bytearray(1),
+                                  dtype=torch.uint8).to(dev)
+            g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(),
+                               stream)
+            children.append(Column(ldt, total_rows, out[:max(total_rows, 1)],
+                                   leaf_validity, scale=leaf.scale,
+                                   null_count=None))
+
+    return Column(DType.STRUCT, total_rows, None, struct_validity, None,
+                  children, null_count=None)
+
+
 def read_table(path: str, columns: Optional[Sequence[str]] = None,
                device="cuda") -> Table:
     """Scan a parquet file into a GPU Table (footer + page decode).
@@ -889,12 +1130,19 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
         footer = footer.prune(columns)
     total_rows = sum(rg.num_rows for rg in footer.row_groups)
     cols = []
-    for fi, f in enumerate(footer.schema):
-        chunks = [rg.columns[fi] for rg in footer.row_groups]
+    leaf = 0
+    for f in footer.schema:
+        if f.is_struct:
+            cols.append(_read_struct_column(raw, f, footer.row_groups, leaf,
+                                            total_rows, device))
+            leaf += len(f.children)
+            continue
+        chunks = [rg.columns[leaf] for rg in footer.row_groups]
         if f.is_list:
             cols.append(_read_list_column(raw, f, chunks, total_rows, device))
         else:
             cols.append(_read_column(raw, f, chunks, total_rows, device))
+        leaf += 1
     return Table(cols)
 
 

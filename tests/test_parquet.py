@@ -333,3 +333,28 @@ def test_thrift_typed_fuzz_roundtrip():
         back, end = host.thrift_parse_typed(raw, 0)
         assert end == len(raw)
         assert norm(back) == norm(tree)
+
+
+@pytest.mark.gpu
+def test_decode_struct_column(tmp_path):
+    rows = [None if i % 13 == 4 else
+            {"x": None if i % 7 == 2 else i * 3,
+             "y": None if i % 5 == 1 else f"s{i}"}
+            for i in range(3000)]
+    t = pa.table({"st": pa.array(rows, type=pa.struct(
+        [("x", pa.int64()), ("y", pa.string())])),
+        "i": pa.array(list(range(3000)), type=pa.int64())})
+    p = str(tmp_path / "st.parquet")
+    pq.write_table(t, p, compression="NONE", row_group_size=1100)
+    got = srj_pq.read_table(p, device="cuda")
+    st = got.columns[0]
+    assert st.dtype.name == "STRUCT"
+    xs = st.children[0].to_pylist()
+    ys = st.children[1].to_pylist()
+    for i, r in enumerate(rows):
+        if r is None:
+            assert not st.is_valid_host(i), i
+        else:
+            assert st.is_valid_host(i), i
+            assert xs[i] == r["x"] and ys[i] == r["y"], i
+    assert got.columns[1].to_pylist() == list(range(3000))
