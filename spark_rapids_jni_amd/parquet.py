@@ -51,6 +51,7 @@ class SchemaField:
     scale: int = 0
     precision: int = 0
     logical: Optional[dict] = None
+    type_length: int = 0     # FIXED_LEN_BYTE_ARRAY byte width
     # LIST<primitive> (parquet 3-level encoding): element leaf + level depths
     is_list: bool = False
     element: Optional["SchemaField"] = None
@@ -144,7 +145,7 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                 name=se[4].decode(), physical_type=se.get(1, -1),
                 repetition=se.get(3, 0), converted_type=se.get(6),
                 scale=se.get(7, 0), precision=se.get(8, 0),
-                logical=se.get(10)))
+                logical=se.get(10), type_length=se.get(2, 0)))
             i += 1
             continue
         if not (se.get(6) == 3 or (se.get(10) and 3 in se.get(10, {}))):
@@ -159,7 +160,7 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                     name=lf[4].decode(), physical_type=lf.get(1, -1),
                     repetition=lf.get(3, 0), converted_type=lf.get(6),
                     scale=lf.get(7, 0), precision=lf.get(8, 0),
-                    logical=lf.get(10),
+                    logical=lf.get(10), type_length=lf.get(2, 0),
                     max_def=outer_opt + (1 if lf.get(3, 0) == 1 else 0)))
             fields.append(SchemaField(
                 name=se[4].decode(), physical_type=-1,
@@ -221,14 +222,15 @@ def _field_dtype(f: SchemaField) -> DType:
         return DType.DATE32
     if f.physical_type == T_INT64 and f.converted_type in (9, 10):
         return DType.TIMESTAMP_US
-    if f.converted_type == 5:  # DECIMAL backed by int32/int64 physical
+    if f.converted_type == 5:  # DECIMAL
         if f.physical_type == T_INT32:
             return DType.DECIMAL32
         if f.physical_type == T_INT64:
             return DType.DECIMAL64
-        raise NotImplementedError(
-            "FIXED_LEN_BYTE_ARRAY decimals not supported (write with "
-            "store_decimal_as_integer / Spark int-backed decimals)")
+        if (f.physical_type == T_FIXED_LEN_BYTE_ARRAY
+                and 0 < f.type_length <= 16):
+            return DType.DECIMAL128
+        raise NotImplementedError("unsupported DECIMAL physical type")
     return _PHYS_TO_DTYPE[f.physical_type]
 
 
@@ -606,9 +608,11 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
         return Column(DType.BOOL8, total_rows, out[:total_rows], validity,
                       null_count=None)
 
-    width = _PHYS_WIDTH[f.physical_type]
+    flba_dec = f.physical_type == T_FIXED_LEN_BYTE_ARRAY
+    width = f.type_length if flba_dec else _PHYS_WIDTH[f.physical_type]
     from .columnar import TORCH_DTYPE
-    out = torch.zeros(max(total_rows, 1), dtype=TORCH_DTYPE[dtype], device=dev)
+    numel = (total_rows * 2) if flba_dec else total_rows
+    out = torch.zeros(max(numel, 1), dtype=TORCH_DTYPE[dtype], device=dev)
     sc_descs = bytearray()
     for i, p in enumerate(pages):
         bo = _body_off(i, p)
@@ -623,12 +627,16 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                                     int(row_starts[i]), p.num_values,
                                     page_vbase[i], width, 1)
     st = torch.frombuffer(sc_descs or bytearray(1), dtype=torch.uint8).to(dev)
-    g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(), stream)
+    if flba_dec:
+        # big-endian W-byte decimals -> 2x int64 words per row
+        g.pq_flba_dec128(st.data_ptr(), len(pages), out.data_ptr(), stream)
+    else:
+        g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(), stream)
     validity = None
     if nullable:
         validity = make_validity(total_rows, dev)
         g.pq_def_to_validity(def_ptr, total_rows, validity.data_ptr(), stream)
-    return Column(dtype, total_rows, out[:total_rows], validity,
+    return Column(dtype, total_rows, out[:max(numel, 1)], validity,
                   scale=f.scale, null_count=None)
 
 

@@ -10,6 +10,7 @@ void srj_string_copy(const void*, int32_t, int32_t, int32_t*, const int32_t*,
 void srj_def_to_validity(const uint8_t*, int64_t, uint8_t*, hipStream_t);
 void srj_pq_snappy_decomp(const void*, int32_t, hipStream_t);
 void srj_gather_u8_at(const uint64_t*, int64_t, uint8_t*, hipStream_t);
+void srj_pq_flba_dec128(const void*, int32_t, int64_t*, hipStream_t);
 }
 
 void register_parquet(py::module_& m) {
@@ -22,6 +23,12 @@ void register_parquet(py::module_& m) {
     srj_gather_u8_at(as_ptr<uint64_t>(addrs), n, as_ptr<uint8_t>(out),
                      as_stream(stream));
     check_hip("gather_u8_at");
+  });
+  m.def("pq_flba_dec128", [](uintptr_t descs, int32_t npages, uintptr_t out,
+                             uintptr_t stream) {
+    srj_pq_flba_dec128(as_ptr<void>(descs), npages, as_ptr<int64_t>(out),
+                       as_stream(stream));
+    check_hip("pq_flba_dec128");
   });
   m.def("pq_rle_decode", [](uintptr_t descs, int32_t npages, uintptr_t stream) {
     srj_rle_decode(as_ptr<void>(descs), npages, as_stream(stream));

@@ -341,6 +341,44 @@ __global__ void gather_u8_at_kernel(const uint64_t* __restrict__ addrs,
   if (i < n) out[i] = *reinterpret_cast<const uint8_t*>(addrs[i]);
 }
 
+
+// FIXED_LEN_BYTE_ARRAY decimals: W-byte big-endian two's-complement values
+// -> DECIMAL128 (2 x int64 little-endian words, sign-extended). Reuses the
+// ScatterDesc grid (width = FLBA byte length, is_dict selects the source).
+__global__ void flba_dec128_kernel(const ScatterDesc* __restrict__ descs,
+                                   int32_t npages,
+                                   int64_t* __restrict__ out) {
+  for (int32_t page = blockIdx.x; page < npages; page += gridDim.x) {
+    ScatterDesc d = descs[page];
+    for (int64_t i = threadIdx.x; i < d.nrows; i += blockDim.x) {
+      int64_t row = d.row_start + i;
+      bool valid = d.def == nullptr || d.def[row] != 0;
+      if (!valid) continue;
+      int64_t vi = d.def == nullptr ? i : (d.vprefix[row] - d.value_base);
+      int W = d.width;
+      const uint8_t* src;
+      if (d.is_dict) {
+        int32_t idx = reinterpret_cast<const int32_t*>(d.values)[vi];
+        src = d.dict + (int64_t)idx * W;
+      } else {
+        src = d.values + vi * W;
+      }
+      bool neg = src[0] & 0x80;
+      uint64_t hi = neg ? ~0ull : 0ull, lo = neg ? ~0ull : 0ull;
+      for (int k = 0; k < W; ++k) {
+        uint64_t b = src[k];
+        int bit = (W - 1 - k) * 8;
+        if (bit < 64)
+          lo = (lo & ~(0xFFull << bit)) | (b << bit);
+        else
+          hi = (hi & ~(0xFFull << (bit - 64))) | (b << (bit - 64));
+      }
+      out[row * 2] = (int64_t)lo;
+      out[row * 2 + 1] = (int64_t)hi;
+    }
+  }
+}
+
 }  // namespace srj
 
 using namespace srj;
@@ -401,6 +439,13 @@ void srj_pq_snappy_decomp(const void* descs, int32_t n, hipStream_t stream) {
 void srj_gather_u8_at(const uint64_t* addrs, int64_t n, uint8_t* out,
                       hipStream_t stream) {
   gather_u8_at_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(addrs, n, out);
+}
+
+void srj_pq_flba_dec128(const void* descs, int32_t npages, int64_t* out,
+                        hipStream_t stream) {
+  flba_dec128_kernel<<<npages < MAX_GRID ? (npages ? npages : 1) : MAX_GRID,
+                       DEFAULT_BLOCK, 0, stream>>>(
+      reinterpret_cast<const ScatterDesc*>(descs), npages, out);
 }
 
 }  // extern "C"

@@ -358,3 +358,34 @@ def test_decode_struct_column(tmp_path):
             assert st.is_valid_host(i), i
             assert xs[i] == r["x"] and ys[i] == r["y"], i
     assert got.columns[1].to_pylist() == list(range(3000))
+
+
+@pytest.mark.gpu
+def test_decode_decimal_flba(tmp_path):
+    """FIXED_LEN_BYTE_ARRAY decimals (pyarrow's default layout; Spark uses it
+    for precision > 18): big-endian N-byte values -> DECIMAL128."""
+    from decimal import Decimal
+    rows_small = [None if i % 9 == 1 else Decimal(i * 7 - 5000).scaleb(-2)
+                  for i in range(3000)]
+    rows_big = [None if i % 5 == 2 else
+                Decimal((i - 1500) * 10**20 + i).scaleb(-4)
+                for i in range(3000)]
+    t = pa.table({"d7": pa.array(rows_small, type=pa.decimal128(7, 2)),
+                  "d38": pa.array(rows_big, type=pa.decimal128(38, 4))})
+    p = str(tmp_path / "flba.parquet")
+    pq.write_table(t, p, compression="NONE", row_group_size=1300)
+    got = srj_pq.read_table(p, device="cuda")
+    for col, rows, scale in ((got.columns[0], rows_small, 2),
+                             (got.columns[1], rows_big, 4)):
+        assert col.dtype.name == "DECIMAL128" and col.scale == scale
+        words = col.data.cpu().tolist()
+        for i, exp in enumerate(rows):
+            if exp is None:
+                assert not col.is_valid_host(i), i
+                continue
+            lo = words[2 * i] & (2**64 - 1)
+            hi = words[2 * i + 1] & (2**64 - 1)
+            u = (hi << 64) | lo
+            if u >= 2**127:
+                u -= 2**128
+            assert u == int(exp.scaleb(scale)), (i, u, exp)
