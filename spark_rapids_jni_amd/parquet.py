@@ -674,6 +674,8 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
     stream = _native.current_stream()
     dev = torch.device(device)
     elem = f.element
+    if elem.physical_type == T_FIXED_LEN_BYTE_ARRAY:
+        raise NotImplementedError("LIST of FLBA decimals not supported")
     elem_nullable = elem.repetition == 1
     max_def = f.max_def
     def_bw = max(max_def.bit_length(), 1)
