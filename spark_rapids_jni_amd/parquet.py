@@ -1093,10 +1093,13 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
             children.append(Column(DType.STRING, total_rows, chars[:nchars],
                                    leaf_validity, soffs, null_count=None))
         else:
-            width = _PHYS_WIDTH[leaf.physical_type]
+            flba = leaf.physical_type == T_FIXED_LEN_BYTE_ARRAY
+            width = leaf.type_length if flba else \
+                _PHYS_WIDTH[leaf.physical_type]
             from .columnar import TORCH_DTYPE
             ldt = _field_dtype(leaf)
-            out = torch.zeros(max(total_rows, 1), dtype=TORCH_DTYPE[ldt],
+            numel = (total_rows * 2) if flba else total_rows
+            out = torch.zeros(max(numel, 1), dtype=TORCH_DTYPE[ldt],
                               device=dev)
             sc_descs = bytearray()
             for i, p in enumerate(pages):
@@ -1116,9 +1119,13 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                                             width, 1)
             st = torch.frombuffer(sc_descs or bytearray(1),
                                   dtype=torch.uint8).to(dev)
-            g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(),
-                               stream)
-            children.append(Column(ldt, total_rows, out[:max(total_rows, 1)],
+            if flba:
+                g.pq_flba_dec128(st.data_ptr(), len(pages), out.data_ptr(),
+                                 stream)
+            else:
+                g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(),
+                                   stream)
+            children.append(Column(ldt, total_rows, out[:max(numel, 1)],
                                    leaf_validity, scale=leaf.scale,
                                    null_count=None))
 

@@ -341,8 +341,13 @@ def test_decode_struct_column(tmp_path):
             {"x": None if i % 7 == 2 else i * 3,
              "y": None if i % 5 == 1 else f"s{i}"}
             for i in range(3000)]
+    from decimal import Decimal
+    for i, r in enumerate(rows):
+        if r is not None:
+            r["z"] = None if i % 11 == 6 else Decimal(i - 1500).scaleb(-2)
     t = pa.table({"st": pa.array(rows, type=pa.struct(
-        [("x", pa.int64()), ("y", pa.string())])),
+        [("x", pa.int64()), ("y", pa.string()),
+         ("z", pa.decimal128(20, 2))])),
         "i": pa.array(list(range(3000)), type=pa.int64())})
     p = str(tmp_path / "st.parquet")
     pq.write_table(t, p, compression="NONE", row_group_size=1100)
@@ -351,12 +356,23 @@ def test_decode_struct_column(tmp_path):
     assert st.dtype.name == "STRUCT"
     xs = st.children[0].to_pylist()
     ys = st.children[1].to_pylist()
+    zc = st.children[2]
+    zw = zc.data.cpu().tolist()
+    assert zc.dtype.name == "DECIMAL128" and zc.scale == 2
     for i, r in enumerate(rows):
         if r is None:
             assert not st.is_valid_host(i), i
         else:
             assert st.is_valid_host(i), i
             assert xs[i] == r["x"] and ys[i] == r["y"], i
+            if r["z"] is None:
+                assert not zc.is_valid_host(i), i
+            else:
+                u = (zw[2 * i] & (2**64 - 1)) | \
+                    ((zw[2 * i + 1] & (2**64 - 1)) << 64)
+                if u >= 2**127:
+                    u -= 2**128
+                assert u == int(r["z"].scaleb(2)), i
     assert got.columns[1].to_pylist() == list(range(3000))
 
 
