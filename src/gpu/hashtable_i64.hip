@@ -37,9 +37,10 @@ constexpr int PIPE = 8;
 
 __device__ inline uint64_t i64_hash(long long k) { return mix64((uint64_t)k); }
 
-// Slot from the TOP hash bits: the table is then ordered by hash prefix, so
-// keys radix-partitioned by the same prefix probe a contiguous table window
-// (LLC-sized at 512 partitions) instead of the whole table. mask = 2^k - 1.
+// Slot from the TOP hash bits (mask = 2^k - 1). This orders the table by
+// hash prefix; a radix-partitioned probe over it was built and measured NOT
+// to pay on the 10Bx1B config (docs/PERF.md) — the layout is kept because it
+// is equivalent in cost to low-bit indexing and keeps that option open.
 __device__ inline uint64_t slot_of(uint64_t h, uint64_t mask) {
   return h >> __builtin_clzll(mask);
 }

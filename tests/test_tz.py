@@ -57,3 +57,30 @@ def test_tz_convert_device():
     lcol = Column.from_pylist(local, DType.TIMESTAMP_US, "cuda")
     back = db.convert_timestamp_to_utc(lcol, LA).to_pylist()
     assert back == cases_utc
+
+
+def test_tzdb_more_zones_against_zoneinfo():
+    """Cross-check the built transition tables against zoneinfo for zones
+    spanning both hemispheres, half-hour offsets and historic changes."""
+    import bisect
+    import datetime
+    from zoneinfo import ZoneInfo
+    from spark_rapids_jni_amd.tz import extract_transitions
+    zones = ["Australia/Sydney", "America/Sao_Paulo", "Asia/Kathmandu",
+             "Africa/Cairo", "Pacific/Auckland"]
+    probes = [datetime.datetime(y, m, d, h) for (y, m, d, h) in
+              [(1999, 6, 1, 12), (2005, 12, 31, 23), (2012, 3, 11, 2),
+               (2019, 10, 6, 3), (2024, 1, 15, 0)]]
+    for z in zones:
+        try:
+            zi = ZoneInfo(z)
+        except Exception:
+            continue
+        trans = extract_transitions(z)  # [(utc_second, offset_after)]
+        instants = [t[0] for t in trans]
+        for dt in probes:
+            utc = dt.replace(tzinfo=datetime.timezone.utc)
+            ts = int(utc.timestamp())
+            i = bisect.bisect_right(instants, ts) - 1
+            expect = utc.astimezone(zi).utcoffset().total_seconds()
+            assert trans[max(i, 0)][1] == expect, (z, dt)
