@@ -148,6 +148,16 @@ class Column:
         fill = [v if v is not None else 0 for v in values]
         if dtype == DType.BOOL8:
             fill = [1 if v else 0 for v in fill]
+        if dtype == DType.DECIMAL128:
+            import struct as _struct
+            words = []
+            for v in fill:
+                words.extend(_struct.unpack(
+                    "<2q", int(v).to_bytes(16, "little", signed=True)))
+            data = torch.tensor(words, dtype=torch.int64, device=device)
+            v = validity_from_bools(valid, device) if has_null else None
+            return Column(dtype, n, data, v, scale=scale,
+                          null_count=n - sum(valid))
         data = torch.tensor(fill, dtype=tdt, device=device)
         v = validity_from_bools(valid, device) if has_null else None
         return Column(dtype, n, data, v, scale=scale, null_count=n - sum(valid))
@@ -215,6 +225,12 @@ class Column:
             kids = [ch.to_pylist() for ch in c.children]
             return [None if not c.is_valid_host(i) else tuple(k[i] for k in kids)
                     for i in range(c.size)]
+        if self.dtype == DType.DECIMAL128:
+            raw = c.data.numpy().tobytes()
+            vals = [int.from_bytes(raw[i * 16:(i + 1) * 16], "little",
+                                   signed=True) for i in range(c.size)]
+            return [v if c.is_valid_host(i) else None
+                    for i, v in enumerate(vals)]
         vals = c.data.tolist()
         if self.dtype == DType.BOOL8:
             vals = [bool(v) for v in vals]

@@ -112,7 +112,14 @@ def write_partition(columns: Sequence[Column], row_offset: int, num_rows: int,
             b1 = (start + n + 7) // 8
             if b1 <= b0:
                 b1 = b0 + 1  # at least 1 byte per spec
-            validity_parts.append(v[b0:b1].tobytes())
+            part = v[b0:b1].tobytes()
+            # b0 can sit past the end of the validity buffer (0-row slice at a
+            # 64-multiple row count, or any 0-row nullable column): the reader
+            # still consumes >=1 byte, so pad to the advertised length or the
+            # whole body desyncs at merge time.
+            if len(part) < b1 - b0:
+                part += b"\x00" * (b1 - b0 - len(part))
+            validity_parts.append(part)
         if has_offsets(c) and n >= 0:
             o = c.offsets.numpy()
             offset_parts.append(o[start:start + n + 1].tobytes()

@@ -108,9 +108,15 @@ class GpuTimeZoneDB:
             return
         utc, local, off, zoffs = [], [], [], [0]
         for z in sorted(self._zones.values(), key=lambda z: z.index):
-            for t, o in zip(z.utc_instants, z.offsets):
+            for i, (t, o) in enumerate(zip(z.utc_instants, z.offsets)):
+                # localInstant follows the reference GpuTimeZoneDB.loadData
+                # isGap/isOverlap split: for an overlap (offset decreases) the
+                # boundary is utc+offsetBefore so ambiguous fall-back local
+                # times resolve to the EARLIER offset; for a gap, offsetAfter.
+                o_before = z.offsets[i - 1] if i > 0 else o
+                lo = o_before if o < o_before else o
                 utc.append(t * 1_000_000)
-                local.append((t + o) * 1_000_000)
+                local.append((t + lo) * 1_000_000)
                 off.append(o)
             zoffs.append(len(utc))
         dev = self.device

@@ -97,6 +97,18 @@ def _to_signed64(v):
     return v - (1 << 64) if v >= (1 << 63) else v
 
 
+def java_bigint_bytes(v: int) -> bytes:
+    """java.math.BigInteger.toByteArray(): minimal big-endian two's
+    complement, at least one byte. Spark hashes DECIMAL128 over these bytes
+    (ref hash/hash.cuh:64 to_java_bigdecimal)."""
+    n = 1
+    while True:
+        try:
+            return int(v).to_bytes(n, "big", signed=True)
+        except OverflowError:
+            n += 1
+
+
 def murmur3_row(values, dtypes, seed=42):
     """Chained murmur3 over one row. values: python values (None = null)."""
     from ..columnar import DType
@@ -106,11 +118,16 @@ def murmur3_row(values, dtypes, seed=42):
             continue
         if dt in (DType.BOOL8,):
             h = murmur3_int(1 if v else 0, h)
-        elif dt in (DType.INT8, DType.INT16, DType.INT32, DType.DATE32,
-                    DType.DECIMAL32):
+        elif dt in (DType.INT8, DType.INT16, DType.INT32, DType.DATE32):
             h = murmur3_int(v, h)
-        elif dt in (DType.INT64, DType.TIMESTAMP_US, DType.DECIMAL64):
+        # Spark hashes decimals with precision<=18 as hashLong of the
+        # unscaled value, so DECIMAL32 promotes to 64-bit too
+        # (ref murmur_hash.cuh:185-197).
+        elif dt in (DType.INT64, DType.TIMESTAMP_US, DType.DECIMAL64,
+                    DType.DECIMAL32):
             h = murmur3_long(v, h)
+        elif dt == DType.DECIMAL128:
+            h = murmur3_bytes(java_bigint_bytes(v), h)
         elif dt == DType.FLOAT32:
             h = murmur3_int(norm_float_bits(v), h)
         elif dt == DType.FLOAT64:
@@ -204,6 +221,8 @@ def xxhash64_row(values, dtypes, seed=42):
             h = xxhash64_bytes(struct.pack("<q", v), h)
         elif dt == DType.DECIMAL32:
             h = xxhash64_bytes(struct.pack("<q", v), h)
+        elif dt == DType.DECIMAL128:
+            h = xxhash64_bytes(java_bigint_bytes(v), h)
         elif dt == DType.FLOAT32:
             h = xxhash64_bytes(struct.pack("<i", norm_float_bits(v)), h)
         elif dt == DType.FLOAT64:

@@ -16,11 +16,26 @@
 
 namespace srj {
 
+// Minimal big-endian two's-complement byte form of a 128-bit little-endian
+// value, matching java.math.BigDecimal.unscaledValue().toByteArray(): strip
+// redundant sign-extension bytes, keep one extra byte when needed to preserve
+// the sign bit, reverse to big-endian. Spark hashes DECIMAL128 (precision>18)
+// over exactly these bytes (ref hash/hash.cuh:64 to_java_bigdecimal).
+__device__ int dec128_java_bytes(const uint8_t* p, uint8_t out[16]) {
+  bool neg = (p[15] & 0x80) != 0;
+  uint8_t ext = neg ? 0xff : 0x00;
+  int len = 16;
+  while (len > 1 && p[len - 1] == ext) --len;
+  if (len < 16 && (neg != ((p[len - 1] & 0x80) != 0))) ++len;
+  for (int i = 0; i < len; ++i) out[i] = p[len - 1 - i];
+  return len;
+}
+
 // ---------------------------------------------------------------------------
 // murmur3 (Spark Murmur3_x86_32)
 // ---------------------------------------------------------------------------
 __device__ uint32_t murmur3_col_row(const ColDesc* cols, const ColDesc& c,
-                                    int64_t row, uint32_t seed);
+                                    int64_t row, uint32_t seed, int depth);
 
 // struct/list members hash their elements in order with the running seed.
 __device__ uint32_t murmur3_children(const ColDesc* cols, const ColDesc& c,
@@ -29,7 +44,7 @@ __device__ uint32_t murmur3_children(const ColDesc* cols, const ColDesc& c,
   if (c.dtype == STRUCT) {
     uint32_t h = seed;
     for (int k = 0; k < c.num_children; ++k) {
-      h = murmur3_col_row(cols, cols[c.child0 + k], row, h);
+      h = murmur3_col_row(cols, cols[c.child0 + k], row, h, depth);
     }
     return h;
   }
@@ -37,7 +52,7 @@ __device__ uint32_t murmur3_children(const ColDesc* cols, const ColDesc& c,
     uint32_t h = seed;
     const ColDesc& child = cols[c.child0];
     for (int32_t j = c.offsets[row]; j < c.offsets[row + 1]; ++j) {
-      h = murmur3_col_row(cols, child, j, h);
+      h = murmur3_col_row(cols, child, j, h, depth);
     }
     return h;
   }
@@ -45,18 +60,26 @@ __device__ uint32_t murmur3_children(const ColDesc* cols, const ColDesc& c,
 }
 
 __device__ uint32_t murmur3_col_row(const ColDesc* cols, const ColDesc& c,
-                                    int64_t row, uint32_t seed) {
+                                    int64_t row, uint32_t seed, int depth = 0) {
   if (!is_valid(c.valid, row)) return seed;  // null: seed passes through
   switch (c.dtype) {
     case BOOL8: return mm3_hash_int(reinterpret_cast<const int8_t*>(c.data)[row] != 0, seed);
     case INT8: return mm3_hash_int(reinterpret_cast<const int8_t*>(c.data)[row], seed);
     case INT16: return mm3_hash_int(reinterpret_cast<const int16_t*>(c.data)[row], seed);
     case INT32:
-    case DATE32:
-    case DECIMAL32: return mm3_hash_int(reinterpret_cast<const int32_t*>(c.data)[row], seed);
+    case DATE32: return mm3_hash_int(reinterpret_cast<const int32_t*>(c.data)[row], seed);
+    // Spark hashes small decimals' unscaled value as a long (hashLong), so
+    // DECIMAL32 sign-extends to 64 bits (ref murmur_hash.cuh:185-197).
+    case DECIMAL32:
+      return mm3_hash_long((int64_t)reinterpret_cast<const int32_t*>(c.data)[row], seed);
     case INT64:
     case TIMESTAMP_US:
     case DECIMAL64: return mm3_hash_long(reinterpret_cast<const int64_t*>(c.data)[row], seed);
+    case DECIMAL128: {
+      uint8_t buf[16];
+      int n = dec128_java_bytes(reinterpret_cast<const uint8_t*>(c.data) + row * 16, buf);
+      return mm3_hash_bytes(reinterpret_cast<const char*>(buf), n, seed);
+    }
     case FLOAT32:
       return mm3_hash_int(norm_float_bits(reinterpret_cast<const float*>(c.data)[row]), seed);
     case FLOAT64:
@@ -66,7 +89,7 @@ __device__ uint32_t murmur3_col_row(const ColDesc* cols, const ColDesc& c,
       return mm3_hash_bytes(s.ptr, s.len, seed);
     }
     case STRUCT:
-    case LIST: return murmur3_children(cols, c, row, seed, 0);
+    case LIST: return murmur3_children(cols, c, row, seed, depth + 1);
     default: return seed;
   }
 }
@@ -88,28 +111,29 @@ __global__ void murmur3_kernel(const ColDesc* __restrict__ cols,
 // xxhash64
 // ---------------------------------------------------------------------------
 __device__ uint64_t xxhash64_col_row(const ColDesc* cols, const ColDesc& c,
-                                     int64_t row, uint64_t seed);
+                                     int64_t row, uint64_t seed, int depth);
 
 __device__ uint64_t xxhash64_children(const ColDesc* cols, const ColDesc& c,
-                                      int64_t row, uint64_t seed) {
+                                      int64_t row, uint64_t seed, int depth) {
+  if (depth > 8) return seed;
   if (c.dtype == STRUCT) {
     uint64_t h = seed;
     for (int k = 0; k < c.num_children; ++k)
-      h = xxhash64_col_row(cols, cols[c.child0 + k], row, h);
+      h = xxhash64_col_row(cols, cols[c.child0 + k], row, h, depth);
     return h;
   }
   if (c.dtype == LIST) {
     uint64_t h = seed;
     const ColDesc& child = cols[c.child0];
     for (int32_t j = c.offsets[row]; j < c.offsets[row + 1]; ++j)
-      h = xxhash64_col_row(cols, child, j, h);
+      h = xxhash64_col_row(cols, child, j, h, depth);
     return h;
   }
   return seed;
 }
 
 __device__ uint64_t xxhash64_col_row(const ColDesc* cols, const ColDesc& c,
-                                     int64_t row, uint64_t seed) {
+                                     int64_t row, uint64_t seed, int depth = 0) {
   if (!is_valid(c.valid, row)) return seed;
   switch (c.dtype) {
     case BOOL8:
@@ -128,6 +152,11 @@ __device__ uint64_t xxhash64_col_row(const ColDesc* cols, const ColDesc& c,
       return xxhash64_fixed((uint64_t)reinterpret_cast<const int32_t*>(c.data)[row], 8, seed);
     case DECIMAL64:
       return xxhash64_fixed((uint64_t)reinterpret_cast<const int64_t*>(c.data)[row], 8, seed);
+    case DECIMAL128: {
+      uint8_t buf[16];
+      int n = dec128_java_bytes(reinterpret_cast<const uint8_t*>(c.data) + row * 16, buf);
+      return xxhash64_bytes(reinterpret_cast<const char*>(buf), n, seed);
+    }
     case FLOAT32: {
       int32_t b = norm_float_bits(reinterpret_cast<const float*>(c.data)[row]);
       return xxhash64_fixed((uint32_t)b, 4, seed);
@@ -141,7 +170,7 @@ __device__ uint64_t xxhash64_col_row(const ColDesc* cols, const ColDesc& c,
       return xxhash64_bytes(s.ptr, s.len, seed);
     }
     case STRUCT:
-    case LIST: return xxhash64_children(cols, c, row, seed);
+    case LIST: return xxhash64_children(cols, c, row, seed, depth + 1);
     default: return seed;
   }
 }
@@ -162,26 +191,30 @@ __global__ void xxhash64_kernel(const ColDesc* __restrict__ cols,
 // ---------------------------------------------------------------------------
 // hive hash (Java Object.hashCode semantics; combine: h = 31*h + colhash)
 // ---------------------------------------------------------------------------
-__device__ int32_t hive_col_row(const ColDesc* cols, const ColDesc& c, int64_t row);
+__device__ int32_t hive_col_row(const ColDesc* cols, const ColDesc& c, int64_t row,
+                                int depth);
 
-__device__ int32_t hive_children(const ColDesc* cols, const ColDesc& c, int64_t row) {
+__device__ int32_t hive_children(const ColDesc* cols, const ColDesc& c, int64_t row,
+                                 int depth) {
+  if (depth > 8) return 0;
   if (c.dtype == STRUCT) {
     int32_t h = 0;
     for (int k = 0; k < c.num_children; ++k)
-      h = 31 * h + hive_col_row(cols, cols[c.child0 + k], row);
+      h = 31 * h + hive_col_row(cols, cols[c.child0 + k], row, depth);
     return h;
   }
   if (c.dtype == LIST) {
     int32_t h = 0;
     const ColDesc& child = cols[c.child0];
     for (int32_t j = c.offsets[row]; j < c.offsets[row + 1]; ++j)
-      h = 31 * h + hive_col_row(cols, child, j);
+      h = 31 * h + hive_col_row(cols, child, j, depth);
     return h;
   }
   return 0;
 }
 
-__device__ int32_t hive_col_row(const ColDesc* cols, const ColDesc& c, int64_t row) {
+__device__ int32_t hive_col_row(const ColDesc* cols, const ColDesc& c, int64_t row,
+                                int depth = 0) {
   if (!is_valid(c.valid, row)) return 0;
   switch (c.dtype) {
     case BOOL8: return reinterpret_cast<const int8_t*>(c.data)[row] != 0 ? 1 : 0;
@@ -224,7 +257,7 @@ __device__ int32_t hive_col_row(const ColDesc* cols, const ColDesc& c, int64_t r
       return h;
     }
     case STRUCT:
-    case LIST: return hive_children(cols, c, row);
+    case LIST: return hive_children(cols, c, row, depth + 1);
     default: return 0;
   }
 }

@@ -145,3 +145,61 @@ def test_murmur3_host_matches_gpu():
     gc = Column.from_torch(keys.cuda())
     gpu = hashing.murmur3([gc]).data.cpu()
     assert torch.equal(out, gpu)
+
+
+def _decimal_values(dt, n):
+    lim = {DType.DECIMAL32: 2**31, DType.DECIMAL64: 2**63,
+           DType.DECIMAL128: 2**127}[dt]
+    special = [0, 1, -1, 127, 128, -128, -129, lim - 1, -lim,
+               255, 256, -255, -256, 2**32, -2**32]
+    out = [s for s in special if -lim <= s < lim]
+    while len(out) < n:
+        v = random.randint(-lim, lim - 1)
+        out.append(None if len(out) % 9 == 4 else v)
+    return out[:n]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dt", [DType.DECIMAL32, DType.DECIMAL64,
+                                DType.DECIMAL128], ids=lambda d: d.name)
+def test_murmur3_gpu_decimal(dt):
+    # ADVICE fix: DECIMAL32 hashes the unscaled value as a long; DECIMAL128
+    # hashes the java BigDecimal minimal byte form.
+    from spark_rapids_jni_amd.ops import hashing
+    n = 500
+    vals = _decimal_values(dt, n)
+    col = Column.from_pylist(vals, dt, device="cuda")
+    got = hashing.murmur3([col]).to_pylist()
+    for i in range(n):
+        exp = ref.murmur3_row([vals[i]], [dt])
+        assert got[i] == exp, f"row {i}: {vals[i]!r} -> {got[i]} != {exp}"
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dt", [DType.DECIMAL32, DType.DECIMAL64,
+                                DType.DECIMAL128], ids=lambda d: d.name)
+def test_xxhash64_gpu_decimal(dt):
+    from spark_rapids_jni_amd.ops import hashing
+    n = 500
+    vals = _decimal_values(dt, n)
+    col = Column.from_pylist(vals, dt, device="cuda")
+    got = hashing.xxhash64([col]).to_pylist()
+    for i in range(n):
+        exp = ref.xxhash64_row([vals[i]], [dt])
+        assert got[i] == exp, f"row {i}: {vals[i]!r} -> {got[i]} != {exp}"
+
+
+def test_java_bigint_bytes_oracle():
+    # matches java.math.BigInteger.toByteArray() on the boundary cases the
+    # reference's to_java_bigdecimal comments call out (hash/hash.cuh:64)
+    jb = ref.java_bigint_bytes
+    assert jb(0) == b"\x00"
+    assert jb(-1) == b"\xff"
+    assert jb(127) == b"\x7f"
+    assert jb(128) == b"\x00\x80"
+    assert jb(-128) == b"\x80"
+    assert jb(-129) == b"\xff\x7f"
+    assert jb(255) == b"\x00\xff"
+    assert jb(-256) == b"\xff\x00"
+    assert jb(2**127 - 1) == b"\x7f" + b"\xff" * 15
+    assert jb(-2**127) == b"\x80" + b"\x00" * 15

@@ -230,3 +230,23 @@ def test_kudo_fuzz_roundtrip():
             if col.dtype == DType.STRUCT:
                 got = [tuple(g) if g is not None else None for g in got]
             assert got == want, f"trial {trial}"
+
+
+def test_zero_row_slice_at_validity_boundary():
+    # ADVICE fix: a 0-row slice whose start byte lies past the end of the
+    # validity buffer (64-row nullable column sliced at row 64) must still
+    # emit the 1 advertised validity byte, or the reader desyncs.
+    vals = [None if i % 3 == 0 else i for i in range(64)]
+    c = Column.from_pylist(vals, DType.INT64)
+    slices = [(0, 64), (64, 0)]
+    got = _roundtrip([c], slices)
+    assert got[0] == _expected([c], slices)[0]
+
+
+def test_zero_row_nullable_column():
+    # A 0-row nullable column: validity_nbytes(0)==0 but the wire spec says
+    # at least 1 byte travels.
+    vals = [None, 1, 2, None]
+    c = Column.from_pylist(vals, DType.INT32)
+    got = _roundtrip([c], [(2, 0), (0, 4), (4, 0)])
+    assert got[0] == _expected([c], [(2, 0), (0, 4), (4, 0)])[0]

@@ -84,3 +84,51 @@ def test_tzdb_more_zones_against_zoneinfo():
             i = bisect.bisect_right(instants, ts) - 1
             expect = utc.astimezone(zi).utcoffset().total_seconds()
             assert trans[max(i, 0)][1] == expect, (z, dt)
+
+
+def test_overlap_local_instant_uses_offset_before():
+    # ADVICE fix: reference GpuTimeZoneDB.loadData stores local = utc +
+    # offsetBefore for overlap transitions so ambiguous fall-back local
+    # times resolve to the earlier offset; gaps keep offsetAfter.
+    from spark_rapids_jni_amd.tz import GpuTimeZoneDB
+    db = GpuTimeZoneDB(device="cpu")
+    db.load("America/New_York")
+    db._materialize()
+    utc = db._utc_t.tolist()
+    local = db._local_t.tolist()
+    off = db._off_t.tolist()
+    # 2024-11-03 06:00 UTC: EDT(-4h) -> EST(-5h), overlap
+    fall = int(datetime.datetime(2024, 11, 3, 6, tzinfo=datetime.timezone.utc)
+               .timestamp()) * 1_000_000
+    i = utc.index(fall)
+    assert off[i] == -5 * 3600
+    assert off[i - 1] == -4 * 3600
+    assert local[i] == fall + off[i - 1] * 1_000_000  # offsetBefore
+    # 2024-03-10 07:00 UTC: EST(-5h) -> EDT(-4h), gap
+    spring = int(datetime.datetime(2024, 3, 10, 7, tzinfo=datetime.timezone.utc)
+                 .timestamp()) * 1_000_000
+    j = utc.index(spring)
+    assert off[j] == -4 * 3600
+    assert local[j] == spring + off[j] * 1_000_000  # offsetAfter
+
+
+def test_ambiguous_local_time_resolution_matches_java():
+    # During the repeated 1:00-2:00 AM hour on 2024-11-03 in New York the
+    # earlier offset (EDT, -4h) must win, matching java.time and Spark.
+    import bisect
+    from spark_rapids_jni_amd.tz import GpuTimeZoneDB
+    db = GpuTimeZoneDB(device="cpu")
+    db.load("America/New_York")
+    db._materialize()
+    local = db._local_t.tolist()
+    off = db._off_t.tolist()
+
+    def to_utc(local_us):
+        k = bisect.bisect_right(local, local_us) - 1
+        return local_us - off[k] * 1_000_000
+
+    # 01:30 local on 2024-11-03 is ambiguous; java resolves to EDT (-4h)
+    amb = int(datetime.datetime(2024, 11, 3, 1, 30).replace(
+        tzinfo=datetime.timezone.utc).timestamp()) * 1_000_000
+    expect = amb + 4 * 3600 * 1_000_000  # EDT: utc = local + 4h
+    assert to_utc(amb) == expect
