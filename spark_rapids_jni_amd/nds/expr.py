@@ -129,6 +129,15 @@ class Coalesce(Expr):
         self.exprs = [_wrap(e) for e in exprs]
 
 
+class Abs(Expr):
+    def __init__(self, e):
+        self.e = _wrap(e)
+
+
+def abs_(e) -> Abs:
+    return Abs(e)
+
+
 def col(name: str) -> Col:
     return Col(name)
 
@@ -186,6 +195,10 @@ def eval_expr(e: Expr, env: dict, nrows: int, device) -> Val:
         if v is None:
             data = torch.zeros(nrows, dtype=torch.int64, device=device)
             return Val(data, torch.zeros(nrows, dtype=torch.bool, device=device))
+        if isinstance(v, str):
+            # standalone string literal: a single-entry-dictionary column
+            return Val(torch.zeros(nrows, dtype=torch.int64, device=device),
+                       None, [v])
         if isinstance(v, bool):
             dt = torch.bool
         elif isinstance(v, int):
@@ -193,9 +206,11 @@ def eval_expr(e: Expr, env: dict, nrows: int, device) -> Val:
         elif isinstance(v, float):
             dt = torch.float64
         else:
-            raise TypeError(f"literal {v!r}: string literals only combine with "
-                            "dict-encoded columns via ==/isin/like")
+            raise TypeError(f"unsupported literal {v!r}")
         return Val(torch.full((nrows,), v, dtype=dt, device=device))
+    if isinstance(e, Abs):
+        v = eval_expr(e.e, env, nrows, device)
+        return Val(v.data.abs(), v.valid)
     if isinstance(e, BinOp):
         return _eval_binop(e, env, nrows, device)
     if isinstance(e, Not):
@@ -310,6 +325,18 @@ def _eval_binop(e: BinOp, env, nrows, device) -> Val:
             return Val(res, col_v.valid)
     l = eval_expr(e.l, env, nrows, device)
     r = eval_expr(e.r, env, nrows, device)
+    if (l.dict is not None and r.dict is not None
+            and l.dict is not r.dict and e.op in ("==", "!=")):
+        # two dict-encoded columns with different dictionaries: remap both
+        # into a merged dictionary before comparing codes
+        merged = {s: i for i, s in enumerate(sorted(set(l.dict) |
+                                                    set(r.dict)))}
+        lmap = torch.tensor([merged[s] for s in l.dict], dtype=torch.int64,
+                            device=device)
+        rmap = torch.tensor([merged[s] for s in r.dict], dtype=torch.int64,
+                            device=device)
+        l = Val(lmap[l.data.long()], l.valid)
+        r = Val(rmap[r.data.long()], r.valid)
     if e.op == "and":
         ld, rd = l.data.bool(), r.data.bool()
         data = ld & rd

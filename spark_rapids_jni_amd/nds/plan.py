@@ -47,6 +47,7 @@ class Filter(Plan):
 class Project(Plan):
     child: Plan
     outs: List[Tuple[str, Expr]]   # (name, expr); replaces the schema
+    extend: bool = False           # True: keep existing columns, add outs
 
 
 @dataclass
@@ -54,7 +55,7 @@ class Join(Plan):
     left: Plan
     right: Plan
     on: List[Tuple[str, str]]      # (left_col, right_col)
-    how: str = "inner"             # inner | left | semi | anti
+    how: str = "inner"             # inner | left | full | semi | anti
     # build side is always `right`; put the smaller input there
 
 
@@ -121,6 +122,13 @@ class Frame:
             safe = idx.clamp(min=0)
         for name, v in self.cols.items():
             if null_for_neg:
+                if v.data.numel() == 0:
+                    # empty side of an outer join: all rows are null
+                    data = torch.zeros(n, dtype=v.data.dtype,
+                                       device=idx.device)
+                    out[name] = Val(data, torch.zeros(
+                        n, dtype=torch.bool, device=idx.device), v.dict)
+                    continue
                 data = v.data[safe]
                 valid = v.valid[safe] if v.valid is not None else \
                     torch.ones(n, dtype=torch.bool, device=data.device)
@@ -199,6 +207,11 @@ class GpuBackend:
         if how in ("semi", "anti"):
             sel = tbl.semi_join(pcols, anti=(how == "anti"))
             return sel, None
+        if how == "full":
+            from ..ops.join import make_full_outer, make_left_outer
+            bi, pi, matched = tbl.inner_join(pcols, track_build_matches=True)
+            fb, fp = make_full_outer(matched, bi, pi, nleft)
+            return fp, fb  # -1 entries on both sides
         bi, pi = tbl.inner_join(pcols)
         if how == "left":
             from ..ops.join import make_left_outer
@@ -301,10 +314,17 @@ class CpuBackend:
         offs = np.repeat(lo, cnt) + (np.arange(cnt.sum()) -
                                      np.repeat(np.cumsum(cnt) - cnt, cnt))
         ri = rs_idx[offs]
-        if how == "left":
+        if how in ("left", "full"):
             miss = np.nonzero(cnt == 0)[0]
             li = np.concatenate([li, miss])
             ri = np.concatenate([ri, np.full(len(miss), -1, dtype=ri.dtype)])
+        if how == "full":
+            hit = np.zeros(nr, dtype=bool)
+            hit[ri[ri >= 0]] = True
+            miss_r = np.nonzero(~hit)[0]
+            li = np.concatenate([li, np.full(len(miss_r), -1,
+                                             dtype=li.dtype)])
+            ri = np.concatenate([ri, miss_r])
         return (torch.from_numpy(li.astype(np.int64)).to(dev),
                 torch.from_numpy(ri.astype(np.int64)).to(dev))
 
@@ -408,6 +428,28 @@ class Engine:
     def register(self, name: str, frame: Frame):
         self.temps[name] = frame
 
+    def const_frame(self, **cols) -> Frame:
+        """One-row frame from python scalars (scalar-subquery composition
+        queries like q9/q28/q88)."""
+        out = {}
+        for name, v in cols.items():
+            if v is None:
+                out[name] = Val(torch.zeros(1, dtype=torch.float64,
+                                            device=self.device),
+                                torch.zeros(1, dtype=torch.bool,
+                                            device=self.device))
+            elif isinstance(v, str):
+                out[name] = Val(torch.zeros(1, dtype=torch.int64,
+                                            device=self.device), None, [v])
+            elif isinstance(v, int):
+                out[name] = Val(torch.full((1,), v, dtype=torch.int64,
+                                           device=self.device))
+            else:
+                out[name] = Val(torch.full((1,), float(v),
+                                           dtype=torch.float64,
+                                           device=self.device))
+        return Frame(out, 1)
+
     def scalar(self, plan: Plan, col: Optional[str] = None):
         f = self.run(plan)
         name = col or f.names()[0]
@@ -459,7 +501,7 @@ class Engine:
 
     def _exec_project(self, p: Project) -> Frame:
         f = self._exec(p.child)
-        out = {}
+        out = dict(f.cols) if p.extend else {}
         for name, e in p.outs:
             out[name] = eval_expr(_wrap(e), f.cols, f.nrows, self.device)
         return Frame(out, f.nrows, f.sharded)
@@ -477,13 +519,15 @@ class Engine:
             out = lf.gather(li)
             out.sharded = sharded
             return out
-        left_out = lf.gather(li)
-        rnames = {b for _, b in p.on}
+        left_out = lf.gather(li, null_for_neg=(p.how == "full"))
+        # right join-key columns equal the left keys for inner/left; drop
+        # them (full outer keeps both so queries can coalesce)
+        rnames = {b for _, b in p.on} if p.how != "full" else set()
         out = dict(left_out.cols)
-        right_g = rf.gather(ri, null_for_neg=(p.how == "left"))
+        right_g = rf.gather(ri, null_for_neg=(p.how in ("left", "full")))
         for name, v in right_g.cols.items():
             if name in rnames:
-                continue  # equal to the left key for inner; drop
+                continue
             if name in out:
                 raise ValueError(f"join output column clash: {name} "
                                  "(Project/rename one side first)")
@@ -510,6 +554,11 @@ class Engine:
             keynames = keys + list(extra_keys.keys())
         else:
             keynames = list(keys)
+        dummy_key = not kv
+        if dummy_key:  # global aggregate: constant key, dropped at the end
+            kv = [Val(torch.zeros(f.nrows, dtype=torch.int64,
+                                  device=self.device))]
+            keynames = ["__k0"]
 
         native = []          # (fn, Val|None)
         slots = []           # per plain agg: dict of native indices
@@ -574,10 +623,30 @@ class Engine:
                 var = (ssq.data - mean_sq) / (cnt - 1).clamp(min=1)
                 out[name] = Val(var.clamp(min=0).sqrt(), ok)
         result = Frame(out, ng, sharded=False)
+        if dummy_key and ng == 0:
+            # SQL global aggregate over empty input still yields one row:
+            # counts are 0, everything else null
+            cols = {"__k0": Val(torch.zeros(1, dtype=torch.int64,
+                                            device=self.device))}
+            for n_, f_, _x in plain:
+                if f_ == "count":
+                    cols[n_] = Val(torch.zeros(1, dtype=torch.int64,
+                                               device=self.device))
+                else:
+                    cols[n_] = Val(torch.zeros(1, dtype=torch.float64,
+                                               device=self.device),
+                                   torch.zeros(1, dtype=torch.bool,
+                                               device=self.device))
+            result = Frame(cols, 1)
+            ng = 1
 
         for name, fn, e in countd:
             d = self._count_distinct(f, keys, e, extra_keys)
-            result = self._merge_on_keys(result, d, keynames, name)
+            result = self._merge_on_keys(result, d,
+                                         keynames if not dummy_key else [],
+                                         name)
+        if dummy_key:
+            result.cols.pop("__k0", None)
         return result
 
     @staticmethod
@@ -630,6 +699,10 @@ class Engine:
             kvals, _r, ng = self.backend.groupby(kv2, [("count", None)])
             dist = Frame(self._kv_dict(keynames + ["__v"], kvals, []), ng)
         # count per key
+        if not keynames:  # global count distinct
+            return Frame({"__cd": Val(torch.full(
+                (1,), dist.nrows, dtype=torch.int64,
+                device=self.device))}, 1)
         kv3 = [dist.cols[k] for k in keynames]
         kvals, rvals, ng = self.backend.groupby(kv3, [("count", None)])
         out = self._kv_dict(keynames, kvals, [])
@@ -843,6 +916,20 @@ class Engine:
         total = sum(f.nrows for f in frames)
         for name in names:
             protos = [f.cols[name] for f in frames]
+            dicts = [v.dict for v in protos]
+            d = None
+            if any(x is not None for x in dicts):
+                assert all(x is not None for x in dicts), \
+                    f"union mixes dict and plain column {name}"
+                if all(x is dicts[0] for x in dicts):
+                    d = dicts[0]
+                else:  # merge dictionaries, remap codes per branch
+                    d = sorted(set().union(*[set(x) for x in dicts]))
+                    code = {s: i for i, s in enumerate(d)}
+                    protos = [Val(torch.tensor(
+                        [code[s] for s in v.dict], dtype=torch.int64,
+                        device=self.device)[v.data.long()], v.valid, d)
+                        for v in protos]
             dt = protos[0].data.dtype
             for v in protos[1:]:
                 dt = torch.promote_types(dt, v.data.dtype)
@@ -854,11 +941,6 @@ class Engine:
                     for v, f in zip(protos, frames)])
             else:
                 valid = None
-            d = None
-            for v in protos:
-                if v.dict is not None:
-                    d = v.dict
-                    break
             out[name] = Val(data, valid, d)
         return Frame(out, total)
 

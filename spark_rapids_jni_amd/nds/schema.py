@@ -302,7 +302,9 @@ def _gen_customer_address(g: _Gen) -> Dataset:
           dic=[f"{10000 + 89 * k % 90000:05d}" for k in range(nzip)])
     d.add("ca_country", np.zeros(n, dtype=np.int64),
           valid=r.random(n) >= 0.005, dic=["United States"])
-    d.add("ca_gmt_offset", r.integers(-10, -4, n).astype(np.float64),
+    # cyclic so every offset value is guaranteed present (queries filter
+    # on gmt_offset == -5)
+    d.add("ca_gmt_offset", (-5.0 - (np.arange(n) % 4)).astype(np.float64),
           valid=r.random(n) >= 0.01)
     d.add("ca_location_type", *_dict_col(
         r, n, ["apartment", "condo", "single family"], 0.01))
@@ -368,7 +370,7 @@ def _gen_store(g: _Gen) -> Dataset:
     d.add("s_company_id", comp)
     d.add("s_company_name", comp - 1,
           dic=[f"company_{k+1}" for k in range(ncomp)])
-    d.add("s_gmt_offset", r.integers(-10, -4, n).astype(np.float64))
+    d.add("s_gmt_offset", (-5.0 - (np.arange(n) % 4)).astype(np.float64))
     d.add("s_market_id", r.integers(1, 11, n))
     return d
 
@@ -475,25 +477,33 @@ def _gen_store_sales(g: _Gen) -> Dataset:
     r = g.rng("store_sales")
     n = g.nrows("store_sales")
     d = Dataset("store_sales", n, sharded=True)
-    dk, dkv = g.date_fk(r, n, 0.02)
-    d.add("ss_sold_date_sk", dk, dkv)
-    d.add("ss_sold_time_sk", r.integers(0, 86_400, n),
-          valid=r.random(n) >= 0.02)
+    # ticket-level attributes: all line items of one ticket share date,
+    # time, customer, demographics, address and store (as dsdgen does) —
+    # group-by-ticket queries (q34/q46/q68/q73/q79) depend on this
+    nt = max(n // 12, 1)
+    tk = r.integers(0, nt, n)
+
+    def per_ticket(vals, valid):
+        return vals[tk], (valid[tk] if valid is not None else None)
+
+    d.add("ss_sold_date_sk", *per_ticket(*g.date_fk(r, nt, 0.02)))
+    tt = r.integers(0, 86_400, nt)
+    ttv = r.random(nt) >= 0.02
+    d.add("ss_sold_time_sk", tt[tk], ttv[tk])
     d.add("ss_item_sk", *g.fk(r, n, g.sizes["item"]))
-    ck, ckv = g.fk(r, n, g.sizes["customer"], 0.03)
-    d.add("ss_customer_sk", ck, ckv)
-    cd, cdv = g.fk(r, n, g.sizes["customer_demographics"], 0.03)
-    d.add("ss_cdemo_sk", cd, cdv)
-    hd, hdv = g.fk(r, n, g.sizes["household_demographics"], 0.03)
-    d.add("ss_hdemo_sk", hd, hdv)
-    ca, cav = g.fk(r, n, g.sizes["customer_address"], 0.03)
-    d.add("ss_addr_sk", ca, cav)
-    st, stv = g.fk(r, n, g.sizes["store"], 0.02)
-    d.add("ss_store_sk", st, stv)
+    d.add("ss_customer_sk",
+          *per_ticket(*g.fk(r, nt, g.sizes["customer"], 0.03)))
+    d.add("ss_cdemo_sk",
+          *per_ticket(*g.fk(r, nt, g.sizes["customer_demographics"], 0.03)))
+    d.add("ss_hdemo_sk",
+          *per_ticket(*g.fk(r, nt, g.sizes["household_demographics"],
+                            0.03)))
+    d.add("ss_addr_sk",
+          *per_ticket(*g.fk(r, nt, g.sizes["customer_address"], 0.03)))
+    d.add("ss_store_sk", *per_ticket(*g.fk(r, nt, g.sizes["store"], 0.02)))
     pr, prv = g.fk(r, n, g.sizes["promotion"], 0.02)
     d.add("ss_promo_sk", pr, prv)
-    d.add("ss_ticket_number",
-          r.integers(1, max(n // 2, 2), n, dtype=np.int64))
+    d.add("ss_ticket_number", tk + 1)
     d.add("ss_quantity", r.integers(1, 101, n), valid=r.random(n) >= 0.02)
     whole = g.money(r, n, 1, 100)
     lst = np.round(whole * r.uniform(1.0, 2.0, n), 2)
@@ -566,39 +576,49 @@ def _gen_catalog_sales(g: _Gen) -> Dataset:
     r = g.rng("catalog_sales")
     n = g.nrows("catalog_sales")
     d = Dataset("catalog_sales", n, sharded=True)
-    dk, dkv = g.date_fk(r, n, 0.02)
-    d.add("cs_sold_date_sk", dk, dkv)
-    d.add("cs_sold_time_sk", r.integers(0, 86_400, n))
-    d.add("cs_ship_date_sk", dk + r.integers(1, 90, n), dkv)
-    ck, ckv = g.fk(r, n, g.sizes["customer"], 0.02)
-    d.add("cs_bill_customer_sk", ck, ckv)
-    cd, cdv = g.fk(r, n, g.sizes["customer_demographics"], 0.02)
-    d.add("cs_bill_cdemo_sk", cd, cdv)
-    hd, hdv = g.fk(r, n, g.sizes["household_demographics"], 0.02)
-    d.add("cs_bill_hdemo_sk", hd, hdv)
-    ba, bav = g.fk(r, n, g.sizes["customer_address"], 0.02)
-    d.add("cs_bill_addr_sk", ba, bav)
+    # order-level attributes (dates, customers, addresses, call center,
+    # ship mode) are shared by an order's line items; warehouse and
+    # catalog page stay per-item (q16/q94-style multi-warehouse orders)
+    nt = max(n // 10, 1)
+    tk = r.integers(0, nt, n)
+
+    def per_order(vals, valid):
+        return vals[tk], (valid[tk] if valid is not None else None)
+
+    dk, dkv = g.date_fk(r, nt, 0.02)
+    d.add("cs_sold_date_sk", *per_order(dk, dkv))
+    tt = r.integers(0, 86_400, nt)
+    d.add("cs_sold_time_sk", tt[tk])
+    sh = dk + r.integers(1, 90, nt)
+    d.add("cs_ship_date_sk", *per_order(sh, dkv))
+    ck, ckv = g.fk(r, nt, g.sizes["customer"], 0.02)
+    d.add("cs_bill_customer_sk", *per_order(ck, ckv))
+    cd, cdv = g.fk(r, nt, g.sizes["customer_demographics"], 0.02)
+    d.add("cs_bill_cdemo_sk", *per_order(cd, cdv))
+    hd, hdv = g.fk(r, nt, g.sizes["household_demographics"], 0.02)
+    d.add("cs_bill_hdemo_sk", *per_order(hd, hdv))
+    ba, bav = g.fk(r, nt, g.sizes["customer_address"], 0.02)
+    d.add("cs_bill_addr_sk", *per_order(ba, bav))
     # ship customer == bill customer for ~85% of orders
     ship = ck.copy()
-    other = r.integers(1, g.sizes["customer"] + 1, n)
-    swap = r.random(n) < 0.15
+    other = r.integers(1, g.sizes["customer"] + 1, nt)
+    swap = r.random(nt) < 0.15
     ship[swap] = other[swap]
-    d.add("cs_ship_customer_sk", ship, ckv)
-    sa, sav = g.fk(r, n, g.sizes["customer_address"], 0.02)
-    d.add("cs_ship_addr_sk", sa, sav)
-    cc, ccv = g.fk(r, n, g.sizes["call_center"], 0.02)
-    d.add("cs_call_center_sk", cc, ccv)
+    d.add("cs_ship_customer_sk", *per_order(ship, ckv))
+    sa, sav = g.fk(r, nt, g.sizes["customer_address"], 0.02)
+    d.add("cs_ship_addr_sk", *per_order(sa, sav))
+    cc, ccv = g.fk(r, nt, g.sizes["call_center"], 0.02)
+    d.add("cs_call_center_sk", *per_order(cc, ccv))
     cp, cpv = g.fk(r, n, g.sizes["catalog_page"], 0.02)
     d.add("cs_catalog_page_sk", cp, cpv)
-    sm, smv = g.fk(r, n, g.sizes["ship_mode"], 0.02)
-    d.add("cs_ship_mode_sk", sm, smv)
+    sm, smv = g.fk(r, nt, g.sizes["ship_mode"], 0.02)
+    d.add("cs_ship_mode_sk", *per_order(sm, smv))
     wh, whv = g.fk(r, n, g.sizes["warehouse"], 0.02)
     d.add("cs_warehouse_sk", wh, whv)
     d.add("cs_item_sk", *g.fk(r, n, g.sizes["item"]))
     pr, prv = g.fk(r, n, g.sizes["promotion"], 0.02)
     d.add("cs_promo_sk", pr, prv)
-    d.add("cs_order_number",
-          r.integers(1, max(n // 2, 2), n, dtype=np.int64))
+    d.add("cs_order_number", tk + 1)
     d.add("cs_quantity", r.integers(1, 101, n), valid=r.random(n) >= 0.02)
     whole = g.money(r, n, 1, 100)
     lst = np.round(whole * r.uniform(1.0, 2.0, n), 2)
@@ -671,38 +691,43 @@ def _gen_web_sales(g: _Gen) -> Dataset:
     r = g.rng("web_sales")
     n = g.nrows("web_sales")
     d = Dataset("web_sales", n, sharded=True)
-    dk, dkv = g.date_fk(r, n, 0.02)
-    d.add("ws_sold_date_sk", dk, dkv)
-    d.add("ws_sold_time_sk", r.integers(0, 86_400, n))
-    d.add("ws_ship_date_sk", dk + r.integers(1, 90, n), dkv)
+    nt = max(n // 8, 1)
+    tk = r.integers(0, nt, n)
+
+    def per_order(vals, valid):
+        return vals[tk], (valid[tk] if valid is not None else None)
+
+    dk, dkv = g.date_fk(r, nt, 0.02)
+    d.add("ws_sold_date_sk", *per_order(dk, dkv))
+    d.add("ws_sold_time_sk", r.integers(0, 86_400, nt)[tk])
+    d.add("ws_ship_date_sk", *per_order(dk + r.integers(1, 90, nt), dkv))
     d.add("ws_item_sk", *g.fk(r, n, g.sizes["item"]))
-    ck, ckv = g.fk(r, n, g.sizes["customer"], 0.02)
-    d.add("ws_bill_customer_sk", ck, ckv)
-    cd, cdv = g.fk(r, n, g.sizes["customer_demographics"], 0.02)
-    d.add("ws_bill_cdemo_sk", cd, cdv)
-    hd, hdv = g.fk(r, n, g.sizes["household_demographics"], 0.02)
-    d.add("ws_bill_hdemo_sk", hd, hdv)
-    ba, bav = g.fk(r, n, g.sizes["customer_address"], 0.02)
-    d.add("ws_bill_addr_sk", ba, bav)
+    ck, ckv = g.fk(r, nt, g.sizes["customer"], 0.02)
+    d.add("ws_bill_customer_sk", *per_order(ck, ckv))
+    cd, cdv = g.fk(r, nt, g.sizes["customer_demographics"], 0.02)
+    d.add("ws_bill_cdemo_sk", *per_order(cd, cdv))
+    hd, hdv = g.fk(r, nt, g.sizes["household_demographics"], 0.02)
+    d.add("ws_bill_hdemo_sk", *per_order(hd, hdv))
+    ba, bav = g.fk(r, nt, g.sizes["customer_address"], 0.02)
+    d.add("ws_bill_addr_sk", *per_order(ba, bav))
     ship = ck.copy()
-    other = r.integers(1, g.sizes["customer"] + 1, n)
-    swap = r.random(n) < 0.15
+    other = r.integers(1, g.sizes["customer"] + 1, nt)
+    swap = r.random(nt) < 0.15
     ship[swap] = other[swap]
-    d.add("ws_ship_customer_sk", ship, ckv)
-    sa, sav = g.fk(r, n, g.sizes["customer_address"], 0.02)
-    d.add("ws_ship_addr_sk", sa, sav)
-    wp, wpv = g.fk(r, n, g.sizes["web_page"], 0.02)
-    d.add("ws_web_page_sk", wp, wpv)
-    wsit, wsitv = g.fk(r, n, g.sizes["web_site"], 0.02)
-    d.add("ws_web_site_sk", wsit, wsitv)
-    sm, smv = g.fk(r, n, g.sizes["ship_mode"], 0.02)
-    d.add("ws_ship_mode_sk", sm, smv)
+    d.add("ws_ship_customer_sk", *per_order(ship, ckv))
+    sa, sav = g.fk(r, nt, g.sizes["customer_address"], 0.02)
+    d.add("ws_ship_addr_sk", *per_order(sa, sav))
+    wp, wpv = g.fk(r, nt, g.sizes["web_page"], 0.02)
+    d.add("ws_web_page_sk", *per_order(wp, wpv))
+    wsit, wsitv = g.fk(r, nt, g.sizes["web_site"], 0.02)
+    d.add("ws_web_site_sk", *per_order(wsit, wsitv))
+    sm, smv = g.fk(r, nt, g.sizes["ship_mode"], 0.02)
+    d.add("ws_ship_mode_sk", *per_order(sm, smv))
     wh, whv = g.fk(r, n, g.sizes["warehouse"], 0.02)
     d.add("ws_warehouse_sk", wh, whv)
     pr, prv = g.fk(r, n, g.sizes["promotion"], 0.02)
     d.add("ws_promo_sk", pr, prv)
-    d.add("ws_order_number",
-          r.integers(1, max(n // 2, 2), n, dtype=np.int64))
+    d.add("ws_order_number", tk + 1)
     d.add("ws_quantity", r.integers(1, 101, n), valid=r.random(n) >= 0.02)
     whole = g.money(r, n, 1, 100)
     lst = np.round(whole * r.uniform(1.0, 2.0, n), 2)
