@@ -137,11 +137,18 @@ def main():
     ap.add_argument("--probe-rows", type=int, default=PROBE_ROWS_DEFAULT,
                     help="probe rows per GPU per step (default = named config)")
     ap.add_argument("--chunk-rows", type=int, default=CHUNK_ROWS)
-    ap.add_argument("--op", choices=["join", "groupby", "q3", "parquet"],
+    ap.add_argument("--op", choices=["join", "groupby", "q3", "parquet",
+                                     "nds"],
                     default="join",
                     help="flagship join (default), BASELINE config[1] "
-                         "group-by, the NDS q3-shaped pipeline, or the "
-                         "config[3] parquet scan->filter->project")
+                         "group-by, the NDS q3-shaped pipeline, the "
+                         "config[3] parquet scan->filter->project, or the "
+                         "config[4] NDS q1-q99 power run")
+    ap.add_argument("--sf", type=float, default=10.0,
+                    help="scale factor for --op nds")
+    ap.add_argument("--queries", type=str, default="",
+                    help="comma-separated query numbers for --op nds "
+                         "(default all 99)")
     ap.add_argument("--groups", type=int, default=1_000_000,
                     help="distinct groups for --op groupby")
     ap.add_argument("--codec", choices=["none", "snappy"], default="none",
@@ -149,6 +156,8 @@ def main():
     args = ap.parse_args()
 
     world, rank, local_rank = setup_dist(args)
+    if args.op == "nds":
+        return run_nds(args, world, rank, local_rank)
     if args.op != "join":
         return run_secondary(args, world, rank, local_rank)
     if args.gpus > 1 and world != args.gpus:
@@ -232,6 +241,59 @@ def main():
                 "seq_len": 0,
                 "parallelism": f"1 executor/GPU x{world}, RCCL all-to-all shuffle",
             },
+        }))
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+def run_nds(args, world, rank, local_rank):
+    """BASELINE config[4]: NDS q1-q99 power run. One step = one full power
+    run of all (or --queries selected) queries; data is generated once per
+    rank (facts sharded, dims replicated) and stays GPU-resident. Fixed SF
+    split over N ranks = strong scaling."""
+    device = torch.device("cuda", local_rank)
+    from spark_rapids_jni_amd.nds import runner as nds_runner
+    qs = ([int(x) for x in args.queries.split(",") if x] or None)
+    log(rank, f"[bench] generating NDS catalog sf={args.sf} "
+              f"(world={world}, rank={rank})")
+    eng = nds_runner.make_engine(args.sf, device=f"cuda:{local_rank}",
+                                 world=world, rank=rank)
+    log(rank, "[bench] warmup")
+    for _ in range(args.warmup):
+        nds_runner.power_run(eng, qs, quiet=True)
+    barrier_sync(world)
+    t0 = time.perf_counter()
+    per = None
+    for _ in range(args.steps):
+        per = nds_runner.power_run(eng, qs, quiet=True)
+    barrier_sync(world)
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    nq = per["n_queries"] if per else 0
+    if rank == 0:
+        print(json.dumps({
+            "metric": "nds_power_run_seconds",
+            "value": elapsed / args.steps,
+            "unit": "s",
+            "n_gpus": world, "steps": args.steps, "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "int64+fp64",
+            "data": "synthetic",
+            "config": {"model": f"NDS q1-q99 power run (synthetic "
+                                f"TPC-DS-shaped, SF{args.sf:g})",
+                       "n_queries": nq,
+                       "global_batch": 0, "seq_len": 0,
+                       "parallelism": f"1 executor/GPU x{world}, "
+                                      "RCCL exchange",
+                       "per_query": per["queries"] if per else {}},
         }))
     if world > 1:
         import torch.distributed as dist

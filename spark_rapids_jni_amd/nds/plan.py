@@ -891,7 +891,12 @@ class Engine:
         if f.sharded and self.world > 1:
             f = self._allgather(f)
             f.sharded = False
-        perm = self._argsort(f, p.by)
+        # append the remaining columns as tie-breakers so a following Limit
+        # cuts deterministically regardless of execution order (both
+        # backends and any rank count agree on the kept rows)
+        named = {n for n, _a in p.by}
+        by = list(p.by) + [(n, True) for n in f.names() if n not in named]
+        perm = self._argsort(f, by)
         return f.gather(perm)
 
     def _exec_limit(self, p: Limit) -> Frame:

@@ -1981,3 +1981,1090 @@ def q65(e: Engine) -> Frame:
           [("ss_item_sk", "i_item_sk")])
     return e.run(Limit(Sort(p, [("s_store_name", True),
                                 ("i_item_desc", True)]), 100))
+
+
+def q66(e: Engine) -> Frame:
+    chans = []
+    for tbl, pre, price, qty in (
+            ("web_sales", "ws", "ws_ext_sales_price", "ws_quantity"),
+            ("catalog_sales", "cs", "cs_ext_sales_price", "cs_quantity")):
+        p = J(S(tbl, f"{pre}_sold_date_sk", f"{pre}_sold_time_sk",
+                f"{pre}_warehouse_sk", f"{pre}_ship_mode_sk", price, qty),
+              dd(col("d_year") == 2001, "d_year", "d_moy"),
+              [(f"{pre}_sold_date_sk", "d_date_sk")])
+        p = J(p, F(S("time_dim", "t_time_sk", "t_time"),
+                   col("t_time").between(30838, 30838 + 28800)),
+              [(f"{pre}_sold_time_sk", "t_time_sk")])
+        p = J(p, F(S("ship_mode", "sm_ship_mode_sk", "sm_carrier"),
+                   col("sm_carrier").isin(["carrier_1", "carrier_2"])),
+              [(f"{pre}_ship_mode_sk", "sm_ship_mode_sk")])
+        p = J(p, S("warehouse", "w_warehouse_sk", "w_warehouse_name",
+                   "w_warehouse_sq_ft", "w_city", "w_county", "w_state",
+                   "w_country"),
+              [(f"{pre}_warehouse_sk", "w_warehouse_sk")])
+        aggs = []
+        for m in range(1, 13):
+            aggs.append((f"sales_m{m}", "sum",
+                         case_when((col("d_moy") == m, col(price)),
+                                   otherwise=lit(0.0))))
+            aggs.append((f"net_m{m}", "sum",
+                         case_when((col("d_moy") == m,
+                                    col(qty).cast_float()),
+                                   otherwise=lit(0.0))))
+        chans.append(A(p, ["w_warehouse_name", "w_warehouse_sq_ft",
+                           "w_city", "w_county", "w_state", "w_country"],
+                       aggs))
+    p = A(Union(chans), ["w_warehouse_name", "w_warehouse_sq_ft", "w_city",
+                         "w_county", "w_state", "w_country"],
+          [(f"{k}_m{m}", "sum", col(f"{k}_m{m}"))
+           for m in range(1, 13) for k in ("sales", "net")])
+    return e.run(Limit(Sort(p, [("w_warehouse_name", True)]), 100))
+
+
+def q67(e: Engine) -> Frame:
+    p = J(S("store_sales", "ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+            "ss_sales_price", "ss_quantity"),
+          dd(col("d_month_seq").between(1200, 1211), "d_month_seq",
+             "d_year", "d_qoy", "d_moy"),
+          [("ss_sold_date_sk", "d_date_sk")])
+    p = J(p, S("store", "s_store_sk", "s_store_id"),
+          [("ss_store_sk", "s_store_sk")])
+    p = J(p, S("item", "i_item_sk", "i_category", "i_class", "i_brand",
+               "i_product_name"),
+          [("ss_item_sk", "i_item_sk")])
+    p = A(p, ["i_category", "i_class", "i_brand", "i_product_name",
+              "d_year", "d_qoy", "d_moy", "s_store_id"],
+          [("sumsales", "sum", col("ss_sales_price") *
+            col("ss_quantity").cast_float())], rollup=True)
+    p = Window(p, ["i_category"],
+               [("rk", "rank", None, [("sumsales", False)])])
+    p = F(p, col("rk") <= 100)
+    return e.run(Limit(Sort(p, [("i_category", True), ("rk", True)]), 100))
+
+
+def q68(e: Engine) -> Frame:
+    p = J(S("store_sales", "ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+            "ss_addr_sk", "ss_customer_sk", "ss_ticket_number",
+            "ss_ext_sales_price", "ss_ext_list_price", "ss_ext_tax"),
+          dd(col("d_dom").between(1, 2) & col("d_year").isin(
+              [1999, 2000, 2001]), "d_dom", "d_year"),
+          [("ss_sold_date_sk", "d_date_sk")])
+    p = J(p, F(S("store", "s_store_sk", "s_city"),
+               col("s_city").isin(["city_00", "city_01"])),
+          [("ss_store_sk", "s_store_sk")])
+    p = J(p, F(S("household_demographics", "hd_demo_sk", "hd_dep_count",
+                 "hd_vehicle_count"),
+               (col("hd_dep_count") == 4) | (col("hd_vehicle_count") == 3)),
+          [("ss_hdemo_sk", "hd_demo_sk")])
+    p = J(p, P(S("customer_address", "ca_address_sk", "ca_city"),
+               ("ca_address_sk", col("ca_address_sk")),
+               ("bought_city", col("ca_city"))),
+          [("ss_addr_sk", "ca_address_sk")])
+    p = A(p, ["ss_ticket_number", "ss_customer_sk", "bought_city"],
+          [("extended_price", "sum", col("ss_ext_sales_price")),
+           ("list_price", "sum", col("ss_ext_list_price")),
+           ("extended_tax", "sum", col("ss_ext_tax"))])
+    p = J(p, S("customer", "c_customer_sk", "c_current_addr_sk",
+               "c_last_name", "c_first_name"),
+          [("ss_customer_sk", "c_customer_sk")])
+    p = J(p, P(S("customer_address", "ca_address_sk", "ca_city"),
+               ("ca2", col("ca_address_sk")),
+               ("current_city", col("ca_city"))),
+          [("c_current_addr_sk", "ca2")])
+    p = F(p, ~(col("current_city") == col("bought_city")))
+    return e.run(Limit(Sort(p, [("c_last_name", True),
+                                ("ss_ticket_number", True)]), 100))
+
+
+def q69(e: Engine) -> Frame:
+    win = dd((col("d_year") == 2001) & col("d_moy").between(4, 6),
+             "d_year", "d_moy")
+    c = J(S("customer", "c_customer_sk", "c_current_addr_sk",
+            "c_current_cdemo_sk"),
+          F(S("customer_address", "ca_address_sk", "ca_state"),
+            col("ca_state").isin(["KY", "GA", "NM"])),
+          [("c_current_addr_sk", "ca_address_sk")])
+    ss = J(S("store_sales", "ss_customer_sk", "ss_sold_date_sk"), win,
+           [("ss_sold_date_sk", "d_date_sk")])
+    c = J(c, P(ss, ("cust", col("ss_customer_sk"))),
+          [("c_customer_sk", "cust")], how="semi")
+    ws = J(S("web_sales", "ws_bill_customer_sk", "ws_sold_date_sk"), win,
+           [("ws_sold_date_sk", "d_date_sk")])
+    c = J(c, P(ws, ("cust", col("ws_bill_customer_sk"))),
+          [("c_customer_sk", "cust")], how="anti")
+    cs = J(S("catalog_sales", "cs_ship_customer_sk", "cs_sold_date_sk"),
+           win, [("cs_sold_date_sk", "d_date_sk")])
+    c = J(c, P(cs, ("cust", col("cs_ship_customer_sk"))),
+          [("c_customer_sk", "cust")], how="anti")
+    p = J(c, S("customer_demographics", "cd_demo_sk", "cd_gender",
+               "cd_marital_status", "cd_education_status",
+               "cd_purchase_estimate", "cd_credit_rating"),
+          [("c_current_cdemo_sk", "cd_demo_sk")])
+    p = A(p, ["cd_gender", "cd_marital_status", "cd_education_status",
+              "cd_purchase_estimate", "cd_credit_rating"],
+          [("cnt1", "count", None)])
+    return e.run(Limit(Sort(p, [("cd_gender", True),
+                                ("cd_marital_status", True),
+                                ("cd_education_status", True)]), 100))
+
+
+def q70(e: Engine) -> Frame:
+    base = J(S("store_sales", "ss_sold_date_sk", "ss_store_sk",
+               "ss_net_profit"),
+             dd(col("d_month_seq").between(1200, 1211), "d_month_seq"),
+             [("ss_sold_date_sk", "d_date_sk")])
+    by_state = A(J(base, S("store", "s_store_sk", "s_state"),
+                   [("ss_store_sk", "s_store_sk")]),
+                 ["s_state"], [("profit", "sum", col("ss_net_profit"))])
+    top5 = Window(by_state, [], [("rk", "rank", None,
+                                  [("profit", False)])])
+    top5 = P(F(top5, col("rk") <= 5), ("top_state", col("s_state")))
+    e.register("q70top", e.run(top5))
+    p = J(base, S("store", "s_store_sk", "s_state", "s_county"),
+          [("ss_store_sk", "s_store_sk")])
+    p = J(p, S("q70top"), [("s_state", "top_state")], how="semi")
+    p = A(p, ["s_state", "s_county"],
+          [("total_sum", "sum", col("ss_net_profit"))], rollup=True)
+    p = P(p, ("s_state", col("s_state")), ("s_county", col("s_county")),
+          ("lochierarchy", lit(2) - col("__lvl")),
+          ("total_sum", col("total_sum")),
+          ("rank_state", case_when((col("__lvl") == 2, col("s_state")),
+                                   otherwise=lit(None))))
+    p = Window(p, ["lochierarchy", "rank_state"],
+               [("rank_within_parent", "rank", None,
+                 [("total_sum", False)])])
+    return e.run(Limit(Sort(p, [("lochierarchy", False),
+                                ("s_state", True),
+                                ("rank_within_parent", True)]), 100))
+
+
+def q71(e: Engine) -> Frame:
+    chans = []
+    for tbl, pre in (("web_sales", "ws"), ("catalog_sales", "cs"),
+                     ("store_sales", "ss")):
+        p = J(S(tbl, f"{pre}_sold_date_sk", f"{pre}_sold_time_sk",
+                f"{pre}_item_sk", f"{pre}_ext_sales_price"),
+              dd((col("d_moy") == 11) & (col("d_year") == 1999), "d_moy",
+                 "d_year"),
+              [(f"{pre}_sold_date_sk", "d_date_sk")])
+        chans.append(P(p, ("ext_price", col(f"{pre}_ext_sales_price")),
+                       ("item_sk", col(f"{pre}_item_sk")),
+                       ("time_sk", col(f"{pre}_sold_time_sk"))))
+    p = J(Union(chans), F(S("item", "i_item_sk", "i_brand_id", "i_brand",
+                            "i_manager_id"), col("i_manager_id") == 1),
+          [("item_sk", "i_item_sk")])
+    p = J(p, F(S("time_dim", "t_time_sk", "t_hour", "t_minute",
+                 "t_meal_time"),
+               (col("t_meal_time") == "breakfast") |
+               (col("t_meal_time") == "dinner")),
+          [("time_sk", "t_time_sk")])
+    p = A(p, ["i_brand_id", "i_brand", "t_hour", "t_minute"],
+          [("ext_price", "sum", col("ext_price"))])
+    return e.run(Limit(Sort(p, [("ext_price", False),
+                                ("i_brand_id", True)]), 100))
+
+
+def q72(e: Engine) -> Frame:
+    p = J(S("catalog_sales", "cs_sold_date_sk", "cs_ship_date_sk",
+            "cs_item_sk", "cs_order_number", "cs_quantity",
+            "cs_bill_cdemo_sk", "cs_bill_hdemo_sk", "cs_promo_sk"),
+          dd(col("d_year") == 2001, "d_year", "d_week_seq", "d_date"),
+          [("cs_sold_date_sk", "d_date_sk")])
+    p = J(p, F(S("customer_demographics", "cd_demo_sk",
+                 "cd_marital_status"), col("cd_marital_status") == "D"),
+          [("cs_bill_cdemo_sk", "cd_demo_sk")])
+    p = J(p, F(S("household_demographics", "hd_demo_sk",
+                 "hd_buy_potential"),
+               col("hd_buy_potential") == ">10000"),
+          [("cs_bill_hdemo_sk", "hd_demo_sk")])
+    # inventory joined on (item, week): inventory date -> week via date_dim
+    inv = J(S("inventory", "inv_date_sk", "inv_item_sk",
+              "inv_quantity_on_hand"),
+            P(S("date_dim", "d_date_sk", "d_week_seq"),
+              ("invdsk", col("d_date_sk")), ("inv_week", col("d_week_seq"))),
+            [("inv_date_sk", "invdsk")])
+    p = J(p, P(inv, ("inv_item", col("inv_item_sk")),
+               ("inv_week", col("inv_week")),
+               ("inv_qoh", col("inv_quantity_on_hand"))),
+          [("cs_item_sk", "inv_item"), ("d_week_seq", "inv_week")])
+    p = F(p, (col("inv_qoh") < col("cs_quantity")) &
+          (col("cs_ship_date_sk") > col("cs_sold_date_sk") + 5))
+    p = J(p, S("item", "i_item_sk", "i_item_desc"),
+          [("cs_item_sk", "i_item_sk")])
+    p = J(p, S("warehouse", "w_warehouse_sk", "w_warehouse_name"), [],
+          how="inner") if False else p
+    p = J(p, P(S("promotion", "p_promo_sk"), ("pp", col("p_promo_sk")),
+               ("has_promo", lit(1))),
+          [("cs_promo_sk", "pp")], how="left")
+    p = J(p, P(S("catalog_returns", "cr_item_sk", "cr_order_number"),
+               ("cri", col("cr_item_sk")), ("cro", col("cr_order_number")),
+               ("returned", lit(1))),
+          [("cs_item_sk", "cri"), ("cs_order_number", "cro")], how="left")
+    p = A(p, ["i_item_desc", "d_week_seq"],
+          [("no_promo", "sum", case_when((is_null(col("has_promo")),
+                                          lit(1)), otherwise=lit(0))),
+           ("promo", "sum", case_when((is_not_null(col("has_promo")),
+                                       lit(1)), otherwise=lit(0))),
+           ("total_cnt", "count", None)])
+    return e.run(Limit(Sort(p, [("total_cnt", False),
+                                ("i_item_desc", True),
+                                ("d_week_seq", True)]), 100))
+
+
+def q73(e: Engine) -> Frame:
+    p = J(S("store_sales", "ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+            "ss_customer_sk", "ss_ticket_number"),
+          dd((col("d_dom").between(1, 2)) & col("d_year").isin(
+              [1999, 2000, 2001]), "d_dom", "d_year"),
+          [("ss_sold_date_sk", "d_date_sk")])
+    p = J(p, F(S("store", "s_store_sk", "s_county"),
+               col("s_county").isin(["county_00", "county_01", "county_02",
+                                     "county_03"])),
+          [("ss_store_sk", "s_store_sk")])
+    p = J(p, F(S("household_demographics", "hd_demo_sk",
+                 "hd_buy_potential", "hd_vehicle_count", "hd_dep_count"),
+               col("hd_buy_potential").isin([">10000", "Unknown"]) &
+               (col("hd_vehicle_count") > 0) &
+               (col("hd_dep_count").cast_float() /
+                col("hd_vehicle_count").cast_float() > 1.0)),
+          [("ss_hdemo_sk", "hd_demo_sk")])
+    p = F(A(p, ["ss_ticket_number", "ss_customer_sk"],
+            [("cnt", "count", None)]), col("cnt").between(1, 5))
+    p = J(p, S("customer", "c_customer_sk", "c_last_name", "c_first_name",
+               "c_salutation", "c_preferred_cust_flag"),
+          [("ss_customer_sk", "c_customer_sk")])
+    return e.run(Limit(Sort(p, [("cnt", False), ("c_last_name", True)]),
+                       1000))
+
+
+def q74(e: Engine) -> Frame:
+    def yt(tbl, pre, cc, year):
+        p = J(S(tbl, f"{pre}_sold_date_sk", cc, f"{pre}_net_paid"),
+              dd(col("d_year") == year, "d_year"),
+              [(f"{pre}_sold_date_sk", "d_date_sk")])
+        p = J(p, S("customer", "c_customer_sk", "c_customer_id",
+                   "c_first_name", "c_last_name"),
+              [(cc, "c_customer_sk")])
+        return A(p, ["c_customer_id", "c_first_name", "c_last_name"],
+                 [("year_total", "sum", col(f"{pre}_net_paid"))])
+    e.register("t74_s1", e.run(yt("store_sales", "ss", "ss_customer_sk",
+                                  2001)))
+    e.register("t74_s2", e.run(yt("store_sales", "ss", "ss_customer_sk",
+                                  2002)))
+    e.register("t74_w1", e.run(yt("web_sales", "ws", "ws_bill_customer_sk",
+                                  2001)))
+    e.register("t74_w2", e.run(yt("web_sales", "ws", "ws_bill_customer_sk",
+                                  2002)))
+    p = J(P(F(S("t74_s1"), col("year_total") > 0), ("cid", col("c_customer_id")),
+            ("fn", col("c_first_name")), ("ln", col("c_last_name")),
+            ("ts1", col("year_total"))),
+          P(S("t74_s2"), ("cid2", col("c_customer_id")),
+            ("ts2", col("year_total"))),
+          [("cid", "cid2")])
+    p = J(p, P(F(S("t74_w1"), col("year_total") > 0),
+               ("cid3", col("c_customer_id")), ("tw1", col("year_total"))),
+          [("cid", "cid3")])
+    p = J(p, P(S("t74_w2"), ("cid4", col("c_customer_id")),
+               ("tw2", col("year_total"))),
+          [("cid", "cid4")])
+    p = F(p, col("tw2") / col("tw1") > col("ts2") / col("ts1"))
+    p = P(p, ("customer_id", col("cid")), ("customer_first_name", col("fn")),
+          ("customer_last_name", col("ln")))
+    return e.run(Limit(Sort(p, [("customer_id", True)]), 100))
+
+
+def q75(e: Engine) -> Frame:
+    chans = []
+    for tbl, pre, rtbl, rpre, okey in (
+            ("store_sales", "ss", "store_returns", "sr", "ticket_number"),
+            ("catalog_sales", "cs", "catalog_returns", "cr",
+             "order_number"),
+            ("web_sales", "ws", "web_returns", "wr", "order_number")):
+        amt = ("cr_return_amount" if rpre == "cr" else
+               f"{rpre}_return_amt")
+        p = J(S(tbl, f"{pre}_sold_date_sk", f"{pre}_item_sk",
+                f"{pre}_{okey}", f"{pre}_quantity", f"{pre}_ext_sales_price"),
+              dd(col("d_year").isin([2001, 2002]), "d_year"),
+              [(f"{pre}_sold_date_sk", "d_date_sk")])
+        p = J(p, F(S("item", "i_item_sk", "i_brand_id", "i_class_id",
+                     "i_category_id", "i_category", "i_manufact_id"),
+                   col("i_category") == "Books"),
+              [(f"{pre}_item_sk", "i_item_sk")])
+        p = J(p, P(S(rtbl, f"{rpre}_item_sk", f"{rpre}_{okey}",
+                     f"{rpre}_return_quantity", amt),
+                   ("ri", col(f"{rpre}_item_sk")),
+                   ("ro", col(f"{rpre}_{okey}")),
+                   ("rq", col(f"{rpre}_return_quantity")),
+                   ("ra", col(amt))),
+              [(f"{pre}_item_sk", "ri"), (f"{pre}_{okey}", "ro")],
+              how="left")
+        chans.append(P(p, ("d_year", col("d_year")),
+                       ("i_brand_id", col("i_brand_id")),
+                       ("i_class_id", col("i_class_id")),
+                       ("i_category_id", col("i_category_id")),
+                       ("i_manufact_id", col("i_manufact_id")),
+                       ("qty", col(f"{pre}_quantity") -
+                        coalesce(col("rq"), lit(0))),
+                       ("amt", col(f"{pre}_ext_sales_price") -
+                        coalesce(col("ra"), lit(0.0)))))
+    total = A(Union(chans), ["d_year", "i_brand_id", "i_class_id",
+                             "i_category_id", "i_manufact_id"],
+              [("sales_cnt", "sum", col("qty")),
+               ("sales_amt", "sum", col("amt"))])
+    e.register("q75all", e.run(total))
+    cur = P(F(S("q75all"), col("d_year") == 2002),
+            ("b2", col("i_brand_id")), ("c2", col("i_class_id")),
+            ("g2", col("i_category_id")), ("m2", col("i_manufact_id")),
+            ("cnt2", col("sales_cnt")), ("amt2", col("sales_amt")))
+    prev = P(F(S("q75all"), col("d_year") == 2001),
+             ("b1", col("i_brand_id")), ("c1", col("i_class_id")),
+             ("g1", col("i_category_id")), ("m1", col("i_manufact_id")),
+             ("cnt1", col("sales_cnt")), ("amt1", col("sales_amt")))
+    p = J(cur, prev, [("b2", "b1"), ("c2", "c1"), ("g2", "g1"),
+                      ("m2", "m1")])
+    p = F(p, col("cnt2").cast_float() / col("cnt1").cast_float() < 0.9)
+    p = P(p, ("prev_year", lit(2001)), ("year_", lit(2002)),
+          ("i_brand_id", col("b2")), ("i_class_id", col("c2")),
+          ("i_category_id", col("g2")), ("i_manufact_id", col("m2")),
+          ("prev_cnt", col("cnt1")), ("curr_cnt", col("cnt2")),
+          ("sales_cnt_diff", col("cnt2") - col("cnt1")),
+          ("sales_amt_diff", col("amt2") - col("amt1")))
+    return e.run(Limit(Sort(p, [("sales_cnt_diff", True)]), 100))
+
+
+def q76(e: Engine) -> Frame:
+    chans = []
+    specs = (("store_sales", "ss", "ss_store_sk", "ss_item_sk",
+              "ss_ext_sales_price", "ss_sold_date_sk", "store"),
+             ("web_sales", "ws", "ws_ship_customer_sk", "ws_item_sk",
+              "ws_ext_sales_price", "ws_sold_date_sk", "web"),
+             ("catalog_sales", "cs", "cs_ship_addr_sk", "cs_item_sk",
+              "cs_ext_sales_price", "cs_sold_date_sk", "catalog"))
+    for tbl, pre, nullcol, isk, price, datecol, label in specs:
+        p = F(S(tbl, nullcol, isk, price, datecol),
+              is_null(col(nullcol)))
+        p = J(p, S("item", "i_item_sk", "i_category"),
+              [(isk, "i_item_sk")])
+        p = J(p, S("date_dim", "d_date_sk", "d_year", "d_qoy"),
+              [(datecol, "d_date_sk")])
+        chans.append(P(p, ("channel", lit(label)),
+                       ("col_name", lit(nullcol)),
+                       ("d_year", col("d_year")), ("d_qoy", col("d_qoy")),
+                       ("i_category", col("i_category")),
+                       ("ext_sales_price", col(price))))
+    p = A(Union(chans), ["channel", "col_name", "d_year", "d_qoy",
+                         "i_category"],
+          [("sales_cnt", "count", None),
+           ("sales_amt", "sum", col("ext_sales_price"))])
+    return e.run(Limit(Sort(p, [("channel", True), ("col_name", True),
+                                ("d_year", True), ("d_qoy", True),
+                                ("i_category", True)]), 100))
+
+
+def q77(e: Engine) -> Frame:
+    lo, hi = dsk("2000-08-23"), dsk("2000-09-22")
+    win = lambda c: col(c).between(lo, hi)
+    ss = A(J(F(S("store_sales", "ss_sold_date_sk", "ss_store_sk",
+               "ss_ext_sales_price", "ss_net_profit"),
+              win("ss_sold_date_sk")),
+            S("store", "s_store_sk"), [("ss_store_sk", "s_store_sk")]),
+           ["ss_store_sk"],
+           [("sales", "sum", col("ss_ext_sales_price")),
+            ("profit", "sum", col("ss_net_profit"))])
+    sr = A(J(F(S("store_returns", "sr_returned_date_sk", "sr_store_sk",
+               "sr_return_amt", "sr_net_loss"),
+              win("sr_returned_date_sk")),
+            S("store", "s_store_sk"), [("sr_store_sk", "s_store_sk")]),
+           ["sr_store_sk"],
+           [("returns_", "sum", col("sr_return_amt")),
+            ("profit_loss", "sum", col("sr_net_loss"))])
+    st = J(ss, P(sr, ("srsk", col("sr_store_sk")),
+                 ("returns_", col("returns_")),
+                 ("profit_loss", col("profit_loss"))),
+           [("ss_store_sk", "srsk")], how="left")
+    st = P(st, ("channel", lit("store channel")), ("id", col("ss_store_sk")),
+           ("sales", col("sales")),
+           ("returns_", coalesce(col("returns_"), lit(0.0))),
+           ("profit", col("profit") - coalesce(col("profit_loss"),
+                                               lit(0.0))))
+    cs_sales = e.run(A(F(S("catalog_sales", "cs_sold_date_sk",
+                           "cs_call_center_sk", "cs_ext_sales_price",
+                           "cs_net_profit"), win("cs_sold_date_sk")),
+                       ["cs_call_center_sk"],
+                       [("sales", "sum", col("cs_ext_sales_price")),
+                        ("profit", "sum", col("cs_net_profit"))]))
+    e.register("q77cs", cs_sales)
+    cr_tot = e.run(A(F(S("catalog_returns", "cr_returned_date_sk",
+                         "cr_return_amount", "cr_net_loss"),
+                       win("cr_returned_date_sk")), [],
+                     [("returns_", "sum", col("cr_return_amount")),
+                      ("profit_loss", "sum", col("cr_net_loss"))]))
+    rr = cr_tot.to_rows()
+    cr_r, cr_l = (rr[0] if rr else (0.0, 0.0))
+    ct = P(S("q77cs"), ("channel", lit("catalog channel")),
+           ("id", col("cs_call_center_sk")), ("sales", col("sales")),
+           ("returns_", lit(float(cr_r or 0.0))),
+           ("profit", col("profit") - lit(float(cr_l or 0.0))))
+    ws = A(J(F(S("web_sales", "ws_sold_date_sk", "ws_web_page_sk",
+               "ws_ext_sales_price", "ws_net_profit"),
+              win("ws_sold_date_sk")),
+            S("web_page", "wp_web_page_sk"),
+            [("ws_web_page_sk", "wp_web_page_sk")]),
+           ["ws_web_page_sk"],
+           [("sales", "sum", col("ws_ext_sales_price")),
+            ("profit", "sum", col("ws_net_profit"))])
+    wr = A(J(F(S("web_returns", "wr_returned_date_sk", "wr_web_page_sk",
+               "wr_return_amt", "wr_net_loss"),
+              win("wr_returned_date_sk")),
+            S("web_page", "wp_web_page_sk"),
+            [("wr_web_page_sk", "wp_web_page_sk")]),
+           ["wr_web_page_sk"],
+           [("returns_", "sum", col("wr_return_amt")),
+            ("profit_loss", "sum", col("wr_net_loss"))])
+    wt = J(ws, P(wr, ("wrsk", col("wr_web_page_sk")),
+                 ("returns_", col("returns_")),
+                 ("profit_loss", col("profit_loss"))),
+           [("ws_web_page_sk", "wrsk")], how="left")
+    wt = P(wt, ("channel", lit("web channel")), ("id", col("ws_web_page_sk")),
+           ("sales", col("sales")),
+           ("returns_", coalesce(col("returns_"), lit(0.0))),
+           ("profit", col("profit") - coalesce(col("profit_loss"),
+                                               lit(0.0))))
+    p = A(Union([st, ct, wt]), ["channel", "id"],
+          [("sales", "sum", col("sales")),
+           ("returns_", "sum", col("returns_")),
+           ("profit", "sum", col("profit"))], rollup=True)
+    return e.run(Limit(Sort(p, [("__lvl", True), ("channel", True),
+                                ("id", True)]), 100))
+
+
+def q78(e: Engine) -> Frame:
+    def no_return(tbl, pre, rtbl, rpre, okey, cust):
+        p = J(S(tbl, f"{pre}_sold_date_sk", f"{pre}_item_sk", cust,
+                f"{pre}_{okey}", f"{pre}_quantity", f"{pre}_wholesale_cost",
+                f"{pre}_sales_price"),
+              P(S(rtbl, f"{rpre}_item_sk", f"{rpre}_{okey}"),
+                ("ri", col(f"{rpre}_item_sk")),
+                ("ro", col(f"{rpre}_{okey}"))),
+              [(f"{pre}_item_sk", "ri"), (f"{pre}_{okey}", "ro")],
+              how="anti")
+        p = J(p, S("date_dim", "d_date_sk", "d_year"),
+              [(f"{pre}_sold_date_sk", "d_date_sk")])
+        return A(p, ["d_year", f"{pre}_item_sk", cust],
+                 [(f"{pre}_qty", "sum", col(f"{pre}_quantity")),
+                  (f"{pre}_wc", "sum", col(f"{pre}_wholesale_cost")),
+                  (f"{pre}_sp", "sum", col(f"{pre}_sales_price"))])
+    ss = no_return("store_sales", "ss", "store_returns", "sr",
+                   "ticket_number", "ss_customer_sk")
+    ws = no_return("web_sales", "ws", "web_returns", "wr", "order_number",
+                   "ws_bill_customer_sk")
+    cs = no_return("catalog_sales", "cs", "catalog_returns", "cr",
+                   "order_number", "cs_bill_customer_sk")
+    p = J(F(ss, col("d_year") == 2000),
+          P(F(ws, col("d_year") == 2000), ("wyi", col("ws_item_sk")),
+            ("wyc", col("ws_bill_customer_sk")), ("ws_qty", col("ws_qty")),
+            ("ws_wc", col("ws_wc")), ("ws_sp", col("ws_sp"))),
+          [("ss_item_sk", "wyi"), ("ss_customer_sk", "wyc")])
+    p = J(p, P(F(cs, col("d_year") == 2000), ("cyi", col("cs_item_sk")),
+               ("cyc", col("cs_bill_customer_sk")),
+               ("cs_qty", col("cs_qty")), ("cs_wc", col("cs_wc")),
+               ("cs_sp", col("cs_sp"))),
+          [("ss_item_sk", "cyi"), ("ss_customer_sk", "cyc")], how="left")
+    p = F(p, coalesce(col("ws_qty"), lit(0)) +
+          coalesce(col("cs_qty"), lit(0)) > 0)
+    p = P(p, ("d_year", col("d_year")), ("ss_item_sk", col("ss_item_sk")),
+          ("ss_customer_sk", col("ss_customer_sk")),
+          ("ratio", col("ss_qty").cast_float() /
+           (coalesce(col("ws_qty"), lit(0)) +
+            coalesce(col("cs_qty"), lit(0))).cast_float()),
+          ("store_qty", col("ss_qty")),
+          ("store_wholesale_cost", col("ss_wc")),
+          ("store_sales_price", col("ss_sp")),
+          ("other_chan_qty", coalesce(col("ws_qty"), lit(0)) +
+           coalesce(col("cs_qty"), lit(0))))
+    return e.run(Limit(Sort(p, [("ratio", True), ("ss_qty", True)
+                                if False else ("store_qty", True)]), 100))
+
+
+def q79(e: Engine) -> Frame:
+    p = J(S("store_sales", "ss_sold_date_sk", "ss_store_sk", "ss_hdemo_sk",
+            "ss_customer_sk", "ss_ticket_number", "ss_addr_sk",
+            "ss_coupon_amt", "ss_net_profit"),
+          dd((col("d_dow") == 1) & col("d_year").isin([1999, 2000, 2001]),
+             "d_dow", "d_year"),
+          [("ss_sold_date_sk", "d_date_sk")])
+    p = J(p, F(S("store", "s_store_sk", "s_number_employees", "s_city"),
+               col("s_number_employees").between(200, 295)),
+          [("ss_store_sk", "s_store_sk")])
+    p = J(p, F(S("household_demographics", "hd_demo_sk", "hd_dep_count",
+                 "hd_vehicle_count"),
+               (col("hd_dep_count") == 6) | (col("hd_vehicle_count") > 2)),
+          [("ss_hdemo_sk", "hd_demo_sk")])
+    p = A(p, ["ss_ticket_number", "ss_customer_sk", "s_city"],
+          [("amt", "sum", col("ss_coupon_amt")),
+           ("profit", "sum", col("ss_net_profit"))])
+    p = J(p, S("customer", "c_customer_sk", "c_last_name", "c_first_name"),
+          [("ss_customer_sk", "c_customer_sk")])
+    return e.run(Limit(Sort(p, [("c_last_name", True),
+                                ("c_first_name", True),
+                                ("ss_ticket_number", True)]), 100))
+
+
+def q80(e: Engine) -> Frame:
+    lo, hi = dsk("2000-08-23"), dsk("2000-09-22")
+    chans = []
+    for (tbl, pre, rtbl, rpre, okey, sitecol, sitetbl, sitesk, siteid,
+         label) in (
+            ("store_sales", "ss", "store_returns", "sr", "ticket_number",
+             "ss_store_sk", "store", "s_store_sk", "s_store_id", "store"),
+            ("catalog_sales", "cs", "catalog_returns", "cr",
+             "order_number", "cs_catalog_page_sk", "catalog_page",
+             "cp_catalog_page_sk", "cp_catalog_page_id", "catalog"),
+            ("web_sales", "ws", "web_returns", "wr", "order_number",
+             "ws_web_site_sk", "web_site", "web_site_sk", "web_site_id",
+             "web")):
+        amt = ("cr_return_amount" if rpre == "cr" else
+               f"{rpre}_return_amt")
+        p = J(F(S(tbl, f"{pre}_sold_date_sk", f"{pre}_item_sk",
+                  f"{pre}_{okey}", f"{pre}_promo_sk", sitecol,
+                  f"{pre}_ext_sales_price", f"{pre}_net_profit"),
+                col(f"{pre}_sold_date_sk").between(lo, hi)),
+              F(S("item", "i_item_sk", "i_current_price"),
+                col("i_current_price") > 50.0),
+              [(f"{pre}_item_sk", "i_item_sk")])
+        p = J(p, F(S("promotion", "p_promo_sk", "p_channel_tv"),
+                   col("p_channel_tv") == "N"),
+              [(f"{pre}_promo_sk", "p_promo_sk")])
+        p = J(p, P(S(rtbl, f"{rpre}_item_sk", f"{rpre}_{okey}", amt,
+                     f"{rpre}_net_loss"),
+                   ("ri", col(f"{rpre}_item_sk")),
+                   ("ro", col(f"{rpre}_{okey}")),
+                   ("ramt", col(amt)), ("rloss", col(f"{rpre}_net_loss"))),
+              [(f"{pre}_item_sk", "ri"), (f"{pre}_{okey}", "ro")],
+              how="left")
+        p = J(p, S(sitetbl, sitesk, siteid), [(sitecol, sitesk)])
+        g = A(p, [siteid],
+              [("sales", "sum", col(f"{pre}_ext_sales_price")),
+               ("returns_", "sum", coalesce(col("ramt"), lit(0.0))),
+               ("profit", "sum", col(f"{pre}_net_profit") -
+                coalesce(col("rloss"), lit(0.0)))])
+        chans.append(P(g, ("channel", lit(label + " channel")),
+                       ("id", col(siteid)), ("sales", col("sales")),
+                       ("returns_", col("returns_")),
+                       ("profit", col("profit"))))
+    p = A(Union(chans), ["channel", "id"],
+          [("sales", "sum", col("sales")),
+           ("returns_", "sum", col("returns_")),
+           ("profit", "sum", col("profit"))], rollup=True)
+    return e.run(Limit(Sort(p, [("__lvl", True), ("channel", True),
+                                ("id", True)]), 100))
+
+
+def q81(e: Engine) -> Frame:
+    ctr = J(S("catalog_returns", "cr_returned_date_sk",
+              "cr_returning_customer_sk", "cr_returning_addr_sk",
+              "cr_return_amt_inc_tax"),
+            dd(col("d_year") == 2000, "d_year"),
+            [("cr_returned_date_sk", "d_date_sk")])
+    ctr = J(ctr, S("customer_address", "ca_address_sk", "ca_state"),
+            [("cr_returning_addr_sk", "ca_address_sk")])
+    ctr = A(ctr, ["cr_returning_customer_sk", "ca_state"],
+            [("ctr_total_return", "sum", col("cr_return_amt_inc_tax"))])
+    e.register("ctr81", e.run(ctr))
+    st_avg = A(S("ctr81"), ["ca_state"],
+               [("avg_ret", "avg", col("ctr_total_return"))])
+    p = J(S("ctr81"), P(st_avg, ("st2", col("ca_state")),
+                        ("avg_ret", col("avg_ret"))),
+          [("ca_state", "st2")])
+    p = F(p, col("ctr_total_return") > col("avg_ret") * 1.2)
+    cust = J(S("customer", "c_customer_sk", "c_customer_id",
+               "c_salutation", "c_first_name", "c_last_name",
+               "c_current_addr_sk"),
+             F(S("customer_address", "ca_address_sk", "ca_state",
+                 "ca_city", "ca_zip", "ca_country", "ca_location_type"),
+               col("ca_state") == "GA"),
+             [("c_current_addr_sk", "ca_address_sk")])
+    cust = P(cust, ("c_customer_sk", col("c_customer_sk")),
+             ("c_customer_id", col("c_customer_id")),
+             ("c_salutation", col("c_salutation")),
+             ("c_first_name", col("c_first_name")),
+             ("c_last_name", col("c_last_name")),
+             ("cust_city", col("ca_city")), ("cust_zip", col("ca_zip")),
+             ("cust_country", col("ca_country")),
+             ("cust_loc", col("ca_location_type")))
+    p = J(p, cust, [("cr_returning_customer_sk", "c_customer_sk")])
+    p = P(p, ("c_customer_id", col("c_customer_id")),
+          ("c_salutation", col("c_salutation")),
+          ("c_first_name", col("c_first_name")),
+          ("c_last_name", col("c_last_name")),
+          ("ca_city", col("cust_city")), ("ca_zip", col("cust_zip")),
+          ("ca_country", col("cust_country")),
+          ("ca_location_type", col("cust_loc")),
+          ("ctr_total_return", col("ctr_total_return")))
+    return e.run(Limit(Sort(p, [("c_customer_id", True)]), 100))
+
+
+def q82(e: Engine) -> Frame:
+    lo = dsk("2000-05-25")
+    it = F(S("item", "i_item_sk", "i_item_id", "i_item_desc",
+             "i_current_price", "i_manufact_id"),
+           col("i_current_price").between(62.0, 92.0) &
+           col("i_manufact_id").isin([12, 13, 14, 15]))
+    p = J(it, F(S("inventory", "inv_item_sk", "inv_date_sk",
+                  "inv_quantity_on_hand"),
+                col("inv_quantity_on_hand").between(100, 500) &
+                col("inv_date_sk").between(lo, lo + 60)),
+          [("i_item_sk", "inv_item_sk")])
+    p = J(p, P(S("store_sales", "ss_item_sk"), ("ssk", col("ss_item_sk"))),
+          [("i_item_sk", "ssk")], how="semi")
+    p = Distinct(P(p, ("i_item_id", col("i_item_id")),
+                   ("i_item_desc", col("i_item_desc")),
+                   ("i_current_price", col("i_current_price"))))
+    return e.run(Limit(Sort(p, [("i_item_id", True)]), 100))
+
+
+def q83(e: Engine) -> Frame:
+    dates = [date_lit("2000-06-30").value, date_lit("2000-09-27").value,
+             date_lit("2000-11-17").value]
+    weeks = Distinct(P(F(S("date_dim", "d_date_sk", "d_date",
+                           "d_week_seq"), col("d_date").isin(dates)),
+                       ("wk", col("d_week_seq"))))
+    e.register("q83w", e.run(weeks))
+    wk_dates = J(S("date_dim", "d_date_sk", "d_week_seq"),
+                 S("q83w"), [("d_week_seq", "wk")], how="semi")
+    e.register("q83d", e.run(P(wk_dates, ("dsk", col("d_date_sk")))))
+    outs = []
+    for rtbl, rpre in (("store_returns", "sr"), ("catalog_returns", "cr"),
+                       ("web_returns", "wr")):
+        p = J(S(rtbl, f"{rpre}_returned_date_sk", f"{rpre}_item_sk",
+                f"{rpre}_return_quantity"),
+              S("q83d"), [(f"{rpre}_returned_date_sk", "dsk")], how="semi")
+        p = J(p, S("item", "i_item_sk", "i_item_id"),
+              [(f"{rpre}_item_sk", "i_item_sk")])
+        outs.append(A(p, ["i_item_id"],
+                      [(f"{rpre}_qty", "sum",
+                        col(f"{rpre}_return_quantity"))]))
+    p = J(outs[0], P(outs[1], ("iid2", col("i_item_id")),
+                     ("cr_qty", col("cr_qty"))), [("i_item_id", "iid2")])
+    p = J(p, P(outs[2], ("iid3", col("i_item_id")),
+               ("wr_qty", col("wr_qty"))), [("i_item_id", "iid3")])
+    tot = (col("sr_qty") + col("cr_qty") + col("wr_qty")).cast_float()
+    p = P(p, ("item_id", col("i_item_id")), ("sr_item_qty", col("sr_qty")),
+          ("sr_dev", col("sr_qty").cast_float() / tot / 3.0 * 100.0),
+          ("cr_item_qty", col("cr_qty")),
+          ("cr_dev", col("cr_qty").cast_float() / tot / 3.0 * 100.0),
+          ("wr_item_qty", col("wr_qty")),
+          ("wr_dev", col("wr_qty").cast_float() / tot / 3.0 * 100.0),
+          ("average", tot / 3.0))
+    return e.run(Limit(Sort(p, [("item_id", True), ("sr_item_qty", True)]),
+                       100))
+
+
+def q84(e: Engine) -> Frame:
+    p = J(S("customer", "c_customer_sk", "c_customer_id", "c_first_name",
+            "c_last_name", "c_current_addr_sk", "c_current_cdemo_sk",
+            "c_current_hdemo_sk"),
+          F(S("customer_address", "ca_address_sk", "ca_city"),
+            col("ca_city") == "city_10"),
+          [("c_current_addr_sk", "ca_address_sk")])
+    p = J(p, S("household_demographics", "hd_demo_sk",
+               "hd_income_band_sk"),
+          [("c_current_hdemo_sk", "hd_demo_sk")])
+    p = J(p, F(S("income_band", "ib_income_band_sk", "ib_lower_bound",
+                 "ib_upper_bound"),
+               (col("ib_lower_bound") >= 30000) &
+               (col("ib_upper_bound") <= 80000)),
+          [("hd_income_band_sk", "ib_income_band_sk")])
+    p = J(p, P(S("store_returns", "sr_cdemo_sk"),
+               ("srcd", col("sr_cdemo_sk"))),
+          [("c_current_cdemo_sk", "srcd")], how="semi")
+    p = P(p, ("customer_id", col("c_customer_id")),
+          ("customername", col("c_last_name")),
+          ("c_first_name", col("c_first_name")))
+    return e.run(Limit(Sort(p, [("customer_id", True)]), 100))
+
+
+def q85(e: Engine) -> Frame:
+    p = J(S("web_returns", "wr_item_sk", "wr_order_number",
+            "wr_refunded_cdemo_sk", "wr_returning_cdemo_sk",
+            "wr_refunded_addr_sk", "wr_reason_sk", "wr_fee",
+            "wr_refunded_cash", "wr_return_quantity"),
+          S("web_sales", "ws_item_sk", "ws_order_number",
+            "ws_web_page_sk", "ws_sold_date_sk", "ws_quantity",
+            "ws_sales_price", "ws_net_profit"),
+          [("wr_item_sk", "ws_item_sk"),
+           ("wr_order_number", "ws_order_number")])
+    p = J(p, S("web_page", "wp_web_page_sk"),
+          [("ws_web_page_sk", "wp_web_page_sk")])
+    p = J(p, dd(col("d_year") == 2000, "d_year"),
+          [("ws_sold_date_sk", "d_date_sk")])
+    p = J(p, P(S("customer_demographics", "cd_demo_sk",
+                 "cd_marital_status", "cd_education_status"),
+               ("cd1sk", col("cd_demo_sk")),
+               ("ms1", col("cd_marital_status")),
+               ("ed1", col("cd_education_status"))),
+          [("wr_refunded_cdemo_sk", "cd1sk")])
+    p = J(p, P(S("customer_demographics", "cd_demo_sk",
+                 "cd_marital_status", "cd_education_status"),
+               ("cd2sk", col("cd_demo_sk")),
+               ("ms2", col("cd_marital_status")),
+               ("ed2", col("cd_education_status"))),
+          [("wr_returning_cdemo_sk", "cd2sk")])
+    p = J(p, S("customer_address", "ca_address_sk", "ca_country",
+               "ca_state"),
+          [("wr_refunded_addr_sk", "ca_address_sk")])
+    p = J(p, S("reason", "r_reason_sk", "r_reason_desc"),
+          [("wr_reason_sk", "r_reason_sk")])
+    m1 = ((col("ms1") == "M") & (col("ed1") == "4 yr Degree") &
+          col("ws_sales_price").between(100.0, 150.0))
+    m2 = ((col("ms1") == "S") & (col("ed1") == "College") &
+          col("ws_sales_price").between(50.0, 100.0))
+    m3 = ((col("ms1") == "W") & (col("ed1") == "2 yr Degree") &
+          col("ws_sales_price").between(150.0, 200.0))
+    s1 = (col("ca_country") == "United States") & \
+        col("ca_state").isin(["IN", "OH", "NJ"]) & \
+        col("ws_net_profit").between(100, 200)
+    s2 = (col("ca_country") == "United States") & \
+        col("ca_state").isin(["WI", "CT", "KY"]) & \
+        col("ws_net_profit").between(150, 300)
+    s3 = (col("ca_country") == "United States") & \
+        col("ca_state").isin(["LA", "IA", "AR"]) & \
+        col("ws_net_profit").between(50, 250)
+    p = F(p, ((col("ms1") == col("ms2")) & (m1 | m2 | m3)) & (s1 | s2 | s3))
+    p = A(p, ["r_reason_desc"],
+          [("avg_qty", "avg", col("ws_quantity")),
+           ("avg_fee", "avg", col("wr_fee")),
+           ("avg_cash", "avg", col("wr_refunded_cash"))])
+    return e.run(Limit(Sort(p, [("avg_qty", True), ("avg_fee", True),
+                                ("r_reason_desc", True)]), 100))
+
+
+def q86(e: Engine) -> Frame:
+    p = J(S("web_sales", "ws_sold_date_sk", "ws_item_sk", "ws_net_paid"),
+          dd(col("d_month_seq").between(1200, 1211), "d_month_seq"),
+          [("ws_sold_date_sk", "d_date_sk")])
+    p = J(p, S("item", "i_item_sk", "i_category", "i_class"),
+          [("ws_item_sk", "i_item_sk")])
+    p = A(p, ["i_category", "i_class"],
+          [("total_sum", "sum", col("ws_net_paid"))], rollup=True)
+    p = P(p, ("i_category", col("i_category")), ("i_class", col("i_class")),
+          ("lochierarchy", lit(2) - col("__lvl")),
+          ("total_sum", col("total_sum")),
+          ("rank_cat", case_when((col("__lvl") == 2, col("i_category")),
+                                 otherwise=lit(None))))
+    p = Window(p, ["lochierarchy", "rank_cat"],
+               [("rank_within_parent", "rank", None,
+                 [("total_sum", False)])])
+    return e.run(Limit(Sort(p, [("lochierarchy", False),
+                                ("i_category", True),
+                                ("rank_within_parent", True)]), 100))
+
+
+def q87(e: Engine) -> Frame:
+    win = dd(col("d_month_seq").between(1200, 1211), "d_month_seq",
+             "d_date")
+    frames = []
+    for tbl, pre, cc in (("store_sales", "ss", "ss_customer_sk"),
+                         ("catalog_sales", "cs", "cs_bill_customer_sk"),
+                         ("web_sales", "ws", "ws_bill_customer_sk")):
+        p = J(S(tbl, f"{pre}_sold_date_sk", cc), win,
+              [(f"{pre}_sold_date_sk", "d_date_sk")])
+        p = J(p, S("customer", "c_customer_sk", "c_last_name",
+                   "c_first_name"),
+              [(cc, "c_customer_sk")])
+        frames.append(e.run(Distinct(P(p, ("ln", col("c_last_name")),
+                                       ("fn", col("c_first_name")),
+                                       ("dt", col("d_date"))))))
+    e.register("q87a", frames[0])
+    e.register("q87b", frames[1])
+    e.register("q87c", frames[2])
+    p = J(S("q87a"), P(S("q87b"), ("ln2", col("ln")), ("fn2", col("fn")),
+                       ("dt2", col("dt"))),
+          [("ln", "ln2"), ("fn", "fn2"), ("dt", "dt2")], how="anti")
+    p = J(p, P(S("q87c"), ("ln3", col("ln")), ("fn3", col("fn")),
+               ("dt3", col("dt"))),
+          [("ln", "ln3"), ("fn", "fn3"), ("dt", "dt3")], how="anti")
+    return e.run(A(p, [], [("cnt", "count", None)]))
+
+
+def q88(e: Engine) -> Frame:
+    slots = [(8, 30), (9, 0), (9, 30), (10, 0), (10, 30), (11, 0),
+             (11, 30), (12, 0)]
+    out = {}
+    for i, (h, m) in enumerate(slots, 1):
+        p = J(S("store_sales", "ss_sold_time_sk", "ss_hdemo_sk",
+                "ss_store_sk"),
+              F(S("time_dim", "t_time_sk", "t_hour", "t_minute"),
+                (col("t_hour") == h) &
+                (col("t_minute").between(m, m + 29))),
+              [("ss_sold_time_sk", "t_time_sk")])
+        p = J(p, F(S("household_demographics", "hd_demo_sk",
+                     "hd_dep_count", "hd_vehicle_count"),
+                   ((col("hd_dep_count") == 4) &
+                    (col("hd_vehicle_count") <= 6)) |
+                   ((col("hd_dep_count") == 2) &
+                    (col("hd_vehicle_count") <= 4)) |
+                   ((col("hd_dep_count") == 0) &
+                    (col("hd_vehicle_count") <= 2))),
+              [("ss_hdemo_sk", "hd_demo_sk")])
+        p = J(p, F(S("store", "s_store_sk", "s_store_name"),
+                   col("s_store_name") == "store_a"),
+              [("ss_store_sk", "s_store_sk")])
+        out[f"h{h}_{m}"] = int(e.scalar(A(p, [], [("c", "count", None)]),
+                                        "c") or 0)
+    return e.const_frame(**out)
+
+
+def q89(e: Engine) -> Frame:
+    in1 = (col("i_category").isin(["Books", "Children", "Electronics"]) &
+           col("i_class").isin(["class06", "class07", "class08"]))
+    in2 = (col("i_category").isin(["Women", "Music", "Men"]) &
+           col("i_class").isin(["class09", "class10", "class11"]))
+    p = J(S("store_sales", "ss_sold_date_sk", "ss_item_sk", "ss_store_sk",
+            "ss_sales_price"),
+          dd(col("d_year") == 1999, "d_year", "d_moy"),
+          [("ss_sold_date_sk", "d_date_sk")])
+    p = J(p, F(S("item", "i_item_sk", "i_category", "i_class", "i_brand"),
+               in1 | in2),
+          [("ss_item_sk", "i_item_sk")])
+    p = J(p, S("store", "s_store_sk", "s_store_name", "s_company_name"),
+          [("ss_store_sk", "s_store_sk")])
+    p = A(p, ["i_category", "i_class", "i_brand", "s_store_name",
+              "s_company_name", "d_moy"],
+          [("sum_sales", "sum", col("ss_sales_price"))])
+    p = Window(p, ["i_category", "i_brand", "s_store_name",
+                   "s_company_name"],
+               [("avg_monthly_sales", "avg", "sum_sales")])
+    p = F(p, case_when((~(col("avg_monthly_sales") == 0.0),
+                        abs_(col("sum_sales") -
+                             col("avg_monthly_sales")) /
+                        col("avg_monthly_sales")),
+                       otherwise=lit(None)) > 0.1)
+    return e.run(Limit(Sort(p, [("sum_sales", True),
+                                ("s_store_name", True)]), 100))
+
+
+def q90(e: Engine) -> Frame:
+    def half(h1, h2):
+        p = J(S("web_sales", "ws_sold_time_sk", "ws_ship_hdemo_sk",
+                "ws_web_page_sk"),
+              F(S("time_dim", "t_time_sk", "t_hour"),
+                col("t_hour").between(h1, h2)),
+              [("ws_sold_time_sk", "t_time_sk")])
+            
+        p = J(p, F(S("household_demographics", "hd_demo_sk",
+                     "hd_dep_count"), col("hd_dep_count") == 6),
+              [("ws_ship_hdemo_sk", "hd_demo_sk")])
+        p = J(p, F(S("web_page", "wp_web_page_sk", "wp_char_count"),
+                   col("wp_char_count").between(5000, 5200)),
+              [("ws_web_page_sk", "wp_web_page_sk")])
+        return int(e.scalar(A(p, [], [("c", "count", None)]), "c") or 0)
+    am = half(8, 9)
+    pm = half(19, 20)
+    return e.const_frame(amc=am, pmc=pm,
+                         am_pm_ratio=(am / pm) if pm else None)
+
+
+def q91(e: Engine) -> Frame:
+    p = J(S("catalog_returns", "cr_returned_date_sk", "cr_call_center_sk",
+            "cr_returning_customer_sk", "cr_net_loss"),
+          dd((col("d_year") == 1998) & (col("d_moy") == 11), "d_year",
+             "d_moy"),
+          [("cr_returned_date_sk", "d_date_sk")])
+    p = J(p, S("call_center", "cc_call_center_sk", "cc_call_center_id",
+               "cc_name", "cc_manager"),
+          [("cr_call_center_sk", "cc_call_center_sk")])
+    p = J(p, S("customer", "c_customer_sk", "c_current_cdemo_sk",
+               "c_current_hdemo_sk", "c_current_addr_sk"),
+          [("cr_returning_customer_sk", "c_customer_sk")])
+    p = J(p, F(S("customer_demographics", "cd_demo_sk", "cd_marital_status",
+                 "cd_education_status"),
+               ((col("cd_marital_status") == "M") &
+                (col("cd_education_status") == "Unknown")) |
+               ((col("cd_marital_status") == "W") &
+                (col("cd_education_status") == "Advanced Degree"))),
+          [("c_current_cdemo_sk", "cd_demo_sk")])
+    p = J(p, F(S("household_demographics", "hd_demo_sk",
+                 "hd_buy_potential"),
+               col("hd_buy_potential").isin(["0-500"])),
+          [("c_current_hdemo_sk", "hd_demo_sk")])
+    p = J(p, F(S("customer_address", "ca_address_sk", "ca_gmt_offset"),
+               col("ca_gmt_offset") == -7.0),
+          [("c_current_addr_sk", "ca_address_sk")])
+    p = A(p, ["cc_call_center_id", "cc_name", "cc_manager",
+              "cd_marital_status", "cd_education_status"],
+          [("returns_loss", "sum", col("cr_net_loss"))])
+    return e.run(Sort(p, [("returns_loss", False)]))
+
+
+def q92(e: Engine) -> Frame:
+    lo = dsk("2000-01-27")
+    base = J(F(S("web_sales", "ws_sold_date_sk", "ws_item_sk",
+                 "ws_ext_discount_amt"),
+               col("ws_sold_date_sk").between(lo, lo + 90)),
+             F(S("item", "i_item_sk", "i_manufact_id"),
+               col("i_manufact_id") == 9),
+             [("ws_item_sk", "i_item_sk")])
+    e.register("q92base", e.run(base))
+    avg_d = A(S("q92base"), ["ws_item_sk"],
+              [("avg_disc", "avg", col("ws_ext_discount_amt"))])
+    p = J(S("q92base"), P(avg_d, ("isk2", col("ws_item_sk")),
+                          ("avg_disc", col("avg_disc"))),
+          [("ws_item_sk", "isk2")])
+    p = F(p, col("ws_ext_discount_amt") > col("avg_disc") * 1.3)
+    return e.run(A(p, [], [("excess_discount", "sum",
+                            col("ws_ext_discount_amt"))]))
+
+
+def q93(e: Engine) -> Frame:
+    rr = J(S("store_returns", "sr_item_sk", "sr_ticket_number",
+             "sr_reason_sk", "sr_return_quantity"),
+           F(S("reason", "r_reason_sk", "r_reason_desc"),
+             col("r_reason_desc") == "reason 28"),
+           [("sr_reason_sk", "r_reason_sk")])
+    p = J(S("store_sales", "ss_item_sk", "ss_ticket_number",
+            "ss_customer_sk", "ss_quantity", "ss_sales_price"),
+          P(rr, ("ri", col("sr_item_sk")), ("rt", col("sr_ticket_number")),
+            ("ret_qty", col("sr_return_quantity"))),
+          [("ss_item_sk", "ri"), ("ss_ticket_number", "rt")], how="left")
+    act = case_when(
+        (is_not_null(col("ret_qty")),
+         (col("ss_quantity") - col("ret_qty")).cast_float() *
+         col("ss_sales_price")),
+        otherwise=col("ss_quantity").cast_float() * col("ss_sales_price"))
+    p = A(p, ["ss_customer_sk"], [("sumsales", "sum", act)])
+    return e.run(Limit(Sort(p, [("sumsales", True),
+                                ("ss_customer_sk", True)]), 100))
+
+
+def q94(e: Engine) -> Frame:
+    lo, hi = dsk("1999-02-01"), dsk("1999-04-02")
+    ws1 = J(F(S("web_sales", "ws_ship_date_sk", "ws_ship_addr_sk",
+               "ws_web_site_sk", "ws_order_number", "ws_warehouse_sk",
+               "ws_ext_ship_cost", "ws_net_profit"),
+              col("ws_ship_date_sk").between(lo, hi)),
+            F(S("customer_address", "ca_address_sk", "ca_state"),
+              col("ca_state") == "IL"),
+            [("ws_ship_addr_sk", "ca_address_sk")])
+    ws1 = J(ws1, F(S("web_site", "web_site_sk", "web_company_name"),
+                   col("web_company_name") == "webco_0"),
+            [("ws_web_site_sk", "web_site_sk")])
+    multi_wh = P(F(A(S("web_sales", "ws_order_number", "ws_warehouse_sk"),
+                     ["ws_order_number"],
+                     [("nwh", "countd", col("ws_warehouse_sk"))]),
+                   col("nwh") > 1),
+                 ("ono", col("ws_order_number")))
+    ws1 = J(ws1, multi_wh, [("ws_order_number", "ono")], how="semi")
+    ws1 = J(ws1, P(S("web_returns", "wr_order_number"),
+                   ("rno", col("wr_order_number"))),
+            [("ws_order_number", "rno")], how="anti")
+    return e.run(A(ws1, [],
+                   [("order_count", "countd", col("ws_order_number")),
+                    ("total_shipping_cost", "sum", col("ws_ext_ship_cost")),
+                    ("total_net_profit", "sum", col("ws_net_profit"))]))
+
+
+def q95(e: Engine) -> Frame:
+    lo, hi = dsk("1999-02-01"), dsk("1999-04-02")
+    ws1 = J(F(S("web_sales", "ws_ship_date_sk", "ws_ship_addr_sk",
+               "ws_web_site_sk", "ws_order_number", "ws_ext_ship_cost",
+               "ws_net_profit"),
+              col("ws_ship_date_sk").between(lo, hi)),
+            F(S("customer_address", "ca_address_sk", "ca_state"),
+              col("ca_state") == "IL"),
+            [("ws_ship_addr_sk", "ca_address_sk")])
+    ws1 = J(ws1, F(S("web_site", "web_site_sk", "web_company_name"),
+                   col("web_company_name") == "webco_0"),
+            [("ws_web_site_sk", "web_site_sk")])
+    ws_wh = P(F(A(S("web_sales", "ws_order_number", "ws_warehouse_sk"),
+                  ["ws_order_number"],
+                  [("nwh", "countd", col("ws_warehouse_sk"))]),
+                col("nwh") > 1),
+              ("ono", col("ws_order_number")))
+    e.register("q95wh", e.run(ws_wh))
+    ws1 = J(ws1, S("q95wh"), [("ws_order_number", "ono")], how="semi")
+    ret = J(S("web_returns", "wr_order_number"),
+            S("q95wh"), [("wr_order_number", "ono")], how="semi")
+    ws1 = J(ws1, P(ret, ("rno", col("wr_order_number"))),
+            [("ws_order_number", "rno")], how="semi")
+    return e.run(A(ws1, [],
+                   [("order_count", "countd", col("ws_order_number")),
+                    ("total_shipping_cost", "sum", col("ws_ext_ship_cost")),
+                    ("total_net_profit", "sum", col("ws_net_profit"))]))
+
+
+def q96(e: Engine) -> Frame:
+    p = J(S("store_sales", "ss_sold_time_sk", "ss_hdemo_sk", "ss_store_sk"),
+          F(S("time_dim", "t_time_sk", "t_hour", "t_minute"),
+            (col("t_hour") == 20) & (col("t_minute") >= 30)),
+          [("ss_sold_time_sk", "t_time_sk")])
+    p = J(p, F(S("household_demographics", "hd_demo_sk", "hd_dep_count"),
+               col("hd_dep_count") == 7),
+          [("ss_hdemo_sk", "hd_demo_sk")])
+    p = J(p, F(S("store", "s_store_sk", "s_store_name"),
+               col("s_store_name") == "store_a"),
+          [("ss_store_sk", "s_store_sk")])
+    return e.run(A(p, [], [("cnt", "count", None)]))
+
+
+def q97(e: Engine) -> Frame:
+    win = dd(col("d_month_seq").between(1200, 1211), "d_month_seq")
+    ssci = A(J(S("store_sales", "ss_sold_date_sk", "ss_customer_sk",
+                 "ss_item_sk"), win, [("ss_sold_date_sk", "d_date_sk")]),
+             ["ss_customer_sk", "ss_item_sk"], [("c1", "count", None)])
+    csci = A(J(S("catalog_sales", "cs_sold_date_sk", "cs_bill_customer_sk",
+                 "cs_item_sk"), win, [("cs_sold_date_sk", "d_date_sk")]),
+             ["cs_bill_customer_sk", "cs_item_sk"],
+             [("c2", "count", None)])
+    p = J(P(ssci, ("cust", col("ss_customer_sk")),
+            ("item", col("ss_item_sk")), ("s_mark", lit(1))),
+          P(csci, ("cust2", col("cs_bill_customer_sk")),
+            ("item2", col("cs_item_sk")), ("c_mark", lit(1))),
+          [("cust", "cust2"), ("item", "item2")], how="full")
+    p = A(p, [], [
+        ("store_only", "sum",
+         case_when((is_not_null(col("s_mark")) & is_null(col("c_mark")),
+                    lit(1)), otherwise=lit(0))),
+        ("catalog_only", "sum",
+         case_when((is_null(col("s_mark")) & is_not_null(col("c_mark")),
+                    lit(1)), otherwise=lit(0))),
+        ("store_and_catalog", "sum",
+         case_when((is_not_null(col("s_mark")) &
+                    is_not_null(col("c_mark")), lit(1)),
+                   otherwise=lit(0)))])
+    return e.run(p)
+
+
+def q98(e: Engine) -> Frame:
+    return e.run(_ratio_by_class(e, "store_sales", "ss", "1999-02-22",
+                                 ["Sports", "Books", "Home"]))
+
+
+def q99(e: Engine) -> Frame:
+    p = J(S("catalog_sales", "cs_ship_date_sk", "cs_sold_date_sk",
+            "cs_warehouse_sk", "cs_ship_mode_sk", "cs_call_center_sk"),
+          dd(col("d_month_seq").between(1200, 1211), "d_month_seq"),
+          [("cs_ship_date_sk", "d_date_sk")])
+    p = J(p, S("warehouse", "w_warehouse_sk", "w_warehouse_name"),
+          [("cs_warehouse_sk", "w_warehouse_sk")])
+    p = J(p, S("ship_mode", "sm_ship_mode_sk", "sm_type"),
+          [("cs_ship_mode_sk", "sm_ship_mode_sk")])
+    p = J(p, S("call_center", "cc_call_center_sk", "cc_name"),
+          [("cs_call_center_sk", "cc_call_center_sk")])
+    lag = col("cs_ship_date_sk") - col("cs_sold_date_sk")
+    p = A(p, ["w_warehouse_name", "sm_type", "cc_name"],
+          [("d30", "sum", case_when((lag <= 30, lit(1)), otherwise=lit(0))),
+           ("d31_60", "sum", case_when(((lag > 30) & (lag <= 60), lit(1)),
+                                       otherwise=lit(0))),
+           ("d61_90", "sum", case_when(((lag > 60) & (lag <= 90), lit(1)),
+                                       otherwise=lit(0))),
+           ("d91_120", "sum", case_when(((lag > 90) & (lag <= 120),
+                                         lit(1)), otherwise=lit(0))),
+           ("dgt120", "sum", case_when((lag > 120, lit(1)),
+                                       otherwise=lit(0)))])
+    return e.run(Limit(Sort(p, [("w_warehouse_name", True),
+                                ("sm_type", True), ("cc_name", True)]),
+                       100))
+
+
+QUERIES = {n: globals()[f"q{n}"] for n in range(1, 100)}

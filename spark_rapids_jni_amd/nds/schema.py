@@ -107,13 +107,25 @@ def table_rows(name: str, sf: float) -> int:
     if name == "date_dim":
         return N_DATES
     base, exp, _sh = TABLES[name]
+    if name == "customer_demographics" and sf < 1.0:
+        # spec-fixed at 1.92M, but scaled below SF1 so tiny validation
+        # catalogs stay tiny (documented delta)
+        return max(int(base * sf ** 0.5), 64)
     if exp == 0.0:
         return base
     return max(int(base * sf ** exp), 8)
 
 
-def _shard(n: int, world: int, rank: int) -> int:
-    return n // world + (1 if rank < n % world else 0)
+# Fact tables are generated in NBLOCKS deterministic blocks (block b is
+# seeded by (table, b)); rank r of a world-w run owns blocks b % w == r.
+# The union of all ranks' shards is therefore byte-identical to the
+# world=1 table for ANY world size — distributed results must equal
+# single-GPU results, and the driver's scaling curve runs the same data.
+NBLOCKS = 8
+
+
+def _block_rows(n: int, b: int) -> int:
+    return n // NBLOCKS + (1 if b < n % NBLOCKS else 0)
 
 
 class _Gen:
@@ -128,15 +140,70 @@ class _Gen:
         # Python string hashing is randomized per process
         import zlib
         seed = (zlib.crc32(table.encode()) & 0xFFFF) * 10007 + 7
-        if TABLES[table][2]:  # sharded tables differ per rank
-            seed += self.rank * 131
+        return np.random.default_rng(seed)
+
+    def nrows(self, table):
+        return self.sizes[table]
+
+    # -- column helpers ----------------------------------------------------
+    @staticmethod
+    def money(r, n, lo, hi):
+        return np.round(r.uniform(lo, hi, n), 2)
+
+    @staticmethod
+    def fk(r, n, domain, null_frac=0.0):
+        arr = r.integers(1, domain + 1, n, dtype=np.int64)
+        if null_frac <= 0:
+            return arr, None
+        valid = r.random(n) >= null_frac
+        return arr, valid
+
+    def date_fk(self, r, n, null_frac=0.0):
+        day = r.integers(SELL_FIRST, SELL_LAST + 1, n, dtype=np.int64)
+        arr = day + DATE_BASE_SK
+        if null_frac <= 0:
+            return arr, None
+        valid = r.random(n) >= null_frac
+        return arr, valid
+
+
+class _BlockGen(_Gen):
+    """View of the generator for one fact block: sharded tables size to the
+    block and seed by (table, block)."""
+
+    def __init__(self, g: _Gen, block: int):
+        self.sf = g.sf
+        self.world = g.world
+        self.rank = g.rank
+        self.sizes = g.sizes
+        self.block = block
+
+    def rng(self, table):
+        import zlib
+        seed = ((zlib.crc32(table.encode()) & 0xFFFF) * 10007 + 7 +
+                self.block * 131071)
         return np.random.default_rng(seed)
 
     def nrows(self, table):
         n = self.sizes[table]
-        if TABLES[table][2] and self.world > 1:
-            return _shard(n, self.world, self.rank)
+        if TABLES[table][2]:
+            return _block_rows(n, self.block)
         return n
+
+
+def _concat_datasets(name: str, parts: List[Dataset]) -> Dataset:
+    total = sum(p.nrows for p in parts)
+    out = Dataset(name, total, sharded=True)
+    for cname in parts[0].columns:
+        arr = np.concatenate([p.columns[cname] for p in parts])
+        if any(p.valid[cname] is not None for p in parts):
+            valid = np.concatenate([
+                p.valid[cname] if p.valid[cname] is not None
+                else np.ones(p.nrows, dtype=bool) for p in parts])
+        else:
+            valid = None
+        out.add(cname, arr, valid, parts[0].dicts[cname])
+    return out
 
     # -- column helpers ----------------------------------------------------
     @staticmethod
@@ -708,6 +775,8 @@ def _gen_web_sales(g: _Gen) -> Dataset:
     d.add("ws_bill_cdemo_sk", *per_order(cd, cdv))
     hd, hdv = g.fk(r, nt, g.sizes["household_demographics"], 0.02)
     d.add("ws_bill_hdemo_sk", *per_order(hd, hdv))
+    shd, shdv = g.fk(r, nt, g.sizes["household_demographics"], 0.02)
+    d.add("ws_ship_hdemo_sk", *per_order(shd, shdv))
     ba, bav = g.fk(r, nt, g.sizes["customer_address"], 0.02)
     d.add("ws_bill_addr_sk", *per_order(ba, bav))
     ship = ck.copy()
@@ -822,26 +891,55 @@ _DIM_GENS = {
 }
 
 
+# fact -> (sales generator, ticket/order column, line items per ticket,
+#          paired returns table + generator)
+_FACTS = {
+    "store_sales": (_gen_store_sales, "ss_ticket_number", 12,
+                    "store_returns", _gen_store_returns),
+    "catalog_sales": (_gen_catalog_sales, "cs_order_number", 10,
+                      "catalog_returns", _gen_catalog_returns),
+    "web_sales": (_gen_web_sales, "ws_order_number", 8,
+                  "web_returns", _gen_web_returns),
+}
+
+
 def gen_catalog(sf: float = 0.01, world: int = 1, rank: int = 0,
                 tables: Optional[List[str]] = None) -> Dict[str, Dataset]:
-    """Generate the full NDS catalog (or a subset) at scale factor `sf`."""
+    """Generate the full NDS catalog (or a subset) at scale factor `sf`.
+
+    Sharded (fact) tables are built from this rank's deterministic blocks;
+    the union over ranks equals the world=1 table exactly.
+    """
     g = _Gen(sf, world, rank)
     want = set(tables) if tables else set(TABLES)
-    # returns need their sales table
-    if "store_returns" in want:
-        want.add("store_sales")
-    if "catalog_returns" in want:
-        want.add("catalog_sales")
-    if "web_returns" in want:
-        want.add("web_sales")
+    for sales, (_f, _t, _k, rets, _rf) in _FACTS.items():
+        if rets in want:
+            want.add(sales)
     out: Dict[str, Dataset] = {}
     for name, fn in _DIM_GENS.items():
+        if TABLES[name][2]:
+            continue  # facts handled block-wise below
         if name in want:
             out[name] = fn(g)
-    if "store_returns" in want:
-        out["store_returns"] = _gen_store_returns(g, out["store_sales"])
-    if "catalog_returns" in want:
-        out["catalog_returns"] = _gen_catalog_returns(g, out["catalog_sales"])
-    if "web_returns" in want:
-        out["web_returns"] = _gen_web_returns(g, out["web_sales"])
+    my_blocks = [b for b in range(NBLOCKS) if b % world == rank]
+    if "inventory" in want:
+        out["inventory"] = _concat_datasets(
+            "inventory", [_gen_inventory(_BlockGen(g, b))
+                          for b in my_blocks])
+    for sales, (sfn, tcol, k, rets, rfn) in _FACTS.items():
+        if sales not in want:
+            continue
+        # per-block order-id stride keeps ticket numbers globally unique
+        stride = g.sizes[sales] // NBLOCKS // k + 2
+        sparts, rparts = [], []
+        for b in my_blocks:
+            bg = _BlockGen(g, b)
+            sb = sfn(bg)
+            sb.columns[tcol] = sb.columns[tcol] + b * stride
+            sparts.append(sb)
+            if rets in want:
+                rparts.append(rfn(bg, sb))
+        out[sales] = _concat_datasets(sales, sparts)
+        if rets in want:
+            out[rets] = _concat_datasets(rets, rparts)
     return out

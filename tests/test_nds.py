@@ -1,0 +1,179 @@
+"""NDS harness tests.
+
+CPU tier: every one of the 99 queries executes on the oracle backend and a
+few have hand-checked result invariants; the distributed (world=2, gloo)
+run of representative queries matches the world=1 run exactly.
+
+GPU tier (@pytest.mark.gpu): all 99 queries cross-checked GPU kernels vs
+CPU oracle at SF 0.05 (VERDICT r01 item 1 acceptance criterion).
+"""
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+from spark_rapids_jni_amd.nds import gen_catalog
+from spark_rapids_jni_amd.nds.plan import CpuBackend, Engine
+from spark_rapids_jni_amd.nds.queries import QUERIES
+from spark_rapids_jni_amd.nds.runner import (compare_frames,
+                                             frame_sorted_rows, power_run,
+                                             verify)
+
+SF = 0.01
+
+
+@pytest.fixture(scope="module")
+def catalog():
+    return gen_catalog(sf=SF)
+
+
+def test_all_99_queries_defined():
+    assert sorted(QUERIES) == list(range(1, 100))
+
+
+@pytest.mark.parametrize("n", sorted(QUERIES))
+def test_query_runs_cpu(catalog, n):
+    e = Engine(catalog, device="cpu")
+    f = QUERIES[n](e)
+    assert f.nrows >= 0
+    # materialization must succeed (decodes dictionaries, applies nulls)
+    rows = f.to_rows()
+    assert len(rows) == f.nrows
+
+
+def test_q3_invariants(catalog):
+    e = Engine(catalog, device="cpu")
+    f = QUERIES[3](e)
+    rows = f.to_rows()
+    assert f.names() == ["d_year", "i_brand", "i_brand_id", "sum_agg"]
+    # all brands must come from manufacturer 1 items sold in November
+    sums = [r[3] for r in rows]
+    assert all(s is None or s > 0 for s in sums)
+    # sorted by year then sum desc
+    years = [r[0] for r in rows]
+    assert years == sorted(years)
+
+
+def test_q96_count_matches_manual(catalog):
+    e = Engine(catalog, device="cpu")
+    f = QUERIES[96](e)
+    (cnt,) = f.to_rows()[0]
+    # manual recompute with numpy
+    import numpy as np
+    ss = catalog["store_sales"]
+    hd = catalog["household_demographics"]
+    st = catalog["store"]
+    tsk = ss.columns["ss_sold_time_sk"]
+    tv = ss.valid["ss_sold_time_sk"]
+    hour = tsk // 3600
+    minute = (tsk % 3600) // 60
+    time_ok = (hour == 20) & (minute >= 30)
+    if tv is not None:
+        time_ok &= tv
+    dep7 = set(np.nonzero(hd.columns["hd_dep_count"] == 7)[0] + 1)
+    names = st.dicts["s_store_name"]
+    stores = {i + 1 for i, c in enumerate(st.columns["s_store_name"])
+              if names[c] == "store_a"}
+    hds = ss.columns["ss_hdemo_sk"]
+    hdv = ss.valid["ss_hdemo_sk"]
+    sts = ss.columns["ss_store_sk"]
+    stv = ss.valid["ss_store_sk"]
+    manual = 0
+    for i in np.nonzero(time_ok)[0]:
+        if hdv is not None and not hdv[i]:
+            continue
+        if stv is not None and not stv[i]:
+            continue
+        if int(hds[i]) in dep7 and int(sts[i]) in stores:
+            manual += 1
+    assert cnt == manual
+
+
+def test_sort_limit_deterministic(catalog):
+    # two engines over the same catalog produce identical Limit cuts
+    a = QUERIES[42](Engine(catalog, device="cpu"))
+    b = QUERIES[42](Engine(catalog, device="cpu"))
+    assert frame_sorted_rows(a) == frame_sorted_rows(b)
+
+
+def test_power_run_subset(catalog):
+    e = Engine(catalog, device="cpu")
+    out = power_run(e, queries=[3, 7, 42], quiet=True)
+    assert out["n_queries"] == 3
+    assert all(q["seconds"] >= 0 for q in out["queries"].values())
+
+
+# --- distributed (world=2, gloo) -------------------------------------------
+
+_DIST_QUERIES = [3, 6, 15, 16, 28, 38, 51, 95]
+
+_WORKER = r"""
+import os, sys, json
+import torch
+import torch.distributed as dist
+sys.path.insert(0, {repo!r})
+from spark_rapids_jni_amd.nds import gen_catalog
+from spark_rapids_jni_amd.nds.plan import Engine
+from spark_rapids_jni_amd.nds.queries import QUERIES
+from spark_rapids_jni_amd.nds.runner import frame_sorted_rows
+
+rank = int(os.environ["RANK"])
+dist.init_process_group("gloo")
+world = dist.get_world_size()
+cat = gen_catalog(sf={sf}, world=world, rank=rank)
+out = {{}}
+for n in {queries}:
+    e = Engine(cat, device="cpu", world=world, rank=rank)
+    f = QUERIES[n](e)
+    out[n] = frame_sorted_rows(f)
+if rank == 0:
+    with open({out!r}, "w") as fh:
+        json.dump({{str(k): [[repr(x) for x in row] for row in v]
+                   for k, v in out.items()}}, fh)
+dist.barrier()
+dist.destroy_process_group()
+"""
+
+
+def test_distributed_matches_single(tmp_path, catalog):
+    """world=2 gloo run (sharded facts + exchange/partial-agg paths) must
+    reproduce the world=1 result for representative queries."""
+    import json
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = str(tmp_path / "dist.json")
+    script = tmp_path / "worker.py"
+    script.write_text(_WORKER.format(repo=repo, sf=SF,
+                                     queries=_DIST_QUERIES, out=out))
+    env = dict(os.environ)
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29517",
+                "WORLD_SIZE": "2"})
+    procs = []
+    for r in range(2):
+        env_r = dict(env)
+        env_r["RANK"] = str(r)
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env_r,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE))
+    for p in procs:
+        so, se = p.communicate(timeout=600)
+        assert p.returncode == 0, se.decode()[-3000:]
+    with open(out) as fh:
+        dist_rows = json.load(fh)
+    for n in _DIST_QUERIES:
+        e = Engine(catalog, device="cpu")
+        f = QUERIES[n](e)
+        single = [[repr(x) for x in row] for row in frame_sorted_rows(f)]
+        got = dist_rows[str(n)]
+        assert got == single, \
+            f"q{n}: world=2 differs from world=1 " \
+            f"({len(got)} vs {len(single)} rows)"
+
+
+# --- GPU tier ---------------------------------------------------------------
+
+@pytest.mark.gpu
+def test_gpu_matches_cpu_oracle_all_queries():
+    failures = verify(sf=0.05)
+    assert not failures, f"GPU/CPU mismatches: {failures}"
