@@ -779,16 +779,18 @@ class Engine:
             perm = self._argsort(f, sort_keys) if sort_keys else perm
             inv = torch.empty_like(perm)
             inv[perm] = torch.arange(n, dtype=torch.int64, device=dev)
-            # segment boundaries on partition keys
+            # segment boundaries on partition keys (nulls group together:
+            # a data difference only counts when both rows are valid)
             bnd = torch.zeros(n, dtype=torch.bool, device=dev)
             bnd[0] = True
             for k in p.partition:
                 v = f.cols[k]
                 d = v.data[perm]
-                bnd[1:] |= d[1:] != d[:-1]
+                diff = d[1:] != d[:-1]
                 if v.valid is not None:
                     vv = v.valid[perm]
-                    bnd[1:] |= vv[1:] != vv[:-1]
+                    diff = (diff & vv[1:] & vv[:-1]) | (vv[1:] != vv[:-1])
+                bnd[1:] |= diff
             seg = torch.cumsum(bnd.to(torch.int64), 0) - 1
             nseg = int(seg[-1].item()) + 1
             seg_start = torch.zeros(nseg, dtype=torch.int64, device=dev)
@@ -845,10 +847,12 @@ class Engine:
                     for c, _asc in order:
                         v = f.cols[c]
                         d = v.data[perm]
-                        change[1:] |= d[1:] != d[:-1]
+                        diff = d[1:] != d[:-1]
                         if v.valid is not None:
                             vv = v.valid[perm]
-                            change[1:] |= vv[1:] != vv[:-1]
+                            diff = (diff & vv[1:] & vv[:-1]) | \
+                                (vv[1:] != vv[:-1])
+                        change[1:] |= diff
                     if fn == "rank":
                         idx = torch.arange(n, dtype=torch.int64, device=dev)
                         last_change = torch.cummax(

@@ -71,8 +71,11 @@ def compare_frames(got: Frame, exp: Frame, q: int, rel_tol: float = 1e-6):
     for i, (rg, rx) in enumerate(zip(g, x)):
         for a, b in zip(rg, rx):
             if isinstance(a, float) and isinstance(b, float):
-                denom = max(abs(a), abs(b), 1.0)
-                assert abs(a - b) / denom < rel_tol, \
+                # rows carry values rounded to 4 decimals; a true 1e-9
+                # accumulation difference can still land across a rounding
+                # boundary, so allow one ulp of the rounding grid
+                tol = max(2e-4, rel_tol * max(abs(a), abs(b)))
+                assert abs(a - b) <= tol, \
                     f"q{q} row {i}: {a} != {b}\n got={rg}\n exp={rx}"
             else:
                 assert a == b, f"q{q} row {i}: {a!r} != {b!r}\n" \

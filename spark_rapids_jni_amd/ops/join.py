@@ -259,7 +259,10 @@ class HashJoinTable:
                 slots = torch.zeros(self.capacity, dtype=torch.int64,
                                     device=self.build_keys[0].device)
                 desc, top, keep = pack_descriptors(self.build_keys)
-                g.join_build(desc.data_ptr(), top.data_ptr(), 1, n,
+                # hash over ALL key columns — a single-column hash here
+                # silently misses every multi-key match (r2 fix)
+                g.join_build(desc.data_ptr(), top.data_ptr(),
+                             len(self.build_keys), n,
                              slots.data_ptr(), self.capacity,
                              _native.current_stream())
                 self._generic = HashJoinTable(self.build_keys, slots,

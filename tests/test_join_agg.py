@@ -493,3 +493,28 @@ def test_empty_inputs_all_paths():
     assert kt.num_rows == 0
     kt2, res2 = groupby(e64, [(Agg.COUNT_ALL, None)], num_groups_hint=8)
     assert kt2.num_rows == 0
+
+
+@pytest.mark.gpu
+def test_multikey_semi_anti_join():
+    # r2 regression: the generic-table rebuild inside semi_join hashed only
+    # the first key column, silently missing every multi-key match
+    import random as _r
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    _r.seed(7)
+    n_b, n_p = 500, 2000
+    bk1 = [_r.randint(0, 50) for _ in range(n_b)]
+    bk2 = [_r.randint(0, 50) for _ in range(n_b)]
+    pk1 = [_r.randint(0, 60) for _ in range(n_p)]
+    pk2 = [_r.randint(0, 60) for _ in range(n_p)]
+    bset = set(zip(bk1, bk2))
+    exp_semi = sorted(i for i in range(n_p) if (pk1[i], pk2[i]) in bset)
+    b = [Column.from_pylist(bk1, DType.INT64, device="cuda"),
+         Column.from_pylist(bk2, DType.INT64, device="cuda")]
+    p = [Column.from_pylist(pk1, DType.INT64, device="cuda"),
+         Column.from_pylist(pk2, DType.INT64, device="cuda")]
+    tbl = HashJoinTable.build(b)
+    semi = sorted(tbl.semi_join(p).cpu().tolist())
+    anti = sorted(tbl.semi_join(p, anti=True).cpu().tolist())
+    assert semi == exp_semi
+    assert anti == sorted(set(range(n_p)) - set(exp_semi))
