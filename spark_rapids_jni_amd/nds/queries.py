@@ -2036,8 +2036,13 @@ def q67(e: Engine) -> Frame:
               "d_year", "d_qoy", "d_moy", "s_store_id"],
           [("sumsales", "sum", col("ss_sales_price") *
             col("ss_quantity").cast_float())], rollup=True)
+    # rank on exact cents: the sum is a multiple of 0.01, so quantizing
+    # removes float accumulation jitter that would break ties differently
+    # between execution orders
+    p = Project(p, [("rank_key", (col("sumsales") * 100.0 + 0.5)
+                     .cast_int())], extend=True)
     p = Window(p, ["i_category"],
-               [("rk", "rank", None, [("sumsales", False)])])
+               [("rk", "rank", None, [("rank_key", False)])])
     p = F(p, col("rk") <= 100)
     return e.run(Limit(Sort(p, [("i_category", True), ("rk", True)]), 100))
 
