@@ -203,10 +203,21 @@ class GpuBackend:
         from ..ops.join import HashJoinTable
         bcols = [_val_to_column(v) for v in right_keys]
         pcols = [_val_to_column(v) for v in left_keys]
-        tbl = HashJoinTable.build(bcols)
         if how in ("semi", "anti"):
+            nright = right_keys[0].data.numel() if right_keys else 0
+            if nright > 8 * max(nleft, 1):
+                # right-semi: build on the (small) left side and mark
+                # matches while streaming the huge right side — avoids a
+                # hash table over the big input for a handful of probes
+                ltbl = HashJoinTable.build(pcols)
+                matched = ltbl.mark_matches(bcols).bool()
+                sel = torch.nonzero(matched if how == "semi" else ~matched,
+                                    as_tuple=False).view(-1)
+                return sel, None
+            tbl = HashJoinTable.build(bcols)
             sel = tbl.semi_join(pcols, anti=(how == "anti"))
             return sel, None
+        tbl = HashJoinTable.build(bcols)
         if how == "full":
             from ..ops.join import make_full_outer, make_left_outer
             bi, pi, matched = tbl.inner_join(pcols, track_build_matches=True)
@@ -219,7 +230,7 @@ class GpuBackend:
             return lp, lb  # (left_idx, right_idx with -1 for unmatched)
         return pi, bi.long()
 
-    def groupby(self, keys: List[Val], aggs):
+    def groupby(self, keys: List[Val], aggs, hint=None):
         """aggs: list of (fn∈{count,count_valid,sum,min,max}, Val|None).
         Returns (key Vals, result Vals)."""
         from ..columnar import validity_to_bool
@@ -231,7 +242,7 @@ class GpuBackend:
         for fn, v in aggs:
             gaggs.append((fnmap[fn], _val_to_column(v) if v is not None
                           else None))
-        kt, res = gb(kcols, gaggs)
+        kt, res = gb(kcols, gaggs, num_groups_hint=hint)
         ngroups = kt.num_rows
         out_keys = []
         for c, orig in zip(kt.columns, keys):
@@ -328,7 +339,7 @@ class CpuBackend:
         return (torch.from_numpy(li.astype(np.int64)).to(dev),
                 torch.from_numpy(ri.astype(np.int64)).to(dev))
 
-    def groupby(self, keys: List[Val], aggs):
+    def groupby(self, keys: List[Val], aggs, hint=None):
         import numpy as np
         n = keys[0].data.numel() if keys else (
             aggs[0][1].data.numel() if aggs and aggs[0][1] is not None else 0)
@@ -590,7 +601,20 @@ class Engine:
                 raise ValueError(fn)
         if not native:
             native.append(("count", None))  # keys-only: need group discovery
-        kvals, rvals, ng = self.backend.groupby(kv, native)
+        # a global aggregate has exactly one group: the hint engages the
+        # LDS pre-aggregation kernel instead of a per-row contended CAS
+        # table; chunk to <=3 aggs per pass (the LDS kernel's limit)
+        if dummy_key:
+            kvals = rvals = None
+            ng = 0
+            out_r = []
+            for i in range(0, len(native), 3):
+                kvals, rv, ng = self.backend.groupby(kv, native[i:i + 3],
+                                                     hint=1)
+                out_r.extend(rv)
+            rvals = out_r
+        else:
+            kvals, rvals, ng = self.backend.groupby(kv, native)
 
         if distributed:
             # partials -> replicated frame -> re-reduce

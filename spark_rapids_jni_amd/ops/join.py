@@ -249,6 +249,47 @@ class HashJoinTable:
             return out_build, out_probe, matched
         return out_build, out_probe
 
+    def mark_matches(self, probe) -> torch.Tensor:
+        """Stream the probe rows and return per-BUILD-row matched flags
+        (uint8) without materializing join pairs. This is the right-semi
+        building block: to semi/anti-filter a small table against a huge
+        one, build on the small side and mark while probing the huge side
+        (reference get_matched_rows, join_primitives.hpp:237)."""
+        pcols = _keys(probe)
+        nprobe = pcols[0].size
+        g = _native.gpu()
+        stream = _native.current_stream()
+        dev = pcols[0].device
+        matched = torch.zeros(self.num_build_rows, dtype=torch.uint8,
+                              device=dev)
+        counter = torch.zeros(1, dtype=torch.int64, device=dev)
+        pack = getattr(self, "_pack", None)
+        fast = self.i64_fast and (
+            (pack is not None and len(pcols) == len(self.build_keys))
+            or (pack is None and _is_i64_fast(pcols)))
+        if fast and pack is not None:
+            packed, pvalid = _pack_keys(pcols, *pack)
+            pcols = [Column(pcols[0].dtype, nprobe, packed, pvalid,
+                            null_count=None)]
+        elif fast:
+            pcols = [_as_i64_keys(pcols[0])]
+        if fast:
+            p = pcols[0]
+            g.join_probe_i64(p.data.data_ptr(),
+                             p.validity.data_ptr() if p.validity is not None
+                             else 0,
+                             nprobe, self.slots.data_ptr(), self.capacity,
+                             counter.data_ptr(), 0, 0, 0,
+                             matched.data_ptr(), 1, 0, stream)
+        else:
+            bdesc, btop, pdesc, ptop, keep = self._descs(pcols)
+            g.join_probe_fill(bdesc.data_ptr(), btop.data_ptr(),
+                              pdesc.data_ptr(), ptop.data_ptr(),
+                              len(pcols), nprobe, self.slots.data_ptr(),
+                              self.capacity, counter.data_ptr(), 0, 0, 0,
+                              matched.data_ptr(), stream)
+        return matched
+
     def semi_join(self, probe, anti: bool = False) -> torch.Tensor:
         """Left semi/anti join: probe-side row indices with (no) match."""
         if self.i64_fast:
