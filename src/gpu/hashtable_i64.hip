@@ -200,8 +200,10 @@ __global__ void join_probe_i64_kernel(
           if (pos < out_capacity) {
             out_build[pos] = (int32_t)(hit1[b] - 1);
             out_probe[pos] = idxmap ? (int64_t)idxmap[base + b] : base + b;
-            if (build_matched) build_matched[hit1[b] - 1] = 1;
           }
+          // matched flags are valid even when the pair output overflows
+          // (capacity-0 mark-only probes depend on this)
+          if (build_matched) build_matched[hit1[b] - 1] = 1;
           ++pos;
           continue;
         }
@@ -213,8 +215,8 @@ __global__ void join_probe_i64_kernel(
             if (pos < out_capacity) {
               out_build[pos] = (int32_t)(r1 - 1);
               out_probe[pos] = idxmap ? (int64_t)idxmap[base + b] : base + b;
-              if (build_matched) build_matched[r1 - 1] = 1;
             }
+            if (build_matched) build_matched[r1 - 1] = 1;
             ++pos;
           }
           if (!(cur.row1 & SLOT_CHAIN)) break;

@@ -518,3 +518,55 @@ def test_multikey_semi_anti_join():
     anti = sorted(tbl.semi_join(p, anti=True).cpu().tolist())
     assert semi == exp_semi
     assert anti == sorted(set(range(n_p)) - set(exp_semi))
+
+
+@pytest.mark.gpu
+def test_mark_matches_small_build_huge_probe():
+    # right-semi building block: matched flags must be set even with a
+    # capacity-0 (mark-only) probe
+    from spark_rapids_jni_amd.ops.join import HashJoinTable
+    import random as _r
+    _r.seed(11)
+    build_vals = list(range(0, 200, 2))  # evens < 200
+    probe_vals = [_r.randint(0, 400) for _ in range(500_000)]
+    b = Column.from_pylist(build_vals, DType.INT64, device="cuda")
+    p = Column.from_pylist(probe_vals, DType.INT64, device="cuda")
+    tbl = HashJoinTable.build(b)
+    m = tbl.mark_matches(p).cpu().tolist()
+    pset = set(probe_vals)
+    exp = [1 if v in pset else 0 for v in build_vals]
+    assert m == exp
+
+
+@pytest.mark.gpu
+def test_global_agg_chunked_lds():
+    # the NDS engine computes global aggregates via 1-group LDS passes;
+    # verify sum/avg/min/max/count against a plain torch fp64 reference
+    import torch as _t
+    from spark_rapids_jni_amd.nds.plan import Engine, Frame, Val
+    from spark_rapids_jni_amd.nds.plan import Agg as NAgg
+    from spark_rapids_jni_amd.nds.expr import col
+    n = 1_000_000
+    g = _t.Generator(device="cuda").manual_seed(5)
+    vals = _t.rand(n, dtype=_t.float64, device="cuda", generator=g) * 100
+    valid = _t.rand(n, device="cuda", generator=g) > 0.1
+    iv = _t.randint(0, 1000, (n,), dtype=_t.int64, device="cuda",
+                    generator=g)
+    f = Frame({"x": Val(vals, valid), "i": Val(iv)}, n)
+    e = Engine({}, device="cuda")
+    e.register("t", f)
+    from spark_rapids_jni_amd.nds.plan import Scan
+    out = e.run(NAgg(Scan("t"), [], [
+        ("s", "sum", col("x")), ("a", "avg", col("x")),
+        ("mn", "min", col("x")), ("mx", "max", col("x")),
+        ("c", "count", col("x")), ("si", "sum", col("i")),
+        ("call", "count", None)]))
+    row = out.to_rows()[0]
+    xs = vals[valid]
+    exp = (float(xs.sum()), float(xs.mean()), float(xs.min()),
+           float(xs.max()), int(valid.sum()), int(iv.sum()), n)
+    for got, want in zip(row, exp):
+        if isinstance(want, float):
+            assert abs(got - want) / max(abs(want), 1) < 1e-9, (row, exp)
+        else:
+            assert got == want, (row, exp)
