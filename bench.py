@@ -24,6 +24,10 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+# large-SF NDS runs allocate/free many odd-sized tensors; expandable
+# segments avoid the fragmentation OOMs the default allocator hits
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 
 BUILD_ROWS_DEFAULT = 1_000_000_000

@@ -480,27 +480,29 @@ class Engine:
     def _exec_scan(self, p: Scan) -> Frame:
         if p.table in self.temps:
             f = self.temps[p.table]
-        else:
-            f = self._load(p.table)
-        if p.columns:
-            cols = {c: f.cols[c] for c in p.columns}
-            return Frame(cols, f.nrows, f.sharded)
-        return f
+            if p.columns:
+                return Frame({c: f.cols[c] for c in p.columns}, f.nrows,
+                             f.sharded)
+            return f
+        return self._load(p.table, p.columns)
 
-    def _load(self, name: str) -> Frame:
-        if name in self._scan_cache:
-            return self._scan_cache[name]
+    def _load(self, name: str, columns=None) -> Frame:
+        """Columns load lazily (only what queries reference) and stay
+        cached on the device across queries."""
         ds = self.catalog[name]
-        cols = {}
-        for cname, arr in ds.columns.items():
-            t = torch.as_tensor(arr).to(self.device)
+        cache = self._scan_cache.setdefault(name, {})
+        want = list(columns) if columns else list(ds.columns.keys())
+        for cname in want:
+            if cname in cache:
+                continue
+            t = torch.as_tensor(ds.columns[cname]).to(self.device)
             valid = None
             if ds.valid.get(cname) is not None:
                 valid = torch.as_tensor(ds.valid[cname]).to(self.device)
-            cols[cname] = Val(t, valid, ds.dicts.get(cname))
-        f = Frame(cols, ds.nrows, sharded=(ds.sharded and self.world > 1))
-        self._scan_cache[name] = f
-        return f
+            cache[cname] = Val(t, valid, ds.dicts.get(cname))
+        cols = {c: cache[c] for c in want}
+        return Frame(cols, ds.nrows,
+                     sharded=(ds.sharded and self.world > 1))
 
     def _exec_filter(self, p: Filter) -> Frame:
         f = self._exec(p.child)
@@ -908,7 +910,8 @@ class Engine:
                 if d.dtype.is_floating_point:
                     fill = float("-inf") if asc else float("inf")
                 else:
-                    fill = -(2**62) if asc else (2**62)
+                    ii = torch.iinfo(d.dtype)
+                    fill = ii.min if asc else ii.max
                 d = torch.where(vm, d, torch.full_like(d, fill))
             o = torch.argsort(d, stable=True, descending=not asc)
             perm = perm[o]

@@ -32,6 +32,13 @@ class Dataset:
     sharded: bool = False
 
     def add(self, name, arr, valid=None, dic=None):
+        if arr.dtype == np.int64 and arr.size:
+            # every NDS integer domain (surrogate keys, ticket numbers up
+            # to SF3000, codes, quantities) fits 32 bits: downcasting
+            # halves HBM footprint and scan bandwidth
+            mn, mx = arr.min(), arr.max()
+            if -2**31 <= mn and mx < 2**31:
+                arr = arr.astype(np.int32)
         self.columns[name] = arr
         self.valid[name] = valid
         self.dicts[name] = dic
