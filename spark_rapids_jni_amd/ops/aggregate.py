@@ -104,114 +104,136 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
                 packed = packed * w + code
             i64_fast = True
             i64_keys = packed
-    cap_groups = num_groups_hint if num_groups_hint else n
-    capacity = max(_next_pow2(min(cap_groups, n) * 2), 64)
-    if i64_fast:
-        # interleaved {key, row1} pairs; keys init to the EMPTY sentinel
-        # (INT64_MIN), +1 reserved slot for rows whose key equals it
-        slots = torch.zeros(2 * (capacity + 1), dtype=torch.int64, device=dev)
-        slots.view(-1, 2)[:, 0] = -2**63
-        nstates = capacity + 1
-    else:
-        slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
-        nstates = capacity
-
     # native agg list = user aggs + a hidden COUNT_VALID per nullable SUM col
     # (to derive the all-null-group -> null result in the SAME compaction pass,
-    # since compaction order is nondeterministic).
-    native: List[Tuple[int, Optional[Column], torch.Tensor]] = []
+    # since compaction order is nondeterministic). States are (re)allocated
+    # inside the capacity loop below.
+    native: List[Tuple[int, Optional[Column]]] = []
     metas = []  # (user op, is_float, native_index, hidden_count_index|None)
 
-    def add_native(native_op, col, st):
-        native.append((native_op, col, st))
+    def add_native(native_op, col):
+        native.append((native_op, col))
         return len(native) - 1
 
-    def _hidden_count(col, add_native):
+    def _hidden_count(col):
         # all-null-group detection needs a per-group valid count — but a
         # column with no validity buffer can never produce an all-null
         # group, so skip the extra atomic stream entirely (the bench shape:
         # one atomicAdd per row saved)
         if col.validity is None:
             return None
-        stc = torch.zeros(nstates, dtype=torch.int64, device=dev)
-        return add_native(1, col, stc)
+        return add_native(1, col)
 
     for op, col in aggs:
         if op in (Agg.COUNT_ALL, Agg.COUNT_VALID):
-            st = torch.zeros(nstates, dtype=torch.int64, device=dev)
-            idx = add_native(int(op), col, st)
+            idx = add_native(int(op), col)
             metas.append((op, False, idx, None))
             continue
         assert col is not None
         is_float = col.dtype in (DType.FLOAT32, DType.FLOAT64)
         if op == Agg.SUM:
-            st = torch.zeros(nstates, dtype=torch.float64 if is_float
-                             else torch.int64, device=dev)
-            idx = add_native(3 if is_float else 2, col, st)
-            metas.append((op, is_float, idx, _hidden_count(col, add_native)))
+            idx = add_native(3 if is_float else 2, col)
         elif op == Agg.MIN:
-            st = torch.full((nstates,), float("inf") if is_float else 2**63 - 1,
-                            dtype=torch.float64 if is_float else torch.int64,
-                            device=dev)
-            idx = add_native(6 if is_float else 4, col, st)
-            metas.append((op, is_float, idx, _hidden_count(col, add_native)))
+            idx = add_native(6 if is_float else 4, col)
         else:  # MAX
-            st = torch.full((nstates,), float("-inf") if is_float else -2**63,
-                            dtype=torch.float64 if is_float else torch.int64,
-                            device=dev)
-            idx = add_native(7 if is_float else 5, col, st)
-            metas.append((op, is_float, idx, _hidden_count(col, add_native)))
+            idx = add_native(7 if is_float else 5, col)
+        metas.append((op, is_float, idx, _hidden_count(col)))
 
     naggs = len(native)
-    raw = bytearray(max(naggs, 1) * _AGGDESC_SZ)
-    for i, (native_op, col, st) in enumerate(native):
-        struct.pack_into(
-            _AGGDESC_FMT, raw, i * _AGGDESC_SZ, native_op,
-            int(col.dtype) if col is not None else 0,
-            col.data.data_ptr() if col is not None else 0,
-            (col.validity.data_ptr() if col is not None and col.validity is not None
-             else 0),
-            st.data_ptr())
-    agg_desc = torch.frombuffer(raw, dtype=torch.uint8).to(dev)
+
+    def _state_for(native_op, is_float_dtype, nstates):
+        if native_op in (0, 1):
+            return torch.zeros(nstates, dtype=torch.int64, device=dev)
+        if native_op in (2, 3):
+            return torch.zeros(nstates, dtype=torch.float64 if native_op == 3
+                               else torch.int64, device=dev)
+        if native_op == 4:
+            return torch.full((nstates,), 2**63 - 1, dtype=torch.int64,
+                              device=dev)
+        if native_op == 5:
+            return torch.full((nstates,), -2**63, dtype=torch.int64,
+                              device=dev)
+        if native_op == 6:
+            return torch.full((nstates,), float("inf"), dtype=torch.float64,
+                              device=dev)
+        return torch.full((nstates,), float("-inf"), dtype=torch.float64,
+                          device=dev)
+
+    # capacity schedule: start from the hint (or a 4M-slot guess) and grow
+    # 16x on overflow up to the exact bound. A 288M-row aggregation with
+    # thousands of groups must not allocate 2n-slot tables (that is ~17 GB
+    # of slots plus an 8.6 GB state per aggregate).
+    full_cap = max(_next_pow2(n * 2), 64)
+    if num_groups_hint:
+        capacity = max(_next_pow2(min(num_groups_hint, n) * 2), 64)
+    else:
+        capacity = min(full_cap, 1 << 22)
+    lds_ok = (num_groups_hint is not None and num_groups_hint <= 1024
+              and naggs <= 3)
+    if not i64_fast:
+        kdesc, ktop, keep = pack_descriptors(kcols)
+    while True:
+        nstates = capacity + 1 if i64_fast else capacity
+        if i64_fast:
+            # interleaved {key, row1} pairs; keys init to the EMPTY sentinel
+            # (INT64_MIN), +1 reserved slot for rows whose key equals it
+            slots = torch.zeros(2 * (capacity + 1), dtype=torch.int64,
+                                device=dev)
+            slots.view(-1, 2)[:, 0] = -2**63
+        else:
+            slots = torch.zeros(capacity, dtype=torch.int64, device=dev)
+        states = [_state_for(op, None, nstates) for op, _c in native]
+        raw = bytearray(max(naggs, 1) * _AGGDESC_SZ)
+        for i, ((native_op, col), st) in enumerate(zip(native, states)):
+            struct.pack_into(
+                _AGGDESC_FMT, raw, i * _AGGDESC_SZ, native_op,
+                int(col.dtype) if col is not None else 0,
+                col.data.data_ptr() if col is not None else 0,
+                (col.validity.data_ptr()
+                 if col is not None and col.validity is not None else 0),
+                st.data_ptr())
+        agg_desc = torch.frombuffer(raw, dtype=torch.uint8).to(dev)
+        overflow = torch.zeros(1, dtype=torch.int32, device=dev)
+        if i64_fast:
+            if lds_ok:
+                # low-cardinality: per-workgroup LDS pre-aggregation
+                # collapses per-row global atomics into per-(block x
+                # group) merges
+                _IDENT = {0: 0, 1: 0, 2: 0, 3: 0,
+                          4: 2**63 - 1, 5: -2**63,
+                          6: 0x7FF0000000000000,           # +inf bits
+                          7: 0xFFF0000000000000 - 2**64}   # -inf bits
+                idents = torch.tensor([_IDENT[op] for op, _c in native],
+                                      dtype=torch.int64).to(dev)
+                g.groupby_i64_lds(i64_keys.data_ptr(), n, slots.data_ptr(),
+                                  capacity, agg_desc.data_ptr(), naggs,
+                                  idents.data_ptr(), overflow.data_ptr(),
+                                  stream)
+            else:
+                g.groupby_i64(i64_keys.data_ptr(), n, slots.data_ptr(),
+                              capacity, agg_desc.data_ptr(), naggs,
+                              overflow.data_ptr(), stream)
+        else:
+            g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n,
+                      slots.data_ptr(), capacity, agg_desc.data_ptr(),
+                      naggs, overflow.data_ptr(), stream)
+        if not int(overflow.item()):
+            break
+        assert capacity < full_cap, "groupby overflow at full capacity"
+        capacity = min(capacity * 16, full_cap)
+        lds_ok = False
+        del slots, states, agg_desc
 
     counter = torch.zeros(1, dtype=torch.int64, device=dev)
     out_repr = torch.empty(nstates, dtype=torch.int64, device=dev)
-    out_agg = torch.empty(max(naggs, 1) * nstates, dtype=torch.int64, device=dev)
+    out_agg = torch.empty(max(naggs, 1) * nstates, dtype=torch.int64,
+                          device=dev)
     if i64_fast:
-        overflow = torch.zeros(1, dtype=torch.int32, device=dev)
-        lds_ok = (num_groups_hint is not None and num_groups_hint <= 1024
-                  and naggs <= 3)
-        if lds_ok:
-            # low-cardinality: per-workgroup LDS pre-aggregation collapses
-            # per-row global atomics into per-(block x group) merges
-            _IDENT = {0: 0, 1: 0, 2: 0, 3: 0,
-                      4: 2**63 - 1, 5: -2**63,
-                      6: 0x7FF0000000000000,           # +inf bits
-                      7: 0xFFF0000000000000 - 2**64}   # -inf bits
-            idents = torch.tensor([_IDENT[op] for op, _c, _s in native],
-                                  dtype=torch.int64).to(dev)
-            g.groupby_i64_lds(i64_keys.data_ptr(), n, slots.data_ptr(),
-                              capacity, agg_desc.data_ptr(), naggs,
-                              idents.data_ptr(), overflow.data_ptr(), stream)
-        else:
-            g.groupby_i64(i64_keys.data_ptr(), n, slots.data_ptr(), capacity,
-                          agg_desc.data_ptr(), naggs, overflow.data_ptr(),
-                          stream)
-        if int(overflow.item()):
-            # cardinality hint undersized the table: re-run unhinted
-            return groupby(keys, aggs, num_groups_hint=None)
         g.groupby_compact_i64(slots.data_ptr(), capacity + 1,
                               agg_desc.data_ptr(), naggs, counter.data_ptr(),
                               out_repr.data_ptr(), out_agg.data_ptr(), nstates,
                               stream)
     else:
-        overflow = torch.zeros(1, dtype=torch.int32, device=dev)
-        kdesc, ktop, keep = pack_descriptors(kcols)
-        g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n,
-                  slots.data_ptr(), capacity, agg_desc.data_ptr(), naggs,
-                  overflow.data_ptr(), stream)
-        if int(overflow.item()):
-            return groupby(keys, aggs, num_groups_hint=None)
         g.groupby_compact(slots.data_ptr(), capacity, agg_desc.data_ptr(),
                           naggs, counter.data_ptr(), out_repr.data_ptr(),
                           out_agg.data_ptr(), nstates, stream)
