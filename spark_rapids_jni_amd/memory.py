@@ -92,18 +92,60 @@ class RmmSpark:
 
     # -- lifecycle (reference RmmSpark.setEventHandler / clearEventHandler) --
     @classmethod
-    def set_event_handler(cls, pool_limit: int = -1, host_limit: int = -1):
+    def set_event_handler(cls, pool_limit: int = -1, host_limit: int = -1,
+                          real: bool = False):
+        """Install the adaptor. With real=True the adaptor also becomes the
+        target of the torch pluggable-allocator bridge: every REAL device
+        allocation (hipMallocAsync under torch) runs the pre/post state
+        machine and `pool_limit` caps actual HBM bytes — the reference's
+        RMM-resource wrapping (SparkResourceAdaptorJni.cpp:2113). Requires
+        use_real_allocator() to have been called before any CUDA work."""
         with cls._lock:
             if cls._adaptor is not None:
                 raise RuntimeError("event handler already set")
-            cls._adaptor = _native.host().SparkResourceAdaptor(pool_limit,
-                                                               host_limit)
+            cls._adaptor = _native.host().SparkResourceAdaptor(
+                -1 if real else pool_limit, host_limit)
+            if real:
+                host = _native.host()
+                gpu = _native.gpu()
+                host.install_as_current(cls._adaptor)
+                gpu.install_ra_hooks(*host.hook_addrs())
+                gpu.set_device_pool_limit(pool_limit)
             return cls._adaptor
 
     @classmethod
     def clear_event_handler(cls):
         with cls._lock:
+            if cls._adaptor is not None:
+                try:
+                    _native.host().clear_current()
+                    g = _native.gpu_or_none()
+                    if g is not None:
+                        g.clear_ra_hooks()
+                        g.set_device_pool_limit(-1)
+                except Exception:
+                    pass
             cls._adaptor = None
+
+    _real_allocator = None
+
+    @classmethod
+    def use_real_allocator(cls):
+        """Route ALL torch device allocations through srj_torch_malloc/free
+        (src/gpu/torch_alloc.hip). Must run before the first CUDA
+        allocation in the process; typically paired with
+        set_event_handler(real=True)."""
+        if cls._real_allocator is not None:
+            return
+        import os
+        import torch
+        gpu = _native.gpu()  # ensure the .so is importable
+        path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "_gpu.so")
+        alloc = torch.cuda.memory.CUDAPluggableAllocator(
+            path, "srj_torch_malloc", "srj_torch_free")
+        torch.cuda.memory.change_current_allocator(alloc)
+        cls._real_allocator = alloc
 
     @classmethod
     def adaptor(cls):
@@ -211,15 +253,18 @@ def with_retry(fn, spill=None, split=None, max_retries: int = 16,
     while True:
         try:
             return fn()
-        except GpuSplitAndRetryOOM:
-            if split is None:
+        except (GpuSplitAndRetryOOM, GpuRetryOOM, RuntimeError) as e:
+            kind = _oom_kind(e)
+            if kind == "split":
+                if split is None:
+                    raise
+                attempts += 1
+                if attempts > max_retries:
+                    raise
+                split()
+                continue  # split made room; retry immediately
+            if kind != "retry":
                 raise
-            attempts += 1
-            if attempts > max_retries:
-                raise
-            split()
-            continue  # split made room; retry immediately
-        except GpuRetryOOM:
             attempts += 1
             if attempts > max_retries:
                 raise
@@ -229,3 +274,19 @@ def with_retry(fn, spill=None, split=None, max_retries: int = 16,
                 RmmSpark.block_thread_until_ready(thread_id)
             except Exception:
                 pass
+
+
+def _oom_kind(e) -> Optional[str]:
+    """Classify both the typed exceptions (simulated pool) and the marker-
+    prefixed RuntimeErrors the real-allocator bridge raises through torch
+    (torch_alloc.hip throw_code)."""
+    if isinstance(e, GpuSplitAndRetryOOM):
+        return "split"
+    if isinstance(e, GpuRetryOOM):
+        return "retry"
+    msg = str(e)
+    if "GpuSplitAndRetryOOM" in msg:
+        return "split"
+    if "GpuRetryOOM" in msg:
+        return "retry"
+    return None

@@ -6,8 +6,23 @@
 
 extern "C" void srj_pb_decode(const void*, int64_t, void*, int32_t, uint8_t*,
                               int32_t, hipStream_t);
+extern "C" void srj_install_ra_hooks(uintptr_t, uintptr_t, uintptr_t,
+                                     uintptr_t);
+extern "C" void srj_clear_ra_hooks();
+extern "C" void srj_set_device_pool_limit(long long);
+extern "C" long long srj_device_pool_used();
 
 void register_tools(py::module_& m) {
+  // torch pluggable-allocator bridge (src/gpu/torch_alloc.hip)
+  m.def("install_ra_hooks", [](uintptr_t a, uintptr_t b, uintptr_t c,
+                               uintptr_t d) {
+    srj_install_ra_hooks(a, b, c, d);
+  });
+  m.def("clear_ra_hooks", [] { srj_clear_ra_hooks(); });
+  m.def("set_device_pool_limit", [](long long b) {
+    srj_set_device_pool_limit(b);
+  });
+  m.def("device_pool_used", [] { return srj_device_pool_used(); });
   m.def("roctx_range_push", [](const std::string& name) {
     return roctxRangePush(name.c_str());
   });

@@ -579,8 +579,53 @@ class SparkResourceAdaptor {
 
 }  // namespace srj_host
 
+// ---------------------------------------------------------------------------
+// Bridge for the torch pluggable allocator (src/gpu/torch_alloc.hip):
+// stable C entry points over the CURRENT adaptor so real hipMallocAsync
+// allocations run the same pre/post state machine the reference wraps
+// around every RMM allocation (SparkResourceAdaptorJni.cpp:2113-2141).
+// The GPU-side allocator receives these function addresses from Python
+// (no cross-.so link needed).
+// ---------------------------------------------------------------------------
+static srj_host::SparkResourceAdaptor* g_current_adaptor = nullptr;
+static std::mutex g_current_mu;
+
+extern "C" int srj_ra_pre_alloc(long tid) {
+  auto* a = g_current_adaptor;
+  return a ? a->pre_alloc_external(tid) : 0;
+}
+extern "C" int srj_ra_post_alloc_failed(long tid) {
+  auto* a = g_current_adaptor;
+  return a ? a->post_alloc_failed_external(tid, /*was_oom=*/true) : 6;
+}
+extern "C" void srj_ra_post_alloc_success(long tid, long nbytes) {
+  auto* a = g_current_adaptor;
+  if (a) a->post_alloc_success_external(tid, nbytes);
+}
+extern "C" void srj_ra_dealloc(long tid, long nbytes) {
+  auto* a = g_current_adaptor;
+  if (a) a->dealloc_external(tid, nbytes);
+}
+
 void register_resource_adaptor(py::module_& m) {
   using srj_host::SparkResourceAdaptor;
+  m.def("install_as_current",
+        [](SparkResourceAdaptor& a) {
+          std::lock_guard<std::mutex> lk(g_current_mu);
+          g_current_adaptor = &a;
+        },
+        "Route the torch-allocator bridge through this adaptor (the Python "
+        "owner must keep it alive until clear_current).");
+  m.def("clear_current", []() {
+    std::lock_guard<std::mutex> lk(g_current_mu);
+    g_current_adaptor = nullptr;
+  });
+  m.def("hook_addrs", []() {
+    return py::make_tuple((uintptr_t)&srj_ra_pre_alloc,
+                          (uintptr_t)&srj_ra_post_alloc_failed,
+                          (uintptr_t)&srj_ra_post_alloc_success,
+                          (uintptr_t)&srj_ra_dealloc);
+  });
   py::class_<SparkResourceAdaptor>(m, "SparkResourceAdaptor")
       .def(py::init<int64_t, int64_t>(), py::arg("pool_limit") = -1,
            py::arg("host_limit") = -1)
