@@ -270,6 +270,15 @@ def with_retry(fn, spill=None, split=None, max_retries: int = 16,
                 raise
             if spill is not None:
                 spill()
+                # torch's caching layer sits above the pluggable
+                # allocator: blocks freed by the spill stay cached until
+                # explicitly flushed back to the device pool
+                try:
+                    import torch
+                    if torch.cuda.is_initialized():
+                        torch.cuda.empty_cache()
+                except Exception:
+                    pass
             try:
                 RmmSpark.block_thread_until_ready(thread_id)
             except Exception:
