@@ -270,9 +270,11 @@ def with_retry(fn, spill=None, split=None, max_retries: int = 16,
                 raise
             if spill is not None:
                 spill()
-                # torch's caching layer sits above the pluggable
-                # allocator: blocks freed by the spill stay cached until
-                # explicitly flushed back to the device pool
+                # slow path hygiene (the JVM plugin does the same on OOM):
+                # break python reference cycles that can briefly pin the
+                # spilled device tensors, then flush any allocator cache
+                import gc
+                gc.collect()
                 try:
                     import torch
                     if torch.cuda.is_initialized():
