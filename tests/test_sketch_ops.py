@@ -147,15 +147,18 @@ def test_gbk_decode():
 
 
 @pytest.mark.gpu
-def test_hll_bias_correction_midrange():
-    """Mid-range cardinalities (the HLL++ bias region) must estimate within
-    ~2% with the generated bias tables (raw HLL alone is ~5-10% high there)."""
+def test_hll_estimate_midrange():
+    """Reference-exact finalizer (cuco semantics: linear counting below
+    2.5m, raw HLL above — no appendix bias tables, same as the reference's
+    estimate path): linear-counting range is tight, the uncorrected band
+    above 2.5m carries the documented ~+5% raw-HLL bias."""
     from spark_rapids_jni_amd.ops.sketch import HyperLogLogPlusPlus
     p = 12
     m = 1 << p
-    for true_n in (int(m * 1.5), int(m * 3)):
+    for true_n, tol in ((int(m * 1.5), 0.02), (int(m * 3), 0.07),
+                        (int(m * 30), 0.03)):
         h = HyperLogLogPlusPlus(precision=p)
         keys = torch.arange(true_n, dtype=torch.int64, device="cuda") * 977 + 13
         h.update(Column.from_torch(keys))
         est = h.estimate()
-        assert abs(est - true_n) / true_n < 0.025, (true_n, est)
+        assert abs(est - true_n) / true_n < tol, (true_n, est)
