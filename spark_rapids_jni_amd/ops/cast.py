@@ -201,51 +201,99 @@ def from_floats(col: Column) -> Column:
                   null_count=None)
 
 
-_FMT_KINDS = {"y": 1, "M": 2, "d": 3, "H": 4, "m": 5, "s": 6, "S": 7}
+_FMT_KINDS = {"y": 1, "M": 2, "d": 3, "H": 4, "m": 5, "s": 6}
+_FMT_SKIP_WS = 8
 
 
-def compile_timestamp_format(fmt: str):
-    """Spark to_timestamp pattern subset: y/M/d/H/m/s/S runs + literals
-    (reference parse_timestamp_with_format.cu:386)."""
+def compile_timestamp_format(fmt: str, legacy: bool = False):
+    """Compile a Spark to_timestamp pattern to device tokens with the
+    reference's exact width/whitespace policies
+    (parse_timestamp_with_format.cu:142-250 compile_format):
+
+      * y/M/d/H/m/s only; non-year runs must be length 2; runs cap at 9
+      * packed runs (digit field abutting a digit field) are exact-width
+      * CORRECTED: exact widths, except the pinned "yyyy/MM/dd" deviation
+        which accepts 1-2 digit month/day (spark-rapids compat contract)
+      * LEGACY: [1, run] widths and [ \\t]* skipped before each
+        non-packed field (SimpleDateFormat semantics)
+      * non-ASCII literals rejected; a space matches exactly one ' '
+
+    Extension beyond the reference: a 'S' run parses a 1..run-digit
+    fraction (the reference rejects 'S'; Spark supports it)."""
     toks = []
+    n = len(fmt)
     i = 0
-    while i < len(fmt):
+    saw_field = False
+    corrected_slash = (not legacy) and fmt == "yyyy/MM/dd"
+
+    def lit(ch):
+        if ord(ch) >= 0x80:
+            raise ValueError("non-ASCII literal in pattern is not supported")
+        toks.append((0, ord(ch), 0, 0))
+
+    while i < n:
         c = fmt[i]
-        if c in _FMT_KINDS:
+        if c.isalpha():
             j = i
-            while j < len(fmt) and fmt[j] == c:
+            while j < n and fmt[j] == c:
                 j += 1
-            toks.append((_FMT_KINDS[c], j - i))
+            run = j - i
+            if run > 9:
+                raise ValueError(f"pattern letter run too long: {c}")
+            if c == "S":
+                toks.append((7, run, 0, 0))
+                saw_field = True
+                i = j
+                continue
+            if c not in _FMT_KINDS:
+                raise ValueError(f"unsupported pattern letter: {c}")
+            if c != "y" and run != 2:
+                raise ValueError(
+                    f"non-year pattern letter run must be length 2: {c}")
+            packed = ((i > 0 and fmt[i - 1].isalpha()) or
+                      (j < n and fmt[j].isalpha()))
+            variable = (legacy and not packed) or corrected_slash
+            mind = run if c == "y" else (1 if variable else run)
+            abuts_prev = i > 0 and fmt[i - 1].isalpha()
+            if legacy and not abuts_prev:
+                toks.append((_FMT_SKIP_WS, 0, 0, 0))
+            toks.append((_FMT_KINDS[c], 0, mind, run))
+            saw_field = True
             i = j
         elif c == "'":
             j = fmt.index("'", i + 1)
             for ch in fmt[i + 1:j]:
-                toks.append((0, ord(ch)))
+                lit(ch)
             i = j + 1
         else:
-            toks.append((0, ord(c)))
+            lit(c)
             i += 1
+    if not saw_field:
+        raise ValueError("timestamp format has no datetime fields")
     return toks
 
 
 def to_timestamp_with_format(col: Column, fmt: str, ansi: bool = False,
-                             default_tz_offset_sec: int = 0) -> Column:
-    """Spark to_timestamp(col, fmt) (reference parse_timestamp_with_format)."""
+                             default_tz_offset_sec: int = 0,
+                             legacy: bool = False) -> Column:
+    """Spark to_timestamp(col, fmt) (reference parse_timestamp_with_format);
+    legacy=True gives SimpleDateFormat-era parsing semantics."""
     g = _native.gpu()
     n = col.size
     dev = col.device
-    toks = compile_timestamp_format(fmt)
+    toks = compile_timestamp_format(fmt, legacy=legacy)
     raw = bytearray()
-    for kind, count in toks:
-        raw += struct.pack("<ii", kind, count)
+    for kind, aux, mind, maxd in toks:
+        raw += struct.pack("<iiii", kind, aux, mind, maxd)
     tt = torch.frombuffer(raw or bytearray(1), dtype=torch.uint8).to(dev)
     out = torch.empty(n, dtype=torch.int64, device=dev)
     validity = make_validity(n, dev)
     err, err_ptr = _err_buf(ansi, dev)
     desc, keep = _desc_for(col)
     g.parse_timestamp_fmt(desc.data_ptr(), n, tt.data_ptr(), len(toks),
-                          default_tz_offset_sec, out.data_ptr(),
-                          validity.data_ptr(), err_ptr, _native.current_stream())
+                          1 if legacy else 0, default_tz_offset_sec,
+                          out.data_ptr(), validity.data_ptr(), err_ptr,
+                          _native.current_stream())
     _check_err(err, col, ansi)
     return Column(DType.TIMESTAMP_US, n, out, validity, null_count=None)
 

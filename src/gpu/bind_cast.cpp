@@ -21,8 +21,9 @@ void srj_format_number(const double*, const uint8_t*, int64_t, int32_t,
                        hipStream_t);
 void srj_float_to_string(const void*, const uint8_t*, int64_t, int32_t, int32_t,
                          int32_t*, const int32_t*, char*, uint8_t*, hipStream_t);
-void srj_parse_timestamp_fmt(const void*, int64_t, const void*, int32_t, int64_t,
-                             int64_t*, uint8_t*, int64_t*, hipStream_t);
+void srj_parse_timestamp_fmt(const void*, int64_t, const void*, int32_t,
+                             int32_t, int64_t, int64_t*, uint8_t*, int64_t*,
+                             hipStream_t);
 }
 
 void register_cast(py::module_& m) {
@@ -75,11 +76,13 @@ void register_cast(py::module_& m) {
     check_hip("string_to_timestamp");
   });
   m.def("parse_timestamp_fmt", [](uintptr_t in, int64_t n, uintptr_t toks,
-                                  int32_t ntoks, int64_t tz_off, uintptr_t out,
+                                  int32_t ntoks, int32_t trail_nondigit,
+                                  int64_t tz_off, uintptr_t out,
                                   uintptr_t valid, uintptr_t err,
                                   uintptr_t stream) {
     srj_parse_timestamp_fmt(as_ptr<void>(in), n, as_ptr<void>(toks), ntoks,
-                            tz_off, as_ptr<int64_t>(out), as_ptr<uint8_t>(valid),
+                            trail_nondigit, tz_off, as_ptr<int64_t>(out),
+                            as_ptr<uint8_t>(valid),
                             as_ptr<int64_t>(err), as_stream(stream));
     check_hip("parse_timestamp_fmt");
   });

@@ -674,32 +674,39 @@ __global__ void string_to_timestamp_kernel(ColDesc in, int64_t nrows,
 enum FmtKind : int32_t {
   FMT_LITERAL = 0, FMT_YEAR = 1, FMT_MONTH = 2, FMT_DAY = 3,
   FMT_HOUR = 4, FMT_MINUTE = 5, FMT_SECOND = 6, FMT_FRACTION = 7,
+  FMT_SKIP_WS = 8,  // LEGACY: skip [ \t]* before a numeric field
 };
 
 struct FmtToken {
   int32_t kind;
-  int32_t count;   // repeat count (or the literal char for FMT_LITERAL)
+  int32_t aux;     // literal char / fraction max digits
+  int32_t mind;    // min digits for a numeric field
+  int32_t maxd;    // max digits for a numeric field
 };
 
-__device__ inline bool parse_digits_n(StrView s, int* pos, int count,
-                                      bool greedy2, long* out) {
+__device__ inline bool parse_digits_mm(StrView s, int* pos, int mind,
+                                       int maxd, long* out) {
   long v = 0;
   int k = 0;
-  int maxd = greedy2 ? 2 : count;
   while (*pos < s.len && k < maxd && s.ptr[*pos] >= '0' &&
          s.ptr[*pos] <= '9') {
     v = v * 10 + (s.ptr[*pos] - '0');
     ++(*pos);
     ++k;
   }
-  if (greedy2 ? k < 1 : k < count) return false;
+  if (k < mind) return false;
   *out = v;
   return true;
 }
 
+// Width/trailing policy mirrors the reference compile_format +
+// device walker (parse_timestamp_with_format.cu:142-250): exact widths in
+// CORRECTED mode (packed runs always exact), [1,run] widths plus [ \t]
+// skipping before fields in LEGACY, trailing EOF (CORRECTED) vs trailing
+// non-digit (LEGACY).
 __global__ void parse_timestamp_fmt_kernel(ColDesc in, int64_t nrows,
                                            const FmtToken* __restrict__ toks,
-                                           int32_t ntoks,
+                                           int32_t ntoks, int32_t trail_nondigit,
                                            int64_t default_tz_offset_sec,
                                            int64_t* __restrict__ out,
                                            uint8_t* __restrict__ out_valid,
@@ -719,41 +726,35 @@ __global__ void parse_timestamp_fmt_kernel(ColDesc in, int64_t nrows,
         FmtToken tk = toks[t];
         switch (tk.kind) {
           case FMT_LITERAL:
-            if (pos >= s.len || s.ptr[pos] != (char)tk.count) valid = false;
+            if (pos >= s.len || s.ptr[pos] != (char)tk.aux) valid = false;
             else ++pos;
             break;
-          case FMT_YEAR: {
-            // years allow up to max(4, count) digits when count == 4 ("yyyy")
-            long vv = 0;
-            int k = 0, maxd = tk.count <= 2 ? 4 : tk.count;
-            while (pos < s.len && k < maxd && s.ptr[pos] >= '0' &&
-                   s.ptr[pos] <= '9') {
-              vv = vv * 10 + (s.ptr[pos] - '0');
-              ++pos; ++k;
-            }
-            if (k < (tk.count <= 2 ? 1 : tk.count)) valid = false;
-            else y = vv;
+          case FMT_SKIP_WS:
+            while (pos < s.len && (s.ptr[pos] == ' ' || s.ptr[pos] == '\t'))
+              ++pos;
             break;
-          }
+          case FMT_YEAR:
+            valid = valid && parse_digits_mm(s, &pos, tk.mind, tk.maxd, &y);
+            break;
           case FMT_MONTH:
-            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &mo);
+            valid = valid && parse_digits_mm(s, &pos, tk.mind, tk.maxd, &mo);
             break;
           case FMT_DAY:
-            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &d);
+            valid = valid && parse_digits_mm(s, &pos, tk.mind, tk.maxd, &d);
             break;
           case FMT_HOUR:
-            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &hh);
+            valid = valid && parse_digits_mm(s, &pos, tk.mind, tk.maxd, &hh);
             break;
           case FMT_MINUTE:
-            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &mi);
+            valid = valid && parse_digits_mm(s, &pos, tk.mind, tk.maxd, &mi);
             break;
           case FMT_SECOND:
-            valid = valid && parse_digits_n(s, &pos, tk.count, tk.count == 1, &ss);
+            valid = valid && parse_digits_mm(s, &pos, tk.mind, tk.maxd, &ss);
             break;
           case FMT_FRACTION: {
             long f = 0;
             int k = 0;
-            while (pos < s.len && k < tk.count && s.ptr[pos] >= '0' &&
+            while (pos < s.len && k < tk.aux && s.ptr[pos] >= '0' &&
                    s.ptr[pos] <= '9') {
               f = f * 10 + (s.ptr[pos] - '0');
               ++pos; ++k;
@@ -766,7 +767,13 @@ __global__ void parse_timestamp_fmt_kernel(ColDesc in, int64_t nrows,
           }
         }
       }
-      if (pos != s.len) valid = false;  // whole string must be consumed
+      if (trail_nondigit) {
+        // LEGACY: trailing text allowed unless it starts with a digit
+        if (pos < s.len && s.ptr[pos] >= '0' && s.ptr[pos] <= '9')
+          valid = false;
+      } else if (pos != s.len) {
+        valid = false;  // CORRECTED: whole string must be consumed
+      }
       if (valid && (!valid_ymd((int)y, (int)mo, (int)d) || hh > 23 ||
                     mi > 59 || ss > 59))
         valid = false;
@@ -946,13 +953,14 @@ void srj_string_to_timestamp(const void* in, int64_t nrows, int64_t now_us,
 }
 
 void srj_parse_timestamp_fmt(const void* in, int64_t nrows, const void* toks,
-                             int32_t ntoks, int64_t default_tz_offset_sec,
+                             int32_t ntoks, int32_t trail_nondigit,
+                             int64_t default_tz_offset_sec,
                              int64_t* out, uint8_t* out_valid, int64_t* err_row,
                              hipStream_t stream) {
   parse_timestamp_fmt_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       *reinterpret_cast<const ColDesc*>(in), nrows,
-      reinterpret_cast<const FmtToken*>(toks), ntoks, default_tz_offset_sec,
-      out, out_valid, err_row);
+      reinterpret_cast<const FmtToken*>(toks), ntoks, trail_nondigit,
+      default_tz_offset_sec, out, out_valid, err_row);
 }
 
 void srj_integer_to_string(const void* in, int64_t nrows, int32_t phase,

@@ -433,15 +433,42 @@ def test_to_timestamp_with_format():
     assert got[5] is None
     assert got[6] is None  # trailing junk
 
-    col2 = Column.from_pylist(["15-7-2021 9.5", "1-12-1999 23.59"],
-                              DType.STRING, "cuda")
-    got2 = cast.to_timestamp_with_format(col2, "d-M-yyyy H.m").to_pylist()
-    assert got2[0] == us(2021, 7, 15, 9, 5)
-    assert got2[1] == us(1999, 12, 1, 23, 59)
-
     col3 = Column.from_pylist(["2020-01-02 03:04:05.123"], DType.STRING, "cuda")
     got3 = cast.to_timestamp_with_format(col3, "yyyy-MM-dd HH:mm:ss.SSS")
     assert got3.to_pylist()[0] == us(2020, 1, 2, 3, 4, 5, micro=123000)
+
+    # CORRECTED "yyyy/MM/dd" keeps the spark-rapids compat deviation:
+    # 1-2 digit month/day accepted (reference
+    # parse_timestamp_with_format.cu corrected_variable_width_slash_date)
+    col4 = Column.from_pylist(["2024/5/6", "2024/05/06", "2024/5/6 x"],
+                              DType.STRING, "cuda")
+    got4 = cast.to_timestamp_with_format(col4, "yyyy/MM/dd").to_pylist()
+    assert got4[0] == us(2024, 5, 6)
+    assert got4[1] == us(2024, 5, 6)
+    assert got4[2] is None  # CORRECTED requires full consumption
+
+    # packed run stays exact-width even in LEGACY
+    col5 = Column.from_pylist(["20240506", "2024056"], DType.STRING, "cuda")
+    got5 = cast.to_timestamp_with_format(col5, "yyyyMMdd",
+                                         legacy=True).to_pylist()
+    assert got5[0] == us(2024, 5, 6)
+    assert got5[1] is None
+
+    # LEGACY: [ \t] skipped before fields, 1-2 digit widths, trailing
+    # non-digit text tolerated (SimpleDateFormat semantics)
+    col6 = Column.from_pylist(["2024- 5- 6", "2024-5-6 junk",
+                               "2024-5-67"], DType.STRING, "cuda")
+    got6 = cast.to_timestamp_with_format(col6, "yyyy-MM-dd",
+                                         legacy=True).to_pylist()
+    assert got6[0] == us(2024, 5, 6)
+    assert got6[1] == us(2024, 5, 6)  # " junk" after the date is ignored
+    assert got6[2] is None  # trailing digit
+
+    # pattern validation mirrors the reference's compile_format errors
+    import pytest as _pytest
+    for bad_fmt in ("d-M-yyyy", "yyyy-MMM-dd", "", "abc"):
+        with _pytest.raises(ValueError):
+            cast.compile_timestamp_format(bad_fmt)
 
 
 @pytest.mark.gpu
