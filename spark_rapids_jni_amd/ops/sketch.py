@@ -174,23 +174,34 @@ class UriPart(IntEnum):
     QUERY_KEY = 4
 
 
-def parse_uri(col: Column, part: UriPart, query_key: str = "") -> Column:
-    """Spark parse_url (reference parse_uri.cu / ParseURI.java:38-174)."""
+def parse_uri(col: Column, part: UriPart, query_key="") -> Column:
+    """Spark parse_url (reference parse_uri.cu / ParseURI.java:38-174).
+
+    query_key may be a literal str (parseURIQueryWithLiteral) or a string
+    Column of per-row keys (parseURIQueryWithColumn)."""
     g = _native.gpu()
     n = col.size
     dev = col.device
     desc, top, keep = pack_descriptors([col])
-    kb = query_key.encode()
+    qcol_ptr = 0
+    kb = b""
+    if isinstance(query_key, Column):
+        assert query_key.size == n, "key column must match the URI column"
+        qdesc, qtop, qkeep = pack_descriptors([query_key])
+        qcol_ptr = qdesc.data_ptr()
+        part = UriPart.QUERY_KEY
+    else:
+        kb = query_key.encode()
     kt = torch.frombuffer(bytearray(kb) or bytearray(1), dtype=torch.uint8).to(dev)
 
     def lens_fn(lens, stream):
-        g.parse_uri(desc.data_ptr(), n, int(part), kt.data_ptr(), len(kb), 0,
-                    lens.data_ptr(), 0, 0, 0, stream)
+        g.parse_uri(desc.data_ptr(), n, int(part), kt.data_ptr(), len(kb),
+                    qcol_ptr, 0, lens.data_ptr(), 0, 0, 0, stream)
 
     def write_fn(offsets, chars, validity, stream):
-        g.parse_uri(desc.data_ptr(), n, int(part), kt.data_ptr(), len(kb), 1, 0,
-                    offsets.data_ptr(), chars.data_ptr(), validity.data_ptr(),
-                    stream)
+        g.parse_uri(desc.data_ptr(), n, int(part), kt.data_ptr(), len(kb),
+                    qcol_ptr, 1, 0, offsets.data_ptr(), chars.data_ptr(),
+                    validity.data_ptr(), stream)
 
     return _str_output(n, dev, lens_fn, write_fn)
 

@@ -14,7 +14,8 @@ void srj_percentile_from_histogram(const int32_t*, const double*, const int64_t*
                                    uint8_t*, hipStream_t);
 void srj_conv(const void*, int64_t, int32_t, int32_t, int32_t, int32_t*,
               const int32_t*, char*, uint8_t*, hipStream_t);
-void srj_parse_uri(const void*, int64_t, int32_t, const char*, int32_t, int32_t,
+void srj_parse_uri(const void*, int64_t, int32_t, const char*, int32_t,
+                   const void*, int32_t,
                    int32_t*, const int32_t*, char*, uint8_t*, hipStream_t);
 void srj_gbk_decode(const void*, int64_t, int32_t, int32_t, int32_t*,
                     const int32_t*, char*, uint8_t*, int64_t*, hipStream_t);
@@ -73,10 +74,11 @@ void register_misc2(py::module_& m) {
     check_hip("conv");
   });
   m.def("parse_uri", [](uintptr_t in, int64_t n, int32_t part, uintptr_t qkey,
-                        int32_t qkey_len, int32_t phase, uintptr_t lens,
-                        uintptr_t offsets, uintptr_t chars, uintptr_t valid,
-                        uintptr_t stream) {
-    srj_parse_uri(as_ptr<void>(in), n, part, as_ptr<char>(qkey), qkey_len, phase,
+                        int32_t qkey_len, uintptr_t qcol, int32_t phase,
+                        uintptr_t lens, uintptr_t offsets, uintptr_t chars,
+                        uintptr_t valid, uintptr_t stream) {
+    srj_parse_uri(as_ptr<void>(in), n, part, as_ptr<char>(qkey), qkey_len,
+                  as_ptr<void>(qcol), phase,
                   as_ptr<int32_t>(lens), as_ptr<int32_t>(offsets),
                   as_ptr<char>(chars), as_ptr<uint8_t>(valid), as_stream(stream));
     check_hip("parse_uri");

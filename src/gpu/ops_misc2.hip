@@ -219,15 +219,443 @@ enum UriPart : int32_t {
   URI_QUERY_KEY = 4,
 };
 
-__device__ inline bool uri_scheme_char(char c, bool first) {
-  bool alpha = (c | 32) >= 'a' && (c | 32) <= 'z';
-  if (first) return alpha;
-  return alpha || (c >= '0' && c <= '9') || c == '+' || c == '-' || c == '.';
+// Full RFC-shaped validation machine with the reference's exact semantics
+// (reference parse_uri.cu:87-756): per-chunk character allowlists,
+// percent-escape and UTF-8 validation with the unicode-whitespace
+// blacklist, IPv4/IPv6/domain host validation (3-state: VALID / host-only
+// INVALID / FATAL), userinfo/port splitting with the reference's quirky
+// last_colon bookkeeping, opaque vs hierarchical URIs, and fragment
+// validation. Bug-compat notes are marked inline.
+
+struct USpan { const char* p; int n; };
+
+__device__ inline bool u_alpha(char c) { return (c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z'); }
+__device__ inline bool u_num(char c) { return c >= '0' && c <= '9'; }
+__device__ inline bool u_alnum(char c) { return u_alpha(c) || u_num(c); }
+__device__ inline bool u_hex(char c) {
+  return u_num(c) || (c >= 'a' && c <= 'f') || (c >= 'A' && c <= 'F');
+}
+
+// Validate (and skip) percent escapes and multibyte UTF-8 at position *i.
+// Packed-value whitespace blacklist matches the reference's char_utf8
+// comparisons (parse_uri.cu:101-140).
+__device__ inline bool uri_skip_special(const char* s, int len, int* i,
+                                        bool allow_bad_escape) {
+  while (*i < len) {
+    unsigned char b = (unsigned char)s[*i];
+    if (b == '%' && !allow_bad_escape) {
+      if (*i + 2 >= len) return false;
+      if (!u_hex(s[*i + 1]) || !u_hex(s[*i + 2])) return false;
+      *i += 3;
+    } else if (b >= 0xC0) {
+      int nb = (b & 0xE0) == 0xC0 ? 2 : (b & 0xF0) == 0xE0 ? 3
+               : (b & 0xF8) == 0xF0 ? 4 : 1;
+      if (nb == 1) return false;  // 0xF8+ invalid lead
+      if (*i + nb > len) return false;
+      uint32_t packed = b;
+      for (int k = 1; k < nb; ++k)
+        packed = (packed << 8) | (unsigned char)s[*i + k];
+      if (nb > 1 && (packed & 0xC0) != 0x80) return false;
+      if (nb > 2 && (packed & 0xC000) != 0x8000) return false;
+      if (nb > 3 && (packed & 0xC00000) != 0x800000) return false;
+      if ((packed >= 0xc280 && packed <= 0xc2a0) || packed == 0xe19a80 ||
+          (packed >= 0xe28080 && packed <= 0xe2808a) || packed == 0xe280af ||
+          packed == 0xe280a8 || packed == 0xe2819f || packed == 0xe38080)
+        return false;
+      *i += nb;
+    } else {
+      break;  // plain byte (incl. stray continuation): the allowlist decides
+    }
+  }
+  return true;
+}
+
+template <typename Pred>
+__device__ inline bool uri_chunk_ok(USpan s, Pred pred,
+                                    bool allow_bad_escape = false) {
+  int i = 0;
+  if (!uri_skip_special(s.p, s.n, &i, allow_bad_escape)) return false;
+  while (i < s.n) {
+    if (!pred(s.p[i])) return false;
+    ++i;
+    if (!uri_skip_special(s.p, s.n, &i, allow_bad_escape)) return false;
+  }
+  return true;
+}
+
+__device__ inline bool uri_valid_scheme(USpan s) {
+  if (s.n <= 0 || !u_alpha(s.p[0])) return false;
+  for (int i = 1; i < s.n; ++i) {
+    char c = s.p[i];
+    if (!u_alnum(c) && c != '+' && c != '-' && c != '.') return false;
+  }
+  return true;
+}
+
+__device__ inline bool uri_valid_ipv6(USpan s) {
+  constexpr int max_colons = 8;
+  if (s.n < 2) return false;
+  bool found_double_colon = false;
+  int open_br = 0, close_br = 0, periods = 0, colons = 0, percents = 0;
+  char prev = 0;
+  int address = 0, addr_chars = 0;
+  bool addr_hex = false;
+  for (int i = 0; i < s.n; ++i) {
+    char c = s.p[i];
+    switch (c) {
+      case '[':
+        if (++open_br > 1) return false;
+        break;
+      case ']':
+        if (++close_br > 1) return false;
+        if (periods > 0 && (addr_hex || address > 255)) return false;
+        break;
+      case ':':
+        ++colons;
+        if (prev == ':') {
+          if (found_double_colon) return false;
+          found_double_colon = true;
+        }
+        address = 0; addr_hex = false; addr_chars = 0;
+        if (colons > max_colons ||
+            (colons == max_colons && !found_double_colon))
+          return false;
+        if (periods > 0 || percents > 0) return false;
+        break;
+      case '.':
+        ++periods;
+        if (percents > 0) return false;
+        if (periods > 3) return false;
+        if (addr_hex) return false;
+        if (address > 255) return false;
+        if (colons != 6 && !found_double_colon) return false;
+        if (colons >= max_colons) return false;
+        address = 0; addr_hex = false; addr_chars = 0;
+        break;
+      case '%':
+        // zone id suffix (%eth0); anything goes after it
+        if (++percents > 1) return false;
+        if (periods > 0 && (addr_hex || address > 255)) return false;
+        address = 0; addr_hex = false; addr_chars = 0;
+        break;
+      default:
+        if (percents == 0) {
+          if (addr_chars > 3) return false;
+          ++addr_chars;
+          address *= 10;
+          if (c >= 'a' && c <= 'f') {
+            address += 10 + (c - 'a');
+            addr_hex = true;
+          } else if (c >= 'A' && c <= 'Z') {
+            // reference quirk: A..Z (not just A..F) count as hex digits
+            address += 10 + (c - 'A');
+            addr_hex = true;
+          } else if (u_num(c)) {
+            address += c - '0';
+          } else {
+            return false;
+          }
+        }
+        break;
+    }
+    prev = c;
+  }
+  return true;
+}
+
+__device__ inline bool uri_valid_ipv4(USpan s) {
+  int address = 0, addr_chars = 0, dots = 0;
+  for (int i = 0; i < s.n; ++i) {
+    char c = s.p[i];
+    if (!u_num(c) && (i == 0 || c != '.')) return false;
+    if (c == '.') {
+      if (addr_chars == 0) return false;
+      address = 0; addr_chars = 0; ++dots;
+      continue;
+    }
+    ++addr_chars;
+    address = address * 10 + (c - '0');
+    if (address > 255) return false;
+  }
+  if (addr_chars == 0) return false;
+  return dots == 3;
+}
+
+__device__ inline bool uri_valid_domain(USpan s) {
+  bool last_dash = false, last_dot = false, numeric_start = false;
+  int chars_before_dot = 0;
+  for (int i = 0; i < s.n; ++i) {
+    char c = s.p[i];
+    if (!u_alnum(c) && c != '-' && c != '.') return false;
+    numeric_start = last_dot && u_num(c);
+    if (c == '-') {
+      if (last_dot || i == 0 || i == s.n - 1) return false;
+      last_dash = true;
+      last_dot = false;
+    } else if (c == '.') {
+      if (last_dash || last_dot || chars_before_dot == 0) return false;
+      last_dot = true;
+      last_dash = false;
+      chars_before_dot = 0;
+    } else {
+      last_dot = false;
+      last_dash = false;
+      ++chars_before_dot;
+    }
+  }
+  return !numeric_start;
+}
+
+enum UriHostValidity : int { UHOST_FATAL = 0, UHOST_INVALID = 1, UHOST_VALID = 2 };
+
+__device__ inline int uri_valid_host(USpan host) {
+  if (host.n > 0 && host.p[0] == '[') {
+    if (host.p[host.n - 1] != ']') return UHOST_FATAL;
+    return uri_valid_ipv6(host) ? UHOST_VALID : UHOST_FATAL;
+  }
+  int last_open = -1, last_close = -1, last_dot = -1;
+  for (int i = 0; i < host.n; ++i) {
+    char c = host.p[i];
+    if (c == '[') last_open = i;
+    else if (c == ']') last_close = i;
+    else if (c == '.') last_dot = i;
+  }
+  if (last_open >= 0 || last_close >= 0) return UHOST_FATAL;
+  if (last_dot < 0 || last_dot == host.n - 1 || !u_num(host.p[last_dot + 1])) {
+    if (uri_valid_domain(host)) return UHOST_VALID;
+  } else if (uri_valid_ipv4(host)) {
+    return UHOST_VALID;
+  }
+  return UHOST_INVALID;
+}
+
+struct UriQueryPred {
+  __device__ bool operator()(char c) const {
+    return c == '!' || c == '"' || c == '$' || (c >= '&' && c <= ';') ||
+           c == '=' || (c >= '?' && c <= ']' && c != '\\') ||
+           (c >= 'a' && c <= 'z') || c == '_' || c == '~';
+  }
+};
+struct UriAuthorityPred {
+  bool allow_bad_escape;
+  __device__ bool operator()(char c) const {
+    return c == '!' || c == '$' || (c >= '&' && c <= ';' && c != '/') ||
+           c == '=' || (c >= '@' && c <= '_' && c != '^' && c != '\\') ||
+           (c >= 'a' && c <= 'z') || c == '~' ||
+           (allow_bad_escape && c == '%');
+  }
+};
+struct UriUserinfoPred {
+  __device__ bool operator()(char c) const { return c != '[' && c != ']'; }
+};
+struct UriPortPred {
+  // bug-compat: the reference's condition (c < '0' && c > '9') can never
+  // be true, so ports accept any character
+  __device__ bool operator()(char c) const { return true; }
+};
+struct UriPathPred {
+  __device__ bool operator()(char c) const {
+    return c == '!' || c == '$' || (c >= '&' && c <= ';') || c == '=' ||
+           (c >= '@' && c <= 'Z') || c == '_' || (c >= 'a' && c <= 'z') ||
+           c == '~';
+  }
+};
+struct UriOpaqueFragPred {
+  __device__ bool operator()(char c) const {
+    return c == '!' || c == '$' || (c >= '&' && c <= ';') || c == '=' ||
+           (c >= '?' && c <= ']' && c != '\\') || c == '_' || c == '~' ||
+           (c >= 'a' && c <= 'z');
+  }
+};
+
+// query-parameter extraction (reference find_query_part)
+__device__ inline bool uri_find_query_part(USpan q, const char* key,
+                                           int key_len, USpan* out) {
+  const char* h = q.p;
+  const char* h_end = q.p + q.n;
+  while (h + key_len < h_end) {
+    bool match = true;
+    for (int j = 0; j < key_len; ++j) {
+      if (h[j] != key[j]) { match = false; break; }
+    }
+    if (match && h[key_len] == '=') {
+      h += key_len + 1;
+      const char* start = h;
+      int n = 0;
+      while (h < h_end && *h != '&') { ++n; ++h; }
+      *out = {start, n};
+      return true;
+    }
+    while (h + key_len < h_end && *h != '&') ++h;
+    ++h;
+  }
+  return false;
+}
+
+struct UriParts {
+  USpan scheme, host, authority, path, fragment, query, userinfo, port, opaque;
+  uint32_t valid = 0;  // bit per chunk below
+};
+enum UriChunkBit : int {
+  UB_PROTOCOL = 0, UB_HOST = 1, UB_AUTHORITY = 2, UB_PATH = 3,
+  UB_FRAGMENT = 4, UB_QUERY = 5, UB_USERINFO = 6, UB_PORT = 7, UB_OPAQUE = 8,
+};
+
+__device__ inline UriParts uri_validate(const char* str, int len,
+                                        const char* qkey, int qkey_len) {
+  UriParts ret;
+  const char* original = str;
+  int col = -1, slash = -1, hash = -1, question = -1;
+  for (int i = 0; i < len &&
+       (col == -1 || slash == -1 || hash == -1 || question == -1); ++i) {
+    switch (str[i]) {
+      case ':': if (col == -1) col = i; break;
+      case '/': if (slash == -1) slash = i; break;
+      case '#': if (hash == -1) hash = i; break;
+      case '?': if (question == -1) question = i; break;
+      default: break;
+    }
+  }
+  if (hash >= 0) {
+    ret.fragment = {str + hash + 1, len - hash - 1};
+    if (!uri_chunk_ok(ret.fragment, UriOpaqueFragPred{})) {
+      ret.valid = 0;
+      return ret;
+    }
+    ret.valid |= 1u << UB_FRAGMENT;
+    len = hash;
+    if (col > hash) col = -1;
+    if (slash > hash) slash = -1;
+    if (question > hash) question = -1;
+  }
+  bool const has_scheme = (col != -1) && (slash == -1 || col < slash) &&
+                          (hash == -1 || col < hash);
+  if (has_scheme) {
+    ret.scheme = {str, col};
+    if (!uri_valid_scheme(ret.scheme)) {
+      ret.valid = 0;
+      return ret;
+    }
+    ret.valid |= 1u << UB_PROTOCOL;
+    int const skip = col + 1;
+    str += skip;
+    len -= skip;
+    question -= skip;
+    hash -= skip;
+    slash -= skip;
+  }
+  if (len <= 0) {
+    // scheme-only is invalid; empty/fragment-only counts as an empty path
+    ret.valid = (has_scheme ? 0u : 1u) << UB_PATH;
+    return ret;
+  }
+  bool const hierarchical = str[0] == '/' || str == original;
+  if (hierarchical) {
+    if (question >= 0) {
+      ret.query = {str + question + 1, len - question - 1};
+      if (!uri_chunk_ok(ret.query, UriQueryPred{})) {
+        ret.valid = 0;
+        return ret;
+      }
+      if (qkey != nullptr) {
+        USpan part;
+        if (!uri_find_query_part(ret.query, qkey, qkey_len, &part)) {
+          ret.valid = 0;
+          return ret;
+        }
+        ret.query = part;
+      }
+      ret.valid |= 1u << UB_QUERY;
+    }
+    int const path_len = question >= 0 ? question : len;
+    if (len >= 2 && str[0] == '/' && str[1] == '/') {
+      int next_slash = -1;
+      for (int i = 2; i < path_len; ++i) {
+        if (str[i] == '/') { next_slash = i; break; }
+      }
+      ret.authority = {str + 2,
+                       next_slash == -1
+                           ? (question < 0 ? len - 2 : question - 2)
+                           : next_slash - 2};
+      if (next_slash > 0) ret.path = {str + next_slash, path_len - next_slash};
+      if (ret.authority.n > 0) {
+        bool const ipv6_addr = ret.authority.n > 2 && ret.authority.p[0] == '[';
+        if (!uri_chunk_ok(ret.authority, UriAuthorityPred{ipv6_addr},
+                          ipv6_addr)) {
+          ret.valid = 0;
+          return ret;
+        }
+        ret.valid |= 1u << UB_AUTHORITY;
+        const char* auth = ret.authority.p;
+        int auth_size = ret.authority.n;
+        int amp = -1, closing_bracket = -1, last_colon = -1;
+        for (int i = 0; i < auth_size; ++i) {
+          switch (auth[i]) {
+            case '@':
+              if (amp == -1) {
+                amp = i;
+                if (last_colon > 0) last_colon = -1;
+                if (closing_bracket > 0) closing_bracket = -1;
+              }
+              break;
+            case ':': last_colon = amp > 0 ? i - amp - 1 : i; break;
+            case ']':
+              if (closing_bracket == -1)
+                closing_bracket = amp > 0 ? i - amp : i;
+              break;
+          }
+        }
+        if (amp > 0) {
+          ret.userinfo = {auth, amp};
+          if (!uri_chunk_ok(ret.userinfo, UriUserinfoPred{})) {
+            ret.valid = 0;
+            return ret;
+          }
+          ret.valid |= 1u << UB_USERINFO;
+          ++amp;
+          auth += amp;
+          auth_size -= amp;
+        }
+        if (last_colon > 0 && last_colon > closing_bracket) {
+          ret.port = {auth + last_colon + 1, auth_size - last_colon - 1};
+          if (!uri_chunk_ok(ret.port, UriPortPred{})) {
+            ret.valid = 0;
+            return ret;
+          }
+          ret.valid |= 1u << UB_PORT;
+          ret.host = {auth, last_colon};
+        } else {
+          ret.host = {auth, auth_size};
+        }
+        switch (uri_valid_host(ret.host)) {
+          case UHOST_FATAL: ret.valid = 0; return ret;
+          case UHOST_INVALID: ret.host = {nullptr, 0}; break;
+          case UHOST_VALID: ret.valid |= 1u << UB_HOST; break;
+        }
+      }
+    } else {
+      ret.path = {str, path_len};
+    }
+    if (!uri_chunk_ok(ret.path, UriPathPred{})) {
+      ret.valid = 0;
+      return ret;
+    }
+    ret.valid |= 1u << UB_PATH;
+  } else {
+    ret.opaque = {str, len};
+    if (!uri_chunk_ok(ret.opaque, UriOpaqueFragPred{})) {
+      ret.valid = 0;
+      return ret;
+    }
+    ret.valid |= 1u << UB_OPAQUE;
+  }
+  return ret;
 }
 
 template <bool WRITE>
 __global__ void parse_uri_kernel(ColDesc in, int64_t nrows, int32_t part,
                                  const char* __restrict__ qkey, int32_t qkey_len,
+                                 ColDesc qcol, int32_t has_qcol,
                                  int32_t* __restrict__ lens,
                                  const int32_t* __restrict__ offsets,
                                  char* __restrict__ chars,
@@ -242,115 +670,40 @@ __global__ void parse_uri_kernel(ColDesc in, int64_t nrows, int32_t part,
     int32_t out_n = 0;
     if (valid) {
       StrView s = get_string(in, row);
-      // scheme
-      int i = 0;
-      while (i < s.len && s.ptr[i] != ':') {
-        if (!uri_scheme_char(s.ptr[i], i == 0)) { i = -1; break; }
-        ++i;
+      bool want_key = part == URI_QUERY_KEY;
+      const char* key = qkey;
+      int32_t key_len = qkey_len;
+      if (want_key && has_qcol) {
+        // per-row key column (reference parseURIQueryWithColumn; a null
+        // key row yields a null result)
+        if (!is_valid(qcol.valid, row)) {
+          valid = false;
+          key = nullptr;
+        } else {
+          StrView k = get_string(qcol, row);
+          key = k.ptr;
+          key_len = k.len;
+        }
       }
-      if (i <= 0 || i >= s.len) {
+      UriParts parts{};
+      if (valid)
+        parts = uri_validate(s.ptr, s.len, want_key ? key : nullptr,
+                             key_len);
+      if (!valid) parts.valid = 0;
+      USpan sel{nullptr, 0};
+      int bit = -1;
+      switch (part) {
+        case URI_PROTOCOL: sel = parts.scheme; bit = UB_PROTOCOL; break;
+        case URI_HOST: sel = parts.host; bit = UB_HOST; break;
+        case URI_PATH: sel = parts.path; bit = UB_PATH; break;
+        case URI_QUERY:
+        case URI_QUERY_KEY: sel = parts.query; bit = UB_QUERY; break;
+      }
+      if (bit < 0 || !(parts.valid & (1u << bit))) {
         valid = false;
       } else {
-        int scheme_end = i;  // ':' position
-        int p = i + 1;
-        int auth_start = -1, auth_end = -1;
-        if (p + 1 < s.len && s.ptr[p] == '/' && s.ptr[p + 1] == '/') {
-          auth_start = p + 2;
-          auth_end = auth_start;
-          while (auth_end < s.len && s.ptr[auth_end] != '/' &&
-                 s.ptr[auth_end] != '?' && s.ptr[auth_end] != '#')
-            ++auth_end;
-          p = auth_end;
-        }
-        int path_start = p, path_end = p;
-        while (path_end < s.len && s.ptr[path_end] != '?' &&
-               s.ptr[path_end] != '#')
-          ++path_end;
-        p = path_end;
-        int q_start = -1, q_end = -1;
-        if (p < s.len && s.ptr[p] == '?') {
-          q_start = p + 1;
-          q_end = q_start;
-          while (q_end < s.len && s.ptr[q_end] != '#') ++q_end;
-        }
-        // basic validation: no spaces/control chars anywhere
-        for (int k = 0; k < s.len && valid; ++k) {
-          unsigned char c = (unsigned char)s.ptr[k];
-          if (c <= ' ' || c == '<' || c == '>' || c == '"' || c == '`' ||
-              c == '{' || c == '}' || c == '|' || c == '\\' || c == '^')
-            valid = false;
-        }
-        if (valid) {
-          switch (part) {
-            case URI_PROTOCOL:
-              out_p = s.ptr;
-              out_n = scheme_end;
-              break;
-            case URI_HOST: {
-              if (auth_start < 0) { valid = false; break; }
-              int hs = auth_start, he = auth_end;
-              for (int k = auth_start; k < auth_end; ++k)
-                if (s.ptr[k] == '@') hs = k + 1;
-              // strip port (but not inside [])
-              bool brac = hs < he && s.ptr[hs] == '[';
-              if (brac) {
-                int k = hs;
-                while (k < he && s.ptr[k] != ']') ++k;
-                out_p = s.ptr + hs;
-                out_n = k + 1 <= he ? k + 1 - hs : he - hs;
-              } else {
-                int pe = he;
-                for (int k = hs; k < he; ++k)
-                  if (s.ptr[k] == ':') { pe = k; break; }
-                out_p = s.ptr + hs;
-                out_n = pe - hs;
-                // validate host chars
-                for (int k = hs; k < pe && valid; ++k) {
-                  char c = s.ptr[k];
-                  bool ok = (c | 32) >= 'a' && (c | 32) <= 'z';
-                  ok = ok || (c >= '0' && c <= '9') || c == '.' || c == '-' ||
-                       c == '_' || c == '%';
-                  if (!ok) valid = false;
-                }
-              }
-              if (out_n == 0) valid = false;
-              break;
-            }
-            case URI_PATH:
-              out_p = s.ptr + path_start;
-              out_n = path_end - path_start;
-              break;
-            case URI_QUERY:
-              if (q_start < 0) { valid = false; break; }
-              out_p = s.ptr + q_start;
-              out_n = q_end - q_start;
-              break;
-            case URI_QUERY_KEY: {
-              if (q_start < 0) { valid = false; break; }
-              valid = false;
-              int k = q_start;
-              while (k < q_end) {
-                int amp = k;
-                while (amp < q_end && s.ptr[amp] != '&') ++amp;
-                int eq = k;
-                while (eq < amp && s.ptr[eq] != '=') ++eq;
-                if (eq - k == qkey_len) {
-                  bool m = true;
-                  for (int t2 = 0; t2 < qkey_len; ++t2)
-                    if (s.ptr[k + t2] != qkey[t2]) { m = false; break; }
-                  if (m && eq < amp) {
-                    out_p = s.ptr + eq + 1;
-                    out_n = amp - eq - 1;
-                    valid = true;
-                    break;
-                  }
-                }
-                k = amp + 1;
-              }
-              break;
-            }
-          }
-        }
+        out_p = sel.p;
+        out_n = sel.n;
       }
     }
     if (WRITE) {
@@ -527,16 +880,20 @@ void srj_conv(const void* in, int64_t nrows, int32_t from_base, int32_t to_base,
 }
 
 void srj_parse_uri(const void* in, int64_t nrows, int32_t part, const char* qkey,
-                   int32_t qkey_len, int32_t phase, int32_t* lens,
-                   const int32_t* offsets, char* chars, uint8_t* out_valid,
-                   hipStream_t stream) {
+                   int32_t qkey_len, const void* qcol, int32_t phase,
+                   int32_t* lens, const int32_t* offsets, char* chars,
+                   uint8_t* out_valid, hipStream_t stream) {
   ColDesc c = *reinterpret_cast<const ColDesc*>(in);
+  ColDesc kc{};
+  if (qcol) kc = *reinterpret_cast<const ColDesc*>(qcol);
   if (phase == 0)
     parse_uri_kernel<false><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
-        c, nrows, part, qkey, qkey_len, lens, nullptr, nullptr, nullptr);
+        c, nrows, part, qkey, qkey_len, kc, qcol ? 1 : 0, lens, nullptr,
+        nullptr, nullptr);
   else
     parse_uri_kernel<true><<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
-        c, nrows, part, qkey, qkey_len, nullptr, offsets, chars, out_valid);
+        c, nrows, part, qkey, qkey_len, kc, qcol ? 1 : 0, nullptr, offsets,
+        chars, out_valid);
 }
 
 void srj_gbk_decode(const void* in, int64_t nrows, int32_t report, int32_t phase,
