@@ -227,7 +227,7 @@ enum UriPart : int32_t {
 // last_colon bookkeeping, opaque vs hierarchical URIs, and fragment
 // validation. Bug-compat notes are marked inline.
 
-struct USpan { const char* p; int n; };
+struct USpan { const char* p = nullptr; int n = 0; };
 
 __device__ inline bool u_alpha(char c) { return (c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z'); }
 __device__ inline bool u_num(char c) { return c >= '0' && c <= '9'; }
