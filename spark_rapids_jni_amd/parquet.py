@@ -34,6 +34,9 @@ ENC_RLE_DICTIONARY = 8
 
 CODEC_UNCOMPRESSED = 0
 CODEC_SNAPPY = 1
+CODEC_GZIP = 2
+CODEC_ZSTD = 6
+CODEC_LZ4_RAW = 7
 
 _SCATTER_FMT = "<QQQQqqqii"     # ScatterDesc (64B)
 _RLE_FMT = "<QqQqii"            # RleDesc (40B)
@@ -242,19 +245,66 @@ class _Page:
     data: bytes              # page payload (still compressed when comp=True)
     def_bytes: int = 0       # v2: definition level byte length
     num_nulls: int = -1      # v2 only
-    comp: bool = False       # True: snappy payload decompressed ON DEVICE
+    comp: bool = False       # True: payload not yet decompressed
     uncomp: int = 0          # uncompressed size when comp=True
     rep_bytes: int = 0       # v2: repetition level byte length
+    codec: int = CODEC_UNCOMPRESSED
 
 
+_HOST_CODECS = {CODEC_SNAPPY: "snappy", CODEC_GZIP: "gzip",
+                CODEC_ZSTD: "zstd", CODEC_LZ4_RAW: "lz4_raw"}
+
+# Codec policy (measured, documented): SNAPPY decompresses ON DEVICE (the
+# format is byte-oriented LZ77 with no entropy stage — one wave per page
+# saturates it). ZSTD/GZIP carry an entropy stage (FSE/Huffman, DEFLATE)
+# whose bit-serial decode is a poor fit for wave64 SIMD; they decompress
+# on the HOST through a thread pool (pyarrow codecs release the GIL), which
+# keeps the page pipeline full while staying honest about where the work
+# runs. See profiles/ for the measured SF1K ZSTD scan.
 def _decompress(codec, data, uncompressed_size):
     if codec == CODEC_UNCOMPRESSED:
         return data
-    if codec == CODEC_SNAPPY:
-        import pyarrow as pa
-        return pa.Codec("snappy").decompress(
-            data, decompressed_size=uncompressed_size).to_pybytes()
-    raise NotImplementedError(f"parquet codec {codec}")
+    name = _HOST_CODECS.get(codec)
+    if name is None:
+        raise NotImplementedError(f"parquet codec {codec}")
+    import pyarrow as pa
+    return pa.Codec(name).decompress(
+        data, decompressed_size=uncompressed_size).to_pybytes()
+
+
+_DECOMP_POOL = None
+
+
+def _decomp_pool():
+    global _DECOMP_POOL
+    if _DECOMP_POOL is None:
+        import concurrent.futures
+        import os
+        _DECOMP_POOL = concurrent.futures.ThreadPoolExecutor(
+            max_workers=min(32, (os.cpu_count() or 8)))
+    return _DECOMP_POOL
+
+
+def _bulk_host_decompress(pages):
+    """Decompress host-codec pages in parallel, in place."""
+    import pyarrow as pa
+    jobs = []
+    for p in pages:
+        if p.comp and p.codec in (CODEC_GZIP, CODEC_ZSTD, CODEC_LZ4_RAW):
+            jobs.append(p)
+    if not jobs:
+        return
+    codecs = {c: pa.Codec(n) for c, n in _HOST_CODECS.items()}
+
+    def run(p):
+        lv = p.rep_bytes + p.def_bytes if p.kind == 2 else 0
+        body = codecs[p.codec].decompress(
+            bytes(p.data[lv:]), decompressed_size=p.uncomp - lv).to_pybytes()
+        p.data = (bytes(p.data[:lv]) + body) if lv else body
+        p.comp = False
+        return None
+
+    list(_decomp_pool().map(run, jobs))
 
 
 def _parse_page_header(h, raw, pos):
@@ -288,18 +338,19 @@ def _walk_pages(raw, chunk: ColumnChunkMeta,
         pos = end + comp
         if ptype == 0:  # DATA_PAGE v1
             dph = ph[5]
-            if keep_compressed and chunk.codec == CODEC_SNAPPY:
+            if keep_compressed and chunk.codec != CODEC_UNCOMPRESSED:
                 pages.append(_Page(0, dph[1], dph[2], payload, comp=True,
-                                   uncomp=uncomp))
+                                   uncomp=uncomp, codec=chunk.codec))
             else:
                 data = _decompress(chunk.codec, payload, uncomp)
                 pages.append(_Page(0, dph[1], dph[2], data))
             values_seen += dph[1]
         elif ptype == 2:  # DICTIONARY_PAGE
             dph = ph[7]
-            if keep_compressed and chunk.codec == CODEC_SNAPPY:
+            if keep_compressed and chunk.codec != CODEC_UNCOMPRESSED:
                 pages.append(_Page(1, dph[1], dph.get(2, ENC_PLAIN), payload,
-                                   comp=True, uncomp=uncomp))
+                                   comp=True, uncomp=uncomp,
+                                   codec=chunk.codec))
             else:
                 data = _decompress(chunk.codec, payload, uncomp)
                 pages.append(_Page(1, dph[1], dph.get(2, ENC_PLAIN), data))
@@ -308,15 +359,23 @@ def _walk_pages(raw, chunk: ColumnChunkMeta,
             nv = dph[1]
             dlen = dph.get(5, 0)
             rlen = dph.get(6, 0)
-            # v2 layout: [rep levels][def levels][body]
+            # v2 layout: [rep levels (uncompressed)][def levels
+            # (uncompressed)][body (maybe compressed)]
             lv = rlen + dlen
-            if dph.get(7, True) and chunk.codec != CODEC_UNCOMPRESSED:
-                body = _decompress(chunk.codec, payload[lv:], uncomp - lv)
-                data = bytes(payload[:lv]) + bytes(body)
+            body_comp = dph.get(7, True) and chunk.codec != CODEC_UNCOMPRESSED
+            if body_comp and keep_compressed:
+                pages.append(_Page(2, nv, dph[4], payload, def_bytes=dlen,
+                                   num_nulls=dph.get(2, -1), rep_bytes=rlen,
+                                   comp=True, uncomp=uncomp,
+                                   codec=chunk.codec))
             else:
-                data = payload  # levels + body are already contiguous
-            pages.append(_Page(2, nv, dph[4], data, def_bytes=dlen,
-                               num_nulls=dph.get(2, -1), rep_bytes=rlen))
+                if body_comp:
+                    body = _decompress(chunk.codec, payload[lv:], uncomp - lv)
+                    data = bytes(payload[:lv]) + bytes(body)
+                else:
+                    data = payload  # levels + body are already contiguous
+                pages.append(_Page(2, nv, dph[4], data, def_bytes=dlen,
+                                   num_nulls=dph.get(2, -1), rep_bytes=rlen))
             values_seen += nv
         else:
             pass  # index page: skip
@@ -332,7 +391,9 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     dtype = _field_dtype(f)
 
     # flatten pages across row groups; snappy payloads stay compressed and
-    # are decompressed ON DEVICE (one wave per page) for non-BOOLEAN columns
+    # are decompressed ON DEVICE (one wave per page) for non-BOOLEAN
+    # columns; ZSTD/GZIP/LZ4 pages decompress on the host thread pool
+    # (entropy-coded formats — see _decompress policy note)
     keep_comp = dev.type == "cuda" and f.physical_type != T_BOOLEAN
     pages: List[_Page] = []
     dict_per_page: List[int] = []   # index into dict list, -1 none
@@ -346,6 +407,7 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
             else:
                 pages.append(p)
                 dict_per_page.append(cur_dict)
+    _bulk_host_decompress(pages + dicts)
 
     # upload all page payloads in one buffer: stage through PINNED host
     # memory (CachingHostAllocator keeps the source alive across the async
@@ -377,7 +439,9 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     dict_base_idx = len(pages)
 
     # device snappy decompression of compressed blobs; pbase is repointed at
-    # the decompressed copies so everything downstream is codec-agnostic
+    # the decompressed copies so everything downstream is codec-agnostic.
+    # v2 pages keep their uncompressed level prefix: the levels bytes are
+    # copied device-side and only the body feeds the snappy kernel.
     allp = pages + dicts
     comp_ids = [j for j, p in enumerate(allp) if p.comp]
     if comp_ids:
@@ -388,8 +452,16 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                            device=dev)
         sd = bytearray()
         for k, j in enumerate(comp_ids):
-            sd += struct.pack(_SNAP_FMT, pbase[j], len(allp[j].data),
-                              big2.data_ptr() + int(offs2[k]), allp[j].uncomp)
+            p = allp[j]
+            lv = p.rep_bytes + p.def_bytes if p.kind == 2 else 0
+            if lv:
+                src_idx = pbase[j] - big.data_ptr()
+                dst_idx = int(offs2[k])
+                big2[dst_idx:dst_idx + lv] = big[src_idx:src_idx + lv]
+            sd += struct.pack(_SNAP_FMT, pbase[j] + lv,
+                              len(p.data) - lv,
+                              big2.data_ptr() + int(offs2[k]) + lv,
+                              p.uncomp - lv)
         sdt = torch.frombuffer(sd, dtype=torch.uint8).to(dev)
         g.pq_snappy_decomp(sdt.data_ptr(), len(comp_ids), stream)
         for k, j in enumerate(comp_ids):

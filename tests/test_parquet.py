@@ -405,3 +405,49 @@ def test_decode_decimal_flba(tmp_path):
             if u >= 2**127:
                 u -= 2**128
             assert u == int(exp.scaleb(scale)), (i, u, exp)
+
+
+@pytest.mark.parametrize("codec", ["ZSTD", "GZIP", "LZ4"])
+@pytest.mark.gpu
+def test_decode_host_codecs(tmp_path, codec):
+    """ZSTD/GZIP/LZ4 pages decompress on the host thread pool before the
+    device decode kernels run."""
+    t, oracle = _make_table(3000)
+    p = str(tmp_path / f"{codec}.parquet")
+    pq.write_table(t, p, compression=codec, use_dictionary=True,
+                   row_group_size=900)
+    _check(p, oracle)
+
+
+@pytest.mark.parametrize("codec", ["ZSTD", "GZIP"])
+@pytest.mark.parametrize("pagever", ["1.0", "2.0"])
+@pytest.mark.gpu
+def test_decode_host_codecs_gpu(tmp_path, codec, pagever):
+    t, oracle = _make_table(3000)
+    p = str(tmp_path / f"{codec}_{pagever}.parquet")
+    pq.write_table(t, p, compression=codec, use_dictionary=True,
+                   data_page_version=pagever, row_group_size=900)
+    _check(p, oracle)
+
+
+@pytest.mark.gpu
+def test_decode_snappy_v2_device(tmp_path):
+    """v2 pages keep their uncompressed level prefix; the snappy body now
+    decompresses on device (previously host)."""
+    t, oracle = _make_table(4000)
+    p = str(tmp_path / "snv2.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=True,
+                   data_page_version="2.0", row_group_size=1100)
+    _check(p, oracle)
+
+
+@pytest.mark.gpu
+def test_decode_zstd_list(tmp_path):
+    vals = [None if i % 9 == 4 else
+            [random.randint(0, 100) for _ in range(i % 5)]
+            for i in range(1200)]
+    t = pa.table({"lst": pa.array(vals, type=pa.list_(pa.int64()))})
+    p = str(tmp_path / "zl.parquet")
+    pq.write_table(t, p, compression="ZSTD", row_group_size=500)
+    got = srj_pq.read_table(p, device="cuda")
+    assert got.columns[0].to_pylist() == vals
