@@ -155,8 +155,8 @@ def main():
                          "(default all 99)")
     ap.add_argument("--groups", type=int, default=1_000_000,
                     help="distinct groups for --op groupby")
-    ap.add_argument("--codec", choices=["none", "snappy"], default="none",
-                    help="parquet codec for --op parquet")
+    ap.add_argument("--codec", choices=["none", "snappy", "zstd", "gzip"],
+                    default="none", help="parquet codec for --op parquet")
     args = ap.parse_args()
 
     world, rank, local_rank = setup_dist(args)
@@ -304,6 +304,28 @@ def run_nds(args, world, rank, local_rank):
         dist.destroy_process_group()
 
 
+def _write_store_sales_file(job):
+    """Worker: write one store_sales-shaped parquet file (seeded by index)."""
+    path, idx, rows, codec = job
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as papq
+    rng = np.random.default_rng(1000 + idx)
+    tbl = pa.table({
+        "ss_sold_date_sk": rng.integers(2450000, 2452555, rows,
+                                        dtype=np.int64),
+        "ss_item_sk": rng.integers(1, 300_000, rows, dtype=np.int64),
+        "ss_quantity": rng.integers(1, 100, rows, dtype=np.int64),
+        "ss_sales_price": rng.random(rows) * 100.0,
+        "ss_ext_sales_price": rng.random(rows) * 1000.0,
+    })
+    kw = {"compression": codec, "row_group_size": 8_000_000}
+    if codec == "ZSTD":
+        kw["compression_level"] = 1
+    papq.write_table(tbl, path, **kw)
+    return path
+
+
 def run_secondary(args, world, rank, local_rank):
     """Secondary benches: BASELINE config[1] hash-aggregate and the q3-shaped
     pipeline. Single-GPU metrics (weak-scaled aggregate when launched under
@@ -328,43 +350,52 @@ def run_secondary(args, world, rank, local_rank):
         rows_per_step = n
     elif args.op == "parquet":
         # BASELINE config[3]: parquet scan -> filter -> project over a
-        # store_sales-shaped file. The file is written once (pyarrow, outside
-        # the timed region); each step re-reads it from the page cache,
-        # decodes the pages on the GPU, filters a date window and projects a
-        # price sum. Row count set by --probe-rows (default 100M; SF1K's
-        # 2.88B-row table would not fit a bench-length write).
-        import numpy as np
-        import pyarrow as pa
-        import pyarrow.parquet as papq
+        # store_sales-shaped dataset. Files are written once (pyarrow,
+        # parallel worker processes, outside the timed region) as a
+        # multi-file dataset — SF1K (2.88B rows: --probe-rows 2880000000)
+        # writes 50M-row files; each step decodes every file on the GPU,
+        # filters a date window and projects a price sum, accumulating
+        # across files (bounded device memory, Spark-partition style).
         from spark_rapids_jni_amd import parquet as srj_pq
         n = args.probe_rows if args.probe_rows != PROBE_ROWS_DEFAULT \
             else 100_000_000
-        codec = "SNAPPY" if args.codec == "snappy" else "NONE"
-        path = f"/tmp/bench_store_sales_{rank}_{args.codec}.parquet"
-        rng = np.random.default_rng(3)
-        log(rank, f"[bench] writing {n}-row store_sales parquet (one-time)")
-        papq.write_table(pa.table({
-            "ss_sold_date_sk": rng.integers(2450000, 2452555, n,
-                                            dtype=np.int64),
-            "ss_item_sk": rng.integers(1, 300_000, n, dtype=np.int64),
-            "ss_quantity": rng.integers(1, 100, n, dtype=np.int64),
-            "ss_sales_price": rng.random(n) * 100.0,
-            "ss_ext_sales_price": rng.random(n) * 1000.0,
-        }), path, compression=codec, row_group_size=8_000_000)
+        codec = {"snappy": "SNAPPY", "zstd": "ZSTD",
+                 "gzip": "GZIP"}.get(args.codec, "NONE")
+        per_file = min(n, 50_000_000)
+        nfiles = (n + per_file - 1) // per_file
+        pdir = f"/tmp/bench_store_sales_{rank}_{args.codec}"
+        os.makedirs(pdir, exist_ok=True)
+        paths = [os.path.join(pdir, f"part-{i:04d}.parquet")
+                 for i in range(nfiles)]
+        todo = [(p, i, per_file if i < nfiles - 1 or n % per_file == 0
+                 else n % per_file, codec)
+                for i, p in enumerate(paths) if not os.path.exists(p)]
+        if todo:
+            log(rank, f"[bench] writing {len(todo)} x {per_file}-row "
+                      f"store_sales files ({codec}, parallel)")
+            import concurrent.futures
+            with concurrent.futures.ProcessPoolExecutor(
+                    max_workers=min(16, len(todo))) as ex:
+                list(ex.map(_write_store_sales_file, todo))
 
         def step():
-            t = srj_pq.read_table(path, columns=["ss_sold_date_sk",
-                                                 "ss_quantity",
-                                                 "ss_sales_price"],
-                                  device=str(device))
-            d = t.columns[0].data
-            mask = (d >= 2450500) & (d < 2451500)
-            price = t.columns[2].data[mask]
-            s = (price * t.columns[1].data[mask]).sum()
-            del t, mask, price, s
+            total = torch.zeros(1, dtype=torch.float64, device=device)
+            rows = 0
+            for p in paths:
+                t = srj_pq.read_table(p, columns=["ss_sold_date_sk",
+                                                  "ss_quantity",
+                                                  "ss_sales_price"],
+                                      device=str(device))
+                d = t.columns[0].data
+                mask = (d >= 2450500) & (d < 2451500)
+                price = t.columns[2].data[mask]
+                total += (price * t.columns[1].data[mask]).sum()
+                rows += t.num_rows
+                del t, mask, price
+            assert rows == n, (rows, n)
         metric, model = "parquet_scan_rows_per_sec", \
             "parquet scan->filter->project, store_sales-shaped " \
-            f"({n} rows, 5 cols, {args.codec})"
+            f"({n} rows, 5 cols, {args.codec}, {nfiles} files)"
         rows_per_step = n
     else:  # q3
         from spark_rapids_jni_amd import exec as ex
