@@ -264,6 +264,11 @@ _HOST_CODECS = {CODEC_SNAPPY: "snappy", CODEC_GZIP: "gzip",
 def _decompress(codec, data, uncompressed_size):
     if codec == CODEC_UNCOMPRESSED:
         return data
+    if codec == CODEC_GZIP:
+        # zlib handles gzip framing directly (pyarrow's GZipCodec trips
+        # over its own length check on valid page streams)
+        import zlib
+        return zlib.decompress(bytes(data), wbits=47)
     name = _HOST_CODECS.get(codec)
     if name is None:
         raise NotImplementedError(f"parquet codec {codec}")
@@ -298,8 +303,13 @@ def _bulk_host_decompress(pages):
 
     def run(p):
         lv = p.rep_bytes + p.def_bytes if p.kind == 2 else 0
-        body = codecs[p.codec].decompress(
-            bytes(p.data[lv:]), decompressed_size=p.uncomp - lv).to_pybytes()
+        if p.codec == CODEC_GZIP:
+            import zlib
+            body = zlib.decompress(bytes(p.data[lv:]), wbits=47)
+        else:
+            body = codecs[p.codec].decompress(
+                bytes(p.data[lv:]),
+                decompressed_size=p.uncomp - lv).to_pybytes()
         p.data = (bytes(p.data[:lv]) + body) if lv else body
         p.comp = False
         return None
