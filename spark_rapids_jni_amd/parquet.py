@@ -60,9 +60,13 @@ class SchemaField:
     element: Optional["SchemaField"] = None
     max_def: int = 0         # leaf max definition level
     max_rep: int = 0         # leaf max repetition level
-    # flat STRUCT group: leaf fields (each mapping to one column chunk)
+    # STRUCT group: child fields (leaves or nested structs)
     is_struct: bool = False
     children: Optional[list] = None
+    # MAP group (LIST<STRUCT<key,value>> assembly from two leaf chunks)
+    is_map: bool = False
+    map_key: Optional["SchemaField"] = None
+    map_value: Optional["SchemaField"] = None
 
 
 @dataclass
@@ -137,59 +141,85 @@ def read_footer(path_or_bytes) -> ParquetFooter:
 
     schema_elems = fmd[2]
     root = schema_elems[0]
-    fields = []
-    i = 1
-    nchild_root = root.get(5, 0)
-    for _ in range(nchild_root):
+
+    def leaf_field(se, d_above):
+        own = 1 if se.get(3, 0) == 1 else 0
+        return SchemaField(
+            name=se[4].decode(), physical_type=se.get(1, -1),
+            repetition=se.get(3, 0), converted_type=se.get(6),
+            scale=se.get(7, 0), precision=se.get(8, 0),
+            logical=se.get(10), type_length=se.get(2, 0),
+            max_def=d_above + own)
+
+    def is_list_group(se):
+        return se.get(6) == 3 or (se.get(10) and 3 in se.get(10, {}))
+
+    def is_map_group(se):
+        return se.get(6) in (1, 2) or (se.get(10) and 2 in se.get(10, {}))
+
+    def parse_field(i, d_above):
+        """Parse schema element i (and subtree) -> (SchemaField, next_i).
+        d_above = definition level contributed by ancestors."""
         se = schema_elems[i]
         nch = se.get(5, 0)
         if nch == 0:
-            fields.append(SchemaField(
-                name=se[4].decode(), physical_type=se.get(1, -1),
-                repetition=se.get(3, 0), converted_type=se.get(6),
-                scale=se.get(7, 0), precision=se.get(8, 0),
-                logical=se.get(10), type_length=se.get(2, 0)))
-            i += 1
-            continue
-        if not (se.get(6) == 3 or (se.get(10) and 3 in se.get(10, {}))):
-            # plain group = STRUCT of leaves (non-repeated)
-            outer_opt = 1 if se.get(3, 0) == 1 else 0
-            leaves = []
-            for k in range(nch):
-                lf = schema_elems[i + 1 + k]
-                assert lf.get(5, 0) == 0, \
-                    "nested struct-of-struct not supported"
-                leaves.append(SchemaField(
-                    name=lf[4].decode(), physical_type=lf.get(1, -1),
-                    repetition=lf.get(3, 0), converted_type=lf.get(6),
-                    scale=lf.get(7, 0), precision=lf.get(8, 0),
-                    logical=lf.get(10), type_length=lf.get(2, 0),
-                    max_def=outer_opt + (1 if lf.get(3, 0) == 1 else 0)))
-            fields.append(SchemaField(
-                name=se[4].decode(), physical_type=-1,
-                repetition=se.get(3, 0), converted_type=None, is_struct=True,
-                children=leaves))
-            i += 1 + nch
-            continue
-        # 3-level LIST group: optional group (LIST) { repeated group list {
-        #   <element leaf> } }
-        rep_grp = schema_elems[i + 1]
-        assert rep_grp.get(3, 0) == 2 and rep_grp.get(5, 0) == 1, \
-            "unrecognized LIST encoding"
-        leaf = schema_elems[i + 2]
-        assert leaf.get(5, 0) == 0, "LIST of nested types not supported"
-        elem = SchemaField(
-            name=leaf[4].decode(), physical_type=leaf.get(1, -1),
-            repetition=leaf.get(3, 0), converted_type=leaf.get(6),
-            scale=leaf.get(7, 0), precision=leaf.get(8, 0),
-            logical=leaf.get(10))
+            return leaf_field(se, d_above), i + 1
         outer_opt = 1 if se.get(3, 0) == 1 else 0
-        elem_opt = 1 if leaf.get(3, 0) == 1 else 0
-        fields.append(SchemaField(
-            name=se[4].decode(), physical_type=-1, repetition=se.get(3, 0),
-            converted_type=3, is_list=True, element=elem,
-            max_def=outer_opt + 1 + elem_opt, max_rep=1))
-        i += 3
+        if is_list_group(se):
+            # 3-level LIST: optional group (LIST) { repeated group list {
+            #   <element> } }
+            rep_grp = schema_elems[i + 1]
+            assert rep_grp.get(3, 0) == 2 and rep_grp.get(5, 0) == 1, \
+                "unrecognized LIST encoding"
+            el = schema_elems[i + 2]
+            if el.get(5, 0) != 0:
+                raise NotImplementedError(
+                    "LIST of nested types (list-of-list / list-of-struct) "
+                    "is not supported yet")
+            elem = leaf_field(el, 0)
+            elem_opt = 1 if el.get(3, 0) == 1 else 0
+            return SchemaField(
+                name=se[4].decode(), physical_type=-1,
+                repetition=se.get(3, 0), converted_type=3, is_list=True,
+                element=elem, max_def=d_above + outer_opt + 1 + elem_opt,
+                max_rep=1), i + 3
+        if is_map_group(se):
+            # MAP: optional group (MAP) { repeated group key_value {
+            #   required <key>; optional <value> } }
+            kv = schema_elems[i + 1]
+            assert kv.get(3, 0) == 2 and kv.get(5, 0) == 2, \
+                "unrecognized MAP encoding"
+            kse = schema_elems[i + 2]
+            vse = schema_elems[i + 3]
+            if kse.get(5, 0) != 0 or vse.get(5, 0) != 0:
+                raise NotImplementedError("MAP of nested key/value types "
+                                          "is not supported yet")
+            key = leaf_field(kse, 0)
+            val = leaf_field(vse, 0)
+            return SchemaField(
+                name=se[4].decode(), physical_type=-1,
+                repetition=se.get(3, 0), converted_type=2, is_list=False,
+                is_map=True, map_key=key, map_value=val,
+                max_def=d_above + outer_opt + 1, max_rep=1), i + 4
+        # plain group = STRUCT; children may be leaves or nested STRUCTs
+        children = []
+        j = i + 1
+        for _ in range(nch):
+            ch, j = parse_field(j, d_above + outer_opt)
+            if ch.is_list:
+                raise NotImplementedError(
+                    "LIST nested inside STRUCT is not supported yet")
+            children.append(ch)
+        return SchemaField(
+            name=se[4].decode(), physical_type=-1,
+            repetition=se.get(3, 0), converted_type=None, is_struct=True,
+            max_def=d_above + outer_opt, children=children), j
+
+    fields = []
+    i = 1
+    for _ in range(root.get(5, 0)):
+        f, i = parse_field(i, 0)
+        fields.append(f)
     assert i == len(schema_elems), "unsupported schema shape"
 
     row_groups = []
@@ -988,21 +1018,34 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                   null_count=None)
 
 
+def _flatten_struct_leaves(f: SchemaField, out=None):
+    """DFS leaf fields of a (possibly nested) struct — parquet chunk order."""
+    if out is None:
+        out = []
+    for ch in f.children:
+        if ch.is_struct:
+            _flatten_struct_leaves(ch, out)
+        else:
+            out.append(ch)
+    return out
+
+
 def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                         total_rows: int, device) -> Column:
-    """Flat STRUCT<primitive|string> decode: every leaf is a plain column
-    whose definition levels carry BOTH the struct's and the leaf's
-    nullability (position space == row space, no repetition). Values reuse
-    the flat-path kernels; the struct validity falls out of the first
-    leaf's levels (def >= 1 means the struct is present)."""
+    """STRUCT decode (arbitrarily nested structs of primitives/strings):
+    every leaf is a plain column whose definition levels carry the whole
+    ancestor chain's nullability plus its own (position space == row
+    space, no repetition). Values reuse the flat-path kernels; each struct
+    level's validity falls out of its first descendant leaf's levels
+    (def >= node.max_def means that struct level is present)."""
     g = _native.gpu()
     stream = _native.current_stream()
     dev = torch.device(device)
-    outer_opt = f.repetition == 1
+    leaves = _flatten_struct_leaves(f)
     children = []
-    struct_validity = None
+    leaf_levs = []  # per-leaf int64 def levels (row space), None if max_def==0
 
-    for li, leaf in enumerate(f.children):
+    for li, leaf in enumerate(leaves):
         chunks = [rg.columns[leaf0 + li] for rg in row_groups]
         max_def = leaf.max_def
         def_bw = max(max_def.bit_length(), 1)
@@ -1059,10 +1102,8 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
             body_offs = [0] * len(pages)
             lev_t.fill_(0)
 
-        if struct_validity is None and outer_opt:
-            from .ops.aggregate import _validity_from_bool
-            struct_validity = _validity_from_bool(
-                lev_t[:total_rows].to(torch.int64) >= 1)
+        leaf_levs.append(lev_t[:max(total_rows, 1)].to(torch.int64)
+                         if max_def > 0 else None)
 
         leaf_def = (lev_t[:max(total_rows, 1)].to(torch.int64) ==
                     max_def).to(torch.uint8) if max_def > 0 else \
@@ -1211,8 +1252,61 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                                    leaf_validity, scale=leaf.scale,
                                    null_count=None))
 
-    return Column(DType.STRUCT, total_rows, None, struct_validity, None,
-                  children, null_count=None)
+    from .ops.aggregate import _validity_from_bool
+
+    def first_leaf_index(node, base):
+        return base
+
+    def assemble(node, li):
+        if not node.is_struct:
+            return children[li], li + 1
+        first = li
+        kids = []
+        for ch in node.children:
+            col, li = assemble(ch, li)
+            kids.append(col)
+        validity = None
+        if node.repetition == 1 and total_rows and \
+                leaf_levs[first] is not None:
+            validity = _validity_from_bool(
+                leaf_levs[first][:total_rows] >= node.max_def)
+        return Column(DType.STRUCT, total_rows, None, validity, None,
+                      kids, null_count=None), li
+
+    col, used = assemble(f, 0)
+    assert used == len(leaves)
+    return col
+
+
+def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
+                     total_rows: int, device) -> Column:
+    """MAP decode by composition: each of the key/value leaves has exactly
+    the level structure of a 3-level LIST of that leaf (outer optional
+    group + repeated key_value), so both decode through the LIST machinery
+    and zip into LIST<STRUCT<key, value>> — the Spark map layout. The
+    entry offsets come from the key side (identical on both by
+    construction)."""
+    outer_opt = 1 if f.repetition == 1 else 0
+    key_f = SchemaField(
+        name=f.name, physical_type=-1, repetition=f.repetition,
+        converted_type=3, is_list=True, element=f.map_key,
+        max_def=outer_opt + 1 + (1 if f.map_key.repetition == 1 else 0),
+        max_rep=1)
+    val_f = SchemaField(
+        name=f.name, physical_type=-1, repetition=f.repetition,
+        converted_type=3, is_list=True, element=f.map_value,
+        max_def=outer_opt + 1 + (1 if f.map_value.repetition == 1 else 0),
+        max_rep=1)
+    kchunks = [rg.columns[leaf0] for rg in row_groups]
+    vchunks = [rg.columns[leaf0 + 1] for rg in row_groups]
+    klist = _read_list_column(raw, key_f, kchunks, total_rows, device)
+    vlist = _read_list_column(raw, val_f, vchunks, total_rows, device)
+    nentries = klist.children[0].size
+    entries = Column(DType.STRUCT, nentries, None, None, None,
+                     [klist.children[0], vlist.children[0]],
+                     null_count=None)
+    return Column(DType.LIST, total_rows, None, klist.validity,
+                  klist.offsets, [entries], null_count=None)
 
 
 def read_table(path: str, columns: Optional[Sequence[str]] = None,
@@ -1234,7 +1328,12 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
         if f.is_struct:
             cols.append(_read_struct_column(raw, f, footer.row_groups, leaf,
                                             total_rows, device))
-            leaf += len(f.children)
+            leaf += len(_flatten_struct_leaves(f))
+            continue
+        if f.is_map:
+            cols.append(_read_map_column(raw, f, footer.row_groups, leaf,
+                                         total_rows, device))
+            leaf += 2
             continue
         chunks = [rg.columns[leaf] for rg in footer.row_groups]
         if f.is_list:

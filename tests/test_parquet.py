@@ -451,3 +451,70 @@ def test_decode_zstd_list(tmp_path):
     pq.write_table(t, p, compression="ZSTD", row_group_size=500)
     got = srj_pq.read_table(p, device="cuda")
     assert got.columns[0].to_pylist() == vals
+
+
+@pytest.mark.gpu
+def test_decode_struct_of_struct(tmp_path):
+    n = 2500
+    vals = []
+    for i in range(n):
+        if i % 11 == 3:
+            vals.append(None)
+        elif i % 7 == 2:
+            vals.append({"a": i, "inner": None,
+                         "s": None if i % 5 == 0 else f"x{i}"})
+        else:
+            vals.append({"a": i,
+                         "inner": {"b": i * 2,
+                                   "c": None if i % 3 == 0 else i * 1.5},
+                         "s": f"x{i}"})
+    typ = pa.struct([("a", pa.int64()),
+                     ("inner", pa.struct([("b", pa.int64()),
+                                          ("c", pa.float64())])),
+                     ("s", pa.string())])
+    t = pa.table({"st": pa.array(vals, type=typ)})
+    p = str(tmp_path / "sos.parquet")
+    pq.write_table(t, p, compression="SNAPPY", row_group_size=800)
+    got = srj_pq.read_table(p, device="cuda").columns[0].to_pylist()
+    for i, (g, x) in enumerate(zip(got, vals)):
+        if x is None:
+            assert g is None, i
+            continue
+        assert g is not None, i
+        a, inner, s = g
+        assert a == x["a"] and s == x["s"], i
+        if x["inner"] is None:
+            assert inner is None, i
+        else:
+            b, c = inner
+            assert b == x["inner"]["b"], i
+            if x["inner"]["c"] is None:
+                assert c is None, i
+            else:
+                assert abs(c - x["inner"]["c"]) < 1e-9, i
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("codec", ["NONE", "ZSTD"])
+def test_decode_map_column(tmp_path, codec):
+    import random as _r
+    _r.seed(4)
+    n = 2000
+    vals = []
+    for i in range(n):
+        if i % 13 == 5:
+            vals.append(None)
+        else:
+            m = [(f"k{j}", None if (i + j) % 7 == 0 else i * 10 + j)
+                 for j in range(i % 4)]
+            vals.append(m)
+    t = pa.table({"m": pa.array(vals, type=pa.map_(pa.string(),
+                                                   pa.int64()))})
+    p = str(tmp_path / f"map_{codec}.parquet")
+    pq.write_table(t, p, compression=codec, row_group_size=700)
+    got = srj_pq.read_table(p, device="cuda").columns[0].to_pylist()
+    for i, (g, x) in enumerate(zip(got, vals)):
+        if x is None:
+            assert g is None, i
+        else:
+            assert g == [tuple(kv) for kv in x], (i, g, x)
