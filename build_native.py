@@ -133,6 +133,23 @@ def build(verbose=False):
             if r.returncode != 0:
                 raise RuntimeError(f"faultinj build failed:\n{r.stderr}")
             cache[ck] = key
+    # kernel-activity capture lib (roctracer; reference CUPTI activity)
+    ka = ROOT / "src" / "tools" / "kernel_activity.cpp"
+    if ka.exists():
+        out = PKG / "libsrjktrace.so"
+        key = _key(ka, CXX_FLAGS, [])
+        ck = str(out)
+        if not out.exists() or cache.get(ck) != key:
+            r = subprocess.run(
+                [CXX] + CXX_FLAGS + ["-I/opt/rocm/include",
+                                     "-I/opt/rocm/include/roctracer",
+                                     "-shared", str(ka), "-o", str(out),
+                                     "-L/opt/rocm/lib", "-lroctracer64",
+                                     "-Wl,-rpath,/opt/rocm/lib"],
+                capture_output=True, text=True)
+            if r.returncode != 0:
+                raise RuntimeError(f"kernel_activity build failed:\n{r.stderr}")
+            cache[ck] = key
     if host_objs:
         link(host_objs, PKG / "_host.so")
     CACHE.write_text(json.dumps(cache))

@@ -93,6 +93,64 @@ class Profiler:
             self._flush_locked()
 
 
+class KernelTracer:
+    """Kernel-level activity capture (reference ProfilerJni.cpp:267-292
+    CUPTI activity analog): roctracer's HIP_OPS activity domain delivers
+    per-kernel dispatch begin/end timestamps and names in-process through
+    libsrjktrace.so. Records flow into the active Profiler stream as
+    kind=2 (kernel) so converted traces attribute time to `srj::` kernels,
+    not just ops."""
+
+    def __init__(self):
+        import ctypes
+        path = os.path.join(os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))), "libsrjktrace.so")
+        self._lib = ctypes.CDLL(path)
+        self._lib.srj_ktrace_count.restype = ctypes.c_long
+        self._ct = ctypes
+
+    def start(self):
+        rc = self._lib.srj_ktrace_start()
+        if rc != 0:
+            raise RuntimeError(f"roctracer activity start failed ({rc})")
+
+    def stop(self):
+        self._lib.srj_ktrace_stop()
+
+    def clear(self):
+        self._lib.srj_ktrace_clear()
+
+    def records(self):
+        ct = self._ct
+        n = self._lib.srj_ktrace_count()
+        name = ct.create_string_buffer(1024)
+        out = (ct.c_uint64 * 6)()
+        recs = []
+        for i in range(n):
+            if self._lib.srj_ktrace_get(ct.c_long(i), name, 1024, out) == 0:
+                recs.append(dict(name=name.value.decode(errors="replace"),
+                                 op=int(out[0]), begin_ns=int(out[1]),
+                                 end_ns=int(out[2]), correlation=int(out[3]),
+                                 device=int(out[4]), queue=int(out[5])))
+        return recs
+
+    def drain_into(self, profiler: "Profiler"):
+        """Append captured kernel records to a Profiler stream (kind=2,
+        tid encodes the device queue so kernels get their own track)."""
+        for r in self.records():
+            nb = r["name"].encode()
+            rec = struct.pack(
+                "<BQQIH", 2, r["begin_ns"],
+                max(r["end_ns"] - r["begin_ns"], 0),
+                0x40000000 | (r["device"] << 8) | (r["queue"] & 0xFF),
+                len(nb)) + nb
+            with profiler._lock:
+                profiler._buf.write(struct.pack("<I", len(rec)))
+                profiler._buf.write(rec)
+        profiler.flush()
+        self.clear()
+
+
 @contextlib.contextmanager
 def srj_func_range(name: str, use_roctx: bool = True):
     """SRJ_FUNC_RANGE() analog: rocTX push/pop (visible in rocprofv3) plus a
@@ -154,7 +212,9 @@ def convert_to_json(path: str, out_path: Optional[str] = None) -> str:
                            "tid": r["tid"], "ts": r["start_ns"] / 1000.0,
                            "s": "t"})
         else:
-            events.append({"name": r["name"], "ph": "X", "pid": 0,
+            # kernel records (kind 2) render on their own "GPU" process row
+            events.append({"name": r["name"], "ph": "X",
+                           "pid": 1 if r["kind"] == 2 else 0,
                            "tid": r["tid"], "ts": r["start_ns"] / 1000.0,
                            "dur": r["dur_ns"] / 1000.0})
     doc = json.dumps({"traceEvents": events})

@@ -91,3 +91,43 @@ def test_fault_injection_shim(tmp_path):
     r2 = faultinj.run_with_faults([sys.executable, "-c", code], cfg,
                                   capture_output=True, text=True, timeout=300)
     assert "OK" in r2.stdout, (r2.stdout, r2.stderr)
+
+
+@pytest.mark.gpu
+def test_kernel_activity_capture(tmp_path):
+    """Kernel-level capture (reference CUPTI activity analog): traces must
+    attribute time to named srj:: kernels, not just op ranges."""
+    import torch
+    from spark_rapids_jni_amd.columnar import Column
+    from spark_rapids_jni_amd.ops import hashing
+    from spark_rapids_jni_amd.tools.profiler import (KernelTracer, Profiler,
+                                                     convert_to_json,
+                                                     read_records)
+    kt = KernelTracer()
+    kt.start()
+    col = Column.from_torch(torch.arange(1_000_000, dtype=torch.int64,
+                                         device="cuda"))
+    h = hashing.murmur3([col])
+    torch.cuda.synchronize()
+    kt.stop()
+    recs = kt.records()
+    assert recs, "no kernel activity captured"
+    names = " ".join(r["name"] for r in recs)
+    assert "murmur3" in names, names[:500]
+    assert all(r["end_ns"] >= r["begin_ns"] for r in recs)
+
+    # drain into a profiler stream and convert: kernel rows appear
+    path = str(tmp_path / "prof.bin")
+    p = Profiler.init(path)
+    p.start()
+    kt2 = KernelTracer()
+    kt2.start()
+    hashing.murmur3([col])
+    torch.cuda.synchronize()
+    kt2.stop()
+    kt2.drain_into(p)
+    p.shutdown()
+    rr = read_records(path)
+    assert any(r["kind"] == 2 and "murmur3" in r["name"] for r in rr)
+    doc = convert_to_json(path)
+    assert "murmur3" in doc
