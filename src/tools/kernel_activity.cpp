@@ -46,11 +46,17 @@ void activity_cb(const char* begin, const char* end, void* /*arg*/) {
   while (record < end_record) {
     if (record->domain == ACTIVITY_DOMAIN_HIP_OPS) {
       KRecord r;
-      r.name = record->kernel_name
-                   ? std::string(record->kernel_name)
-                   : std::string(roctracer_op_string(record->domain,
-                                                     record->op,
-                                                     record->kind));
+      // the record union carries kernel_name ONLY for dispatch ops; copy
+      // and barrier records overlay `bytes` there (reading it as a
+      // pointer faults)
+      bool const is_dispatch = record->op == HIP_OP_ID_DISPATCH;
+      const char* opname =
+          roctracer_op_string(record->domain, record->op, record->kind);
+      if (is_dispatch && record->kernel_name != nullptr) {
+        r.name = record->kernel_name;
+      } else {
+        r.name = opname ? opname : "hip_op";
+      }
       r.op = record->op;
       r.begin_ns = record->begin_ns;
       r.end_ns = record->end_ns;
