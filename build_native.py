@@ -150,6 +150,21 @@ def build(verbose=False):
             if r.returncode != 0:
                 raise RuntimeError(f"kernel_activity build failed:\n{r.stderr}")
             cache[ck] = key
+    # JNI-mirroring C ABI surface (docs/JNI_ABI.md)
+    ja = ROOT / "src" / "host" / "jni_abi.cpp"
+    if ja.exists():
+        out = PKG / "_jniabi.so"
+        gen = ROOT / "src" / "host" / "jni_abi_generated.inc"
+        key = _key(ja, CXX_FLAGS, []) + (
+            _key(gen, [], []) if gen.exists() else "")
+        ck = str(out)
+        if not out.exists() or cache.get(ck) != key:
+            r = subprocess.run([CXX] + CXX_FLAGS + ["-shared", str(ja),
+                                "-o", str(out)], capture_output=True,
+                               text=True)
+            if r.returncode != 0:
+                raise RuntimeError(f"jni_abi build failed:\n{r.stderr}")
+            cache[ck] = key
     if host_objs:
         link(host_objs, PKG / "_host.so")
     CACHE.write_text(json.dumps(cache))
