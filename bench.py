@@ -44,6 +44,14 @@ def setup_dist(args):
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if getattr(args, "dry_run", False):
+        # CPU/gloo dry run: exercises the full world>1 control flow
+        # (rendezvous, hash partition, all-to-all, max-over-ranks timing)
+        # without a GPU — VERDICT r01 item 2 readiness check
+        if world > 1:
+            import torch.distributed as dist
+            dist.init_process_group("gloo")
+        return world, rank, local_rank
     if world > 1:
         import torch.distributed as dist
         torch.cuda.set_device(local_rank)
@@ -53,11 +61,35 @@ def setup_dist(args):
     return world, rank, local_rank
 
 
-def barrier_sync(world):
+def barrier_sync(world, dry_run=False):
     if world > 1:
         import torch.distributed as dist
         dist.barrier()
-    torch.cuda.synchronize()
+    if not dry_run:
+        torch.cuda.synchronize()
+
+
+def shuffle_exchange_dry(keys: torch.Tensor, world: int):
+    """gloo/CPU stand-in for shuffle_exchange: same partition->a2a control
+    flow, torch-side hash (the HIP murmur3 kernel needs a GPU)."""
+    import torch.distributed as dist
+    h = (keys * 0x9E3779B97F4A7C15) & (2**63 - 1)
+    pids = (h % world).to(torch.int64)
+    order = torch.argsort(pids, stable=True)
+    send = keys[order]
+    counts = torch.bincount(pids, minlength=world)
+    recv_counts = torch.empty_like(counts)
+    gathered = [torch.zeros_like(counts) for _ in range(world)]
+    dist.all_gather(gathered, counts)
+    me = dist.get_rank()
+    recv_counts = torch.tensor([int(g[me]) for g in gathered])
+    out = torch.empty(int(recv_counts.sum()), dtype=keys.dtype)
+    in_splits = counts.tolist()
+    out_splits = recv_counts.tolist()
+    from spark_rapids_jni_amd.parallel.exchange import _exchange_bytes
+    b, _os = _exchange_bytes(send.view(torch.uint8).view(-1),
+                             [s * 8 for s in in_splits])
+    return b.view(torch.int64)
 
 
 def shuffle_exchange(keys: torch.Tensor, world: int):
@@ -83,14 +115,27 @@ def shuffle_exchange(keys: torch.Tensor, world: int):
     return out
 
 
-def make_build_shard(n_build: int, world: int, rank: int, device):
+def make_build_shard(n_build: int, world: int, rank: int, device,
+                     dry_run=False):
     """Rank r's build shard: global key space [0, world*n_build) routed to the
     owning rank by pmod(murmur3), like Spark's build-side shuffle (one-time)."""
     local = torch.arange(rank * n_build, (rank + 1) * n_build, dtype=torch.int64,
                          device=device)
     if world == 1:
         return local
+    if dry_run:
+        return shuffle_exchange_dry(local, world)
     return shuffle_exchange(local, world)
+
+
+def one_step_dry(build, chunks, world):
+    """CPU stand-in for one_step: exchange + torch-side join probe count
+    (control flow identical; HIP kernels need a GPU)."""
+    bset = torch.zeros(1, dtype=torch.int64)
+    for chunk in chunks:
+        probe = shuffle_exchange_dry(chunk, world) if world > 1 else chunk
+        bset += torch.isin(probe, build).sum()
+    return int(bset.item())
 
 
 def one_step(build_col, chunks, world, out_hint, comm_stream=None):
@@ -153,6 +198,9 @@ def main():
     ap.add_argument("--queries", type=str, default="",
                     help="comma-separated query numbers for --op nds "
                          "(default all 99)")
+    ap.add_argument("--dry-run", action="store_true",
+                    help="CPU/gloo dry run of the world>1 control flow "
+                         "(no GPU needed; readiness check only)")
     ap.add_argument("--groups", type=int, default=1_000_000,
                     help="distinct groups for --op groupby")
     ap.add_argument("--codec", choices=["none", "snappy", "zstd", "gzip"],
@@ -168,6 +216,8 @@ def main():
         raise RuntimeError(
             f"--gpus {args.gpus} needs torchrun with --nproc-per-node {args.gpus} "
             f"(WORLD_SIZE is {world})")
+    if args.dry_run:
+        return run_join_dry(args, world, rank)
     device = torch.device("cuda", local_rank)
 
     free, total = torch.cuda.mem_get_info(device)
@@ -245,6 +295,56 @@ def main():
                 "seq_len": 0,
                 "parallelism": f"1 executor/GPU x{world}, RCCL all-to-all shuffle",
             },
+        }))
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+def run_join_dry(args, world, rank):
+    """World>1 readiness dry run (gloo/CPU): same phases as the real
+    bench — build-shard exchange, warmup, timed loop with barriers, max
+    elapsed over ranks, one JSON line — with torch stand-ins for the HIP
+    kernels."""
+    build = make_build_shard(args.build_rows, world, rank, "cpu",
+                             dry_run=True)
+    chunks = []
+    done = 0
+    key_space = world * args.build_rows
+    g = torch.Generator().manual_seed(11 + rank)
+    while done < args.probe_rows:
+        m = min(args.chunk_rows, args.probe_rows - done)
+        chunks.append(torch.randint(0, key_space, (m,), dtype=torch.int64,
+                                    generator=g))
+        done += m
+    for _ in range(args.warmup):
+        one_step_dry(build, chunks, world)
+    barrier_sync(world, dry_run=True)
+    t0 = time.perf_counter()
+    matches = 0
+    for _ in range(args.steps):
+        matches = one_step_dry(build, chunks, world)
+    barrier_sync(world, dry_run=True)
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    if rank == 0:
+        print(json.dumps({
+            "metric": "hash_join_probe_rows_per_sec", "dry_run": True,
+            "value": world * args.probe_rows * args.steps / elapsed,
+            "unit": "rows/s", "n_gpus": world, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "int64",
+            "data": "synthetic (CPU/gloo DRY RUN — not a measurement)",
+            "config": {"model": "hash-join dry run", "matches": matches,
+                       "global_batch": world * args.probe_rows,
+                       "seq_len": 0,
+                       "parallelism": f"x{world} gloo dry-run"},
         }))
     if world > 1:
         import torch.distributed as dist

@@ -58,3 +58,40 @@ def test_gloo_shuffle_exchange_world2():
     all_keys = sorted(results[0][0] + results[1][0])
     expect = sorted([r * 10000 + i for r in range(world) for i in range(200)])
     assert all_keys == expect
+
+
+def test_bench_world2_dry_run(tmp_path):
+    """`torchrun --nproc-per-node 2 bench.py --gpus 2`
+    must reach the timed loop on a CPU/gloo dry run (rendezvous, build
+    exchange, per-step all-to-all, max-over-ranks timing, JSON output)."""
+    import json
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29533",
+                "WORLD_SIZE": "2"})
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(repo, "bench.py"), "--gpus", "2",
+             "--dry-run", "--steps", "1", "--warmup", "1",
+             "--build-rows", "20000", "--probe-rows", "40000",
+             "--chunk-rows", "20000"],
+            env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            cwd=repo))
+    outs = []
+    for p in procs:
+        so, se = p.communicate(timeout=300)
+        assert p.returncode == 0, se.decode()[-2000:]
+        outs.append(so.decode())
+    js = [l for l in outs[0].splitlines() if l.startswith("{")]
+    assert js, outs[0]
+    d = json.loads(js[-1])
+    assert d["n_gpus"] == 2 and d["dry_run"] is True
+    # probe keys cover the whole build key space: every row matches
+    assert d["config"]["matches"] > 0
