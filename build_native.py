@@ -81,7 +81,7 @@ def _compile(src: Path, flags, headers, cache, log):
     return obj, r.stderr or None
 
 
-def build(verbose=False):
+def build(verbose=False, asan_host=False):
     OBJ.mkdir(parents=True, exist_ok=True)
     cache = _load_cache()
     log = []
@@ -108,6 +108,32 @@ def build(verbose=False):
     for s in sorted(SRC_GPU.glob("*.cpp")):
         jobs.append((s, gpu_headers, CXX_FLAGS + pyb))
     gpu_objs = batch(jobs, None)
+
+    if asan_host:
+        # sanitizer tier (reference pom.xml test-with-sanitizer profile;
+        # the MI355X analog is an ASAN lane over the host-native code):
+        # compile the host sources with AddressSanitizer into a side .so
+        # the tests load via SRJ_HOST_SO (see ci/sanitize.sh)
+        asan_dir = ROOT / ".build" / "asan"
+        asan_dir.mkdir(parents=True, exist_ok=True)
+        aflags = CXX_FLAGS + pyb + ["-fsanitize=address",
+                                    "-fno-omit-frame-pointer", "-g"]
+        objs = []
+        for s in sorted(SRC_HOST.glob("*.cpp")):
+            obj = asan_dir / (s.stem + ".o")
+            r = subprocess.run([CXX] + aflags + ["-c", str(s), "-o",
+                                str(obj)], capture_output=True, text=True)
+            if r.returncode != 0:
+                raise RuntimeError(f"asan compile failed:\n{r.stderr}")
+            objs.append(obj)
+        out = asan_dir / "_host_asan.so"
+        r = subprocess.run([CXX, "-shared", "-fsanitize=address"] +
+                           [str(o) for o in objs] + ["-o", str(out),
+                            "-pthread"], capture_output=True, text=True)
+        if r.returncode != 0:
+            raise RuntimeError(f"asan link failed:\n{r.stderr}")
+        print(f"asan host extension: {out}")
+        return
 
     host_jobs = [(s, host_headers, CXX_FLAGS + pyb) for s in sorted(SRC_HOST.glob("*.cpp"))]
     host_objs = batch(host_jobs, None)
@@ -173,4 +199,4 @@ def build(verbose=False):
 
 
 if __name__ == "__main__":
-    build(verbose=True)
+    build(verbose=True, asan_host="--asan-host" in sys.argv)

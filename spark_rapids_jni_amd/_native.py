@@ -48,11 +48,22 @@ def gpu_or_none():
 
 
 def host():
-    """Return the _host extension (CPU-only subsystems)."""
+    """Return the _host extension (CPU-only subsystems). SRJ_HOST_SO
+    overrides the module path — the ASAN sanitizer lane (ci/sanitize.sh)
+    points it at the AddressSanitizer build."""
     global _host_mod, _host_err
     if _host_mod is None and _host_err is None:
         try:
-            _host_mod = importlib.import_module("spark_rapids_jni_amd._host")
+            override = os.environ.get("SRJ_HOST_SO")
+            if override:
+                import importlib.util
+                spec = importlib.util.spec_from_file_location(
+                    "spark_rapids_jni_amd._host", override)
+                _host_mod = importlib.util.module_from_spec(spec)
+                spec.loader.exec_module(_host_mod)
+            else:
+                _host_mod = importlib.import_module(
+                    "spark_rapids_jni_amd._host")
         except Exception as e:  # pragma: no cover
             _host_err = e
     if _host_mod is None:
