@@ -53,10 +53,10 @@ def host():
     points it at the AddressSanitizer build."""
     global _host_mod, _host_err
     if _host_mod is None and _host_err is None:
+        import importlib.util
         try:
             override = os.environ.get("SRJ_HOST_SO")
             if override:
-                import importlib.util
                 spec = importlib.util.spec_from_file_location(
                     "spark_rapids_jni_amd._host", override)
                 _host_mod = importlib.util.module_from_spec(spec)
