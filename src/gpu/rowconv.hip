@@ -64,6 +64,71 @@ __device__ inline void copy_elem(uint8_t* dst, const uint8_t* src, int w) {
   }
 }
 
+
+// 4B-granular variants for the PITCHED LDS tile: the tile row pitch is
+// padded to an odd dword count to spread strided row accesses across all
+// 32 LDS banks (SQ_LDS_BANK_CONFLICT measured ~40x per instruction with
+// the natural multiple-of-8 pitch), which caps tile alignment at 4B.
+__device__ inline void copy_elem_z4(uint8_t* dst, const uint8_t* src, int w,
+                                    bool valid) {
+  switch (w) {
+    case 1: *dst = valid ? *src : 0; break;
+    case 2: *reinterpret_cast<uint16_t*>(dst) =
+                valid ? *reinterpret_cast<const uint16_t*>(src) : 0;
+            break;
+    case 4: *reinterpret_cast<uint32_t*>(dst) =
+                valid ? *reinterpret_cast<const uint32_t*>(src) : 0;
+            break;
+    case 8: {
+      uint64_t v = valid ? *reinterpret_cast<const uint64_t*>(src) : 0;
+      reinterpret_cast<uint32_t*>(dst)[0] = (uint32_t)v;
+      reinterpret_cast<uint32_t*>(dst)[1] = (uint32_t)(v >> 32);
+      break;
+    }
+    case 16: {
+      uint64_t lo = *reinterpret_cast<const uint64_t*>(src);
+      uint64_t hi = *reinterpret_cast<const uint64_t*>(src + 8);
+      if (!valid) { lo = 0; hi = 0; }
+      reinterpret_cast<uint32_t*>(dst)[0] = (uint32_t)lo;
+      reinterpret_cast<uint32_t*>(dst)[1] = (uint32_t)(lo >> 32);
+      reinterpret_cast<uint32_t*>(dst)[2] = (uint32_t)hi;
+      reinterpret_cast<uint32_t*>(dst)[3] = (uint32_t)(hi >> 32);
+      break;
+    }
+  }
+}
+
+__device__ inline void copy_elem4(uint8_t* dst, const uint8_t* src, int w) {
+  switch (w) {
+    case 1: *dst = *src; break;
+    case 2: *reinterpret_cast<uint16_t*>(dst) =
+                *reinterpret_cast<const uint16_t*>(src);
+            break;
+    case 4: *reinterpret_cast<uint32_t*>(dst) =
+                *reinterpret_cast<const uint32_t*>(src);
+            break;
+    case 8: {
+      uint64_t v = ((uint64_t)reinterpret_cast<const uint32_t*>(src)[1]
+                    << 32) | reinterpret_cast<const uint32_t*>(src)[0];
+      *reinterpret_cast<uint64_t*>(dst) = v;
+      break;
+    }
+    case 16: {
+      const uint32_t* s = reinterpret_cast<const uint32_t*>(src);
+      uint64_t lo = ((uint64_t)s[1] << 32) | s[0];
+      uint64_t hi = ((uint64_t)s[3] << 32) | s[2];
+      *reinterpret_cast<uint64_t*>(dst) = lo;
+      *reinterpret_cast<uint64_t*>(dst + 8) = hi;
+      break;
+    }
+  }
+}
+
+__device__ inline int32_t lds_pitch(int32_t row_size) {
+  // odd dword count -> stride coprime with the 32 LDS banks
+  return ((row_size >> 2) & 1) ? row_size : row_size + 4;
+}
+
 __global__ void to_rows_kernel(const RowColDesc* __restrict__ cols, int32_t ncols,
                                int64_t nrows, int32_t row_size,
                                int32_t validity_off, uint8_t* __restrict__ out) {
@@ -138,14 +203,16 @@ __global__ void to_rows_tiled_kernel(const RowColDesc* __restrict__ cols,
                                      uint8_t* __restrict__ out) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint8_t* tile = reinterpret_cast<uint8_t*>(smem);
+  int32_t const pitch = lds_pitch(row_size);
+  int32_t const rs8 = row_size >> 3;
   int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
   for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
     int64_t row0 = t * TILE_ROWS;
     int32_t m = (int32_t)(nrows - row0 < TILE_ROWS ? nrows - row0 : TILE_ROWS);
     // zero the tile (padding + null slots are zero by contract)
-    for (int32_t i = threadIdx.x * 16; i < TILE_ROWS * row_size;
-         i += blockDim.x * 16)
-      *reinterpret_cast<uint4*>(tile + i) = uint4{0, 0, 0, 0};
+    for (int32_t i = threadIdx.x * 4; i < TILE_ROWS * pitch;
+         i += blockDim.x * 4)
+      *reinterpret_cast<uint32_t*>(tile + i) = 0;
     __syncthreads();
     // data: flat (column, tile-row) work — each wave covers one column's 64
     // consecutive rows, so the global column loads are fully coalesced and
@@ -156,15 +223,15 @@ __global__ void to_rows_tiled_kernel(const RowColDesc* __restrict__ cols,
       if (r >= m) continue;
       int64_t row = row0 + r;
       const RowColDesc& d = cols[c];
-      copy_elem_z(tile + (int64_t)r * row_size + d.row_off,
-                  reinterpret_cast<const uint8_t*>(d.data) + row * d.width,
-                  d.width, is_valid(d.valid, row));
+      copy_elem_z4(tile + (int64_t)r * pitch + d.row_off,
+                   reinterpret_cast<const uint8_t*>(d.data) + row * d.width,
+                   d.width, is_valid(d.valid, row));
     }
     // validity bytes: one thread per tile row (no cross-thread byte RMW)
     for (int32_t r = threadIdx.x; r < TILE_ROWS; r += blockDim.x) {
       if (r >= m) continue;
       int64_t row = row0 + r;
-      uint8_t* dst = tile + (int64_t)r * row_size;
+      uint8_t* dst = tile + (int64_t)r * pitch;
       uint8_t vbyte = 0;
       for (int32_t c = 0; c < ncols; ++c) {
         vbyte |= (uint8_t)is_valid(cols[c].valid, row) << (c & 7);
@@ -176,64 +243,80 @@ __global__ void to_rows_tiled_kernel(const RowColDesc* __restrict__ cols,
       if (ncols & 7) dst[validity_off + (ncols >> 3)] = vbyte;
     }
     __syncthreads();
-    // coalesced wide copy LDS -> global
+    // coalesced copy LDS -> global in 8B grains (global u64 stores stay
+    // wide; the pitched LDS side reads as 2x u32)
     uint8_t* gdst = out + row0 * row_size;
-    int32_t nbytes = m * row_size;
-    int32_t i = threadIdx.x * 16;
-    for (; i + 16 <= nbytes; i += blockDim.x * 16)
-      *reinterpret_cast<uint4*>(gdst + i) = *reinterpret_cast<uint4*>(tile + i);
-    // tail (row_size is a multiple of 8, so 8B granularity suffices)
-    if (threadIdx.x == 0)
-      for (int32_t j = nbytes & ~15; j < nbytes; j += 8)
-        *reinterpret_cast<uint64_t*>(gdst + j) =
-            *reinterpret_cast<uint64_t*>(tile + j);
+    int32_t ngrains = m * rs8;
+    for (int32_t g = threadIdx.x; g < ngrains; g += blockDim.x) {
+      int32_t r = g / rs8;
+      int32_t o = (g - r * rs8) << 3;
+      const uint32_t* lp =
+          reinterpret_cast<const uint32_t*>(tile + (int64_t)r * pitch + o);
+      uint64_t v = ((uint64_t)lp[1] << 32) | lp[0];
+      *reinterpret_cast<uint64_t*>(gdst + (int64_t)r * row_size + o) = v;
+    }
     __syncthreads();
   }
 }
 
+// TR (tile rows) is a template parameter: bigger tiles turn the column
+// write side into longer contiguous bursts (24 column streams x 512B at
+// TR=64 round-robin poorly through HBM write combining; TR=256 makes
+// them 2KB) at the cost of LDS per block.
+template <int TR>
 __global__ void from_rows_tiled_kernel(const RowColDesc* __restrict__ cols,
                                        int32_t ncols, int64_t nrows,
                                        int32_t row_size, int32_t validity_off,
                                        const uint8_t* __restrict__ in) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   uint8_t* tile = reinterpret_cast<uint8_t*>(smem);
-  int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
+  int64_t ntiles = (nrows + TR - 1) / TR;
   for (int64_t t = blockIdx.x; t < ntiles; t += gridDim.x) {
-    int64_t row0 = t * TILE_ROWS;
-    int32_t m = (int32_t)(nrows - row0 < TILE_ROWS ? nrows - row0 : TILE_ROWS);
+    int64_t row0 = t * TR;
+    int32_t m = (int32_t)(nrows - row0 < TR ? nrows - row0 : TR);
     const uint8_t* gsrc = in + row0 * row_size;
-    int32_t nbytes = m * row_size;
-    for (int32_t i = threadIdx.x * 16; i + 16 <= nbytes; i += blockDim.x * 16)
-      *reinterpret_cast<uint4*>(tile + i) =
-          *reinterpret_cast<const uint4*>(gsrc + i);
-    if (threadIdx.x == 0)
-      for (int32_t j = nbytes & ~15; j < nbytes; j += 8)
-        *reinterpret_cast<uint64_t*>(tile + j) =
-            *reinterpret_cast<const uint64_t*>(gsrc + j);
+    int32_t const pitch = lds_pitch(row_size);
+    int32_t const rs8 = row_size >> 3;
+    // coalesced copy global -> LDS in 8B grains (pitched LDS writes as
+    // 2x u32)
+    int32_t ngrains = m * rs8;
+    for (int32_t g = threadIdx.x; g < ngrains; g += blockDim.x) {
+      int32_t r = g / rs8;
+      int32_t o = (g - r * rs8) << 3;
+      uint64_t v =
+          *reinterpret_cast<const uint64_t*>(gsrc + (int64_t)r * row_size + o);
+      uint32_t* lp =
+          reinterpret_cast<uint32_t*>(tile + (int64_t)r * pitch + o);
+      lp[0] = (uint32_t)v;
+      lp[1] = (uint32_t)(v >> 32);
+    }
     __syncthreads();
     // data: flat (column, tile-row) work — coalesced global column writes
-    for (int32_t w = threadIdx.x; w < ncols * TILE_ROWS; w += blockDim.x) {
-      int32_t c = w >> 6;
-      int32_t r = w & (TILE_ROWS - 1);
+    for (int32_t w = threadIdx.x; w < ncols * TR; w += blockDim.x) {
+      int32_t c = w / TR;
+      int32_t r = w % TR;
       if (r >= m) continue;
       const RowColDesc& d = cols[c];
-      copy_elem(const_cast<uint8_t*>(
-                    reinterpret_cast<const uint8_t*>(d.data)) +
-                    (row0 + r) * d.width,
-                tile + (int64_t)r * row_size + d.row_off, d.width);
+      copy_elem4(const_cast<uint8_t*>(
+                     reinterpret_cast<const uint8_t*>(d.data)) +
+                     (row0 + r) * d.width,
+                 tile + (int64_t)r * pitch + d.row_off, d.width);
     }
-    // validity: columns round-robin over ALL waves (TILE_ROWS == WAVE,
-    // rows map 1:1 onto lanes, so each wave's ballot covers the tile) —
-    // a single wave doing every column serialized ~ncols ballots while
-    // the rest of the block idled (measured 2x gap vs to-rows)
+    // validity: (column, 64-row subtile) pairs round-robin over all waves
+    // (each wave's 64 lanes map 1:1 onto the subtile's rows, so one
+    // ballot covers one validity word)
     {
       int32_t wave = threadIdx.x / WAVE;
       int32_t nwaves = blockDim.x / WAVE;
-      int32_t r = threadIdx.x & (WAVE - 1);
-      bool in_range = r < m;
-      const uint8_t* src = tile + (int64_t)(in_range ? r : 0) * row_size;
-      int64_t row = row0 + r;
-      for (int32_t c = wave; c < ncols; c += nwaves) {
+      int32_t lane = threadIdx.x & (WAVE - 1);
+      int32_t nsub = TR / WAVE;
+      for (int32_t cs = wave; cs < ncols * nsub; cs += nwaves) {
+        int32_t c = cs / nsub;
+        int32_t sub = cs % nsub;
+        int32_t r = sub * WAVE + lane;
+        bool in_range = r < m;
+        const uint8_t* src = tile + (int64_t)(in_range ? r : 0) * pitch;
+        int64_t row = row0 + r;
         const RowColDesc& d = cols[c];
         bool valid =
             in_range && ((src[validity_off + (c >> 3)] >> (c & 7)) & 1);
@@ -370,7 +453,8 @@ extern "C" {
 
 void srj_to_rows(const void* cols, int32_t ncols, int64_t nrows, int32_t row_size,
                  int32_t validity_off, uint8_t* out, hipStream_t stream) {
-  size_t lds = (size_t)TILE_ROWS * row_size;
+  int32_t pitch = ((row_size >> 2) & 1) ? row_size : row_size + 4;
+  size_t lds = (size_t)TILE_ROWS * pitch;
   if (lds <= 64 * 1024) {
     int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
     int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
@@ -384,16 +468,46 @@ void srj_to_rows(const void* cols, int32_t ncols, int64_t nrows, int32_t row_siz
       validity_off, out);
 }
 
+template <int TR>
+static void launch_from_rows_tiled(const void* cols, int32_t ncols,
+                                   int64_t nrows, int32_t row_size,
+                                   int32_t validity_off, const uint8_t* in,
+                                   hipStream_t stream) {
+  int32_t pitch = ((row_size >> 2) & 1) ? row_size : row_size + 4;
+  size_t lds = (size_t)TR * pitch;
+  int64_t ntiles = (nrows + TR - 1) / TR;
+  int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
+  from_rows_tiled_kernel<TR><<<nblk, DEFAULT_BLOCK, lds, stream>>>(
+      reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
+      validity_off, in);
+}
+
 void srj_from_rows(const void* cols, int32_t ncols, int64_t nrows,
                    int32_t row_size, int32_t validity_off, const uint8_t* in,
                    hipStream_t stream) {
-  size_t lds = (size_t)TILE_ROWS * row_size;
-  if (lds <= 64 * 1024) {
-    int64_t ntiles = (nrows + TILE_ROWS - 1) / TILE_ROWS;
-    int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
-    from_rows_tiled_kernel<<<nblk, DEFAULT_BLOCK, lds, stream>>>(
-        reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
-        validity_off, in);
+  // biggest tile whose LDS fits 64 KB: larger tiles give longer
+  // per-column write bursts (SRJ_FROM_ROWS_TILE overrides for tuning)
+  int tr = 0;
+  if (const char* env = getenv("SRJ_FROM_ROWS_TILE")) tr = atoi(env);
+  size_t l64 = 64 * 1024;
+  size_t pitch = ((row_size >> 2) & 1) ? row_size : row_size + 4;
+  if (tr == 0)
+    tr = (256 * pitch <= l64) ? 256
+         : (128 * pitch <= l64) ? 128
+         : (64 * pitch <= l64) ? 64 : 0;
+  if (tr == 256 && 256 * pitch <= l64) {
+    launch_from_rows_tiled<256>(cols, ncols, nrows, row_size, validity_off,
+                                in, stream);
+    return;
+  }
+  if (tr == 128 && 128 * pitch <= l64) {
+    launch_from_rows_tiled<128>(cols, ncols, nrows, row_size, validity_off,
+                                in, stream);
+    return;
+  }
+  if (tr == 64 && 64 * pitch <= l64) {
+    launch_from_rows_tiled<64>(cols, ncols, nrows, row_size, validity_off,
+                               in, stream);
     return;
   }
   from_rows_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
