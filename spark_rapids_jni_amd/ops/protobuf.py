@@ -2,10 +2,11 @@
 ProtobufSchemaDescriptor.java flattened field tables).
 
 Schema: list of (field_number, kind) where kind in {"int64","int32","bool",
-"sint64","double","float","string","bytes","repeated_int64"} or
-("message", child_schema). Nested messages decode recursively to STRUCT
-columns; repeated_int64 (packed or unpacked varints) decodes to
-LIST<INT64>.
+"sint64","double","float","string","bytes"} or "repeated_<k>" for k in
+{int64,int32,bool,sint64,double,float,string,bytes} (packed or unpacked
+encodings -> LIST columns; reference protobuf_kernels.cuh:150-361 batched
+variants) or ("message", child_schema) — nested messages decode
+recursively to STRUCT columns.
 """
 import struct
 from typing import List, Tuple
@@ -24,7 +25,18 @@ _KINDS = {"int64": (0, DType.INT64, torch.int64),
           "string": (6, DType.STRING, None),
           "bytes": (6, DType.STRING, None)}
 
-_FIELD_FMT = "<iiiiQQQQQ"  # fnum, kind, rep_slot, pad, data, valid, lens, offsets, chars
+# fnum, kind, rep_slot, pad, data, valid, lens, offsets, chars,
+# lens2, elem_offsets, char_base
+_FIELD_FMT = "<iiiiQQQQQQQQ"
+
+_REP_KINDS = {"repeated_int64": (7, DType.INT64, torch.int64),
+              "repeated_int32": (8, DType.INT32, torch.int32),
+              "repeated_bool": (9, DType.BOOL8, torch.int8),
+              "repeated_sint64": (10, DType.INT64, torch.int64),
+              "repeated_double": (11, DType.FLOAT64, torch.float64),
+              "repeated_float": (12, DType.FLOAT32, torch.float32),
+              "repeated_string": (13, DType.STRING, None),
+              "repeated_bytes": (13, DType.STRING, None)}
 
 
 def decode(col: Column, schema) -> Table:
@@ -61,12 +73,16 @@ def decode(col: Column, schema) -> Table:
     outs = []
     nrep = 0
     for fnum, kindname in schema:
-        if kindname == "repeated_int64":
+        if kindname in _REP_KINDS:
             assert nrep < 8, "at most 8 repeated fields per message"
-            outs.append({"kind": 7, "dtype": DType.LIST, "fnum": fnum,
-                         "rep_slot": nrep,
-                         "lens": torch.zeros(n, dtype=torch.int32, device=dev),
-                         "valid": make_validity(n, dev, fill_valid=False)})
+            kind, edt, etdt = _REP_KINDS[kindname]
+            o = {"kind": kind, "dtype": DType.LIST, "fnum": fnum,
+                 "rep_slot": nrep, "elem_dtype": edt, "elem_tdt": etdt,
+                 "lens": torch.zeros(n, dtype=torch.int32, device=dev),
+                 "valid": make_validity(n, dev, fill_valid=False)}
+            if kind == 13:
+                o["lens2"] = torch.zeros(n, dtype=torch.int32, device=dev)
+            outs.append(o)
             nrep += 1
             continue
         kind, dt, tdt = _KINDS[kindname]
@@ -88,7 +104,10 @@ def decode(col: Column, schema) -> Table:
                 o["valid"].data_ptr(),
                 o["lens"].data_ptr() if "lens" in o and phase == 0 else 0,
                 o["offsets"].data_ptr() if "offsets" in o else 0,
-                o["chars"].data_ptr() if "chars" in o else 0)
+                o["chars"].data_ptr() if "chars" in o else 0,
+                o["lens2"].data_ptr() if "lens2" in o and phase == 0 else 0,
+                o["elem_offsets"].data_ptr() if "elem_offsets" in o else 0,
+                o["char_base"].data_ptr() if "char_base" in o else 0)
         return torch.frombuffer(raw, dtype=torch.uint8).to(dev)
 
     row_ok = torch.zeros(n, dtype=torch.uint8, device=dev)
@@ -104,8 +123,20 @@ def decode(col: Column, schema) -> Table:
             torch.cumsum(o["lens"], 0, out=offsets[1:].view(n))
             o["offsets"] = offsets
             nch = int(offsets[-1].item())
-            if o["kind"] == 7:
-                o["data"] = torch.empty(max(nch, 1), dtype=torch.int64,
+            if o["kind"] == 13:
+                # LIST<STRING>: list offsets + per-element char offsets
+                char_base = torch.zeros(n + 1, dtype=torch.int32, device=dev)
+                torch.cumsum(o["lens2"], 0, out=char_base[1:].view(n))
+                o["char_base"] = char_base
+                tot_chars = int(char_base[-1].item())
+                o["tot_chars"] = tot_chars
+                o["elem_offsets"] = torch.zeros(max(nch, 1) + 1,
+                                                dtype=torch.int32,
+                                                device=dev)
+                o["chars"] = torch.empty(max(tot_chars, 1),
+                                         dtype=torch.uint8, device=dev)
+            elif 7 <= o["kind"] <= 12:
+                o["data"] = torch.empty(max(nch, 1), dtype=o["elem_tdt"],
                                         device=dev)
             else:
                 o["chars"] = torch.empty(max(nch, 1), dtype=torch.uint8,
@@ -116,9 +147,17 @@ def decode(col: Column, schema) -> Table:
                     0, 1, stream)
     cols = []
     for o in outs:
-        if o["kind"] == 7:
+        if o["kind"] == 13:
+            nel = int(o["offsets"][-1].item())
+            o["elem_offsets"][nel] = o["tot_chars"]
+            child = Column(DType.STRING, nel, o["chars"][:max(
+                o["tot_chars"], 1)], None, o["elem_offsets"][:nel + 1],
+                null_count=0)
+            cols.append(Column(DType.LIST, n, None, o["valid"],
+                               o["offsets"], [child], null_count=None))
+        elif 7 <= o["kind"] <= 12:
             nch = int(o["offsets"][-1].item())
-            child = Column(DType.INT64, nch, o["data"][:max(nch, 1)])
+            child = Column(o["elem_dtype"], nch, o["data"][:max(nch, 1)])
             cols.append(Column(DType.LIST, n, None, o["valid"],
                                o["offsets"], [child], null_count=None))
         elif "offsets" in o:

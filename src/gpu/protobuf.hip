@@ -21,10 +21,22 @@ enum PbKind : int32_t {
   PB_FIXED64_DOUBLE = 4,
   PB_FIXED32_FLOAT = 5,
   PB_BYTES = 6,          // string/bytes/submessage blob
-  PB_REP_I64 = 7,        // repeated int64 varint (packed or unpacked) -> LIST
+  // repeated fields -> LIST columns (packed or unpacked encodings;
+  // reference protobuf_kernels.cuh:150-361 batched variants)
+  PB_REP_I64 = 7,
+  PB_REP_I32 = 8,
+  PB_REP_BOOL = 9,
+  PB_REP_SINT64 = 10,
+  PB_REP_DOUBLE = 11,
+  PB_REP_FLOAT = 12,
+  PB_REP_BYTES = 13,     // repeated string/bytes -> LIST<STRING>
 };
 
 constexpr int PB_MAX_REPEATED = 8;
+
+__device__ inline bool pb_is_rep(int32_t k) {
+  return k >= PB_REP_I64 && k <= PB_REP_BYTES;
+}
 
 struct PbField {
   int32_t field_number;
@@ -34,9 +46,32 @@ struct PbField {
   void* data;        // output fixed-width data (null for BYTES in phase 0)
   uint8_t* valid;    // output validity
   int32_t* lens;     // BYTES/REP phase 0: per-row length/count
-  const int32_t* offsets;  // BYTES/REP phase 1: column offsets
-  char* chars;       // BYTES phase 1
+  const int32_t* offsets;  // BYTES/REP phase 1: column/list offsets
+  char* chars;       // BYTES/REP_BYTES phase 1: char output
+  int32_t* lens2;    // REP_BYTES phase 0: per-row char total
+  int32_t* elem_offsets;   // REP_BYTES phase 1: per-element char offsets
+  const int32_t* char_base;  // REP_BYTES phase 1: per-row char base
 };
+
+// append one repeated numeric element (phase-1 store by kind)
+__device__ inline void pb_store_rep(const PbField& f, int64_t at,
+                                    uint64_t v, int32_t kind) {
+  switch (kind) {
+    case PB_REP_I64:
+      reinterpret_cast<int64_t*>(f.data)[at] = (int64_t)v;
+      break;
+    case PB_REP_I32:
+      reinterpret_cast<int32_t*>(f.data)[at] = (int32_t)v;
+      break;
+    case PB_REP_BOOL:
+      reinterpret_cast<int8_t*>(f.data)[at] = v != 0;
+      break;
+    case PB_REP_SINT64:
+      reinterpret_cast<int64_t*>(f.data)[at] =
+          (int64_t)(v >> 1) ^ -(int64_t)(v & 1);
+      break;
+  }
+}
 
 __device__ inline bool pb_varint(const uint8_t* p, int64_t len, int64_t* pos,
                                  uint64_t* out) {
@@ -66,9 +101,10 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
     bool valid = in_range && is_valid(in.valid, row);
     // per-field presence this row (up to 64 fields via bitmask)
     uint64_t present = 0;
-    int32_t rcnt[PB_MAX_REPEATED];
+    int32_t rcnt[PB_MAX_REPEATED];   // element counts per repeated slot
+    int32_t ccnt[PB_MAX_REPEATED];   // char counts (REP_BYTES)
 #pragma unroll
-    for (int k = 0; k < PB_MAX_REPEATED; ++k) rcnt[k] = 0;
+    for (int k = 0; k < PB_MAX_REPEATED; ++k) { rcnt[k] = 0; ccnt[k] = 0; }
     bool ok = valid;
     if (valid) {
       StrView s = get_string(in, row);
@@ -91,11 +127,13 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
               PbField& f = fields[fi];
               present |= 1ull << fi;
               switch (f.kind) {
-                case PB_REP_I64: {
+                case PB_REP_I64:
+                case PB_REP_I32:
+                case PB_REP_BOOL:
+                case PB_REP_SINT64: {
                   int32_t c0 = rcnt[f.rep_slot]++;
                   if (WRITE_BYTES)
-                    reinterpret_cast<int64_t*>(f.data)[f.offsets[row] + c0] =
-                        (int64_t)v;
+                    pb_store_rep(f, (int64_t)f.offsets[row] + c0, v, f.kind);
                   break;
                 }
                 case PB_VARINT_I64:
@@ -124,6 +162,15 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
               __builtin_memcpy(&d, p + pos, 8);
               reinterpret_cast<double*>(fields[fi].data)[row] = d;
               present |= 1ull << fi;
+            } else if (fi >= 0 && fields[fi].kind == PB_REP_DOUBLE) {
+              PbField& f = fields[fi];
+              present |= 1ull << fi;
+              int32_t c0 = rcnt[f.rep_slot]++;
+              if (WRITE_BYTES) {
+                double d;
+                __builtin_memcpy(&d, p + pos, 8);
+                reinterpret_cast<double*>(f.data)[f.offsets[row] + c0] = d;
+              }
             }
             pos += 8;
             break;
@@ -135,6 +182,15 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
               __builtin_memcpy(&d, p + pos, 4);
               reinterpret_cast<float*>(fields[fi].data)[row] = d;
               present |= 1ull << fi;
+            } else if (fi >= 0 && fields[fi].kind == PB_REP_FLOAT) {
+              PbField& f = fields[fi];
+              present |= 1ull << fi;
+              int32_t c0 = rcnt[f.rep_slot]++;
+              if (WRITE_BYTES) {
+                float d;
+                __builtin_memcpy(&d, p + pos, 4);
+                reinterpret_cast<float*>(f.data)[f.offsets[row] + c0] = d;
+              }
             }
             pos += 4;
             break;
@@ -155,7 +211,8 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
               } else {
                 f.lens[row] = (int32_t)blen;
               }
-            } else if (fi >= 0 && fields[fi].kind == PB_REP_I64) {
+            } else if (fi >= 0 && fields[fi].kind >= PB_REP_I64 &&
+                       fields[fi].kind <= PB_REP_SINT64) {
               // packed repeated varints
               PbField& f = fields[fi];
               present |= 1ull << fi;
@@ -165,9 +222,44 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
                 if (!pb_varint(p, bend, &bpos, &v)) { ok = false; break; }
                 int32_t c0 = rcnt[f.rep_slot]++;
                 if (WRITE_BYTES)
-                  reinterpret_cast<int64_t*>(f.data)[f.offsets[row] + c0] =
-                      (int64_t)v;
+                  pb_store_rep(f, (int64_t)f.offsets[row] + c0, v, f.kind);
               }
+            } else if (fi >= 0 && fields[fi].kind == PB_REP_DOUBLE) {
+              // packed fixed64 elements
+              PbField& f = fields[fi];
+              present |= 1ull << fi;
+              for (uint64_t b = 0; b + 8 <= blen; b += 8) {
+                int32_t c0 = rcnt[f.rep_slot]++;
+                if (WRITE_BYTES) {
+                  double d;
+                  __builtin_memcpy(&d, p + pos + b, 8);
+                  reinterpret_cast<double*>(f.data)[f.offsets[row] + c0] = d;
+                }
+              }
+            } else if (fi >= 0 && fields[fi].kind == PB_REP_FLOAT) {
+              // packed fixed32 elements
+              PbField& f = fields[fi];
+              present |= 1ull << fi;
+              for (uint64_t b = 0; b + 4 <= blen; b += 4) {
+                int32_t c0 = rcnt[f.rep_slot]++;
+                if (WRITE_BYTES) {
+                  float d;
+                  __builtin_memcpy(&d, p + pos + b, 4);
+                  reinterpret_cast<float*>(f.data)[f.offsets[row] + c0] = d;
+                }
+              }
+            } else if (fi >= 0 && fields[fi].kind == PB_REP_BYTES) {
+              // one element per length-delimited occurrence
+              PbField& f = fields[fi];
+              present |= 1ull << fi;
+              int32_t c0 = rcnt[f.rep_slot]++;
+              if (WRITE_BYTES) {
+                int32_t off = f.char_base[row] + ccnt[f.rep_slot];
+                for (uint64_t k = 0; k < blen; ++k)
+                  f.chars[off + k] = (char)p[pos + k];
+                f.elem_offsets[f.offsets[row] + c0] = off;
+              }
+              ccnt[f.rep_slot] += (int32_t)blen;
             }
             pos += blen;
             break;
@@ -184,8 +276,11 @@ __global__ void pb_decode_kernel(ColDesc in, int64_t nrows,
       ballot_write_validity(fields[k].valid, row, fv);
       if (in_range && !fv && fields[k].kind == PB_BYTES && !WRITE_BYTES)
         fields[k].lens[row] = 0;
-      if (in_range && fields[k].kind == PB_REP_I64 && !WRITE_BYTES)
+      if (in_range && pb_is_rep(fields[k].kind) && !WRITE_BYTES) {
         fields[k].lens[row] = fv ? rcnt[fields[k].rep_slot] : 0;
+        if (fields[k].kind == PB_REP_BYTES)
+          fields[k].lens2[row] = fv ? ccnt[fields[k].rep_slot] : 0;
+      }
     }
   }
 }

@@ -445,6 +445,20 @@ __global__ void from_rows_var_kernel(const RowColDesc* __restrict__ cols,
   }
 }
 
+template <int TR>
+void launch_from_rows_tiled(const void* cols, int32_t ncols,
+                                   int64_t nrows, int32_t row_size,
+                                   int32_t validity_off, const uint8_t* in,
+                                   hipStream_t stream) {
+  int32_t pitch = ((row_size >> 2) & 1) ? row_size : row_size + 4;
+  size_t lds = (size_t)TR * pitch;
+  int64_t ntiles = (nrows + TR - 1) / TR;
+  int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
+  from_rows_tiled_kernel<TR><<<nblk, DEFAULT_BLOCK, lds, stream>>>(
+      reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
+      validity_off, in);
+}
+
 }  // namespace srj
 
 using namespace srj;
@@ -468,19 +482,6 @@ void srj_to_rows(const void* cols, int32_t ncols, int64_t nrows, int32_t row_siz
       validity_off, out);
 }
 
-template <int TR>
-static void launch_from_rows_tiled(const void* cols, int32_t ncols,
-                                   int64_t nrows, int32_t row_size,
-                                   int32_t validity_off, const uint8_t* in,
-                                   hipStream_t stream) {
-  int32_t pitch = ((row_size >> 2) & 1) ? row_size : row_size + 4;
-  size_t lds = (size_t)TR * pitch;
-  int64_t ntiles = (nrows + TR - 1) / TR;
-  int64_t nblk = ntiles < MAX_GRID ? ntiles : MAX_GRID;
-  from_rows_tiled_kernel<TR><<<nblk, DEFAULT_BLOCK, lds, stream>>>(
-      reinterpret_cast<const RowColDesc*>(cols), ncols, nrows, row_size,
-      validity_off, in);
-}
 
 void srj_from_rows(const void* cols, int32_t ncols, int64_t nrows,
                    int32_t row_size, int32_t validity_off, const uint8_t* in,
@@ -496,17 +497,17 @@ void srj_from_rows(const void* cols, int32_t ncols, int64_t nrows,
          : (128 * pitch <= l64) ? 128
          : (64 * pitch <= l64) ? 64 : 0;
   if (tr == 256 && 256 * pitch <= l64) {
-    launch_from_rows_tiled<256>(cols, ncols, nrows, row_size, validity_off,
+    srj::launch_from_rows_tiled<256>(cols, ncols, nrows, row_size, validity_off,
                                 in, stream);
     return;
   }
   if (tr == 128 && 128 * pitch <= l64) {
-    launch_from_rows_tiled<128>(cols, ncols, nrows, row_size, validity_off,
+    srj::launch_from_rows_tiled<128>(cols, ncols, nrows, row_size, validity_off,
                                 in, stream);
     return;
   }
   if (tr == 64 && 64 * pitch <= l64) {
-    launch_from_rows_tiled<64>(cols, ncols, nrows, row_size, validity_off,
+    srj::launch_from_rows_tiled<64>(cols, ncols, nrows, row_size, validity_off,
                                in, stream);
     return;
   }

@@ -149,3 +149,78 @@ def test_protobuf_repeated_int64():
             assert got[i] == vals, i
         else:
             assert got[i] is None, i
+
+
+@pytest.mark.gpu
+def test_protobuf_repeated_all_types():
+    """Reference protobuf_kernels.cuh:150-361 batched variants: repeated
+    int32/bool/sint64/double/float decode to LISTs (packed and unpacked),
+    repeated string to LIST<STRING>."""
+    import struct as st
+    from spark_rapids_jni_amd.ops import protobuf as pb
+
+    def enc_row(i):
+        out = b""
+        # f1 repeated int32: unpacked for even rows, packed for odd
+        vals32 = [i, i * 7 + 1, -i][: (i % 4)]
+        if i % 2 == 0:
+            for v in vals32:
+                out += _tag(1, 0) + _varint(v & (2**64 - 1))
+        else:
+            blob = b"".join(_varint(v & (2**64 - 1)) for v in vals32)
+            if blob:
+                out += _tag(1, 2) + _varint(len(blob)) + blob
+        # f2 repeated bool (packed)
+        bools = [bool((i >> k) & 1) for k in range(i % 3)]
+        blob = b"".join(_varint(int(b)) for b in bools)
+        if blob:
+            out += _tag(2, 2) + _varint(len(blob)) + blob
+        # f3 repeated sint64 (unpacked, zigzag)
+        sints = [-(i * 3), i * 5][: (i % 3)]
+        for v in sints:
+            out += _tag(3, 0) + _varint(((v << 1) ^ (v >> 63)) & (2**64 - 1))
+        # f4 repeated double: packed even rows, unpacked odd
+        dbls = [i * 1.5, -i / 3.0][: (i % 3)]
+        if i % 2 == 0:
+            blob = b"".join(st.pack("<d", d) for d in dbls)
+            if blob:
+                out += _tag(4, 2) + _varint(len(blob)) + blob
+        else:
+            for d in dbls:
+                out += _tag(4, 1) + st.pack("<d", d)
+        # f5 repeated float (packed)
+        flts = [float(i), float(i) * 0.5, 9.25][: (i % 4)]
+        blob = b"".join(st.pack("<f", f) for f in flts)
+        if blob:
+            out += _tag(5, 2) + _varint(len(blob)) + blob
+        # f6 repeated string
+        strs = [f"s{i}", "", f"xx{i*3}"][: (i % 4)]
+        for s in strs:
+            b = s.encode()
+            out += _tag(6, 2) + _varint(len(b)) + b
+        return out, (vals32, bools, sints, dbls, flts, strs)
+
+    n = 500
+    rows = [enc_row(i) for i in range(n)]
+    col = Column.from_pylist([r[0] for r in rows], DType.STRING, "cuda")
+    tbl = pb.decode(col, [(1, "repeated_int32"), (2, "repeated_bool"),
+                          (3, "repeated_sint64"), (4, "repeated_double"),
+                          (5, "repeated_float"), (6, "repeated_string")])
+    got = [c.to_pylist() for c in tbl.columns]
+    for i in range(n):
+        vals32, bools, sints, dbls, flts, strs = rows[i][1]
+        assert (got[0][i] or []) == (vals32 if vals32 else []) or \
+            (got[0][i] is None and not vals32), (i, got[0][i], vals32)
+        if vals32:
+            assert got[0][i] == vals32, i
+        if bools:
+            assert got[1][i] == bools, i
+        if sints:
+            assert got[2][i] == sints, i
+        if dbls:
+            assert len(got[3][i]) == len(dbls) and all(
+                abs(a - b) < 1e-12 for a, b in zip(got[3][i], dbls)), i
+        if flts:
+            assert len(got[4][i]) == len(flts), i
+        if strs:
+            assert got[5][i] == strs, i
