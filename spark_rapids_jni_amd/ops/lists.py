@@ -169,9 +169,19 @@ def iceberg_bucket(col: Column, nbuckets: int) -> Column:
     out = torch.empty(n, dtype=torch.int32, device=dev)
     validity = make_validity(n, dev)
     if col.dtype == DType.STRING:
+        # strings AND binary: raw UTF-8/byte payload
         desc, top, keep = pack_descriptors([col])
         g.iceberg_bucket_string(desc.data_ptr(), n, nbuckets, out.data_ptr(),
                                 validity.data_ptr(), _native.current_stream())
+    elif col.dtype in (DType.DECIMAL32, DType.DECIMAL64, DType.DECIMAL128):
+        # minimal big-endian two's-complement unscaled bytes (iceberg spec)
+        width = {DType.DECIMAL32: 4, DType.DECIMAL64: 8,
+                 DType.DECIMAL128: 16}[col.dtype]
+        g.iceberg_bucket_decimal(
+            col.data.data_ptr(),
+            col.validity.data_ptr() if col.validity is not None else 0,
+            n, width, nbuckets, out.data_ptr(), validity.data_ptr(),
+            _native.current_stream())
     else:
         data = col.data
         if col.dtype in (DType.INT32, DType.DATE32):

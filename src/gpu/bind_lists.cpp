@@ -16,6 +16,9 @@ void srj_map_zip(const int32_t*, const int32_t*, int64_t, const void*, int32_t,
                  int64_t*, hipStream_t);
 void srj_iceberg_bucket_long(const int64_t*, const uint8_t*, int64_t, int32_t,
                              int32_t*, uint8_t*, hipStream_t);
+void srj_iceberg_bucket_decimal(const void*, const uint8_t*, int64_t,
+                                int32_t, int32_t, int32_t*, uint8_t*,
+                                hipStream_t);
 void srj_iceberg_bucket_string(const void*, int64_t, int32_t, int32_t*,
                                uint8_t*, hipStream_t);
 void srj_iceberg_truncate_long(const int64_t*, const uint8_t*, int64_t, int64_t,
@@ -71,6 +74,15 @@ void register_lists(py::module_& m) {
                 as_ptr<int64_t>(kmap), as_ptr<int64_t>(v1), as_ptr<int64_t>(v2),
                 as_stream(stream));
     check_hip("map_zip");
+  });
+  m.def("iceberg_bucket_decimal", [](uintptr_t in, uintptr_t valid,
+                                     int64_t n, int32_t width, int32_t nb,
+                                     uintptr_t out, uintptr_t ov,
+                                     uintptr_t stream) {
+    srj_iceberg_bucket_decimal(as_ptr<void>(in), as_ptr<uint8_t>(valid), n,
+                               width, nb, as_ptr<int32_t>(out),
+                               as_ptr<uint8_t>(ov), as_stream(stream));
+    check_hip("iceberg_bucket_decimal");
   });
   m.def("iceberg_bucket_long", [](uintptr_t in, uintptr_t valid, int64_t n,
                                   int32_t nb, uintptr_t out, uintptr_t ov,

@@ -16,20 +16,6 @@
 
 namespace srj {
 
-// Minimal big-endian two's-complement byte form of a 128-bit little-endian
-// value, matching java.math.BigDecimal.unscaledValue().toByteArray(): strip
-// redundant sign-extension bytes, keep one extra byte when needed to preserve
-// the sign bit, reverse to big-endian. Spark hashes DECIMAL128 (precision>18)
-// over exactly these bytes (ref hash/hash.cuh:64 to_java_bigdecimal).
-__device__ int dec128_java_bytes(const uint8_t* p, uint8_t out[16]) {
-  bool neg = (p[15] & 0x80) != 0;
-  uint8_t ext = neg ? 0xff : 0x00;
-  int len = 16;
-  while (len > 1 && p[len - 1] == ext) --len;
-  if (len < 16 && (neg != ((p[len - 1] & 0x80) != 0))) ++len;
-  for (int i = 0; i < len; ++i) out[i] = p[len - 1 - i];
-  return len;
-}
 
 // ---------------------------------------------------------------------------
 // murmur3 (Spark Murmur3_x86_32)

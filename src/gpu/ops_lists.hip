@@ -222,6 +222,37 @@ __global__ void iceberg_bucket_string_kernel(ColDesc in, int64_t n,
   }
 }
 
+// decimal bucket: murmur3_x86_32 of the unscaled value's minimal
+// big-endian two's-complement bytes (iceberg spec appendix B; reference
+// iceberg_bucket.hpp decimal path)
+__global__ void iceberg_bucket_decimal_kernel(
+    const uint8_t* __restrict__ in, const uint8_t* __restrict__ valid,
+    int64_t n, int32_t width, int32_t nbuckets, int32_t* __restrict__ out,
+    uint8_t* __restrict__ out_valid) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t npad = (n + WAVE - 1) & ~(int64_t)(WAVE - 1);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < npad;
+       i += stride) {
+    bool in_range = i < n;
+    bool v = in_range && is_valid(valid, i);
+    int32_t b = 0;
+    if (v) {
+      // sign-extend the little-endian unscaled value to 16 bytes, then
+      // take the java-minimal big-endian form
+      uint8_t le[16];
+      const uint8_t* p = in + i * width;
+      uint8_t ext = (p[width - 1] & 0x80) ? 0xFF : 0x00;
+      for (int k = 0; k < 16; ++k) le[k] = k < width ? p[k] : ext;
+      uint8_t be[16];
+      int len = dec128_java_bytes(le, be);
+      uint32_t h = mm3_hash_bytes(reinterpret_cast<const char*>(be), len, 0);
+      b = (int32_t)((h & 0x7FFFFFFFu) % (uint32_t)nbuckets);
+    }
+    if (in_range) out[i] = b;
+    ballot_write_validity(out_valid, i, v);
+  }
+}
+
 __global__ void iceberg_truncate_long_kernel(const int64_t* __restrict__ in,
                                              const uint8_t* __restrict__ valid,
                                              int64_t n, int64_t width,
@@ -481,6 +512,15 @@ void srj_iceberg_bucket_string(const void* in, int64_t n, int32_t nbuckets,
                                hipStream_t stream) {
   iceberg_bucket_string_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(
       *reinterpret_cast<const ColDesc*>(in), n, nbuckets, out, out_valid);
+}
+
+void srj_iceberg_bucket_decimal(const void* in, const uint8_t* valid,
+                                int64_t n, int32_t width, int32_t nbuckets,
+                                int32_t* out, uint8_t* out_valid,
+                                hipStream_t stream) {
+  iceberg_bucket_decimal_kernel<<<grid_1d(n), DEFAULT_BLOCK, 0, stream>>>(
+      reinterpret_cast<const uint8_t*>(in), valid, n, width, nbuckets, out,
+      out_valid);
 }
 
 void srj_iceberg_truncate_long(const int64_t* in, const uint8_t* valid,

@@ -244,3 +244,35 @@ def test_q3_pipeline_vs_torch_oracle():
     assert set(got) == set(exp)
     for b in exp:
         assert abs(got[b] - exp[b]) < 1e-6 * max(1.0, abs(exp[b]))
+
+
+@pytest.mark.gpu
+def test_iceberg_bucket_decimal_and_binary():
+    """Iceberg spec Appendix B vectors: decimal buckets hash the minimal
+    big-endian two's-complement unscaled bytes; binary hashes raw bytes."""
+    from spark_rapids_jni_amd.ops.lists import iceberg_bucket
+    from spark_rapids_jni_amd.utils import sparkref as ref
+
+    # decimal 14.20 -> unscaled 1420 -> bytes 0x05 0x8C -> murmur3 seed 0
+    # = -500754589 (iceberg spec test value)
+    assert ref.murmur3_bytes(b"\x05\x8c", 0) & 0xFFFFFFFF == \
+        (-500754589) & 0xFFFFFFFF
+    for dt in (DType.DECIMAL32, DType.DECIMAL64, DType.DECIMAL128):
+        col = Column.from_pylist([1420, None, -7, 0, 2**30], dt,
+                                 device="cuda", scale=2)
+        got = iceberg_bucket(col, 16).to_pylist()
+        import struct as _s
+        for i, v in enumerate([1420, None, -7, 0, 2**30]):
+            if v is None:
+                assert got[i] is None
+                continue
+            exp_bytes = ref.java_bigint_bytes(v)
+            h = ref.murmur3_bytes(exp_bytes, 0)
+            assert got[i] == (h & 0x7FFFFFFF) % 16, (dt, i)
+
+    # binary payloads go through the string/bytes path
+    col = Column.from_pylist([b"\x00\x01\x02\x03", b"", None], DType.STRING,
+                             device="cuda")
+    got = iceberg_bucket(col, 128).to_pylist()
+    exp = (ref.murmur3_bytes(b"\x00\x01\x02\x03", 0) & 0x7FFFFFFF) % 128
+    assert got[0] == exp
