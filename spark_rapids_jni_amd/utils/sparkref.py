@@ -97,6 +97,42 @@ def _to_signed64(v):
     return v - (1 << 64) if v >= (1 << 63) else v
 
 
+def murmur3_bytes_std(b: bytes, seed):
+    """STANDARD MurmurHash3_x86_32 (proper 1-3 byte tail mixing) — what
+    Iceberg's bucket transform uses (the Spark variant above processes
+    tail bytes as full int blocks instead)."""
+    h = seed & M32
+    n = len(b)
+    aligned = n & ~3
+    for i in range(0, aligned, 4):
+        k = struct.unpack_from("<I", b, i)[0]
+        k = (k * 0xCC9E2D51) & M32
+        k = _rotl32(k, 15)
+        k = (k * 0x1B873593) & M32
+        h ^= k
+        h = _rotl32(h, 13)
+        h = (h * 5 + 0xE6546B64) & M32
+    k = 0
+    rem = n & 3
+    if rem >= 3:
+        k ^= b[aligned + 2] << 16
+    if rem >= 2:
+        k ^= b[aligned + 1] << 8
+    if rem >= 1:
+        k ^= b[aligned]
+        k = (k * 0xCC9E2D51) & M32
+        k = _rotl32(k, 15)
+        k = (k * 0x1B873593) & M32
+        h ^= k
+    h ^= n
+    h ^= h >> 16
+    h = (h * 0x85EBCA6B) & M32
+    h ^= h >> 13
+    h = (h * 0xC2B2AE35) & M32
+    h ^= h >> 16
+    return h
+
+
 def java_bigint_bytes(v: int) -> bytes:
     """java.math.BigInteger.toByteArray(): minimal big-endian two's
     complement, at least one byte. Spark hashes DECIMAL128 over these bytes

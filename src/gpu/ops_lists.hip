@@ -214,7 +214,8 @@ __global__ void iceberg_bucket_string_kernel(ColDesc in, int64_t n,
     int32_t r = 0;
     if (v) {
       StrView s = get_string(in, i);
-      int32_t h = (int32_t)mm3_hash_bytes(s.ptr, s.len, 0);
+      // standard murmur3 (iceberg spec), not the Spark tail variant
+      int32_t h = (int32_t)mm3_hash_bytes_std(s.ptr, s.len, 0);
       r = (h & 0x7FFFFFFF) % nbuckets;
     }
     if (in_range) out[i] = r;
@@ -245,7 +246,7 @@ __global__ void iceberg_bucket_decimal_kernel(
       for (int k = 0; k < 16; ++k) le[k] = k < width ? p[k] : ext;
       uint8_t be[16];
       int len = dec128_java_bytes(le, be);
-      uint32_t h = mm3_hash_bytes(reinterpret_cast<const char*>(be), len, 0);
+      uint32_t h = mm3_hash_bytes_std(reinterpret_cast<const char*>(be), len, 0);
       b = (int32_t)((h & 0x7FFFFFFFu) % (uint32_t)nbuckets);
     }
     if (in_range) out[i] = b;

@@ -214,6 +214,33 @@ __device__ __host__ inline uint32_t mm3_hash_bytes(const char* p, int32_t len,
   return mm3_fmix(h1, (uint32_t)len);
 }
 
+// STANDARD MurmurHash3_x86_32 (proper 1-3 byte tail) — Iceberg's bucket
+// transform requires this variant (the Spark mm3_hash_bytes above mixes
+// tail bytes as full int blocks; reference iceberg_bucket.cu delegates
+// to cuco::detail::MurmurHash3_32 which is the standard algorithm).
+__device__ __host__ inline uint32_t mm3_hash_bytes_std(const char* p,
+                                                       int32_t len,
+                                                       uint32_t seed) {
+  uint32_t h1 = seed;
+  int32_t aligned = len & ~3;
+  for (int32_t i = 0; i < aligned; i += 4) {
+    uint32_t b = (uint8_t)p[i] | ((uint8_t)p[i + 1] << 8) |
+                 ((uint8_t)p[i + 2] << 16) | ((uint8_t)p[i + 3] << 24);
+    h1 = mm3_mix_h1(h1, mm3_mix_k1(b));
+  }
+  uint32_t k1 = 0;
+  switch (len & 3) {
+    case 3: k1 ^= (uint32_t)(uint8_t)p[aligned + 2] << 16; [[fallthrough]];
+    case 2: k1 ^= (uint32_t)(uint8_t)p[aligned + 1] << 8; [[fallthrough]];
+    case 1:
+      k1 ^= (uint32_t)(uint8_t)p[aligned];
+      h1 ^= mm3_mix_k1(k1);
+      break;
+    default: break;
+  }
+  return mm3_fmix(h1, (uint32_t)len);
+}
+
 // normalization Spark applies before hashing floats (NaN -> canonical NaN,
 // -0.0 -> +0.0). Bit-based so it is immune to fast-math flags.
 __device__ __host__ inline int32_t norm_float_bits(float f) {

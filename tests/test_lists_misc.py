@@ -253,10 +253,13 @@ def test_iceberg_bucket_decimal_and_binary():
     from spark_rapids_jni_amd.ops.lists import iceberg_bucket
     from spark_rapids_jni_amd.utils import sparkref as ref
 
-    # decimal 14.20 -> unscaled 1420 -> bytes 0x05 0x8C -> murmur3 seed 0
-    # = -500754589 (iceberg spec test value)
-    assert ref.murmur3_bytes(b"\x05\x8c", 0) & 0xFFFFFFFF == \
+    # iceberg spec Appendix B vectors (STANDARD murmur3, not Spark's
+    # tail-as-int variant): decimal 14.20 -> bytes 0x05 0x8C ->
+    # -500754589; "iceberg" -> 1210000089
+    assert ref.murmur3_bytes_std(b"\x05\x8c", 0) & 0xFFFFFFFF == \
         (-500754589) & 0xFFFFFFFF
+    assert ref.murmur3_bytes_std(b"iceberg", 0) & 0xFFFFFFFF == \
+        1210000089
     for dt in (DType.DECIMAL32, DType.DECIMAL64, DType.DECIMAL128):
         col = Column.from_pylist([1420, None, -7, 0, 2**30], dt,
                                  device="cuda", scale=2)
@@ -267,12 +270,23 @@ def test_iceberg_bucket_decimal_and_binary():
                 assert got[i] is None
                 continue
             exp_bytes = ref.java_bigint_bytes(v)
-            h = ref.murmur3_bytes(exp_bytes, 0)
+            h = ref.murmur3_bytes_std(exp_bytes, 0)
             assert got[i] == (h & 0x7FFFFFFF) % 16, (dt, i)
+
+    # strings: spec vector "iceberg" -> 1210000089 through the kernel
+    scol = Column.from_pylist(["iceberg", "abc", None], DType.STRING,
+                              device="cuda")
+    sgot = iceberg_bucket(scol, 64).to_pylist()
+    assert sgot[0] == (1210000089 & 0x7FFFFFFF) % 64
+    assert sgot[1] == (ref.murmur3_bytes_std(b"abc", 0) & 0x7FFFFFFF) % 64
+    assert sgot[2] is None
 
     # binary payloads go through the string/bytes path
     col = Column.from_pylist([b"\x00\x01\x02\x03", b"", None], DType.STRING,
                              device="cuda")
     got = iceberg_bucket(col, 128).to_pylist()
-    exp = (ref.murmur3_bytes(b"\x00\x01\x02\x03", 0) & 0x7FFFFFFF) % 128
+    exp = (ref.murmur3_bytes_std(b"\x00\x01\x02\x03", 0) & 0x7FFFFFFF) % 128
     assert got[0] == exp
+    # spec: binary 0x00 01 02 03 hashes to -188683207
+    assert ref.murmur3_bytes_std(b"\x00\x01\x02\x03", 0) & 0xFFFFFFFF == \
+        (-188683207) & 0xFFFFFFFF
