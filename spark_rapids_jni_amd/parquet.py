@@ -173,9 +173,27 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                 "unrecognized LIST encoding"
             el = schema_elems[i + 2]
             if el.get(5, 0) != 0:
-                raise NotImplementedError(
-                    "LIST of nested types (list-of-list / list-of-struct) "
-                    "is not supported yet")
+                # nested element: LIST<LIST<...>> or LIST<STRUCT<...>>.
+                # Parse the element subtree with the list chain's def levels
+                # accumulated, so leaf max_def values are ABSOLUTE (the
+                # repeated group contributes +1).
+                if is_map_group(el):
+                    raise NotImplementedError(
+                        "LIST of MAP is not supported yet")
+                inner, j = parse_field(i + 2, d_above + outer_opt + 1)
+                if inner.is_list:
+                    return SchemaField(
+                        name=se[4].decode(), physical_type=-1,
+                        repetition=se.get(3, 0), converted_type=3,
+                        is_list=True, element=inner, max_def=inner.max_def,
+                        max_rep=inner.max_rep + 1), j
+                assert inner.is_struct
+                leaf_max = max(lf.max_def
+                               for lf in _flatten_struct_leaves(inner))
+                return SchemaField(
+                    name=se[4].decode(), physical_type=-1,
+                    repetition=se.get(3, 0), converted_type=3, is_list=True,
+                    element=inner, max_def=leaf_max, max_rep=1), j
             elem = leaf_field(el, 0)
             elem_opt = 1 if el.get(3, 0) == 1 else 0
             return SchemaField(
@@ -774,22 +792,45 @@ def _mmap_file(path: str) -> memoryview:
     return mv
 
 
+def _list_chain(f: SchemaField):
+    """Walk a (possibly nested) LIST chain: one (node_opt, slot_def) pair per
+    repetition level, outermost first, where slot_def is the ABSOLUTE
+    definition level at which an element slot of that level exists. Returns
+    (chain, deepest_non_list_element)."""
+    chain = []
+    d = 0
+    cur = f
+    while True:
+        opt = 1 if cur.repetition == 1 else 0
+        d += opt + 1
+        chain.append((opt, d))
+        if cur.element.is_list:
+            cur = cur.element
+        else:
+            return chain, cur.element
+
+
 def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
-                      total_rows: int, device) -> Column:
-    """LIST<primitive|string> decode (parquet 3-level encoding).
+                      total_rows: int, device, return_elem_def=False):
+    """LIST decode (parquet 3-level encoding), any repetition depth:
+    LIST<leaf>, LIST<LIST<...<leaf>>>, and (via _read_list_struct_column's
+    per-leaf synthetic fields) the leaves of LIST<STRUCT<...>>.
 
     Levels (repetition + definition) decode with the same RLE kernel as the
-    flat path; the row structure (list offsets, null/empty lists) is derived
-    from them with torch scans, and the VALUE decode reuses the flat-path
-    kernels verbatim by treating ELEMENT space as row space."""
+    flat path; the row structure (per-level list offsets, null/empty lists)
+    is derived from them with torch scans, and the VALUE decode reuses the
+    flat-path kernels verbatim by treating deepest-ELEMENT space as row
+    space. With return_elem_def the element-space absolute def levels are
+    returned too (struct assembly needs them for struct-level validity)."""
     g = _native.gpu()
     stream = _native.current_stream()
     dev = torch.device(device)
-    elem = f.element
+    chain, elem = _list_chain(f)
     if elem.physical_type == T_FIXED_LEN_BYTE_ARRAY:
         raise NotImplementedError("LIST of FLBA decimals not supported")
-    elem_nullable = elem.repetition == 1
     max_def = f.max_def
+    elem_nullable = max_def > chain[-1][1]
+    rep_bw = max(len(chain).bit_length(), 1)
     def_bw = max(max_def.bit_length(), 1)
 
     pages: List[_Page] = []
@@ -842,7 +883,7 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
             body_offs.append(p.rep_bytes + p.def_bytes)
         rle_descs += struct.pack(_RLE_FMT, rep_src, rep_len,
                                  rep_t.data_ptr() + int(pos_starts[i]),
-                                 p.num_values, 1, 0)
+                                 p.num_values, rep_bw, 0)
         rle_descs += struct.pack(_RLE_FMT, def_src, def_len,
                                  def_t.data_ptr() + int(pos_starts[i]),
                                  p.num_values, def_bw, 0)
@@ -856,14 +897,36 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
     row_mask = rep64 == 0
     nrows_here = int(row_mask.sum().item())
     assert nrows_here == total_rows, (nrows_here, total_rows)
-    row_id = torch.cumsum(row_mask.to(torch.int64), 0) - 1
-    elem_slot_def = max_def - (1 if elem_nullable else 0)
-    elem_mask = def64 >= elem_slot_def
-    lengths = torch.bincount(row_id[elem_mask], minlength=total_rows)
-    offsets = torch.zeros(total_rows + 1, dtype=torch.int32, device=dev)
-    if total_rows:
-        offsets[1:] = torch.cumsum(lengths, 0).to(torch.int32)
-    total_elems = int(offsets[-1].item())
+
+    # per repetition level, outer->inner: slots of level L start at positions
+    # with rep <= L whose def reaches that level's slot_def; their container
+    # is the most recent level-(L-1) slot (rows for L=1). A level is null at
+    # its container iff def stops one short of slot_def (and the node is
+    # optional); equal-to-slot_def-minus-... == slot_def-1 with opt=0 means
+    # present-but-empty.
+    from .ops.aggregate import _validity_from_bool
+    parent_mask = row_mask
+    parent_ids = torch.cumsum(row_mask.to(torch.int64), 0) - 1
+    nparents = total_rows
+    levels = []  # (nparents, offsets, validity, nslots)
+    for lvl, (opt, slot_def) in enumerate(chain, 1):
+        slot_mask = (rep64 <= lvl) & (def64 >= slot_def)
+        offsets_l = torch.zeros(nparents + 1, dtype=torch.int32, device=dev)
+        if nparents:
+            lengths = torch.bincount(parent_ids[slot_mask],
+                                     minlength=nparents)
+            offsets_l[1:] = torch.cumsum(lengths, 0).to(torch.int32)
+        validity_l = None
+        if opt and nparents:
+            validity_l = _validity_from_bool(
+                def64[parent_mask] >= slot_def - 1)
+        nslots = int(offsets_l[-1].item())
+        levels.append((nparents, offsets_l, validity_l, nslots))
+        parent_mask = slot_mask
+        parent_ids = torch.cumsum(slot_mask.to(torch.int64), 0) - 1
+        nparents = nslots
+    elem_mask = parent_mask
+    total_elems = nparents
 
     # element-space def (1 = element non-null) + exclusive valid prefix
     elem_def = (def64[elem_mask] == max_def).to(torch.uint8)
@@ -1008,14 +1071,14 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
         child = Column(edt, total_elems, out[:max(total_elems, 1)],
                        elem_validity, scale=elem.scale, null_count=None)
 
-    # ---- list validity (null vs empty/populated lists) --------------------
-    validity = None
-    if f.repetition == 1 and total_rows:
-        from .ops.aggregate import _validity_from_bool
-        def_at_row = def64[row_mask]
-        validity = _validity_from_bool(def_at_row >= 1)
-    return Column(DType.LIST, total_rows, None, validity, offsets, [child],
-                  null_count=None)
+    # ---- wrap the offsets chain inner->outer ------------------------------
+    col = child
+    for (np_l, offsets_l, validity_l, _ns) in reversed(levels):
+        col = Column(DType.LIST, np_l, None, validity_l, offsets_l, [col],
+                     null_count=None)
+    if return_elem_def:
+        return col, def64[elem_mask]
+    return col
 
 
 def _flatten_struct_leaves(f: SchemaField, out=None):
@@ -1278,6 +1341,54 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
     return col
 
 
+def _read_list_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
+                             total_rows: int, device) -> Column:
+    """LIST<STRUCT<...>> decode by composition (like MAP): every struct leaf
+    has exactly the level structure of a LIST of that leaf — same repetition
+    chain, deeper definition chain — so each decodes through the generic
+    LIST machinery with a synthetic per-leaf field whose max_def is the
+    leaf's ABSOLUTE level. The list offsets/validity are identical across
+    leaves (taken from the first); struct-level validity at each node falls
+    out of the first descendant leaf's element-space def levels
+    (def >= node.max_def means that struct level is present)."""
+    elem = f.element
+    leaves = _flatten_struct_leaves(elem)
+    cols = []
+    edefs = []
+    for li, leaf in enumerate(leaves):
+        chunks = [rg.columns[leaf0 + li] for rg in row_groups]
+        synth = SchemaField(
+            name=f.name, physical_type=-1, repetition=f.repetition,
+            converted_type=3, is_list=True, element=leaf,
+            max_def=leaf.max_def, max_rep=1)
+        col, edef = _read_list_column(raw, synth, chunks, total_rows, device,
+                                      return_elem_def=True)
+        cols.append(col)
+        edefs.append(edef)
+
+    from .ops.aggregate import _validity_from_bool
+    nentries = cols[0].children[0].size
+
+    def assemble(node, li):
+        if not node.is_struct:
+            return cols[li].children[0], li + 1
+        first = li
+        kids = []
+        for ch in node.children:
+            c, li = assemble(ch, li)
+            kids.append(c)
+        validity = None
+        if node.repetition == 1 and nentries:
+            validity = _validity_from_bool(edefs[first] >= node.max_def)
+        return Column(DType.STRUCT, nentries, None, validity, None, kids,
+                      null_count=None), li
+
+    entries, used = assemble(elem, 0)
+    assert used == len(leaves)
+    return Column(DType.LIST, total_rows, None, cols[0].validity,
+                  cols[0].offsets, [entries], null_count=None)
+
+
 def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
                      total_rows: int, device) -> Column:
     """MAP decode by composition: each of the key/value leaves has exactly
@@ -1334,6 +1445,11 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
             cols.append(_read_map_column(raw, f, footer.row_groups, leaf,
                                          total_rows, device))
             leaf += 2
+            continue
+        if f.is_list and f.element.is_struct:
+            cols.append(_read_list_struct_column(
+                raw, f, footer.row_groups, leaf, total_rows, device))
+            leaf += len(_flatten_struct_leaves(f.element))
             continue
         chunks = [rg.columns[leaf] for rg in footer.row_groups]
         if f.is_list:

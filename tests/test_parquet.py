@@ -518,3 +518,131 @@ def test_decode_map_column(tmp_path, codec):
             assert g is None, i
         else:
             assert g == [tuple(kv) for kv in x], (i, g, x)
+
+
+def _lstruct_rows(n, seed):
+    rng = random.Random(seed)
+    rows = []
+    for i in range(n):
+        if i % 17 == 3:
+            rows.append(None)
+            continue
+        lst = []
+        for j in range(rng.randrange(0, 4)):
+            if (i + j) % 11 == 5:
+                lst.append(None)  # null struct element
+            else:
+                lst.append({"a": None if (i * 3 + j) % 7 == 2 else i + j,
+                            "b": None if j % 5 == 1 else f"s{i % 40}_{j}"})
+        rows.append(lst)
+    return rows
+
+
+def test_footer_nested_list_schema(tmp_path):
+    """Schema parse (CPU): LIST<STRUCT> and LIST<LIST> shapes and levels."""
+    typ_ls = pa.list_(pa.struct([("a", pa.int64()), ("b", pa.string())]))
+    typ_ll = pa.list_(pa.list_(pa.int64()))
+    t = pa.table({"ls": pa.array([[{"a": 1, "b": "x"}]], type=typ_ls),
+                  "ll": pa.array([[[1, 2], []]], type=typ_ll)})
+    p = str(tmp_path / "sch.parquet")
+    pq.write_table(t, p, compression="NONE")
+    f = srj_pq.read_footer(p)
+    ls, ll = f.schema
+    assert ls.is_list and ls.element.is_struct and ls.max_rep == 1
+    # optional list + repeated + (a,b optional leaves under the elem struct)
+    leaves = srj_pq._flatten_struct_leaves(ls.element)
+    assert [lf.max_def for lf in leaves] == [ls.max_def] * 2
+    assert ll.is_list and ll.element.is_list and ll.max_rep == 2
+    assert ll.element.element.physical_type == srj_pq.T_INT64
+
+
+@pytest.mark.parametrize("pagever", ["1.0", "2.0"])
+@pytest.mark.gpu
+def test_decode_list_of_struct(tmp_path, pagever):
+    rows = _lstruct_rows(2500, 7)
+    typ = pa.list_(pa.struct([("a", pa.int64()), ("b", pa.string())]))
+    t = pa.table({"ls": pa.array(rows, type=typ)})
+    p = str(tmp_path / "ls.parquet")
+    pq.write_table(t, p, compression="NONE", data_page_version=pagever,
+                   row_group_size=900)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    exp = [None if r is None else
+           [None if e is None else (e["a"], e["b"]) for e in r] for r in rows]
+    assert col.to_pylist() == exp
+
+
+@pytest.mark.gpu
+def test_decode_list_of_nested_struct(tmp_path):
+    """struct-in-struct inside a list element; snappy + dictionary pages."""
+    rows = []
+    for i, base in enumerate(_lstruct_rows(1500, 13)):
+        if base is None or i % 3 == 0:
+            rows.append(base)
+            continue
+        rows.append([None if e is None else
+                     {"inner": e, "c": (i * 7) % 23} for e in base])
+    typ = pa.list_(pa.struct([
+        ("inner", pa.struct([("a", pa.int64()), ("b", pa.string())])),
+        ("c", pa.int32())]))
+    fixed = [None if r is None else
+             [e if e is None or "inner" in e else {"inner": e, "c": None}
+              for e in r] for r in rows]
+    t = pa.table({"ls": pa.array(fixed, type=typ)})
+    p = str(tmp_path / "lns.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=True,
+                   row_group_size=700)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+
+    def conv(e):
+        if e is None:
+            return None
+        inner = e["inner"]
+        return ((None if inner is None else (inner["a"], inner["b"])),
+                e["c"])
+    exp = [None if r is None else [conv(e) for e in r] for r in fixed]
+    assert col.to_pylist() == exp
+
+
+@pytest.mark.parametrize("pagever", ["1.0", "2.0"])
+@pytest.mark.gpu
+def test_decode_list_of_list(tmp_path, pagever):
+    rng = random.Random(11)
+    rows = []
+    for i in range(2500):
+        if i % 19 == 2:
+            rows.append(None)
+            continue
+        outer = []
+        for j in range(rng.randrange(0, 4)):
+            if (i + j) % 13 == 6:
+                outer.append(None)
+                continue
+            outer.append([None if (i * 7 + j + k) % 5 == 1 else i + j * 10 + k
+                          for k in range(rng.randrange(0, 3))])
+        rows.append(outer)
+    t = pa.table({"ll": pa.array(rows, type=pa.list_(pa.list_(pa.int64())))})
+    p = str(tmp_path / "ll.parquet")
+    pq.write_table(t, p, compression="NONE", data_page_version=pagever,
+                   row_group_size=800)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    assert col.to_pylist() == rows
+
+
+@pytest.mark.gpu
+def test_decode_list_of_list_strings(tmp_path):
+    """3-deep def chain + dictionary-encoded string leaves."""
+    rng = random.Random(29)
+    vocab = [f"w{v}" for v in range(50)]
+    rows = []
+    for i in range(2000):
+        rows.append(None if i % 23 == 7 else
+                    [None if (i + j) % 9 == 4 else
+                     [None if (i + j + k) % 6 == 2 else rng.choice(vocab)
+                      for k in range(rng.randrange(0, 3))]
+                     for j in range(rng.randrange(0, 3))])
+    t = pa.table({"ll": pa.array(rows, type=pa.list_(pa.list_(pa.string())))})
+    p = str(tmp_path / "lls.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=True,
+                   row_group_size=650)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    assert col.to_pylist() == rows
