@@ -67,6 +67,9 @@ class SchemaField:
     is_map: bool = False
     map_key: Optional["SchemaField"] = None
     map_value: Optional["SchemaField"] = None
+    # def level contributed by ancestors ABOVE a LIST node (0 at top level;
+    # >0 for a LIST nested inside a STRUCT) — the chain walk starts here
+    d_base: int = 0
 
 
 @dataclass
@@ -186,21 +189,22 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                         name=se[4].decode(), physical_type=-1,
                         repetition=se.get(3, 0), converted_type=3,
                         is_list=True, element=inner, max_def=inner.max_def,
-                        max_rep=inner.max_rep + 1), j
+                        max_rep=inner.max_rep + 1, d_base=d_above), j
                 assert inner.is_struct
                 leaf_max = max(lf.max_def
                                for lf in _flatten_struct_leaves(inner))
                 return SchemaField(
                     name=se[4].decode(), physical_type=-1,
                     repetition=se.get(3, 0), converted_type=3, is_list=True,
-                    element=inner, max_def=leaf_max, max_rep=1), j
+                    element=inner, max_def=leaf_max, max_rep=1,
+                    d_base=d_above), j
             elem = leaf_field(el, 0)
             elem_opt = 1 if el.get(3, 0) == 1 else 0
             return SchemaField(
                 name=se[4].decode(), physical_type=-1,
                 repetition=se.get(3, 0), converted_type=3, is_list=True,
                 element=elem, max_def=d_above + outer_opt + 1 + elem_opt,
-                max_rep=1), i + 3
+                max_rep=1, d_base=d_above), i + 3
         if is_map_group(se):
             # MAP: optional group (MAP) { repeated group key_value {
             #   required <key>; optional <value> } }
@@ -224,9 +228,17 @@ def read_footer(path_or_bytes) -> ParquetFooter:
         j = i + 1
         for _ in range(nch):
             ch, j = parse_field(j, d_above + outer_opt)
-            if ch.is_list:
+            if ch.is_map:
                 raise NotImplementedError(
-                    "LIST nested inside STRUCT is not supported yet")
+                    "MAP nested inside STRUCT is not supported yet")
+            if ch.is_list:
+                tip = ch.element
+                while tip.is_list:
+                    tip = tip.element
+                if tip.is_struct:
+                    raise NotImplementedError(
+                        "LIST of STRUCT nested inside STRUCT "
+                        "is not supported yet")
             children.append(ch)
         return SchemaField(
             name=se[4].decode(), physical_type=-1,
@@ -798,7 +810,7 @@ def _list_chain(f: SchemaField):
     definition level at which an element slot of that level exists. Returns
     (chain, deepest_non_list_element)."""
     chain = []
-    d = 0
+    d = f.d_base
     cur = f
     while True:
         opt = 1 if cur.repetition == 1 else 0
@@ -811,7 +823,8 @@ def _list_chain(f: SchemaField):
 
 
 def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
-                      total_rows: int, device, return_elem_def=False):
+                      total_rows: int, device, return_elem_def=False,
+                      return_row_def=False):
     """LIST decode (parquet 3-level encoding), any repetition depth:
     LIST<leaf>, LIST<LIST<...<leaf>>>, and (via _read_list_struct_column's
     per-leaf synthetic fields) the leaves of LIST<STRUCT<...>>.
@@ -1078,6 +1091,8 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                      null_count=None)
     if return_elem_def:
         return col, def64[elem_mask]
+    if return_row_def:
+        return col, def64[row_mask]
     return col
 
 
@@ -1110,6 +1125,15 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
 
     for li, leaf in enumerate(leaves):
         chunks = [rg.columns[leaf0 + li] for rg in row_groups]
+        if leaf.is_list:
+            # LIST child of a STRUCT: the full list machinery applies with
+            # the struct chain's def levels as the base; its row-space def
+            # levels double as the ancestor-struct validity source.
+            lcol, row_def = _read_list_column(raw, leaf, chunks, total_rows,
+                                              device, return_row_def=True)
+            children.append(lcol)
+            leaf_levs.append(row_def)
+            continue
         max_def = leaf.max_def
         def_bw = max(max_def.bit_length(), 1)
         pages: List[_Page] = []

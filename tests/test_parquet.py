@@ -646,3 +646,53 @@ def test_decode_list_of_list_strings(tmp_path):
                    row_group_size=650)
     col = srj_pq.read_table(p, device="cuda").columns[0]
     assert col.to_pylist() == rows
+
+
+@pytest.mark.gpu
+def test_decode_struct_with_list(tmp_path):
+    """LIST nested inside STRUCT (struct validity from the list leaf too)."""
+    rng = random.Random(41)
+    rows = []
+    for i in range(2200):
+        if i % 13 == 4:
+            rows.append(None)
+            continue
+        tags = (None if i % 9 == 2 else
+                [None if (i + k) % 7 == 3 else (i * 3 + k) % 1000
+                 for k in range(rng.randrange(0, 4))])
+        rows.append({"x": None if i % 5 == 1 else f"n{i % 60}",
+                     "tags": tags})
+    typ = pa.struct([("x", pa.string()), ("tags", pa.list_(pa.int64()))])
+    t = pa.table({"st": pa.array(rows, type=typ)})
+    p = str(tmp_path / "swl.parquet")
+    pq.write_table(t, p, compression="NONE", row_group_size=750)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    exp = [None if r is None else (r["x"], r["tags"]) for r in rows]
+    assert col.to_pylist() == exp
+
+
+@pytest.mark.gpu
+def test_decode_struct_with_list_first_leaf(tmp_path):
+    """The LIST is the struct's FIRST leaf — struct validity must come from
+    the list's row-space def levels; inner struct adds a level."""
+    rows = []
+    for i in range(1500):
+        if i % 11 == 3:
+            rows.append(None)
+            continue
+        inner = (None if i % 17 == 5 else
+                 {"ll": None if i % 7 == 1 else
+                  [[j, None, i % 50] for j in range(i % 3)]})
+        rows.append({"inner": inner, "y": i % 100})
+    typ = pa.struct([
+        ("inner", pa.struct([("ll", pa.list_(pa.list_(pa.int32())))])),
+        ("y", pa.int32())])
+    t = pa.table({"st": pa.array(rows, type=typ)})
+    p = str(tmp_path / "swl2.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=True,
+                   row_group_size=600)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    exp = [None if r is None else
+           ((None if r["inner"] is None else (r["inner"]["ll"],)), r["y"])
+           for r in rows]
+    assert col.to_pylist() == exp
