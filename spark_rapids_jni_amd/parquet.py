@@ -839,6 +839,9 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
     stream = _native.current_stream()
     dev = torch.device(device)
     chain, elem = _list_chain(f)
+    if elem.is_struct:
+        raise NotImplementedError(
+            "STRUCT below more than one LIST level is not supported yet")
     if elem.physical_type == T_FIXED_LEN_BYTE_ARRAY:
         raise NotImplementedError("LIST of FLBA decimals not supported")
     max_def = f.max_def
@@ -1427,21 +1430,47 @@ def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
         converted_type=3, is_list=True, element=f.map_key,
         max_def=outer_opt + 1 + (1 if f.map_key.repetition == 1 else 0),
         max_rep=1)
-    val_f = SchemaField(
-        name=f.name, physical_type=-1, repetition=f.repetition,
-        converted_type=3, is_list=True, element=f.map_value,
-        max_def=outer_opt + 1 + (1 if f.map_value.repetition == 1 else 0),
-        max_rep=1)
     kchunks = [rg.columns[leaf0] for rg in row_groups]
-    vchunks = [rg.columns[leaf0 + 1] for rg in row_groups]
     klist = _read_list_column(raw, key_f, kchunks, total_rows, device)
-    vlist = _read_list_column(raw, val_f, vchunks, total_rows, device)
+    mv = f.map_value
+    if mv.is_struct:
+        # MAP<k, STRUCT<...>>: the value side is exactly LIST<STRUCT> with
+        # the key_value chain as the (single) repetition level
+        synth = SchemaField(
+            name=f.name, physical_type=-1, repetition=f.repetition,
+            converted_type=3, is_list=True, element=mv,
+            max_def=max(lf.max_def for lf in _flatten_struct_leaves(mv)),
+            max_rep=1)
+        vlist = _read_list_struct_column(raw, synth, row_groups, leaf0 + 1,
+                                         total_rows, device)
+    elif mv.is_list:
+        # MAP<k, LIST<...>>: wrap the parsed value-list node in a synthetic
+        # outer list for the key_value repetition level (chain depth + 1)
+        synth = SchemaField(
+            name=f.name, physical_type=-1, repetition=f.repetition,
+            converted_type=3, is_list=True, element=mv, max_def=mv.max_def,
+            max_rep=mv.max_rep + 1)
+        vchunks = [rg.columns[leaf0 + 1] for rg in row_groups]
+        vlist = _read_list_column(raw, synth, vchunks, total_rows, device)
+    else:
+        val_f = SchemaField(
+            name=f.name, physical_type=-1, repetition=f.repetition,
+            converted_type=3, is_list=True, element=mv,
+            max_def=outer_opt + 1 + (1 if mv.repetition == 1 else 0),
+            max_rep=1)
+        vchunks = [rg.columns[leaf0 + 1] for rg in row_groups]
+        vlist = _read_list_column(raw, val_f, vchunks, total_rows, device)
     nentries = klist.children[0].size
     entries = Column(DType.STRUCT, nentries, None, None, None,
                      [klist.children[0], vlist.children[0]],
                      null_count=None)
     return Column(DType.LIST, total_rows, None, klist.validity,
                   klist.offsets, [entries], null_count=None)
+
+
+def _map_leaf_count(f: SchemaField) -> int:
+    mv = f.map_value
+    return 1 + (len(_flatten_struct_leaves(mv)) if mv.is_struct else 1)
 
 
 def read_table(path: str, columns: Optional[Sequence[str]] = None,
@@ -1468,7 +1497,7 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
         if f.is_map:
             cols.append(_read_map_column(raw, f, footer.row_groups, leaf,
                                          total_rows, device))
-            leaf += 2
+            leaf += _map_leaf_count(f)
             continue
         if f.is_list and f.element.is_struct:
             cols.append(_read_list_struct_column(

@@ -696,3 +696,56 @@ def test_decode_struct_with_list_first_leaf(tmp_path):
            ((None if r["inner"] is None else (r["inner"]["ll"],)), r["y"])
            for r in rows]
     assert col.to_pylist() == exp
+
+
+@pytest.mark.gpu
+def test_decode_map_of_list_value(tmp_path):
+    """MAP<string, LIST<int>>: null maps, null/empty value lists."""
+    rng = random.Random(53)
+    rows = []
+    for i in range(1800):
+        if i % 15 == 6:
+            rows.append(None)
+            continue
+        ents = []
+        for j in range(rng.randrange(0, 4)):
+            v = (None if (i + j) % 8 == 3 else
+                 [None if (i + j + k) % 6 == 1 else i + 10 * j + k
+                  for k in range(rng.randrange(0, 3))])
+            ents.append((f"k{j}", v))
+        rows.append(ents)
+    typ = pa.map_(pa.string(), pa.list_(pa.int64()))
+    t = pa.table({"m": pa.array(rows, type=typ)})
+    p = str(tmp_path / "mlv.parquet")
+    pq.write_table(t, p, compression="NONE", row_group_size=700)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    assert col.to_pylist() == rows
+
+
+@pytest.mark.gpu
+def test_decode_map_of_struct_value(tmp_path):
+    """MAP<int, STRUCT<a,b>>: null struct values and null struct fields."""
+    rng = random.Random(59)
+    rows = []
+    for i in range(1800):
+        if i % 21 == 8:
+            rows.append(None)
+            continue
+        ents = []
+        for j in range(rng.randrange(0, 3)):
+            v = (None if (i + j) % 7 == 2 else
+                 {"a": None if (i * 5 + j) % 9 == 4 else i + j,
+                  "b": None if j % 4 == 1 else f"v{(i + j) % 30}"})
+            ents.append((i * 10 + j, v))
+        rows.append(ents)
+    typ = pa.map_(pa.int64(), pa.struct([("a", pa.int64()),
+                                         ("b", pa.string())]))
+    t = pa.table({"m": pa.array(rows, type=typ)})
+    p = str(tmp_path / "msv.parquet")
+    pq.write_table(t, p, compression="SNAPPY", use_dictionary=True,
+                   row_group_size=650)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    exp = [None if r is None else
+           [(k, None if v is None else (v["a"], v["b"])) for k, v in r]
+           for r in rows]
+    assert col.to_pylist() == exp
