@@ -212,17 +212,27 @@ def read_footer(path_or_bytes) -> ParquetFooter:
             assert kv.get(3, 0) == 2 and kv.get(5, 0) == 2, \
                 "unrecognized MAP encoding"
             kse = schema_elems[i + 2]
-            vse = schema_elems[i + 3]
-            if kse.get(5, 0) != 0 or vse.get(5, 0) != 0:
-                raise NotImplementedError("MAP of nested key/value types "
-                                          "is not supported yet")
+            if kse.get(5, 0) != 0:
+                raise NotImplementedError("MAP with a nested KEY type "
+                                          "is not supported")
+            d_elem = d_above + outer_opt + 1
             key = leaf_field(kse, 0)
-            val = leaf_field(vse, 0)
+            vse = schema_elems[i + 3]
+            if vse.get(5, 0) == 0:
+                val = leaf_field(vse, 0)
+                j = i + 4
+            else:
+                # nested value: LIST or STRUCT, parsed with the map chain's
+                # def levels accumulated (absolute leaf max_def)
+                val, j = parse_field(i + 3, d_elem)
+                if val.is_map:
+                    raise NotImplementedError(
+                        "MAP with a MAP value is not supported yet")
             return SchemaField(
                 name=se[4].decode(), physical_type=-1,
                 repetition=se.get(3, 0), converted_type=2, is_list=False,
                 is_map=True, map_key=key, map_value=val,
-                max_def=d_above + outer_opt + 1, max_rep=1), i + 4
+                max_def=d_above + outer_opt + 1, max_rep=1), j
         # plain group = STRUCT; children may be leaves or nested STRUCTs
         children = []
         j = i + 1
