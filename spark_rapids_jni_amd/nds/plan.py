@@ -191,7 +191,15 @@ def _val_to_column(v: Val):
     if data.dtype == torch.bool:
         data = data.to(torch.int8)
     dt = _TORCH2DT[data.dtype]
-    return Column(dt, data.numel(), data, _pack_validity(v.valid),
+    valid = v.valid
+    # expressions like case_when(..., otherwise=lit(0)) produce an all-true
+    # mask: drop it, or every SUM downstream pays a hidden per-group
+    # count_valid atomic stream AND global aggregates lose the LDS
+    # pre-aggregation path (naggs<=3 limit) — q97's final agg was 2x255 ms
+    # on a single contended CAS slot because of exactly this
+    if valid is not None and data.numel() and bool(valid.all().item()):
+        valid = None
+    return Column(dt, data.numel(), data, _pack_validity(valid),
                   null_count=None)
 
 
@@ -607,12 +615,33 @@ class Engine:
         # LDS pre-aggregation kernel instead of a per-row contended CAS
         # table; chunk to <=3 aggs per pass (the LDS kernel's limit)
         if dummy_key:
+            # chunk by EFFECTIVE agg count: a nullable SUM/MIN/MAX carries a
+            # hidden count_valid stream inside the kernel, so it costs 2 of
+            # the LDS kernel's <=3 agg slots (q97 postmortem: 3 sums over
+            # masked values -> naggs 6 -> no LDS -> 255 ms on one CAS slot)
+            def _cost(fn_v):
+                fn, v = fn_v
+                if fn in ("count", "count_valid") or v is None:
+                    return 1
+                if v.valid is None or (v.data.numel()
+                                       and bool(v.valid.all().item())):
+                    return 1
+                return 2
+            chunks, cur, budget = [], [], 3
+            for item in native:
+                c = _cost(item)
+                if cur and budget < c:
+                    chunks.append(cur)
+                    cur, budget = [], 3
+                cur.append(item)
+                budget -= c
+            if cur:
+                chunks.append(cur)
             kvals = rvals = None
             ng = 0
             out_r = []
-            for i in range(0, len(native), 3):
-                kvals, rv, ng = self.backend.groupby(kv, native[i:i + 3],
-                                                     hint=1)
+            for chunk in chunks:
+                kvals, rv, ng = self.backend.groupby(kv, chunk, hint=1)
                 out_r.extend(rv)
             rvals = out_r
         else:
