@@ -787,10 +787,26 @@ class Engine:
         return out_f
 
     def _rollup(self, f: Frame, p: Agg) -> Frame:
+        # cascade re-aggregation when every agg is decomposable: level L-1
+        # re-aggregates level L's RESULT (sum of sums, min of mins, count ->
+        # sum) instead of rescanning the fact input once per level — q67's
+        # 9-level rollup over 58M rows becomes one 58M pass + 8 passes over
+        # group-count-sized inputs. Both backends run the same cascade, so
+        # GPU-vs-oracle verification is unaffected.
+        from .expr import col as _col
+        cascade = all(fn in ("sum", "min", "max", "count")
+                      for _n, fn, _e in p.aggs)
         frames = []
+        src = None
         for lvl in range(len(p.keys), -1, -1):
             keys = p.keys[:lvl]
-            g = self._agg_frame(f, keys, p.aggs)
+            if src is None or not cascade:
+                g = self._agg_frame(f, keys, p.aggs)
+            else:
+                reaggs = [(n, "sum" if fn == "count" else fn, _col(n))
+                          for n, fn, _e in p.aggs]
+                g = self._agg_frame(src, keys, reaggs)
+            src = g
             cols = dict(g.cols)
             for miss in p.keys[lvl:]:
                 proto = f.cols[miss]

@@ -177,3 +177,43 @@ def test_distributed_matches_single(tmp_path, catalog):
 def test_gpu_matches_cpu_oracle_all_queries():
     failures = verify(sf=0.05)
     assert not failures, f"GPU/CPU mismatches: {failures}"
+
+
+def test_rollup_cascade_matches_bruteforce(catalog):
+    """The cascade re-aggregation in Engine._rollup (level L-1 from level
+    L's result) must equal a brute-force per-level aggregation computed in
+    plain Python — sums, counts, min/max, null keys included."""
+    from collections import defaultdict
+    from spark_rapids_jni_amd.nds.expr import col
+    from spark_rapids_jni_amd.nds.queries import A, S
+
+    e = Engine(catalog, device="cpu")
+    plan = A(S("item", "i_category", "i_class", "i_current_price",
+               "i_brand_id"),
+             ["i_category", "i_class"],
+             [("s", "sum", col("i_current_price")),
+              ("mx", "max", col("i_brand_id")),
+              ("c", "count", None)], rollup=True)
+    f = e.run(plan)
+    got = {}
+    for cat_, cls, s, mx, c, lvl in f.to_rows():
+        got[(lvl, cat_, cls)] = (None if s is None else round(s, 4), mx, c)
+
+    base = e.run(S("item", "i_category", "i_class", "i_current_price",
+                   "i_brand_id")).to_rows()
+    exp = {}
+    for lvl in (2, 1, 0):
+        acc = defaultdict(lambda: [0.0, False, None, 0])
+        for cat_, cls, price, brand in base:
+            key = (lvl, cat_ if lvl >= 1 else None,
+                   cls if lvl >= 2 else None)
+            a = acc[key]
+            if price is not None:
+                a[0] += price
+                a[1] = True
+            if brand is not None:
+                a[2] = brand if a[2] is None else max(a[2], brand)
+            a[3] += 1
+        for key, (s, any_s, mx, c) in acc.items():
+            exp[key] = (round(s, 4) if any_s else None, mx, c)
+    assert got == exp
