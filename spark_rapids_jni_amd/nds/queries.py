@@ -41,6 +41,18 @@ def A(child, keys, aggs, rollup=False):
     return Agg(child, keys, aggs, rollup)
 
 
+def day_pivot_sums(child, keys, val_expr, days):
+    """sum(case_when(d_day_name = X, v)) pivot, decomposed: ONE pass over
+    the fact input grouped by keys + d_day_name, then the 7 case-pivots
+    over the group-count-sized intermediate (sum distributes over the
+    finer partition, so results are exact)."""
+    pre = A(child, keys + ["d_day_name"], [("__pv", "sum", val_expr)])
+    aggs = [(dy.lower()[:3] + "_sales", "sum",
+             case_when((col("d_day_name") == dy, col("__pv")),
+                       otherwise=lit(0.0))) for dy in days]
+    return A(pre, keys, aggs)
+
+
 def dd(cond, *cols):
     want = set(cols) | {"d_date_sk"}
     return F(S("date_dim", *sorted(want)), cond)
@@ -84,10 +96,8 @@ def q2(e: Engine) -> Frame:
                [("sold_date_sk", "d_date_sk")])
     days = ["Sunday", "Monday", "Tuesday", "Wednesday", "Thursday",
             "Friday", "Saturday"]
-    aggs = [(dy.lower()[:3] + "_sales", "sum",
-             case_when((col("d_day_name") == dy, col("sales_price")),
-                       otherwise=lit(0.0))) for dy in days]
-    wswscs = A(joined, ["d_week_seq", "d_year"], aggs)
+    wswscs = day_pivot_sums(joined, ["d_week_seq", "d_year"],
+                            col("sales_price"), days)
     e.register("wswscs", e.run(wswscs))
     y = P(F(S("wswscs"), col("d_year") == 2001),
           ("week1", col("d_week_seq")),
@@ -1763,14 +1773,12 @@ def q58(e: Engine) -> Frame:
 def q59(e: Engine) -> Frame:
     days = ["Sunday", "Monday", "Tuesday", "Wednesday", "Thursday",
             "Friday", "Saturday"]
-    wss = A(J(S("store_sales", "ss_sold_date_sk", "ss_store_sk",
-               "ss_sales_price"),
-             S("date_dim", "d_date_sk", "d_week_seq", "d_day_name"),
-             [("ss_sold_date_sk", "d_date_sk")]),
-            ["d_week_seq", "ss_store_sk"],
-            [(dy.lower()[:3] + "_sales", "sum",
-              case_when((col("d_day_name") == dy, col("ss_sales_price")),
-                        otherwise=lit(0.0))) for dy in days])
+    wss = day_pivot_sums(
+        J(S("store_sales", "ss_sold_date_sk", "ss_store_sk",
+            "ss_sales_price"),
+          S("date_dim", "d_date_sk", "d_week_seq", "d_day_name"),
+          [("ss_sold_date_sk", "d_date_sk")]),
+        ["d_week_seq", "ss_store_sk"], col("ss_sales_price"), days)
     e.register("q59wss", e.run(wss))
     wk1 = e.scalar(P(dd(col("d_date") == date_lit("2001-01-01").value,
                         "d_date", "d_week_seq"), ("wk", col("d_week_seq"))),
