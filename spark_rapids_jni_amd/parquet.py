@@ -30,7 +30,11 @@ T_BOOLEAN, T_INT32, T_INT64, T_INT96, T_FLOAT, T_DOUBLE, T_BYTE_ARRAY, \
 ENC_PLAIN = 0
 ENC_PLAIN_DICTIONARY = 2
 ENC_RLE = 3
+ENC_DELTA_BINARY_PACKED = 5
+ENC_DELTA_LENGTH_BYTE_ARRAY = 6
+ENC_DELTA_BYTE_ARRAY = 7
 ENC_RLE_DICTIONARY = 8
+ENC_BYTE_STREAM_SPLIT = 9
 
 CODEC_UNCOMPRESSED = 0
 CODEC_SNAPPY = 1
@@ -39,6 +43,10 @@ CODEC_ZSTD = 6
 CODEC_LZ4_RAW = 7
 
 _SCATTER_FMT = "<QQQQqqqii"     # ScatterDesc (64B)
+_DELTA_FMT = "<QqQqQii"         # DeltaDesc (48B)
+_STROFF_FMT = "<QQQqqQQQ"       # StrOffDesc (64B)
+_DBA_FMT = "<QQQQQqQ"           # DbaDesc (56B)
+_BSS_FMT = "<QqQii"             # BssDesc (32B)
 _RLE_FMT = "<QqQqii"            # RleDesc (40B)
 _STRIDX_FMT = "<QqqQQ"          # StrIndexDesc (40B)
 _STRCPY_FMT = "<QQQQQQqqq"      # StrCopyDesc (72B)
@@ -693,6 +701,11 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
             st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
             g.pq_string_plain_index(st.data_ptr(), len(plain_meta), stream)
         pm = {i: (voff, vlen) for i, voff, vlen in plain_meta}
+        delta_idx = [i for i, p in enumerate(pages)
+                     if p.encoding in (ENC_DELTA_LENGTH_BYTE_ARRAY,
+                                       ENC_DELTA_BYTE_ARRAY)]
+        delta_str, _delta_hold = _decode_delta_strings(
+            g, stream, dev, pages, pbase, _plen, _body_off, delta_idx)
 
         cp_descs = bytearray()
         for i, p in enumerate(pages):
@@ -701,6 +714,12 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
                 voff, vlen = pm[i]
                 cp_descs += struct.pack(
                     _STRCPY_FMT, pbase[i] + bo, voff.data_ptr(), vlen.data_ptr(),
+                    0, def_ptr, vprefix_ptr, int(row_starts[i]), p.num_values,
+                    page_vbase[i])
+            elif i in delta_str:
+                sp, voff, vlen = delta_str[i]
+                cp_descs += struct.pack(
+                    _STRCPY_FMT, sp, voff.data_ptr(), vlen.data_ptr(),
                     0, def_ptr, vprefix_ptr, int(row_starts[i]), p.num_values,
                     page_vbase[i])
             else:
@@ -765,6 +784,35 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     from .columnar import TORCH_DTYPE
     numel = (total_rows * 2) if flba_dec else total_rows
     out = torch.zeros(max(numel, 1), dtype=TORCH_DTYPE[dtype], device=dev)
+    # DELTA_BINARY_PACKED / BYTE_STREAM_SPLIT pages decode into a dense
+    # valid-order temp first, then ride the standard scatter as if PLAIN
+    delta_tmp = {}
+    db_descs, bss_descs = bytearray(), bytearray()
+    ndb = nbss = 0
+    for i, p in enumerate(pages):
+        bo = _body_off(i, p)
+        if p.encoding == ENC_DELTA_BINARY_PACKED:
+            nv = max(p.num_values, 1)
+            tmp = torch.empty(nv * width, dtype=torch.uint8, device=dev)
+            delta_tmp[i] = tmp
+            db_descs += struct.pack(_DELTA_FMT, pbase[i] + bo,
+                                    _plen(i) - bo, tmp.data_ptr(),
+                                    p.num_values, 0, width, 0)
+            ndb += 1
+        elif p.encoding == ENC_BYTE_STREAM_SPLIT:
+            nv = (_plen(i) - bo) // width
+            tmp = torch.empty(max(nv * width, 1), dtype=torch.uint8,
+                              device=dev)
+            delta_tmp[i] = tmp
+            bss_descs += struct.pack(_BSS_FMT, pbase[i] + bo, nv,
+                                     tmp.data_ptr(), width, 0)
+            nbss += 1
+    if ndb:
+        t = torch.frombuffer(db_descs, dtype=torch.uint8).to(dev)
+        g.pq_delta_binpack(t.data_ptr(), ndb, stream)
+    if nbss:
+        t = torch.frombuffer(bss_descs, dtype=torch.uint8).to(dev)
+        g.pq_bss(t.data_ptr(), nbss, stream)
     sc_descs = bytearray()
     for i, p in enumerate(pages):
         bo = _body_off(i, p)
@@ -772,6 +820,11 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
             sc_descs += struct.pack(_SCATTER_FMT, pbase[i] + bo, 0, def_ptr,
                                     vprefix_ptr, int(row_starts[i]),
                                     p.num_values, page_vbase[i], width, 0)
+        elif i in delta_tmp:
+            sc_descs += struct.pack(_SCATTER_FMT, delta_tmp[i].data_ptr(), 0,
+                                    def_ptr, vprefix_ptr,
+                                    int(row_starts[i]), p.num_values,
+                                    page_vbase[i], width, 0)
         else:
             di = dict_per_page[i]
             sc_descs += struct.pack(_SCATTER_FMT, idx_tensors[i].data_ptr(),
@@ -793,6 +846,134 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
 
 
 _MMAP_CACHE: Dict[tuple, memoryview] = {}
+
+
+def _decode_delta_strings(g, stream, dev, pages, pbase, plen_of, body_of,
+                          idxs):
+    """DELTA_LENGTH_BYTE_ARRAY / DELTA_BYTE_ARRAY page decode.
+
+    Returns ({page_idx: (src_ptr, val_off, val_len)}, holders): descriptors
+    usable by the StrCopy kernels, plus tensors that must outlive the copy
+    kernels. DLBA = [lengths: DELTA_BINARY_PACKED int32][concatenated
+    bytes]; DBA = [prefix lens: DBP][suffix: DLBA], values reconstructed
+    into a scratch buffer (value v = value[v-1][:plen[v]] + suffix[v])."""
+    out = {}
+    holders = []
+    if not idxs:
+        return out, holders
+    lens1 = {}
+    consumed1 = torch.zeros(len(idxs), dtype=torch.int64, device=dev)
+    ki = {i: k for k, i in enumerate(idxs)}
+    descs = bytearray()
+    for k, i in enumerate(idxs):
+        p = pages[i]
+        bo = body_of(i, p)
+        nv = max(p.num_values, 1)
+        l1 = torch.empty(nv, dtype=torch.int32, device=dev)
+        lens1[i] = l1
+        descs += struct.pack(_DELTA_FMT, pbase[i] + bo, plen_of(i) - bo,
+                             l1.data_ptr(), p.num_values,
+                             consumed1.data_ptr() + 8 * k, 4, 0)
+    t = torch.frombuffer(descs, dtype=torch.uint8).to(dev)
+    g.pq_delta_binpack(t.data_ptr(), len(idxs), stream)
+    holders += [consumed1, t] + list(lens1.values())
+
+    dlba = [i for i in idxs
+            if pages[i].encoding == ENC_DELTA_LENGTH_BYTE_ARRAY]
+    dba = [i for i in idxs if pages[i].encoding == ENC_DELTA_BYTE_ARRAY]
+
+    soff_descs = bytearray()
+    nso = 0
+    for i in dlba:
+        p = pages[i]
+        nv = max(p.num_values, 1)
+        voff = torch.empty(nv, dtype=torch.int64, device=dev)
+        vlen = torch.empty(nv, dtype=torch.int32, device=dev)
+        out[i] = (pbase[i] + body_of(i, p), voff, vlen)
+        soff_descs += struct.pack(_STROFF_FMT, lens1[i].data_ptr(), 0,
+                                  consumed1.data_ptr() + 8 * ki[i], 0,
+                                  p.num_values, voff.data_ptr(),
+                                  vlen.data_ptr(), 0)
+        nso += 1
+    if nso:
+        t = torch.frombuffer(soff_descs, dtype=torch.uint8).to(dev)
+        g.pq_str_off(t.data_ptr(), nso, stream)
+        holders.append(t)
+    if not dba:
+        return out, holders
+
+    # DBA: suffix-length DBP section starts where the prefix one ended
+    c1 = consumed1.cpu().tolist()
+    lens2 = {}
+    consumed2 = torch.zeros(len(dba), dtype=torch.int64, device=dev)
+    totals = torch.zeros(len(dba), dtype=torch.int64, device=dev)
+    vo = {}
+    vl = {}
+    descs = bytearray()
+    soff_descs = bytearray()
+    for k, i in enumerate(dba):
+        p = pages[i]
+        bo = body_of(i, p)
+        nv = max(p.num_values, 1)
+        l2 = torch.empty(nv, dtype=torch.int32, device=dev)
+        lens2[i] = l2
+        start2 = bo + int(c1[ki[i]])
+        descs += struct.pack(_DELTA_FMT, pbase[i] + start2,
+                             plen_of(i) - start2, l2.data_ptr(),
+                             p.num_values, consumed2.data_ptr() + 8 * k, 4, 0)
+        vo[i] = torch.empty(nv, dtype=torch.int64, device=dev)
+        vl[i] = torch.empty(nv, dtype=torch.int32, device=dev)
+        soff_descs += struct.pack(_STROFF_FMT, lens1[i].data_ptr(),
+                                  l2.data_ptr(), 0, 0, p.num_values,
+                                  vo[i].data_ptr(), vl[i].data_ptr(),
+                                  totals.data_ptr() + 8 * k)
+    t = torch.frombuffer(descs, dtype=torch.uint8).to(dev)
+    g.pq_delta_binpack(t.data_ptr(), len(dba), stream)
+    t2 = torch.frombuffer(soff_descs, dtype=torch.uint8).to(dev)
+    g.pq_str_off(t2.data_ptr(), len(dba), stream)
+    holders += [consumed2, t, t2] + list(lens2.values())
+    back = torch.cat([consumed2, totals]).cpu().tolist()
+    c2 = back[:len(dba)]
+    tot = back[len(dba):]
+
+    scratch_off = []
+    total_chars = 0
+    for k in range(len(dba)):
+        scratch_off.append(total_chars)
+        total_chars += int(tot[k])
+    scratch = torch.empty(max(total_chars, 1), dtype=torch.uint8, device=dev)
+    holders.append(scratch)
+    soff_descs = bytearray()
+    dba_descs = bytearray()
+    suf = {}
+    for k, i in enumerate(dba):
+        p = pages[i]
+        bo = body_of(i, p)
+        nv = max(p.num_values, 1)
+        start2 = bo + int(c1[ki[i]])
+        # scratch-absolute value offsets (recomputed with the page base)
+        soff_descs += struct.pack(_STROFF_FMT, lens1[i].data_ptr(),
+                                  lens2[i].data_ptr(), 0, scratch_off[k],
+                                  p.num_values, vo[i].data_ptr(),
+                                  vl[i].data_ptr(), 0)
+        suf[i] = torch.empty(nv, dtype=torch.int64, device=dev)
+        sl_dummy = torch.empty(nv, dtype=torch.int32, device=dev)
+        holders.append(sl_dummy)
+        # suffix byte offsets, absolute within the page payload
+        soff_descs += struct.pack(_STROFF_FMT, lens2[i].data_ptr(), 0, 0,
+                                  start2 + int(c2[k]), p.num_values,
+                                  suf[i].data_ptr(), sl_dummy.data_ptr(), 0)
+        dba_descs += struct.pack(_DBA_FMT, pbase[i], lens1[i].data_ptr(),
+                                 lens2[i].data_ptr(), suf[i].data_ptr(),
+                                 vo[i].data_ptr(), p.num_values,
+                                 scratch.data_ptr())
+        out[i] = (scratch.data_ptr(), vo[i], vl[i])
+    t = torch.frombuffer(soff_descs, dtype=torch.uint8).to(dev)
+    g.pq_str_off(t.data_ptr(), 2 * len(dba), stream)
+    t2 = torch.frombuffer(dba_descs, dtype=torch.uint8).to(dev)
+    g.pq_dba_reconstruct(t2.data_ptr(), len(dba), stream)
+    holders += [t, t2] + list(suf.values())
+    return out, holders
 
 
 def _mmap_file(path: str) -> memoryview:
@@ -873,6 +1054,13 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                 dict_per_page.append(cur_dict)
 
     # upload payloads
+    _DELTA_ENCS = (ENC_DELTA_BINARY_PACKED, ENC_DELTA_LENGTH_BYTE_ARRAY,
+                   ENC_DELTA_BYTE_ARRAY, ENC_BYTE_STREAM_SPLIT)
+    for p_ in pages:
+        if p_.encoding in _DELTA_ENCS:
+            raise NotImplementedError(
+                "DELTA/BYTE_STREAM_SPLIT encodings inside nested columns "
+                "are not supported yet (flat columns decode them)")
     blobs = [p.data for p in pages] + [p.data for p in dicts]
     offs = np.zeros(len(blobs) + 1, dtype=np.int64)
     for i, b in enumerate(blobs):
@@ -1161,6 +1349,13 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                 else:
                     pages.append(p)
                     dict_per_page.append(cur_dict)
+        _DELTA_ENCS = (ENC_DELTA_BINARY_PACKED, ENC_DELTA_LENGTH_BYTE_ARRAY,
+                           ENC_DELTA_BYTE_ARRAY, ENC_BYTE_STREAM_SPLIT)
+        for p_ in pages:
+            if p_.encoding in _DELTA_ENCS:
+                raise NotImplementedError(
+                    "DELTA/BYTE_STREAM_SPLIT encodings inside nested columns "
+                    "are not supported yet (flat columns decode them)")
         blobs = [p.data for p in pages] + [p.data for p in dicts]
         offs = np.zeros(len(blobs) + 1, dtype=np.int64)
         for i, b in enumerate(blobs):

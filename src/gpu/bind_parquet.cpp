@@ -11,9 +11,30 @@ void srj_def_to_validity(const uint8_t*, int64_t, uint8_t*, hipStream_t);
 void srj_pq_snappy_decomp(const void*, int32_t, hipStream_t);
 void srj_gather_u8_at(const uint64_t*, int64_t, uint8_t*, hipStream_t);
 void srj_pq_flba_dec128(const void*, int32_t, int64_t*, hipStream_t);
+void srj_pq_delta_binpack(const void*, int32_t, hipStream_t);
+void srj_pq_str_off(const void*, int32_t, hipStream_t);
+void srj_pq_dba_reconstruct(const void*, int32_t, hipStream_t);
+void srj_pq_bss(const void*, int32_t, hipStream_t);
 }
 
 void register_parquet(py::module_& m) {
+  m.def("pq_delta_binpack", [](uintptr_t descs, int32_t n, uintptr_t stream) {
+    srj_pq_delta_binpack(as_ptr<void>(descs), n, as_stream(stream));
+    check_hip("pq_delta_binpack");
+  });
+  m.def("pq_str_off", [](uintptr_t descs, int32_t n, uintptr_t stream) {
+    srj_pq_str_off(as_ptr<void>(descs), n, as_stream(stream));
+    check_hip("pq_str_off");
+  });
+  m.def("pq_dba_reconstruct", [](uintptr_t descs, int32_t n,
+                                 uintptr_t stream) {
+    srj_pq_dba_reconstruct(as_ptr<void>(descs), n, as_stream(stream));
+    check_hip("pq_dba_reconstruct");
+  });
+  m.def("pq_bss", [](uintptr_t descs, int32_t n, uintptr_t stream) {
+    srj_pq_bss(as_ptr<void>(descs), n, as_stream(stream));
+    check_hip("pq_bss");
+  });
   m.def("pq_snappy_decomp", [](uintptr_t descs, int32_t n, uintptr_t stream) {
     srj_pq_snappy_decomp(as_ptr<void>(descs), n, as_stream(stream));
     check_hip("pq_snappy_decomp");

@@ -383,7 +383,274 @@ __global__ void flba_dec128_kernel(const ScatterDesc* __restrict__ descs,
 
 using namespace srj;
 
+// ---------------------------------------------------------------------------
+// DELTA encodings (parquet-mr v2 defaults; reference decodes via libcudf):
+// DELTA_BINARY_PACKED ints, DELTA_LENGTH_BYTE_ARRAY and DELTA_BYTE_ARRAY
+// strings, BYTE_STREAM_SPLIT floats. One wave per page; miniblocks unpack
+// lane-parallel with a wave-wide prefix sum carrying the running value.
+// ---------------------------------------------------------------------------
+struct DeltaDesc {
+  const uint8_t* src;   // DBP section start (page body)
+  int64_t src_len;
+  uint8_t* out;         // dense values, `width` bytes each, valid-order
+  int64_t expected;     // capacity of out (upper bound on value count)
+  int64_t* consumed;    // out: bytes consumed by the DBP section (or null)
+  int32_t width;        // 4 or 8
+  int32_t pad;
+};
+
+__device__ inline uint64_t dbp_varint(const uint8_t* p, int64_t len,
+                                      int64_t& pos) {
+  uint64_t v = 0;
+  int sh = 0;
+  while (pos < len && sh < 64) {
+    uint8_t b = p[pos++];
+    v |= (uint64_t)(b & 0x7F) << sh;
+    if (!(b & 0x80)) break;
+    sh += 7;
+  }
+  return v;
+}
+
+__device__ inline int64_t dbp_zigzag(uint64_t u) {
+  return (int64_t)((u >> 1) ^ (~(u & 1) + 1));
+}
+
+// w bits starting at bit `bitoff` of p[0..plen), little-endian packing
+__device__ inline uint64_t load_bits(const uint8_t* p, int64_t plen,
+                                     int64_t bitoff, int w) {
+  if (w <= 0) return 0;
+  int64_t b0 = bitoff >> 3;
+  int shift = (int)(bitoff & 7);
+  unsigned __int128 acc = 0;
+  int need = (shift + w + 7) >> 3;
+  for (int b = 0; b < need && b0 + b < plen; ++b)
+    acc |= (unsigned __int128)p[b0 + b] << (8 * b);
+  uint64_t v = (uint64_t)(acc >> shift);
+  if (w < 64) v &= (((uint64_t)1 << w) - 1);
+  return v;
+}
+
+__device__ inline uint64_t shfl_u64(uint64_t v, int lane) {
+  return (uint64_t)__shfl((long long)v, lane, 64);
+}
+
+__global__ void delta_binpack_kernel(const DeltaDesc* __restrict__ descs,
+                                     int32_t npages) {
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  int wpb = blockDim.x >> 6;
+  for (int32_t page = (int32_t)blockIdx.x * wpb + wave; page < npages;
+       page += gridDim.x * wpb) {
+    DeltaDesc d = descs[page];
+    uint64_t block_size = 0, nmini = 0, total = 0;
+    int64_t first = 0, pos = 0;
+    if (lane == 0) {
+      block_size = dbp_varint(d.src, d.src_len, pos);
+      nmini = dbp_varint(d.src, d.src_len, pos);
+      total = dbp_varint(d.src, d.src_len, pos);
+      first = dbp_zigzag(dbp_varint(d.src, d.src_len, pos));
+    }
+    block_size = shfl_u64(block_size, 0);
+    nmini = shfl_u64(nmini, 0);
+    total = shfl_u64(total, 0);
+    first = (int64_t)shfl_u64((uint64_t)first, 0);
+    pos = (int64_t)shfl_u64((uint64_t)pos, 0);
+    int64_t n = (int64_t)total < d.expected ? (int64_t)total : d.expected;
+    int64_t vpm = nmini ? (int64_t)(block_size / nmini) : 0;
+    if (n > 0 && lane == 0) {
+      if (d.width == 8) reinterpret_cast<uint64_t*>(d.out)[0] = (uint64_t)first;
+      else reinterpret_cast<uint32_t*>(d.out)[0] = (uint32_t)first;
+    }
+    uint64_t prev = (uint64_t)first;
+    int64_t produced = 1;
+    int64_t remaining = n > 0 ? n - 1 : 0;
+    while (remaining > 0 && vpm > 0 && pos < d.src_len) {
+      uint64_t md = 0;
+      int64_t p2 = pos;
+      if (lane == 0) md = dbp_varint(d.src, d.src_len, p2);
+      md = shfl_u64(md, 0);
+      p2 = (int64_t)shfl_u64((uint64_t)p2, 0);
+      int64_t min_delta = dbp_zigzag(md);
+      int64_t bw_pos = p2;
+      int64_t data_pos = bw_pos + (int64_t)nmini;
+      for (uint32_t mb = 0; mb < (uint32_t)nmini && remaining > 0; ++mb) {
+        int w = (bw_pos + mb < d.src_len) ? d.src[bw_pos + mb] : 0;
+        int64_t take = vpm < remaining ? vpm : remaining;
+        for (int64_t base = 0; base < take; base += 64) {
+          int64_t k = base + lane;
+          uint64_t delta = 0;
+          if (k < take)
+            delta = load_bits(d.src + data_pos, d.src_len - data_pos,
+                              k * (int64_t)w, w) + (uint64_t)min_delta;
+          uint64_t scan = delta;
+          for (int off = 1; off < 64; off <<= 1) {
+            uint64_t nb = (uint64_t)__shfl_up((long long)scan, off, 64);
+            if (lane >= off) scan += nb;
+          }
+          uint64_t v = prev + scan;
+          if (k < take) {
+            if (d.width == 8)
+              reinterpret_cast<uint64_t*>(d.out)[produced + k] = v;
+            else
+              reinterpret_cast<uint32_t*>(d.out)[produced + k] = (uint32_t)v;
+          }
+          int64_t lastl = take - base - 1;
+          prev = shfl_u64(v, (int)(lastl < 63 ? lastl : 63));
+        }
+        produced += take;
+        remaining -= take;
+        data_pos += (vpm * (int64_t)w + 7) >> 3;  // miniblocks are padded full
+      }
+      pos = data_pos;
+    }
+    // zero-fill any tail so downstream prefix sums see 0-length entries
+    for (int64_t k = produced + lane; k < d.expected; k += 64) {
+      if (d.width == 8) reinterpret_cast<uint64_t*>(d.out)[k] = 0;
+      else reinterpret_cast<uint32_t*>(d.out)[k] = 0;
+    }
+    if (n == 0 && lane == 0 && d.expected > 0) {
+      if (d.width == 8) reinterpret_cast<uint64_t*>(d.out)[0] = 0;
+      else reinterpret_cast<uint32_t*>(d.out)[0] = 0;
+    }
+    if (lane == 0 && d.consumed) *d.consumed = pos;
+  }
+}
+
+// generic per-page exclusive prefix over int32 lengths -> int64 offsets:
+// val_len[v] = lens[v] (+ lens2[v]); val_off[v] = extra + *base_off + prefix
+struct StrOffDesc {
+  const int32_t* lens;
+  const int32_t* lens2;     // null unless DELTA_BYTE_ARRAY (prefix+suffix)
+  const int64_t* base_off;  // null or ptr to a device int64 (e.g. consumed)
+  int64_t extra;
+  int64_t num_values;
+  int64_t* val_off;
+  int32_t* val_len;
+  int64_t* total_out;       // null or out: total bytes
+};
+
+__global__ void str_off_kernel(const StrOffDesc* __restrict__ descs,
+                               int32_t npages) {
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  int wpb = blockDim.x >> 6;
+  for (int32_t page = (int32_t)blockIdx.x * wpb + wave; page < npages;
+       page += gridDim.x * wpb) {
+    StrOffDesc d = descs[page];
+    int64_t base = d.extra + (d.base_off ? *d.base_off : 0);
+    int64_t run = 0;
+    for (int64_t b = 0; b < d.num_values; b += 64) {
+      int64_t v = b + lane;
+      int64_t len = 0;
+      if (v < d.num_values) {
+        len = d.lens[v];
+        if (d.lens2) len += d.lens2[v];
+      }
+      uint64_t scan = (uint64_t)len;  // inclusive
+      for (int off = 1; off < 64; off <<= 1) {
+        uint64_t nb = (uint64_t)__shfl_up((long long)scan, off, 64);
+        if (lane >= off) scan += nb;
+      }
+      if (v < d.num_values) {
+        d.val_off[v] = base + run + (int64_t)scan - len;  // exclusive
+        d.val_len[v] = (int32_t)len;
+      }
+      int64_t lastl = d.num_values - b - 1;
+      run += (int64_t)shfl_u64(scan, (int)(lastl < 63 ? lastl : 63));
+    }
+    if (lane == 0 && d.total_out) *d.total_out = run;
+  }
+}
+
+// DELTA_BYTE_ARRAY reconstruction: value v = value[v-1][0:plen[v]] + suffix
+struct DbaDesc {
+  const uint8_t* src;        // page body (suffix bytes addressed absolutely)
+  const int32_t* plen;
+  const int32_t* slen;
+  const int64_t* suf_off;    // per-value absolute offset of suffix in src
+  const int64_t* val_off;    // per-value absolute offset in scratch
+  int64_t num_values;
+  uint8_t* scratch;
+};
+
+__global__ void dba_reconstruct_kernel(const DbaDesc* __restrict__ descs,
+                                       int32_t npages) {
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  int wpb = blockDim.x >> 6;
+  for (int32_t page = (int32_t)blockIdx.x * wpb + wave; page < npages;
+       page += gridDim.x * wpb) {
+    DbaDesc d = descs[page];
+    for (int64_t v = 0; v < d.num_values; ++v) {
+      uint8_t* dst = d.scratch + d.val_off[v];
+      int32_t pl = d.plen[v], sl = d.slen[v];
+      if (v > 0) {
+        const uint8_t* prevp = d.scratch + d.val_off[v - 1];
+        for (int32_t k = lane; k < pl; k += 64) dst[k] = prevp[k];
+      }
+      const uint8_t* suf = d.src + d.suf_off[v];
+      for (int32_t k = lane; k < sl; k += 64) dst[pl + k] = suf[k];
+      __builtin_amdgcn_s_waitcnt(0);  // writes visible before next iteration
+    }
+  }
+}
+
+// BYTE_STREAM_SPLIT: byte plane b of value v at src[b*n + v]
+struct BssDesc {
+  const uint8_t* src;
+  int64_t n;           // value count (body_len / width)
+  uint8_t* out;
+  int32_t width;
+  int32_t pad;
+};
+
+__global__ void bss_kernel(const BssDesc* __restrict__ descs,
+                           int32_t npages) {
+  for (int32_t page = blockIdx.x; page < npages; page += gridDim.x) {
+    BssDesc d = descs[page];
+    for (int64_t v = threadIdx.x; v < d.n; v += blockDim.x)
+      for (int b = 0; b < d.width; ++b)
+        d.out[v * d.width + b] = d.src[(int64_t)b * d.n + v];
+  }
+}
+
 extern "C" {
+
+void srj_pq_delta_binpack(const void* descs, int32_t npages,
+                          hipStream_t stream) {
+  if (npages == 0) return;
+  int wpb = DEFAULT_BLOCK / 64;
+  int blocks = (npages + wpb - 1) / wpb;
+  delta_binpack_kernel<<<blocks < MAX_GRID ? blocks : MAX_GRID, DEFAULT_BLOCK,
+                         0, stream>>>(
+      reinterpret_cast<const DeltaDesc*>(descs), npages);
+}
+
+void srj_pq_str_off(const void* descs, int32_t npages, hipStream_t stream) {
+  if (npages == 0) return;
+  int wpb = DEFAULT_BLOCK / 64;
+  int blocks = (npages + wpb - 1) / wpb;
+  str_off_kernel<<<blocks < MAX_GRID ? blocks : MAX_GRID, DEFAULT_BLOCK, 0,
+                   stream>>>(reinterpret_cast<const StrOffDesc*>(descs),
+                             npages);
+}
+
+void srj_pq_dba_reconstruct(const void* descs, int32_t npages,
+                            hipStream_t stream) {
+  if (npages == 0) return;
+  int wpb = DEFAULT_BLOCK / 64;
+  int blocks = (npages + wpb - 1) / wpb;
+  dba_reconstruct_kernel<<<blocks < MAX_GRID ? blocks : MAX_GRID,
+                           DEFAULT_BLOCK, 0, stream>>>(
+      reinterpret_cast<const DbaDesc*>(descs), npages);
+}
+
+void srj_pq_bss(const void* descs, int32_t npages, hipStream_t stream) {
+  if (npages == 0) return;
+  bss_kernel<<<npages < MAX_GRID ? npages : MAX_GRID, DEFAULT_BLOCK, 0,
+               stream>>>(reinterpret_cast<const BssDesc*>(descs), npages);
+}
 
 void srj_rle_decode(const void* descs, int32_t npages, hipStream_t stream) {
   if (npages == 0) return;
