@@ -784,35 +784,8 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
     from .columnar import TORCH_DTYPE
     numel = (total_rows * 2) if flba_dec else total_rows
     out = torch.zeros(max(numel, 1), dtype=TORCH_DTYPE[dtype], device=dev)
-    # DELTA_BINARY_PACKED / BYTE_STREAM_SPLIT pages decode into a dense
-    # valid-order temp first, then ride the standard scatter as if PLAIN
-    delta_tmp = {}
-    db_descs, bss_descs = bytearray(), bytearray()
-    ndb = nbss = 0
-    for i, p in enumerate(pages):
-        bo = _body_off(i, p)
-        if p.encoding == ENC_DELTA_BINARY_PACKED:
-            nv = max(p.num_values, 1)
-            tmp = torch.empty(nv * width, dtype=torch.uint8, device=dev)
-            delta_tmp[i] = tmp
-            db_descs += struct.pack(_DELTA_FMT, pbase[i] + bo,
-                                    _plen(i) - bo, tmp.data_ptr(),
-                                    p.num_values, 0, width, 0)
-            ndb += 1
-        elif p.encoding == ENC_BYTE_STREAM_SPLIT:
-            nv = (_plen(i) - bo) // width
-            tmp = torch.empty(max(nv * width, 1), dtype=torch.uint8,
-                              device=dev)
-            delta_tmp[i] = tmp
-            bss_descs += struct.pack(_BSS_FMT, pbase[i] + bo, nv,
-                                     tmp.data_ptr(), width, 0)
-            nbss += 1
-    if ndb:
-        t = torch.frombuffer(db_descs, dtype=torch.uint8).to(dev)
-        g.pq_delta_binpack(t.data_ptr(), ndb, stream)
-    if nbss:
-        t = torch.frombuffer(bss_descs, dtype=torch.uint8).to(dev)
-        g.pq_bss(t.data_ptr(), nbss, stream)
+    delta_tmp = _decode_delta_fixed(g, stream, dev, pages, pbase, _plen,
+                                    _body_off, width)
     sc_descs = bytearray()
     for i, p in enumerate(pages):
         bo = _body_off(i, p)
@@ -846,6 +819,42 @@ def _read_column(raw: bytes, f: SchemaField, chunks: List[ColumnChunkMeta],
 
 
 _MMAP_CACHE: Dict[tuple, memoryview] = {}
+
+
+def _decode_delta_fixed(g, stream, dev, pages, pbase, plen_of, body_of,
+                        width):
+    """DELTA_BINARY_PACKED / BYTE_STREAM_SPLIT pages -> dense valid-order
+    temp buffers keyed by page index; the standard scatter then treats
+    them exactly like PLAIN bodies."""
+    tmp = {}
+    db, bss = bytearray(), bytearray()
+    ndb = nbss = 0
+    for i, p in enumerate(pages):
+        if p.kind == 1:
+            continue
+        bo = body_of(i, p)
+        if p.encoding == ENC_DELTA_BINARY_PACKED:
+            nv = max(p.num_values, 1)
+            t_ = torch.empty(nv * width, dtype=torch.uint8, device=dev)
+            tmp[i] = t_
+            db += struct.pack(_DELTA_FMT, pbase[i] + bo, plen_of(i) - bo,
+                              t_.data_ptr(), p.num_values, 0, width, 0)
+            ndb += 1
+        elif p.encoding == ENC_BYTE_STREAM_SPLIT:
+            nv = (plen_of(i) - bo) // width
+            t_ = torch.empty(max(nv * width, 1), dtype=torch.uint8,
+                             device=dev)
+            tmp[i] = t_
+            bss += struct.pack(_BSS_FMT, pbase[i] + bo, nv, t_.data_ptr(),
+                               width, 0)
+            nbss += 1
+    if ndb:
+        t_ = torch.frombuffer(db, dtype=torch.uint8).to(dev)
+        g.pq_delta_binpack(t_.data_ptr(), ndb, stream)
+    if nbss:
+        t_ = torch.frombuffer(bss, dtype=torch.uint8).to(dev)
+        g.pq_bss(t_.data_ptr(), nbss, stream)
+    return tmp
 
 
 def _decode_delta_strings(g, stream, dev, pages, pbase, plen_of, body_of,
@@ -1054,13 +1063,6 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                 dict_per_page.append(cur_dict)
 
     # upload payloads
-    _DELTA_ENCS = (ENC_DELTA_BINARY_PACKED, ENC_DELTA_LENGTH_BYTE_ARRAY,
-                   ENC_DELTA_BYTE_ARRAY, ENC_BYTE_STREAM_SPLIT)
-    for p_ in pages:
-        if p_.encoding in _DELTA_ENCS:
-            raise NotImplementedError(
-                "DELTA/BYTE_STREAM_SPLIT encodings inside nested columns "
-                "are not supported yet (flat columns decode them)")
     blobs = [p.data for p in pages] + [p.data for p in dicts]
     offs = np.zeros(len(blobs) + 1, dtype=np.int64)
     for i, b in enumerate(blobs):
@@ -1229,6 +1231,12 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
             st = torch.frombuffer(sidx_descs, dtype=torch.uint8).to(dev)
             g.pq_string_plain_index(st.data_ptr(), len(plain_meta), stream)
         pm = {i: (voff, vlen) for i, voff, vlen in plain_meta}
+        delta_str, _dh = _decode_delta_strings(
+            g, stream, dev, pages, pbase, lambda j: len(pages[j].data),
+            lambda j, pp: body_offs[j],
+            [i for i, p in enumerate(pages)
+             if p.encoding in (ENC_DELTA_LENGTH_BYTE_ARRAY,
+                               ENC_DELTA_BYTE_ARRAY)])
         cp_descs = bytearray()
         for i, p in enumerate(pages):
             if p.encoding == ENC_PLAIN:
@@ -1237,6 +1245,12 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                     _STRCPY_FMT, pbase[i] + body_offs[i], voff.data_ptr(),
                     vlen.data_ptr(), 0, def_ptr, vprefix_ptr, elem_starts[i],
                     elems_in_page[i], vbase[i])
+            elif i in delta_str:
+                sp, voff, vlen = delta_str[i]
+                cp_descs += struct.pack(
+                    _STRCPY_FMT, sp, voff.data_ptr(), vlen.data_ptr(), 0,
+                    def_ptr, vprefix_ptr, elem_starts[i], elems_in_page[i],
+                    vbase[i])
             else:
                 di = dict_per_page[i]
                 base, voff, vlen = dict_str[di]
@@ -1265,10 +1279,18 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
         edt = _field_dtype(elem)
         out = torch.zeros(max(total_elems, 1), dtype=TORCH_DTYPE[edt],
                           device=dev)
+        delta_tmp = _decode_delta_fixed(g, stream, dev, pages, pbase,
+                                        lambda j: len(pages[j].data),
+                                        lambda j, pp: body_offs[j], width)
         sc_descs = bytearray()
         for i, p in enumerate(pages):
             if p.encoding == ENC_PLAIN:
                 sc_descs += struct.pack(_SCATTER_FMT, pbase[i] + body_offs[i],
+                                        0, def_ptr, vprefix_ptr,
+                                        elem_starts[i], elems_in_page[i],
+                                        vbase[i], width, 0)
+            elif i in delta_tmp:
+                sc_descs += struct.pack(_SCATTER_FMT, delta_tmp[i].data_ptr(),
                                         0, def_ptr, vprefix_ptr,
                                         elem_starts[i], elems_in_page[i],
                                         vbase[i], width, 0)
@@ -1349,13 +1371,6 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                 else:
                     pages.append(p)
                     dict_per_page.append(cur_dict)
-        _DELTA_ENCS = (ENC_DELTA_BINARY_PACKED, ENC_DELTA_LENGTH_BYTE_ARRAY,
-                           ENC_DELTA_BYTE_ARRAY, ENC_BYTE_STREAM_SPLIT)
-        for p_ in pages:
-            if p_.encoding in _DELTA_ENCS:
-                raise NotImplementedError(
-                    "DELTA/BYTE_STREAM_SPLIT encodings inside nested columns "
-                    "are not supported yet (flat columns decode them)")
         blobs = [p.data for p in pages] + [p.data for p in dicts]
         offs = np.zeros(len(blobs) + 1, dtype=np.int64)
         for i, b in enumerate(blobs):
@@ -1479,6 +1494,12 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                 g.pq_string_plain_index(st.data_ptr(), len(plain_meta),
                                         stream)
             pm = {i: (voff, vlen) for i, voff, vlen in plain_meta}
+            delta_str, _dh = _decode_delta_strings(
+                g, stream, dev, pages, pbase, lambda j: len(pages[j].data),
+                lambda j, pp: body_offs[j],
+                [i for i, p in enumerate(pages)
+                 if p.encoding in (ENC_DELTA_LENGTH_BYTE_ARRAY,
+                                   ENC_DELTA_BYTE_ARRAY)])
             cp_descs = bytearray()
             for i, p in enumerate(pages):
                 if p.encoding == ENC_PLAIN:
@@ -1487,6 +1508,12 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
                         _STRCPY_FMT, pbase[i] + body_offs[i], voff.data_ptr(),
                         vlen.data_ptr(), 0, def_ptr, vprefix_ptr,
                         int(row_starts[i]), p.num_values, int(vbase[i]))
+                elif i in delta_str:
+                    sp, voff, vlen = delta_str[i]
+                    cp_descs += struct.pack(
+                        _STRCPY_FMT, sp, voff.data_ptr(), vlen.data_ptr(), 0,
+                        def_ptr, vprefix_ptr, int(row_starts[i]),
+                        p.num_values, int(vbase[i]))
                 else:
                     di = dict_per_page[i]
                     base, voff, vlen = dict_str[di]
@@ -1519,11 +1546,20 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
             numel = (total_rows * 2) if flba else total_rows
             out = torch.zeros(max(numel, 1), dtype=TORCH_DTYPE[ldt],
                               device=dev)
+            delta_tmp = _decode_delta_fixed(
+                g, stream, dev, pages, pbase, lambda j: len(pages[j].data),
+                lambda j, pp: body_offs[j], width)
             sc_descs = bytearray()
             for i, p in enumerate(pages):
                 if p.encoding == ENC_PLAIN:
                     sc_descs += struct.pack(_SCATTER_FMT,
                                             pbase[i] + body_offs[i], 0,
+                                            def_ptr, vprefix_ptr,
+                                            int(row_starts[i]), p.num_values,
+                                            int(vbase[i]), width, 0)
+                elif i in delta_tmp:
+                    sc_descs += struct.pack(_SCATTER_FMT,
+                                            delta_tmp[i].data_ptr(), 0,
                                             def_ptr, vprefix_ptr,
                                             int(row_starts[i]), p.num_values,
                                             int(vbase[i]), width, 0)

@@ -823,3 +823,45 @@ def test_decode_byte_stream_split(tmp_path):
         assert (g_ is None) == (e_ is None) and (g_ is None or g_ == e_)
     for g_, e_ in zip(got.columns[1].to_pylist(), f64):
         assert (g_ is None) == (e_ is None) and (g_ is None or g_ == e_)
+
+
+@pytest.mark.gpu
+def test_decode_delta_nested(tmp_path):
+    """DELTA encodings inside LIST and STRUCT columns ride the same
+    temp-buffer composition as flat pages."""
+    rng = random.Random(83)
+    ll, ls, st_rows = [], [], []
+    cur = 0
+    for i in range(2200):
+        if i % 13 == 4:
+            ll.append(None)
+            ls.append(None)
+            st_rows.append(None)
+            continue
+        row = []
+        for k in range(rng.randrange(0, 4)):
+            cur += rng.randint(-5, 9)
+            row.append(None if (i + k) % 7 == 2 else cur)
+        ll.append(row)
+        ls.append([None if (i + k) % 8 == 3 else
+                   f"pre{(i + k) % 5}_" + str(rng.randrange(500))
+                   for k in range(rng.randrange(0, 3))])
+        st_rows.append({"a": None if i % 9 == 5 else i * 3 - 1000,
+                        "b": None if i % 6 == 1 else f"w{(i * 7) % 90}"})
+    t = pa.table({
+        "ll": pa.array(ll, type=pa.list_(pa.int64())),
+        "ls": pa.array(ls, type=pa.list_(pa.string())),
+        "st": pa.array(st_rows, type=pa.struct([("a", pa.int64()),
+                                                ("b", pa.string())]))})
+    p = str(tmp_path / "nd.parquet")
+    pq.write_table(t, p, compression="NONE", use_dictionary=False,
+                   data_page_size=4096, row_group_size=800,
+                   column_encoding={"ll.list.element": "DELTA_BINARY_PACKED",
+                                    "ls.list.element": "DELTA_BYTE_ARRAY",
+                                    "st.a": "DELTA_BINARY_PACKED",
+                                    "st.b": "DELTA_LENGTH_BYTE_ARRAY"})
+    got = srj_pq.read_table(p, device="cuda")
+    assert got.columns[0].to_pylist() == ll
+    assert got.columns[1].to_pylist() == ls
+    exp_st = [None if r is None else (r["a"], r["b"]) for r in st_rows]
+    assert got.columns[2].to_pylist() == exp_st
