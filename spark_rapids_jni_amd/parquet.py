@@ -1042,8 +1042,6 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
     if elem.is_struct:
         raise NotImplementedError(
             "STRUCT below more than one LIST level is not supported yet")
-    if elem.physical_type == T_FIXED_LEN_BYTE_ARRAY:
-        raise NotImplementedError("LIST of FLBA decimals not supported")
     max_def = f.max_def
     elem_nullable = max_def > chain[-1][1]
     rep_bw = max(len(chain).bit_length(), 1)
@@ -1274,10 +1272,12 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
         child = Column(DType.STRING, total_elems, chars[:nchars],
                        elem_validity, soffs, null_count=None)
     else:
-        width = _PHYS_WIDTH[elem.physical_type]
+        flba = elem.physical_type == T_FIXED_LEN_BYTE_ARRAY
+        width = elem.type_length if flba else _PHYS_WIDTH[elem.physical_type]
         from .columnar import TORCH_DTYPE
         edt = _field_dtype(elem)
-        out = torch.zeros(max(total_elems, 1), dtype=TORCH_DTYPE[edt],
+        numel = (total_elems * 2) if flba else total_elems
+        out = torch.zeros(max(numel, 1), dtype=TORCH_DTYPE[edt],
                           device=dev)
         delta_tmp = _decode_delta_fixed(g, stream, dev, pages, pbase,
                                         lambda j: len(pages[j].data),
@@ -1303,8 +1303,14 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                                         elems_in_page[i], vbase[i], width, 1)
         st = torch.frombuffer(sc_descs or bytearray(1),
                               dtype=torch.uint8).to(dev)
-        g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(), stream)
-        child = Column(edt, total_elems, out[:max(total_elems, 1)],
+        if flba:
+            # big-endian W-byte decimals -> 2x int64 words per element
+            g.pq_flba_dec128(st.data_ptr(), len(pages), out.data_ptr(),
+                             stream)
+        else:
+            g.pq_scatter_fixed(st.data_ptr(), len(pages), out.data_ptr(),
+                               stream)
+        child = Column(edt, total_elems, out[:max(numel, 1)],
                        elem_validity, scale=elem.scale, null_count=None)
 
     # ---- wrap the offsets chain inner->outer ------------------------------

@@ -865,3 +865,29 @@ def test_decode_delta_nested(tmp_path):
     assert got.columns[1].to_pylist() == ls
     exp_st = [None if r is None else (r["a"], r["b"]) for r in st_rows]
     assert got.columns[2].to_pylist() == exp_st
+
+
+@pytest.mark.gpu
+def test_decode_list_of_flba_decimal(tmp_path):
+    """LIST<DECIMAL128 (FIXED_LEN_BYTE_ARRAY)>: big-endian N-byte decimals
+    inside list elements -> 2x int64 words, same kernel as the flat path."""
+    from decimal import Decimal
+    rng = random.Random(89)
+    rows = []
+    for i in range(1800):
+        if i % 11 == 3:
+            rows.append(None)
+            continue
+        rows.append([None if (i + k) % 7 == 2 else
+                     Decimal((i - 900) * 10**19 + k * 37).scaleb(-4)
+                     for k in range(rng.randrange(0, 4))])
+    t = pa.table({"ld": pa.array(rows, type=pa.list_(pa.decimal128(38, 4)))})
+    p = str(tmp_path / "lflba.parquet")
+    pq.write_table(t, p, compression="NONE", row_group_size=700)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+    child = col.children[0]
+    assert child.dtype.name == "DECIMAL128" and child.scale == 4
+    exp = [None if r is None else
+           [None if v is None else int(v.scaleb(4)) for v in r]
+           for r in rows]
+    assert col.to_pylist() == exp
