@@ -224,10 +224,10 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                 raise NotImplementedError("MAP with a nested KEY type "
                                           "is not supported")
             d_elem = d_above + outer_opt + 1
-            key = leaf_field(kse, 0)
+            key = leaf_field(kse, d_elem)
             vse = schema_elems[i + 3]
             if vse.get(5, 0) == 0:
-                val = leaf_field(vse, 0)
+                val = leaf_field(vse, d_elem)
                 j = i + 4
             else:
                 # nested value: LIST or STRUCT, parsed with the map chain's
@@ -240,23 +240,13 @@ def read_footer(path_or_bytes) -> ParquetFooter:
                 name=se[4].decode(), physical_type=-1,
                 repetition=se.get(3, 0), converted_type=2, is_list=False,
                 is_map=True, map_key=key, map_value=val,
-                max_def=d_above + outer_opt + 1, max_rep=1), j
+                max_def=d_above + outer_opt + 1, max_rep=1,
+                d_base=d_above), j
         # plain group = STRUCT; children may be leaves or nested STRUCTs
         children = []
         j = i + 1
         for _ in range(nch):
             ch, j = parse_field(j, d_above + outer_opt)
-            if ch.is_map:
-                raise NotImplementedError(
-                    "MAP nested inside STRUCT is not supported yet")
-            if ch.is_list:
-                tip = ch.element
-                while tip.is_list:
-                    tip = tip.element
-                if tip.is_struct:
-                    raise NotImplementedError(
-                        "LIST of STRUCT nested inside STRUCT "
-                        "is not supported yet")
             children.append(ch)
         return SchemaField(
             name=se[4].decode(), physical_type=-1,
@@ -1024,7 +1014,7 @@ def _list_chain(f: SchemaField):
 
 def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
                       total_rows: int, device, return_elem_def=False,
-                      return_row_def=False):
+                      return_row_def=False, return_defs=False):
     """LIST decode (parquet 3-level encoding), any repetition depth:
     LIST<leaf>, LIST<LIST<...<leaf>>>, and (via _read_list_struct_column's
     per-leaf synthetic fields) the leaves of LIST<STRUCT<...>>.
@@ -1318,6 +1308,8 @@ def _read_list_column(raw, f: SchemaField, chunks: List[ColumnChunkMeta],
     for (np_l, offsets_l, validity_l, _ns) in reversed(levels):
         col = Column(DType.LIST, np_l, None, validity_l, offsets_l, [col],
                      null_count=None)
+    if return_defs:
+        return col, def64[elem_mask], def64[row_mask]
     if return_elem_def:
         return col, def64[elem_mask]
     if return_row_def:
@@ -1352,14 +1344,35 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
     children = []
     leaf_levs = []  # per-leaf int64 def levels (row space), None if max_def==0
 
+    cursor = leaf0
     for li, leaf in enumerate(leaves):
-        chunks = [rg.columns[leaf0 + li] for rg in row_groups]
-        if leaf.is_list:
-            # LIST child of a STRUCT: the full list machinery applies with
-            # the struct chain's def levels as the base; its row-space def
+        base0 = cursor
+        cursor += _field_chunk_count(leaf)
+        chunks = [rg.columns[base0] for rg in row_groups]
+        if leaf.is_map:
+            # MAP child of a STRUCT: the map reader runs with the struct
+            # chain's def levels as the base; its key-side row-space def
             # levels double as the ancestor-struct validity source.
-            lcol, row_def = _read_list_column(raw, leaf, chunks, total_rows,
-                                              device, return_row_def=True)
+            mcol, row_def = _read_map_column(raw, leaf, row_groups, base0,
+                                             total_rows, device,
+                                             return_row_def=True)
+            children.append(mcol)
+            leaf_levs.append(row_def)
+            continue
+        if leaf.is_list:
+            tip = leaf.element
+            while tip.is_list:
+                tip = tip.element
+            if tip.is_struct:
+                lcol, row_def = _read_list_struct_column(
+                    raw, leaf, row_groups, base0, total_rows, device,
+                    return_row_def=True)
+            else:
+                # LIST child of a STRUCT: the full list machinery applies
+                # with the struct chain's def levels as the base
+                lcol, row_def = _read_list_column(raw, leaf, chunks,
+                                                  total_rows, device,
+                                                  return_row_def=True)
             children.append(lcol)
             leaf_levs.append(row_def)
             continue
@@ -1616,7 +1629,7 @@ def _read_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
 
 
 def _read_list_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
-                             total_rows: int, device) -> Column:
+                             total_rows: int, device, return_row_def=False):
     """LIST<STRUCT<...>> decode by composition (like MAP): every struct leaf
     has exactly the level structure of a LIST of that leaf — same repetition
     chain, deeper definition chain — so each decodes through the generic
@@ -1627,16 +1640,24 @@ def _read_list_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
     (def >= node.max_def means that struct level is present)."""
     elem = f.element
     leaves = _flatten_struct_leaves(elem)
+    if any(lf.is_map or lf.is_list for lf in leaves):
+        raise NotImplementedError(
+            "MAP/LIST inside a repeated STRUCT element is not supported yet")
     cols = []
     edefs = []
+    row_def = None
     for li, leaf in enumerate(leaves):
         chunks = [rg.columns[leaf0 + li] for rg in row_groups]
         synth = SchemaField(
             name=f.name, physical_type=-1, repetition=f.repetition,
             converted_type=3, is_list=True, element=leaf,
-            max_def=leaf.max_def, max_rep=1)
-        col, edef = _read_list_column(raw, synth, chunks, total_rows, device,
-                                      return_elem_def=True)
+            max_def=leaf.max_def, max_rep=1, d_base=f.d_base)
+        if li == 0:
+            col, edef, row_def = _read_list_column(
+                raw, synth, chunks, total_rows, device, return_defs=True)
+        else:
+            col, edef = _read_list_column(raw, synth, chunks, total_rows,
+                                          device, return_elem_def=True)
         cols.append(col)
         edefs.append(edef)
 
@@ -1659,26 +1680,28 @@ def _read_list_struct_column(raw, f: SchemaField, row_groups, leaf0: int,
 
     entries, used = assemble(elem, 0)
     assert used == len(leaves)
-    return Column(DType.LIST, total_rows, None, cols[0].validity,
+    lcol = Column(DType.LIST, total_rows, None, cols[0].validity,
                   cols[0].offsets, [entries], null_count=None)
+    if return_row_def:
+        return lcol, row_def
+    return lcol
 
 
 def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
-                     total_rows: int, device) -> Column:
+                     total_rows: int, device, return_row_def=False):
     """MAP decode by composition: each of the key/value leaves has exactly
     the level structure of a 3-level LIST of that leaf (outer optional
     group + repeated key_value), so both decode through the LIST machinery
     and zip into LIST<STRUCT<key, value>> — the Spark map layout. The
     entry offsets come from the key side (identical on both by
     construction)."""
-    outer_opt = 1 if f.repetition == 1 else 0
     key_f = SchemaField(
         name=f.name, physical_type=-1, repetition=f.repetition,
         converted_type=3, is_list=True, element=f.map_key,
-        max_def=outer_opt + 1 + (1 if f.map_key.repetition == 1 else 0),
-        max_rep=1)
+        max_def=f.map_key.max_def, max_rep=1, d_base=f.d_base)
     kchunks = [rg.columns[leaf0] for rg in row_groups]
-    klist = _read_list_column(raw, key_f, kchunks, total_rows, device)
+    klist, _ke, row_def = _read_list_column(raw, key_f, kchunks, total_rows,
+                                            device, return_defs=True)
     mv = f.map_value
     if mv.is_struct:
         # MAP<k, STRUCT<...>>: the value side is exactly LIST<STRUCT> with
@@ -1687,7 +1710,7 @@ def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
             name=f.name, physical_type=-1, repetition=f.repetition,
             converted_type=3, is_list=True, element=mv,
             max_def=max(lf.max_def for lf in _flatten_struct_leaves(mv)),
-            max_rep=1)
+            max_rep=1, d_base=f.d_base)
         vlist = _read_list_struct_column(raw, synth, row_groups, leaf0 + 1,
                                          total_rows, device)
     elif mv.is_list:
@@ -1696,28 +1719,44 @@ def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
         synth = SchemaField(
             name=f.name, physical_type=-1, repetition=f.repetition,
             converted_type=3, is_list=True, element=mv, max_def=mv.max_def,
-            max_rep=mv.max_rep + 1)
+            max_rep=mv.max_rep + 1, d_base=f.d_base)
         vchunks = [rg.columns[leaf0 + 1] for rg in row_groups]
         vlist = _read_list_column(raw, synth, vchunks, total_rows, device)
     else:
         val_f = SchemaField(
             name=f.name, physical_type=-1, repetition=f.repetition,
             converted_type=3, is_list=True, element=mv,
-            max_def=outer_opt + 1 + (1 if mv.repetition == 1 else 0),
-            max_rep=1)
+            max_def=mv.max_def, max_rep=1, d_base=f.d_base)
         vchunks = [rg.columns[leaf0 + 1] for rg in row_groups]
         vlist = _read_list_column(raw, val_f, vchunks, total_rows, device)
     nentries = klist.children[0].size
     entries = Column(DType.STRUCT, nentries, None, None, None,
                      [klist.children[0], vlist.children[0]],
                      null_count=None)
-    return Column(DType.LIST, total_rows, None, klist.validity,
+    mcol = Column(DType.LIST, total_rows, None, klist.validity,
                   klist.offsets, [entries], null_count=None)
+    if return_row_def:
+        return mcol, row_def
+    return mcol
 
 
 def _map_leaf_count(f: SchemaField) -> int:
     mv = f.map_value
     return 1 + (len(_flatten_struct_leaves(mv)) if mv.is_struct else 1)
+
+
+def _field_chunk_count(f: SchemaField) -> int:
+    """Number of parquet column chunks (leaf columns) a field spans."""
+    if f.is_map:
+        return _map_leaf_count(f)
+    if f.is_list:
+        tip = f.element
+        while tip.is_list:
+            tip = tip.element
+        return len(_flatten_struct_leaves(tip)) if tip.is_struct else 1
+    if f.is_struct:
+        return sum(_field_chunk_count(c) for c in f.children)
+    return 1
 
 
 def read_table(path: str, columns: Optional[Sequence[str]] = None,
@@ -1739,17 +1778,17 @@ def read_table(path: str, columns: Optional[Sequence[str]] = None,
         if f.is_struct:
             cols.append(_read_struct_column(raw, f, footer.row_groups, leaf,
                                             total_rows, device))
-            leaf += len(_flatten_struct_leaves(f))
+            leaf += _field_chunk_count(f)
             continue
         if f.is_map:
             cols.append(_read_map_column(raw, f, footer.row_groups, leaf,
                                          total_rows, device))
-            leaf += _map_leaf_count(f)
+            leaf += _field_chunk_count(f)
             continue
         if f.is_list and f.element.is_struct:
             cols.append(_read_list_struct_column(
                 raw, f, footer.row_groups, leaf, total_rows, device))
-            leaf += len(_flatten_struct_leaves(f.element))
+            leaf += _field_chunk_count(f)
             continue
         chunks = [rg.columns[leaf] for rg in footer.row_groups]
         if f.is_list:

@@ -891,3 +891,41 @@ def test_decode_list_of_flba_decimal(tmp_path):
            [None if v is None else int(v.scaleb(4)) for v in r]
            for r in rows]
     assert col.to_pylist() == exp
+
+
+@pytest.mark.gpu
+def test_decode_struct_with_map_and_liststruct(tmp_path):
+    """MAP and LIST<STRUCT> nested INSIDE a struct: chunk-cursor routing +
+    ancestor validity from the nested child's row-space def levels."""
+    rng = random.Random(97)
+    rows = []
+    for i in range(1600):
+        if i % 13 == 5:
+            rows.append(None)
+            continue
+        m = (None if i % 9 == 2 else
+             [(f"k{j}", None if (i + j) % 6 == 1 else i * 2 + j)
+              for j in range(rng.randrange(0, 3))])
+        ls = (None if i % 7 == 3 else
+              [None if (i + j) % 11 == 4 else
+               {"x": None if (i * 3 + j) % 5 == 2 else i + j,
+                "y": f"s{(i + j) % 40}"}
+               for j in range(rng.randrange(0, 3))])
+        rows.append({"m": m, "ls": ls, "z": i % 1000})
+    typ = pa.struct([
+        ("m", pa.map_(pa.string(), pa.int64())),
+        ("ls", pa.list_(pa.struct([("x", pa.int64()), ("y", pa.string())]))),
+        ("z", pa.int32())])
+    t = pa.table({"st": pa.array(rows, type=typ)})
+    p = str(tmp_path / "swml.parquet")
+    pq.write_table(t, p, compression="SNAPPY", row_group_size=600)
+    col = srj_pq.read_table(p, device="cuda").columns[0]
+
+    def conv(r):
+        if r is None:
+            return None
+        ls = (None if r["ls"] is None else
+              [None if e is None else (e["x"], e["y"]) for e in r["ls"]])
+        m = None if r["m"] is None else [tuple(kv) for kv in r["m"]]
+        return (m, ls, r["z"])
+    assert col.to_pylist() == [conv(r) for r in rows]
