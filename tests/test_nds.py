@@ -107,7 +107,7 @@ def test_power_run_subset(catalog):
 
 # --- distributed (world=2, gloo) -------------------------------------------
 
-_DIST_QUERIES = [3, 6, 15, 16, 28, 38, 51, 95]
+_DIST_QUERIES = [2, 3, 6, 15, 16, 28, 38, 51, 59, 67, 78, 95, 97]
 
 _WORKER = r"""
 import os, sys, json
