@@ -24,6 +24,11 @@ void srj_groupby_i64(const long long*, int64_t, void*, int64_t, const void*,
 void srj_groupby_i64_lds(const long long*, int64_t, void*, int64_t,
                          const void*, int32_t, const int64_t*, int32_t*,
                          hipStream_t);
+void srj_groupby_compact_i64_count(const void*, int64_t, int64_t*,
+                                   hipStream_t);
+void srj_groupby_compact_i64_fill(const void*, int64_t, const void*, int32_t,
+                                  const int64_t*, int64_t*, int64_t*, int64_t,
+                                  hipStream_t);
 void srj_groupby_compact_i64(const void*, int64_t, const void*, int32_t,
                              uint64_t*, int64_t*, int64_t*, int64_t,
                              hipStream_t);
@@ -116,6 +121,26 @@ void register_hashtable(py::module_& m) {
                                   as_ptr<int64_t>(out_agg), out_capacity,
                                   as_stream(stream));
           check_hip("groupby_compact_i64");
+        });
+  m.def("groupby_compact_i64_count",
+        [](uintptr_t slots, int64_t capacity1, uintptr_t blk_counts,
+           uintptr_t stream) {
+          srj_groupby_compact_i64_count(as_ptr<void>(slots), capacity1,
+                                        as_ptr<int64_t>(blk_counts),
+                                        as_stream(stream));
+          check_hip("groupby_compact_i64_count");
+        });
+  m.def("groupby_compact_i64_fill",
+        [](uintptr_t slots, int64_t capacity1, uintptr_t aggs, int32_t naggs,
+           uintptr_t blk_bases, uintptr_t out_repr, uintptr_t out_agg,
+           int64_t out_capacity, uintptr_t stream) {
+          srj_groupby_compact_i64_fill(as_ptr<void>(slots), capacity1,
+                                       as_ptr<void>(aggs), naggs,
+                                       as_ptr<int64_t>(blk_bases),
+                                       as_ptr<int64_t>(out_repr),
+                                       as_ptr<int64_t>(out_agg), out_capacity,
+                                       as_stream(stream));
+          check_hip("groupby_compact_i64_fill");
         });
   m.def("join_probe_i64",
         [](uintptr_t probe, uintptr_t pvalid, int64_t nprobe, uintptr_t slots,
