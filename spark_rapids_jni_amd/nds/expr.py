@@ -19,12 +19,13 @@ import torch
 class Val:
     """An evaluated expression: data tensor + optional validity (bool, True =
     valid) + optional string dictionary (data holds codes)."""
-    __slots__ = ("data", "valid", "dict")
+    __slots__ = ("data", "valid", "dict", "_all_true")
 
     def __init__(self, data, valid=None, dict=None):
         self.data = data
         self.valid = valid
         self.dict = dict
+        self._all_true = None  # memo: valid mask known all-true (1 sync max)
 
     def valid_mask(self):
         if self.valid is None:

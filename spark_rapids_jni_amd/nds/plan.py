@@ -197,10 +197,33 @@ def _val_to_column(v: Val):
     # count_valid atomic stream AND global aggregates lose the LDS
     # pre-aggregation path (naggs<=3 limit) — q97's final agg was 2x255 ms
     # on a single contended CAS slot because of exactly this
-    if valid is not None and data.numel() and bool(valid.all().item()):
-        valid = None
+    if valid is not None and data.numel():
+        at = getattr(v, "_all_true", None)
+        if at is None:
+            at = bool(valid.all().item())
+            try:
+                v._all_true = at
+            except AttributeError:
+                pass
+        if at:
+            valid = None
     return Column(dt, data.numel(), data, _pack_validity(valid),
                   null_count=None)
+
+
+def _precheck_all_true(vals):
+    """Resolve the all-true memo for many Vals in ONE device sync."""
+    cand = [v for v in vals
+            if v is not None and v.valid is not None
+            and getattr(v, "_all_true", False) is None and v.data.numel()]
+    if not cand:
+        return
+    flags = torch.stack([v.valid.all() for v in cand]).cpu().tolist()
+    for v, fl in zip(cand, flags):
+        try:
+            v._all_true = bool(fl)
+        except AttributeError:
+            pass
 
 
 class GpuBackend:
@@ -209,6 +232,7 @@ class GpuBackend:
     def join(self, left_keys: List[Val], right_keys: List[Val], how: str,
              nleft: int):
         from ..ops.join import HashJoinTable
+        _precheck_all_true(list(left_keys) + list(right_keys))
         bcols = [_val_to_column(v) for v in right_keys]
         pcols = [_val_to_column(v) for v in left_keys]
         if how in ("semi", "anti"):
@@ -243,6 +267,7 @@ class GpuBackend:
         Returns (key Vals, result Vals)."""
         from ..columnar import validity_to_bool
         from ..ops.aggregate import Agg as A, groupby as gb
+        _precheck_all_true(list(keys) + [v for _fn, v in aggs])
         kcols = [_val_to_column(v) for v in keys]
         fnmap = {"count": A.COUNT_ALL, "count_valid": A.COUNT_VALID,
                  "sum": A.SUM, "min": A.MIN, "max": A.MAX}
@@ -619,12 +644,13 @@ class Engine:
             # hidden count_valid stream inside the kernel, so it costs 2 of
             # the LDS kernel's <=3 agg slots (q97 postmortem: 3 sums over
             # masked values -> naggs 6 -> no LDS -> 255 ms on one CAS slot)
+            _precheck_all_true([v for _f, v in native])
+
             def _cost(fn_v):
                 fn, v = fn_v
                 if fn in ("count", "count_valid") or v is None:
                     return 1
-                if v.valid is None or (v.data.numel()
-                                       and bool(v.valid.all().item())):
+                if v.valid is None or getattr(v, "_all_true", False):
                     return 1
                 return 2
             chunks, cur, budget = [], [], 3

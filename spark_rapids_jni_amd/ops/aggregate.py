@@ -77,9 +77,11 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
         widths = []
         mins = []
         ok = True
-        for c in kcols:
-            lo, hi = torch.aminmax(c.data)
-            lo, hi = int(lo.item()), int(hi.item())
+        pairs = [torch.aminmax(c.data) for c in kcols]
+        flat = torch.stack(
+            [t for lo_hi in pairs for t in lo_hi]).cpu().tolist()
+        for k in range(len(kcols)):  # ONE device sync for all columns
+            lo, hi = int(flat[2 * k]), int(flat[2 * k + 1])
             span = hi - lo + 2  # +1 for the null code
             if span <= 0:
                 ok = False
@@ -217,7 +219,23 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
             g.groupby(kdesc.data_ptr(), ktop.data_ptr(), len(kcols), n,
                       slots.data_ptr(), capacity, agg_desc.data_ptr(),
                       naggs, overflow.data_ptr(), stream)
-        if not int(overflow.item()):
+        if i64_fast:
+            # speculative compact-count: enqueue the two-pass compaction's
+            # counting half WITH the groupby so the overflow flag and the
+            # group count come back in ONE device sync (host gaps, not
+            # kernel time, dominate the NDS power run at this point)
+            cap1 = capacity + 1
+            nblk = max(1, min((cap1 + 255) // 256, 2048))  # mirrors grid_1d
+            blk = torch.empty(nblk, dtype=torch.int64, device=dev)
+            g.groupby_compact_i64_count(slots.data_ptr(), cap1,
+                                        blk.data_ptr(), stream)
+            csum = torch.cumsum(blk, 0)
+            probe = torch.cat([overflow.to(torch.int64),
+                               csum[-1:]]).cpu().tolist()
+            ovf, ngroups = int(probe[0]), int(probe[1])
+        else:
+            ovf = int(overflow.item())
+        if not ovf:
             break
         assert capacity < full_cap, "groupby overflow at full capacity"
         # a second guess would fill to ~100% load before overflowing again
@@ -231,22 +249,15 @@ def groupby(keys, aggs: Sequence[Tuple[Agg, Optional[Column]]],
     out_agg = torch.empty(max(naggs, 1) * nstates, dtype=torch.int64,
                           device=dev)
     if i64_fast:
-        # two-pass compaction: per-block occupancy counts -> cumsum bases ->
-        # LDS-cursor fill. The single global counter a wave-leader atomicAdd
-        # hammered was the bottleneck on big tables (101 ms on a 512M-slot
-        # scan); this runs at slot-scan bandwidth and is deterministic.
-        cap1 = capacity + 1
-        nblk = max(1, min((cap1 + 255) // 256, 2048))  # mirrors grid_1d
-        blk = torch.empty(nblk, dtype=torch.int64, device=dev)
-        g.groupby_compact_i64_count(slots.data_ptr(), cap1, blk.data_ptr(),
-                                    stream)
-        csum = torch.cumsum(blk, 0)
+        # two-pass compaction fill (counting half ran inside the loop): the
+        # single global counter a wave-leader atomicAdd hammered was the
+        # bottleneck on big tables (101 ms on a 512M-slot scan); per-block
+        # bases + an LDS cursor run at slot-scan bandwidth, deterministic.
         bases = csum - blk
         g.groupby_compact_i64_fill(slots.data_ptr(), cap1,
                                    agg_desc.data_ptr(), naggs,
                                    bases.data_ptr(), out_repr.data_ptr(),
                                    out_agg.data_ptr(), nstates, stream)
-        ngroups = int(csum[-1].item())
     else:
         counter = torch.zeros(1, dtype=torch.int64, device=dev)
         g.groupby_compact(slots.data_ptr(), capacity, agg_desc.data_ptr(),

@@ -69,11 +69,12 @@ def _pack_ranges(cols: List[Column]):
     if the range product overflows."""
     if cols[0].size == 0:
         return None  # nothing to scan; generic path handles empties
+    pairs = [torch.aminmax(c.data) for c in cols]
+    flat = torch.stack([t for lo_hi in pairs for t in lo_hi]).cpu().tolist()
     mins, widths = [], []
     total = 1
-    for c in cols:
-        lo, hi = torch.aminmax(c.data)
-        lo, hi = int(lo.item()), int(hi.item())
+    for k in range(len(cols)):  # ONE device sync for all columns
+        lo, hi = int(flat[2 * k]), int(flat[2 * k + 1])
         span = hi - lo + 1
         if span <= 0:
             return None
