@@ -1741,19 +1741,22 @@ def _read_map_column(raw, f: SchemaField, row_groups, leaf0: int,
 
 
 def _map_leaf_count(f: SchemaField) -> int:
-    mv = f.map_value
-    return 1 + (len(_flatten_struct_leaves(mv)) if mv.is_struct else 1)
+    return 1 + _field_chunk_count(f.map_value)
 
 
 def _field_chunk_count(f: SchemaField) -> int:
-    """Number of parquet column chunks (leaf columns) a field spans."""
+    """Number of parquet column chunks (leaf columns) a field spans.
+
+    Recurses fully — a MAP inside a list-element struct spans 2 chunks,
+    not 1, so flattened-leaf counting is NOT equivalent (caught by
+    tests/test_parquet.py::test_footer_random_nested_schemas)."""
     if f.is_map:
-        return _map_leaf_count(f)
+        return 1 + _field_chunk_count(f.map_value)
     if f.is_list:
         tip = f.element
         while tip.is_list:
             tip = tip.element
-        return len(_flatten_struct_leaves(tip)) if tip.is_struct else 1
+        return _field_chunk_count(tip) if tip.is_struct else 1
     if f.is_struct:
         return sum(_field_chunk_count(c) for c in f.children)
     return 1
