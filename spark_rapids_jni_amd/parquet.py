@@ -1744,6 +1744,33 @@ def _map_leaf_count(f: SchemaField) -> int:
     return 1 + _field_chunk_count(f.map_value)
 
 
+def _leaf_levels(f: SchemaField, rep: int = 0):
+    """(max_def, max_rep) per leaf chunk in chunk order — the parser's
+    absolute level accounting, cross-checked against pyarrow's column
+    descriptors by the randomized schema test."""
+    if f.is_map:
+        yield (f.map_key.max_def, rep + 1)
+        yield from _leaf_levels(f.map_value, rep + 1)
+        return
+    if f.is_list:
+        depth = 1
+        tip = f.element
+        while tip.is_list:
+            tip = tip.element
+            depth += 1
+        if tip.is_struct:
+            for c in tip.children:
+                yield from _leaf_levels(c, rep + depth)
+            return
+        yield (f.max_def, rep + depth)
+        return
+    if f.is_struct:
+        for c in f.children:
+            yield from _leaf_levels(c, rep)
+        return
+    yield (f.max_def, rep)
+
+
 def _field_chunk_count(f: SchemaField) -> int:
     """Number of parquet column chunks (leaf columns) a field spans.
 

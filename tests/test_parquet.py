@@ -963,7 +963,7 @@ def test_footer_random_nested_schemas(tmp_path):
         return 1
 
     parsed = skipped = 0
-    for trial in range(60):
+    for trial in range(150):
         typ = rand_type(0)
         t = pa.table({"c": pa.array([dummy(typ), None], type=typ)})
         p = str(tmp_path / f"r{trial}.parquet")
@@ -976,6 +976,13 @@ def test_footer_random_nested_schemas(tmp_path):
             continue
         got = sum(srj_pq._field_chunk_count(fl) for fl in f.schema)
         assert got == len(f.row_groups[0].columns), (trial, typ)
+        # absolute def/rep levels must match pyarrow's column descriptors
+        sch = pq.ParquetFile(p).schema
+        exp = [(sch.column(i).max_definition_level,
+                sch.column(i).max_repetition_level)
+               for i in range(len(sch))]
+        ours = [lv for fl in f.schema for lv in srj_pq._leaf_levels(fl)]
+        assert ours == exp, (trial, typ, ours, exp)
         parsed += 1
     # the generator must exercise both outcomes
-    assert parsed >= 20 and skipped >= 1, (parsed, skipped)
+    assert parsed >= 60 and skipped >= 1, (parsed, skipped)
