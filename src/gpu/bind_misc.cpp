@@ -19,7 +19,8 @@ void srj_uuid(int64_t, uint64_t, char*, hipStream_t);
 void srj_substring_index(const void*, const char*, int32_t, int32_t, int64_t,
                          int32_t, int32_t*, const int32_t*, char*, uint8_t*,
                          hipStream_t);
-void srj_literal_range(const void*, const char*, int32_t, int32_t, char, char,
+void srj_literal_range(const void*, const char*, int32_t, int32_t,
+                       uint32_t, uint32_t,
                        int64_t, uint8_t*, uint8_t*, hipStream_t);
 void srj_extract_chunk32(const int64_t*, const uint8_t*, int64_t, int32_t,
                          int64_t*, hipStream_t);
@@ -97,7 +98,7 @@ void register_misc(py::module_& m) {
                             int32_t range_len, int32_t rs, int32_t re, int64_t n,
                             uintptr_t out, uintptr_t valid, uintptr_t stream) {
     srj_literal_range(as_ptr<void>(in), as_ptr<char>(lit), lit_len, range_len,
-                      (char)rs, (char)re, n, as_ptr<uint8_t>(out),
+                      (uint32_t)rs, (uint32_t)re, n, as_ptr<uint8_t>(out),
                       as_ptr<uint8_t>(valid), as_stream(stream));
     check_hip("literal_range");
   });

@@ -338,9 +338,36 @@ __global__ void substring_index_kernel(ColDesc in, const char* __restrict__ deli
 // literal_range_pattern: does the string contain <literal><len chars in
 // [range_start..range_end]> (regex-free rewrite of %lit[a-b]{n}% patterns)
 // ---------------------------------------------------------------------------
+// UTF-8 codepoint decode (advances i past the sequence); malformed bytes
+// decode as U+FFFD one byte at a time
+__device__ inline uint32_t lr_utf8_decode(const char* p, int32_t len,
+                                          int32_t& i) {
+  uint8_t c0 = (uint8_t)p[i];
+  if (c0 < 0x80) { i += 1; return c0; }
+  if ((c0 >> 5) == 0x6 && i + 1 < len) {
+    uint32_t v = ((uint32_t)(c0 & 0x1F) << 6) | ((uint8_t)p[i + 1] & 0x3F);
+    i += 2; return v;
+  }
+  if ((c0 >> 4) == 0xE && i + 2 < len) {
+    uint32_t v = ((uint32_t)(c0 & 0x0F) << 12) |
+                 (((uint8_t)p[i + 1] & 0x3F) << 6) |
+                 ((uint8_t)p[i + 2] & 0x3F);
+    i += 3; return v;
+  }
+  if ((c0 >> 3) == 0x1E && i + 3 < len) {
+    uint32_t v = ((uint32_t)(c0 & 0x07) << 18) |
+                 (((uint8_t)p[i + 1] & 0x3F) << 12) |
+                 (((uint8_t)p[i + 2] & 0x3F) << 6) |
+                 ((uint8_t)p[i + 3] & 0x3F);
+    i += 4; return v;
+  }
+  i += 1;
+  return 0xFFFD;
+}
+
 __global__ void literal_range_kernel(ColDesc in, const char* __restrict__ lit,
                                      int32_t lit_len, int32_t range_len,
-                                     char range_start, char range_end,
+                                     uint32_t range_start, uint32_t range_end,
                                      int64_t nrows, uint8_t* __restrict__ out,
                                      uint8_t* __restrict__ out_valid) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -351,16 +378,28 @@ __global__ void literal_range_kernel(ColDesc in, const char* __restrict__ lit,
     bool valid = in_range && is_valid(in.valid, row);
     bool found = false;
     if (valid) {
+      // codepoint semantics (reference regex_rewrite_utils.cu operates on
+      // cudf's codepoint iterator): the range chars after the literal are
+      // CODEPOINTS in [range_start..range_end] — e.g. a CJK class — and
+      // candidate starts advance codepoint-wise
       StrView s = get_string(in, row);
-      for (int32_t i = 0; i + lit_len + range_len <= s.len && !found; ++i) {
+      int32_t i = 0;
+      while (i + lit_len <= s.len && !found) {
         bool m = true;
         for (int32_t k = 0; k < lit_len && m; ++k)
           if (s.ptr[i + k] != lit[k]) m = false;
-        for (int32_t k = 0; k < range_len && m; ++k) {
-          char c = s.ptr[i + lit_len + k];
-          if (c < range_start || c > range_end) m = false;
+        if (m) {
+          int32_t j = i + lit_len;
+          for (int32_t k = 0; k < range_len && m; ++k) {
+            if (j >= s.len) { m = false; break; }
+            uint32_t cp = lr_utf8_decode(s.ptr, s.len, j);
+            if (cp < range_start || cp > range_end) m = false;
+          }
+          found = m;
         }
-        found = m;
+        int32_t adv = i;
+        lr_utf8_decode(s.ptr, s.len, adv);
+        i = adv;
       }
     }
     if (in_range) out[row] = found;
@@ -644,9 +683,9 @@ void srj_substring_index(const void* in, const char* delim, int32_t delim_len,
 }
 
 void srj_literal_range(const void* in, const char* lit, int32_t lit_len,
-                       int32_t range_len, char range_start, char range_end,
-                       int64_t nrows, uint8_t* out, uint8_t* out_valid,
-                       hipStream_t stream) {
+                       int32_t range_len, uint32_t range_start,
+                       uint32_t range_end, int64_t nrows, uint8_t* out,
+                       uint8_t* out_valid, hipStream_t stream) {
   literal_range_kernel<<<grid_1d(nrows), DEFAULT_BLOCK, 0, stream>>>(
       *reinterpret_cast<const ColDesc*>(in), lit, lit_len, range_len, range_start,
       range_end, nrows, out, out_valid);

@@ -108,6 +108,15 @@ def test_iceberg_transforms():
     b = iceberg_bucket(col, 16).to_pylist()
     assert b[0] == (2017239379 & 0x7FFFFFFF) % 16
     assert b[1] is None
+    # spec Appendix B: date 2017-11-16 (17486 days) hashes to -653330422,
+    # timestamp 2017-11-16T22:31:08 (micros) to -2047944441 — both via the
+    # int64 path (dates/timestamps bucket as longs)
+    dcol2 = Column.from_pylist([17486], DType.DATE32, "cuda")
+    assert iceberg_bucket(dcol2, 128).to_pylist() == \
+        [(-653330422 & 0x7FFFFFFF) % 128]
+    tcol = Column.from_pylist([1510871468000000], DType.TIMESTAMP_US, "cuda")
+    assert iceberg_bucket(tcol, 128).to_pylist() == \
+        [(-2047944441 & 0x7FFFFFFF) % 128]
     # truncate: W=10: 11->10, -11->-20 (floored)
     t = iceberg_truncate(Column.from_pylist([11, -11, 0], DType.INT64, "cuda"),
                          10).to_pylist()

@@ -157,6 +157,15 @@ def test_substring_index():
         ["apache.org", "a.b", "nodot", "", None, "c.d"]
     assert substring_index(col, ".", 0).to_pylist() == \
         ["", "", "", "", None, ""]
+    # reference GpuSubstringIndexUtilsTest vectors: count beyond the number
+    # of delimiters returns the whole string; multi-char and multi-byte
+    # UTF-8 delimiters count as units
+    assert substring_index(col, ".", 3).to_pylist()[0] == "www.apache.org"
+    assert substring_index(col, ".", -3).to_pylist()[0] == "www.apache.org"
+    cn = Column.from_pylist(["大千世界大千世界"], DType.STRING, "cuda")
+    assert substring_index(cn, "千", 2).to_pylist() == ["大千世界大"]
+    bars = Column.from_pylist(["www||apache||org"], DType.STRING, "cuda")
+    assert substring_index(bars, "||", 2).to_pylist() == ["www||apache"]
 
 
 @pytest.mark.gpu
@@ -299,3 +308,19 @@ def test_sort_merge_join_matches_hash_join():
     hb, hp = HashJoinTable.build(b).inner_join(p)
     assert set(zip(sb.cpu().tolist(), sp.cpu().tolist())) == \
         set(zip(hb.cpu().tolist(), hp.cpu().tolist()))
+
+
+@pytest.mark.gpu
+def test_literal_range_pattern_codepoints():
+    """Reference RegexRewriteUtilsTest vectors: digit class and a CJK
+    codepoint class (19968..40869) — ranges are CODEPOINTS, not bytes."""
+    from spark_rapids_jni_amd.ops.misc import literal_range_pattern
+    col = Column.from_pylist(["abc123", "aabc123", "aabc12", "abc1232",
+                              "aabc1232"], DType.STRING, "cuda")
+    got = literal_range_pattern(col, "abc", 3, "0", "9").to_pylist()
+    assert got == [True, True, False, True, True]
+    cn = Column.from_pylist(["数据砖块", "火花-急流英伟达", "英伟达Nvidia",
+                             "火花-急流"], DType.STRING, "cuda")
+    got = literal_range_pattern(cn, "英", 2, chr(19968),
+                                chr(40869)).to_pylist()
+    assert got == [False, True, True, False]
