@@ -21,13 +21,10 @@
 
 namespace srj {
 
+#include "fp_parse.inc"
+
 __device__ inline bool is_space(char c) { return (unsigned char)c <= ' '; }
 
-__device__ inline StrView trim_all(StrView s) {
-  while (s.len > 0 && is_space(s.ptr[0])) { ++s.ptr; --s.len; }
-  while (s.len > 0 && is_space(s.ptr[s.len - 1])) --s.len;
-  return s;
-}
 
 __device__ inline void record_error(int64_t* err_row, int64_t row) {
   if (err_row)
@@ -138,203 +135,6 @@ __global__ void string_to_bool_kernel(ColDesc in, int64_t nrows,
     ballot_write_validity(out_valid, row, valid);
   }
 }
-
-// ---------------------------------------------------------------------------
-// string -> float
-// ---------------------------------------------------------------------------
-__device__ inline double pow10_pos(int e) {
-  // exact for e <= 22; split larger exponents
-  static const double tbl[23] = {1e0, 1e1, 1e2, 1e3, 1e4, 1e5, 1e6, 1e7, 1e8,
-                                 1e9, 1e10, 1e11, 1e12, 1e13, 1e14, 1e15, 1e16,
-                                 1e17, 1e18, 1e19, 1e20, 1e21, 1e22};
-  double r = 1.0;
-  while (e > 22) { r *= 1e22; e -= 22; }
-  return r * tbl[e];
-}
-
-#include "exact_fp.inc"
-
-struct DecParts {
-  uint64_t mant;
-  int exp_adj;
-  bool neg;
-  bool dropped_nonzero;
-  bool dropped;
-};
-
-// Shared decimal-literal scan: returns 0 invalid, 1 number, 2 inf, 3 nan.
-__device__ int parse_dec_parts(StrView s, DecParts* o) {
-  s = trim_all(s);
-  if (s.len == 0) return 0;
-  int i = 0;
-  bool neg = false;
-  if (s.ptr[0] == '+' || s.ptr[0] == '-') { neg = s.ptr[0] == '-'; i = 1; }
-  // special literals (Spark Cast: infinity/inf/nan, case-insensitive)
-  if (i < s.len) {
-    char c0 = s.ptr[i] | 32;
-    if (c0 == 'i' || c0 == 'n') {
-      char buf[9];
-      int n = s.len - i;
-      if (n <= 8) {
-        for (int k = 0; k < n; ++k) buf[k] = s.ptr[i + k] | 32;
-        auto eq = [&](const char* lit, int ln) {
-          if (n != ln) return false;
-          for (int k = 0; k < ln; ++k)
-            if (buf[k] != lit[k]) return false;
-          return true;
-        };
-        if (eq("inf", 3) || eq("infinity", 8)) {
-          o->neg = neg;
-          return 2;
-        }
-        if (eq("nan", 3)) {
-          return 3;
-        }
-      }
-      if (c0 == 'n' || c0 == 'i') return 0;
-    }
-  }
-  uint64_t mant = 0;
-  int ndig = 0, exp_adj = 0;
-  bool any = false, dot = false, dropped_nonzero = false, dropped = false;
-  for (; i < s.len; ++i) {
-    char c = s.ptr[i];
-    if (c >= '0' && c <= '9') {
-      any = true;
-      if (ndig < 19) {
-        mant = mant * 10 + (c - '0');
-        if (mant) ++ndig;
-        if (dot) --exp_adj;
-      } else {
-        dropped = true;
-        if (c != '0') dropped_nonzero = true;
-        if (!dot) ++exp_adj;
-      }
-    } else if (c == '.') {
-      if (dot) return 0;
-      dot = true;
-    } else if (c == 'e' || c == 'E') {
-      if (!any) return 0;
-      ++i;
-      bool eneg = false;
-      if (i < s.len && (s.ptr[i] == '+' || s.ptr[i] == '-')) {
-        eneg = s.ptr[i] == '-';
-        ++i;
-      }
-      if (i >= s.len) return 0;
-      long e = 0;
-      for (; i < s.len; ++i) {
-        char ec = s.ptr[i];
-        if (ec < '0' || ec > '9') {
-          if ((ec == 'd' || ec == 'D' || ec == 'f' || ec == 'F') &&
-              i == s.len - 1)
-            break;  // Java-style suffix accepted by Spark's parse
-          return 0;
-        }
-        if (e < 100000) e = e * 10 + (ec - '0');
-      }
-      exp_adj += (int)(eneg ? -e : e);
-      break;
-    } else if ((c == 'd' || c == 'D' || c == 'f' || c == 'F') &&
-               i == s.len - 1) {
-      break;  // trailing type suffix
-    } else {
-      return 0;
-    }
-  }
-  if (!any) return 0;
-  o->mant = mant;
-  o->exp_adj = exp_adj;
-  o->neg = neg;
-  o->dropped_nonzero = dropped_nonzero;
-  o->dropped = dropped;
-  return 1;
-}
-
-__device__ inline double approx_pow10_scale(uint64_t mant, int exp_adj,
-                                            bool neg) {
-  // fallback (rare ambiguous cases): within-1-ulp approximation
-  double d = (double)mant;
-  if (exp_adj > 0) {
-    if (exp_adj > 308 + 19) return neg ? -INFINITY : INFINITY;
-    d *= pow10_pos(exp_adj);
-  } else if (exp_adj < 0) {
-    int e = -exp_adj;
-    if (e > 342 + 19) return neg ? -0.0 : 0.0;
-    while (e > 22) { d /= 1e22; e -= 22; }
-    d /= pow10_pos(e);
-  }
-  return neg ? -d : d;
-}
-
-__device__ bool parse_double(StrView s, double* out) {
-  DecParts p;
-  int r = parse_dec_parts(s, &p);
-  if (r == 0) return false;
-  if (r == 2) { *out = p.neg ? -INFINITY : INFINITY; return true; }
-  if (r == 3) { *out = NAN; return true; }
-  // exact path: Eisel-Lemire; bracket w/w+1 when digits were truncated
-  double exact;
-  if (!p.dropped || !p.dropped_nonzero) {
-    if (eisel_lemire(p.mant, p.exp_adj, p.neg, &exact)) {
-      *out = exact;
-      return true;
-    }
-  } else {
-    double lo_v, hi_v;
-    if (eisel_lemire(p.mant, p.exp_adj, p.neg, &lo_v) &&
-        eisel_lemire(p.mant + 1, p.exp_adj, p.neg, &hi_v)) {
-      int64_t lb, hb;
-      __builtin_memcpy(&lb, &lo_v, 8);
-      __builtin_memcpy(&hb, &hi_v, 8);
-      if (lb == hb) {
-        *out = lo_v;
-        return true;
-      }
-    }
-  }
-  *out = approx_pow10_scale(p.mant, p.exp_adj, p.neg);
-  return true;
-}
-
-// Direct exactly-rounded string->float (no double intermediate, so no
-// double-rounding; reference cast_string_to_float.cu parses per-type too).
-__device__ bool parse_float(StrView s, float* out) {
-  DecParts p;
-  int r = parse_dec_parts(s, &p);
-  if (r == 0) return false;
-  if (r == 2) { *out = p.neg ? -INFINITY : INFINITY; return true; }
-  if (r == 3) { *out = NAN; return true; }
-  float exact;
-  if (!p.dropped || !p.dropped_nonzero) {
-    if (eisel_lemire_f32(p.mant, p.exp_adj, p.neg, &exact)) {
-      *out = exact;
-      return true;
-    }
-  } else {
-    float lo_v, hi_v;
-    if (eisel_lemire_f32(p.mant, p.exp_adj, p.neg, &lo_v) &&
-        eisel_lemire_f32(p.mant + 1, p.exp_adj, p.neg, &hi_v)) {
-      int32_t lb, hb;
-      __builtin_memcpy(&lb, &lo_v, 4);
-      __builtin_memcpy(&hb, &hi_v, 4);
-      if (lb == hb) {
-        *out = lo_v;
-        return true;
-      }
-    }
-  }
-  *out = (float)approx_pow10_scale(p.mant, p.exp_adj, p.neg);
-  return true;
-}
-
-__device__ inline bool parse_fp(StrView s, double* out) {
-  return parse_double(s, out);
-}
-__device__ inline bool parse_fp(StrView s, float* out) {
-  return parse_float(s, out);
-}
-
 
 template <typename T>
 __global__ void string_to_float_kernel(ColDesc in, int64_t nrows,

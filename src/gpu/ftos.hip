@@ -15,347 +15,7 @@
 
 namespace srj {
 
-struct Dec64 {
-  uint64_t digits;  // decimal significand, no trailing zeros
-  int32_t exponent; // value = digits * 10^exponent
-};
-
-__device__ inline uint32_t pow5_factor64(uint64_t v) {
-  uint32_t count = 0;
-  while (v > 0) {
-    if (v % 5 != 0) return count;
-    v /= 5;
-    ++count;
-  }
-  return 0;
-}
-
-__device__ inline bool multiple_of_pow5_64(uint64_t v, uint32_t p) {
-  return pow5_factor64(v) >= p;
-}
-
-__device__ inline bool multiple_of_pow2_64(uint64_t v, uint32_t p) {
-  return (v & ((1ull << p) - 1)) == 0;
-}
-
-__device__ inline uint64_t mulshift64(uint64_t m, const unsigned long long* mul,
-                                      int32_t j) {
-  // (m * (mul[1]:mul[0])) >> j, j in (64, 128)
-  uint64_t b0_lo = m * mul[0];
-  uint64_t b0_hi = __umul64hi(m, mul[0]);
-  uint64_t b2_lo = m * mul[1];
-  uint64_t b2_hi = __umul64hi(m, mul[1]);
-  uint64_t mid = b0_hi + b2_lo;
-  uint64_t hi = b2_hi + (mid < b0_hi);
-  // result = (hi:mid) >> (j - 64)
-  int32_t s = j - 64;
-  return (mid >> s) | (hi << (64 - s));
-}
-
-__device__ inline int32_t log10pow2(int32_t e) { return (e * 78913) >> 18; }
-__device__ inline int32_t log10pow5(int32_t e) { return (e * 732923) >> 20; }
-__device__ inline int32_t pow5bits(int32_t e) {
-  return ((e * 1217359) >> 19) + 1;
-}
-
-// core Ryu d2d: ieee mantissa (52 explicit bits) + biased exponent
-__device__ Dec64 ryu_d2d(uint64_t ieee_mantissa, uint32_t ieee_exponent) {
-  int32_t e2;
-  uint64_t m2;
-  if (ieee_exponent == 0) {
-    e2 = 1 - 1023 - 52 - 2;
-    m2 = ieee_mantissa;
-  } else {
-    e2 = (int32_t)ieee_exponent - 1023 - 52 - 2;
-    m2 = (1ull << 52) | ieee_mantissa;
-  }
-  bool even = (m2 & 1) == 0;
-  bool accept_bounds = even;
-
-  uint64_t mv = 4 * m2;
-  uint32_t mm_shift = ieee_mantissa != 0 || ieee_exponent <= 1;
-  uint64_t vr, vp, vm;
-  int32_t e10;
-  bool vm_trailing = false, vr_trailing = false;
-  if (e2 >= 0) {
-    uint32_t q = log10pow2(e2) - (e2 > 3);
-    e10 = (int32_t)q;
-    int32_t k = 125 + pow5bits((int32_t)q) - 1;
-    int32_t i = -e2 + (int32_t)q + k;
-    vr = mulshift64(mv, RYU_D_POW5_INV[q], i);
-    vp = mulshift64(mv + 2, RYU_D_POW5_INV[q], i);
-    vm = mulshift64(mv - 1 - mm_shift, RYU_D_POW5_INV[q], i);
-    if (q <= 21) {
-      if (mv % 5 == 0) {
-        vr_trailing = multiple_of_pow5_64(mv, q);
-      } else if (accept_bounds) {
-        vm_trailing = multiple_of_pow5_64(mv - 1 - mm_shift, q);
-      } else {
-        vp -= multiple_of_pow5_64(mv + 2, q);
-      }
-    }
-  } else {
-    uint32_t q = log10pow5(-e2) - (-e2 > 1);
-    e10 = (int32_t)q + e2;
-    int32_t i = -e2 - (int32_t)q;
-    int32_t k = pow5bits(i) - 125;
-    int32_t j = (int32_t)q - k;
-    vr = mulshift64(mv, RYU_D_POW5[i], j);
-    vp = mulshift64(mv + 2, RYU_D_POW5[i], j);
-    vm = mulshift64(mv - 1 - mm_shift, RYU_D_POW5[i], j);
-    if (q <= 1) {
-      vr_trailing = true;
-      if (accept_bounds) {
-        vm_trailing = mm_shift == 1;
-      } else {
-        --vp;
-      }
-    } else if (q < 63) {
-      vr_trailing = multiple_of_pow2_64(mv, q);
-    }
-  }
-
-  int32_t removed = 0;
-  uint8_t last_removed = 0;
-  uint64_t output;
-  if (vm_trailing || vr_trailing) {
-    while (vp / 10 > vm / 10) {
-      vm_trailing &= vm % 10 == 0;
-      vr_trailing &= last_removed == 0;
-      last_removed = (uint8_t)(vr % 10);
-      vr /= 10; vp /= 10; vm /= 10;
-      ++removed;
-    }
-    if (vm_trailing) {
-      while (vm % 10 == 0) {
-        vr_trailing &= last_removed == 0;
-        last_removed = (uint8_t)(vr % 10);
-        vr /= 10; vp /= 10; vm /= 10;
-        ++removed;
-      }
-    }
-    if (vr_trailing && last_removed == 5 && vr % 2 == 0) {
-      last_removed = 4;  // round even
-    }
-    output = vr + ((vr == vm && (!accept_bounds || !vm_trailing)) ||
-                   last_removed >= 5);
-  } else {
-    bool round_up = false;
-    if (vp / 100 > vm / 100) {
-      round_up = vr % 100 >= 50;
-      vr /= 100; vp /= 100; vm /= 100;
-      removed += 2;
-    }
-    while (vp / 10 > vm / 10) {
-      round_up = vr % 10 >= 5;
-      vr /= 10; vp /= 10; vm /= 10;
-      ++removed;
-    }
-    output = vr + (vr == vm || round_up);
-  }
-  return Dec64{output, e10 + removed};
-}
-
-__device__ Dec64 ryu_f2d(uint32_t ieee_mantissa, uint32_t ieee_exponent) {
-  int32_t e2;
-  uint32_t m2;
-  if (ieee_exponent == 0) {
-    e2 = 1 - 127 - 23 - 2;
-    m2 = ieee_mantissa;
-  } else {
-    e2 = (int32_t)ieee_exponent - 127 - 23 - 2;
-    m2 = (1u << 23) | ieee_mantissa;
-  }
-  bool even = (m2 & 1) == 0;
-  bool accept_bounds = even;
-  uint32_t mv = 4 * m2;
-  uint32_t mp = mv + 2;
-  uint32_t mm_shift = ieee_mantissa != 0 || ieee_exponent <= 1;
-  uint32_t mm = mv - 1 - mm_shift;
-  uint32_t vr, vp, vm;
-  int32_t e10;
-  bool vm_trailing = false, vr_trailing = false;
-  uint8_t last_removed = 0;
-
-  auto mulshift32 = [](uint32_t m, uint64_t factor, int32_t shift) {
-    uint64_t lo = (uint64_t)m * (uint32_t)factor;
-    uint64_t hi = (uint64_t)m * (uint32_t)(factor >> 32);
-    uint64_t sum = (lo >> 32) + hi;
-    return (uint32_t)(sum >> (shift - 32));
-  };
-
-  if (e2 >= 0) {
-    uint32_t q = (uint32_t)log10pow2(e2);
-    e10 = (int32_t)q;
-    int32_t k = 59 + pow5bits((int32_t)q) - 1;
-    int32_t i = -e2 + (int32_t)q + k;
-    vr = mulshift32(mv, RYU_F_POW5_INV[q], i);
-    vp = mulshift32(mp, RYU_F_POW5_INV[q], i);
-    vm = mulshift32(mm, RYU_F_POW5_INV[q], i);
-    if (q != 0 && (vp - 1) / 10 <= vm / 10) {
-      int32_t l = 59 + pow5bits((int32_t)q - 1) - 1;
-      last_removed =
-          (uint8_t)(mulshift32(mv, RYU_F_POW5_INV[q - 1], -e2 + (int32_t)q - 1 + l) % 10);
-    }
-    if (q <= 9) {
-      if (mv % 5 == 0) vr_trailing = multiple_of_pow5_64(mv, q);
-      else if (accept_bounds) vm_trailing = multiple_of_pow5_64(mm, q);
-      else vp -= multiple_of_pow5_64(mp, q);
-    }
-  } else {
-    uint32_t q = (uint32_t)log10pow5(-e2);
-    e10 = (int32_t)q + e2;
-    int32_t i = -e2 - (int32_t)q;
-    int32_t k = pow5bits(i) - 61;
-    int32_t j = (int32_t)q - k;
-    vr = mulshift32(mv, RYU_F_POW5[i], j);
-    vp = mulshift32(mp, RYU_F_POW5[i], j);
-    vm = mulshift32(mm, RYU_F_POW5[i], j);
-    if (q != 0 && (vp - 1) / 10 <= vm / 10) {
-      int32_t jj = (int32_t)q - 1 - (pow5bits(i + 1) - 61);
-      last_removed = (uint8_t)(mulshift32(mv, RYU_F_POW5[i + 1], jj) % 10);
-    }
-    if (q <= 1) {
-      vr_trailing = true;
-      if (accept_bounds) vm_trailing = mm_shift == 1;
-      else --vp;
-    } else if (q < 31) {
-      vr_trailing = multiple_of_pow2_64(mv, q - 1);
-    }
-  }
-
-  int32_t removed = 0;
-  uint32_t output;
-  if (vm_trailing || vr_trailing) {
-    while (vp / 10 > vm / 10) {
-      vm_trailing &= vm % 10 == 0;
-      vr_trailing &= last_removed == 0;
-      last_removed = (uint8_t)(vr % 10);
-      vr /= 10; vp /= 10; vm /= 10;
-      ++removed;
-    }
-    if (vm_trailing) {
-      while (vm % 10 == 0) {
-        vr_trailing &= last_removed == 0;
-        last_removed = (uint8_t)(vr % 10);
-        vr /= 10; vp /= 10; vm /= 10;
-        ++removed;
-      }
-    }
-    if (vr_trailing && last_removed == 5 && vr % 2 == 0) last_removed = 4;
-    output = vr + ((vr == vm && (!accept_bounds || !vm_trailing)) ||
-                   last_removed >= 5);
-  } else {
-    while (vp / 10 > vm / 10) {
-      last_removed = (uint8_t)(vr % 10);
-      vr /= 10; vp /= 10; vm /= 10;
-      ++removed;
-    }
-    output = vr + (vr == vm || last_removed >= 5);
-  }
-  return Dec64{output, e10 + removed};
-}
-
-// ---------------------------------------------------------------------------
-// Java-style formatting of (digits, exponent)
-// ---------------------------------------------------------------------------
-__device__ inline int format_java(uint64_t digits, int32_t exp10, bool neg,
-                                  char* buf) {
-  // value = digits * 10^exp10, digits has no trailing zeros
-  char dig[20];
-  int nd = 0;
-  uint64_t d = digits;
-  do {
-    dig[nd++] = (char)('0' + d % 10);
-    d /= 10;
-  } while (d);
-  // dig is reversed; decimal point position after first digit: value =
-  // 0.d1d2... * 10^(nd + exp10)
-  int32_t point = nd + exp10;  // digits before the decimal point
-  int len = 0;
-  if (neg) buf[len++] = '-';
-  if (point > 0 && point <= 7) {
-    // plain decimal, Java style
-    for (int i = 0; i < point; ++i)
-      buf[len++] = i < nd ? dig[nd - 1 - i] : '0';
-    buf[len++] = '.';
-    if (point >= nd) {
-      buf[len++] = '0';
-    } else {
-      for (int i = point; i < nd; ++i) buf[len++] = dig[nd - 1 - i];
-    }
-  } else if (point <= 0 && point > -3) {
-    buf[len++] = '0';
-    buf[len++] = '.';
-    for (int i = 0; i < -point; ++i) buf[len++] = '0';
-    for (int i = 0; i < nd; ++i) buf[len++] = dig[nd - 1 - i];
-  } else {
-    // scientific: d.dddE[-]x with exponent = point - 1
-    buf[len++] = dig[nd - 1];
-    buf[len++] = '.';
-    if (nd == 1) buf[len++] = '0';
-    else
-      for (int i = 1; i < nd; ++i) buf[len++] = dig[nd - 1 - i];
-    buf[len++] = 'E';
-    int32_t e = point - 1;
-    if (e < 0) {
-      buf[len++] = '-';
-      e = -e;
-    }
-    char ebuf[4];
-    int en = 0;
-    do {
-      ebuf[en++] = (char)('0' + e % 10);
-      e /= 10;
-    } while (e);
-    for (int i = en - 1; i >= 0; --i) buf[len++] = ebuf[i];
-  }
-  return len;
-}
-
-__device__ inline int format_double(double v, char* buf) {
-  uint64_t bits;
-  __builtin_memcpy(&bits, &v, 8);
-  bool neg = bits >> 63;
-  uint32_t e = (uint32_t)((bits >> 52) & 0x7FF);
-  uint64_t m = bits & ((1ull << 52) - 1);
-  if (e == 0x7FF) {
-    const char* s = m ? "NaN" : (neg ? "-Infinity" : "Infinity");
-    int n = 0;
-    while (s[n]) { buf[n] = s[n]; ++n; }
-    return n;
-  }
-  if (e == 0 && m == 0) {
-    int len = 0;
-    if (neg) buf[len++] = '-';
-    buf[len++] = '0'; buf[len++] = '.'; buf[len++] = '0';
-    return len;
-  }
-  Dec64 dec = ryu_d2d(m, e);
-  return format_java(dec.digits, dec.exponent, neg, buf);
-}
-
-__device__ inline int format_float(float v, char* buf) {
-  uint32_t bits;
-  __builtin_memcpy(&bits, &v, 4);
-  bool neg = bits >> 31;
-  uint32_t e = (bits >> 23) & 0xFF;
-  uint32_t m = bits & ((1u << 23) - 1);
-  if (e == 0xFF) {
-    const char* s = m ? "NaN" : (neg ? "-Infinity" : "Infinity");
-    int n = 0;
-    while (s[n]) { buf[n] = s[n]; ++n; }
-    return n;
-  }
-  if (e == 0 && m == 0) {
-    int len = 0;
-    if (neg) buf[len++] = '-';
-    buf[len++] = '0'; buf[len++] = '.'; buf[len++] = '0';
-    return len;
-  }
-  Dec64 dec = ryu_f2d(m, e);
-  return format_java(dec.digits, dec.exponent, neg, buf);
-}
+#include "ryu_format.inc"
 
 template <typename T, bool WRITE>
 __global__ void float_to_string_kernel(const T* __restrict__ in,

@@ -10,8 +10,12 @@
 //     container match -> raw JSON span; multiple wildcard matches ->
 //     "[m1,m2,...]"; missing/invalid -> null.
 #include "srj_common.hpp"
+#include "ryu_tables.inc"
 
 namespace srj {
+
+#include "fp_parse.inc"
+#include "ryu_format.inc"
 
 constexpr int JSON_MAX_DEPTH = 64;   // reference json_parser max nesting
 constexpr int MAX_PATH_DEPTH = 16;   // reference JSONUtils.MAX_PATH_DEPTH
@@ -26,14 +30,16 @@ __device__ inline const char* j_skip_ws(const char* p, const char* e) {
   return p;
 }
 
-// returns end of string token (after closing quote), or null on error
+// returns end of string token (after closing quote), or null on error;
+// accepts " and ' quoting (Spark's lenient parser)
 __device__ inline const char* j_skip_string(const char* p, const char* e) {
   // p points at opening quote
+  char q = *p;
   ++p;
   while (p < e) {
     if (*p == '\\') {
       p += 2;
-    } else if (*p == '"') {
+    } else if (*p == q) {
       return p + 1;
     } else {
       ++p;
@@ -50,7 +56,7 @@ __device__ const char* j_skip_value(const char* p, const char* e) {
   do {
     if (p >= e) return nullptr;
     char c = *p;
-    if (c == '"') {
+    if (c == '"' || c == '\'') {
       p = j_skip_string(p, e);
       if (!p) return nullptr;
     } else if (c == '{' || c == '[') {
@@ -175,7 +181,7 @@ __device__ bool j_match(const char* p, const char* e, MatchCtx& ctx,
     while (true) {
       p = j_skip_ws(p, e);
       if (p < e && *p == '}') return true;
-      if (p >= e || *p != '"') return false;
+      if (p >= e || (*p != '"' && *p != '\'')) return false;
       const char* kend = j_skip_string(p, e);
       if (!kend) return false;
       const char* kbody = p + 1;
@@ -224,7 +230,7 @@ __device__ bool j_match(const char* p, const char* e, MatchCtx& ctx,
     while (true) {
       p = j_skip_ws(p, e);
       if (p < e && *p == '}') return true;
-      if (p >= e || *p != '"') return false;
+      if (p >= e || (*p != '"' && *p != '\'')) return false;
       const char* kend = j_skip_string(p, e);
       if (!kend) return false;
       p = j_skip_ws(kend, e);
@@ -243,8 +249,250 @@ __device__ bool j_match(const char* p, const char* e, MatchCtx& ctx,
   return true;  // structure mismatch: no match, not an error
 }
 
-__device__ void emit_match(Sink& sink, JsonSpan m, bool as_element) {
-  if (m.len >= 2 && m.p[0] == '"' && !as_element) {
+
+// ---------------------------------------------------------------------------
+// normalized re-serialization: the reference emits results through its JSON
+// generator — whitespace stripped, strings double-quoted with canonical
+// escapes (\uXXXX decoded to UTF-8, surrogate pairs combined), numbers
+// through the exact parse + Java Double.toString pipeline (shared
+// fp_parse.inc / ryu_format.inc), single-quoted input accepted, numbers
+// with leading zeros rejected (GetJsonObjectTest_Number_Normalization,
+// _Escape, _Test_leading_zeros).
+// ---------------------------------------------------------------------------
+
+__device__ inline void j_emit_char(Sink& sink, char c, bool as_json) {
+  if (!as_json) {
+    sink.put(c);
+    return;
+  }
+  switch (c) {
+    case '"': sink.put('\\'); sink.put('"'); return;
+    case '\\': sink.put('\\'); sink.put('\\'); return;
+    case '\n': sink.put('\\'); sink.put('n'); return;
+    case '\t': sink.put('\\'); sink.put('t'); return;
+    case '\r': sink.put('\\'); sink.put('r'); return;
+    case '\b': sink.put('\\'); sink.put('b'); return;
+    case '\f': sink.put('\\'); sink.put('f'); return;
+    default:
+      if ((unsigned char)c < 0x20) {
+        const char* hx = "0123456789abcdef";
+        sink.put('\\'); sink.put('u'); sink.put('0'); sink.put('0');
+        sink.put(hx[((unsigned char)c >> 4) & 15]);
+        sink.put(hx[(unsigned char)c & 15]);
+      } else {
+        sink.put(c);
+      }
+  }
+}
+
+__device__ inline bool j_hex4(const char* p, const char* e, uint32_t* out) {
+  if (p + 4 > e) return false;
+  uint32_t v = 0;
+  for (int k = 0; k < 4; ++k) {
+    char h = p[k];
+    uint32_t d;
+    if (h >= '0' && h <= '9') d = h - '0';
+    else if ((h | 32) >= 'a' && (h | 32) <= 'f') d = (h | 32) - 'a' + 10;
+    else return false;
+    v = v * 16 + d;
+  }
+  *out = v;
+  return true;
+}
+
+// p at the opening quote; emits the normalized string (JSON-escaped when
+// as_json, raw bytes for a top-level string result); advances p past the
+// closing quote
+__device__ bool j_str_norm(Sink& sink, const char*& p, const char* e,
+                           bool as_json) {
+  char q = *p;
+  ++p;
+  if (as_json) sink.put('"');
+  while (p < e && *p != q) {
+    char c = *p;
+    if (c == '\\') {
+      if (p + 1 >= e) return false;
+      char x = p[1];
+      p += 2;
+      switch (x) {
+        case 'n': j_emit_char(sink, '\n', as_json); break;
+        case 't': j_emit_char(sink, '\t', as_json); break;
+        case 'r': j_emit_char(sink, '\r', as_json); break;
+        case 'b': j_emit_char(sink, '\b', as_json); break;
+        case 'f': j_emit_char(sink, '\f', as_json); break;
+        case '/': j_emit_char(sink, '/', as_json); break;
+        case '"': j_emit_char(sink, '"', as_json); break;
+        case '\'': j_emit_char(sink, '\'', as_json); break;
+        case '\\': j_emit_char(sink, '\\', as_json); break;
+        case 'u': {
+          uint32_t cp;
+          if (!j_hex4(p, e, &cp)) return false;
+          p += 4;
+          if (cp >= 0xD800 && cp < 0xDC00 && p + 6 <= e && p[0] == '\\' &&
+              p[1] == 'u') {
+            uint32_t lo;
+            if (j_hex4(p + 2, e, &lo) && lo >= 0xDC00 && lo < 0xE000) {
+              cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
+              p += 6;
+            }
+          }
+          if (cp < 0x80) {
+            j_emit_char(sink, (char)cp, as_json);
+          } else if (cp < 0x800) {
+            sink.put((char)(0xC0 | (cp >> 6)));
+            sink.put((char)(0x80 | (cp & 0x3F)));
+          } else if (cp < 0x10000) {
+            sink.put((char)(0xE0 | (cp >> 12)));
+            sink.put((char)(0x80 | ((cp >> 6) & 0x3F)));
+            sink.put((char)(0x80 | (cp & 0x3F)));
+          } else {
+            sink.put((char)(0xF0 | (cp >> 18)));
+            sink.put((char)(0x80 | ((cp >> 12) & 0x3F)));
+            sink.put((char)(0x80 | ((cp >> 6) & 0x3F)));
+            sink.put((char)(0x80 | (cp & 0x3F)));
+          }
+          break;
+        }
+        default: j_emit_char(sink, x, as_json);
+      }
+    } else {
+      j_emit_char(sink, c, as_json);
+      ++p;
+    }
+  }
+  if (p >= e) return false;
+  ++p;
+  if (as_json) sink.put('"');
+  return true;
+}
+
+// number token: integers verbatim (-0 -> 0; leading zeros invalid), anything
+// with . or e/E through exact parse + Java Double.toString; +/-Infinity and
+// NaN become quoted strings (Spark emits them as JSON strings)
+__device__ bool j_num_norm(Sink& sink, const char*& p, const char* e) {
+  const char* st = p;
+  if (p < e && *p == '-') ++p;
+  const char* ds = p;
+  while (p < e && *p >= '0' && *p <= '9') ++p;
+  int32_t nint = (int32_t)(p - ds);
+  if (nint == 0) return false;
+  if (nint > 1 && ds[0] == '0') return false;  // leading zeros invalid
+  bool isflt = false;
+  if (p < e && *p == '.') {
+    isflt = true;
+    ++p;
+    const char* fs = p;
+    while (p < e && *p >= '0' && *p <= '9') ++p;
+    if (p == fs) return false;
+  }
+  if (p < e && (*p == 'e' || *p == 'E')) {
+    isflt = true;
+    ++p;
+    if (p < e && (*p == '+' || *p == '-')) ++p;
+    const char* es = p;
+    while (p < e && *p >= '0' && *p <= '9') ++p;
+    if (p == es) return false;
+  }
+  if (!isflt) {
+    if (st[0] == '-' && nint == 1 && ds[0] == '0') {
+      sink.put('0');  // "-0" normalizes to 0
+      return true;
+    }
+    sink.span(st, (int32_t)(p - st));
+    return true;
+  }
+  double d;
+  StrView sv{st, (int32_t)(p - st)};
+  if (!parse_double(sv, &d)) return false;
+  char buf[40];
+  int n = format_double(d, buf);
+  bool special = buf[0] == 'N' || buf[n - 1] == 'y';  // NaN / ...Infinity
+  if (special) {
+    sink.put('"');
+    sink.span(buf, n);
+    sink.put('"');
+  } else {
+    sink.span(buf, n);
+  }
+  return true;
+}
+
+// whole-value re-serializer; advances p past the value
+__device__ bool emit_norm_value(Sink& sink, const char*& p, const char* e,
+                                int depth) {
+  p = j_skip_ws(p, e);
+  if (p >= e || depth > JSON_MAX_DEPTH) return false;
+  char c = *p;
+  if (c == '"' || c == '\'') return j_str_norm(sink, p, e, true);
+  if (c == '{') {
+    ++p;
+    sink.put('{');
+    p = j_skip_ws(p, e);
+    if (p < e && *p == '}') { ++p; sink.put('}'); return true; }
+    while (true) {
+      p = j_skip_ws(p, e);
+      if (p >= e || (*p != '"' && *p != '\'')) return false;
+      if (!j_str_norm(sink, p, e, true)) return false;
+      p = j_skip_ws(p, e);
+      if (p >= e || *p != ':') return false;
+      ++p;
+      sink.put(':');
+      if (!emit_norm_value(sink, p, e, depth + 1)) return false;
+      p = j_skip_ws(p, e);
+      if (p < e && *p == ',') { ++p; sink.put(','); continue; }
+      if (p < e && *p == '}') { ++p; sink.put('}'); return true; }
+      return false;
+    }
+  }
+  if (c == '[') {
+    ++p;
+    sink.put('[');
+    p = j_skip_ws(p, e);
+    if (p < e && *p == ']') { ++p; sink.put(']'); return true; }
+    while (true) {
+      if (!emit_norm_value(sink, p, e, depth + 1)) return false;
+      p = j_skip_ws(p, e);
+      if (p < e && *p == ',') { ++p; sink.put(','); continue; }
+      if (p < e && *p == ']') { ++p; sink.put(']'); return true; }
+      return false;
+    }
+  }
+  if (c == 't' && e - p >= 4 && p[1] == 'r' && p[2] == 'u' && p[3] == 'e') {
+    sink.span(p, 4);
+    p += 4;
+    return true;
+  }
+  if (c == 'f' && e - p >= 5 && p[1] == 'a' && p[2] == 'l' && p[3] == 's' &&
+      p[4] == 'e') {
+    sink.span(p, 5);
+    p += 5;
+    return true;
+  }
+  if (c == 'n' && e - p >= 4 && p[1] == 'u' && p[2] == 'l' && p[3] == 'l') {
+    sink.span(p, 4);
+    p += 4;
+    return true;
+  }
+  return j_num_norm(sink, p, e);
+}
+
+// emit one matched span through the normalizer; false = invalid token
+__device__ bool emit_match(Sink& sink, JsonSpan m, bool as_element) {
+  const char* p = m.p;
+  const char* e = m.p + m.len;
+  if (m.len >= 2 && (m.p[0] == '"' || m.p[0] == '\'') && !as_element) {
+    return j_str_norm(sink, p, e, false);
+  }
+  return emit_norm_value(sink, p, e, 0);
+}
+
+// from_json paths keep the row and fall back to the raw span when the
+// normalizer rejects a token (deterministic across the measure/write phases)
+__device__ void emit_match_or_raw(Sink& sink, JsonSpan m, bool as_element) {
+  Sink probe{nullptr, 0, false};
+  if (emit_match(probe, m, as_element)) {
+    emit_match(sink, m, as_element);
+  } else if (m.len >= 2 && (m.p[0] == '"' || m.p[0] == '\'') && !as_element) {
     sink.unescaped(m.p + 1, m.len - 2);
   } else {
     sink.span(m.p, m.len);
@@ -282,17 +530,30 @@ __global__ void get_json_object_kernel(ColDesc in, int64_t nrows,
             m.p[3] == 'l') {
           valid = false;
         } else {
-          emit_match(sink, m, false);
-          out_len = sink.pos;
+          Sink probe{nullptr, 0, false};
+          if (!emit_match(probe, m, false)) {
+            valid = false;
+          } else {
+            emit_match(sink, m, false);
+            out_len = sink.pos;
+          }
         }
       } else {
-        sink.put('[');
-        for (int32_t i = 0; i < ctx.nmatches; ++i) {
-          if (i) sink.put(',');
-          emit_match(sink, ctx.matches[i], true);
+        Sink probe{nullptr, 0, false};
+        bool eok = true;
+        for (int32_t i = 0; i < ctx.nmatches && eok; ++i)
+          eok = emit_match(probe, ctx.matches[i], true);
+        if (!eok) {
+          valid = false;
+        } else {
+          sink.put('[');
+          for (int32_t i = 0; i < ctx.nmatches; ++i) {
+            if (i) sink.put(',');
+            emit_match(sink, ctx.matches[i], true);
+          }
+          sink.put(']');
+          out_len = sink.pos;
         }
-        sink.put(']');
-        out_len = sink.pos;
       }
     }
     if (WRITE) {
@@ -356,7 +617,7 @@ __device__ bool j_match_multi(const char* p, const char* e, MultiCtx& ctx,
     while (true) {
       p = j_skip_ws(p, e);
       if (p < e && *p == '}') return true;
-      if (p >= e || *p != '"') return false;
+      if (p >= e || (*p != '"' && *p != '\'')) return false;
       const char* kend = j_skip_string(p, e);
       if (!kend) return false;
       const char* kbody = p + 1;
@@ -469,17 +730,30 @@ __global__ void get_json_multi_kernel(ColDesc in, int64_t nrows,
               m.p[3] == 'l') {
             valid = false;
           } else {
-            emit_match(sink, m, false);
-            out_len = sink.pos;
+            Sink probe{nullptr, 0, false};
+            if (!emit_match(probe, m, false)) {
+              valid = false;
+            } else {
+              emit_match(sink, m, false);
+              out_len = sink.pos;
+            }
           }
         } else {
-          sink.put('[');
-          for (int32_t k = 0; k < (int32_t)ctx.nmatches[i]; ++k) {
-            if (k) sink.put(',');
-            emit_match(sink, ctx.matches[i][k], true);
+          Sink probe{nullptr, 0, false};
+          bool eok = true;
+          for (int32_t k = 0; k < (int32_t)ctx.nmatches[i] && eok; ++k)
+            eok = emit_match(probe, ctx.matches[i][k], true);
+          if (!eok) {
+            valid = false;
+          } else {
+            sink.put('[');
+            for (int32_t k = 0; k < (int32_t)ctx.nmatches[i]; ++k) {
+              if (k) sink.put(',');
+              emit_match(sink, ctx.matches[i][k], true);
+            }
+            sink.put(']');
+            out_len = sink.pos;
           }
-          sink.put(']');
-          out_len = sink.pos;
         }
       }
       if (ctx.overflow[i] && in_range) atomicOr(o.overflow + 0, 1);
@@ -526,7 +800,7 @@ __global__ void json_to_map_kernel(ColDesc in, int64_t nrows,
         while (valid) {
           p = j_skip_ws(p, e);
           if (p < e && *p == '}') break;
-          if (p >= e || *p != '"') { valid = false; break; }
+          if (p >= e || (*p != '"' && *p != '\'')) { valid = false; break; }
           const char* kend = j_skip_string(p, e);
           if (!kend) { valid = false; break; }
           const char* kbody = p + 1;
@@ -545,14 +819,14 @@ __global__ void json_to_map_kernel(ColDesc in, int64_t nrows,
             ks.unescaped(kbody, klen);
             Sink vs{val_chars + val_offsets[ei], 0, true};
             JsonSpan m{vstart, (int32_t)(vend - vstart)};
-            emit_match(vs, m, false);
+            emit_match_or_raw(vs, m, false);
           } else {
             // measure
             Sink ks{nullptr, 0, false};
             ks.unescaped(kbody, klen);
             Sink vs{nullptr, 0, false};
             JsonSpan m{vstart, (int32_t)(vend - vstart)};
-            emit_match(vs, m, false);
+            emit_match_or_raw(vs, m, false);
             if (key_lens) key_lens[in_range ? row : 0] = 0;  // placeholder
           }
           ++nent;
@@ -587,7 +861,7 @@ __global__ void json_map_entry_lens_kernel(ColDesc in, int64_t nrows,
     while (true) {
       p = j_skip_ws(p, e);
       if (p >= e || *p == '}') break;
-      if (*p != '"') break;
+      if (*p != '"' && *p != '\'') break;
       const char* kend = j_skip_string(p, e);
       if (!kend) break;
       Sink ks{nullptr, 0, false};
@@ -599,7 +873,7 @@ __global__ void json_map_entry_lens_kernel(ColDesc in, int64_t nrows,
       const char* vend = j_skip_value(p, e);
       if (!vend) break;
       Sink vs{nullptr, 0, false};
-      emit_match(vs, {p, (int32_t)(vend - p)}, false);
+      emit_match_or_raw(vs, {p, (int32_t)(vend - p)}, false);
       key_lens[ei] = ks.pos;
       val_lens[ei] = vs.pos;
       ++ei;

@@ -167,3 +167,54 @@ def test_from_json_nested_struct():
             assert not b.is_valid_host(i)
         else:
             assert xs[i] == i * 2 and ys[i] == f"s{i}"
+
+
+@pytest.mark.gpu
+def test_get_json_object_normalization():
+    """Reference GetJsonObjectTest Escape/Number_Normalization/leading-zeros
+    vectors: output is re-serialized through the JSON generator — whitespace
+    stripped, single quotes accepted, strings re-escaped canonically,
+    numbers through the Java Double.toString formatter."""
+    from spark_rapids_jni_amd.ops.json import get_json_object
+
+    bs = "\\"
+    doc5 = ("'" + bs + "u4e2d" + bs + "u56FD" + bs + '"' + bs + "'" +
+            bs + bs + bs + "/" + bs + "b" + bs + "f" + bs + "n" + bs + "r" +
+            bs + "t" + bs + "b'")
+    exp5 = '中国"\'' + bs + "/\b\f\n\r\t\b"
+    doc6 = "[" + doc5 + "]"
+    exp6 = ('["中国' + bs + '"' + "'" + bs + bs + "/" + bs + "b" + bs + "f" +
+            bs + "n" + bs + "r" + bs + "t" + bs + 'b"]')
+
+    cases = [
+        ('{ "a": "A" }', '{"a":"A"}'),
+        ("{'a':'A\"'}", '{"a":"A' + bs + '""}'),
+        ("{'a':\"B'\"}", '{"a":"B\'"}'),
+        ("['a','b','\"C\"']", '["a","b","' + bs + '"C' + bs + '""]'),
+        (doc5, exp5),
+        (doc6, exp6),
+        # number normalization
+        ("[100.0,200.000,351.980]", "[100.0,200.0,351.98]"),
+        ("[12345678900000000000.0]", "[1.23456789E19]"),
+        ("[0.0]", "[0.0]"),
+        ("[-0.0]", "[-0.0]"),
+        ("[-0]", "[0]"),
+        ("[12345678999999999999999999]", "[12345678999999999999999999]"),
+        ("[9.299999257686047e-0005603333574677677]", "[0.0]"),
+        ("9.299999257686047e0005603333574677677", '"Infinity"'),
+        ("[1E308]", "[1.0E308]"),
+        ("[1.0E309,-1E309,1E5000]",
+         '["Infinity","-Infinity","Infinity"]'),
+        ("0.3", "0.3"),
+        ("0.03", "0.03"),
+        ("0.003", "0.003"),
+        ("0.0003", "3.0E-4"),
+        ("0.00003", "3.0E-5"),
+        # leading zeros invalid
+        ("00", None), ("01", None), ("02", None), ("000", None),
+        ("-01", None), ("-00", None), ("-02", None),
+    ]
+    col = Column.from_pylist([d for d, _ in cases], DType.STRING, "cuda")
+    got = get_json_object(col, "$").to_pylist()
+    for (doc, exp), g in zip(cases, got):
+        assert g == exp, (doc, g, exp)
