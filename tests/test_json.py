@@ -39,7 +39,7 @@ DOCS = [
 @pytest.mark.gpu
 @pytest.mark.parametrize("path,expect", [
     ("$.a", ["b", "1", '[{"x":1},{"x":2},{"y":3}]', "line\nbreak é",
-             None, None, None, None, None, '{"deep": {"nest": "v"}}', None]),
+             None, None, None, None, None, '{"deep":{"nest":"v"}}', None]),
     ("$.b.c[1]", [None, "20", None, None, None, None, None, None, None, None,
                   None]),
     ("$.a[*].x", [None, None, "[1,2]", None, None, None, None, None, None,
