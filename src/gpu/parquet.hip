@@ -420,6 +420,7 @@ __device__ inline int64_t dbp_zigzag(uint64_t u) {
 __device__ inline uint64_t load_bits(const uint8_t* p, int64_t plen,
                                      int64_t bitoff, int w) {
   if (w <= 0) return 0;
+  if (w > 64) w = 64;  // malformed miniblock width: clamp (no UB shifts)
   int64_t b0 = bitoff >> 3;
   int shift = (int)(bitoff & 7);
   unsigned __int128 acc = 0;
