@@ -217,3 +217,28 @@ def test_rollup_cascade_matches_bruteforce(catalog):
         for key, (s, any_s, mx, c) in acc.items():
             exp[key] = (round(s, 4) if any_s else None, mx, c)
     assert got == exp
+
+
+def test_day_pivot_decomposition_equivalent(catalog):
+    """day_pivot_sums (two-stage q2/q59 rewrite) must produce EXACTLY the
+    single-stage sum(case_when(d_day_name=X)) aggregation it replaced."""
+    from spark_rapids_jni_amd.nds.expr import case_when, col, lit
+    from spark_rapids_jni_amd.nds.queries import A, J, S, day_pivot_sums
+
+    days = ["Sunday", "Monday", "Tuesday", "Wednesday", "Thursday",
+            "Friday", "Saturday"]
+    child = J(S("store_sales", "ss_sold_date_sk", "ss_store_sk",
+                "ss_sales_price"),
+              S("date_dim", "d_date_sk", "d_week_seq", "d_day_name"),
+              [("ss_sold_date_sk", "d_date_sk")])
+    keys = ["d_week_seq", "ss_store_sk"]
+    e1 = Engine(catalog, device="cpu")
+    direct = e1.run(A(child, keys,
+                      [(dy.lower()[:3] + "_sales", "sum",
+                        case_when((col("d_day_name") == dy,
+                                   col("ss_sales_price")),
+                                  otherwise=lit(0.0))) for dy in days]))
+    e2 = Engine(catalog, device="cpu")
+    pivot = e2.run(day_pivot_sums(child, keys, col("ss_sales_price"), days))
+    assert frame_sorted_rows(direct) == frame_sorted_rows(pivot)
+    assert direct.nrows > 0
