@@ -242,3 +242,54 @@ def test_day_pivot_decomposition_equivalent(catalog):
     pivot = e2.run(day_pivot_sums(child, keys, col("ss_sales_price"), days))
     assert frame_sorted_rows(direct) == frame_sorted_rows(pivot)
     assert direct.nrows > 0
+
+
+def test_q78_filter_pushdown_equivalent(catalog):
+    """q78's d_year=2000 pushdown below the aggregate must not change the
+    result: filtering AFTER the (year-keyed) aggregation gives the same
+    rows as the pushed-down plan used by the query."""
+    from spark_rapids_jni_amd.nds.expr import col
+    from spark_rapids_jni_amd.nds.queries import A, F, J, S
+
+    child = J(J(S("store_sales", "ss_sold_date_sk", "ss_item_sk",
+                  "ss_customer_sk", "ss_quantity"),
+                S("date_dim", "d_date_sk", "d_year"),
+                [("ss_sold_date_sk", "d_date_sk")]),
+              S("item", "i_item_sk"), [("ss_item_sk", "i_item_sk")])
+    keys = ["d_year", "ss_item_sk", "ss_customer_sk"]
+    aggs = [("q", "sum", col("ss_quantity"))]
+    e1 = Engine(catalog, device="cpu")
+    post = e1.run(F(A(child, keys, aggs), col("d_year") == 2000))
+    e2 = Engine(catalog, device="cpu")
+    pushed = e2.run(A(F(child, col("d_year") == 2000), keys, aggs))
+    assert frame_sorted_rows(post) == frame_sorted_rows(pushed)
+    assert post.nrows > 0
+
+
+def test_q16_runtime_semifilter_equivalent(catalog):
+    """q16's runtime semi-filter before the countd aggregate must not
+    change which orders qualify as multi-warehouse."""
+    from spark_rapids_jni_amd.nds.expr import col
+    from spark_rapids_jni_amd.nds.queries import A, F, J, P, S, dsk
+
+    lo, hi = dsk("2002-02-01"), dsk("2002-04-02")
+    cs1 = F(S("catalog_sales", "cs_ship_date_sk", "cs_order_number"),
+            col("cs_ship_date_sk").between(lo, hi))
+    base = S("catalog_sales", "cs_order_number", "cs_warehouse_sk")
+    agg = [("nwh", "countd", col("cs_warehouse_sk"))]
+
+    def qualifying(plan_child, eng):
+        f = eng.run(P(F(A(plan_child, ["cs_order_number"], agg),
+                        col("nwh") > 1), ("ono", col("cs_order_number"))))
+        return sorted(r[0] for r in f.to_rows())
+
+    e1 = Engine(catalog, device="cpu")
+    full = qualifying(base, e1)
+    e2 = Engine(catalog, device="cpu")
+    filtered = qualifying(
+        J(base, P(cs1, ("fono", col("cs_order_number"))),
+          [("cs_order_number", "fono")], how="semi"), e2)
+    e3 = Engine(catalog, device="cpu")
+    cs1_orders = {r[0] for r in
+                  e3.run(P(cs1, ("o", col("cs_order_number")))).to_rows()}
+    assert filtered == [o for o in full if o in cs1_orders]
