@@ -190,11 +190,8 @@ __device__ bool j_match(const char* p, const char* e, MatchCtx& ctx,
       if (p >= e || *p != ':') return false;
       ++p;
       p = j_skip_ws(p, e);
-      bool match = klen == ins.key_len;
-      if (match) {
-        for (int32_t i = 0; i < klen; ++i)
-          if (kbody[i] != ctx.keychars[ins.key_off + i]) { match = false; break; }
-      }
+      bool match = j_key_equals(kbody, klen,
+                                 ctx.keychars + ins.key_off, ins.key_len);
       if (match) {
         if (!j_match(p, e, ctx, step + 1, depth + 1)) return false;
       }
@@ -298,6 +295,70 @@ __device__ inline bool j_hex4(const char* p, const char* e, uint32_t* out) {
   }
   *out = v;
   return true;
+}
+
+
+// compare an escaped key body (between quotes) against a plain UTF-8 path
+// key (reference unescapes names before matching —
+// GetJsonObjectTest_NamesWithEscapedCharacters)
+__device__ inline bool j_key_equals(const char* kbody, int32_t klen,
+                                    const char* key, int32_t key_len) {
+  int32_t i = 0, j = 0;
+  while (i < klen) {
+    char outbuf[4];
+    int nout = 0;
+    char c = kbody[i];
+    if (c == '\\' && i + 1 < klen) {
+      char x = kbody[i + 1];
+      i += 2;
+      switch (x) {
+        case 'n': outbuf[nout++] = '\n'; break;
+        case 't': outbuf[nout++] = '\t'; break;
+        case 'r': outbuf[nout++] = '\r'; break;
+        case 'b': outbuf[nout++] = '\b'; break;
+        case 'f': outbuf[nout++] = '\f'; break;
+        case 'u': {
+          uint32_t cp;
+          if (!j_hex4(kbody + i, kbody + klen, &cp)) return false;
+          i += 4;
+          if (cp >= 0xD800 && cp < 0xDC00 && i + 6 <= klen &&
+              kbody[i] == '\\' && kbody[i + 1] == 'u') {
+            uint32_t lo;
+            if (j_hex4(kbody + i + 2, kbody + klen, &lo) && lo >= 0xDC00 &&
+                lo < 0xE000) {
+              cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
+              i += 6;
+            }
+          }
+          if (cp < 0x80) {
+            outbuf[nout++] = (char)cp;
+          } else if (cp < 0x800) {
+            outbuf[nout++] = (char)(0xC0 | (cp >> 6));
+            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
+          } else if (cp < 0x10000) {
+            outbuf[nout++] = (char)(0xE0 | (cp >> 12));
+            outbuf[nout++] = (char)(0x80 | ((cp >> 6) & 0x3F));
+            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
+          } else {
+            outbuf[nout++] = (char)(0xF0 | (cp >> 18));
+            outbuf[nout++] = (char)(0x80 | ((cp >> 12) & 0x3F));
+            outbuf[nout++] = (char)(0x80 | ((cp >> 6) & 0x3F));
+            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
+          }
+          break;
+        }
+        default: outbuf[nout++] = x;
+      }
+    } else {
+      outbuf[nout++] = c;
+      ++i;
+    }
+    for (int k = 0; k < nout; ++k) {
+      if (j >= key_len || key[j] != outbuf[k]) return false;
+      ++j;
+    }
+  }
+  return j == key_len;
 }
 
 // p at the opening quote; emits the normalized string (JSON-escaped when
@@ -632,11 +693,10 @@ __device__ bool j_match_multi(const char* p, const char* e, MultiCtx& ctx,
         const PathInstr ins = ctx.instrs[ctx.path_off[i] + depth];
         if (ins.kind == 2) {
           m2 |= 1u << i;
-        } else if (ins.kind == 0 && ins.key_len == klen) {
-          bool eq = true;
-          for (int32_t k = 0; k < klen; ++k)
-            if (kbody[k] != ctx.keychars[ins.key_off + k]) { eq = false; break; }
-          if (eq) m2 |= 1u << i;
+        } else if (ins.kind == 0 &&
+                   j_key_equals(kbody, klen, ctx.keychars + ins.key_off,
+                                ins.key_len)) {
+          m2 |= 1u << i;
         }
       }
       if (m2) {

@@ -218,3 +218,23 @@ def test_get_json_object_normalization():
     got = get_json_object(col, "$").to_pylist()
     for (doc, exp), g in zip(cases, got):
         assert g == exp, (doc, g, exp)
+
+
+@pytest.mark.gpu
+def test_get_json_object_escaped_keys_and_surrogates():
+    """Reference NamesWithEscapedCharacters + testUCS2Surrogates: \\uXXXX
+    escapes in KEY names match plain path keys; surrogate pairs in values
+    decode to the astral codepoint."""
+    from spark_rapids_jni_amd.ops.json import (get_json_object,
+                                               get_json_object_multiple_paths)
+    col = Column.from_pylist(
+        ["{'data': 'TEST1'}", "{'\\u0064\\u0061t\\u0061': 'TEST2'}"],
+        DType.STRING, "cuda")
+    assert get_json_object(col, "$.data").to_pylist() == ["TEST1", "TEST2"]
+    got = get_json_object_multiple_paths(col, ["$.data"])[0].to_pylist()
+    assert got == ["TEST1", "TEST2"]
+    col2 = Column.from_pylist(
+        ["{'package_name': 'TEST1'}", "{'package_name': '\\uD83E\\uDD66'}"],
+        DType.STRING, "cuda")
+    assert get_json_object(col2, "$.package_name").to_pylist() == \
+        ["TEST1", "\U0001F966"]
