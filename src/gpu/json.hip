@@ -80,6 +80,84 @@ __device__ const char* j_skip_value(const char* p, const char* e) {
   return p;
 }
 
+__device__ inline bool j_hex4(const char* p, const char* e, uint32_t* out) {
+  if (p + 4 > e) return false;
+  uint32_t v = 0;
+  for (int k = 0; k < 4; ++k) {
+    char h = p[k];
+    uint32_t d;
+    if (h >= '0' && h <= '9') d = h - '0';
+    else if ((h | 32) >= 'a' && (h | 32) <= 'f') d = (h | 32) - 'a' + 10;
+    else return false;
+    v = v * 16 + d;
+  }
+  *out = v;
+  return true;
+}
+
+// compare an escaped key body (between quotes) against a plain UTF-8 path
+// key (reference unescapes names before matching —
+// GetJsonObjectTest_NamesWithEscapedCharacters)
+__device__ inline bool j_key_equals(const char* kbody, int32_t klen,
+                                    const char* key, int32_t key_len) {
+  int32_t i = 0, j = 0;
+  while (i < klen) {
+    char outbuf[4];
+    int nout = 0;
+    char c = kbody[i];
+    if (c == '\\' && i + 1 < klen) {
+      char x = kbody[i + 1];
+      i += 2;
+      switch (x) {
+        case 'n': outbuf[nout++] = '\n'; break;
+        case 't': outbuf[nout++] = '\t'; break;
+        case 'r': outbuf[nout++] = '\r'; break;
+        case 'b': outbuf[nout++] = '\b'; break;
+        case 'f': outbuf[nout++] = '\f'; break;
+        case 'u': {
+          uint32_t cp;
+          if (!j_hex4(kbody + i, kbody + klen, &cp)) return false;
+          i += 4;
+          if (cp >= 0xD800 && cp < 0xDC00 && i + 6 <= klen &&
+              kbody[i] == '\\' && kbody[i + 1] == 'u') {
+            uint32_t lo;
+            if (j_hex4(kbody + i + 2, kbody + klen, &lo) && lo >= 0xDC00 &&
+                lo < 0xE000) {
+              cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
+              i += 6;
+            }
+          }
+          if (cp < 0x80) {
+            outbuf[nout++] = (char)cp;
+          } else if (cp < 0x800) {
+            outbuf[nout++] = (char)(0xC0 | (cp >> 6));
+            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
+          } else if (cp < 0x10000) {
+            outbuf[nout++] = (char)(0xE0 | (cp >> 12));
+            outbuf[nout++] = (char)(0x80 | ((cp >> 6) & 0x3F));
+            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
+          } else {
+            outbuf[nout++] = (char)(0xF0 | (cp >> 18));
+            outbuf[nout++] = (char)(0x80 | ((cp >> 12) & 0x3F));
+            outbuf[nout++] = (char)(0x80 | ((cp >> 6) & 0x3F));
+            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
+          }
+          break;
+        }
+        default: outbuf[nout++] = x;
+      }
+    } else {
+      outbuf[nout++] = c;
+      ++i;
+    }
+    for (int k = 0; k < nout; ++k) {
+      if (j >= key_len || key[j] != outbuf[k]) return false;
+      ++j;
+    }
+  }
+  return j == key_len;
+}
+
 // path instruction: kind 0 = key, 1 = index, 2 = wildcard ([*] or .*)
 struct PathInstr {
   int32_t kind;
@@ -282,84 +360,7 @@ __device__ inline void j_emit_char(Sink& sink, char c, bool as_json) {
   }
 }
 
-__device__ inline bool j_hex4(const char* p, const char* e, uint32_t* out) {
-  if (p + 4 > e) return false;
-  uint32_t v = 0;
-  for (int k = 0; k < 4; ++k) {
-    char h = p[k];
-    uint32_t d;
-    if (h >= '0' && h <= '9') d = h - '0';
-    else if ((h | 32) >= 'a' && (h | 32) <= 'f') d = (h | 32) - 'a' + 10;
-    else return false;
-    v = v * 16 + d;
-  }
-  *out = v;
-  return true;
-}
 
-
-// compare an escaped key body (between quotes) against a plain UTF-8 path
-// key (reference unescapes names before matching —
-// GetJsonObjectTest_NamesWithEscapedCharacters)
-__device__ inline bool j_key_equals(const char* kbody, int32_t klen,
-                                    const char* key, int32_t key_len) {
-  int32_t i = 0, j = 0;
-  while (i < klen) {
-    char outbuf[4];
-    int nout = 0;
-    char c = kbody[i];
-    if (c == '\\' && i + 1 < klen) {
-      char x = kbody[i + 1];
-      i += 2;
-      switch (x) {
-        case 'n': outbuf[nout++] = '\n'; break;
-        case 't': outbuf[nout++] = '\t'; break;
-        case 'r': outbuf[nout++] = '\r'; break;
-        case 'b': outbuf[nout++] = '\b'; break;
-        case 'f': outbuf[nout++] = '\f'; break;
-        case 'u': {
-          uint32_t cp;
-          if (!j_hex4(kbody + i, kbody + klen, &cp)) return false;
-          i += 4;
-          if (cp >= 0xD800 && cp < 0xDC00 && i + 6 <= klen &&
-              kbody[i] == '\\' && kbody[i + 1] == 'u') {
-            uint32_t lo;
-            if (j_hex4(kbody + i + 2, kbody + klen, &lo) && lo >= 0xDC00 &&
-                lo < 0xE000) {
-              cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
-              i += 6;
-            }
-          }
-          if (cp < 0x80) {
-            outbuf[nout++] = (char)cp;
-          } else if (cp < 0x800) {
-            outbuf[nout++] = (char)(0xC0 | (cp >> 6));
-            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
-          } else if (cp < 0x10000) {
-            outbuf[nout++] = (char)(0xE0 | (cp >> 12));
-            outbuf[nout++] = (char)(0x80 | ((cp >> 6) & 0x3F));
-            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
-          } else {
-            outbuf[nout++] = (char)(0xF0 | (cp >> 18));
-            outbuf[nout++] = (char)(0x80 | ((cp >> 12) & 0x3F));
-            outbuf[nout++] = (char)(0x80 | ((cp >> 6) & 0x3F));
-            outbuf[nout++] = (char)(0x80 | (cp & 0x3F));
-          }
-          break;
-        }
-        default: outbuf[nout++] = x;
-      }
-    } else {
-      outbuf[nout++] = c;
-      ++i;
-    }
-    for (int k = 0; k < nout; ++k) {
-      if (j >= key_len || key[j] != outbuf[k]) return false;
-      ++j;
-    }
-  }
-  return j == key_len;
-}
 
 // p at the opening quote; emits the normalized string (JSON-escaped when
 // as_json, raw bytes for a top-level string result); advances p past the
