@@ -203,3 +203,30 @@ def test_java_bigint_bytes_oracle():
     assert jb(-256) == b"\xff\x00"
     assert jb(2**127 - 1) == b"\x7f" + b"\xff" * 15
     assert jb(-2**127) == b"\x80" + b"\x00" * 15
+
+
+def test_murmur3_std_oracle_vectors():
+    """STANDARD MurmurHash3_x86_32 oracle (iceberg bucket transform) vs the
+    published spec/iceberg vectors — distinct from Spark's tail-as-int
+    variant for lengths not divisible by 4."""
+    import struct as st
+    from spark_rapids_jni_amd.utils.sparkref import (murmur3_bytes,
+                                                     murmur3_bytes_std,
+                                                     _to_signed32)
+    vecs = [  # iceberg spec Appendix B
+        (st.pack("<q", 34), 2017239379),            # int/long 34
+        (b"\x05\x8c", -500754589),                  # decimal 14.20 bytes
+        (st.pack("<q", 17486), -653330422),         # date 2017-11-16
+        (st.pack("<q", 1510871468000000), -2047944441),  # ts micros
+        (b"iceberg", 1210000089),                   # string
+        (b"\x00\x01\x02\x03", -188683207),          # binary/fixed
+    ]
+    for data, exp in vecs:
+        assert _to_signed32(murmur3_bytes_std(data, 0)) == exp, data
+    # the Spark variant diverges exactly when len % 4 != 0
+    for data, exp in vecs:
+        spark = _to_signed32(murmur3_bytes(data, 0))
+        if len(data) % 4 == 0:
+            assert spark == exp
+        else:
+            assert spark != exp
