@@ -293,3 +293,23 @@ def test_q16_runtime_semifilter_equivalent(catalog):
     cs1_orders = {r[0] for r in
                   e3.run(P(cs1, ("o", col("cs_order_number")))).to_rows()}
     assert filtered == [o for o in full if o in cs1_orders]
+
+
+def test_all_true_memo_and_drop():
+    """_precheck_all_true resolves memos in one batch; _val_to_column drops
+    all-true masks (keeps real ones) and caches the decision per Val."""
+    from spark_rapids_jni_amd.nds.expr import Val
+    from spark_rapids_jni_amd.nds.plan import (_precheck_all_true,
+                                               _val_to_column)
+    a = Val(torch.arange(10), torch.ones(10, dtype=torch.bool))
+    b = Val(torch.arange(10), torch.tensor([True] * 9 + [False]))
+    c = Val(torch.arange(10))  # no mask
+    _precheck_all_true([a, b, c, None])
+    assert a._all_true is True and b._all_true is False
+    ca = _val_to_column(a)
+    cb = _val_to_column(b)
+    cc = _val_to_column(c)
+    assert ca.validity is None          # all-true mask dropped
+    assert cb.validity is not None      # real nulls preserved
+    assert not cb.is_valid_host(9) and cb.is_valid_host(0)
+    assert cc.validity is None
