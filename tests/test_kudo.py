@@ -250,3 +250,41 @@ def test_zero_row_nullable_column():
     c = Column.from_pylist(vals, DType.INT32)
     got = _roundtrip([c], [(2, 0), (0, 4), (4, 0)])
     assert got[0] == _expected([c], [(2, 0), (0, 4), (4, 0)])[0]
+
+
+def test_list_of_struct_deep_roundtrip():
+    """LIST<STRUCT<int, LIST<string>>> host kudo round-trip — the nesting
+    shape the parquet reader now produces (nulls at every level)."""
+    import torch
+    n = 24
+    rng = random.Random(17)
+    inner_vals, inner_offs = [], [0]
+    n_entries = 40
+    for i in range(n_entries):
+        ln = rng.randint(0, 3)
+        inner_vals.extend(f"s{i}_{k}" if (i + k) % 5 else None
+                          for k in range(ln))
+        inner_offs.append(inner_offs[-1] + ln)
+    inner_child = Column.from_pylist(inner_vals, DType.STRING)
+    inner = Column(DType.LIST, n_entries, None, None,
+                   torch.tensor(inner_offs, dtype=torch.int32),
+                   [inner_child])
+    ints = Column.from_pylist([None if i % 7 == 3 else i * 11
+                               for i in range(n_entries)], DType.INT32)
+    entries = Column(DType.STRUCT, n_entries, None, None, None,
+                     [ints, inner])
+    outer_offs = [0]
+    for i in range(n):
+        outer_offs.append(min(outer_offs[-1] + rng.randint(0, 3), n_entries))
+    col = Column(DType.LIST, n, None, None,
+                 torch.tensor(outer_offs, dtype=torch.int32), [entries])
+    expected = col.to_pylist()
+    slices = [(0, 7), (9, 12)]
+    bufs = []
+    for off, cnt in slices:
+        out = io.BytesIO()
+        kudo.write_partition([col], off, cnt, out)
+        bufs.append(out.getvalue())
+    merged = kudo.merge_on_host(bufs, [col])[0]
+    exp = [v for off, cnt in slices for v in expected[off:off + cnt]]
+    assert merged.to_pylist() == exp
