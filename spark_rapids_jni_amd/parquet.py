@@ -110,12 +110,28 @@ class RowGroupMeta:
 
 @dataclass
 class ParquetFooter:
-    """reference ParquetFooter.java — parsed footer with prune support."""
+    """reference ParquetFooter.java — parsed footer with prune support.
+
+    row_index_offsets mirrors NativeParquetJni.cpp:628-634: the cumulative
+    starting row index of each (surviving) row group, computed over ALL
+    original row groups so it stays correct after byte-range filtering —
+    Spark uses it to reconstruct global row indices per split."""
     version: int
     num_rows: int
     schema: List[SchemaField]
     row_groups: List[RowGroupMeta]
     created_by: Optional[str] = None
+    row_index_offsets: Optional[List[int]] = None
+
+    def _offsets(self):
+        if self.row_index_offsets is not None:
+            return self.row_index_offsets
+        out = []
+        cum = 0
+        for rg in self.row_groups:
+            out.append(cum)
+            cum += rg.num_rows
+        return out
 
     def prune(self, keep_columns: Sequence[str]) -> "ParquetFooter":
         keep = {c.lower() for c in keep_columns}
@@ -125,22 +141,30 @@ class ParquetFooter:
             cols = [c for c in rg.columns if c.path[0].lower() in keep]
             rgs.append(RowGroupMeta(cols, rg.num_rows, rg.total_byte_size))
         return ParquetFooter(self.version, self.num_rows, fields, rgs,
-                             self.created_by)
+                             self.created_by, self._offsets())
 
     def filter_row_groups(self, part_offset: int, part_length: int
                           ) -> "ParquetFooter":
         """Keep row groups whose midpoint falls in [offset, offset+length)
-        (Spark split semantics, reference NativeParquetJni.cpp:795-810)."""
+        (Spark split semantics, reference NativeParquetJni.cpp
+        filter_groups): midpoint = first chunk's start offset + the row
+        group's COMPRESSED size / 2 (parquet-mr's rule — using the
+        uncompressed total_byte_size would assign different groups to a
+        split than Spark does)."""
         rgs = []
-        for rg in self.row_groups:
+        offsets = []
+        base = self._offsets()
+        for i, rg in enumerate(self.row_groups):
             if not rg.columns:
                 continue
-            start = min(c.start_offset for c in rg.columns)
-            mid = start + rg.total_byte_size // 2
+            start = rg.columns[0].start_offset
+            total = sum(c.total_compressed_size for c in rg.columns)
+            mid = start + total // 2
             if part_offset <= mid < part_offset + part_length:
                 rgs.append(rg)
+                offsets.append(base[i])
         return ParquetFooter(self.version, sum(r.num_rows for r in rgs),
-                             self.schema, rgs, self.created_by)
+                             self.schema, rgs, self.created_by, offsets)
 
 
 def read_footer(path_or_bytes) -> ParquetFooter:

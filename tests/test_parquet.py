@@ -1058,3 +1058,39 @@ def test_decode_random_nested(tmp_path):
         assert col.to_pylist() == [conv(typ, r) for r in rows], (trial, typ)
         parsed += 1
     assert parsed >= 10, (parsed, skipped)
+
+
+def test_filter_row_groups_reference_vectors():
+    """Reference ParquetFooterTest row-index-offset vectors: midpoint =
+    first chunk offset + COMPRESSED size / 2; cumulative row offsets are
+    computed over all original groups, so they survive filtering."""
+    def rg(rows, dpo, csize):
+        col = srj_pq.ColumnChunkMeta(
+            path=("c",), physical_type=2, encodings=[0], codec=0,
+            num_values=rows, total_compressed_size=csize,
+            total_uncompressed_size=csize * 3,  # NOT what midpoints use
+            data_page_offset=dpo, dictionary_page_offset=None)
+        return srj_pq.RowGroupMeta([col], rows, csize * 3)
+
+    f = srj_pq.ParquetFooter(1, 3500, [], [rg(1000, 100, 200),
+                                           rg(2000, 400, 200),
+                                           rg(500, 700, 200)])
+    # midpoints: 200, 500, 800; cumulative row offsets: 0, 1000, 3000
+    def sel(off, length):
+        g = f.filter_row_groups(off, length)
+        return ([r.num_rows for r in g.row_groups], g.row_index_offsets,
+                g.num_rows)
+
+    assert sel(0, 10**9) == ([1000, 2000, 500], [0, 1000, 3000], 3500)
+    assert sel(0, 300) == ([1000], [0], 1000)
+    assert sel(300, 300) == ([2000], [1000], 2000)
+    assert sel(600, 300) == ([500], [3000], 500)
+    assert sel(0, 600) == ([1000, 2000], [0, 1000], 3000)
+    assert sel(10000, 100) == ([], [], 0)
+    # single-group + byte-range-filtered-away
+    f1 = srj_pq.ParquetFooter(1, 1000, [], [rg(1000, 100, 200)])
+    assert f1.filter_row_groups(0, 300).row_index_offsets == [0]
+    assert f1.filter_row_groups(300, 100).row_index_offsets == []
+    # prune preserves the offsets
+    g = f.filter_row_groups(300, 10**9)
+    assert g.prune(["c"]).row_index_offsets == [1000, 3000]
