@@ -288,3 +288,35 @@ def test_list_of_struct_deep_roundtrip():
     merged = kudo.merge_on_host(bufs, [col])[0]
     exp = [v for off, cnt in slices for v in expected[off:off + cnt]]
     assert merged.to_pylist() == exp
+
+
+def test_write_simple_reference_byte_counts():
+    """Reference KudoSerializerTest.testWriteSimple: the EXACT byte counts
+    and header fields for the 7-column (a,b,c,c1,d,d1,d2) 4-row table —
+    172 bytes total, validity 7, offsets 40, body 143, validity bitset
+    10111110."""
+    import torch
+    from spark_rapids_jni_amd.columnar import validity_from_bools
+    vb = validity_from_bools([True, True, False, True], "cpu")
+    a = Column.from_pylist([1, 2, 3, 4], DType.INT32)
+    b = Column.from_pylist(["1", "12", None, "45"], DType.STRING)
+    c1 = Column.from_pylist([1, None, 3, 4, 5, 6, 7, 8, 9], DType.INT32)
+    c = Column(DType.LIST, 4, None, vb,
+               torch.tensor([0, 3, 6, 6, 9], dtype=torch.int32), [c1])
+    d1 = Column.from_pylist([1, 2, None, 3], DType.INT8)
+    d2 = Column.from_pylist([11, None, None, 33], DType.INT64)
+    d = Column(DType.STRUCT, 4, None, vb.clone(), None, [d1, d2])
+    out = io.BytesIO()
+    kudo.write_partition([a, b, c, d], 0, 4, out)
+    buf = out.getvalue()
+    assert len(buf) == 172
+    hdr = kudo.KudoTableHeader.read(io.BytesIO(buf))
+    assert hdr.num_columns == 7
+    assert hdr.offset == 0
+    assert hdr.num_rows == 4
+    assert hdr.validity_len == 7
+    assert hdr.offset_len == 40
+    assert hdr.total_len == 143
+    assert not hdr.has_validity_buffer(0)
+    for i in range(1, 7):
+        assert hdr.has_validity_buffer(i), i
