@@ -89,6 +89,15 @@ def _pad4(n: int) -> int:
 # write
 # ---------------------------------------------------------------------------
 
+def write_row_count(out: BinaryIO, num_rows: int) -> int:
+    """Header-only batch carrying just a row count (reference
+    KudoSerializer.writeRowCountToStream, 28 bytes): Spark emits these for
+    zero-column tables, e.g. a COUNT(*) over an empty projection."""
+    h = KudoTableHeader(0, num_rows, 0, 0, 0, 0, b"")
+    h.write(out)
+    return h.header_len()
+
+
 def write_partition(columns: Sequence[Column], row_offset: int, num_rows: int,
                     out: BinaryIO) -> int:
     """Serialize rows [row_offset, row_offset+num_rows) of a host table.

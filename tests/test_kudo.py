@@ -320,3 +320,13 @@ def test_write_simple_reference_byte_counts():
     assert not hdr.has_validity_buffer(0)
     for i in range(1, 7):
         assert hdr.has_validity_buffer(i), i
+
+
+def test_row_count_only_reference_bytes():
+    """Reference testRowCountOnly: a 28-byte header-only batch."""
+    out = io.BytesIO()
+    n = kudo.write_row_count(out, 5)
+    assert n == 28 and len(out.getvalue()) == 28
+    hdr = kudo.KudoTableHeader.read(io.BytesIO(out.getvalue()))
+    assert (hdr.num_columns, hdr.offset, hdr.num_rows, hdr.validity_len,
+            hdr.offset_len, hdr.total_len) == (0, 0, 5, 0, 0, 0)
