@@ -330,3 +330,14 @@ def test_row_count_only_reference_bytes():
     hdr = kudo.KudoTableHeader.read(io.BytesIO(out.getvalue()))
     assert (hdr.num_columns, hdr.offset, hdr.num_rows, hdr.validity_len,
             hdr.offset_len, hdr.total_len) == (0, 0, 5, 0, 0, 0)
+
+
+def test_serialize_validity_unaligned_tail_slice():
+    """Reference testSerializeValidity: a 3-row slice at offset 509 of a
+    512-row nullable column — validity bits cross byte boundaries at an
+    unaligned offset and must merge back exactly."""
+    col = Column.from_pylist([None, None] + list(range(2, 512)), DType.INT32)
+    out = io.BytesIO()
+    kudo.write_partition([col], 509, 3, out)
+    merged = kudo.merge_on_host([out.getvalue()], [col])[0]
+    assert merged.to_pylist() == [509, 510, 511]
