@@ -50,6 +50,20 @@ def decode(col: Column, schema) -> Table:
     def _is_msg(k):
         return isinstance(k, tuple) and k[0] == "message"
 
+    # reference ProtobufSchemaDescriptor rejects duplicate field numbers
+    # under the same parent message (they would double-decode); different
+    # parents may reuse numbers — validated recursively BEFORE any kernel
+    # work so invalid schemas fail fast on any host
+    def _check(sch):
+        fnums = [f for f, _k in sch]
+        if len(set(fnums)) != len(fnums):
+            raise ValueError(
+                "duplicate protobuf field numbers in one message")
+        for _f, k in sch:
+            if _is_msg(k):
+                _check(k[1])
+    _check(schema)
+
     if any(_is_msg(k) for _, k in schema):
         flat = [(f, "bytes" if _is_msg(k) else k) for f, k in schema]
         tbl = decode(col, flat)

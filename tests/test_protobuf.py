@@ -224,3 +224,18 @@ def test_protobuf_repeated_all_types():
             assert len(got[4][i]) == len(flts), i
         if strs:
             assert got[5][i] == strs, i
+
+
+def test_duplicate_field_numbers_rejected():
+    """Reference ProtobufSchemaDescriptorTest: duplicate field numbers under
+    ONE parent are invalid; the same number under different parents is
+    fine (validated per message level, no GPU needed to reject)."""
+    import pytest as _pt
+    from spark_rapids_jni_amd.ops import protobuf as pb
+    from spark_rapids_jni_amd.columnar import Column, DType
+    col = Column.from_pylist([b""], DType.STRING)
+    with _pt.raises(ValueError, match="duplicate"):
+        pb.decode(col, [(1, "int64"), (1, "string")])
+    with _pt.raises(ValueError, match="duplicate"):
+        pb.decode(col, [(1, "int64"),
+                        (2, ("message", [(3, "int64"), (3, "bool")]))])
