@@ -290,3 +290,42 @@ def test_with_retry_split_path(rmm):
     got = with_retry(alloc, split=lambda: halves.append(1), thread_id=tid)
     assert got == 1 and len(halves) == 1
     rmm.task_done(11)
+
+
+def test_shuffle_thread_survives_deadlock(rmm):
+    """Reference testShuffleBlocking semantics: shuffle pool threads get the
+    HIGHEST priority (priority -1), so in a deadlock the dedicated task
+    thread rolls back, never the shuffle thread."""
+    _adaptor(rmm, pool=100)
+    t_task, t_shuf = 31, 32
+    rmm.start_dedicated_task_thread(t_task, 7)
+    rmm.adaptor().pool_thread_working_on_tasks(t_shuf, [7], True)
+    rmm.alloc(50, thread_id=t_shuf)
+    rmm.alloc(40, thread_id=t_task)
+    errors = {}
+
+    def try_alloc(tid, amount):
+        try:
+            rmm.alloc(amount, thread_id=tid)
+            errors[tid] = None
+        except Exception as e:
+            errors[tid] = e
+
+    th_s = threading.Thread(target=try_alloc, args=(t_shuf, 40))
+    th_s.start()
+    deadline = time.time() + 5
+    while rmm.get_state_of(t_shuf) != "THREAD_BLOCKED" and \
+            time.time() < deadline:
+        time.sleep(0.01)
+    th_t = threading.Thread(target=try_alloc, args=(t_task, 40))
+    th_t.start()
+    th_t.join(timeout=5)
+    # the DEDICATED thread (priority >= 0) is rolled back; shuffle survives
+    assert isinstance(errors.get(t_task), GpuRetryOOM)
+    rmm.dealloc(40, thread_id=t_task)
+    th_s.join(timeout=5)
+    assert errors.get(t_shuf, "unset") is None
+    rmm.dealloc(40, thread_id=t_shuf)
+    rmm.dealloc(50, thread_id=t_shuf)
+    rmm.adaptor().pool_thread_finished_for_tasks(t_shuf, [7])
+    rmm.task_done(7)
