@@ -163,7 +163,8 @@ class GpuTimeZoneDB:
 class DstRule:
     month: int
     week: int        # 1..5, 5 = last
-    day_of_week: int  # 0=Sunday (ORC convention)
+    day_of_week: int  # 0=Sunday (ORC convention); week == -1 means LAST
+                      # occurrence in the month (DOW >= monthLength-6)
     seconds_of_day: int
     offset_after: int
 
@@ -196,7 +197,14 @@ def extract_dst_rules(name: str, probe_year: int = 2060) -> List[DstRule]:
                         hi = mid
                 local = hi + prev
                 dt = _EPOCH + datetime.timedelta(seconds=local)
-                week = (dt.day - 1) // 7 + 1
+                # last-occurrence rules ("last Sunday of October") must not
+                # encode an ordinal week — it differs between years with 4
+                # vs 5 such weekdays. The reference encodes them as
+                # DOW >= monthLength-6 (OrcDstRuleExtractor.decodeTransition
+                # isLastOccurrence); week == -1 carries that here.
+                import calendar
+                mlen = calendar.monthrange(dt.year, dt.month)[1]
+                week = -1 if dt.day + 7 > mlen else (dt.day - 1) // 7 + 1
                 dow = (dt.weekday() + 1) % 7  # ORC: 0 = Sunday
                 rules.append(DstRule(dt.month, week, dow,
                                      dt.hour * 3600 + dt.minute * 60 + dt.second,

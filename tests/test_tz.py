@@ -132,3 +132,20 @@ def test_ambiguous_local_time_resolution_matches_java():
         tzinfo=datetime.timezone.utc).timestamp()) * 1_000_000
     expect = amb + 4 * 3600 * 1_000_000  # EDT: utc = local + 4h
     assert to_utc(amb) == expect
+
+
+def test_extract_dst_rules_hemispheres_and_no_dst():
+    """Reference OrcDstRuleExtractor tests: northern (Europe/London) and
+    southern (Australia/Sydney) hemisphere rules, and zones without DST
+    return no rules; rules stay stable at the reference's probe years."""
+    from spark_rapids_jni_amd.tz import extract_dst_rules, validate_rules_stable
+    lon = extract_dst_rules("Europe/London")
+    assert len(lon) == 2 and sorted(r.month for r in lon) == [3, 10]
+    # both London transitions are LAST-Sunday rules -> week -1
+    assert all(r.week == -1 and r.day_of_week == 0 for r in lon)
+    syd = extract_dst_rules("Australia/Sydney")
+    assert len(syd) == 2 and sorted(r.month for r in syd) == [4, 10]
+    assert extract_dst_rules("Asia/Shanghai") == []
+    assert extract_dst_rules("UTC") == []
+    for z in ("Europe/London", "Australia/Sydney"):
+        assert validate_rules_stable(z)
