@@ -18,6 +18,10 @@ class SeekableInputStream(ABC):
     def tell(self) -> int: ...
 
     def read_fully(self, pos: int, n: int) -> bytes:
+        """Exactly n bytes at pos: EOFError on a short read, ValueError on
+        negative arguments (reference RapidsInputFileTest semantics)."""
+        if pos < 0 or n < 0:
+            raise ValueError(f"negative read: pos={pos} n={n}")
         self.seek(pos)
         out = self.read(n)
         if len(out) != n:
@@ -31,6 +35,15 @@ class InputFile(ABC):
 
     @abstractmethod
     def open(self) -> SeekableInputStream: ...
+
+    def read_tail(self, n: int) -> bytes:
+        """Last n bytes of the file (reference RapidsInputFile.readTail):
+        EOFError when n exceeds the file length — footer readers rely on
+        this to fail cleanly on truncated files."""
+        ln = self.length()
+        if n > ln:
+            raise EOFError(f"readTail({n}) > file length {ln}")
+        return self.open().read_fully(ln - n, n)
 
 
 class OutputFile(ABC):

@@ -193,6 +193,17 @@ def test_fileio_local(tmp_path):
     assert inf.length() == 10
     s = inf.open()
     assert s.read_fully(3, 4) == b"3456"
+    # reference RapidsInputFileTest: tail reads + clean failures
+    assert inf.read_tail(4) == b"6789"
+    assert inf.read_tail(10) == b"0123456789"
+    with pytest.raises(EOFError):
+        inf.read_tail(11)
+    with pytest.raises(EOFError):
+        s.read_fully(8, 4)
+    with pytest.raises(ValueError):
+        s.read_fully(-1, 2)
+    with pytest.raises(ValueError):
+        s.read_fully(0, -2)
 
 
 def test_version_gates():
